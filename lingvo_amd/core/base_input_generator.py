@@ -1,0 +1,83 @@
+"""Input generators (reference lingvo/core/base_input_generator.py).
+
+BaseInputGenerator yields NestedMap batches. On a GPU box, ToDevice()
+moves batches host->HBM with pinned memory on a dedicated HIP copy stream
+(the MI355X replacement for TPU infeed, reference
+base_input_generator.py:446-671); on CPU it is a no-op.
+"""
+
+from __future__ import annotations
+
+from typing import Iterator, List, Optional
+
+import torch
+
+from lingvo_amd.core.base_layer import BaseLayer
+from lingvo_amd.core.nested_map import NestedMap
+
+
+class BaseInputGenerator(BaseLayer):
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('batch_size', 16, 'Per-device batch size.')
+    p.Define('num_samples', 0, 'Dataset size (0 = infinite/synthetic).')
+    p.Define('resettable', False, 'Whether the source can be reset for eval.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    self._copy_stream = None
+    self._batch_count = 0
+
+  def InfeedBatchSize(self) -> int:
+    return self.p.batch_size
+
+  def _InputBatch(self) -> NestedMap:
+    """Subclass: produce one CPU batch."""
+    raise NotImplementedError
+
+  def GetPreprocessedInputBatch(self) -> NestedMap:
+    batch = self._InputBatch()
+    self._batch_count += 1
+    return batch
+
+  def Reset(self) -> None:
+    self._batch_count = 0
+
+  def __iter__(self) -> Iterator[NestedMap]:
+    while True:
+      yield self.GetPreprocessedInputBatch()
+
+  # ---- host->device (infeed replacement) --------------------------------
+  def ToDevice(self, batch: NestedMap, device) -> NestedMap:
+    """Moves a batch to the device, overlapping the copy on a side stream."""
+    if str(device).startswith('cuda') and torch.cuda.is_available():
+      if self._copy_stream is None:
+        self._copy_stream = torch.cuda.Stream(device=device)
+      with torch.cuda.stream(self._copy_stream):
+        moved = batch.Transform(
+            lambda t: t.pin_memory().to(device, non_blocking=True)
+            if isinstance(t, torch.Tensor) else t)
+      torch.cuda.current_stream(device).wait_stream(self._copy_stream)
+      return moved
+    return batch.Transform(
+        lambda t: t.to(device) if isinstance(t, torch.Tensor) else t)
+
+
+class BaseSequenceInputGenerator(BaseInputGenerator):
+  """Adds length-bucketing params (reference base_input_generator.py:1457)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('bucket_upper_bound', [], 'Bucket length upper bounds.')
+    p.Define('bucket_batch_limit', [], 'Per-bucket batch sizes.')
+    p.Define('source_max_length', None, 'Max source length.')
+    p.Define('target_max_length', 300, 'Max target length.')
+    p.Define('pad_to_max_seq_length', False, 'Static-shape padding.')
+    return p
+
+  def scaled_bucket_batch_limit(self) -> List[int]:
+    return list(self.p.bucket_batch_limit)

@@ -1,0 +1,354 @@
+"""Foundation utilities: weight init specs, padding ops, deterministic RNG.
+
+Covers the capability surface of the reference's py_utils
+(lingvo/core/py_utils.py): WeightInit/WeightParams (py_utils.py:1085,1250),
+padded-sequence ops (4282-4539), deterministic per-step seeds (3933), and
+numeric checks (208). All tensor math is torch; sequence layout is
+batch-major [B, T, ...] with a float `paddings` tensor [B, T] where 1.0
+marks padded positions (the reference's convention).
+"""
+
+from __future__ import annotations
+
+import contextlib
+import math
+import threading
+from typing import Any, List, Optional, Sequence, Tuple, Union
+
+import torch
+
+from lingvo_amd.core.hyperparams import Params
+from lingvo_amd.core.nested_map import NestedMap
+
+
+# --------------------------------------------------------------------------
+# Weight initialization
+# --------------------------------------------------------------------------
+class WeightInit:
+  """Factory for weight-initialization specs (method + scale)."""
+
+  @staticmethod
+  def _Spec(method: str, scale: float) -> Params:
+    p = Params()
+    p.Define('method', method, 'Initialization method.')
+    p.Define('scale', scale, 'Initialization scale.')
+    return p
+
+  @staticmethod
+  def Gaussian(scale: float = 1.0) -> Params:
+    return WeightInit._Spec('gaussian', scale)
+
+  @staticmethod
+  def Uniform(scale: float = 1.0) -> Params:
+    return WeightInit._Spec('uniform', scale)
+
+  @staticmethod
+  def Constant(scale: float = 0.0) -> Params:
+    return WeightInit._Spec('constant', scale)
+
+  @staticmethod
+  def Xavier(scale: float = 1.0) -> Params:
+    return WeightInit._Spec('xavier', scale)
+
+  @staticmethod
+  def GeoMeanXavier(scale: float = 1.0) -> Params:
+    return WeightInit._Spec('geo_mean_xavier', scale)
+
+  @staticmethod
+  def TruncatedGaussian(scale: float = 1.0) -> Params:
+    return WeightInit._Spec('truncated_gaussian', scale)
+
+  @staticmethod
+  def GaussianSqrtDim(scale: float = 1.0) -> Params:
+    return WeightInit._Spec('gaussian_sqrt_dim', scale)
+
+  @staticmethod
+  def UniformSqrtDim(scale: float = 1.0) -> Params:
+    return WeightInit._Spec('uniform_sqrt_dim', scale)
+
+  @staticmethod
+  def UniformUnitScaling(scale: float = 1.0) -> Params:
+    return WeightInit._Spec('uniform_unit_scaling', scale)
+
+  @staticmethod
+  def TruncatedGaussianSqrtFanIn(scale: float = 1.0) -> Params:
+    return WeightInit._Spec('truncated_gaussian_sqrt_fanin', scale)
+
+  @staticmethod
+  def TruncatedGaussianSqrtFanOut(scale: float = 1.0) -> Params:
+    return WeightInit._Spec('truncated_gaussian_sqrt_fanout', scale)
+
+
+def WeightParams(shape: Sequence[int],
+                 init: Optional[Params] = None,
+                 dtype: torch.dtype = torch.float32,
+                 collections: Optional[List[str]] = None) -> Params:
+  """Describes a trainable weight (reference py_utils.py:1250)."""
+  p = Params()
+  p.Define('shape', list(shape), 'Weight shape.')
+  p.Define('init', init or WeightInit.Xavier(1.0), 'Initialization spec.')
+  p.Define('dtype', dtype, 'Weight dtype.')
+  p.Define('collections', collections or [], 'Weight collections.')
+  return p
+
+
+def _FanInFanOut(shape: Sequence[int]) -> Tuple[int, int]:
+  if not shape:
+    return 1, 1
+  if len(shape) == 1:
+    return shape[0], shape[0]
+  # Conv kernels: [..spatial.., in, out]; matmul: [in, out].
+  receptive = 1
+  for d in shape[:-2]:
+    receptive *= d
+  return shape[-2] * receptive, shape[-1] * receptive
+
+
+def InitWeight(shape: Sequence[int], spec: Params,
+               generator: Optional[torch.Generator] = None,
+               dtype: torch.dtype = torch.float32) -> torch.Tensor:
+  """Materializes an initialized weight tensor from a WeightInit spec."""
+  method, scale = spec.method, spec.scale
+  shape = list(shape)
+  dim0 = shape[0] if shape else 1
+  fan_in, fan_out = _FanInFanOut(shape)
+
+  def randn():
+    return torch.randn(shape, generator=generator, dtype=torch.float32)
+
+  def rand():
+    return torch.rand(shape, generator=generator, dtype=torch.float32) * 2 - 1
+
+  if method == 'constant':
+    w = torch.full(shape, float(scale), dtype=torch.float32)
+  elif method == 'gaussian':
+    w = randn() * scale
+  elif method == 'uniform':
+    w = rand() * scale
+  elif method == 'gaussian_sqrt_dim':
+    w = randn() * (scale / math.sqrt(max(dim0, 1)))
+  elif method == 'uniform_sqrt_dim':
+    w = rand() * (scale / math.sqrt(max(dim0, 1)))
+  elif method == 'xavier':
+    limit = scale * math.sqrt(6.0 / (fan_in + fan_out))
+    w = rand() * limit
+  elif method == 'geo_mean_xavier':
+    limit = scale * math.sqrt(3.0 / math.sqrt(fan_in * fan_out))
+    w = rand() * limit
+  elif method == 'uniform_unit_scaling':
+    w = rand() * (scale * math.sqrt(3.0 / max(fan_in, 1)))
+  elif method in ('truncated_gaussian', 'truncated_gaussian_sqrt_fanin',
+                  'truncated_gaussian_sqrt_fanout'):
+    std = scale
+    if method.endswith('fanin'):
+      std = scale / math.sqrt(max(fan_in, 1))
+    elif method.endswith('fanout'):
+      std = scale / math.sqrt(max(fan_out, 1))
+    w = torch.empty(shape, dtype=torch.float32)
+    torch.nn.init.trunc_normal_(w, std=std, a=-2 * std, b=2 * std,
+                                generator=generator)
+  else:
+    raise ValueError(f'Unknown init method {method!r}')
+  return w.to(dtype)
+
+
+# --------------------------------------------------------------------------
+# Deterministic step-seeded RNG (reference py_utils.py:3933)
+# --------------------------------------------------------------------------
+_RNG_STATE = threading.local()
+
+
+def _GetRngStack() -> List[NestedMap]:
+  if not hasattr(_RNG_STATE, 'stack'):
+    _RNG_STATE.stack = [NestedMap(global_seed=1234, step=0, op_counter=[0])]
+  return _RNG_STATE.stack
+
+
+@contextlib.contextmanager
+def StepSeedScope(global_seed: int, step: int):
+  """Makes GenerateStepSeedPair deterministic in (global_seed, step)."""
+  stack = _GetRngStack()
+  stack.append(NestedMap(global_seed=int(global_seed), step=int(step),
+                         op_counter=[0]))
+  try:
+    yield
+  finally:
+    stack.pop()
+
+
+def GenerateStepSeedPair(op_seed: Optional[int] = None) -> Tuple[int, int]:
+  """Returns a deterministic (seed1, seed2) for the current step scope.
+
+  Each call without op_seed increments a per-scope op counter, so multiple
+  stochastic ops in one step get distinct but reproducible seeds.
+  """
+  top = _GetRngStack()[-1]
+  if op_seed is None:
+    op_seed = top.op_counter[0]
+    top.op_counter[0] += 1
+  mixed = (top.global_seed * 1000003 + op_seed) & 0x7FFFFFFF
+  return mixed, top.step
+
+
+def MakeStepGenerator(device: Union[str, torch.device],
+                      op_seed: Optional[int] = None) -> torch.Generator:
+  s1, s2 = GenerateStepSeedPair(op_seed)
+  g = torch.Generator(device=device)
+  g.manual_seed((s1 * 2654435761 + s2) & 0x7FFFFFFFFFFFFFFF)
+  return g
+
+
+def DeterministicDropout(x: torch.Tensor, keep_prob: float,
+                         op_seed: Optional[int] = None) -> torch.Tensor:
+  """Dropout reproducible under StepSeedScope (reference py_utils.py:3978)."""
+  if keep_prob >= 1.0:
+    return x
+  g = MakeStepGenerator(x.device, op_seed)
+  mask = (torch.rand(x.shape, generator=g, device=x.device,
+                     dtype=torch.float32) < keep_prob)
+  return x * mask.to(x.dtype) / keep_prob
+
+
+# --------------------------------------------------------------------------
+# Padded-sequence utilities ([B, T] paddings, 1.0 == padded)
+# --------------------------------------------------------------------------
+def PaddingsFromLengths(lengths: torch.Tensor, maxlen: int) -> torch.Tensor:
+  """[B] lengths -> [B, maxlen] float paddings."""
+  pos = torch.arange(maxlen, device=lengths.device)[None, :]
+  return (pos >= lengths[:, None]).float()
+
+
+def LengthsFromPaddings(paddings: torch.Tensor) -> torch.Tensor:
+  """[B, T] paddings -> [B] int lengths (reference py_utils.py:4429)."""
+  return (1.0 - paddings).sum(dim=1).round().long()
+
+
+def ApplyPadding(padding: torch.Tensor, x: torch.Tensor,
+                 padded_value: float = 0.0) -> torch.Tensor:
+  """Zeroes (or sets) x where padding==1; padding broadcast against x."""
+  while padding.dim() < x.dim():
+    padding = padding.unsqueeze(-1)
+  padding = padding.to(x.dtype)
+  if padded_value == 0.0:
+    return x * (1.0 - padding)
+  return x * (1.0 - padding) + padded_value * padding
+
+
+def PadSequenceDimension(x: torch.Tensor, length: int, pad_val: float = 0.0,
+                         axis: int = 1) -> torch.Tensor:
+  """Pads axis `axis` of x up to `length` (reference py_utils.py:4282)."""
+  cur = x.shape[axis]
+  if cur == length:
+    return x
+  if cur > length:
+    raise ValueError(f'Cannot pad dim {axis} from {cur} down to {length}')
+  pad_shape = list(x.shape)
+  pad_shape[axis] = length - cur
+  pad = torch.full(pad_shape, pad_val, dtype=x.dtype, device=x.device)
+  return torch.cat([x, pad], dim=axis)
+
+
+def ConcatenatePaddedSequences(x: torch.Tensor, y: torch.Tensor,
+                               px: torch.Tensor, py: torch.Tensor
+                               ) -> Tuple[torch.Tensor, torch.Tensor]:
+  """Concats per-example valid segments of two padded [B,T,...] batches."""
+  b = x.shape[0]
+  lx = LengthsFromPaddings(px)
+  ly = LengthsFromPaddings(py)
+  tot = int((lx + ly).max().item())
+  feat = x.shape[2:]
+  out = torch.zeros((b, tot) + tuple(feat), dtype=x.dtype, device=x.device)
+  pout = torch.ones((b, tot), dtype=px.dtype, device=px.device)
+  for i in range(b):
+    n1, n2 = int(lx[i]), int(ly[i])
+    out[i, :n1] = x[i, :n1]
+    out[i, n1:n1 + n2] = y[i, :n2]
+    pout[i, :n1 + n2] = 0.0
+  return out, pout
+
+
+# --------------------------------------------------------------------------
+# Numeric checks & misc
+# --------------------------------------------------------------------------
+def CheckNumerics(x: torch.Tensor, message: str = '') -> torch.Tensor:
+  """Raises if x contains NaN/Inf (reference py_utils.py:208)."""
+  if not torch.isfinite(x).all():
+    raise FloatingPointError(f'Tensor has NaN/Inf: {message}')
+  return x
+
+
+def HasNanOrInf(nmap_or_tensor) -> bool:
+  if isinstance(nmap_or_tensor, torch.Tensor):
+    return not bool(torch.isfinite(nmap_or_tensor).all())
+  tensors = [v for v in nmap_or_tensor.Flatten()
+             if isinstance(v, torch.Tensor) and v.is_floating_point()]
+  return any(not bool(torch.isfinite(t).all()) for t in tensors)
+
+
+def GlobalGradNorm(grads: List[torch.Tensor]) -> torch.Tensor:
+  """L2 norm over a list of grads (reference learner.py global clip)."""
+  device = grads[0].device if grads else 'cpu'
+  if not grads:
+    return torch.zeros((), device=device)
+  norms = torch.stack([g.detach().float().norm(2) for g in grads])
+  return norms.norm(2)
+
+
+def WeightedAvg(values: torch.Tensor, weights: torch.Tensor
+                ) -> Tuple[torch.Tensor, torch.Tensor]:
+  w = weights.float()
+  total = w.sum()
+  avg = (values.float() * w).sum() / total.clamp_min(1e-8)
+  return avg, total
+
+
+def WeightedAvgOfMetrics(metrics_list: List[NestedMap]) -> NestedMap:
+  """Averages a list of {name: (value, weight)} metric maps
+  (reference base_model.py:648)."""
+  ret = NestedMap()
+  if not metrics_list:
+    return ret
+  for name in metrics_list[0].keys():
+    vals = torch.stack([torch.as_tensor(m[name][0]).float()
+                        for m in metrics_list])
+    wts = torch.stack([torch.as_tensor(m[name][1]).float()
+                       for m in metrics_list])
+    total_w = wts.sum()
+    avg = (vals * wts).sum() / total_w.clamp_min(1e-8)
+    ret[name] = (avg, total_w)
+  return ret
+
+
+def ToScalar(x) -> float:
+  if isinstance(x, torch.Tensor):
+    return float(x.detach().cpu().item())
+  return float(x)
+
+
+class Timer:
+  """Accumulating wall-clock timer (reference py_utils.py:6890)."""
+
+  def __init__(self):
+    self._total = 0.0
+    self._start = None
+
+  def Start(self):
+    import time
+    self._start = time.perf_counter()
+    return self
+
+  def Stop(self):
+    import time
+    if self._start is not None:
+      self._total += time.perf_counter() - self._start
+      self._start = None
+    return self
+
+  def Duration(self) -> float:
+    return self._total
+
+  def __enter__(self):
+    return self.Start()
+
+  def __exit__(self, *a):
+    self.Stop()

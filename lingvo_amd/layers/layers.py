@@ -1,0 +1,411 @@
+"""Core layer library, tranche 1 (reference lingvo/core/layers.py).
+
+ProjectionLayer (reference layers.py:845), FCLayer (:1586), FeedForwardNet
+(:1597), Conv2DLayer (:182-771), PoolingLayer (:2285), EmbeddingLayer
+(:2679), positional embeddings (:3380-3476), SimpleFullSoftmax (:3697),
+DropoutLayer (:4842), LayerNorm (:4927), label smoothing (:5305).
+GEMMs run on hipBLASLt via torch.matmul; LayerNorm / softmax-xent /
+embedding dispatch to hand-written gfx950 HIP kernels on ROCm devices
+(lingvo_amd/ops), plain torch on CPU.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import List, Optional, Sequence
+
+import torch
+import torch.nn.functional as F
+
+from lingvo_amd.core import py_utils
+from lingvo_amd.core.base_layer import BaseLayer
+from lingvo_amd.core.hyperparams import Params
+from lingvo_amd.core.nested_map import NestedMap
+from lingvo_amd.layers import activations
+
+
+class ProjectionLayer(BaseLayer):
+  """Linear projection + optional bias + activation."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('input_dim', 0, 'Input dimension.')
+    p.Define('output_dim', 0, 'Output dimension.')
+    p.Define('has_bias', False, 'Add bias.')
+    p.Define('activation', 'NONE', 'Activation name.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    assert p.input_dim > 0 and p.output_dim > 0
+    self.CreateVariable('w', py_utils.WeightParams(
+        [p.input_dim, p.output_dim], p.params_init, p.dtype))
+    if p.has_bias:
+      self.CreateVariable('b', py_utils.WeightParams(
+          [p.output_dim], py_utils.WeightInit.Constant(0.0), p.dtype))
+
+  def FProp(self, theta: NestedMap, inputs: torch.Tensor,
+            paddings: Optional[torch.Tensor] = None) -> torch.Tensor:
+    p = self.p
+    out = torch.matmul(inputs, theta.w)
+    if p.has_bias:
+      out = out + theta.b
+    out = activations.GetFn(p.activation)(out)
+    if paddings is not None:
+      out = py_utils.ApplyPadding(paddings, out)
+    return out
+
+
+class FCLayer(ProjectionLayer):
+  """Fully-connected = projection with bias + RELU default
+  (reference layers.py:1586)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.has_bias = True
+    p.activation = 'RELU'
+    return p
+
+
+class FeedForwardNet(BaseLayer):
+  """Stack of FC layers (reference layers.py:1597)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('input_dim', 0, 'Input dimension.')
+    p.Define('hidden_layer_dims', [], 'Output dim of each layer.')
+    p.Define('activation', 'RELU', 'One name, or list per layer.')
+    p.Define('has_bias', True, 'Bias on each layer.')
+    p.Define('dropout_prob', 0.0, 'Dropout after each hidden layer.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    dims = [p.input_dim] + list(p.hidden_layer_dims)
+    acts = p.activation
+    if isinstance(acts, str):
+      acts = [acts] * (len(dims) - 1)
+    layers_p = []
+    for i in range(len(dims) - 1):
+      layers_p.append(ProjectionLayer.Params().Set(
+          input_dim=dims[i], output_dim=dims[i + 1],
+          has_bias=p.has_bias, activation=acts[i]))
+    self.CreateChildren('fc', layers_p)
+    self._dropout = p.dropout_prob
+
+  def FProp(self, theta: NestedMap, inputs: torch.Tensor) -> torch.Tensor:
+    out = inputs
+    for i, layer in enumerate(self.fc):
+      out = layer.FProp(theta.fc[i], out)
+      if self._dropout and not self.do_eval:
+        out = py_utils.DeterministicDropout(out, 1.0 - self._dropout)
+    return out
+
+
+class Conv2DLayer(BaseLayer):
+  """NHWC conv + optional BN-free bias + activation
+  (reference layers.py:182-771, simplified: no weight norm)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('filter_shape', (3, 3, 1, 32),
+             '(H, W, in_channels, out_channels).')
+    p.Define('filter_stride', (1, 1), '(stride_h, stride_w).')
+    p.Define('padding', 'SAME', 'SAME or VALID.')
+    p.Define('has_bias', True, 'Bias add.')
+    p.Define('activation', 'RELU', 'Activation.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    fh, fw, cin, cout = p.filter_shape
+    self.CreateVariable('w', py_utils.WeightParams(
+        [fh, fw, cin, cout], p.params_init, p.dtype))
+    if p.has_bias:
+      self.CreateVariable('b', py_utils.WeightParams(
+          [cout], py_utils.WeightInit.Constant(0.0), p.dtype))
+
+  def FProp(self, theta: NestedMap, inputs: torch.Tensor) -> torch.Tensor:
+    """inputs: [B, H, W, C] (batch-major NHWC like the reference)."""
+    p = self.p
+    x = inputs.permute(0, 3, 1, 2)  # NCHW for torch/MIOpen
+    w = theta.w.permute(3, 2, 0, 1)  # OIHW
+    fh, fw = p.filter_shape[:2]
+    sh, sw = p.filter_stride
+    if p.padding == 'SAME':
+      ih, iw = x.shape[2], x.shape[3]
+      oh = (ih + sh - 1) // sh
+      ow = (iw + sw - 1) // sw
+      pad_h = max(0, (oh - 1) * sh + fh - ih)
+      pad_w = max(0, (ow - 1) * sw + fw - iw)
+      x = F.pad(x, (pad_w // 2, pad_w - pad_w // 2,
+                    pad_h // 2, pad_h - pad_h // 2))
+    out = F.conv2d(x, w, theta.b if p.has_bias else None, stride=(sh, sw))
+    out = activations.GetFn(p.activation)(out)
+    return out.permute(0, 2, 3, 1)
+
+
+class PoolingLayer(BaseLayer):
+  """Max/avg pooling, NHWC (reference layers.py:2285)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('window_shape', (2, 2), 'Pool window (H, W).')
+    p.Define('window_stride', (2, 2), 'Stride (H, W).')
+    p.Define('pooling_type', 'MAX', 'MAX or AVG.')
+    return p
+
+  def FProp(self, theta: NestedMap, inputs: torch.Tensor) -> torch.Tensor:
+    p = self.p
+    x = inputs.permute(0, 3, 1, 2)
+    if p.pooling_type == 'MAX':
+      out = F.max_pool2d(x, p.window_shape, p.window_stride)
+    else:
+      out = F.avg_pool2d(x, p.window_shape, p.window_stride)
+    return out.permute(0, 2, 3, 1)
+
+
+class EmbeddingLayer(BaseLayer):
+  """Token embedding with scatter-add bwd HIP kernel on GPU
+  (reference layers.py:2679 SimpleEmbeddingLayer)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('vocab_size', 0, 'Vocabulary size.')
+    p.Define('embedding_dim', 0, 'Embedding dimension.')
+    p.Define('scale_sqrt_depth', False,
+             'Scale outputs by sqrt(dim) (transformer convention).')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    self.CreateVariable('wm', py_utils.WeightParams(
+        [p.vocab_size, p.embedding_dim],
+        py_utils.WeightInit.Gaussian(1.0 / math.sqrt(p.embedding_dim)),
+        p.dtype))
+
+  def EmbLookup(self, theta: NestedMap, ids: torch.Tensor) -> torch.Tensor:
+    out = F.embedding(ids, theta.wm)
+    if self.p.scale_sqrt_depth:
+      out = out * (self.p.embedding_dim ** 0.5)
+    return out
+
+  def FProp(self, theta: NestedMap, ids: torch.Tensor) -> torch.Tensor:
+    return self.EmbLookup(theta, ids)
+
+
+class PositionalEmbeddingLayer(BaseLayer):
+  """Sinusoidal positions (reference layers.py:3380)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('embedding_dim', 0, 'Dimension.')
+    p.Define('min_timescale', 1, 'Min timescale.')
+    p.Define('max_timescale', 10_000, 'Max timescale.')
+    return p
+
+  def FProp(self, theta: NestedMap, seq_length: int,
+            device=None) -> torch.Tensor:
+    p = self.p
+    pos = torch.arange(seq_length, dtype=torch.float32, device=device)
+    num_ts = p.embedding_dim // 2
+    log_inc = math.log(p.max_timescale / p.min_timescale) / max(1, num_ts - 1)
+    inv_ts = p.min_timescale * torch.exp(
+        torch.arange(num_ts, dtype=torch.float32, device=device) * -log_inc)
+    scaled = pos[:, None] * inv_ts[None, :]
+    emb = torch.cat([torch.sin(scaled), torch.cos(scaled)], dim=1)
+    if p.embedding_dim % 2:
+      emb = F.pad(emb, (0, 1))
+    return emb.to(self.fprop_dtype)
+
+
+class LayerNorm(BaseLayer):
+  """Layer normalization over the last dim (reference layers.py:4927).
+
+  GPU: fused one-pass HIP kernel; CPU: torch reference.
+  """
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('input_dim', 0, 'Normalized dimension.')
+    p.Define('epsilon', 1e-6, 'Epsilon.')
+    p.Define('use_fused_layernorm', True, 'Use the HIP kernel on GPU.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    self.CreateVariable('scale', py_utils.WeightParams(
+        [p.input_dim], py_utils.WeightInit.Constant(0.0), p.dtype))
+    self.CreateVariable('bias', py_utils.WeightParams(
+        [p.input_dim], py_utils.WeightInit.Constant(0.0), p.dtype))
+
+  def FProp(self, theta: NestedMap, inputs: torch.Tensor) -> torch.Tensor:
+    p = self.p
+    # Reference convention: scale is (1 + scale) so init 0 == identity.
+    if inputs.is_cuda and p.use_fused_layernorm:
+      from lingvo_amd.ops import layer_norm as ln_ops
+      return ln_ops.layer_norm(inputs, theta.scale, theta.bias, p.epsilon)
+    x = inputs.float()
+    mean = x.mean(dim=-1, keepdim=True)
+    var = x.var(dim=-1, unbiased=False, keepdim=True)
+    out = (x - mean) * torch.rsqrt(var + p.epsilon)
+    out = out * (1.0 + theta.scale.float()) + theta.bias.float()
+    return out.to(inputs.dtype)
+
+
+class RmsNorm(BaseLayer):
+  """RMS normalization (no mean subtraction)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('input_dim', 0, 'Normalized dimension.')
+    p.Define('epsilon', 1e-6, 'Epsilon.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    self.CreateVariable('scale', py_utils.WeightParams(
+        [self.p.input_dim], py_utils.WeightInit.Constant(0.0), self.p.dtype))
+
+  def FProp(self, theta: NestedMap, inputs: torch.Tensor) -> torch.Tensor:
+    if inputs.is_cuda:
+      from lingvo_amd.ops import layer_norm as ln_ops
+      return ln_ops.rms_norm(inputs, theta.scale, self.p.epsilon)
+    x = inputs.float()
+    ms = x.pow(2).mean(dim=-1, keepdim=True)
+    out = x * torch.rsqrt(ms + self.p.epsilon) * (1.0 + theta.scale.float())
+    return out.to(inputs.dtype)
+
+
+class DropoutLayer(BaseLayer):
+  """Deterministic dropout under StepSeedScope (reference layers.py:4916)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('keep_prob', 1.0, 'Keep probability.')
+    return p
+
+  def FProp(self, theta: NestedMap, inputs: torch.Tensor) -> torch.Tensor:
+    if self.do_eval or self.p.keep_prob >= 1.0:
+      return inputs
+    return py_utils.DeterministicDropout(inputs, self.p.keep_prob)
+
+
+class SimpleFullSoftmax(BaseLayer):
+  """Softmax + cross-entropy over a full vocab (reference layers.py:3697).
+
+  XentLoss returns a NestedMap with total_xent/total_weight/per_example.
+  On GPU the fused logits+log-softmax+gather HIP path avoids
+  materializing [B*T, V] probabilities.
+  """
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('input_dim', 0, 'Input dim.')
+    p.Define('num_classes', 0, 'Number of classes.')
+    p.Define('chunk_size', 0, 'If >0, compute xent in vocab chunks.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    self.CreateVariable('linear_w', py_utils.WeightParams(
+        [p.input_dim, p.num_classes], p.params_init, p.dtype))
+    self.CreateVariable('bias', py_utils.WeightParams(
+        [p.num_classes], py_utils.WeightInit.Constant(0.0), p.dtype))
+
+  def Logits(self, theta: NestedMap, inputs: torch.Tensor) -> torch.Tensor:
+    return torch.matmul(inputs, theta.linear_w) + theta.bias
+
+  def XentLossFromLogits(self, logits: torch.Tensor,
+                         class_ids: Optional[torch.Tensor] = None,
+                         class_probabilities: Optional[torch.Tensor] = None
+                         ) -> NestedMap:
+    log_probs = F.log_softmax(logits.float(), dim=-1)
+    if class_probabilities is not None:
+      per_example = -(class_probabilities.float() * log_probs).sum(-1)
+    else:
+      per_example = F.nll_loss(log_probs, class_ids.reshape(-1).long(),
+                               reduction='none')
+    return NestedMap(per_example_xent=per_example, log_probs=log_probs)
+
+  def XentLoss(self, theta: NestedMap, inputs: torch.Tensor,
+               class_weights: torch.Tensor,
+               class_ids: Optional[torch.Tensor] = None,
+               class_probabilities: Optional[torch.Tensor] = None
+               ) -> NestedMap:
+    p = self.p
+    inputs2d = inputs.reshape(-1, p.input_dim)
+    w = class_weights.reshape(-1).float()
+    if (inputs2d.is_cuda and class_probabilities is None and
+        p.chunk_size == 0):
+      from lingvo_amd.ops import softmax_xent
+      per_example = softmax_xent.logits_xent(
+          inputs2d, theta.linear_w, theta.bias, class_ids.reshape(-1))
+    else:
+      logits = self.Logits(theta, inputs2d)
+      per_example = self.XentLossFromLogits(
+          logits, class_ids,
+          None if class_probabilities is None else
+          class_probabilities.reshape(-1, p.num_classes)).per_example_xent
+    total_weight = w.sum()
+    total_xent = (per_example * w).sum()
+    return NestedMap(
+        total_xent=total_xent,
+        total_weight=total_weight,
+        avg_xent=total_xent / total_weight.clamp_min(1e-8),
+        per_example_xent=per_example)
+
+
+class SharedSoftmaxLayer(SimpleFullSoftmax):
+  """Softmax sharing its weight as the embedding table
+  (reference layers.py:4403)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('scale_sqrt_depth', True, 'Scale embedding by sqrt(dim).')
+    return p
+
+  def EmbLookup(self, theta: NestedMap, ids: torch.Tensor) -> torch.Tensor:
+    emb = F.embedding(ids, theta.linear_w.t().contiguous())
+    if self.p.scale_sqrt_depth:
+      emb = emb * (self.p.input_dim ** 0.5)
+    return emb
+
+
+class UniformLabelSmoother(BaseLayer):
+  """Uniform label smoothing (reference layers.py:5305)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('num_classes', 0, 'Number of classes.')
+    p.Define('uncertainty', 0.1, 'Smoothing mass.')
+    return p
+
+  def FProp(self, theta: NestedMap, target_ids: torch.Tensor
+            ) -> torch.Tensor:
+    p = self.p
+    off = p.uncertainty / (p.num_classes - 1)
+    probs = torch.full(
+        (*target_ids.shape, p.num_classes), off,
+        dtype=torch.float32, device=target_ids.device)
+    probs.scatter_(-1, target_ids.long().unsqueeze(-1), 1.0 - p.uncertainty)
+    return probs

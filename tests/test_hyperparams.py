@@ -1,0 +1,124 @@
+"""Tests for Params/InstantiableParams (reference hyperparams_test.py
+capability)."""
+
+import pytest
+
+from lingvo_amd.core.hyperparams import (CopyParamsTo, InstantiableParams,
+                                         Params)
+
+
+def test_define_get_set():
+  p = Params()
+  p.Define('lr', 0.1, 'learning rate')
+  assert p.lr == 0.1
+  p.lr = 0.2
+  assert p.lr == 0.2
+  p.Set(lr=0.3)
+  assert p.Get('lr') == 0.3
+
+
+def test_unknown_attr_raises():
+  p = Params()
+  p.Define('foo_bar', 1, '')
+  with pytest.raises(AttributeError):
+    p.foo = 2
+  with pytest.raises(AttributeError):
+    _ = p.baz
+
+
+def test_double_define_raises():
+  p = Params()
+  p.Define('x', 1, '')
+  with pytest.raises(AttributeError):
+    p.Define('x', 2, '')
+
+
+def test_nested_and_dotted_paths():
+  inner = Params()
+  inner.Define('dim', 8, '')
+  p = Params()
+  p.Define('encoder', inner, '')
+  assert p.Get('encoder.dim') == 8
+  p.SetPath('encoder.dim', 16)
+  assert p.encoder.dim == 16
+
+
+def test_copy_is_deep():
+  inner = Params()
+  inner.Define('dim', 8, '')
+  p = Params()
+  p.Define('encoder', inner, '')
+  q = p.Copy()
+  q.encoder.dim = 99
+  assert p.encoder.dim == 8
+
+
+def test_freeze():
+  p = Params()
+  p.Define('x', 1, '')
+  p.Freeze()
+  with pytest.raises(TypeError):
+    p.x = 2
+  q = p.Copy()
+  q.x = 2  # copies are mutable
+  assert q.x == 2
+
+
+def test_totext_fromtext_roundtrip():
+  inner = Params()
+  inner.Define('dim', 8, '')
+  inner.Define('tag', 'hello world', '')
+  p = Params()
+  p.Define('encoder', inner, '')
+  p.Define('lr', 0.5, '')
+  text = p.ToText()
+  assert 'encoder.dim : 8' in text
+  assert 'lr : 0.5' in text
+  q = p.Copy()
+  q.encoder.dim = 0
+  q.lr = 0.0
+  q.FromText(text)
+  assert q.encoder.dim == 8
+  assert q.lr == 0.5
+  assert q.encoder.tag == 'hello world'
+
+
+def test_textdiff():
+  p = Params()
+  p.Define('x', 1, '')
+  q = p.Copy()
+  q.x = 2
+  diff = p.TextDiff(q)
+  assert '< x : 1' in diff and '> x : 2' in diff
+  assert p.TextDiff(p.Copy()) == ''
+
+
+def test_instantiable():
+  class Dummy:
+    def __init__(self, params):
+      self.params = params
+
+  p = InstantiableParams(Dummy)
+  obj = p.Instantiate()
+  assert isinstance(obj, Dummy)
+
+
+def test_copy_params_to():
+  a = Params()
+  a.Define('x', 1, '')
+  a.Define('y', 2, '')
+  b = Params()
+  b.Define('x', 0, '')
+  b.Define('z', 3, '')
+  CopyParamsTo(a, b)
+  assert b.x == 1 and b.z == 3
+
+
+def test_list_of_params_totext():
+  sub = Params()
+  sub.Define('d', 1, '')
+  p = Params()
+  p.Define('blocks', [sub, sub.Copy()], '')
+  text = p.ToText()
+  assert 'blocks[0].d : 1' in text
+  assert 'blocks[1].d : 1' in text

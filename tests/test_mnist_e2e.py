@@ -1,0 +1,83 @@
+"""End-to-end: image.mnist.LeNet5 trains on CPU via the trainer CLI
+(BASELINE config 1; reference trainer_test.py capability)."""
+
+import json
+import os
+
+import pytest
+import torch
+
+from lingvo_amd.core import registry
+from lingvo_amd.runtime import trainer as trainer_cli
+
+
+def test_registry_lookup():
+  cls = registry.GetClass('image.mnist.LeNet5')
+  assert cls is not None
+  model_p = registry.GetParams('image.mnist.LeNet5', 'Train')
+  assert model_p.input is not None
+  model = model_p.Instantiate()
+  total = sum(p.numel() for p in model.parameters())
+  assert total > 10_000
+
+
+def test_mnist_loss_decreases(tmp_path):
+  logdir = str(tmp_path / 'log')
+  trainer_cli.main([
+      '--model=image.mnist.LeNet5', f'--logdir={logdir}',
+      '--job=trainer_client', '--max_steps=30', '--device=cpu'])
+  # checkpoint layout contract
+  train_dir = os.path.join(logdir, 'train')
+  assert os.path.exists(os.path.join(train_dir, 'checkpoint'))
+  cks = [f for f in os.listdir(train_dir) if f.startswith('ckpt-')]
+  assert any(f == 'ckpt-00000030.pt' for f in cks), cks
+  # control artifacts
+  assert os.path.exists(os.path.join(logdir, 'control', 'params.txt'))
+  assert os.path.exists(os.path.join(logdir, 'control',
+                                     'model_analysis.txt'))
+  # loss decreased
+  with open(os.path.join(train_dir, 'metrics.jsonl')) as f:
+    recs = [json.loads(l) for l in f]
+  assert recs[-1]['loss'] < recs[0]['loss']
+
+
+def test_resume_from_checkpoint(tmp_path):
+  logdir = str(tmp_path / 'log')
+  trainer_cli.main([
+      '--model=image.mnist.LeNet5', f'--logdir={logdir}',
+      '--job=trainer', '--max_steps=5', '--device=cpu'])
+  trainer_cli.main([
+      '--model=image.mnist.LeNet5', f'--logdir={logdir}',
+      '--job=trainer', '--max_steps=8', '--device=cpu'])
+  train_dir = os.path.join(logdir, 'train')
+  cks = sorted(f for f in os.listdir(train_dir) if f.endswith('.pt'))
+  assert cks[-1] == 'ckpt-00000008.pt'
+  payload = torch.load(os.path.join(train_dir, cks[-1]),
+                       map_location='cpu', weights_only=False)
+  assert payload['step'] == 8
+
+
+def test_evaler_and_decoder_run_once(tmp_path):
+  logdir = str(tmp_path / 'log')
+  trainer_cli.main([
+      '--model=image.mnist.LeNet5', f'--logdir={logdir}',
+      '--job=trainer', '--max_steps=3', '--device=cpu'])
+  trainer_cli.main([
+      '--model=image.mnist.LeNet5', f'--logdir={logdir}',
+      '--job=evaler_test', '--run_once', '--device=cpu'])
+  trainer_cli.main([
+      '--model=image.mnist.LeNet5', f'--logdir={logdir}',
+      '--job=decoder_test', '--run_once', '--device=cpu'])
+  with open(os.path.join(logdir, 'eval_test', 'metrics.jsonl')) as f:
+    rec = json.loads(f.readline())
+  assert 'loss' in rec and rec['step'] == 3
+  with open(os.path.join(logdir, 'decoder_test', 'metrics.jsonl')) as f:
+    rec = json.loads(f.readline())
+  assert rec['step'] == 3
+
+
+def test_inspect_modes(tmp_path, capsys):
+  trainer_cli.main(['--model=image.mnist.LeNet5',
+                    f'--logdir={tmp_path}', '--mode=inspect_params'])
+  out = capsys.readouterr().out
+  assert 'task.hidden_dim : 300' in out
