@@ -1,0 +1,90 @@
+"""GPU numerics tests: HIP kernels vs plain fp32 torch references.
+
+All tests here are @pytest.mark.gpu and run on an MI355X via gpurun.
+Tolerances account for bf16 IO (the kernels accumulate in fp32).
+"""
+
+import pytest
+import torch
+
+gpu = pytest.mark.gpu
+
+
+def _ln_ref(x, scale, bias, eps=1e-6):
+  xf = x.float()
+  mean = xf.mean(-1, keepdim=True)
+  var = xf.var(-1, unbiased=False, keepdim=True)
+  return (xf - mean) * torch.rsqrt(var + eps) * (1 + scale.float()) + \
+      bias.float()
+
+
+def _rms_ref(x, scale, eps=1e-6):
+  xf = x.float()
+  return xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps) * \
+      (1 + scale.float())
+
+
+@gpu
+@pytest.mark.parametrize('shape', [(4, 7, 512), (2, 3, 1024), (8, 2048),
+                                   (3, 5, 384), (2, 130)])
+def test_layer_norm_fwd_matches_ref(shape):
+  from lingvo_amd.ops import layer_norm as ln
+  torch.manual_seed(0)
+  d = shape[-1]
+  x = torch.randn(shape, device='cuda', dtype=torch.bfloat16) * 3 + 1
+  scale = torch.randn(d, device='cuda') * 0.1
+  bias = torch.randn(d, device='cuda') * 0.1
+  y = ln.layer_norm(x, scale, bias)
+  ref = _ln_ref(x, scale, bias)
+  assert (y.float() - ref).abs().max() < 0.05
+
+
+@gpu
+@pytest.mark.parametrize('d', [512, 1024, 384])
+def test_layer_norm_bwd_matches_ref(d):
+  from lingvo_amd.ops import layer_norm as ln
+  torch.manual_seed(1)
+  x = (torch.randn(6, 9, d, device='cuda', dtype=torch.bfloat16)
+       ).requires_grad_(True)
+  scale = (torch.randn(d, device='cuda') * 0.1).requires_grad_(True)
+  bias = (torch.randn(d, device='cuda') * 0.1).requires_grad_(True)
+  y = ln.layer_norm(x, scale, bias)
+  g = torch.randn_like(y)
+  y.backward(g)
+
+  xr = x.detach().float().requires_grad_(True)
+  sr = scale.detach().clone().requires_grad_(True)
+  br = bias.detach().clone().requires_grad_(True)
+  _ln_ref(xr, sr, br).backward(g.float())
+
+  assert (x.grad.float() - xr.grad).abs().max() < 0.1
+  rows = x.numel() // d
+  assert (scale.grad - sr.grad).abs().max() / max(1.0, rows ** 0.5) < 0.3
+  assert (bias.grad - br.grad).abs().max() / max(1.0, rows ** 0.5) < 0.3
+
+
+@gpu
+def test_rms_norm_fwd_bwd():
+  from lingvo_amd.ops import layer_norm as ln
+  torch.manual_seed(2)
+  d = 512
+  x = torch.randn(4, 11, d, device='cuda',
+                  dtype=torch.bfloat16).requires_grad_(True)
+  scale = (torch.randn(d, device='cuda') * 0.1).requires_grad_(True)
+  y = ln.rms_norm(x, scale)
+  ref = _rms_ref(x.detach(), scale.detach())
+  assert (y.float() - ref).abs().max() < 0.05
+  g = torch.randn_like(y)
+  y.backward(g)
+  xr = x.detach().float().requires_grad_(True)
+  sr = scale.detach().clone().requires_grad_(True)
+  _rms_ref(xr, sr).backward(g.float())
+  assert (x.grad.float() - xr.grad).abs().max() < 0.1
+
+
+@gpu
+def test_native_ext_is_loaded():
+  """Guard against silent eager fallback on GPU boxes."""
+  import lingvo_amd.ops._lingvo_ops as ext
+  assert hasattr(ext, 'layer_norm_fwd')
+  assert '/root/' in ext.__file__ or 'lingvo_amd' in ext.__file__
