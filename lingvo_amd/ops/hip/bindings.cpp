@@ -13,7 +13,29 @@ std::vector<torch::Tensor> layer_norm_bwd(torch::Tensor dy, torch::Tensor x,
                                           torch::Tensor mean,
                                           torch::Tensor rstd, bool rms);
 
+// mfma_probe.hip
+torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor bT);
+
+// flash_attn.hip
+std::vector<torch::Tensor> fa_fwd(torch::Tensor q, torch::Tensor k,
+                                  torch::Tensor v,
+                                  c10::optional<torch::Tensor> klen,
+                                  c10::optional<torch::Tensor> bias,
+                                  int64_t win_l, int64_t win_r,
+                                  int64_t bias_clip, double scale);
+std::vector<torch::Tensor> fa_bwd(torch::Tensor dout, torch::Tensor q,
+                                  torch::Tensor k, torch::Tensor v,
+                                  torch::Tensor o, torch::Tensor lse,
+                                  c10::optional<torch::Tensor> klen,
+                                  c10::optional<torch::Tensor> bias,
+                                  bool bias_grad, int64_t win_l,
+                                  int64_t win_r, int64_t bias_clip,
+                                  double scale);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layer_norm_fwd", &layer_norm_fwd, "Fused LayerNorm/RMSNorm fwd");
   m.def("layer_norm_bwd", &layer_norm_bwd, "Fused LayerNorm/RMSNorm bwd");
+  m.def("mfma_probe", &mfma_probe, "MFMA 16x16x32 layout probe");
+  m.def("fa_fwd", &fa_fwd, "Flash attention forward");
+  m.def("fa_bwd", &fa_bwd, "Flash attention backward");
 }

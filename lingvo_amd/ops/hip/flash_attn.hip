@@ -1,0 +1,746 @@
+// Fused flash-attention forward+backward for gfx950 (SURVEY.md K2-K4;
+// reference einsums: lingvo/core/batch_major_attention.py:508-514 and
+// AttenProbs:943 — BTNH layouts, causal/local/padding masks, clipped
+// relative-position bias, GQA).
+//
+// Structure (v1, correctness-first; guide §B "fused attention prefill"):
+//  - fwd: grid (ceil(T/64), N, B); block = 4 waves; each wave owns a
+//    16-row Q strip. K staged in LDS row-major (XOR-swizzled), V staged
+//    transposed [H][64]; online softmax in fp32 with per-row (m, l)
+//    tracked in the 16-lane C-layout groups; P routed through per-wave
+//    LDS for the PV MFMA. Saves LSE for backward.
+//  - bwd: grid (ceil(S/64), N_kv, B); each block owns a 64-key tile,
+//    loops q-heads of the GQA group and q-tiles; recomputes P^T from
+//    LSE; accumulates dK/dV in registers (written once, no atomics);
+//    dQ accumulated via fp32 global atomics; optional clipped
+//    rel-position bias gradient via LDS accumulation.
+//
+// Masks: key k is visible from query q iff
+//   k < klen[b]  AND  (win_l < 0 OR k >= q - win_l)
+//                AND  (win_r < 0 OR k <= q + win_r).
+// Causal = win_r 0. LocalSelfAttention (batch_major_attention.py:2656)
+// = finite (win_l, win_r).
+
+#include <ATen/cuda/CUDAContext.h>
+#include <torch/extension.h>
+
+#include "mfma.h"
+
+namespace {
+
+constexpr int QT = 64;   // q rows per block (fwd) / q rows per inner tile (bwd)
+constexpr int KT = 64;   // keys per tile
+constexpr int NW = 4;    // waves per block
+constexpr int BLOCK = NW * WAVE_SIZE;
+constexpr float NEG_INF = -1e30f;
+
+// ---------------------------------------------------------------------------
+// LDS staging helpers. Row-major tiles are stored with a per-row XOR
+// swizzle on 16B granules (mfma.h swz) so B-fragment ds_read_b128 at
+// row-stride >= 128B doesn't bank-conflict (guide §6 G4).
+// ---------------------------------------------------------------------------
+
+// Stage ROWS x H bf16 from global (row i at src + i*src_stride elems) into
+// LDS row-major with swizzle. Rows >= valid_rows are zeroed.
+template <int H, int ROWS>
+__device__ void stage_regular(const unsigned short* src, long src_stride,
+                              int valid_rows, char* dst) {
+  constexpr int ROWB = H * 2;
+  constexpr int TPR = ROWB / 16;             // threads per row (16B each)
+  constexpr int RPP = BLOCK / TPR;           // rows per pass
+  const int tid = threadIdx.x;
+  const int r0 = tid / TPR;
+  const int byte0 = (tid % TPR) * 16;
+#pragma unroll
+  for (int rp = 0; rp < ROWS / RPP; ++rp) {
+    int row = r0 + rp * RPP;
+    ushortx8 v;
+    if (row < valid_rows) {
+      v = *reinterpret_cast<const ushortx8*>(src + (long)row * src_stride +
+                                             byte0 / 2);
+    } else {
+      for (int j = 0; j < 8; ++j) v[j] = 0;
+    }
+    *reinterpret_cast<ushortx8*>(dst + row * ROWB + swz(row, byte0)) = v;
+  }
+}
+
+// Stage ROWS x H bf16 from global into LDS TRANSPOSED as [H][ROWS]
+// (ROWS=64, 128B rows), swizzled per h-row. Scalar 2B writes.
+template <int H, int ROWS>
+__device__ void stage_transposed(const unsigned short* src, long src_stride,
+                                 int valid_rows, char* dst) {
+  constexpr int ROWB = H * 2;
+  constexpr int TPR = ROWB / 16;
+  constexpr int RPP = BLOCK / TPR;
+  const int tid = threadIdx.x;
+  const int r0 = tid / TPR;
+  const int h0 = (tid % TPR) * 8;
+#pragma unroll
+  for (int rp = 0; rp < ROWS / RPP; ++rp) {
+    int row = r0 + rp * RPP;
+    ushortx8 v;
+    if (row < valid_rows) {
+      v = *reinterpret_cast<const ushortx8*>(src + (long)row * src_stride +
+                                             h0);
+    } else {
+      for (int j = 0; j < 8; ++j) v[j] = 0;
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int h = h0 + j;
+      *reinterpret_cast<unsigned short*>(
+          dst + h * (ROWS * 2) + swz(h, row * 2)) = v[j];
+    }
+  }
+}
+
+// ds_read a B-fragment (8 bf16) from a swizzled row-major LDS tile.
+__device__ __forceinline__ bf16x8 lds_frag(const char* tile, int row,
+                                           int rowb, int byte_in_row) {
+  union {
+    ushortx8 u;
+    bf16x8 h;
+  } cv;
+  cv.u = *reinterpret_cast<const ushortx8*>(tile + row * rowb +
+                                            swz(row, byte_in_row));
+  return cv.h;
+}
+
+__device__ __forceinline__ bool visible(int q, int k, int klen, int win_l,
+                                        int win_r) {
+  if (k >= klen) return false;
+  if (win_l >= 0 && k < q - win_l) return false;
+  if (win_r >= 0 && k > q + win_r) return false;
+  return true;
+}
+
+// ---------------------------------------------------------------------------
+// Forward
+// ---------------------------------------------------------------------------
+template <int H>
+__global__ __launch_bounds__(BLOCK) void fa_fwd_kernel(
+    const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
+    const unsigned short* __restrict__ v,
+    const int* __restrict__ klen_ptr,          // [B] or null
+    const unsigned short* __restrict__ bias,   // [N][2C+1] or null
+    unsigned short* __restrict__ o, float* __restrict__ lse, int B, int T,
+    int S, int N, int NKV, int win_l, int win_r, int bias_clip, float scale) {
+  constexpr int ROWB = H * 2;
+  constexpr int KH = H / 32;   // mfma K-steps over head dim
+  constexpr int HF = H / 16;   // output col frags
+  extern __shared__ char smem[];
+  char* k_lds = smem;                        // [KT][H] swz
+  char* vt_lds = k_lds + KT * ROWB;          // [H][KT] swz
+  char* p_lds = vt_lds + H * KT * 2;         // [NW][16][KT] swz
+
+  const int qt = blockIdx.x;
+  const int n = blockIdx.y;
+  const int b = blockIdx.z;
+  const int nkv = n / (N / NKV);
+  const int wid = threadIdx.x / WAVE_SIZE;
+  const int lane = threadIdx.x & 63;
+  const int g = lane >> 4;       // 16-lane group
+  const int cl = lane & 15;      // col-in-frag
+  const int klen = klen_ptr ? klen_ptr[b] : S;
+
+  const int q0 = qt * QT + wid * 16;  // this wave's first q row
+
+  // Q fragments (A layout): row = cl, k = g*8 + kk*32. Folded zeros for
+  // rows >= T.
+  bf16x8 qfrag[KH];
+#pragma unroll
+  for (int kk = 0; kk < KH; ++kk) {
+    int qrow = q0 + cl;
+    if (qrow < T) {
+      qfrag[kk] = load_bf16x8_bits(
+          q + (((long)b * T + qrow) * N + n) * H + kk * 32 + g * 8);
+    } else {
+      float z[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+      qfrag[kk] = pack_bf16x8(z);
+    }
+  }
+
+  f32x4 acc_o[HF];
+#pragma unroll
+  for (int hf = 0; hf < HF; ++hf) acc_o[hf] = {0.f, 0.f, 0.f, 0.f};
+  float m_run[4], l_run[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    m_run[r] = NEG_INF;
+    l_run[r] = 0.f;
+  }
+
+  // KV tile range from the window.
+  int kmax_excl = min(klen, win_r < 0 ? S : min(S, qt * QT + QT - 1 + win_r + 1));
+  int kmin = win_l < 0 ? 0 : max(0, qt * QT - win_l);
+  const int kt_lo = kmin / KT;
+  const int kt_hi = (max(kmax_excl, 1) - 1) / KT;
+
+  for (int kt = kt_lo; kt <= kt_hi; ++kt) {
+    const int kbase = kt * KT;
+    // Stage K and V^T.
+    stage_regular<H, KT>(k + (((long)b * S + kbase) * NKV + nkv) * H,
+                         (long)NKV * H, klen - kbase, k_lds);
+    stage_transposed<H, KT>(v + (((long)b * S + kbase) * NKV + nkv) * H,
+                            (long)NKV * H, klen - kbase, vt_lds);
+    __syncthreads();
+
+    // S strip: 16 q rows x KT keys, fp32.
+    float s[4][4];  // [nf over keys][r]
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int kk = 0; kk < KH; ++kk) {
+        bf16x8 bfr = lds_frag(k_lds, nf * 16 + cl, ROWB, (kk * 32 + g * 8) * 2);
+        acc = mfma16x16x32_bf16(qfrag[kk], bfr, acc);
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r) s[nf][r] = acc[r];
+    }
+
+    // Scale + mask + bias.
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      const int kcol = kbase + nf * 16 + cl;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int qrow = q0 + g * 4 + r;
+        float val = s[nf][r] * scale;
+        if (bias) {
+          int d = qrow - kcol;
+          d = d < -bias_clip ? -bias_clip : (d > bias_clip ? bias_clip : d);
+          val += bf16_bits_to_float(
+              bias[(long)n * (2 * bias_clip + 1) + d + bias_clip]);
+        }
+        if (!visible(qrow, kcol, klen, win_l, win_r) || qrow >= T)
+          val = NEG_INF;
+        s[nf][r] = val;
+      }
+    }
+
+    // Online softmax. Row r lives across the 16 lanes with this lane's g.
+    float rowmax[4], rowsum[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float mx = fmaxf(fmaxf(s[0][r], s[1][r]), fmaxf(s[2][r], s[3][r]));
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1) mx = fmaxf(mx, __shfl_xor(mx, off));
+      rowmax[r] = mx;
+    }
+    float alpha[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float m_new = fmaxf(m_run[r], rowmax[r]);
+      alpha[r] = (m_run[r] == NEG_INF) ? 0.f : __expf(m_run[r] - m_new);
+      m_run[r] = m_new;
+      float sum = 0.f;
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        float p = (s[nf][r] <= NEG_INF * 0.5f) ? 0.f
+                                               : __expf(s[nf][r] - m_new);
+        s[nf][r] = p;
+        sum += p;
+      }
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1) sum += __shfl_xor(sum, off);
+      rowsum[r] = sum;
+      l_run[r] = l_run[r] * alpha[r] + sum;
+#pragma unroll
+      for (int hf = 0; hf < HF; ++hf) acc_o[hf][r] *= alpha[r];
+    }
+
+    // P -> per-wave LDS (bf16, C layout -> row-major [16][KT], swizzled).
+    char* pw = p_lds + wid * 16 * (KT * 2);
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = g * 4 + r;
+        int col = nf * 16 + cl;
+        *reinterpret_cast<unsigned short*>(
+            pw + row * (KT * 2) + swz(row, col * 2)) =
+            float_to_bf16_bits(s[nf][r]);
+      }
+    }
+    // Same-wave DS ordering makes the reads below safe without a barrier.
+
+    // O += P @ V : A = P (rows=q, k=keys), B from Vt (k=keys, col=h).
+#pragma unroll
+    for (int kk2 = 0; kk2 < KT / 32; ++kk2) {
+      bf16x8 pa = lds_frag(pw, cl, KT * 2, (kk2 * 32 + g * 8) * 2);
+#pragma unroll
+      for (int hf = 0; hf < HF; ++hf) {
+        bf16x8 vb =
+            lds_frag(vt_lds, hf * 16 + cl, KT * 2, (kk2 * 32 + g * 8) * 2);
+        acc_o[hf] = mfma16x16x32_bf16(pa, vb, acc_o[hf]);
+      }
+    }
+    __syncthreads();
+  }
+
+  // Epilogue: divide by l, store O (bf16) and LSE (fp32).
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int qrow = q0 + g * 4 + r;
+    if (qrow >= T) continue;
+    float inv_l = l_run[r] > 0.f ? 1.f / l_run[r] : 0.f;
+    unsigned short* orow = o + (((long)b * T + qrow) * N + n) * H;
+#pragma unroll
+    for (int hf = 0; hf < HF; ++hf) {
+      orow[hf * 16 + cl] = float_to_bf16_bits(acc_o[hf][r] * inv_l);
+    }
+    if (cl == 0) {
+      lse[((long)b * N + n) * T + qrow] =
+          l_run[r] > 0.f ? m_run[r] + __logf(l_run[r]) : NEG_INF;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Backward: delta = rowsum(dO * O)
+// ---------------------------------------------------------------------------
+template <int H>
+__global__ void fa_bwd_delta(const unsigned short* __restrict__ dout,
+                             const unsigned short* __restrict__ o,
+                             float* __restrict__ delta, int B, int T, int N) {
+  // One wave per (b, t, n) row.
+  const long row = (long)blockIdx.x * NW + threadIdx.x / WAVE_SIZE;
+  const int lane = threadIdx.x & 63;
+  if (row >= (long)B * T * N) return;
+  // row enumerates (b, t, n) in memory order of [B,T,N,H].
+  const unsigned short* dp = dout + row * H;
+  const unsigned short* op = o + row * H;
+  float sum = 0.f;
+#pragma unroll
+  for (int i = lane; i < H; i += WAVE_SIZE) {
+    sum += bf16_bits_to_float(dp[i]) * bf16_bits_to_float(op[i]);
+  }
+  sum = wave_reduce_sum(sum);
+  if (lane == 0) {
+    long b = row / ((long)T * N);
+    long tn = row % ((long)T * N);
+    long t = tn / N, n = tn % N;
+    delta[(b * N + n) * T + t] = sum;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Backward main
+// ---------------------------------------------------------------------------
+template <int H, bool BIAS_GRAD>
+__global__ __launch_bounds__(BLOCK) void fa_bwd_kernel(
+    const unsigned short* __restrict__ dout,
+    const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
+    const unsigned short* __restrict__ v, const float* __restrict__ lse,
+    const float* __restrict__ delta, const int* __restrict__ klen_ptr,
+    const unsigned short* __restrict__ bias, float* __restrict__ dq_acc,
+    unsigned short* __restrict__ dk, unsigned short* __restrict__ dv,
+    float* __restrict__ dbias, int B, int T, int S, int N, int NKV,
+    int win_l, int win_r, int bias_clip, float scale) {
+  constexpr int ROWB = H * 2;
+  constexpr int KH = H / 32;
+  constexpr int HF = H / 16;
+  const int nbias = 2 * bias_clip + 1;
+  extern __shared__ char smem[];
+  char* q_lds = smem;                        // [QT][H] swz
+  char* qt_lds = q_lds + QT * ROWB;          // [H][QT] swz
+  char* do_lds = qt_lds + H * QT * 2;        // [QT][H] swz
+  char* dot_lds = do_lds + QT * ROWB;        // [H][QT] swz
+  char* kt_lds = dot_lds + H * QT * 2;       // [H][KT] swz
+  char* ds_lds = kt_lds + H * KT * 2;        // [QT][KT] swz
+  char* a_lds = ds_lds + QT * KT * 2;        // [NW][16][QT] swz
+  float* lse_s = (float*)(a_lds + NW * 16 * QT * 2);   // [QT]
+  float* delta_s = lse_s + QT;                         // [QT]
+  float* dbias_s = delta_s + QT;                       // [nbias] if BIAS_GRAD
+
+  const int kt = blockIdx.x;
+  const int nkv = blockIdx.y;
+  const int b = blockIdx.z;
+  const int group = N / NKV;
+  const int wid = threadIdx.x / WAVE_SIZE;
+  const int lane = threadIdx.x & 63;
+  const int g = lane >> 4;
+  const int cl = lane & 15;
+  const int klen = klen_ptr ? klen_ptr[b] : S;
+  const int kbase = kt * KT;
+  const int k0 = kbase + wid * 16;  // wave's first key
+
+  // Stage K^T once (shared by all heads in the group).
+  stage_transposed<H, KT>(k + (((long)b * S + kbase) * NKV + nkv) * H,
+                          (long)NKV * H, klen - kbase, kt_lds);
+
+  // K, V fragments (A layout) from global: row = key (cl), k = h.
+  bf16x8 kfrag[KH], vfrag[KH];
+  {
+    int key = k0 + cl;
+    bool ok = key < klen;
+#pragma unroll
+    for (int kk = 0; kk < KH; ++kk) {
+      if (ok) {
+        long off = (((long)b * S + key) * NKV + nkv) * H + kk * 32 + g * 8;
+        kfrag[kk] = load_bf16x8_bits(k + off);
+        vfrag[kk] = load_bf16x8_bits(v + off);
+      } else {
+        float z[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+        kfrag[kk] = pack_bf16x8(z);
+        vfrag[kk] = pack_bf16x8(z);
+      }
+    }
+  }
+
+  // q-tile range for this key tile.
+  int qlo = win_r < 0 ? 0 : max(0, kbase - win_r);
+  int qhi = win_l < 0 ? T - 1 : min(T - 1, kbase + KT - 1 + win_l);
+  const int qt_lo = qlo / QT, qt_hi = max(qhi, 0) / QT;
+
+  for (int head = 0; head < group; ++head) {
+    const int n = nkv * group + head;
+
+    f32x4 acc_dk[HF], acc_dv[HF];
+#pragma unroll
+    for (int hf = 0; hf < HF; ++hf) {
+      acc_dk[hf] = {0.f, 0.f, 0.f, 0.f};
+      acc_dv[hf] = {0.f, 0.f, 0.f, 0.f};
+    }
+    if (BIAS_GRAD) {
+      for (int i = threadIdx.x; i < nbias; i += BLOCK) dbias_s[i] = 0.f;
+    }
+
+    for (int qt2 = qt_lo; qt2 <= qt_hi; ++qt2) {
+      const int qb = qt2 * QT;
+      __syncthreads();
+      stage_regular<H, QT>(q + (((long)b * T + qb) * N + n) * H, (long)N * H,
+                           T - qb, q_lds);
+      stage_transposed<H, QT>(q + (((long)b * T + qb) * N + n) * H,
+                              (long)N * H, T - qb, qt_lds);
+      stage_regular<H, QT>(dout + (((long)b * T + qb) * N + n) * H,
+                           (long)N * H, T - qb, do_lds);
+      stage_transposed<H, QT>(dout + (((long)b * T + qb) * N + n) * H,
+                              (long)N * H, T - qb, dot_lds);
+      for (int i = threadIdx.x; i < QT; i += BLOCK) {
+        int qrow = qb + i;
+        lse_s[i] = qrow < T ? lse[((long)b * N + n) * T + qrow] : NEG_INF;
+        delta_s[i] = qrow < T ? delta[((long)b * N + n) * T + qrow] : 0.f;
+      }
+      __syncthreads();
+
+      // S^T strip: rows = 16 keys (this wave), cols = QT queries.
+      float pt[4][4];   // P^T
+      float dlg[4][4];  // dLogits
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int kk = 0; kk < KH; ++kk) {
+          bf16x8 bq =
+              lds_frag(q_lds, nf * 16 + cl, ROWB, (kk * 32 + g * 8) * 2);
+          acc = mfma16x16x32_bf16(kfrag[kk], bq, acc);
+        }
+        const int qcol = qb + nf * 16 + cl;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int key = k0 + g * 4 + r;
+          float val = acc[r] * scale;
+          if (bias) {
+            int d = qcol - key;
+            d = d < -bias_clip ? -bias_clip : (d > bias_clip ? bias_clip : d);
+            val += bf16_bits_to_float(bias[(long)n * nbias + d + bias_clip]);
+          }
+          bool vis = visible(qcol, key, klen, win_l, win_r) && qcol < T;
+          float l = lse_s[nf * 16 + cl];
+          pt[nf][r] =
+              (vis && l > NEG_INF * 0.5f) ? __expf(val - l) : 0.f;
+        }
+      }
+
+      // P^T -> a_lds; dV += P^T @ dO (B from dOt).
+      char* aw = a_lds + wid * 16 * (QT * 2);
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int row = g * 4 + r;
+          int col = nf * 16 + cl;
+          *reinterpret_cast<unsigned short*>(
+              aw + row * (QT * 2) + swz(row, col * 2)) =
+              float_to_bf16_bits(pt[nf][r]);
+        }
+      }
+#pragma unroll
+      for (int kk2 = 0; kk2 < QT / 32; ++kk2) {
+        bf16x8 pa = lds_frag(aw, cl, QT * 2, (kk2 * 32 + g * 8) * 2);
+#pragma unroll
+        for (int hf = 0; hf < HF; ++hf) {
+          bf16x8 bd =
+              lds_frag(dot_lds, hf * 16 + cl, QT * 2, (kk2 * 32 + g * 8) * 2);
+          acc_dv[hf] = mfma16x16x32_bf16(pa, bd, acc_dv[hf]);
+        }
+      }
+
+      // dP^T = V @ dO^T (B from do_lds rows).
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int kk = 0; kk < KH; ++kk) {
+          bf16x8 bd =
+              lds_frag(do_lds, nf * 16 + cl, ROWB, (kk * 32 + g * 8) * 2);
+          acc = mfma16x16x32_bf16(vfrag[kk], bd, acc);
+        }
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float dlogits = pt[nf][r] * (acc[r] - delta_s[nf * 16 + cl]);
+          dlg[nf][r] = dlogits;
+          if (BIAS_GRAD) {
+            const int qcol = qb + nf * 16 + cl;
+            const int key = k0 + g * 4 + r;
+            if (dlogits != 0.f) {
+              int d = qcol - key;
+              d = d < -bias_clip ? -bias_clip
+                                 : (d > bias_clip ? bias_clip : d);
+              atomicAdd(&dbias_s[d + bias_clip], dlogits);
+            }
+          }
+        }
+      }
+
+      // dS^T (scaled, bf16) -> a_lds (overwrite); dK += dS^T @ Q (B=Qt).
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int row = g * 4 + r;
+          int col = nf * 16 + cl;
+          *reinterpret_cast<unsigned short*>(
+              aw + row * (QT * 2) + swz(row, col * 2)) =
+              float_to_bf16_bits(dlg[nf][r] * scale);
+        }
+      }
+#pragma unroll
+      for (int kk2 = 0; kk2 < QT / 32; ++kk2) {
+        bf16x8 da = lds_frag(aw, cl, QT * 2, (kk2 * 32 + g * 8) * 2);
+#pragma unroll
+        for (int hf = 0; hf < HF; ++hf) {
+          bf16x8 bq =
+              lds_frag(qt_lds, hf * 16 + cl, QT * 2, (kk2 * 32 + g * 8) * 2);
+          acc_dk[hf] = mfma16x16x32_bf16(da, bq, acc_dk[hf]);
+        }
+      }
+
+      // Full dS tile to LDS (transposed store: ds_lds[q][key]) for dQ.
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int qrow = nf * 16 + cl;
+          int key = wid * 16 + g * 4 + r;
+          *reinterpret_cast<unsigned short*>(
+              ds_lds + qrow * (KT * 2) + swz(qrow, key * 2)) =
+              float_to_bf16_bits(dlg[nf][r] * scale);
+        }
+      }
+      __syncthreads();
+
+      // dQ strip for this wave: rows q = qb + wid*16 + cl (A from ds_lds),
+      // B = K from kt_lds; atomic-accumulate fp32.
+      f32x4 acc_dq[HF];
+#pragma unroll
+      for (int hf = 0; hf < HF; ++hf) acc_dq[hf] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int kk2 = 0; kk2 < KT / 32; ++kk2) {
+        bf16x8 da =
+            lds_frag(ds_lds, wid * 16 + cl, KT * 2, (kk2 * 32 + g * 8) * 2);
+#pragma unroll
+        for (int hf = 0; hf < HF; ++hf) {
+          bf16x8 bk =
+              lds_frag(kt_lds, hf * 16 + cl, KT * 2, (kk2 * 32 + g * 8) * 2);
+          acc_dq[hf] = mfma16x16x32_bf16(da, bk, acc_dq[hf]);
+        }
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int qrow = qb + wid * 16 + g * 4 + r;
+        if (qrow >= T) continue;
+#pragma unroll
+        for (int hf = 0; hf < HF; ++hf) {
+          if (acc_dq[hf][r] != 0.f) {
+            atomicAdd(
+                dq_acc + (((long)b * T + qrow) * N + n) * H + hf * 16 + cl,
+                acc_dq[hf][r]);
+          }
+        }
+      }
+    }
+
+    // Flush dbias for this head.
+    if (BIAS_GRAD) {
+      __syncthreads();
+      for (int i = threadIdx.x; i < nbias; i += BLOCK) {
+        if (dbias_s[i] != 0.f)
+          atomicAdd(dbias + (long)n * nbias + i, dbias_s[i]);
+      }
+    }
+
+    // dK/dV for this head's contribution: with GQA we accumulate across
+    // heads; write after the head loop. To keep single-write semantics we
+    // only write on the last head, so keep accumulating into global via
+    // registers: here we add to global bf16 once per head via fp32 staging
+    // in registers across heads instead. Simpler: accumulate across heads
+    // in registers by NOT resetting acc_dk/acc_dv — but they are reset per
+    // head above; so write-accumulate here:
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int key = k0 + g * 4 + r;
+      if (key >= S) continue;  // acc is 0 for klen<=key<S: writes zeros
+      long off = (((long)b * S + key) * NKV + nkv) * H;
+#pragma unroll
+      for (int hf = 0; hf < HF; ++hf) {
+        if (group == 1) {
+          dk[off + hf * 16 + cl] = float_to_bf16_bits(acc_dk[hf][r]);
+          dv[off + hf * 16 + cl] = float_to_bf16_bits(acc_dv[hf][r]);
+        } else {
+          // GQA: multiple heads accumulate; use fp32 atomics into dq_acc?
+          // dk/dv are bf16 outputs; for group>1 the host passes fp32
+          // buffers aliased via dk/dv pointers being fp32. Handled on the
+          // host by allocating fp32 and a separate cast. Here: atomicAdd.
+          atomicAdd(reinterpret_cast<float*>(dk) + off + hf * 16 + cl,
+                    acc_dk[hf][r]);
+          atomicAdd(reinterpret_cast<float*>(dv) + off + hf * 16 + cl,
+                    acc_dv[hf][r]);
+        }
+      }
+    }
+    __syncthreads();
+  }
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// Host wrappers
+// ---------------------------------------------------------------------------
+static void check_btnh(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda() && t.is_contiguous() && t.dim() == 4 &&
+                  t.scalar_type() == torch::kBFloat16,
+              name, " must be contiguous bf16 [B,T,N,H]");
+}
+
+std::vector<torch::Tensor> fa_fwd(torch::Tensor q, torch::Tensor k,
+                                  torch::Tensor v,
+                                  c10::optional<torch::Tensor> klen,
+                                  c10::optional<torch::Tensor> bias,
+                                  int64_t win_l, int64_t win_r,
+                                  int64_t bias_clip, double scale) {
+  check_btnh(q, "q");
+  check_btnh(k, "k");
+  check_btnh(v, "v");
+  const int B = q.size(0), T = q.size(1), N = q.size(2), H = q.size(3);
+  const int S = k.size(1), NKV = k.size(2);
+  TORCH_CHECK(H == 64 || H == 128, "H must be 64 or 128, got ", H);
+  TORCH_CHECK(N % NKV == 0, "GQA requires N % NKV == 0");
+  auto o = torch::empty_like(q);
+  auto lse = torch::empty({B, N, T}, q.options().dtype(torch::kFloat32));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  dim3 grid((T + QT - 1) / QT, N, B);
+  const int* klp = klen.has_value() ? klen->data_ptr<int>() : nullptr;
+  const unsigned short* bp =
+      bias.has_value() ? (const unsigned short*)bias->data_ptr() : nullptr;
+  size_t shmem = (size_t)KT * H * 2 + (size_t)H * KT * 2 + NW * 16 * KT * 2;
+#define FA_FWD(HH)                                                          \
+  hipLaunchKernelGGL((fa_fwd_kernel<HH>), grid, dim3(BLOCK), shmem, stream, \
+                     (const unsigned short*)q.data_ptr(),                   \
+                     (const unsigned short*)k.data_ptr(),                   \
+                     (const unsigned short*)v.data_ptr(), klp, bp,          \
+                     (unsigned short*)o.data_ptr(), lse.data_ptr<float>(),  \
+                     B, T, S, N, NKV, (int)win_l, (int)win_r,               \
+                     (int)bias_clip, (float)scale)
+  if (H == 64) {
+    FA_FWD(64);
+  } else {
+    FA_FWD(128);
+  }
+#undef FA_FWD
+  return {o, lse};
+}
+
+std::vector<torch::Tensor> fa_bwd(torch::Tensor dout, torch::Tensor q,
+                                  torch::Tensor k, torch::Tensor v,
+                                  torch::Tensor o, torch::Tensor lse,
+                                  c10::optional<torch::Tensor> klen,
+                                  c10::optional<torch::Tensor> bias,
+                                  bool bias_grad, int64_t win_l,
+                                  int64_t win_r, int64_t bias_clip,
+                                  double scale) {
+  check_btnh(q, "q");
+  const int B = q.size(0), T = q.size(1), N = q.size(2), H = q.size(3);
+  const int S = k.size(1), NKV = k.size(2);
+  const int group = N / NKV;
+  auto stream = at::cuda::getCurrentCUDAStream();
+
+  auto delta = torch::empty({B, N, T}, q.options().dtype(torch::kFloat32));
+  {
+    long rows = (long)B * T * N;
+    dim3 grid((rows + NW - 1) / NW);
+    if (H == 64)
+      hipLaunchKernelGGL((fa_bwd_delta<64>), grid, dim3(BLOCK), 0, stream,
+                         (const unsigned short*)dout.data_ptr(),
+                         (const unsigned short*)o.data_ptr(),
+                         delta.data_ptr<float>(), B, T, N);
+    else
+      hipLaunchKernelGGL((fa_bwd_delta<128>), grid, dim3(BLOCK), 0, stream,
+                         (const unsigned short*)dout.data_ptr(),
+                         (const unsigned short*)o.data_ptr(),
+                         delta.data_ptr<float>(), B, T, N);
+  }
+
+  auto dq_acc = torch::zeros_like(q, q.options().dtype(torch::kFloat32));
+  torch::Tensor dk_t, dv_t;
+  if (group == 1) {
+    dk_t = torch::empty_like(k);
+    dv_t = torch::empty_like(v);
+  } else {
+    dk_t = torch::zeros_like(k, k.options().dtype(torch::kFloat32));
+    dv_t = torch::zeros_like(v, v.options().dtype(torch::kFloat32));
+  }
+  const int nbias = 2 * (int)bias_clip + 1;
+  torch::Tensor dbias_t = torch::zeros(
+      {bias.has_value() ? N : 0, bias.has_value() ? nbias : 0},
+      q.options().dtype(torch::kFloat32));
+
+  const int* klp = klen.has_value() ? klen->data_ptr<int>() : nullptr;
+  const unsigned short* bp =
+      bias.has_value() ? (const unsigned short*)bias->data_ptr() : nullptr;
+  bool bg = bias.has_value() && bias_grad;
+
+  size_t shmem = (size_t)QT * H * 2 * 2   // q_lds + do_lds
+                 + (size_t)H * QT * 2 * 2  // qt_lds + dot_lds
+                 + (size_t)H * KT * 2      // kt_lds
+                 + (size_t)QT * KT * 2     // ds_lds
+                 + (size_t)NW * 16 * QT * 2  // a_lds
+                 + 2 * QT * sizeof(float) + (bg ? nbias * sizeof(float) : 0);
+  dim3 grid((S + KT - 1) / KT, NKV, B);
+#define FA_BWD(HH, BG)                                                       \
+  hipLaunchKernelGGL(                                                        \
+      (fa_bwd_kernel<HH, BG>), grid, dim3(BLOCK), shmem, stream,             \
+      (const unsigned short*)dout.data_ptr(),                                \
+      (const unsigned short*)q.data_ptr(),                                   \
+      (const unsigned short*)k.data_ptr(),                                   \
+      (const unsigned short*)v.data_ptr(), lse.data_ptr<float>(),            \
+      delta.data_ptr<float>(), klp, bp, dq_acc.data_ptr<float>(),            \
+      (unsigned short*)dk_t.data_ptr(), (unsigned short*)dv_t.data_ptr(),    \
+      dbias_t.numel() ? dbias_t.data_ptr<float>() : nullptr, B, T, S, N,     \
+      NKV, (int)win_l, (int)win_r, (int)bias_clip, (float)scale)
+  if (H == 64) {
+    if (bg) FA_BWD(64, true); else FA_BWD(64, false);
+  } else {
+    if (bg) FA_BWD(128, true); else FA_BWD(128, false);
+  }
+#undef FA_BWD
+
+  auto dq = dq_acc.to(torch::kBFloat16);
+  auto dk = group == 1 ? dk_t : dk_t.to(torch::kBFloat16);
+  auto dv = group == 1 ? dv_t : dv_t.to(torch::kBFloat16);
+  return {dq, dk, dv, dbias_t};
+}
