@@ -1,0 +1,130 @@
+"""Flagship benchmark: Librispeech Conformer-L train step (BASELINE.json).
+
+  python bench.py --gpus N --steps K --warmup W
+
+Measures whole-job examples/sec for the Conformer-L ASR train step on
+synthetic Librispeech-shaped data (random-init weights, bf16 compute),
+DP over RCCL/xGMI for N>1. Launched by the driver via
+torch.distributed.run for N>1 (one rank per GPU).
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+import torch
+
+
+def main():
+  ap = argparse.ArgumentParser()
+  ap.add_argument('--gpus', type=int, default=1)
+  ap.add_argument('--steps', type=int, default=20)
+  ap.add_argument('--warmup', type=int, default=5)
+  ap.add_argument('--batch', type=int, default=16,
+                  help='Per-GPU batch size.')
+  ap.add_argument('--model', default='asr.librispeech.'
+                  'Librispeech960WpmConformerL')
+  args = ap.parse_args()
+
+  import torch.distributed as dist
+  from lingvo_amd.core import registry
+  from lingvo_amd.parallel import ddp
+
+  world = int(os.environ.get('WORLD_SIZE', '1'))
+  rank = ddp.InitDistributed()
+  local_rank = int(os.environ.get('LOCAL_RANK', '0'))
+  has_gpu = torch.cuda.is_available()
+  device = f'cuda:{local_rank}' if has_gpu else 'cpu'
+  if has_gpu:
+    torch.cuda.set_device(device)
+
+  model_p = registry.GetParams(args.model, 'Train')
+  model_p.input.batch_size = args.batch
+  if not has_gpu:  # CPU smoke of the bench harness only
+    model_p.task.fprop_dtype = torch.float32
+    model_p.input.frame_len = 80
+    model_p.task.encoder.num_layers = 1
+  model_p.task.random_seed = 1234
+  model = model_p.Instantiate().to(device)
+  task = model.GetTask()
+  sync = ddp.GradSync(task) if world > 1 else None
+  finalize = sync.Finalize if sync else None
+
+  # Pre-generate a handful of synthetic batches on device.
+  gen = task.input_generator
+  batches = []
+  for _ in range(4):
+    b = gen.GetPreprocessedInputBatch()
+    batches.append(b.Transform(
+        lambda t: t.to(device) if isinstance(t, torch.Tensor) else t))
+
+  def step(i):
+    metrics = task.TrainStep(batches[i % len(batches)],
+                             grad_sync_finalize=finalize)
+    return metrics
+
+  for i in range(args.warmup):
+    step(i)
+
+  if world > 1:
+    dist.barrier()
+  if has_gpu:
+    torch.cuda.synchronize()
+  t0 = time.perf_counter()
+  for i in range(args.steps):
+    metrics = step(i)
+  if has_gpu:
+    torch.cuda.synchronize()
+  if world > 1:
+    dist.barrier()
+  elapsed = time.perf_counter() - t0
+
+  # MAX step time over ranks == MIN throughput: reduce elapsed as MAX.
+  if world > 1:
+    e = torch.tensor([elapsed], device=device if has_gpu else 'cpu')
+    dist.all_reduce(e, op=dist.ReduceOp.MAX)
+    elapsed = float(e.item())
+
+  ms_per_step = elapsed / args.steps * 1000.0
+  global_batch = args.batch * world
+  examples_per_sec = global_batch * args.steps / elapsed
+  loss = float(metrics['loss'][0])
+
+  if rank == 0:
+    out = {
+        'metric': 'train step examples/sec (whole node), Librispeech '
+                  'Conformer-L',
+        'value': round(examples_per_sec, 3),
+        'unit': 'examples/sec',
+        'n_gpus': world,
+        'steps': args.steps,
+        'warmup': args.warmup,
+        'ms_per_step': round(ms_per_step, 3),
+        'higher_is_better': True,
+        'scaling': 'weak',
+        'vs_baseline': None,
+        'dtype': 'bf16' if has_gpu else 'fp32',
+        'data': 'synthetic',
+        'config': {
+            'model': 'Conformer-L (17 blocks, d=512, h=8, kernel 32) + '
+                     'LSTM attention decoder',
+            'global_batch': global_batch,
+            'seq_len': model_p.input.frame_len,
+            'target_len': model_p.input.target_len,
+            'parallelism': f'dp{world}',
+            'final_loss': round(loss, 4),
+        },
+    }
+    print(json.dumps(out))
+  if world > 1:
+    dist.destroy_process_group()
+
+
+if __name__ == '__main__':
+  main()

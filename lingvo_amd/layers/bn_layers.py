@@ -1,0 +1,127 @@
+"""Normalization layers with padding awareness
+(reference lingvo/core/bn_layers.py: BatchNormLayer:139, GroupNormLayer:747).
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+from lingvo_amd.core import py_utils
+from lingvo_amd.core.base_layer import BaseLayer
+from lingvo_amd.core.nested_map import NestedMap
+
+
+class BatchNormLayer(BaseLayer):
+  """Padded-aware batch norm over [B, T, D] (moments exclude padded
+  frames, reference bn_layers.py:139). Optional cross-replica sync."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('dim', 0, 'Feature dim.')
+    p.Define('decay', 0.999, 'Running-moment decay.')
+    p.Define('epsilon', 1e-3, 'Epsilon.')
+    p.Define('enable_cross_replica_sum_on_tpu', False,
+             'Sync moments across DP ranks (RCCL all-reduce).')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    self.CreateVariable('beta', py_utils.WeightParams(
+        [p.dim], py_utils.WeightInit.Constant(0.0), p.dtype))
+    self.CreateVariable('gamma', py_utils.WeightParams(
+        [p.dim], py_utils.WeightInit.Constant(0.0), p.dtype))
+    self.register_buffer('moving_mean', torch.zeros(p.dim))
+    self.register_buffer('moving_variance', torch.ones(p.dim))
+
+  def _Moments(self, x: torch.Tensor, paddings: Optional[torch.Tensor]):
+    xf = x.float()
+    if paddings is None:
+      count = torch.tensor(float(x.numel() // x.shape[-1]),
+                           device=x.device)
+      mean = xf.mean(dim=tuple(range(x.dim() - 1)))
+      var = xf.var(dim=tuple(range(x.dim() - 1)), unbiased=False)
+    else:
+      mask = (1.0 - paddings).float()
+      while mask.dim() < x.dim():
+        mask = mask.unsqueeze(-1)
+      count = mask.sum() * 1.0
+      mean = (xf * mask).sum(dim=tuple(range(x.dim() - 1))) / \
+          count.clamp_min(1.0)
+      var = ((xf - mean) ** 2 * mask).sum(
+          dim=tuple(range(x.dim() - 1))) / count.clamp_min(1.0)
+    if self.p.enable_cross_replica_sum_on_tpu:
+      import torch.distributed as dist
+      if dist.is_available() and dist.is_initialized():
+        stats = torch.cat([mean * count, var * count, count.reshape(1)])
+        dist.all_reduce(stats)
+        n = stats[-1].clamp_min(1.0)
+        mean = stats[:self.p.dim] / n
+        var = stats[self.p.dim:2 * self.p.dim] / n
+    return mean, var
+
+  def FProp(self, theta: NestedMap, inputs: torch.Tensor,
+            paddings: Optional[torch.Tensor] = None) -> torch.Tensor:
+    p = self.p
+    if self.do_eval:
+      mean, var = self.moving_mean, self.moving_variance
+    else:
+      mean, var = self._Moments(inputs, paddings)
+      with torch.no_grad():
+        self.moving_mean.mul_(p.decay).add_(mean.detach(),
+                                            alpha=1 - p.decay)
+        self.moving_variance.mul_(p.decay).add_(var.detach(),
+                                                alpha=1 - p.decay)
+    out = (inputs.float() - mean) * torch.rsqrt(var + p.epsilon)
+    out = out * (1.0 + theta.gamma.float()) + theta.beta.float()
+    out = out.to(inputs.dtype)
+    if paddings is not None:
+      out = py_utils.ApplyPadding(paddings, out)
+    return out
+
+
+class GroupNormLayer(BaseLayer):
+  """Group norm over [B, T, D] with paddings (reference bn_layers.py:747)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('dim', 0, 'Feature dim.')
+    p.Define('num_groups', 32, 'Groups.')
+    p.Define('epsilon', 1e-3, 'Epsilon.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    assert p.dim % p.num_groups == 0
+    self.CreateVariable('beta', py_utils.WeightParams(
+        [p.dim], py_utils.WeightInit.Constant(0.0), p.dtype))
+    self.CreateVariable('gamma', py_utils.WeightParams(
+        [p.dim], py_utils.WeightInit.Constant(0.0), p.dtype))
+
+  def FProp(self, theta: NestedMap, inputs: torch.Tensor,
+            paddings: Optional[torch.Tensor] = None) -> torch.Tensor:
+    p = self.p
+    b, t, d = inputs.shape
+    g = p.num_groups
+    xf = inputs.float().reshape(b, t, g, d // g)
+    # Per (b, group) moments over (t, d/g), excluding padded frames.
+    if paddings is not None:
+      mask = (1.0 - paddings).float()[:, :, None, None]
+      count = (mask.sum(dim=(1, 3), keepdim=True) * (d // g)).clamp_min(1.0)
+      mean = (xf * mask).sum(dim=(1, 3), keepdim=True) / count
+      var = ((xf - mean) ** 2 * mask).sum(dim=(1, 3), keepdim=True) / count
+    else:
+      mean = xf.mean(dim=(1, 3), keepdim=True)
+      var = xf.var(dim=(1, 3), unbiased=False, keepdim=True)
+    out = (xf - mean) * torch.rsqrt(var + p.epsilon)
+    out = out.reshape(b, t, d)
+    out = out * (1.0 + theta.gamma.float()) + theta.beta.float()
+    out = out.to(inputs.dtype)
+    if paddings is not None:
+      out = py_utils.ApplyPadding(paddings, out)
+    return out

@@ -1,0 +1,212 @@
+"""Conformer block (reference lingvo/core/conformer_layer.py:35 LConvLayer,
+:471 ConformerLayer, CommonParams :566).
+
+Block order (the paper's and the reference default 'mhsa_before_conv'):
+  x += 0.5*FFN(x); x += MHSA(x); x += LConv(x); x += 0.5*FFN(x); x = LN(x)
+LConv: LN -> pointwise 2D GLU -> depthwise time conv (HIP kernel K7)
+ -> norm -> swish -> pointwise -> dropout.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.nn.functional as F
+
+from lingvo_amd.core import py_utils
+from lingvo_amd.core.base_layer import BaseLayer
+from lingvo_amd.core.nested_map import NestedMap
+from lingvo_amd.layers import attention as attention_lib
+from lingvo_amd.layers import bn_layers
+from lingvo_amd.layers import layers as lingvo_layers
+from lingvo_amd.layers import transformer as transformer_lib
+from lingvo_amd.ops import conv1d as conv1d_ops
+
+
+class LConvLayer(BaseLayer):
+  """Lightweight convolution module (reference conformer_layer.py:35)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('input_dim', 0, 'Model dim.')
+    p.Define('kernel_size', 32, 'Depthwise conv kernel size.')
+    p.Define('is_causal', False, 'Causal depthwise conv.')
+    p.Define('conv_norm', 'group', "One of 'batch'|'group'|'layer'.")
+    p.Define('num_groups', 32, 'Groups for group norm.')
+    p.Define('dropout_prob', 0.0, 'Output dropout.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    d = p.input_dim
+    self.CreateChild('ln', lingvo_layers.LayerNorm.Params().Set(
+        input_dim=d))
+    self.CreateVariable('pw1_w', py_utils.WeightParams(
+        [d, 2 * d], p.params_init, p.dtype))
+    self.CreateVariable('pw1_b', py_utils.WeightParams(
+        [2 * d], py_utils.WeightInit.Constant(0.0), p.dtype))
+    self.CreateVariable('dw_w', py_utils.WeightParams(
+        [p.kernel_size, d], p.params_init, p.dtype))
+    self.CreateVariable('dw_b', py_utils.WeightParams(
+        [d], py_utils.WeightInit.Constant(0.0), p.dtype))
+    if p.conv_norm == 'batch':
+      self.CreateChild('norm', bn_layers.BatchNormLayer.Params().Set(dim=d))
+    elif p.conv_norm == 'group':
+      self.CreateChild('norm', bn_layers.GroupNormLayer.Params().Set(
+          dim=d, num_groups=p.num_groups))
+    else:
+      self.CreateChild('norm', lingvo_layers.LayerNorm.Params().Set(
+          input_dim=d))
+    self.CreateVariable('pw2_w', py_utils.WeightParams(
+        [d, d], p.params_init, p.dtype))
+    self.CreateVariable('pw2_b', py_utils.WeightParams(
+        [d], py_utils.WeightInit.Constant(0.0), p.dtype))
+
+  def FProp(self, theta: NestedMap, inputs: torch.Tensor,
+            paddings: Optional[torch.Tensor] = None) -> torch.Tensor:
+    p = self.p
+    x = self.ln.FProp(theta.ln, inputs)
+    x = torch.matmul(x, theta.pw1_w) + theta.pw1_b
+    a, b = x.chunk(2, dim=-1)
+    x = a * torch.sigmoid(b)  # GLU
+    if paddings is not None:
+      x = py_utils.ApplyPadding(paddings, x)
+    x = conv1d_ops.depthwise_conv1d(x, theta.dw_w, theta.dw_b,
+                                    causal=p.is_causal)
+    if p.conv_norm == 'layer':
+      x = self.norm.FProp(theta.norm, x)
+      if paddings is not None:
+        x = py_utils.ApplyPadding(paddings, x)
+    else:
+      x = self.norm.FProp(theta.norm, x, paddings)
+    x = F.silu(x)
+    x = torch.matmul(x, theta.pw2_w) + theta.pw2_b
+    if p.dropout_prob and not self.do_eval:
+      x = py_utils.DeterministicDropout(x, 1.0 - p.dropout_prob)
+    if paddings is not None:
+      x = py_utils.ApplyPadding(paddings, x)
+    return inputs + x
+
+
+class ConformerLayer(BaseLayer):
+  """½FFN -> MHSA -> LConv -> ½FFN -> LN (reference conformer_layer.py:471)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('input_dim', 0, 'Model dim.')
+    p.Define('atten_num_heads', 8, 'MHSA heads.')
+    p.Define('atten_left_context', -1, 'Local attention left window.')
+    p.Define('atten_right_context', -1, 'Local attention right window.')
+    p.Define('use_relative_atten', True, 'Clipped rel-pos bias in MHSA.')
+    p.Define('rel_pos_clip', 127, 'Rel-pos clip distance.')
+    p.Define('fflayer_hidden_dim', 0, 'FFN hidden (0 = 4x).')
+    p.Define('kernel_size', 32, 'LConv kernel size.')
+    p.Define('is_causal', False, 'Causal conv + attention.')
+    p.Define('conv_norm', 'group', 'LConv norm type.')
+    p.Define('dropout_prob', 0.0, 'Dropout throughout.')
+    p.Define('remat', False, 'Gradient-checkpoint this layer '
+             '(reference conformer_layer.py:548 p.remat).')
+    return p
+
+  @classmethod
+  def CommonParams(cls, input_dim, atten_num_heads=8, kernel_size=32,
+                   is_causal=False, dropout_prob=0.0):
+    return cls.Params().Set(
+        input_dim=input_dim, atten_num_heads=atten_num_heads,
+        kernel_size=kernel_size, is_causal=is_causal,
+        dropout_prob=dropout_prob)
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    d = p.input_dim
+    hidden = p.fflayer_hidden_dim or 4 * d
+    ff = transformer_lib.TransformerFeedForwardLayer.Params().Set(
+        input_dim=d, hidden_dim=hidden, activation='SWISH',
+        residual_weight=0.5, residual_dropout_prob=p.dropout_prob,
+        relu_dropout_prob=p.dropout_prob)
+    self.CreateChild('fflayer_start', ff.Copy())
+    self.CreateChild('fflayer_end', ff.Copy())
+    atten = transformer_lib.TransformerAttentionLayer.Params().Set(
+        input_dim=d, num_heads=p.atten_num_heads, is_masked=p.is_causal,
+        residual_dropout_prob=p.dropout_prob)
+    atten.atten_tpl.rel_pos_bias = p.use_relative_atten
+    atten.atten_tpl.rel_pos_clip = p.rel_pos_clip
+    atten.atten_tpl.left_context = p.atten_left_context
+    atten.atten_tpl.right_context = p.atten_right_context
+    atten.atten_tpl.atten_dropout_prob = p.dropout_prob
+    self.CreateChild('trans_atten', atten)
+    self.CreateChild('lconv', LConvLayer.Params().Set(
+        input_dim=d, kernel_size=p.kernel_size, is_causal=p.is_causal,
+        conv_norm=p.conv_norm, dropout_prob=p.dropout_prob))
+    self.CreateChild('final_ln', lingvo_layers.LayerNorm.Params().Set(
+        input_dim=d))
+
+  def _Body(self, theta, x, paddings):
+    x = self.fflayer_start.FProp(theta.fflayer_start, x, paddings)
+    x = self.trans_atten.FProp(theta.trans_atten, x, paddings)
+    x = self.lconv.FProp(theta.lconv, x, paddings)
+    x = self.fflayer_end.FProp(theta.fflayer_end, x, paddings)
+    return self.final_ln.FProp(theta.final_ln, x)
+
+  def FProp(self, theta: NestedMap, inputs: torch.Tensor,
+            paddings: Optional[torch.Tensor] = None) -> torch.Tensor:
+    if self.p.remat and self.training:
+      return torch.utils.checkpoint.checkpoint(
+          lambda x: self._Body(theta, x, paddings), inputs,
+          use_reentrant=False)
+    return self._Body(theta, inputs, paddings)
+
+
+class ConvSubsampling(BaseLayer):
+  """2x Conv2D stride-2 frontend: [B, T, F] mel -> [B, T/4, D]
+  (reference tasks/asr/encoder conv subsampling)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('input_freq_dim', 80, 'Mel bins.')
+    p.Define('channels', 0, 'Conv channels (defaults to output_dim).')
+    p.Define('output_dim', 512, 'Output model dim.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    ch = p.channels or p.output_dim
+    self._ch = ch
+    self.CreateVariable('conv1_w', py_utils.WeightParams(
+        [3, 3, 1, ch], p.params_init, p.dtype))
+    self.CreateVariable('conv1_b', py_utils.WeightParams(
+        [ch], py_utils.WeightInit.Constant(0.0), p.dtype))
+    self.CreateVariable('conv2_w', py_utils.WeightParams(
+        [3, 3, ch, ch], p.params_init, p.dtype))
+    self.CreateVariable('conv2_b', py_utils.WeightParams(
+        [ch], py_utils.WeightInit.Constant(0.0), p.dtype))
+    freq_out = ((p.input_freq_dim + 1) // 2 + 1) // 2
+    self.CreateVariable('proj_w', py_utils.WeightParams(
+        [freq_out * ch, p.output_dim], p.params_init, p.dtype))
+    self.CreateVariable('proj_b', py_utils.WeightParams(
+        [p.output_dim], py_utils.WeightInit.Constant(0.0), p.dtype))
+
+  def FProp(self, theta: NestedMap, inputs: torch.Tensor,
+            paddings: torch.Tensor):
+    """inputs [B, T, F] -> (out [B, ceil(T/4), D], out_paddings)."""
+    x = inputs.unsqueeze(1)  # [B,1,T,F]
+    w1 = theta.conv1_w.permute(3, 2, 0, 1)
+    x = F.conv2d(x, w1, theta.conv1_b, stride=2, padding=1)
+    x = F.relu(x)
+    w2 = theta.conv2_w.permute(3, 2, 0, 1)
+    x = F.conv2d(x, w2, theta.conv2_b, stride=2, padding=1)
+    x = F.relu(x)  # [B, ch, T/4, F/4]
+    b, ch, t4, f4 = x.shape
+    x = x.permute(0, 2, 3, 1).reshape(b, t4, f4 * ch)
+    out = torch.matmul(x, theta.proj_w) + theta.proj_b
+    out_paddings = paddings[:, ::2][:, ::2]
+    out_paddings = out_paddings[:, :t4]
+    out = py_utils.ApplyPadding(out_paddings, out)
+    return out, out_paddings

@@ -1,0 +1,268 @@
+"""ASR task: Conformer encoder + attention LSTM decoder (LAS-style).
+
+Reference: lingvo/tasks/asr/model.py:30 AsrModel, encoder.py:32,
+decoder.py:48; Conformer blocks from lingvo/core/conformer_layer.py:471.
+The Librispeech Conformer-L config (tasks/asr/params/librispeech.py
+composes the LAS baseline; the Conformer-L encoder follows the Conformer
+paper: 17 blocks, d=512, h=8, conv kernel 32) is the BASELINE.json
+north-star model.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional, Tuple
+
+import torch
+import torch.nn.functional as F
+
+from lingvo_amd.core import py_utils
+from lingvo_amd.core.base_input_generator import BaseSequenceInputGenerator
+from lingvo_amd.core.base_layer import BaseLayer
+from lingvo_amd.core.base_model import BaseTask
+from lingvo_amd.core.nested_map import NestedMap
+from lingvo_amd.layers import conformer as conformer_lib
+from lingvo_amd.layers import layers as lingvo_layers
+from lingvo_amd.layers import rnn_cell
+from lingvo_amd.core import metrics as metrics_lib
+
+
+class SyntheticAsrInput(BaseSequenceInputGenerator):
+  """Synthetic Librispeech-shaped batches: [B, T, 80] log-mel + token ids
+  (no network for real data; shapes match librispeech.py:42 80-dim
+  features)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.batch_size = 16
+    p.Define('frame_len', 1200, 'Input frames T (10ms frames).')
+    p.Define('feature_dim', 80, 'Mel bins.')
+    p.Define('target_len', 64, 'Target tokens L.')
+    p.Define('vocab_size', 1024, 'WPM vocab size.')
+    return p
+
+  def _InputBatch(self) -> NestedMap:
+    p = self.p
+    g = torch.Generator().manual_seed(1000 + self._batch_count)
+    b, t, f, l = p.batch_size, p.frame_len, p.feature_dim, p.target_len
+    src = torch.randn(b, t, f, generator=g)
+    src_lens = torch.randint(int(0.8 * t), t + 1, (b,), generator=g)
+    src_paddings = py_utils.PaddingsFromLengths(src_lens, t)
+    ids = torch.randint(2, p.vocab_size, (b, l), generator=g)
+    tgt_lens = torch.randint(int(0.75 * l), l + 1, (b,), generator=g)
+    tgt_paddings = py_utils.PaddingsFromLengths(tgt_lens, l)
+    ids = (ids * (1 - tgt_paddings).long())
+    return NestedMap(
+        src=NestedMap(src_inputs=src, paddings=src_paddings),
+        tgt=NestedMap(ids=ids, paddings=tgt_paddings,
+                      labels=ids.roll(-1, dims=1) * (1 - tgt_paddings).long(),
+                      weights=1.0 - tgt_paddings))
+
+
+class ConformerEncoder(BaseLayer):
+  """Subsampling frontend + N Conformer blocks
+  (reference tasks/asr/encoder.py shape; core/conformer_layer.py blocks)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('input_dim', 80, 'Mel bins.')
+    p.Define('model_dim', 512, 'Encoder dim.')
+    p.Define('num_layers', 17, 'Conformer blocks.')
+    p.Define('num_heads', 8, 'MHSA heads.')
+    p.Define('kernel_size', 32, 'LConv kernel.')
+    p.Define('dropout_prob', 0.1, 'Dropout.')
+    p.Define('remat', False, 'Checkpoint each block.')
+    p.Define('subsample_channels', 0, 'Frontend conv channels '
+             '(0 = model_dim).')
+    p.Define('conformer_tpl', conformer_lib.ConformerLayer.Params(),
+             'Block template.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    self.CreateChild('sub', conformer_lib.ConvSubsampling.Params().Set(
+        input_freq_dim=p.input_dim, output_dim=p.model_dim,
+        channels=p.subsample_channels or min(p.model_dim, 256)))
+    blocks = []
+    for i in range(p.num_layers):
+      bp = p.conformer_tpl.Copy().Set(
+          name=f'conformer_{i}', input_dim=p.model_dim,
+          atten_num_heads=p.num_heads, kernel_size=p.kernel_size,
+          dropout_prob=p.dropout_prob, remat=p.remat)
+      blocks.append(bp)
+    self.CreateChildren('blocks', blocks)
+
+  def FProp(self, theta: NestedMap, src_inputs: torch.Tensor,
+            paddings: torch.Tensor):
+    x = src_inputs.to(self.fprop_dtype)
+    x, out_pad = self.sub.FProp(theta.sub, x, paddings)
+    for i, block in enumerate(self.blocks):
+      x = block.FProp(theta.blocks[i], x, out_pad)
+    return x, out_pad
+
+
+class AsrDecoder(BaseLayer):
+  """Teacher-forced attention LSTM decoder (LAS-style,
+  reference tasks/asr/decoder.py:48). Dot-product attention over
+  projected encoder outputs, 2 LSTM layers, shared softmax."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('vocab_size', 1024, 'Output vocab.')
+    p.Define('emb_dim', 128, 'Token embedding dim.')
+    p.Define('rnn_cell_dim', 640, 'LSTM dim.')
+    p.Define('num_lstm_layers', 2, 'LSTM layers.')
+    p.Define('source_dim', 512, 'Encoder output dim.')
+    p.Define('dropout_prob', 0.1, 'Dropout.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    self.CreateChild('emb', lingvo_layers.EmbeddingLayer.Params().Set(
+        vocab_size=p.vocab_size, embedding_dim=p.emb_dim))
+    self.CreateVariable('atten_query_w', py_utils.WeightParams(
+        [p.rnn_cell_dim, p.source_dim], p.params_init, p.dtype))
+    cells = []
+    for i in range(p.num_lstm_layers):
+      in_dim = (p.emb_dim + p.source_dim) if i == 0 else p.rnn_cell_dim
+      cells.append(rnn_cell.LSTMCellSimple.Params().Set(
+          name=f'lstm_{i}', num_input_nodes=in_dim,
+          num_output_nodes=p.rnn_cell_dim))
+    self.CreateChildren('rnns', cells)
+    self.CreateChild('softmax', lingvo_layers.SimpleFullSoftmax.Params().Set(
+        input_dim=p.rnn_cell_dim + p.source_dim, num_classes=p.vocab_size))
+
+  def _Attend(self, theta: NestedMap, query_m: torch.Tensor,
+              enc: torch.Tensor, enc_paddings: torch.Tensor):
+    """Dot-product attention: query [B,H] x enc [B,S,D] -> ctx [B,D]."""
+    q = torch.matmul(query_m, theta.atten_query_w)  # [B, D]
+    logits = torch.einsum('bd,bsd->bs', q.float(), enc.float())
+    logits = logits / math.sqrt(enc.shape[-1])
+    logits = logits.masked_fill(enc_paddings > 0.5, -1e30)
+    probs = torch.softmax(logits, dim=-1)
+    return torch.einsum('bs,bsd->bd', probs, enc.float()).to(enc.dtype)
+
+  def ComputePredictions(self, theta: NestedMap, enc: torch.Tensor,
+                         enc_paddings: torch.Tensor,
+                         targets: NestedMap) -> NestedMap:
+    p = self.p
+    b, l = targets.ids.shape
+    emb_all = self.emb.EmbLookup(theta.emb, targets.ids.long()).to(
+        self.fprop_dtype)
+    states = [c.InitState(b, enc.device, self.fprop_dtype)
+              for c in self.rnns]
+    ctx = torch.zeros(b, p.source_dim, device=enc.device,
+                      dtype=self.fprop_dtype)
+    outs = []
+    for t in range(l):
+      x = torch.cat([emb_all[:, t], ctx], dim=-1)
+      for i, cell in enumerate(self.rnns):
+        states[i] = cell.FProp(theta.rnns[i], states[i], NestedMap(act=x))
+        x = states[i].m
+      ctx = self._Attend(theta, x, enc, enc_paddings)
+      outs.append(torch.cat([x, ctx], dim=-1))
+    return NestedMap(atten_vecs=torch.stack(outs, dim=1))  # [B, L, H+D]
+
+  def ComputeLoss(self, theta: NestedMap, predictions: NestedMap,
+                  targets: NestedMap):
+    act = predictions.atten_vecs
+    xent = self.softmax.XentLoss(
+        theta.softmax, act, class_weights=targets.weights,
+        class_ids=targets.labels)
+    metrics = NestedMap(
+        loss=(xent.avg_xent, xent.total_weight),
+        log_pplx=(xent.avg_xent.detach(), xent.total_weight))
+    return metrics, NestedMap(per_example_xent=xent.per_example_xent)
+
+  def GreedyDecode(self, theta: NestedMap, enc: torch.Tensor,
+                   enc_paddings: torch.Tensor, max_len: int = 100,
+                   sos_id: int = 1, eos_id: int = 2) -> torch.Tensor:
+    p = self.p
+    b = enc.shape[0]
+    tok = torch.full((b,), sos_id, dtype=torch.long, device=enc.device)
+    states = [c.InitState(b, enc.device, self.fprop_dtype)
+              for c in self.rnns]
+    ctx = torch.zeros(b, p.source_dim, device=enc.device,
+                      dtype=self.fprop_dtype)
+    done = torch.zeros(b, dtype=torch.bool, device=enc.device)
+    out = []
+    for _ in range(max_len):
+      e = self.emb.EmbLookup(theta.emb, tok).to(self.fprop_dtype)
+      x = torch.cat([e, ctx], dim=-1)
+      for i, cell in enumerate(self.rnns):
+        states[i] = cell.FProp(theta.rnns[i], states[i], NestedMap(act=x))
+        x = states[i].m
+      ctx = self._Attend(theta, x, enc, enc_paddings)
+      logits = self.softmax.Logits(theta.softmax,
+                                   torch.cat([x, ctx], dim=-1))
+      tok = logits.argmax(-1)
+      tok = torch.where(done, torch.full_like(tok, eos_id), tok)
+      done = done | (tok == eos_id)
+      out.append(tok)
+      if bool(done.all()):
+        break
+    return torch.stack(out, dim=1)
+
+
+class AsrModel(BaseTask):
+  """Encoder/decoder ASR task (reference tasks/asr/model.py:30)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('encoder', ConformerEncoder.Params(), 'Encoder params.')
+    p.Define('decoder', AsrDecoder.Params(), 'Decoder params.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    self.CreateChild('encoder', self.p.encoder)
+    self.CreateChild('decoder', self.p.decoder)
+
+  def ComputePredictions(self, theta: NestedMap,
+                         input_batch: NestedMap) -> NestedMap:
+    enc, enc_pad = self.encoder.FProp(
+        theta.encoder, input_batch.src.src_inputs,
+        input_batch.src.paddings)
+    preds = self.decoder.ComputePredictions(theta.decoder, enc, enc_pad,
+                                            input_batch.tgt)
+    preds.encoder_outputs = enc
+    preds.encoder_paddings = enc_pad
+    return preds
+
+  def ComputeLoss(self, theta: NestedMap, predictions: NestedMap,
+                  input_batch: NestedMap):
+    metrics, per_example = self.decoder.ComputeLoss(
+        theta.decoder, predictions, input_batch.tgt)
+    b = input_batch.src.src_inputs.shape[0]
+    metrics.num_samples_in_batch = (
+        torch.tensor(float(b)), torch.ones(()))
+    return metrics, per_example
+
+  def Decode(self, input_batch: NestedMap) -> NestedMap:
+    with torch.no_grad():
+      enc, enc_pad = self.encoder.FProp(
+          self.theta.encoder, input_batch.src.src_inputs,
+          input_batch.src.paddings)
+      hyps = self.decoder.GreedyDecode(self.theta.decoder, enc, enc_pad)
+    return NestedMap(topk_decoded=hyps,
+                     transcripts=input_batch.tgt.ids)
+
+  def CreateDecoderMetrics(self) -> NestedMap:
+    return NestedMap(wer=metrics_lib.WerMetric(),
+                     num_samples_in_batch=metrics_lib.AverageMetric())
+
+  def PostProcessDecodeOut(self, decode_out: NestedMap,
+                           decode_metrics: NestedMap) -> None:
+    hyps = decode_out.topk_decoded
+    refs = decode_out.transcripts
+    for i in range(hyps.shape[0]):
+      hyp = ' '.join(str(int(x)) for x in hyps[i] if int(x) > 2)
+      ref = ' '.join(str(int(x)) for x in refs[i] if int(x) > 2)
+      decode_metrics.wer.Update(ref, hyp)
+    decode_metrics.num_samples_in_batch.Update(float(hyps.shape[0]))

@@ -32,7 +32,15 @@ std::vector<torch::Tensor> fa_bwd(torch::Tensor dout, torch::Tensor q,
                                   int64_t win_r, int64_t bias_clip,
                                   double scale);
 
+// conv1d.hip
+torch::Tensor dwconv1d_fwd(torch::Tensor x, torch::Tensor w,
+                           c10::optional<torch::Tensor> bias, int64_t pad);
+std::vector<torch::Tensor> dwconv1d_bwd(torch::Tensor dy, torch::Tensor x,
+                                        torch::Tensor w, int64_t pad);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("dwconv1d_fwd", &dwconv1d_fwd, "Depthwise time conv fwd");
+  m.def("dwconv1d_bwd", &dwconv1d_bwd, "Depthwise time conv bwd");
   m.def("layer_norm_fwd", &layer_norm_fwd, "Fused LayerNorm/RMSNorm fwd");
   m.def("layer_norm_bwd", &layer_norm_bwd, "Fused LayerNorm/RMSNorm bwd");
   m.def("mfma_probe", &mfma_probe, "MFMA 16x16x32 layout probe");

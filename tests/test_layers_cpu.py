@@ -1,0 +1,167 @@
+"""CPU tests for transformer/conformer/RNN layers (shapes, masks,
+determinism, gradient flow)."""
+
+import pytest
+import torch
+
+from lingvo_amd.core import py_utils
+from lingvo_amd.core.nested_map import NestedMap
+from lingvo_amd.layers import attention as attention_lib
+from lingvo_amd.layers import conformer as conformer_lib
+from lingvo_amd.layers import rnn_cell
+from lingvo_amd.layers import rnn_layers
+from lingvo_amd.layers import transformer as transformer_lib
+
+
+def test_mha_self_attention_shapes():
+  p = attention_lib.MultiHeadedAttention.Params().Set(
+      name='mha', input_dim=128, hidden_dim=128, num_heads=2, random_seed=1)
+  layer = p.Instantiate()
+  x = torch.randn(2, 10, 128)
+  pad = py_utils.PaddingsFromLengths(torch.tensor([10, 7]), 10)
+  out = layer.FProp(layer.theta, x, pad)
+  assert out.shape == (2, 10, 128)
+  # padded positions produce zeros
+  assert out[1, 8:].abs().sum() == 0
+
+
+def test_mha_causal_no_future_leak():
+  p = attention_lib.MultiHeadedAttention.Params().Set(
+      name='mha', input_dim=64, hidden_dim=64, num_heads=1, causal=True,
+      random_seed=1)
+  layer = p.Instantiate()
+  x = torch.randn(1, 8, 64)
+  out1 = layer.FProp(layer.theta, x)
+  x2 = x.clone()
+  x2[0, 5:] = 99.0  # mutate the future
+  out2 = layer.FProp(layer.theta, x2)
+  assert torch.allclose(out1[0, :5], out2[0, :5], atol=1e-4)
+
+
+def test_mha_gqa_and_extend_step_matches_fprop():
+  p = attention_lib.MultiHeadedAttention.Params().Set(
+      name='mha', input_dim=64, hidden_dim=64, num_heads=1, causal=True,
+      random_seed=3)
+  layer = p.Instantiate()
+  layer.eval()
+  x = torch.randn(2, 6, 64)
+  full = layer.FProp(layer.theta, x)
+  states = layer.InitStates(layer.theta, 2, 6, 'cpu', torch.float32)
+  outs = []
+  for t in range(6):
+    o, states = layer.ExtendStep(layer.theta, x[:, t:t + 1], states)
+    outs.append(o)
+  inc = torch.cat(outs, dim=1)
+  assert (full - inc).abs().max() < 1e-3
+
+
+def test_transformer_layer_and_stack():
+  p = transformer_lib.StackedTransformerLayers.Params().Set(
+      name='stack', model_dim=64, num_layers=2, num_heads=1,
+      random_seed=1)
+  stack = p.Instantiate()
+  x = torch.randn(2, 12, 64)
+  pad = torch.zeros(2, 12)
+  out = stack.FProp(stack.theta, x, pad)
+  assert out.shape == x.shape
+  out.sum().backward()
+  grads = [prm.grad for prm in stack.parameters() if prm.requires_grad]
+  assert all(g is not None for g in grads)
+
+
+def test_transformer_decoder_cross_attention():
+  p = transformer_lib.TransformerLayer.Params().Set(
+      name='dec', input_dim=64, num_heads=1, mask_self_atten=True,
+      has_aux_atten=True, random_seed=1)
+  layer = p.Instantiate()
+  x = torch.randn(2, 5, 64)
+  src = torch.randn(2, 9, 64)
+  src_pad = py_utils.PaddingsFromLengths(torch.tensor([9, 4]), 9)
+  out = layer.FProp(layer.theta, x, None, aux_vecs=src,
+                    aux_paddings=src_pad)
+  assert out.shape == x.shape
+
+
+def test_conformer_layer_shapes_and_padding():
+  p = conformer_lib.ConformerLayer.Params().Set(
+      name='conf', input_dim=64, atten_num_heads=1, kernel_size=8,
+      random_seed=1, use_relative_atten=True)
+  layer = p.Instantiate()
+  x = torch.randn(2, 20, 64)
+  pad = py_utils.PaddingsFromLengths(torch.tensor([20, 11]), 20)
+  out = layer.FProp(layer.theta, x, pad)
+  assert out.shape == x.shape
+  assert torch.isfinite(out).all()
+
+
+def test_depthwise_conv1d_cpu_ref():
+  from lingvo_amd.ops import conv1d as conv_ops
+  x = torch.randn(2, 10, 8)
+  w = torch.randn(3, 8)
+  y = conv_ops.depthwise_conv1d(x, w, causal=True)
+  # manual check at t=0: only tap j=K-1 (x[0]) contributes
+  want0 = (x[:, 0] * w[2]).float()
+  assert torch.allclose(y[:, 0].float(), want0, atol=1e-5)
+  # causality: output at t doesn't depend on x[t+1:]
+  x2 = x.clone()
+  x2[:, 5:] = 7.0
+  y2 = conv_ops.depthwise_conv1d(x2, w, causal=True)
+  assert torch.allclose(y[:, :5], y2[:, :5], atol=1e-5)
+
+
+def test_lstm_cell_and_frnn():
+  cell_p = rnn_cell.LSTMCellSimple.Params().Set(
+      name='lstm', num_input_nodes=8, num_output_nodes=16, random_seed=1)
+  frnn = rnn_layers.FRNN.Params().Set(name='frnn', cell=cell_p).Instantiate()
+  x = torch.randn(3, 7, 8)
+  pad = py_utils.PaddingsFromLengths(torch.tensor([7, 4, 1]), 7)
+  out, final = frnn.FProp(frnn.theta, x, pad)
+  assert out.shape == (3, 7, 16)
+  # state frozen after padding starts
+  assert torch.allclose(out[1, 3], out[1, 6], atol=1e-5)
+  out.sum().backward()
+
+
+def test_bidirectional_frnn():
+  mk = lambda name: rnn_cell.LSTMCellSimple.Params().Set(
+      name=name, num_input_nodes=8, num_output_nodes=8, random_seed=1)
+  p = rnn_layers.BidirectionalFRNN.Params().Set(
+      name='bi', fwd=mk('f'), bak=mk('b'))
+  layer = p.Instantiate()
+  x = torch.randn(2, 5, 8)
+  out = layer.FProp(layer.theta, x, torch.zeros(2, 5))
+  assert out.shape == (2, 5, 16)
+
+
+def test_recurrent_remat_matches_plain():
+  from lingvo_amd.core import recurrent
+  cell_p = rnn_cell.LSTMCellSimple.Params().Set(
+      name='c', num_input_nodes=4, num_output_nodes=4, random_seed=1)
+  cell = cell_p.Instantiate()
+  x = torch.randn(6, 2, 4, requires_grad=True)
+  state0 = cell.InitState(2, 'cpu', torch.float32)
+
+  def fn(th, st, inp):
+    return cell.FProp(th, st, inp), NestedMap()
+
+  acc1, _ = recurrent.Recurrent(cell.theta, state0, NestedMap(act=x), fn,
+                                remat=False)
+  g1 = torch.autograd.grad(acc1.m.sum(), x)[0]
+  acc2, _ = recurrent.Recurrent(cell.theta, state0, NestedMap(act=x), fn,
+                                remat=True)
+  g2 = torch.autograd.grad(acc2.m.sum(), x)[0]
+  assert torch.allclose(g1, g2, atol=1e-5)
+
+
+def test_group_norm_padding_invariance():
+  from lingvo_amd.layers import bn_layers
+  p = bn_layers.GroupNormLayer.Params().Set(
+      name='gn', dim=16, num_groups=4, random_seed=1)
+  gn = p.Instantiate()
+  x = torch.randn(2, 10, 16)
+  pad = py_utils.PaddingsFromLengths(torch.tensor([6, 10]), 10)
+  out1 = gn.FProp(gn.theta, x, pad)
+  x2 = x.clone()
+  x2[0, 6:] = 123.0  # changing padded region must not affect output
+  out2 = gn.FProp(gn.theta, x2, pad)
+  assert torch.allclose(out1[0, :6], out2[0, :6], atol=1e-5)
