@@ -38,7 +38,15 @@ torch::Tensor dwconv1d_fwd(torch::Tensor x, torch::Tensor w,
 std::vector<torch::Tensor> dwconv1d_bwd(torch::Tensor dy, torch::Tensor x,
                                         torch::Tensor w, int64_t pad);
 
+// softmax_xent.hip
+std::vector<torch::Tensor> xent_fwd(torch::Tensor logits,
+                                    torch::Tensor labels);
+torch::Tensor xent_bwd(torch::Tensor logits, torch::Tensor labels,
+                       torch::Tensor lse, torch::Tensor gout);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("xent_fwd", &xent_fwd, "Fused softmax-xent fwd");
+  m.def("xent_bwd", &xent_bwd, "Fused softmax-xent bwd");
   m.def("dwconv1d_fwd", &dwconv1d_fwd, "Depthwise time conv fwd");
   m.def("dwconv1d_bwd", &dwconv1d_bwd, "Depthwise time conv bwd");
   m.def("layer_norm_fwd", &layer_norm_fwd, "Fused LayerNorm/RMSNorm fwd");
