@@ -200,13 +200,35 @@ def MakeStepGenerator(device: Union[str, torch.device],
 
 def DeterministicDropout(x: torch.Tensor, keep_prob: float,
                          op_seed: Optional[int] = None) -> torch.Tensor:
-  """Dropout reproducible under StepSeedScope (reference py_utils.py:3978)."""
+  """Dropout reproducible under StepSeedScope (reference py_utils.py:3978).
+
+  GPU: one fused HIP kernel (mask recomputed from the seed in backward);
+  CPU: torch.Generator reference path.
+  """
   if keep_prob >= 1.0:
     return x
+  if x.is_cuda and x.numel() % 8 == 0:
+    from lingvo_amd.ops import dropout as dropout_ops
+    s1, s2 = GenerateStepSeedPair(op_seed)
+    return dropout_ops.dropout(x, keep_prob, (s1 << 32) ^ (s2 & 0xFFFFFFFF))
   g = MakeStepGenerator(x.device, op_seed)
   mask = (torch.rand(x.shape, generator=g, device=x.device,
                      dtype=torch.float32) < keep_prob)
   return x * mask.to(x.dtype) / keep_prob
+
+
+def DeterministicDropoutAdd(x: torch.Tensor, keep_prob: float,
+                            residual: torch.Tensor,
+                            op_seed: Optional[int] = None) -> torch.Tensor:
+  """residual + dropout(x): fused into one HIP kernel on GPU."""
+  if keep_prob >= 1.0:
+    return residual + x
+  if x.is_cuda and x.numel() % 8 == 0:
+    from lingvo_amd.ops import dropout as dropout_ops
+    s1, s2 = GenerateStepSeedPair(op_seed)
+    return dropout_ops.dropout(x, keep_prob, (s1 << 32) ^ (s2 & 0xFFFFFFFF),
+                               residual=residual)
+  return residual + DeterministicDropout(x, keep_prob, op_seed)
 
 
 # --------------------------------------------------------------------------

@@ -84,10 +84,11 @@ class LConvLayer(BaseLayer):
       x = self.norm.FProp(theta.norm, x, paddings)
     x = F.silu(x)
     x = torch.matmul(x, theta.pw2_w) + theta.pw2_b
-    if p.dropout_prob and not self.do_eval:
-      x = py_utils.DeterministicDropout(x, 1.0 - p.dropout_prob)
     if paddings is not None:
       x = py_utils.ApplyPadding(paddings, x)
+    if p.dropout_prob and not self.do_eval:
+      return py_utils.DeterministicDropoutAdd(x, 1.0 - p.dropout_prob,
+                                              inputs)
     return inputs + x
 
 

@@ -57,8 +57,8 @@ class TransformerAttentionLayer(BaseLayer):
       ctx = self.atten.FPropCross(theta.atten, x, source_vecs, source_vecs,
                                   source_paddings)
     if p.residual_dropout_prob and not self.do_eval:
-      ctx = py_utils.DeterministicDropout(ctx,
-                                          1.0 - p.residual_dropout_prob)
+      return py_utils.DeterministicDropoutAdd(
+          ctx, 1.0 - p.residual_dropout_prob, query_vec)
     return query_vec + ctx
 
   def InitStates(self, theta, batch, max_len, device, dtype=torch.bfloat16):
@@ -110,11 +110,14 @@ class TransformerFeedForwardLayer(BaseLayer):
     if p.relu_dropout_prob and not self.do_eval:
       h = py_utils.DeterministicDropout(h, 1.0 - p.relu_dropout_prob)
     out = torch.matmul(h, theta.w2) + theta.b2
-    if p.residual_dropout_prob and not self.do_eval:
-      out = py_utils.DeterministicDropout(out, 1.0 - p.residual_dropout_prob)
+    if p.residual_weight != 1.0:
+      out = out * p.residual_weight
     if paddings is not None:
       out = py_utils.ApplyPadding(paddings, out)
-    return inputs + p.residual_weight * out
+    if p.residual_dropout_prob and not self.do_eval:
+      return py_utils.DeterministicDropoutAdd(
+          out, 1.0 - p.residual_dropout_prob, inputs)
+    return inputs + out
 
 
 class TransformerLayer(BaseLayer):

@@ -151,6 +151,8 @@ class AsrDecoder(BaseLayer):
                          enc_paddings: torch.Tensor,
                          targets: NestedMap) -> NestedMap:
     p = self.p
+    if p.num_lstm_layers == 2:
+      return self._FastPredictions(theta, enc, enc_paddings, targets)
     b, l = targets.ids.shape
     emb_all = self.emb.EmbLookup(theta.emb, targets.ids.long()).to(
         self.fprop_dtype)
@@ -167,6 +169,58 @@ class AsrDecoder(BaseLayer):
       ctx = self._Attend(theta, x, enc, enc_paddings)
       outs.append(torch.cat([x, ctx], dim=-1))
     return NestedMap(atten_vecs=torch.stack(outs, dim=1))  # [B, L, H+D]
+
+  def _FastPredictions(self, theta: NestedMap, enc: torch.Tensor,
+                       enc_paddings: torch.Tensor,
+                       targets: NestedMap) -> NestedMap:
+    """Launch-lean teacher-forced loop (identical math to the cell path):
+    embedding gate contributions precomputed in one GEMM; one GEMM per
+    LSTM layer per step; bf16 bmm attention."""
+    p = self.p
+    dt = self.fprop_dtype
+    b, l = targets.ids.shape
+    h = p.rnn_cell_dim
+    ids = targets.ids.long()
+    emb_all = self.emb.EmbLookup(theta.emb, ids).to(dt)
+    wm0, b0 = theta.rnns[0].wm, theta.rnns[0].b
+    wm1, b1 = theta.rnns[1].wm, theta.rnns[1].b
+    cap = self.rnns[0].p.cell_value_cap
+    e_dim = p.emb_dim
+    emb_gates = torch.matmul(emb_all, wm0[:e_dim]) + b0  # [B, L, 4H]
+    w_cm0 = wm0[e_dim:]  # [(src + H), 4H]
+    wq = theta.atten_query_w
+    inv_sqrt_d = 1.0 / math.sqrt(p.source_dim)
+    neg_mask = (enc_paddings.float() * -1e30).unsqueeze(1)  # [B,1,S]
+
+    fgb = self.rnns[0].p.forget_gate_bias
+
+    def lstm_pointwise(gates, c_prev):
+      i_i, i_g, f_g, o_g = gates.split([h, h, h, h], dim=-1)
+      if fgb:
+        f_g = f_g + fgb
+      c = torch.sigmoid(f_g) * c_prev + torch.sigmoid(i_g) * torch.tanh(i_i)
+      if cap is not None:
+        c = torch.clamp(c, -cap, cap)
+      return c, torch.sigmoid(o_g) * torch.tanh(c)
+
+    zeros = lambda d: torch.zeros(b, d, device=enc.device, dtype=dt)
+    c0, m0, c1, m1 = zeros(h), zeros(h), zeros(h), zeros(h)
+    ctx = zeros(p.source_dim)
+    outs = []
+    for t in range(l):
+      gates0 = torch.addmm(emb_gates[:, t], torch.cat([ctx, m0], dim=-1),
+                           w_cm0)
+      c0, m0 = lstm_pointwise(gates0, c0)
+      gates1 = torch.addmm(b1.unsqueeze(0), torch.cat([m0, m1], dim=-1),
+                           wm1)
+      c1, m1 = lstm_pointwise(gates1, c1)
+      q = torch.matmul(m1, wq)  # [B, D]
+      logits = torch.bmm(enc, q.unsqueeze(-1)).transpose(1, 2).float()
+      logits = logits * inv_sqrt_d + neg_mask  # [B,1,S]
+      probs = torch.softmax(logits, dim=-1).to(dt)
+      ctx = torch.bmm(probs, enc).squeeze(1)  # [B, D]
+      outs.append(torch.cat([m1, ctx], dim=-1))
+    return NestedMap(atten_vecs=torch.stack(outs, dim=1))
 
   def ComputeLoss(self, theta: NestedMap, predictions: NestedMap,
                   targets: NestedMap):

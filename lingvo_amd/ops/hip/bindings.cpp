@@ -44,7 +44,14 @@ std::vector<torch::Tensor> xent_fwd(torch::Tensor logits,
 torch::Tensor xent_bwd(torch::Tensor logits, torch::Tensor labels,
                        torch::Tensor lse, torch::Tensor gout);
 
+// dropout.hip
+torch::Tensor dropout_fwd(torch::Tensor x, c10::optional<torch::Tensor> res,
+                          int64_t seed, double keep);
+torch::Tensor dropout_bwd(torch::Tensor dy, int64_t seed, double keep);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("dropout_fwd", &dropout_fwd, "Fused dropout fwd");
+  m.def("dropout_bwd", &dropout_bwd, "Fused dropout bwd");
   m.def("xent_fwd", &xent_fwd, "Fused softmax-xent fwd");
   m.def("xent_bwd", &xent_bwd, "Fused softmax-xent bwd");
   m.def("dwconv1d_fwd", &dwconv1d_fwd, "Depthwise time conv fwd");

@@ -104,19 +104,26 @@ __global__ void dwconv_bwd_dw(const unsigned short* __restrict__ dy,
   if (d >= D) return;
   const int stripe = blockIdx.y;
   const long rows = (long)B * T;
+  // Contiguous row chunks: consecutive iterations share K-1 of the K
+  // x-taps, so the taps stay L1-resident.
+  const long chunk = (rows + nstripes - 1) / nstripes;
+  const long r0 = stripe * chunk;
+  const long r1 = min(rows, r0 + chunk);
   float dw[MAXK];
   for (int j = 0; j < K; ++j) dw[j] = 0.f;
   float db = 0.f;
-  for (long r = stripe; r < rows; r += nstripes) {
+  for (long r = r0; r < r1; ++r) {
     const int t = (int)(r % T);
     const long b = r / T;
     float g = bf16_bits_to_float(dy[r * D + d]);
-    if (g == 0.f) continue;
     db += g;
+    const long xbase = (b * T) * (long)D + d;
     for (int j = 0; j < K; ++j) {
       int ts = t + j - pad;
-      if (ts < 0 || ts >= T) continue;
-      dw[j] += g * bf16_bits_to_float(x[(b * T + ts) * D + d]);
+      float xv = (ts < 0 || ts >= T)
+                     ? 0.f
+                     : bf16_bits_to_float(x[xbase + (long)ts * D]);
+      dw[j] += g * xv;
     }
   }
   for (int j = 0; j < K; ++j) {
@@ -162,7 +169,7 @@ std::vector<torch::Tensor> dwconv1d_bwd(torch::Tensor dy, torch::Tensor x,
                      (const unsigned short*)dy.data_ptr(),
                      (const unsigned short*)w.data_ptr(),
                      (unsigned short*)dx.data_ptr(), B, T, D, K, (int)pad);
-  int nstripes = (int)std::min<long>(64, std::max<long>(1, (long)B * T / 64));
+  int nstripes = (int)std::min<long>(512, std::max<long>(1, (long)B * T / 8));
   dim3 grid_w((D + 255) / 256, nstripes);
   hipLaunchKernelGGL(dwconv_bwd_dw, grid_w, dim3(256), 0, stream,
                      (const unsigned short*)dy.data_ptr(),

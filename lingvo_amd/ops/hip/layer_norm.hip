@@ -167,18 +167,28 @@ __global__ void ln_bwd_bf16(const unsigned short* __restrict__ dy,
 }
 
 // Reduce partials [2, nwaves, D] -> dscale [D], dbias [D] (fp32).
+// Split-K over the waves dim (grid.y); coalesced column reads; one fp32
+// atomicAdd per (thread, output). dscale/dbias must be zero-initialized.
 __global__ void ln_bwd_reduce(const float* __restrict__ partials,
                               float* __restrict__ dscale,
                               float* __restrict__ dbias, int nwaves, int D) {
   int col = blockIdx.x * blockDim.x + threadIdx.x;
   if (col >= D) return;
+  const int chunk = (nwaves + gridDim.y - 1) / gridDim.y;
+  const int w0 = blockIdx.y * chunk;
+  const int w1 = min(nwaves, w0 + chunk);
   float s = 0.f, b = 0.f;
-  for (int w = 0; w < nwaves; ++w) {
+  for (int w = w0; w < w1; ++w) {
     s += partials[(long)w * D + col];
     b += partials[(long)(nwaves + w) * D + col];
   }
-  dscale[col] = s;
-  dbias[col] = b;
+  if (gridDim.y == 1) {
+    dscale[col] = s;
+    dbias[col] = b;
+  } else {
+    atomicAdd(dscale + col, s);
+    atomicAdd(dbias + col, b);
+  }
 }
 
 // ---- generic-D fallback (scalar, one block per row) -----------------------
@@ -351,8 +361,8 @@ std::vector<torch::Tensor> layer_norm_bwd(torch::Tensor dy, torch::Tensor x,
   const float* rp = rstd.data_ptr<float>();
   unsigned short* dxp = (unsigned short*)dx.data_ptr();
 
-  torch::Tensor dscale = torch::empty({D}, opts);
-  torch::Tensor dbias = torch::empty({D}, opts);
+  torch::Tensor dscale = torch::zeros({D}, opts);
+  torch::Tensor dbias = torch::zeros({D}, opts);
 
   if (use_vec_path(D)) {
     int grid = memory_bound_grid(rows, kWavesPerBlock, 512);
@@ -382,8 +392,9 @@ std::vector<torch::Tensor> layer_norm_bwd(torch::Tensor dy, torch::Tensor x,
         TORCH_CHECK(false, "unhandled NVEC");
     }
 #undef LN_BWD_CASE
-    hipLaunchKernelGGL(ln_bwd_reduce, dim3(cdiv(D, 256)), dim3(256), 0,
-                       stream, partials.data_ptr<float>(),
+    int splitk = (int)std::min<long>(64, std::max<long>(1, nwaves / 8));
+    hipLaunchKernelGGL(ln_bwd_reduce, dim3(cdiv(D, 256), splitk), dim3(256),
+                       0, stream, partials.data_ptr<float>(),
                        dscale.data_ptr<float>(), dbias.data_ptr<float>(),
                        nwaves, (int)D);
   } else {
@@ -397,8 +408,9 @@ std::vector<torch::Tensor> layer_norm_bwd(torch::Tensor dy, torch::Tensor x,
       hipLaunchKernelGGL((ln_bwd_generic<false>), dim3(grid), dim3(kBlock), 0,
                          stream, dyp, xp, sp, mp, rp, dxp,
                          partials.data_ptr<float>(), (int)rows, (int)D);
-    hipLaunchKernelGGL(ln_bwd_reduce, dim3(cdiv(D, 256)), dim3(256), 0,
-                       stream, partials.data_ptr<float>(),
+    int splitk2 = (int)std::min<long>(64, std::max<long>(1, grid / 8));
+    hipLaunchKernelGGL(ln_bwd_reduce, dim3(cdiv(D, 256), splitk2), dim3(256),
+                       0, stream, partials.data_ptr<float>(),
                        dscale.data_ptr<float>(), dbias.data_ptr<float>(),
                        grid, (int)D);
   }

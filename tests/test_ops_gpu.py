@@ -88,3 +88,30 @@ def test_native_ext_is_loaded():
   import lingvo_amd.ops._lingvo_ops as ext
   assert hasattr(ext, 'layer_norm_fwd')
   assert '/root/' in ext.__file__ or 'lingvo_amd' in ext.__file__
+
+
+@gpu
+def test_fused_dropout_fwd_bwd():
+  from lingvo_amd.ops import dropout as dropout_ops
+  torch.manual_seed(0)
+  x = torch.randn(4, 64, 512, device='cuda',
+                  dtype=torch.bfloat16).requires_grad_(True)
+  res = torch.randn_like(x).requires_grad_(True)
+  keep = 0.9
+  y = dropout_ops.dropout(x, keep, seed=42, residual=res)
+  # kept elements are x/keep + res; dropped are res
+  ratio = (y - res).detach()
+  kept_frac = (ratio.abs() > 1e-6).float().mean().item()
+  assert abs(kept_frac - keep) < 0.02
+  # determinism
+  y2 = dropout_ops.dropout(x, keep, seed=42, residual=res)
+  assert torch.equal(y, y2)
+  y3 = dropout_ops.dropout(x, keep, seed=43, residual=res)
+  assert not torch.equal(y, y3)
+  # backward: dx = mask/keep * g, dres = g
+  g = torch.randn_like(y)
+  y.backward(g)
+  assert torch.equal(res.grad, g)
+  mask = (ratio.float().abs() > 1e-6)
+  want_dx = (g.float() * mask / keep).to(torch.bfloat16)
+  assert (x.grad.float() - want_dx.float()).abs().max() < 0.05
