@@ -1,0 +1,182 @@
+"""Batched beam search (reference lingvo/core/beam_search_helper.py:200
+BeamSearchHelper.BeamSearchDecode and the C++ step op
+core/ops/beam_search_step_op_kernels.cc).
+
+The per-step pruning (top-k over [b*k, V] scores, EOS handling with
+valid_eos_max_logit_delta, hypothesis bookkeeping) runs as batched torch
+ops on the GPU; terminated hypotheses are tracked per beam. The callback
+contract mirrors the reference (:203-260):
+
+  init_fn(batch, num_hyps) -> state
+  step_fn(state, prev_ids [B*K]) -> (log_probs [B*K, V], state)
+  (state tensors are reordered by gather when hyps are reshuffled)
+"""
+
+from __future__ import annotations
+
+from typing import Callable, Optional, Tuple
+
+import torch
+
+from lingvo_amd.core.hyperparams import Params
+from lingvo_amd.core.nested_map import NestedMap
+
+
+class BeamSearchHelper:
+
+  @classmethod
+  def Params(cls) -> Params:
+    p = Params()
+    p.Define('num_hyps_per_beam', 8, 'Beam width K.')
+    p.Define('beam_size', 3.0,
+             'Stop when best active score is worse than best terminated '
+             'minus this margin (reference x_ops.cc:116 beam_size).')
+    p.Define('length_normalization', 0.0, 'Length-norm alpha.')
+    p.Define('valid_eos_max_logit_delta', 5.0,
+             'EOS may terminate only if its score is within this delta '
+             'of the best non-EOS extension.')
+    p.Define('target_sos_id', 1, 'SOS token.')
+    p.Define('target_eos_id', 2, 'EOS token.')
+    p.Define('max_steps', 128, 'Maximum decode length.')
+    return p
+
+  def __init__(self, params: Params):
+    self.p = params
+
+  def _Norm(self, length: torch.Tensor) -> torch.Tensor:
+    alpha = self.p.length_normalization
+    if alpha == 0.0:
+      return torch.ones_like(length, dtype=torch.float32)
+    return ((5.0 + length.float()) / 6.0) ** alpha
+
+  def BeamSearchDecode(self, batch: int, init_fn, step_fn,
+                       reorder_fn) -> NestedMap:
+    """Runs the search. reorder_fn(state, gather_idx [B*K]) -> state."""
+    p = self.p
+    k = p.num_hyps_per_beam
+    state = init_fn(batch, k)
+    device = state.device if hasattr(state, 'device') else \
+        next(iter(s for s in state.Flatten()
+                  if isinstance(s, torch.Tensor))).device
+
+    bk = batch * k
+    prev_ids = torch.full((bk,), p.target_sos_id, dtype=torch.long,
+                          device=device)
+    # Only hyp 0 of each beam starts alive (others -inf) so the first
+    # step doesn't produce k duplicates.
+    cum_scores = torch.full((batch, k), -1e30, device=device)
+    cum_scores[:, 0] = 0.0
+    histories = torch.zeros(batch, k, p.max_steps, dtype=torch.long,
+                            device=device)
+    done_scores = torch.full((batch, k), -1e30, device=device)
+    done_norm_scores = torch.full((batch, k), -1e30, device=device)
+    done_ids = torch.zeros(batch, k, p.max_steps, dtype=torch.long,
+                           device=device)
+    done_lens = torch.zeros(batch, k, dtype=torch.long, device=device)
+
+    for t in range(p.max_steps):
+      log_probs, state = step_fn(state, prev_ids)  # [B*K, V]
+      v = log_probs.shape[-1]
+      total = cum_scores.reshape(bk, 1) + log_probs.float()  # [B*K, V]
+
+      # EOS handling: candidate terminations this step.
+      eos_scores = total[:, p.target_eos_id].reshape(batch, k)
+      no_eos = total.clone()
+      no_eos[:, p.target_eos_id] = -1e30
+      best_no_eos = no_eos.max(dim=-1).values.reshape(batch, k)
+
+      # Top-k over the flattened (hyp, vocab) extension space.
+      flat = no_eos.reshape(batch, k * v)
+      top_scores, top_idx = flat.topk(k, dim=-1)  # [B, K]
+      prev_hyp = top_idx // v  # [B, K] index into previous hyps
+      new_tok = top_idx % v
+
+      # Terminations: EOS within delta of the best extension of that hyp.
+      eos_valid = eos_scores >= (best_no_eos -
+                                 p.valid_eos_max_logit_delta)
+      eos_norm = eos_scores * 0 + eos_scores  # placeholder for clarity
+      lens = torch.full((batch, k), t + 1, device=device)
+      eos_norm = eos_scores / self._Norm(lens)
+      improve = eos_valid & (eos_norm > done_norm_scores.min(
+          dim=-1, keepdim=True).values)
+      if bool(improve.any()):
+        for b in range(batch):
+          for h in range(k):
+            if bool(improve[b, h]):
+              slot = int(done_norm_scores[b].argmin())
+              if float(eos_norm[b, h]) > float(done_norm_scores[b, slot]):
+                done_norm_scores[b, slot] = eos_norm[b, h]
+                done_scores[b, slot] = eos_scores[b, h]
+                done_ids[b, slot, :t] = histories[b, h, :t]
+                done_ids[b, slot, t] = p.target_eos_id
+                done_lens[b, slot] = t + 1
+
+      # Reshuffle live hyps.
+      gather = (torch.arange(batch, device=device).unsqueeze(1) * k +
+                prev_hyp).reshape(bk)
+      histories = histories.reshape(bk, -1)[gather].reshape(
+          batch, k, -1)
+      histories[:, :, t] = new_tok
+      cum_scores = top_scores
+      prev_ids = new_tok.reshape(bk)
+      state = reorder_fn(state, gather)
+
+      # Early stop: best possible live score worse than worst kept done.
+      best_live = (cum_scores.max(dim=-1).values /
+                   self._Norm(torch.full((batch,), t + 1, device=device)))
+      worst_done = done_norm_scores.min(dim=-1).values
+      have_all = (done_norm_scores > -1e29).all(dim=-1)
+      if bool((have_all &
+               (best_live + p.beam_size < worst_done)).all()):
+        break
+
+    # Fill any empty done slots with live hyps.
+    live_norm = cum_scores / self._Norm(
+        torch.full((batch, k), p.max_steps, device=device))
+    for b in range(batch):
+      for slot in range(k):
+        if float(done_norm_scores[b, slot]) < -1e29:
+          h = int(live_norm[b].argmax())
+          done_norm_scores[b, slot] = live_norm[b, h]
+          done_scores[b, slot] = cum_scores[b, h]
+          done_ids[b, slot] = histories[b, h]
+          done_lens[b, slot] = p.max_steps
+          live_norm[b, h] = -1e30
+
+    order = done_norm_scores.argsort(dim=-1, descending=True)
+    gather3 = order.unsqueeze(-1).expand_as(done_ids)
+    return NestedMap(
+        topk_ids=done_ids.gather(1, gather3),
+        topk_lens=done_lens.gather(1, order),
+        topk_scores=done_norm_scores.gather(1, order))
+
+
+class GreedySearchHelper:
+  """Greedy variant (reference beam_search_helper.py:752)."""
+
+  def __init__(self, max_steps: int = 128, sos_id: int = 1,
+               eos_id: int = 2):
+    self.max_steps = max_steps
+    self.sos_id = sos_id
+    self.eos_id = eos_id
+
+  def GreedySearchDecode(self, batch: int, init_fn, step_fn) -> NestedMap:
+    state = init_fn(batch, 1)
+    device = next(iter(s for s in state.Flatten()
+                       if isinstance(s, torch.Tensor))).device
+    prev = torch.full((batch,), self.sos_id, dtype=torch.long,
+                      device=device)
+    done = torch.zeros(batch, dtype=torch.bool, device=device)
+    ids = []
+    for _ in range(self.max_steps):
+      log_probs, state = step_fn(state, prev)
+      prev = log_probs.argmax(-1)
+      prev = torch.where(done, torch.full_like(prev, self.eos_id), prev)
+      done = done | (prev == self.eos_id)
+      ids.append(prev)
+      if bool(done.all()):
+        break
+    out = torch.stack(ids, dim=1)
+    lens = (out != self.eos_id).long().sum(-1) + 1
+    return NestedMap(topk_ids=out.unsqueeze(1),
+                     topk_lens=lens.clamp_max(out.shape[1]).unsqueeze(1))

@@ -1,0 +1,191 @@
+"""Language-model tasks (reference lingvo/tasks/lm/model.py:25
+LanguageModel, layers.py:495 RnnLm / TransformerLm).
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+
+from lingvo_amd.core import py_utils
+from lingvo_amd.core.base_input_generator import BaseSequenceInputGenerator
+from lingvo_amd.core.base_layer import BaseLayer
+from lingvo_amd.core.base_model import BaseTask
+from lingvo_amd.core.nested_map import NestedMap
+from lingvo_amd.layers import layers as lingvo_layers
+from lingvo_amd.layers import rnn_cell
+from lingvo_amd.layers import rnn_layers
+from lingvo_amd.layers import transformer as transformer_lib
+
+
+class SyntheticLmInput(BaseSequenceInputGenerator):
+  """Synthetic token batches (reference
+  tasks/lm/params/synthetic_packed_input.py:29 SyntheticTrain)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.batch_size = 8
+    p.Define('seq_len', 1024, 'Sequence length.')
+    p.Define('vocab_size', 32000, 'Vocab size.')
+    return p
+
+  def _InputBatch(self) -> NestedMap:
+    p = self.p
+    g = torch.Generator().manual_seed(7000 + self._batch_count)
+    ids = torch.randint(1, p.vocab_size, (p.batch_size, p.seq_len),
+                        generator=g)
+    labels = ids.roll(-1, dims=1)
+    weights = torch.ones(p.batch_size, p.seq_len)
+    weights[:, -1] = 0.0
+    return NestedMap(ids=ids, labels=labels,
+                     paddings=torch.zeros(p.batch_size, p.seq_len),
+                     weights=weights)
+
+
+class TransformerLm(BaseLayer):
+  """Causal transformer LM (reference tasks/lm/layers.py TransformerLm /
+  GPipeTransformerLm at one_billion_wds.py:181)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('vocab_size', 32000, 'Vocab.')
+    p.Define('model_dim', 1024, 'Model dim.')
+    p.Define('num_layers', 12, 'Layers.')
+    p.Define('num_heads', 16, 'Heads.')
+    p.Define('hidden_dim', 0, 'FFN hidden (0 = 4x).')
+    p.Define('dropout_prob', 0.1, 'Dropout.')
+    p.Define('shared_emb', True, 'Tie softmax and embedding weights.')
+    p.Define('remat', False, 'Checkpoint layers.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    if p.shared_emb:
+      self.CreateChild('softmax',
+                       lingvo_layers.SharedSoftmaxLayer.Params().Set(
+                           input_dim=p.model_dim, num_classes=p.vocab_size))
+    else:
+      self.CreateChild('emb', lingvo_layers.EmbeddingLayer.Params().Set(
+          vocab_size=p.vocab_size, embedding_dim=p.model_dim,
+          scale_sqrt_depth=True))
+      self.CreateChild('softmax',
+                       lingvo_layers.SimpleFullSoftmax.Params().Set(
+                           input_dim=p.model_dim, num_classes=p.vocab_size))
+    self.CreateChild('pos_emb',
+                     lingvo_layers.PositionalEmbeddingLayer.Params().Set(
+                         embedding_dim=p.model_dim))
+    stack_p = transformer_lib.StackedTransformerLayers.Params().Set(
+        model_dim=p.model_dim, num_layers=p.num_layers,
+        num_heads=p.num_heads, hidden_dim=p.hidden_dim,
+        mask_self_atten=True, remat=p.remat)
+    stack_p.transformer_tpl.tr_atten_tpl.residual_dropout_prob = \
+        p.dropout_prob
+    stack_p.transformer_tpl.tr_fflayer_tpl.residual_dropout_prob = \
+        p.dropout_prob
+    stack_p.transformer_tpl.tr_fflayer_tpl.relu_dropout_prob = \
+        p.dropout_prob
+    self.CreateChild('stack', stack_p)
+
+  def _Emb(self, theta, ids):
+    if self.p.shared_emb:
+      return self.softmax.EmbLookup(theta.softmax, ids)
+    return self.emb.EmbLookup(theta.emb, ids)
+
+  def FProp(self, theta: NestedMap, ids: torch.Tensor,
+            paddings: torch.Tensor) -> torch.Tensor:
+    x = self._Emb(theta, ids.long()).to(self.fprop_dtype)
+    pos = self.pos_emb.FProp(theta.pos_emb, ids.shape[1], device=ids.device)
+    x = x + pos.unsqueeze(0).to(x.dtype)
+    if self.p.dropout_prob and not self.do_eval:
+      x = py_utils.DeterministicDropout(x, 1.0 - self.p.dropout_prob)
+    return self.stack.FProp(theta.stack, x, paddings)
+
+  def XentLoss(self, theta, act, labels, weights):
+    return self.softmax.XentLoss(theta.softmax, act, class_weights=weights,
+                                 class_ids=labels)
+
+
+class RnnLm(BaseLayer):
+  """LSTM LM (reference tasks/lm/layers.py:495 RnnLm; the
+  WordLevelOneBwdsSimpleSampledSoftmax baseline model)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('vocab_size', 32000, 'Vocab.')
+    p.Define('emb_dim', 1024, 'Embedding dim.')
+    p.Define('rnn_dims', [2048, 2048], 'Per-layer LSTM dims.')
+    p.Define('rnn_proj', 1024, 'LSTM projection dim (0 = none).')
+    p.Define('dropout_prob', 0.1, 'Dropout.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    self.CreateChild('emb', lingvo_layers.EmbeddingLayer.Params().Set(
+        vocab_size=p.vocab_size, embedding_dim=p.emb_dim))
+    cells = []
+    in_dim = p.emb_dim
+    for i, d in enumerate(p.rnn_dims):
+      out_dim = p.rnn_proj or d
+      cells.append(rnn_cell.LSTMCellSimple.Params().Set(
+          name=f'lstm_{i}', num_input_nodes=in_dim, num_output_nodes=out_dim,
+          num_hidden_nodes=d if p.rnn_proj else 0))
+      in_dim = out_dim
+    self.CreateChild('rnns', rnn_layers.StackedFRNNLayerByLayer.Params().Set(
+        cell_tpl=cells, skip_start=1))
+    self.CreateChild('softmax', lingvo_layers.SimpleFullSoftmax.Params().Set(
+        input_dim=in_dim, num_classes=p.vocab_size))
+
+  def FProp(self, theta, ids, paddings):
+    x = self.emb.EmbLookup(theta.emb, ids.long()).to(self.fprop_dtype)
+    if self.p.dropout_prob and not self.do_eval:
+      x = py_utils.DeterministicDropout(x, 1.0 - self.p.dropout_prob)
+    return self.rnns.FProp(theta.rnns, x, paddings)
+
+  def XentLoss(self, theta, act, labels, weights):
+    return self.softmax.XentLoss(theta.softmax, act, class_weights=weights,
+                                 class_ids=labels)
+
+
+class LanguageModel(BaseTask):
+  """LM task (reference tasks/lm/model.py:25)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('lm', TransformerLm.Params(), 'LM layer params.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    self.CreateChild('lm', self.p.lm)
+
+  def ComputePredictions(self, theta, input_batch):
+    act = self.lm.FProp(theta.lm, input_batch.ids, input_batch.paddings)
+    return NestedMap(activations=act)
+
+  def ComputeLoss(self, theta, predictions, input_batch):
+    xent = self.lm.XentLoss(theta.lm, predictions.activations,
+                            input_batch.labels, input_batch.weights)
+    num_toks = xent.total_weight
+    b = input_batch.ids.shape[0]
+    metrics = NestedMap(
+        loss=(xent.avg_xent, num_toks),
+        log_pplx=(xent.avg_xent.detach(), num_toks),
+        num_samples_in_batch=(torch.tensor(float(b)), torch.ones(())),
+        tokens_per_batch=(num_toks.detach(), torch.ones(())))
+    return metrics, NestedMap(per_example_xent=xent.per_example_xent)
+
+  def Decode(self, input_batch):
+    with torch.no_grad():
+      act = self.lm.FProp(self.theta.lm, input_batch.ids,
+                          input_batch.paddings)
+      xent = self.lm.XentLoss(self.theta.lm, act, input_batch.labels,
+                              input_batch.weights)
+    return NestedMap(log_pplx=xent.avg_xent.reshape(1))
