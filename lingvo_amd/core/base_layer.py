@@ -101,11 +101,14 @@ class BaseLayer(nn.Module):
     return self._params.name or type(self).__name__
 
   def _InitGenerator(self, var_name: str) -> torch.Generator:
+    import zlib
     seed = self._params.random_seed
     if seed is None:
       seed = 1234
-    mixed = (hash((self.layer_name, var_name)) ^ (seed * 2654435761)
-             ) & 0x7FFFFFFFFFFFFFFF
+    # Stable cross-process hash (Python's hash() is salted per process,
+    # which would give DP ranks different initial weights).
+    name_hash = zlib.crc32(f'{self.layer_name}/{var_name}'.encode())
+    mixed = (name_hash ^ (seed * 2654435761)) & 0x7FFFFFFFFFFFFFFF
     g = torch.Generator()
     g.manual_seed(mixed)
     return g
