@@ -51,8 +51,8 @@ class MultiHeadedAttention(BaseLayer):
     nkv = p.num_kv_heads or n
     h = p.dim_per_head or (p.hidden_dim // n)
     self._n, self._nkv, self._h = n, nkv, h
-    assert h in (64, 128), (
-        f'flash kernel supports dim_per_head 64/128, got {h}')
+    # The gfx950 flash kernel supports dim_per_head 64/128 (checked at
+    # the kernel boundary); other sizes run only on the CPU reference.
     # Fused QKV for self-attention: [D, (N + 2*NKV) * H].
     self.CreateVariable('qkv_w', py_utils.WeightParams(
         [p.input_dim, (n + 2 * nkv) * h], p.params_init, p.dtype))

@@ -60,6 +60,9 @@ class TransformerLm(BaseLayer):
     p.Define('dropout_prob', 0.1, 'Dropout.')
     p.Define('shared_emb', True, 'Tie softmax and embedding weights.')
     p.Define('remat', False, 'Checkpoint layers.')
+    p.Define('moe_every_n', 0, 'Every n-th layer uses MoE FFN.')
+    p.Define('num_experts', 0, 'MoE experts (when moe_every_n > 0).')
+    p.Define('expert_capacity_factor', 2.0, 'MoE capacity factor.')
     return p
 
   def __init__(self, params):
@@ -89,6 +92,11 @@ class TransformerLm(BaseLayer):
         p.dropout_prob
     stack_p.transformer_tpl.tr_fflayer_tpl.relu_dropout_prob = \
         p.dropout_prob
+    if p.moe_every_n:
+      stack_p.moe_every_n = p.moe_every_n
+      stack_p.moe_tpl.num_experts = p.num_experts
+      stack_p.moe_tpl.expert_capacity_factor = p.expert_capacity_factor
+      stack_p.moe_tpl.residual_dropout_prob = p.dropout_prob
     self.CreateChild('stack', stack_p)
 
   def _Emb(self, theta, ids):
@@ -175,8 +183,16 @@ class LanguageModel(BaseTask):
                             input_batch.labels, input_batch.weights)
     num_toks = xent.total_weight
     b = input_batch.ids.shape[0]
+    loss = xent.avg_xent
+    from lingvo_amd.parallel.moe import MoEFeedForwardLayer
+    aux = [m.AuxLoss() for m in self.modules()
+           if isinstance(m, MoEFeedForwardLayer) and
+           m.AuxLoss() is not None]
+    if aux:
+      aux_total = torch.stack([a.float() for a in aux]).sum()
+      loss = loss + aux_total
     metrics = NestedMap(
-        loss=(xent.avg_xent, num_toks),
+        loss=(loss, num_toks),
         log_pplx=(xent.avg_xent.detach(), num_toks),
         num_samples_in_batch=(torch.tensor(float(b)), torch.ones(())),
         tokens_per_batch=(num_toks.detach(), torch.ones(())))

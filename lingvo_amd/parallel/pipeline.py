@@ -1,0 +1,178 @@
+"""GPipe-style pipeline parallelism over RCCL/xGMI P2P.
+
+Reference: lingvo/core/gpipe.py:324 PipeliningLayer (microbatch split at
+:539-559, send/recv links via recurrent.py:1142-1155 + sendrecv.py:37).
+MI355X-native design: an explicit fill-drain microbatch scheduler, one
+process per stage, activations moved with torch.distributed send/recv
+(RCCL P2P rides xGMI point-to-point links); backward returns boundary
+gradients the same way. Gradients accumulate across microbatches in
+param.grad, so the per-stage optimizer applies once per step.
+
+Per-microbatch RNG: dropout seeds derive from (global_seed, step,
+op_counter); the op counter keeps advancing across microbatches, so each
+microbatch draws distinct deterministic masks (reference gpipe.py:46-63
+global-step override discipline).
+"""
+
+from __future__ import annotations
+
+from typing import Callable, List, Optional
+
+import torch
+import torch.distributed as dist
+
+from lingvo_amd.core.nested_map import NestedMap
+
+_DTYPE_CODES = {
+    torch.float32: 0, torch.bfloat16: 1, torch.float16: 2, torch.int64: 3,
+    torch.int32: 4, torch.bool: 5,
+}
+_CODE_DTYPES = {v: k for k, v in _DTYPE_CODES.items()}
+_MAX_DIMS = 8
+
+
+def _SendTensor(t: torch.Tensor, dst: int, group, device) -> None:
+  header = torch.zeros(_MAX_DIMS + 2, dtype=torch.int64)
+  header[0] = t.dim()
+  for i, d in enumerate(t.shape):
+    header[1 + i] = d
+  header[_MAX_DIMS + 1] = _DTYPE_CODES[t.dtype]
+  dist.send(header, dst, group=group)
+  dist.send(t.contiguous(), dst, group=group)
+
+
+def _RecvTensor(src: int, group, device) -> torch.Tensor:
+  header = torch.zeros(_MAX_DIMS + 2, dtype=torch.int64)
+  dist.recv(header, src, group=group)
+  ndim = int(header[0])
+  shape = [int(header[1 + i]) for i in range(ndim)]
+  dtype = _CODE_DTYPES[int(header[_MAX_DIMS + 1])]
+  t = torch.empty(shape, dtype=dtype, device=device)
+  dist.recv(t, src, group=group)
+  return t
+
+
+def SendNestedMap(nmap: NestedMap, dst: int, group=None,
+                  device='cpu') -> None:
+  flat = nmap.FlattenItems()
+  count = torch.tensor([len(flat)], dtype=torch.int64)
+  dist.send(count, dst, group=group)
+  for key, val in flat:
+    kb = key.encode()[:64].ljust(64)
+    dist.send(torch.frombuffer(bytearray(kb), dtype=torch.uint8).clone(),
+              dst, group=group)
+    _SendTensor(val, dst, group, device)
+
+
+def RecvNestedMap(src: int, group=None, device='cpu') -> NestedMap:
+  count = torch.zeros(1, dtype=torch.int64)
+  dist.recv(count, src, group=group)
+  out = NestedMap()
+  for _ in range(int(count[0])):
+    kb = torch.zeros(64, dtype=torch.uint8)
+    dist.recv(kb, src, group=group)
+    key = bytes(kb.tolist()).decode().strip()
+    out.Set(key, _RecvTensor(src, group, device))
+  return out
+
+
+class GPipeRunner:
+  """Fill-drain microbatch schedule for one pipeline stage.
+
+  stage_fprop(microbatch_nmap) -> nmap: this stage's forward (already
+  bound to theta). Tensors that require grad at the INPUT boundary
+  receive gradients during drain.
+  """
+
+  def __init__(self, stage_idx: int, num_stages: int,
+               num_micro_batches: int, group=None,
+               device: str = 'cpu'):
+    self.stage_idx = stage_idx
+    self.num_stages = num_stages
+    self.num_micro = num_micro_batches
+    self.group = group
+    self.device = device
+    self._prev = stage_idx - 1 if stage_idx > 0 else None
+    self._next = stage_idx + 1 if stage_idx < num_stages - 1 else None
+
+  @property
+  def is_first(self) -> bool:
+    return self.stage_idx == 0
+
+  @property
+  def is_last(self) -> bool:
+    return self.stage_idx == self.num_stages - 1
+
+  def RunStep(self, stage_fprop: Callable[[NestedMap], NestedMap],
+              input_fn: Optional[Callable[[int], NestedMap]] = None,
+              loss_fn: Optional[Callable[[NestedMap, int], torch.Tensor]]
+              = None) -> Optional[torch.Tensor]:
+    """One full train step = M forward microbatches then M backwards.
+
+    Returns the mean loss on the last stage, None elsewhere.
+    """
+    saved_in: List[NestedMap] = []
+    saved_out: List[NestedMap] = []
+    losses: List[torch.Tensor] = []
+
+    # ---- fill: forward all microbatches ----
+    for m in range(self.num_micro):
+      if self.is_first:
+        assert input_fn is not None
+        inp = input_fn(m)
+      else:
+        inp = RecvNestedMap(self._prev, self.group, self.device)
+        inp = inp.Transform(
+            lambda t: t.requires_grad_(True)
+            if t.is_floating_point() else t)
+      out = stage_fprop(inp)
+      if self.is_last:
+        assert loss_fn is not None
+        losses.append(loss_fn(out, m))
+      else:
+        SendNestedMap(out.Transform(lambda t: t.detach()), self._next,
+                      self.group, self.device)
+        saved_out.append(out)
+      saved_in.append(inp)
+
+    # ---- drain: backward in reverse order ----
+    for m in reversed(range(self.num_micro)):
+      if self.is_last:
+        (losses[m] / self.num_micro).backward()
+      else:
+        grads = RecvNestedMap(self._next, self.group, self.device)
+        out = saved_out[m]
+        gtensors, otensors = [], []
+        for key, val in out.FlattenItems():
+          g = grads.Get(key)
+          if g is not None and isinstance(val, torch.Tensor) and \
+              val.requires_grad:
+            otensors.append(val)
+            gtensors.append(g)
+        torch.autograd.backward(otensors, gtensors)
+      if not self.is_first:
+        gmap = NestedMap()
+        for key, val in saved_in[m].FlattenItems():
+          if isinstance(val, torch.Tensor) and val.requires_grad and \
+              val.grad is not None:
+            gmap.Set(key, val.grad)
+        SendNestedMap(gmap, self._prev, self.group, self.device)
+
+    if self.is_last and losses:
+      return torch.stack([l.detach() for l in losses]).mean()
+    return None
+
+
+def PartitionSequentialLayers(layer_params: List, num_stages: int
+                              ) -> List[List]:
+  """Balanced contiguous partition (reference gpipe.py:179)."""
+  n = len(layer_params)
+  base = n // num_stages
+  rem = n % num_stages
+  out = []
+  idx = 0
+  for s in range(num_stages):
+    take = base + (1 if s < rem else 0)
+    out.append(layer_params[idx:idx + take])
+    idx += take
+  return out
