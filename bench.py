@@ -112,11 +112,13 @@ def main():
         'dtype': 'bf16' if has_gpu else 'fp32',
         'data': 'synthetic',
         'config': {
-            'model': 'Conformer-L (17 blocks, d=512, h=8, kernel 32) + '
-                     'LSTM attention decoder',
+            'model': ('Conformer-L (17 blocks, d=512, h=8, kernel 32) + '
+                      'LSTM attention decoder'
+                      if 'Conformer' in args.model else args.model),
             'global_batch': global_batch,
-            'seq_len': model_p.input.frame_len,
-            'target_len': model_p.input.target_len,
+            'seq_len': (model_p.input.frame_len
+                        if 'frame_len' in model_p.input
+                        else model_p.input.Get('seq_len')),
             'parallelism': f'dp{world}',
             'final_loss': round(loss, 4),
         },

@@ -63,6 +63,10 @@ class BaseTask(BaseLayer):
               'Learner params (or list of).')
     tp.Define('max_steps', None, 'Stop training after this step.')
     tp.Define('ema_decay', 0.0, 'If >0, maintain EMA of weights.')
+    tp.Define('bf16_weights', False,
+              'Keep model weights in bf16 with fp32 masters inside the '
+              'optimizer (MasterAdamW): theta casts become no-ops and '
+              'DP all-reduce runs natively on bf16 grads.')
     tp.Define('start_up_delay_steps', 200, 'Unused on MI355X; kept for '
               'config parity.')
     tp.Define('vn_std', 0.0, 'Variational noise std (0 disables).')
@@ -87,6 +91,7 @@ class BaseTask(BaseLayer):
     self.input_generator = None
     if p.input is not None:
       self.input_generator = p.input.Instantiate()
+    self._bf16_converted = False
 
   @property
   def global_step(self) -> int:
@@ -115,9 +120,22 @@ class BaseTask(BaseLayer):
     return self.FPropTower(theta, input_batch)
 
   # ---- training ---------------------------------------------------------
+  def MaybeConvertBf16Weights(self) -> None:
+    """Converts fp32 params to bf16 in place (p.train.bf16_weights);
+    must run before the optimizer is created. Idempotent."""
+    if self._bf16_converted or not self.p.train.bf16_weights:
+      return
+    if self.fprop_dtype != torch.bfloat16:
+      return  # tests override fprop_dtype to fp32; keep fp32 weights
+    for prm in self.parameters():
+      if prm.dtype == torch.float32:
+        prm.data = prm.data.to(torch.bfloat16)
+    self._bf16_converted = True
+
   def TrainStep(self, input_batch: NestedMap,
                 grad_sync_finalize=None) -> NestedMap:
     """One full train step: forward, backward, update, bookkeeping."""
+    self.MaybeConvertBf16Weights()
     step = self.global_step
     with py_utils.StepSeedScope(self.p.random_seed or 1234, step):
       metrics, _ = self.FProp(self.theta, input_batch)

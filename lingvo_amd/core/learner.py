@@ -124,13 +124,14 @@ class Learner(BaseLayer):
     if p.clip_gradient_norm_to_value:
       scale = p.clip_gradient_norm_to_value / grad_norm.clamp_min(
           p.clip_gradient_norm_to_value)
-      for g in grads:
-        g.mul_(scale.to(g.dtype))
+      torch._foreach_mul_(grads, scale)
     if p.clip_gradient_single_norm_to_value:
-      for g in grads:
-        n = g.float().norm(2)
-        g.mul_((p.clip_gradient_single_norm_to_value /
-                n.clamp_min(p.clip_gradient_single_norm_to_value)).to(g.dtype))
+      norms = torch._foreach_norm(grads, 2)
+      scales = [(p.clip_gradient_single_norm_to_value /
+                 n.float().clamp_min(
+                     p.clip_gradient_single_norm_to_value)).to(g.dtype)
+                for n, g in zip(norms, grads)]
+      torch._foreach_mul_(grads, scales)
 
     lr = self.LearningRate(global_step)
     for group in opt.param_groups:

@@ -100,20 +100,89 @@ class Adam(Base):
   def CreateTorchOptimizer(self, params, lr):
     p = self.p
     params = list(params)
-    use_fused = bool(params) and all(
-        t.is_cuda for t in params if isinstance(t, torch.Tensor))
-    if use_fused:
-      try:
-        from lingvo_amd.ops.fused_adam import FusedAdamW
-        return FusedAdamW(params, lr=lr, betas=(p.beta1, p.beta2),
-                          eps=p.epsilon, weight_decay=p.weight_decay)
-      except Exception:
-        pass
+    if params and any(t.dtype == torch.bfloat16 for t in params):
+      # bf16-weight training: fp32 masters live in optimizer state,
+      # updated with ~10 multi-tensor foreach launches.
+      return MasterAdamW(params, lr=lr, betas=(p.beta1, p.beta2),
+                         eps=p.epsilon, weight_decay=p.weight_decay)
     if p.weight_decay:
       return torch.optim.AdamW(params, lr=lr, betas=(p.beta1, p.beta2),
                                eps=p.epsilon, weight_decay=p.weight_decay)
     return torch.optim.Adam(params, lr=lr, betas=(p.beta1, p.beta2),
                             eps=p.epsilon)
+
+
+class MasterAdamW(torch.optim.Optimizer):
+  """AdamW for bf16 model weights with fp32 master copies.
+
+  The whole update is a fixed number of _foreach launches independent of
+  parameter count: grads cast fp32 -> moments update -> master update ->
+  bf16 copy-back. Masters persist in state_dict for exact resume.
+  """
+
+  def __init__(self, params, lr=1e-3, betas=(0.9, 0.999), eps=1e-8,
+               weight_decay=0.0):
+    super().__init__(params, dict(lr=lr, betas=betas, eps=eps,
+                                  weight_decay=weight_decay))
+
+  def load_state_dict(self, state_dict):
+    # torch's default load casts fp32 state to the (bf16) param dtype,
+    # which would destroy the masters; restore exact tensors instead.
+    params = [p for g in self.param_groups for p in g['params']]
+    saved_ids = [i for g in state_dict['param_groups']
+                 for i in g['params']]
+    for pid, prm in zip(saved_ids, params):
+      if pid in state_dict['state']:
+        self.state[prm] = {
+            k: (v.clone().to(prm.device) if torch.is_tensor(v) else v)
+            for k, v in state_dict['state'][pid].items()
+        }
+    for g, sg in zip(self.param_groups, state_dict['param_groups']):
+      for k, v in sg.items():
+        if k != 'params':
+          g[k] = v
+
+  @torch.no_grad()
+  def step(self, closure=None):
+    for group in self.param_groups:
+      lr = group['lr']
+      beta1, beta2 = group['betas']
+      eps = group['eps']
+      wd = group['weight_decay']
+      params = [prm for prm in group['params'] if prm.grad is not None]
+      if not params:
+        continue
+      grads, masters, ms, vs = [], [], [], []
+      for prm in params:
+        st = self.state[prm]
+        if not st:
+          st['step'] = 0
+          st['master'] = prm.detach().float().clone()
+          st['m'] = torch.zeros_like(st['master'])
+          st['v'] = torch.zeros_like(st['master'])
+          st['g32'] = torch.zeros_like(st['master'])
+        st['step'] += 1
+        grads.append(st['g32'])
+        masters.append(st['master'])
+        ms.append(st['m'])
+        vs.append(st['v'])
+      # One batched cast of all bf16 grads to the fp32 scratch buffers.
+      torch._foreach_copy_(grads, [prm.grad for prm in params])
+      t = self.state[params[0]]['step']
+      bc1 = 1.0 - beta1 ** t
+      bc2 = 1.0 - beta2 ** t
+      torch._foreach_mul_(ms, beta1)
+      torch._foreach_add_(ms, grads, alpha=1.0 - beta1)
+      torch._foreach_mul_(vs, beta2)
+      torch._foreach_addcmul_(vs, grads, grads, value=1.0 - beta2)
+      denom = torch._foreach_sqrt(vs)
+      torch._foreach_div_(denom, bc2 ** 0.5)
+      torch._foreach_add_(denom, eps)
+      if wd:
+        torch._foreach_mul_(masters, 1.0 - lr * wd)
+      torch._foreach_addcdiv_(masters, ms, denom, value=-(lr / bc1))
+      torch._foreach_copy_(params, masters)  # bf16 cast back
+    return None
 
 
 class Adafactor(Base):

@@ -308,12 +308,13 @@ def HasNanOrInf(nmap_or_tensor) -> bool:
 
 
 def GlobalGradNorm(grads: List[torch.Tensor]) -> torch.Tensor:
-  """L2 norm over a list of grads (reference learner.py global clip)."""
-  device = grads[0].device if grads else 'cpu'
+  """L2 norm over a list of grads (reference learner.py:60-75 +
+  SumSquared py_utils.py:4242). Multi-tensor via torch._foreach_norm
+  (SURVEY K15)."""
   if not grads:
-    return torch.zeros((), device=device)
-  norms = torch.stack([g.detach().float().norm(2) for g in grads])
-  return norms.norm(2)
+    return torch.zeros(())
+  norms = torch._foreach_norm([g.detach() for g in grads], 2)
+  return torch.stack([n.float() for n in norms]).norm(2)
 
 
 def WeightedAvg(values: torch.Tensor, weights: torch.Tensor

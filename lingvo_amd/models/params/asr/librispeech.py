@@ -42,6 +42,7 @@ class Librispeech960WpmConformerL(SingleTaskModelParams):
   def Task(self):
     p = asr_model.AsrModel.Params().Set(name='librispeech_conformer_l')
     p.fprop_dtype = torch.bfloat16
+    p.train.bf16_weights = True
     p.encoder.Set(input_dim=80, model_dim=512, num_layers=17, num_heads=8,
                   kernel_size=32, dropout_prob=0.1)
     p.decoder.Set(vocab_size=self.VOCAB, emb_dim=128, rnn_cell_dim=640,

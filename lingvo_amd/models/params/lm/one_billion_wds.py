@@ -44,6 +44,7 @@ class OneBWdsTransformerLm(SingleTaskModelParams):
   def Task(self):
     p = lm_model.LanguageModel.Params().Set(name='1bwds_transformer_lm')
     p.fprop_dtype = torch.bfloat16
+    p.train.bf16_weights = True
     p.lm = lm_model.TransformerLm.Params().Set(
         vocab_size=self.VOCAB, model_dim=self.DIM, num_layers=self.LAYERS,
         num_heads=self.HEADS, hidden_dim=4 * self.DIM, dropout_prob=0.1)
@@ -74,6 +75,7 @@ class WordLevelOneBwdsRnnLm(SingleTaskModelParams):
   def Task(self):
     p = lm_model.LanguageModel.Params().Set(name='1bwds_rnn_lm')
     p.fprop_dtype = torch.bfloat16
+    p.train.bf16_weights = True
     p.lm = lm_model.RnnLm.Params().Set(
         vocab_size=32000, emb_dim=1024, rnn_dims=[2048, 2048],
         rnn_proj=1024, dropout_prob=0.1)

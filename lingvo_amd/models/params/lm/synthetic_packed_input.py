@@ -42,6 +42,7 @@ class DenseLm1B(SingleTaskModelParams):
   def Task(self):
     p = lm_model.LanguageModel.Params().Set(name='dense_lm')
     p.fprop_dtype = torch.bfloat16
+    p.train.bf16_weights = True
     p.lm = lm_model.TransformerLm.Params().Set(
         vocab_size=self.VOCAB, model_dim=self.DIM, num_layers=self.LAYERS,
         num_heads=self.DIM // 128, hidden_dim=4 * self.DIM,

@@ -38,6 +38,7 @@ class WmtEnDeTransformerBase(SingleTaskModelParams):
   def Task(self):
     p = mt_model.TransformerModel.Params().Set(name='wmt14_en_de')
     p.fprop_dtype = torch.bfloat16
+    p.train.bf16_weights = True
     p.encoder.Set(vocab_size=self.VOCAB, model_dim=self.DIM,
                   num_layers=self.LAYERS, num_heads=self.HEADS,
                   hidden_dim=self.FF, dropout_prob=0.1)
