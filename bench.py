@@ -26,8 +26,10 @@ def main():
   ap.add_argument('--gpus', type=int, default=1)
   ap.add_argument('--steps', type=int, default=20)
   ap.add_argument('--warmup', type=int, default=5)
-  ap.add_argument('--batch', type=int, default=16,
+  ap.add_argument('--batch', type=int, default=64,
                   help='Per-GPU batch size.')
+  ap.add_argument('--no-graph', action='store_true',
+                  help='Disable hipGraph step capture (eager steps).')
   ap.add_argument('--model', default='asr.librispeech.'
                   'Librispeech960WpmConformerL')
   args = ap.parse_args()
@@ -64,10 +66,23 @@ def main():
     batches.append(b.Transform(
         lambda t: t.to(device) if isinstance(t, torch.Tensor) else t))
 
+  graphed = None
+  if has_gpu and not args.no_graph:
+    try:
+      from lingvo_amd.runtime.graph_step import GraphedTrainStep
+      graphed = GraphedTrainStep(task, batches[0], grad_sync=sync)
+      if rank == 0:
+        print('# using hipGraph-captured train step', flush=True)
+    except Exception as e:  # fall back to eager steps
+      if rank == 0:
+        print(f'# hipGraph capture failed ({e}); eager steps', flush=True)
+      graphed = None
+
   def step(i):
-    metrics = task.TrainStep(batches[i % len(batches)],
-                             grad_sync_finalize=finalize)
-    return metrics
+    if graphed is not None:
+      return graphed.Step(batches[i % len(batches)])
+    return task.TrainStep(batches[i % len(batches)],
+                          grad_sync_finalize=finalize)
 
   for i in range(args.warmup):
     step(i)

@@ -170,6 +170,9 @@ def StepSeedScope(global_seed: int, step: int):
   stack = _GetRngStack()
   stack.append(NestedMap(global_seed=int(global_seed), step=int(step),
                          op_counter=[0]))
+  if torch.cuda.is_available():
+    from lingvo_amd.ops import dropout as dropout_ops
+    dropout_ops.SetStepSeed(int(step), int(global_seed))
   try:
     yield
   finally:
@@ -209,8 +212,9 @@ def DeterministicDropout(x: torch.Tensor, keep_prob: float,
     return x
   if x.is_cuda and x.numel() % 8 == 0:
     from lingvo_amd.ops import dropout as dropout_ops
-    s1, s2 = GenerateStepSeedPair(op_seed)
-    return dropout_ops.dropout(x, keep_prob, (s1 << 32) ^ (s2 & 0xFFFFFFFF))
+    s1, _ = GenerateStepSeedPair(op_seed)
+    # step-dependence comes from the device step-seed buffer (graph-safe)
+    return dropout_ops.dropout(x, keep_prob, s1)
   g = MakeStepGenerator(x.device, op_seed)
   mask = (torch.rand(x.shape, generator=g, device=x.device,
                      dtype=torch.float32) < keep_prob)
@@ -225,9 +229,8 @@ def DeterministicDropoutAdd(x: torch.Tensor, keep_prob: float,
     return residual + x
   if x.is_cuda and x.numel() % 8 == 0:
     from lingvo_amd.ops import dropout as dropout_ops
-    s1, s2 = GenerateStepSeedPair(op_seed)
-    return dropout_ops.dropout(x, keep_prob, (s1 << 32) ^ (s2 & 0xFFFFFFFF),
-                               residual=residual)
+    s1, _ = GenerateStepSeedPair(op_seed)
+    return dropout_ops.dropout(x, keep_prob, s1, residual=residual)
   return residual + DeterministicDropout(x, keep_prob, op_seed)
 
 

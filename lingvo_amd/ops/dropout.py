@@ -1,12 +1,41 @@
-"""Fused deterministic dropout wrapper (seed-only state)."""
+"""Fused deterministic dropout wrapper (seed-only state).
+
+The mask is a function of (op_seed, step_seed, element index). op_seed is
+baked into the kernel launch; step_seed lives in a small device buffer so
+the SAME captured hipGraph draws fresh masks every replay — the buffer is
+rewritten once per step by SetStepSeed (called from StepSeedScope).
+"""
 
 from __future__ import annotations
 
-from typing import Optional
+from typing import Dict, Optional
 
 import torch
 
 from lingvo_amd.ops import _loader
+
+_STEP_SEED_BUFS: Dict[int, torch.Tensor] = {}
+
+
+def _StepSeedBuf(device: torch.device) -> torch.Tensor:
+  idx = device.index or 0
+  if idx not in _STEP_SEED_BUFS:
+    _STEP_SEED_BUFS[idx] = torch.zeros(1, dtype=torch.int64,
+                                       device=device)
+  return _STEP_SEED_BUFS[idx]
+
+
+def SetStepSeed(step: int, global_seed: int = 0) -> None:
+  """Updates the device step-seed buffers (cheap; once per train step)."""
+  if not torch.cuda.is_available():
+    return
+  val = ((step + 1) * 0x9E3779B97F4A7C15 ^ (global_seed * 2654435761)) \
+      & 0x7FFFFFFFFFFFFFFF
+  for buf in _STEP_SEED_BUFS.values():
+    buf.fill_(val)
+  if not _STEP_SEED_BUFS:
+    dev = torch.device('cuda', torch.cuda.current_device())
+    _StepSeedBuf(dev).fill_(val)
 
 
 class _DropoutFn(torch.autograd.Function):
@@ -14,17 +43,19 @@ class _DropoutFn(torch.autograd.Function):
   @staticmethod
   def forward(ctx, x, residual, seed, keep):
     ext = _loader.get_ext(required=True)
-    y = ext.dropout_fwd(x, residual, seed, keep)
+    buf = _StepSeedBuf(x.device)
+    y = ext.dropout_fwd(x, residual, seed, buf, keep)
     ctx.seed = seed
     ctx.keep = keep
     ctx.has_res = residual is not None
+    ctx.dev = x.device
     return y
 
   @staticmethod
   def backward(ctx, dy):
     ext = _loader.get_ext(required=True)
     dy = dy.contiguous()
-    dx = ext.dropout_bwd(dy, ctx.seed, ctx.keep)
+    dx = ext.dropout_bwd(dy, ctx.seed, _StepSeedBuf(ctx.dev), ctx.keep)
     dres = dy if ctx.has_res else None
     return dx, dres, None, None
 

@@ -28,8 +28,10 @@ __global__ void dropout_fwd_kernel(const unsigned short* __restrict__ x,
                                    const unsigned short* __restrict__ res,
                                    unsigned short* __restrict__ y,
                                    long nvec, unsigned long long seed,
+                                   const long* __restrict__ step_seed,
                                    float keep, float inv_keep) {
   const unsigned int thresh = (unsigned int)(keep * 4294967296.0);
+  if (step_seed) seed ^= (unsigned long long)(*step_seed);
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
        i += (long)gridDim.x * blockDim.x) {
     ushortx8 v = *reinterpret_cast<const ushortx8*>(x + i * 8);
@@ -54,8 +56,10 @@ __global__ void dropout_fwd_kernel(const unsigned short* __restrict__ x,
 __global__ void dropout_bwd_kernel(const unsigned short* __restrict__ dy,
                                    unsigned short* __restrict__ dx,
                                    long nvec, unsigned long long seed,
+                                   const long* __restrict__ step_seed,
                                    float keep, float inv_keep) {
   const unsigned int thresh = (unsigned int)(keep * 4294967296.0);
+  if (step_seed) seed ^= (unsigned long long)(*step_seed);
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
        i += (long)gridDim.x * blockDim.x) {
     ushortx8 v = *reinterpret_cast<const ushortx8*>(dy + i * 8);
@@ -73,40 +77,47 @@ __global__ void dropout_bwd_kernel(const unsigned short* __restrict__ dy,
 }  // namespace
 
 torch::Tensor dropout_fwd(torch::Tensor x, c10::optional<torch::Tensor> res,
-                          int64_t seed, double keep) {
+                          int64_t seed, c10::optional<torch::Tensor> step_seed,
+                          double keep) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() &&
               x.scalar_type() == torch::kBFloat16 && x.numel() % 8 == 0);
   auto y = torch::empty_like(x);
   long nvec = x.numel() / 8;
   auto stream = at::cuda::getCurrentCUDAStream();
+  const long* ssp = step_seed.has_value() ?
+      step_seed->data_ptr<long>() : nullptr;
   if (res.has_value()) {
     hipLaunchKernelGGL((dropout_fwd_kernel<true>),
                        dim3(memory_bound_grid(nvec, 256)), dim3(256), 0,
                        stream, (const unsigned short*)x.data_ptr(),
                        (const unsigned short*)res->data_ptr(),
                        (unsigned short*)y.data_ptr(), nvec,
-                       (unsigned long long)seed, (float)keep,
+                       (unsigned long long)seed, ssp, (float)keep,
                        (float)(1.0 / keep));
   } else {
     hipLaunchKernelGGL((dropout_fwd_kernel<false>),
                        dim3(memory_bound_grid(nvec, 256)), dim3(256), 0,
                        stream, (const unsigned short*)x.data_ptr(), nullptr,
                        (unsigned short*)y.data_ptr(), nvec,
-                       (unsigned long long)seed, (float)keep,
+                       (unsigned long long)seed, ssp, (float)keep,
                        (float)(1.0 / keep));
   }
   return y;
 }
 
-torch::Tensor dropout_bwd(torch::Tensor dy, int64_t seed, double keep) {
+torch::Tensor dropout_bwd(torch::Tensor dy, int64_t seed,
+                          c10::optional<torch::Tensor> step_seed,
+                          double keep) {
   auto dx = torch::empty_like(dy);
   long nvec = dy.numel() / 8;
   auto stream = at::cuda::getCurrentCUDAStream();
+  const long* ssp = step_seed.has_value() ?
+      step_seed->data_ptr<long>() : nullptr;
   hipLaunchKernelGGL(dropout_bwd_kernel,
                      dim3(memory_bound_grid(nvec, 256)), dim3(256), 0,
                      stream, (const unsigned short*)dy.data_ptr(),
                      (unsigned short*)dx.data_ptr(), nvec,
-                     (unsigned long long)seed, (float)keep,
+                     (unsigned long long)seed, ssp, (float)keep,
                      (float)(1.0 / keep));
   return dx;
 }

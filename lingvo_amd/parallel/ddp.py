@@ -115,11 +115,17 @@ class GradSync:
     inv = 1.0 / self._world
     for bkt in self._buckets:
       if bkt.work is None:
-        # Some params had no grad this step: zero their slots and fire.
+        # Hooks didn't complete this bucket — either some params had no
+        # grad, or backward ran inside a captured hipGraph (replay does
+        # not fire python hooks). Pull grads from the params directly.
         for p in bkt.params:
           if id(p) not in bkt.ready:
             off = bkt.offsets[id(p)]
-            bkt.buffer[off:off + p.numel()].zero_()
+            if p.grad is not None:
+              bkt.buffer[off:off + p.numel()].copy_(
+                  p.grad.detach().reshape(-1).to(self._dtype))
+            else:
+              bkt.buffer[off:off + p.numel()].zero_()
         bkt.work = dist.all_reduce(bkt.buffer, op=dist.ReduceOp.SUM,
                                    group=self._pg, async_op=True)
     for bkt in self._buckets:

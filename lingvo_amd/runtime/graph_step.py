@@ -1,0 +1,113 @@
+"""hipGraph-captured train step.
+
+Captures forward + backward of a task into one hipGraph (the MI355X
+replacement for the reference's XLA-compiled TpuTrainStep device loop,
+program.py:545-609): a replay is a single launch-bound-free submission of
+the ~5k kernels of a Conformer/Transformer step. The optimizer, LR
+schedule, gradient clipping and DP all-reduce stay eager (a handful of
+foreach/RCCL calls), so learning-rate schedules and collectives behave
+exactly as in the eager path.
+
+Requirements: static batch shapes (synthetic/bucketed-padded inputs),
+bf16_weights mode (theta is the parameters — no cast ops to re-record),
+and dropout seeded via the device step-seed buffer (ops/dropout.py) so
+every replay draws fresh masks.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+from lingvo_amd.core import py_utils
+from lingvo_amd.core.nested_map import NestedMap
+
+
+class GraphedTrainStep:
+
+  def __init__(self, task, example_batch: NestedMap, grad_sync=None,
+               warmup_iters: int = 3):
+    assert torch.cuda.is_available(), 'GraphedTrainStep needs a GPU'
+    task.MaybeConvertBf16Weights()
+    self.task = task
+    self.grad_sync = grad_sync
+    if grad_sync is not None:
+      # Replay doesn't fire python hooks; use Finalize's pull-from-grad
+      # path instead, and keep hook all-reduces out of the capture.
+      grad_sync.Close()
+    self.learner = task.learners[0]
+    self.opt = self.learner.EnsureOptimizer(task)
+    self.loss_name = self.learner.p.loss_name
+    self._seed = task.p.random_seed or 1234
+
+    self.static_batch = example_batch.Transform(
+        lambda t: t.clone() if isinstance(t, torch.Tensor) else t)
+    self.params = [prm for _, prm in self.learner._trainable]
+    for prm in self.params:
+      if prm.grad is None:
+        prm.grad = torch.zeros_like(prm)
+    self.grads = [prm.grad for prm in self.params]
+
+    def fwd_bwd():
+      torch._foreach_zero_(self.grads)
+      with py_utils.StepSeedScope(self._seed, 0):
+        metrics, _ = task.FProp(task.theta, self.static_batch)
+      loss = metrics[self.loss_name][0]
+      loss.backward()
+      return metrics
+
+    # Warmup on a side stream (materializes workspaces/allocations).
+    side = torch.cuda.Stream()
+    side.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(side):
+      for _ in range(warmup_iters):
+        self.metrics = fwd_bwd()
+    torch.cuda.current_stream().wait_stream(side)
+
+    self.graph = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(self.graph):
+      self.metrics = fwd_bwd()
+
+  def Step(self, batch: NestedMap) -> NestedMap:
+    task = self.task
+    lrn = self.learner
+    p = lrn.p
+    # Copy this step's batch into the captured static buffers.
+    dsts = self.static_batch.Flatten()
+    srcs = batch.Flatten()
+    for dst, src in zip(dsts, srcs):
+      if isinstance(dst, torch.Tensor):
+        dst.copy_(src, non_blocking=True)
+    from lingvo_amd.ops import dropout as dropout_ops
+    dropout_ops.SetStepSeed(task.global_step, self._seed)
+
+    self.graph.replay()
+
+    if self.grad_sync is not None:
+      self.grad_sync.Finalize()  # pulls from (static) param.grad in place
+
+    grad_norm = py_utils.GlobalGradNorm(self.grads)
+    # Sync-free skip-step: non-finite or exploding grads scale to 0
+    # (moments still advance with zero grads; the reference skips the
+    # whole apply — difference documented in GraphedTrainStep docstring).
+    scale = torch.ones((), device=grad_norm.device)
+    if p.clip_gradient_norm_to_value:
+      scale = p.clip_gradient_norm_to_value / grad_norm.clamp_min(
+          p.clip_gradient_norm_to_value)
+    if p.grad_norm_to_clip_to_zero:
+      scale = torch.where(grad_norm < p.grad_norm_to_clip_to_zero, scale,
+                          torch.zeros_like(scale))
+    if p.skip_step_on_non_finite:
+      scale = torch.where(torch.isfinite(grad_norm), scale,
+                          torch.zeros_like(scale))
+    torch._foreach_mul_(self.grads, scale)
+
+    lr = lrn.LearningRate(task.global_step)
+    for group in self.opt.param_groups:
+      group['lr'] = lr
+    self.opt.step()
+    task.global_step_var += 1
+    if task.ema is not None:
+      task.ema.Update(task.named_parameters())
+    return self.metrics
