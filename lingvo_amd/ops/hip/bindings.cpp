@@ -46,8 +46,12 @@ torch::Tensor xent_bwd(torch::Tensor logits, torch::Tensor labels,
 
 // dropout.hip
 torch::Tensor dropout_fwd(torch::Tensor x, c10::optional<torch::Tensor> res,
-                          int64_t seed, double keep);
-torch::Tensor dropout_bwd(torch::Tensor dy, int64_t seed, double keep);
+                          int64_t seed,
+                          c10::optional<torch::Tensor> step_seed,
+                          double keep);
+torch::Tensor dropout_bwd(torch::Tensor dy, int64_t seed,
+                          c10::optional<torch::Tensor> step_seed,
+                          double keep);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dropout_fwd", &dropout_fwd, "Fused dropout fwd");
