@@ -351,6 +351,19 @@ def ToScalar(x) -> float:
   return float(x)
 
 
+def MatmulBias(x: torch.Tensor, w: torch.Tensor,
+               b: Optional[torch.Tensor] = None) -> torch.Tensor:
+  """x @ w (+ b) with the bias fused into the hipBLASLt GEMM epilogue
+  (torch.addmm) instead of a separate elementwise add kernel."""
+  shape = x.shape
+  x2 = x.reshape(-1, shape[-1])
+  if b is None:
+    out = torch.matmul(x2, w)
+  else:
+    out = torch.addmm(b, x2, w)
+  return out.reshape(*shape[:-1], w.shape[-1])
+
+
 class Timer:
   """Accumulating wall-clock timer (reference py_utils.py:6890)."""
 

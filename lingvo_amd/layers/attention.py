@@ -73,9 +73,8 @@ class MultiHeadedAttention(BaseLayer):
     p = self.p
     n, nkv, h = self._n, self._nkv, self._h
     b, t = x.shape[0], x.shape[1]
-    qkv = torch.matmul(x, theta.qkv_w)
-    if p.use_bias:
-      qkv = qkv + theta.qkv_b
+    qkv = py_utils.MatmulBias(x, theta.qkv_w,
+                              theta.qkv_b if p.use_bias else None)
     q, k, v = qkv.split([n * h, nkv * h, nkv * h], dim=-1)
     return (q.reshape(b, t, n, h), k.reshape(b, t, nkv, h),
             v.reshape(b, t, nkv, h))
@@ -99,9 +98,8 @@ class MultiHeadedAttention(BaseLayer):
       out = py_utils.DeterministicDropout(out, 1.0 - p.atten_dropout_prob)
     b, t = out.shape[0], out.shape[1]
     ctx = out.reshape(b, t, self._n * self._h)
-    post = torch.matmul(ctx, theta.post_w)
-    if p.use_bias:
-      post = post + theta.post_b
+    post = py_utils.MatmulBias(ctx, theta.post_w,
+                               theta.post_b if p.use_bias else None)
     if paddings is not None:
       post = py_utils.ApplyPadding(paddings, post)
     return post

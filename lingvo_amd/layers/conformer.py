@@ -69,7 +69,7 @@ class LConvLayer(BaseLayer):
             paddings: Optional[torch.Tensor] = None) -> torch.Tensor:
     p = self.p
     x = self.ln.FProp(theta.ln, inputs)
-    x = torch.matmul(x, theta.pw1_w) + theta.pw1_b
+    x = py_utils.MatmulBias(x, theta.pw1_w, theta.pw1_b)
     a, b = x.chunk(2, dim=-1)
     x = a * torch.sigmoid(b)  # GLU
     if paddings is not None:
@@ -83,7 +83,7 @@ class LConvLayer(BaseLayer):
     else:
       x = self.norm.FProp(theta.norm, x, paddings)
     x = F.silu(x)
-    x = torch.matmul(x, theta.pw2_w) + theta.pw2_b
+    x = py_utils.MatmulBias(x, theta.pw2_w, theta.pw2_b)
     if paddings is not None:
       x = py_utils.ApplyPadding(paddings, x)
     if p.dropout_prob and not self.do_eval:
@@ -194,19 +194,35 @@ class ConvSubsampling(BaseLayer):
     self.CreateVariable('proj_b', py_utils.WeightParams(
         [p.output_dim], py_utils.WeightInit.Constant(0.0), p.dtype))
 
+  @staticmethod
+  def _ConvGemm(x: torch.Tensor, w_oihw: torch.Tensor, bias: torch.Tensor,
+                stride: int = 2, pad: int = 1) -> torch.Tensor:
+    """3x3 strided conv as im2col + hipBLASLt GEMM. MIOpen's algorithm
+    search can fall back to a naive NCHW kernel for these shapes on
+    gfx950 (observed ~1000x regression under rocprof); the unfold+GEMM
+    path always lands on Tensile."""
+    o = w_oihw.shape[0]
+    bsz = x.shape[0]
+    hout = (x.shape[2] + 2 * pad - 3) // stride + 1
+    wout = (x.shape[3] + 2 * pad - 3) // stride + 1
+    cols = F.unfold(x, kernel_size=3, stride=stride, padding=pad)
+    out = torch.baddbmm(
+        bias.reshape(1, o, 1),
+        w_oihw.reshape(1, o, -1).expand(bsz, -1, -1), cols)
+    return out.reshape(bsz, o, hout, wout)
+
   def FProp(self, theta: NestedMap, inputs: torch.Tensor,
             paddings: torch.Tensor):
     """inputs [B, T, F] -> (out [B, ceil(T/4), D], out_paddings)."""
     x = inputs.unsqueeze(1)  # [B,1,T,F]
-    w1 = theta.conv1_w.permute(3, 2, 0, 1)
-    x = F.conv2d(x, w1, theta.conv1_b, stride=2, padding=1)
-    x = F.relu(x)
-    w2 = theta.conv2_w.permute(3, 2, 0, 1)
-    x = F.conv2d(x, w2, theta.conv2_b, stride=2, padding=1)
-    x = F.relu(x)  # [B, ch, T/4, F/4]
+    w1 = theta.conv1_w.permute(3, 2, 0, 1).contiguous()
+    x = F.relu(self._ConvGemm(x, w1, theta.conv1_b))
+    w2 = theta.conv2_w.permute(3, 2, 0, 1).contiguous()
+    x = F.relu(self._ConvGemm(x, w2, theta.conv2_b))
     b, ch, t4, f4 = x.shape
     x = x.permute(0, 2, 3, 1).reshape(b, t4, f4 * ch)
-    out = torch.matmul(x, theta.proj_w) + theta.proj_b
+    out = torch.addmm(theta.proj_b, x.reshape(-1, f4 * ch),
+                      theta.proj_w).reshape(b, t4, -1)
     out_paddings = paddings[:, ::2][:, ::2]
     out_paddings = out_paddings[:, :t4]
     out = py_utils.ApplyPadding(out_paddings, out)

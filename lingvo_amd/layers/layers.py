@@ -49,9 +49,8 @@ class ProjectionLayer(BaseLayer):
   def FProp(self, theta: NestedMap, inputs: torch.Tensor,
             paddings: Optional[torch.Tensor] = None) -> torch.Tensor:
     p = self.p
-    out = torch.matmul(inputs, theta.w)
-    if p.has_bias:
-      out = out + theta.b
+    out = py_utils.MatmulBias(inputs, theta.w,
+                              theta.b if p.has_bias else None)
     out = activations.GetFn(p.activation)(out)
     if paddings is not None:
       out = py_utils.ApplyPadding(paddings, out)
@@ -331,7 +330,7 @@ class SimpleFullSoftmax(BaseLayer):
         [p.num_classes], py_utils.WeightInit.Constant(0.0), p.dtype))
 
   def Logits(self, theta: NestedMap, inputs: torch.Tensor) -> torch.Tensor:
-    return torch.matmul(inputs, theta.linear_w) + theta.bias
+    return py_utils.MatmulBias(inputs, theta.linear_w, theta.bias)
 
   def XentLossFromLogits(self, logits: torch.Tensor,
                          class_ids: Optional[torch.Tensor] = None,

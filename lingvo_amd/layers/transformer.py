@@ -105,11 +105,11 @@ class TransformerFeedForwardLayer(BaseLayer):
             paddings: Optional[torch.Tensor] = None) -> torch.Tensor:
     p = self.p
     x = self.layer_norm.FProp(theta.layer_norm, inputs)
-    h = torch.matmul(x, theta.w1) + theta.b1
+    h = py_utils.MatmulBias(x, theta.w1, theta.b1)
     h = activations.GetFn(p.activation)(h)
     if p.relu_dropout_prob and not self.do_eval:
       h = py_utils.DeterministicDropout(h, 1.0 - p.relu_dropout_prob)
-    out = torch.matmul(h, theta.w2) + theta.b2
+    out = py_utils.MatmulBias(h, theta.w2, theta.b2)
     if p.residual_weight != 1.0:
       out = out * p.residual_weight
     if paddings is not None:

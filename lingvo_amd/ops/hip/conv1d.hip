@@ -93,43 +93,87 @@ __global__ void dwconv_bwd_dx(const unsigned short* __restrict__ dy,
 }
 
 // dw[j,d] = sum_{b,t} dy[b,t,d] * x[b,t+j-pad,d]; db[d] = sum dy.
-// One thread per channel d within a (d-block, row-stripe) grid; K+1 fp32
-// register accumulators; one atomicAdd per (thread, tap) at the end.
-__global__ void dwconv_bwd_dw(const unsigned short* __restrict__ dy,
-                              const unsigned short* __restrict__ x,
-                              float* __restrict__ dw_acc,
-                              float* __restrict__ db_acc, int B, int T,
-                              int D, int K, int pad, int nstripes) {
-  const int d = blockIdx.x * blockDim.x + threadIdx.x;
-  if (d >= D) return;
-  const int stripe = blockIdx.y;
+// LDS-staged: each block owns 128 channels x a 64-row chunk; x and dy
+// tiles are staged once (2-pass global traffic total), then 256 threads
+// (d, tap-parity) accumulate from LDS. K/2 fp32 tap accumulators per
+// thread; one atomicAdd per (tap, channel) per chunk-group.
+constexpr int DW_CHUNK = 64;
+constexpr int DW_DBLK = 128;
+
+__global__ __launch_bounds__(256) void dwconv_bwd_dw(
+    const unsigned short* __restrict__ dy,
+    const unsigned short* __restrict__ x, float* __restrict__ dw_acc,
+    float* __restrict__ db_acc, int B, int T, int D, int K, int pad,
+    int ngroups) {
+  __shared__ unsigned short x_s[DW_CHUNK + MAXK - 1][DW_DBLK];
+  __shared__ unsigned short dy_s[DW_CHUNK][DW_DBLK];
+  const int tid = threadIdx.x;
+  const int d_local = tid % DW_DBLK;
+  const int jg = tid / DW_DBLK;  // tap parity: owns taps j with j%2==jg
+  const int d = blockIdx.x * DW_DBLK + d_local;
   const long rows = (long)B * T;
-  // Contiguous row chunks: consecutive iterations share K-1 of the K
-  // x-taps, so the taps stay L1-resident.
-  const long chunk = (rows + nstripes - 1) / nstripes;
-  const long r0 = stripe * chunk;
-  const long r1 = min(rows, r0 + chunk);
-  float dw[MAXK];
-  for (int j = 0; j < K; ++j) dw[j] = 0.f;
+  const int xrows = DW_CHUNK + K - 1;
+
+  float dw[MAXK / 2 + 1];
+  for (int j = 0; j < MAXK / 2 + 1; ++j) dw[j] = 0.f;
   float db = 0.f;
-  for (long r = r0; r < r1; ++r) {
-    const int t = (int)(r % T);
-    const long b = r / T;
-    float g = bf16_bits_to_float(dy[r * D + d]);
-    db += g;
-    const long xbase = (b * T) * (long)D + d;
-    for (int j = 0; j < K; ++j) {
-      int ts = t + j - pad;
-      float xv = (ts < 0 || ts >= T)
-                     ? 0.f
-                     : bf16_bits_to_float(x[xbase + (long)ts * D]);
-      dw[j] += g * xv;
+  const bool d_ok = d < D;
+
+  const long nchunks = (rows + DW_CHUNK - 1) / DW_CHUNK;
+  for (long c = blockIdx.y; c < nchunks; c += ngroups) {
+    const long r0 = c * DW_CHUNK;
+    // Stage x window [r0 - pad, r0 - pad + xrows) and dy [r0, r0+CHUNK).
+    for (int i = tid; i < xrows * (DW_DBLK / 8); i += 256) {
+      int row = i / (DW_DBLK / 8);
+      int d8 = (i % (DW_DBLK / 8)) * 8;
+      long fr = r0 - pad + row;
+      ushortx8 v;
+      int gd = blockIdx.x * DW_DBLK + d8;
+      if (fr >= 0 && fr < rows && gd + 7 < D) {
+        v = *reinterpret_cast<const ushortx8*>(x + fr * D + gd);
+      } else {
+        for (int e = 0; e < 8; ++e) v[e] = 0;
+      }
+      *reinterpret_cast<ushortx8*>(&x_s[row][d8]) = v;
     }
+    for (int i = tid; i < DW_CHUNK * (DW_DBLK / 8); i += 256) {
+      int row = i / (DW_DBLK / 8);
+      int d8 = (i % (DW_DBLK / 8)) * 8;
+      long fr = r0 + row;
+      ushortx8 v;
+      int gd = blockIdx.x * DW_DBLK + d8;
+      if (fr < rows && gd + 7 < D) {
+        v = *reinterpret_cast<const ushortx8*>(dy + fr * D + gd);
+      } else {
+        for (int e = 0; e < 8; ++e) v[e] = 0;
+      }
+      *reinterpret_cast<ushortx8*>(&dy_s[row][d8]) = v;
+    }
+    __syncthreads();
+
+    if (d_ok) {
+      const int nrows = (int)min((long)DW_CHUNK, rows - r0);
+      for (int lr = 0; lr < nrows; ++lr) {
+        float g = bf16_bits_to_float(dy_s[lr][d_local]);
+        const int t = (int)((r0 + lr) % T);
+        if (jg == 0) db += g;
+        for (int j = jg, ji = 0; j < K; j += 2, ++ji) {
+          int ts = t + j - pad;
+          if (ts >= 0 && ts < T) {
+            dw[ji] += g * bf16_bits_to_float(x_s[lr + j][d_local]);
+          }
+        }
+      }
+    }
+    __syncthreads();
   }
-  for (int j = 0; j < K; ++j) {
-    if (dw[j] != 0.f) atomicAdd(dw_acc + (long)j * D + d, dw[j]);
+
+  if (d_ok) {
+    for (int j = jg, ji = 0; j < K; j += 2, ++ji) {
+      if (dw[ji] != 0.f) atomicAdd(dw_acc + (long)j * D + d, dw[ji]);
+    }
+    if (jg == 0 && db != 0.f) atomicAdd(db_acc + d, db);
   }
-  if (db != 0.f) atomicAdd(db_acc + d, db);
 }
 
 }  // namespace
@@ -169,12 +213,15 @@ std::vector<torch::Tensor> dwconv1d_bwd(torch::Tensor dy, torch::Tensor x,
                      (const unsigned short*)dy.data_ptr(),
                      (const unsigned short*)w.data_ptr(),
                      (unsigned short*)dx.data_ptr(), B, T, D, K, (int)pad);
-  int nstripes = (int)std::min<long>(512, std::max<long>(1, (long)B * T / 8));
-  dim3 grid_w((D + 255) / 256, nstripes);
+  long nchunks = ((long)B * T + DW_CHUNK - 1) / DW_CHUNK;
+  int dblks = (D + DW_DBLK - 1) / DW_DBLK;
+  int ngroups = (int)std::min<long>(std::max<long>(1, 1024 / dblks),
+                                    nchunks);
+  dim3 grid_w(dblks, ngroups);
   hipLaunchKernelGGL(dwconv_bwd_dw, grid_w, dim3(256), 0, stream,
                      (const unsigned short*)dy.data_ptr(),
                      (const unsigned short*)x.data_ptr(),
                      dw.data_ptr<float>(), db.data_ptr<float>(), B, T, D, K,
-                     (int)pad, nstripes);
+                     (int)pad, ngroups);
   return {dx, dw, db};
 }
