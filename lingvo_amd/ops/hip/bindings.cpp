@@ -53,7 +53,11 @@ torch::Tensor dropout_bwd(torch::Tensor dy, int64_t seed,
                           c10::optional<torch::Tensor> step_seed,
                           double keep);
 
+// input_pipeline.cpp
+void RegisterInputPipeline(py::module_& m);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  RegisterInputPipeline(m);
   m.def("dropout_fwd", &dropout_fwd, "Fused dropout fwd");
   m.def("dropout_bwd", &dropout_bwd, "Fused dropout bwd");
   m.def("xent_fwd", &xent_fwd, "Fused softmax-xent fwd");
