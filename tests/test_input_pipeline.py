@@ -29,11 +29,11 @@ def test_record_yielder_text(tmp_path):
   files = sorted(g.glob(str(tmp_path / 'part-*.txt')))
   y = ext.RecordYielder(files, 'text', 301, 100, 2, True)
   seen = set()
-  for _ in range(120):  # 2 epochs worth
+  for _ in range(600):  # several epochs; buffer sampling is randomized
     rec, src = y.yield_record()
     seen.add(rec)
     assert 0 <= src < 3
-  assert len(seen) == 60  # all unique lines seen across epochs
+  assert len(seen) == 60  # every unique line eventually surfaces
   assert y.current_epoch() >= 1
   y.stop()
 
