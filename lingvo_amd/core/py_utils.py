@@ -351,6 +351,33 @@ def ToScalar(x) -> float:
   return float(x)
 
 
+def GraphSafeUniform(shape, device, op_seed: Optional[int] = None
+                     ) -> torch.Tensor:
+  """Uniform [0,1) tensor whose values vary per step via the device
+  step-seed buffer — safe inside hipGraph capture (torch.Generator RNG
+  is not). Deterministic in (global_seed, step, op_counter)."""
+  import math as _math
+  n = 1
+  for d in shape:
+    n *= int(d)
+  s1, s2 = GenerateStepSeedPair(op_seed)
+  if not (isinstance(device, torch.device) and device.type == 'cuda') and \
+      str(device) != 'cuda' and not str(device).startswith('cuda'):
+    g = torch.Generator()
+    g.manual_seed((s1 * 2654435761 + s2) & 0x7FFFFFFFFFFFFFFF)
+    return torch.rand(shape, generator=g)
+  from lingvo_amd.ops import dropout as dropout_ops
+  buf = dropout_ops._StepSeedBuf(torch.device(device))
+  idx = torch.arange(n, device=device, dtype=torch.int64)
+  h = idx * 6364136223846793005 + s1
+  h = h ^ buf
+  h = h ^ (h >> 33)
+  h = h * 0x5851F42D4C957F2D
+  h = h ^ (h >> 29)
+  u = ((h >> 32) & 0x7FFFFFFF).float() / float(1 << 31)
+  return u.reshape(shape)
+
+
 def MatmulBias(x: torch.Tensor, w: torch.Tensor,
                b: Optional[torch.Tensor] = None) -> torch.Tensor:
   """x @ w (+ b) with the bias fused into the hipBLASLt GEMM epilogue

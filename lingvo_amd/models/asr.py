@@ -78,11 +78,16 @@ class ConformerEncoder(BaseLayer):
              '(0 = model_dim).')
     p.Define('conformer_tpl', conformer_lib.ConformerLayer.Params(),
              'Block template.')
+    from lingvo_amd.layers import spectrum_augmenter
+    p.Define('specaug_tpl', spectrum_augmenter.SpectrumAugmenter.Params(),
+             'SpecAugment params; None disables.')
     return p
 
   def __init__(self, params):
     super().__init__(params)
     p = self.p
+    if p.specaug_tpl is not None:
+      self.CreateChild('specaug', p.specaug_tpl)
     self.CreateChild('sub', conformer_lib.ConvSubsampling.Params().Set(
         input_freq_dim=p.input_dim, output_dim=p.model_dim,
         channels=p.subsample_channels or min(p.model_dim, 256)))
@@ -98,6 +103,8 @@ class ConformerEncoder(BaseLayer):
   def FProp(self, theta: NestedMap, src_inputs: torch.Tensor,
             paddings: torch.Tensor):
     x = src_inputs.to(self.fprop_dtype)
+    if self.p.specaug_tpl is not None and not self.do_eval:
+      x = self.specaug.FProp(theta.specaug, x, paddings)
     x, out_pad = self.sub.FProp(theta.sub, x, paddings)
     for i, block in enumerate(self.blocks):
       x = block.FProp(theta.blocks[i], x, out_pad)

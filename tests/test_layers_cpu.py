@@ -165,3 +165,62 @@ def test_group_norm_padding_invariance():
   x2[0, 6:] = 123.0  # changing padded region must not affect output
   out2 = gn.FProp(gn.theta, x2, pad)
   assert torch.allclose(out1[0, :6], out2[0, :6], atol=1e-5)
+
+
+def test_spectrum_augmenter():
+  from lingvo_amd.layers import spectrum_augmenter
+  p = spectrum_augmenter.SpectrumAugmenter.Params().Set(
+      name='sa', freq_mask_max_bins=5, freq_mask_count=1,
+      time_mask_max_frames=4, time_mask_count=1)
+  sa = p.Instantiate()
+  x = torch.ones(2, 20, 16)
+  pad = py_utils.PaddingsFromLengths(torch.tensor([20, 15]), 20)
+  with py_utils.StepSeedScope(1, 0):
+    out = sa.FProp(sa.theta, x, pad)
+  assert out.shape == x.shape
+  assert (out == 0).any()  # something was masked
+  # deterministic per (seed, step)
+  with py_utils.StepSeedScope(1, 0):
+    out2 = sa.FProp(sa.theta, x, pad)
+  assert torch.equal(out, out2)
+  sa.eval()
+  assert torch.equal(sa.FProp(sa.theta, x, pad),
+                     py_utils.ApplyPadding(pad, x) if False else x)
+
+
+def test_mel_frontend():
+  from lingvo_amd.layers import asr_frontend
+  p = asr_frontend.MelAsrFrontend.Params().Set(name='fe', num_bins=40)
+  fe = p.Instantiate()
+  wav = torch.randn(2, 16000)
+  pad = py_utils.PaddingsFromLengths(torch.tensor([16000, 8000]), 16000)
+  mel, out_pad = fe.FProp(fe.theta, wav, pad)
+  assert mel.shape[0] == 2 and mel.shape[2] == 40
+  assert torch.isfinite(mel).all()
+  lens = py_utils.LengthsFromPaddings(out_pad)
+  assert lens[1] < lens[0]
+
+
+def test_conv_layers_with_time_padding():
+  from lingvo_amd.layers import conv_layers_with_time_padding as conv_tp
+  p = conv_tp.Conv2DLayerWithPadding.Params().Set(
+      name='c', filter_shape=(3, 3, 1, 4), filter_stride=(2, 2),
+      random_seed=1)
+  layer = p.Instantiate()
+  x = torch.randn(2, 12, 8, 1)
+  pad = py_utils.PaddingsFromLengths(torch.tensor([12, 6]), 12)
+  out, out_pad = layer.FProp(layer.theta, x, pad)
+  assert out.shape[0] == 2 and out.shape[3] == 4
+  assert out_pad.shape[1] == out.shape[1]
+
+  dp = conv_tp.CausalDepthwiseConv1DLayer.Params().Set(
+      name='d', kernel_size=3, dim=8, random_seed=1)
+  dl = dp.Instantiate()
+  x2 = torch.randn(2, 10, 8)
+  out2, _ = dl.FProp(dl.theta, x2, torch.zeros(2, 10))
+  assert out2.shape == x2.shape
+
+  gp = conv_tp.GlobalPoolingLayer.Params().Set(name='g').Instantiate()
+  pooled = gp.FProp(gp.theta, x2,
+                    py_utils.PaddingsFromLengths(torch.tensor([10, 4]), 10))
+  assert pooled.shape == (2, 8)
