@@ -40,9 +40,11 @@ class SpectrumAugmenter(BaseLayer):
         cap = (lengths.to(x.device).float() *
                self.p.time_mask_max_ratio).long()
         widths = torch.minimum(widths, cap)
+      # NOTE: keep everything device-side factories/scalars — creating a
+      # tensor from a Python scalar is a blocking H2D copy, which is
+      # forbidden inside hipGraph capture.
       starts = (py_utils.GraphSafeUniform((b,), x.device) *
-                (torch.as_tensor(size, device=x.device, dtype=torch.float)
-                 - widths.float()).clamp_min(1)).long()
+                (float(size) - widths.float()).clamp_min(1)).long()
       pos = torch.arange(size, device=x.device)[None, :]
       span = (pos >= starts[:, None]) & (pos < (starts + widths)[:, None])
       mask = mask * (~span).float()
