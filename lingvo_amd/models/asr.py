@@ -314,6 +314,22 @@ class AsrModel(BaseTask):
     return NestedMap(topk_decoded=hyps,
                      transcripts=input_batch.tgt.ids)
 
+  def Inference(self) -> NestedMap:
+    """Named inference subgraphs (reference base_model.py:943)."""
+
+    def default(src_inputs, paddings):
+      enc, enc_pad = self.encoder.FProp(self.theta.encoder,
+                                        src_inputs, paddings)
+      hyps = self.decoder.GreedyDecode(self.theta.decoder, enc, enc_pad)
+      return NestedMap(hyps=hyps)
+
+    def encode(src_inputs, paddings):
+      enc, enc_pad = self.encoder.FProp(self.theta.encoder,
+                                        src_inputs, paddings)
+      return NestedMap(encoded=enc, padding=enc_pad)
+
+    return NestedMap(default=default, encode=encode)
+
   def CreateDecoderMetrics(self) -> NestedMap:
     return NestedMap(wer=metrics_lib.WerMetric(),
                      num_samples_in_batch=metrics_lib.AverageMetric())

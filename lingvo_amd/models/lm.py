@@ -205,3 +205,16 @@ class LanguageModel(BaseTask):
       xent = self.lm.XentLoss(self.theta.lm, act, input_batch.labels,
                               input_batch.weights)
     return NestedMap(log_pplx=xent.avg_xent.reshape(1))
+
+  def Inference(self) -> NestedMap:
+    """Scoring subgraph (reference base_model.py:943)."""
+
+    def default(ids, paddings):
+      act = self.lm.FProp(self.theta.lm, ids, paddings)
+      labels = ids.roll(-1, dims=1)
+      weights = (1.0 - paddings)
+      xent = self.lm.XentLoss(self.theta.lm, act, labels, weights)
+      return NestedMap(per_example_xent=xent.per_example_xent,
+                       avg_xent=xent.avg_xent)
+
+    return NestedMap(default=default)

@@ -1,0 +1,67 @@
+"""Executor/programs, early stop, inference export/predictor tests."""
+
+import json
+import os
+
+import pytest
+import torch
+
+from lingvo_amd.core import registry
+
+
+def test_executor_multi_program(tmp_path):
+  from lingvo_amd.runtime import program as program_lib
+  model_p = registry.GetParams('image.mnist.LeNet5', 'Train')
+  model_p.task.random_seed = 3
+  sched_p = program_lib.SimpleProgramSchedule.Params()
+  sched_p.train_program.steps_per_loop = 3
+  ep = program_lib.EvalProgram.Params().Set(name='eval_dev',
+                                            steps_per_loop=2)
+  ep.Define('cls', program_lib.EvalProgram, 'class')
+  sched_p.eval_programs = [ep]
+  ex = program_lib.Executor(model_p, str(tmp_path), sched_p,
+                            device='cpu', max_steps=6)
+  ex.Start()
+  assert ex.task.global_step == 6
+  with open(tmp_path / 'eval_dev' / 'metrics.jsonl') as f:
+    recs = [json.loads(l) for l in f]
+  assert len(recs) == 2 and 'loss' in recs[0]
+  # checkpoint written
+  assert os.path.exists(tmp_path / 'train' / 'checkpoint')
+
+
+def test_metric_history_and_early_stop(tmp_path):
+  from lingvo_amd.core.early_stop import EarlyStop, MetricHistory
+  mh = MetricHistory(str(tmp_path), 'eval', 'loss', minimize=True)
+  for step, val in [(10, 5.0), (20, 3.0), (30, 3.5), (40, 3.4)]:
+    mh.ConditionalAppend(step, val)
+  assert mh.BestStep() == 20
+  es = EarlyStop(EarlyStop.Params().Set(
+      metric_history=mh, window=15, min_steps=0))
+  assert not es.Stop(30)
+  assert es.Stop(40)   # 40 - 20 > 15
+  assert es.Stop(41)   # latched
+
+
+def test_inference_export_and_predictor(tmp_path):
+  from lingvo_amd.runtime.inference import InferenceGraphExporter, Predictor
+  model_p = registry.GetParams(
+      'asr.librispeech.Librispeech960WpmConformerL', 'Train')
+  model_p.task.fprop_dtype = torch.float32
+  model_p.task.encoder.Set(num_layers=1, model_dim=64, num_heads=1,
+                           kernel_size=4)
+  model_p.task.decoder.Set(rnn_cell_dim=32, source_dim=64, emb_dim=16,
+                           vocab_size=32)
+  model_p.input.Set(batch_size=2, frame_len=32, target_len=6,
+                    vocab_size=32)
+  model_p.task.random_seed = 4
+  path = str(tmp_path / 'inference.pt')
+  InferenceGraphExporter.Export(model_p, path)
+  pred = Predictor(path, device='cpu')
+  assert 'default' in pred.subgraphs and 'encode' in pred.subgraphs
+  src = torch.randn(2, 32, 80)
+  pad = torch.zeros(2, 32)
+  out = pred.Run('default', src_inputs=src, paddings=pad)
+  assert out.hyps.shape[0] == 2
+  enc = pred.Run('encode', src_inputs=src, paddings=pad)
+  assert enc.encoded.shape[-1] == 64
