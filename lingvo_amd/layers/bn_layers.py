@@ -108,6 +108,10 @@ class GroupNormLayer(BaseLayer):
     p = self.p
     b, t, d = inputs.shape
     g = p.num_groups
+    if inputs.is_cuda and d % g == 0 and d // g <= 64:
+      from lingvo_amd.ops import group_norm as gn_ops
+      return gn_ops.group_norm(inputs, theta.gamma, theta.beta, paddings,
+                               g, p.epsilon)
     xf = inputs.float().reshape(b, t, g, d // g)
     # Per (b, group) moments over (t, d/g), excluding padded frames.
     if paddings is not None:

@@ -200,8 +200,13 @@ class AsrDecoder(BaseLayer):
     neg_mask = (enc_paddings.float() * -1e30).unsqueeze(1)  # [B,1,S]
 
     fgb = self.rnns[0].p.forget_gate_bias
+    use_fused = enc.is_cuda and dt == torch.bfloat16
 
     def lstm_pointwise(gates, c_prev):
+      if use_fused:
+        from lingvo_amd.ops import lstm_gates as lstm_ops
+        return lstm_ops.lstm_gates(gates, c_prev, fgb,
+                                   cap if cap is not None else 0.0)
       i_i, i_g, f_g, o_g = gates.split([h, h, h, h], dim=-1)
       if fgb:
         f_g = f_g + fgb

@@ -53,11 +53,37 @@ torch::Tensor dropout_bwd(torch::Tensor dy, int64_t seed,
                           c10::optional<torch::Tensor> step_seed,
                           double keep);
 
+// group_norm.hip
+std::vector<torch::Tensor> group_norm_fwd(torch::Tensor x,
+                                          torch::Tensor gamma,
+                                          torch::Tensor beta,
+                                          c10::optional<torch::Tensor> pad,
+                                          int64_t groups, double eps);
+std::vector<torch::Tensor> group_norm_bwd(torch::Tensor dy, torch::Tensor x,
+                                          torch::Tensor gamma,
+                                          c10::optional<torch::Tensor> pad,
+                                          torch::Tensor mean,
+                                          torch::Tensor rstd,
+                                          int64_t groups);
+
+// lstm_gates.hip
+std::vector<torch::Tensor> lstm_gates_fwd_op(torch::Tensor gates,
+                                             torch::Tensor c0, double fgb,
+                                             double cap);
+std::vector<torch::Tensor> lstm_gates_bwd_op(
+    torch::Tensor gates, torch::Tensor c0, torch::Tensor c1,
+    torch::Tensor dm1, c10::optional<torch::Tensor> dc1, double fgb,
+    double cap);
+
 // input_pipeline.cpp
 void RegisterInputPipeline(py::module_& m);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   RegisterInputPipeline(m);
+  m.def("group_norm_fwd", &group_norm_fwd, "Fused padded GroupNorm fwd");
+  m.def("group_norm_bwd", &group_norm_bwd, "Fused padded GroupNorm bwd");
+  m.def("lstm_gates_fwd", &lstm_gates_fwd_op, "Fused LSTM gates fwd");
+  m.def("lstm_gates_bwd", &lstm_gates_bwd_op, "Fused LSTM gates bwd");
   m.def("dropout_fwd", &dropout_fwd, "Fused dropout fwd");
   m.def("dropout_bwd", &dropout_bwd, "Fused dropout bwd");
   m.def("xent_fwd", &xent_fwd, "Fused softmax-xent fwd");

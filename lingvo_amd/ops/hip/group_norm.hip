@@ -1,0 +1,199 @@
+// Fused padded-aware GroupNorm fwd+bwd for gfx950
+// (reference op: lingvo/core/bn_layers.py:747 GroupNormLayer).
+//
+// x: [B, T, D] bf16, G groups over the channel dim; moments are per
+// (b, g) over (valid T) x (D/G); padded frames are excluded from the
+// moments and zeroed in the output. y = xhat * (1+gamma) + beta.
+//
+// One block (256 threads) per (b, g): block-reduce moments, then a
+// second pass normalizes. dgamma/dbeta accumulate via fp32 atomics
+// (small: D columns).
+
+#include <ATen/cuda/CUDAContext.h>
+#include <torch/extension.h>
+
+#include "common.h"
+
+namespace {
+
+constexpr int GN_BLOCK = 256;
+constexpr int GN_WAVES = GN_BLOCK / WAVE_SIZE;
+
+__global__ __launch_bounds__(GN_BLOCK) void gn_fwd_kernel(
+    const unsigned short* __restrict__ x,
+    const unsigned short* __restrict__ gamma,
+    const unsigned short* __restrict__ beta,
+    const unsigned short* __restrict__ paddings,  // [B,T] bf16 or null
+    unsigned short* __restrict__ y, float* __restrict__ mean_out,
+    float* __restrict__ rstd_out, int B, int T, int D, int G, float eps) {
+  __shared__ float scratch[GN_WAVES];
+  const int b = blockIdx.x / G;
+  const int g = blockIdx.x % G;
+  const int cg = D / G;
+  const int tid = threadIdx.x;
+
+  float sum = 0.f, sumsq = 0.f, count = 0.f;
+  for (int i = tid; i < T * cg; i += GN_BLOCK) {
+    int t = i / cg;
+    int c = g * cg + i % cg;
+    float pad = paddings ? bf16_bits_to_float(paddings[(long)b * T + t])
+                         : 0.f;
+    if (pad < 0.5f) {
+      float v = bf16_bits_to_float(x[((long)b * T + t) * D + c]);
+      sum += v;
+      sumsq += v * v;
+      count += 1.f;
+    }
+  }
+  sum = block_reduce_sum<GN_WAVES>(sum, scratch);
+  sumsq = block_reduce_sum<GN_WAVES>(sumsq, scratch);
+  count = block_reduce_sum<GN_WAVES>(count, scratch);
+  count = fmaxf(count, 1.f);
+  float mu = sum / count;
+  float var = sumsq / count - mu * mu;
+  float rstd = rsqrtf(var + eps);
+  if (tid == 0) {
+    mean_out[(long)b * G + g] = mu;
+    rstd_out[(long)b * G + g] = rstd;
+  }
+
+  for (int i = tid; i < T * cg; i += GN_BLOCK) {
+    int t = i / cg;
+    int c = g * cg + i % cg;
+    float pad = paddings ? bf16_bits_to_float(paddings[(long)b * T + t])
+                         : 0.f;
+    float v = bf16_bits_to_float(x[((long)b * T + t) * D + c]);
+    float w = 1.f + bf16_bits_to_float(gamma[c]);
+    float bb = bf16_bits_to_float(beta[c]);
+    float out = ((v - mu) * rstd * w + bb) * (1.f - pad);
+    y[((long)b * T + t) * D + c] = float_to_bf16_bits(out);
+  }
+}
+
+__global__ __launch_bounds__(GN_BLOCK) void gn_bwd_kernel(
+    const unsigned short* __restrict__ dy,
+    const unsigned short* __restrict__ x,
+    const unsigned short* __restrict__ gamma,
+    const unsigned short* __restrict__ paddings,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    unsigned short* __restrict__ dx, float* __restrict__ dgamma,
+    float* __restrict__ dbeta, int B, int T, int D, int G) {
+  __shared__ float scratch[GN_WAVES];
+  const int b = blockIdx.x / G;
+  const int g = blockIdx.x % G;
+  const int cg = D / G;
+  const int tid = threadIdx.x;
+  const float mu = mean[(long)b * G + g];
+  const float rs = rstd[(long)b * G + g];
+
+  // Pass 1: reductions over the group: s1=sum(dyw), s2=sum(dyw*xhat),
+  // count; and per-channel dgamma/dbeta partials via LDS.
+  __shared__ float dg_s[64];  // cg <= 64 supported
+  __shared__ float db_s[64];
+  for (int i = tid; i < cg; i += GN_BLOCK) {
+    dg_s[i] = 0.f;
+    db_s[i] = 0.f;
+  }
+  __syncthreads();
+  float s1 = 0.f, s2 = 0.f, count = 0.f;
+  for (int i = tid; i < T * cg; i += GN_BLOCK) {
+    int t = i / cg;
+    int ci = i % cg;
+    int c = g * cg + ci;
+    float pad = paddings ? bf16_bits_to_float(paddings[(long)b * T + t])
+                         : 0.f;
+    if (pad < 0.5f) {
+      float xv = bf16_bits_to_float(x[((long)b * T + t) * D + c]);
+      float dyv = bf16_bits_to_float(dy[((long)b * T + t) * D + c]);
+      float w = 1.f + bf16_bits_to_float(gamma[c]);
+      float xhat = (xv - mu) * rs;
+      float dxhat = dyv * w;
+      s1 += dxhat;
+      s2 += dxhat * xhat;
+      count += 1.f;
+      atomicAdd(&dg_s[ci], dyv * xhat);
+      atomicAdd(&db_s[ci], dyv);
+    }
+  }
+  s1 = block_reduce_sum<GN_WAVES>(s1, scratch);
+  s2 = block_reduce_sum<GN_WAVES>(s2, scratch);
+  count = block_reduce_sum<GN_WAVES>(count, scratch);
+  count = fmaxf(count, 1.f);
+  const float inv_n = 1.f / count;
+
+  for (int i = tid; i < cg; i += GN_BLOCK) {
+    int c = g * cg + i;
+    if (dg_s[i] != 0.f) atomicAdd(dgamma + c, dg_s[i]);
+    if (db_s[i] != 0.f) atomicAdd(dbeta + c, db_s[i]);
+  }
+
+  for (int i = tid; i < T * cg; i += GN_BLOCK) {
+    int t = i / cg;
+    int c = g * cg + i % cg;
+    float pad = paddings ? bf16_bits_to_float(paddings[(long)b * T + t])
+                         : 0.f;
+    float xv = bf16_bits_to_float(x[((long)b * T + t) * D + c]);
+    float dyv = bf16_bits_to_float(dy[((long)b * T + t) * D + c]);
+    float w = 1.f + bf16_bits_to_float(gamma[c]);
+    float xhat = (xv - mu) * rs;
+    float dxhat = dyv * w;
+    float val = rs * (dxhat - inv_n * (s1 + xhat * s2)) * (1.f - pad);
+    dx[((long)b * T + t) * D + c] = float_to_bf16_bits(val);
+  }
+}
+
+}  // namespace
+
+std::vector<torch::Tensor> group_norm_fwd(torch::Tensor x,
+                                          torch::Tensor gamma,
+                                          torch::Tensor beta,
+                                          c10::optional<torch::Tensor> pad,
+                                          int64_t groups, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 3 &&
+              x.scalar_type() == torch::kBFloat16);
+  const int B = x.size(0), T = x.size(1), D = x.size(2);
+  const int G = (int)groups;
+  TORCH_CHECK(D % G == 0 && D / G <= 64, "cg must be <= 64");
+  auto y = torch::empty_like(x);
+  auto opts = x.options().dtype(torch::kFloat32);
+  auto mean = torch::empty({(long)B * G}, opts);
+  auto rstd = torch::empty({(long)B * G}, opts);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(gn_fwd_kernel, dim3(B * G), dim3(GN_BLOCK), 0, stream,
+                     (const unsigned short*)x.data_ptr(),
+                     (const unsigned short*)gamma.data_ptr(),
+                     (const unsigned short*)beta.data_ptr(),
+                     pad.has_value()
+                         ? (const unsigned short*)pad->data_ptr()
+                         : nullptr,
+                     (unsigned short*)y.data_ptr(), mean.data_ptr<float>(),
+                     rstd.data_ptr<float>(), B, T, D, G, (float)eps);
+  return {y, mean, rstd};
+}
+
+std::vector<torch::Tensor> group_norm_bwd(torch::Tensor dy, torch::Tensor x,
+                                          torch::Tensor gamma,
+                                          c10::optional<torch::Tensor> pad,
+                                          torch::Tensor mean,
+                                          torch::Tensor rstd,
+                                          int64_t groups) {
+  const int B = x.size(0), T = x.size(1), D = x.size(2);
+  const int G = (int)groups;
+  auto dx = torch::empty_like(x);
+  auto opts = x.options().dtype(torch::kFloat32);
+  auto dgamma = torch::zeros({D}, opts);
+  auto dbeta = torch::zeros({D}, opts);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(gn_bwd_kernel, dim3(B * G), dim3(GN_BLOCK), 0, stream,
+                     (const unsigned short*)dy.data_ptr(),
+                     (const unsigned short*)x.data_ptr(),
+                     (const unsigned short*)gamma.data_ptr(),
+                     pad.has_value()
+                         ? (const unsigned short*)pad->data_ptr()
+                         : nullptr,
+                     mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                     (unsigned short*)dx.data_ptr(),
+                     dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), B,
+                     T, D, G);
+  return {dx, dgamma, dbeta};
+}

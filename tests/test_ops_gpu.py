@@ -116,3 +116,68 @@ def test_fused_dropout_fwd_bwd():
   assert torch.equal(res.grad, g)
   want_dx = (g.float() * mask / keep)
   assert (x.grad.float() - want_dx).abs().max() < 0.05
+
+
+@gpu
+def test_group_norm_matches_ref():
+  from lingvo_amd.ops import group_norm as gn_ops
+  from lingvo_amd.core import py_utils as pu
+  torch.manual_seed(3)
+  b, t, d, g = 3, 25, 64, 4
+  x = torch.randn(b, t, d, device='cuda',
+                  dtype=torch.bfloat16).requires_grad_(True)
+  gamma = (torch.randn(d, device='cuda') * 0.1).requires_grad_(True)
+  beta = (torch.randn(d, device='cuda') * 0.1).requires_grad_(True)
+  pad = pu.PaddingsFromLengths(torch.tensor([25, 17, 9]), t).cuda()
+  y = gn_ops.group_norm(x, gamma, beta, pad, g)
+  gout = torch.randn_like(y)
+  y.backward(gout)
+
+  # fp32 reference (same math as the CPU GroupNormLayer path)
+  xr = x.detach().float().requires_grad_(True)
+  gr = gamma.detach().clone().requires_grad_(True)
+  br = beta.detach().clone().requires_grad_(True)
+  xf = xr.reshape(b, t, g, d // g)
+  mask = (1.0 - pad)[:, :, None, None]
+  count = (mask.sum(dim=(1, 3), keepdim=True) * (d // g)).clamp_min(1.0)
+  mean = (xf * mask).sum(dim=(1, 3), keepdim=True) / count
+  var = ((xf - mean) ** 2 * mask).sum(dim=(1, 3), keepdim=True) / count
+  ref = ((xf - mean) * torch.rsqrt(var + 1e-3)).reshape(b, t, d)
+  ref = (ref * (1 + gr) + br) * (1.0 - pad)[:, :, None]
+  ref.backward(gout.float())
+
+  assert (y.float() - ref.detach()).abs().max() < 0.05
+  assert (x.grad.float() - xr.grad).abs().max() < 0.08
+  for got, want in [(gamma.grad, gr.grad), (beta.grad, br.grad)]:
+    scale = max(1.0, float(want.abs().max()))
+    assert (got - want).abs().max() / scale < 0.05
+
+
+@gpu
+def test_lstm_gates_matches_ref():
+  from lingvo_amd.ops import lstm_gates as lstm_ops
+  torch.manual_seed(4)
+  b, h = 8, 32
+  gates = torch.randn(b, 4 * h, device='cuda',
+                      dtype=torch.bfloat16).requires_grad_(True)
+  c0 = torch.randn(b, h, device='cuda',
+                   dtype=torch.bfloat16).requires_grad_(True)
+  cap, fgb = 10.0, 0.5
+  c1, m1 = lstm_ops.lstm_gates(gates, c0, fgb, cap)
+  g1 = torch.randn_like(c1)
+  g2 = torch.randn_like(m1)
+  (c1 * g1 + m1 * g2).sum().backward()
+
+  gr = gates.detach().float().requires_grad_(True)
+  cr = c0.detach().float().requires_grad_(True)
+  i_i, i_g, f_g, o_g = gr.split([h, h, h, h], dim=-1)
+  c_ref = torch.sigmoid(f_g + fgb) * cr + \
+      torch.sigmoid(i_g) * torch.tanh(i_i)
+  c_ref = torch.clamp(c_ref, -cap, cap)
+  m_ref = torch.sigmoid(o_g) * torch.tanh(c_ref)
+  (c_ref * g1.float() + m_ref * g2.float()).sum().backward()
+
+  assert (c1.float() - c_ref.detach()).abs().max() < 0.03
+  assert (m1.float() - m_ref.detach()).abs().max() < 0.03
+  assert (gates.grad.float() - gr.grad).abs().max() < 0.05
+  assert (c0.grad.float() - cr.grad).abs().max() < 0.05
