@@ -1,0 +1,125 @@
+"""Milan: dual-encoder image/text retrieval task
+(reference lingvo/tasks/milan: dual_encoder.py, score_functions.py;
+EfficientNetB4BertAdapter in params/cxc.py).
+
+MI355X-native composition: an image tower (conv stack) and a text tower
+(transformer) projected into a shared space, trained with a symmetric
+in-batch softmax contrastive loss (the reference's dual-encoder loss).
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.nn.functional as F
+
+from lingvo_amd.core import py_utils
+from lingvo_amd.core.base_input_generator import BaseInputGenerator
+from lingvo_amd.core.base_layer import BaseLayer
+from lingvo_amd.core.base_model import BaseTask
+from lingvo_amd.core.nested_map import NestedMap
+from lingvo_amd.layers import layers as lingvo_layers
+from lingvo_amd.layers import transformer as transformer_lib
+
+
+class SyntheticImageTextInput(BaseInputGenerator):
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.batch_size = 32
+    p.Define('image_size', 64, 'Image side.')
+    p.Define('text_len', 16, 'Caption length.')
+    p.Define('vocab_size', 1000, 'Caption vocab.')
+    return p
+
+  def _InputBatch(self) -> NestedMap:
+    p = self.p
+    g = torch.Generator().manual_seed(2200 + self._batch_count)
+    return NestedMap(
+        image=torch.randn(p.batch_size, p.image_size, p.image_size, 3,
+                          generator=g),
+        text=torch.randint(1, p.vocab_size, (p.batch_size, p.text_len),
+                           generator=g),
+        text_paddings=torch.zeros(p.batch_size, p.text_len))
+
+
+class DualEncoder(BaseTask):
+  """Image tower + text tower + symmetric contrastive loss."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('joint_dim', 256, 'Shared embedding dim.')
+    p.Define('image_channels', [32, 64, 128], 'Conv tower channels.')
+    p.Define('text_dim', 256, 'Text tower dim.')
+    p.Define('text_layers', 2, 'Text transformer layers.')
+    p.Define('vocab_size', 1000, 'Caption vocab.')
+    p.Define('temperature', 0.07, 'Softmax temperature.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    convs = []
+    cin = 3
+    for i, ch in enumerate(p.image_channels):
+      convs.append(lingvo_layers.Conv2DLayer.Params().Set(
+          name=f'conv{i}', filter_shape=(3, 3, cin, ch),
+          filter_stride=(2, 2), activation='RELU'))
+      cin = ch
+    self.CreateChildren('image_convs', convs)
+    self.CreateChild('image_proj', lingvo_layers.ProjectionLayer.Params()
+                     .Set(input_dim=cin, output_dim=p.joint_dim,
+                          has_bias=True))
+    self.CreateChild('text_emb', lingvo_layers.EmbeddingLayer.Params().Set(
+        vocab_size=p.vocab_size, embedding_dim=p.text_dim))
+    self.CreateChild('text_stack',
+                     transformer_lib.StackedTransformerLayers.Params().Set(
+                         model_dim=p.text_dim, num_layers=p.text_layers,
+                         num_heads=max(1, p.text_dim // 64)))
+    self.CreateChild('text_proj', lingvo_layers.ProjectionLayer.Params()
+                     .Set(input_dim=p.text_dim, output_dim=p.joint_dim,
+                          has_bias=True))
+
+  def EncodeImage(self, theta, images):
+    x = images.to(self.fprop_dtype)
+    for i, conv in enumerate(self.image_convs):
+      x = conv.FProp(theta.image_convs[i], x)
+    x = x.mean(dim=(1, 2))  # global average pool
+    x = self.image_proj.FProp(theta.image_proj, x)
+    return F.normalize(x.float(), dim=-1)
+
+  def EncodeText(self, theta, text, paddings):
+    x = self.text_emb.EmbLookup(theta.text_emb, text.long()).to(
+        self.fprop_dtype)
+    x = self.text_stack.FProp(theta.text_stack, x, paddings)
+    mask = (1.0 - paddings).unsqueeze(-1).to(x.dtype)
+    x = (x * mask).sum(1) / mask.sum(1).clamp_min(1.0)
+    x = self.text_proj.FProp(theta.text_proj, x)
+    return F.normalize(x.float(), dim=-1)
+
+  def ComputePredictions(self, theta, input_batch):
+    img = self.EncodeImage(theta, input_batch.image)
+    txt = self.EncodeText(theta, input_batch.text,
+                          input_batch.text_paddings)
+    return NestedMap(image_emb=img, text_emb=txt)
+
+  def ComputeLoss(self, theta, predictions, input_batch):
+    p = self.p
+    sims = predictions.image_emb @ predictions.text_emb.t() / p.temperature
+    labels = torch.arange(sims.shape[0], device=sims.device)
+    loss_i2t = F.cross_entropy(sims, labels)
+    loss_t2i = F.cross_entropy(sims.t(), labels)
+    loss = 0.5 * (loss_i2t + loss_t2i)
+    acc = (sims.argmax(-1) == labels).float().mean()
+    w = torch.tensor(float(sims.shape[0]))
+    metrics = NestedMap(loss=(loss, w),
+                        retrieval_at_1=(acc.detach(), w),
+                        num_samples_in_batch=(w, torch.ones(())))
+    return metrics, NestedMap()
+
+  def Decode(self, input_batch):
+    with torch.no_grad():
+      preds = self.ComputePredictions(self.theta, input_batch)
+      sims = preds.image_emb @ preds.text_emb.t()
+    return NestedMap(ranks=sims.argsort(-1, descending=True))
