@@ -1,0 +1,111 @@
+"""Tests: builder layers, Step API, sampler, distillation."""
+
+import pytest
+import torch
+
+from lingvo_amd.core import py_utils
+from lingvo_amd.core.nested_map import NestedMap
+from lingvo_amd.layers import builder_layers
+
+
+def test_sequential_and_parallel_layers():
+  lin = builder_layers.LinearLayer.Params().Set(
+      input_dims=8, output_dims=8, random_seed=1)
+  bias = builder_layers.BiasLayer.Params().Set(dims=8, random_seed=1)
+  seq_p = builder_layers.SequentialLayer.Params().Set(
+      name='seq', sub=[lin, bias], repeat=2)
+  seq = seq_p.Instantiate()
+  x = torch.randn(3, 8)
+  out = seq.FProp(seq.theta, x)
+  assert out.shape == (3, 8)
+  assert len(seq.seq) == 4  # 2 sublayers x repeat 2
+
+  par_p = builder_layers.ParallelLayer.Params().Set(
+      name='par', sub=[lin.Copy(), lin.Copy()], merge='concat')
+  par = par_p.Instantiate()
+  out2 = par.FProp(par.theta, x)
+  assert out2.shape == (3, 16)
+
+
+def test_repeat_layer_shared_and_remat():
+  lin = builder_layers.LinearLayer.Params().Set(
+      input_dims=8, output_dims=8, random_seed=1)
+  rep = builder_layers.RepeatLayer.Params().Set(
+      name='rep', body=lin, repeat=3, per_layer_vars=False,
+      remat=True).Instantiate()
+  # single shared weight
+  assert sum(1 for _ in rep.parameters()) == 1
+  x = torch.randn(2, 8, requires_grad=True)
+  out = rep.FProp(rep.theta, x)
+  out.sum().backward()
+  assert x.grad is not None
+
+  rep2 = builder_layers.RepeatLayer.Params().Set(
+      name='rep2', body=lin, repeat=3, per_layer_vars=True).Instantiate()
+  assert sum(1 for _ in rep2.parameters()) == 3
+
+
+def test_step_api_rnn_stack():
+  from lingvo_amd.core import step as step_lib
+  from lingvo_amd.layers import rnn_cell
+  cell = rnn_cell.LSTMCellSimple.Params().Set(
+      num_input_nodes=8, num_output_nodes=8, random_seed=1)
+  stack_p = step_lib.StackStep.Params().Set(
+      name='stack',
+      sub=[step_lib.RnnStep.Params().Set(cell=cell.Copy()),
+           step_lib.RnnStep.Params().Set(cell=cell.Copy())],
+      residual_start=1)
+  stack = stack_p.Instantiate()
+  prepared = stack.PrepareExternalInputs(stack.theta, NestedMap())
+  state = stack.ZeroState(stack.theta, prepared, 2, 'cpu', torch.float32)
+  x = torch.randn(2, 8)
+  for _ in range(3):
+    out, state = stack.FProp(stack.theta, prepared,
+                             NestedMap(output=x), torch.zeros(2, 1),
+                             state)
+  assert out.output.shape == (2, 8)
+
+
+def test_target_sequence_sampler():
+  from lingvo_amd.core.target_sequence_sampler import TargetSequenceSampler
+  vocab = 16
+
+  def init_fn(b, k):
+    return NestedMap(dummy=torch.zeros(b))
+
+  def step_fn(state, prev):
+    logits = torch.zeros(prev.shape[0], vocab)
+    logits[:, 5] = 4.0  # strongly prefer token 5
+    logits[:, 2] = 2.0
+    return logits, state
+
+  p = TargetSequenceSampler.Params().Set(max_steps=20, top_k=2,
+                                         random_seed=1)
+  sampler = TargetSequenceSampler(p)
+  out = sampler.Sample(4, init_fn, step_fn)
+  assert out.ids.shape[0] == 4
+  toks = set(out.ids.flatten().tolist())
+  assert toks <= {5, 2}  # top-2 restricted
+
+
+def test_distillation_task():
+  from lingvo_amd.core.distillation_task import DistillationTask
+  from lingvo_amd.models import mnist as mnist_model
+
+  def mk_task(seed):
+    p = mnist_model.ModelV1.Params().Set(
+        name='m', hidden_dim=32, filter_shapes=[(3, 3, 1, 4)],
+        random_seed=seed)
+    p.softmax.num_classes = 10
+    return p
+
+  p = DistillationTask.Params().Set(
+      name='distill', teacher=mk_task(1), student=mk_task(2))
+  p.input = mnist_model.FakeMnistData.Params().Set(batch_size=4)
+  task = p.Instantiate()
+  m = task.TrainStep(task.GetInputBatch())
+  assert torch.isfinite(m['loss'][0])
+  assert 'distill_loss' in m
+  # teacher frozen
+  assert all(not prm.requires_grad
+             for prm in task.teacher.parameters())
