@@ -1,0 +1,121 @@
+"""Quantization-aware training utilities
+(reference lingvo/core/quant_utils.py: QuantizableLayer:62, QDomain:748,
+fake-quant schedules:1316, PassiveAsymQDomain:1606).
+
+Minimal-but-functional surface: QDomain implements symmetric fake-quant
+with a start-step schedule; QuantizableLayer mixes in QWeight/QAct
+wrappers that layers call around weights/activations.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+from lingvo_amd.core.base_layer import BaseLayer
+from lingvo_amd.core.nested_map import NestedMap
+
+
+class _FakeQuantFn(torch.autograd.Function):
+  """Straight-through symmetric fake quantization."""
+
+  @staticmethod
+  def forward(ctx, x, scale, bits):
+    qmax = 2.0 ** (bits - 1) - 1
+    return torch.clamp(torch.round(x / scale), -qmax - 1, qmax) * scale
+
+  @staticmethod
+  def backward(ctx, g):
+    return g, None, None
+
+
+class QDomain(BaseLayer):
+  """Symmetric fake-quant domain with running max calibration."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('bits', 8, 'Quantization bits.')
+    p.Define('start_step', 0, 'Enable fake quant from this step.')
+    p.Define('decay', 0.99, 'Running-max decay for calibration.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    self.register_buffer('running_max', torch.ones(()))
+    self._step = 0
+
+  def SetStep(self, step: int) -> None:
+    self._step = step
+
+  def QuantizeTensor(self, x: torch.Tensor,
+                     calibrate: bool = True) -> torch.Tensor:
+    p = self.p
+    if self._step < p.start_step:
+      return x
+    if calibrate and self.training:
+      with torch.no_grad():
+        cur = x.detach().abs().max().float().clamp_min(1e-6)
+        self.running_max.mul_(p.decay).add_(cur * (1 - p.decay))
+    qmax = 2.0 ** (p.bits - 1) - 1
+    scale = (self.running_max / qmax).to(x.dtype)
+    return _FakeQuantFn.apply(x, scale, p.bits)
+
+
+class QuantizableLayer(BaseLayer):
+  """Layers subclass this and wrap tensors with QWeight/QAct
+  (reference quant_utils.py:62)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('qdomain_default', None,
+             'QDomain params; None disables quantization.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    if self.p.qdomain_default is not None:
+      self.CreateChild('qdomain', self.p.qdomain_default)
+
+  def _Q(self, x: torch.Tensor, calibrate: bool) -> torch.Tensor:
+    if self.p.qdomain_default is None:
+      return x
+    return self.qdomain.QuantizeTensor(x, calibrate=calibrate)
+
+  def QWeight(self, w: torch.Tensor) -> torch.Tensor:
+    return self._Q(w, calibrate=False)
+
+  def QAct(self, name: str, x: torch.Tensor) -> torch.Tensor:
+    return self._Q(x, calibrate=True)
+
+  def PostTrainingStepUpdate(self, global_step: int) -> None:
+    if self.p.qdomain_default is not None:
+      self.qdomain.SetStep(global_step)
+    super().PostTrainingStepUpdate(global_step)
+
+
+class QuantizedProjectionLayer(QuantizableLayer):
+  """Example quantized layer: y = QAct(QWeight(w) @ x + b)."""
+
+  @classmethod
+  def Params(cls):
+    from lingvo_amd.core import py_utils
+    p = super().Params()
+    p.Define('input_dim', 0, 'In.')
+    p.Define('output_dim', 0, 'Out.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    from lingvo_amd.core import py_utils
+    p = self.p
+    self.CreateVariable('w', py_utils.WeightParams(
+        [p.input_dim, p.output_dim], p.params_init, p.dtype))
+    self.CreateVariable('b', py_utils.WeightParams(
+        [p.output_dim], py_utils.WeightInit.Constant(0.0), p.dtype))
+
+  def FProp(self, theta, x):
+    w = self.QWeight(theta.w)
+    return self.QAct('out', torch.matmul(x, w) + theta.b)

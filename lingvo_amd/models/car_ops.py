@@ -1,0 +1,169 @@
+"""Point-cloud 3D detection ops (reference lingvo/tasks/car/ops:
+PairwiseIou3D / NonMaxSuppression3D / AveragePrecision3D / SamplePoints
+at car_ops.cc:22-189, nms_3d_op.cc, sampling_ops.cc).
+
+Torch implementations of the core geometry (axis-aligned + yaw-rotated
+BEV overlap via polygon clipping); furthest-point sampling for
+SamplePoints. The car task model zoo itself is round-2 scope; these ops
+cover the reusable compute surface.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Tuple
+
+import torch
+
+
+def _BoxCorners2D(boxes: torch.Tensor) -> torch.Tensor:
+  """boxes [N, 5] (cx, cy, dx, dy, yaw) -> corners [N, 4, 2]."""
+  cx, cy, dx, dy, yaw = boxes.unbind(-1)
+  cos, sin = torch.cos(yaw), torch.sin(yaw)
+  hx, hy = dx / 2, dy / 2
+  base = torch.stack([
+      torch.stack([hx, hy], -1), torch.stack([-hx, hy], -1),
+      torch.stack([-hx, -hy], -1), torch.stack([hx, -hy], -1)
+  ], dim=1)  # [N, 4, 2]
+  rot = torch.stack([torch.stack([cos, -sin], -1),
+                     torch.stack([sin, cos], -1)], dim=1)  # [N, 2, 2]
+  return torch.einsum('nij,nkj->nki', rot, base) + \
+      torch.stack([cx, cy], -1)[:, None, :]
+
+
+def _PolygonArea(poly) -> float:
+  n = len(poly)
+  if n < 3:
+    return 0.0
+  s = 0.0
+  for i in range(n):
+    x1, y1 = poly[i]
+    x2, y2 = poly[(i + 1) % n]
+    s += x1 * y2 - x2 * y1
+  return abs(s) / 2.0
+
+
+def _ClipPolygon(subject, clip):
+  """Sutherland-Hodgman clipping (convex clip polygon)."""
+  def inside(pt, a, b):
+    return (b[0] - a[0]) * (pt[1] - a[1]) - (b[1] - a[1]) * \
+        (pt[0] - a[0]) >= -1e-9
+
+  def intersect(p1, p2, a, b):
+    dx1, dy1 = p2[0] - p1[0], p2[1] - p1[1]
+    dx2, dy2 = b[0] - a[0], b[1] - a[1]
+    denom = dx1 * dy2 - dy1 * dx2
+    if abs(denom) < 1e-12:
+      return p2
+    t = ((a[0] - p1[0]) * dy2 - (a[1] - p1[1]) * dx2) / denom
+    return (p1[0] + t * dx1, p1[1] + t * dy1)
+
+  output = list(subject)
+  for i in range(len(clip)):
+    a, b = clip[i], clip[(i + 1) % len(clip)]
+    input_list, output = output, []
+    if not input_list:
+      break
+    s = input_list[-1]
+    for e in input_list:
+      if inside(e, a, b):
+        if not inside(s, a, b):
+          output.append(intersect(s, e, a, b))
+        output.append(e)
+      elif inside(s, a, b):
+        output.append(intersect(s, e, a, b))
+      s = e
+  return output
+
+
+def PairwiseIou3D(boxes_a: torch.Tensor,
+                  boxes_b: torch.Tensor) -> torch.Tensor:
+  """boxes [N, 7] (cx, cy, cz, dx, dy, dz, yaw) -> IoU [N, M]
+  (reference car_ops.cc PairwiseIou3D)."""
+  n, m = boxes_a.shape[0], boxes_b.shape[0]
+  out = torch.zeros(n, m)
+  ca = _BoxCorners2D(boxes_a[:, [0, 1, 3, 4, 6]]).tolist()
+  cb = _BoxCorners2D(boxes_b[:, [0, 1, 3, 4, 6]]).tolist()
+  for i in range(n):
+    za0 = float(boxes_a[i, 2] - boxes_a[i, 5] / 2)
+    za1 = float(boxes_a[i, 2] + boxes_a[i, 5] / 2)
+    va = float(boxes_a[i, 3] * boxes_a[i, 4] * boxes_a[i, 5])
+    for j in range(m):
+      zb0 = float(boxes_b[j, 2] - boxes_b[j, 5] / 2)
+      zb1 = float(boxes_b[j, 2] + boxes_b[j, 5] / 2)
+      zo = max(0.0, min(za1, zb1) - max(za0, zb0))
+      if zo <= 0:
+        continue
+      inter2d = _PolygonArea(_ClipPolygon(ca[i], cb[j]))
+      inter = inter2d * zo
+      vb = float(boxes_b[j, 3] * boxes_b[j, 4] * boxes_b[j, 5])
+      union = va + vb - inter
+      if union > 0:
+        out[i, j] = inter / union
+  return out
+
+
+def NonMaxSuppression3D(boxes: torch.Tensor, scores: torch.Tensor,
+                        iou_threshold: float = 0.5,
+                        max_boxes: int = 100) -> torch.Tensor:
+  """Returns kept indices, best-score first (reference nms_3d_op.cc)."""
+  order = scores.argsort(descending=True)
+  keep = []
+  iou = PairwiseIou3D(boxes, boxes)
+  suppressed = torch.zeros(boxes.shape[0], dtype=torch.bool)
+  for idx in order.tolist():
+    if suppressed[idx]:
+      continue
+    keep.append(idx)
+    if len(keep) >= max_boxes:
+      break
+    suppressed |= iou[idx] > iou_threshold
+  return torch.tensor(keep, dtype=torch.long)
+
+
+def AveragePrecision3D(gt_boxes: torch.Tensor, pred_boxes: torch.Tensor,
+                       pred_scores: torch.Tensor,
+                       iou_threshold: float = 0.7) -> float:
+  """11-point interpolated AP (reference average_precision_3d_op.cc)."""
+  if pred_boxes.shape[0] == 0:
+    return 0.0
+  order = pred_scores.argsort(descending=True)
+  iou = PairwiseIou3D(pred_boxes, gt_boxes)
+  matched = torch.zeros(gt_boxes.shape[0], dtype=torch.bool)
+  tps = []
+  for idx in order.tolist():
+    best_j, best = -1, iou_threshold
+    for j in range(gt_boxes.shape[0]):
+      if not matched[j] and iou[idx, j] >= best:
+        best, best_j = iou[idx, j], j
+    if best_j >= 0:
+      matched[best_j] = True
+      tps.append(1.0)
+    else:
+      tps.append(0.0)
+  tps_t = torch.tensor(tps)
+  cum_tp = tps_t.cumsum(0)
+  precision = cum_tp / torch.arange(1, len(tps) + 1)
+  recall = cum_tp / max(1, gt_boxes.shape[0])
+  ap = 0.0
+  for r in [i / 10 for i in range(11)]:
+    mask = recall >= r
+    ap += (precision[mask].max().item() if mask.any() else 0.0) / 11
+  return ap
+
+
+def SamplePoints(points: torch.Tensor, num_samples: int,
+                 seed: int = 0) -> torch.Tensor:
+  """Furthest-point sampling -> indices [num_samples]
+  (reference sampling_ops.cc)."""
+  n = points.shape[0]
+  g = torch.Generator().manual_seed(seed)
+  first = int(torch.randint(0, n, (1,), generator=g))
+  chosen = [first]
+  dists = (points - points[first]).pow(2).sum(-1)
+  for _ in range(min(num_samples, n) - 1):
+    nxt = int(dists.argmax())
+    chosen.append(nxt)
+    dists = torch.minimum(dists,
+                          (points - points[nxt]).pow(2).sum(-1))
+  return torch.tensor(chosen, dtype=torch.long)

@@ -1,0 +1,69 @@
+"""Quantization utils + car 3D ops tests."""
+
+import math
+
+import pytest
+import torch
+
+from lingvo_amd.core import quant_utils
+from lingvo_amd.models import car_ops
+
+
+def test_fake_quant_layer():
+  p = quant_utils.QuantizedProjectionLayer.Params().Set(
+      name='q', input_dim=8, output_dim=8, random_seed=1,
+      qdomain_default=quant_utils.QDomain.Params().Set(bits=8))
+  layer = p.Instantiate()
+  x = torch.randn(4, 8)
+  out = layer.FProp(layer.theta, x)
+  assert out.shape == (4, 8)
+  # straight-through gradient flows
+  x2 = torch.randn(4, 8, requires_grad=True)
+  layer.FProp(layer.theta, x2).sum().backward()
+  assert layer.w.grad is not None
+  # start_step gating: before start, identity
+  p2 = p.Copy().Set(name='q2')
+  p2.qdomain_default.start_step = 100
+  layer2 = p2.Instantiate()
+  w = layer2.theta.w
+  assert torch.equal(layer2.QWeight(w), w)
+
+
+def test_pairwise_iou_3d():
+  a = torch.tensor([[0., 0., 0., 2., 2., 2., 0.]])
+  b = torch.tensor([[0., 0., 0., 2., 2., 2., 0.],      # identical
+                    [1., 0., 0., 2., 2., 2., 0.],      # half x overlap
+                    [10., 0., 0., 2., 2., 2., 0.],     # disjoint
+                    [0., 0., 0., 2., 2., 2., math.pi / 2]])  # rotated 90
+  iou = car_ops.PairwiseIou3D(a, b)
+  assert abs(iou[0, 0] - 1.0) < 1e-4
+  assert abs(iou[0, 1] - (4.0 / 12.0)) < 1e-3  # inter 1x2x2, union 12
+  assert iou[0, 2] == 0.0
+  assert abs(iou[0, 3] - 1.0) < 1e-3  # square rotated 90 == itself
+
+
+def test_nms_3d():
+  boxes = torch.tensor([[0., 0., 0., 2., 2., 2., 0.],
+                        [0.1, 0., 0., 2., 2., 2., 0.],
+                        [5., 0., 0., 2., 2., 2., 0.]])
+  scores = torch.tensor([0.9, 0.8, 0.7])
+  keep = car_ops.NonMaxSuppression3D(boxes, scores, iou_threshold=0.5)
+  assert keep.tolist() == [0, 2]
+
+
+def test_average_precision_3d():
+  gt = torch.tensor([[0., 0., 0., 2., 2., 2., 0.]])
+  pred = torch.tensor([[0., 0., 0., 2., 2., 2., 0.],
+                       [5., 5., 0., 2., 2., 2., 0.]])
+  scores = torch.tensor([0.9, 0.8])
+  ap = car_ops.AveragePrecision3D(gt, pred, scores)
+  assert ap > 0.9  # perfect first detection
+
+
+def test_furthest_point_sampling():
+  pts = torch.tensor([[0., 0., 0.], [0.1, 0., 0.], [10., 0., 0.],
+                      [0., 10., 0.]])
+  idx = car_ops.SamplePoints(pts, 3, seed=1)
+  assert len(set(idx.tolist())) == 3
+  # the far points should be picked
+  assert 2 in idx.tolist() and 3 in idx.tolist()
