@@ -108,6 +108,28 @@ class RecordBatcher:
     self._stop = True
 
 
+class WeightedMixYielder:
+  """Samples records from multiple yielders with given weights
+  (reference weighted_mix_record_yielder.cc)."""
+
+  def __init__(self, yielders, weights, seed: int = 301):
+    import random
+    assert len(yielders) == len(weights)
+    self._yielders = yielders
+    self._weights = [float(w) for w in weights]
+    self._rng = random.Random(seed)
+
+  def yield_record(self):
+    i = self._rng.choices(range(len(self._yielders)),
+                          weights=self._weights, k=1)[0]
+    value, _ = self._yielders[i].yield_record()
+    return value, i  # source_id = yielder index
+
+  def stop(self):
+    for y in self._yielders:
+      y.stop()
+
+
 def GenericInput(processor: Callable, file_pattern: str,
                  bucket_upper_bound: Sequence[int],
                  bucket_batch_limit: Sequence[int],

@@ -109,3 +109,42 @@ def test_distillation_task():
   # teacher frozen
   assert all(not prm.requires_grad
              for prm in task.teacher.parameters())
+
+
+def test_helpers_splits_and_scaling():
+  from lingvo_amd.utils import helpers
+  assert helpers.ComputeSplits(10, 3) == [4, 3, 3]
+  batch = NestedMap(a=torch.arange(10), b=torch.ones(10, 2))
+  parts = helpers.SplitNestedMap(batch, 3)
+  assert [p.a.shape[0] for p in parts] == [4, 3, 3]
+  assert helpers.ScaleInfeedToGlobal(16, 8) == 128
+  assert helpers.ScaleGlobalToInfeed(128, 8) == 16
+  gc = helpers.GradientCombiner([0.5, 2.0])
+  out = gc.Combine([torch.tensor(1.0), torch.tensor(2.0)])
+  assert abs(float(out) - 4.5) < 1e-6
+
+
+def test_mlperf_print(capsys):
+  from lingvo_amd.utils import helpers
+  helpers.mlperf_print('run_start', 1)
+  out = capsys.readouterr().out
+  assert out.startswith(':::MLLOG')
+
+
+def test_weighted_mix_yielder(tmp_path):
+  from lingvo_amd.core.generic_input import WeightedMixYielder
+  from lingvo_amd.ops import _loader
+  ext = _loader.get_ext(required=True)
+  ys = []
+  for i in range(2):
+    p = tmp_path / f'f{i}.txt'
+    with open(p, 'w') as f:
+      f.write(f'src{i}\n' * 50)
+    ys.append(ext.RecordYielder([str(p)], 'text', 1, 10, 1, True))
+  mix = WeightedMixYielder(ys, [0.9, 0.1], seed=3)
+  counts = [0, 0]
+  for _ in range(200):
+    _, src = mix.yield_record()
+    counts[src] += 1
+  assert counts[0] > counts[1] * 3
+  mix.stop()
