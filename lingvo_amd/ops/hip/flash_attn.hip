@@ -42,12 +42,12 @@ constexpr float NEG_INF = -1e30f;
 
 // Stage ROWS x H bf16 from global (row i at src + i*src_stride elems) into
 // LDS row-major with swizzle. Rows >= valid_rows are zeroed.
-template <int H, int ROWS>
+template <int H, int ROWS, int NT = BLOCK>
 __device__ void stage_regular(const unsigned short* src, long src_stride,
                               int valid_rows, char* dst) {
   constexpr int ROWB = H * 2;
   constexpr int TPR = ROWB / 16;             // threads per row (16B each)
-  constexpr int RPP = BLOCK / TPR;           // rows per pass
+  constexpr int RPP = NT / TPR;              // rows per pass
   const int tid = threadIdx.x;
   const int r0 = tid / TPR;
   const int byte0 = (tid % TPR) * 16;
@@ -67,12 +67,12 @@ __device__ void stage_regular(const unsigned short* src, long src_stride,
 
 // Stage ROWS x H bf16 from global into LDS TRANSPOSED as [H][ROWS]
 // (ROWS=64, 128B rows), swizzled per h-row. Scalar 2B writes.
-template <int H, int ROWS>
+template <int H, int ROWS, int NT = BLOCK>
 __device__ void stage_transposed(const unsigned short* src, long src_stride,
                                  int valid_rows, char* dst) {
   constexpr int ROWB = H * 2;
   constexpr int TPR = ROWB / 16;
-  constexpr int RPP = BLOCK / TPR;
+  constexpr int RPP = NT / TPR;
   const int tid = threadIdx.x;
   const int r0 = tid / TPR;
   const int h0 = (tid % TPR) * 8;
@@ -118,8 +118,12 @@ __device__ __forceinline__ bool visible(int q, int k, int klen, int win_l,
 // ---------------------------------------------------------------------------
 // Forward
 // ---------------------------------------------------------------------------
+constexpr int FWD_NW = 8;                 // waves per fwd block
+constexpr int FWD_BLOCK = FWD_NW * WAVE_SIZE;
+constexpr int FQT = FWD_NW * 16;          // q rows per fwd block
+
 template <int H>
-__global__ __launch_bounds__(BLOCK) void fa_fwd_kernel(
+__global__ __launch_bounds__(FWD_BLOCK) void fa_fwd_kernel(
     const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
     const unsigned short* __restrict__ v,
     const int* __restrict__ klen_ptr,          // [B] or null
@@ -132,7 +136,7 @@ __global__ __launch_bounds__(BLOCK) void fa_fwd_kernel(
   extern __shared__ char smem[];
   char* k_lds = smem;                        // [KT][H] swz
   char* vt_lds = k_lds + KT * ROWB;          // [H][KT] swz
-  char* p_lds = vt_lds + H * KT * 2;         // [NW][16][KT] swz
+  char* p_lds = vt_lds + H * KT * 2;         // [FWD_NW][16][KT] swz
 
   const int qt = blockIdx.x;
   const int n = blockIdx.y;
@@ -144,7 +148,7 @@ __global__ __launch_bounds__(BLOCK) void fa_fwd_kernel(
   const int cl = lane & 15;      // col-in-frag
   const int klen = klen_ptr ? klen_ptr[b] : S;
 
-  const int q0 = qt * QT + wid * 16;  // this wave's first q row
+  const int q0 = qt * FQT + wid * 16;  // this wave's first q row
 
   // Q fragments (A layout): row = cl, k = g*8 + kk*32. Folded zeros for
   // rows >= T.
@@ -172,22 +176,26 @@ __global__ __launch_bounds__(BLOCK) void fa_fwd_kernel(
   }
 
   // KV tile range from the window.
-  int kmax_excl = min(klen, win_r < 0 ? S : min(S, qt * QT + QT - 1 + win_r + 1));
-  int kmin = win_l < 0 ? 0 : max(0, qt * QT - win_l);
+  int kmax_excl =
+      min(klen, win_r < 0 ? S : min(S, qt * FQT + FQT - 1 + win_r + 1));
+  int kmin = win_l < 0 ? 0 : max(0, qt * FQT - win_l);
   const int kt_lo = kmin / KT;
   const int kt_hi = (max(kmax_excl, 1) - 1) / KT;
 
   for (int kt = kt_lo; kt <= kt_hi; ++kt) {
     const int kbase = kt * KT;
     // Stage K and V^T.
-    stage_regular<H, KT>(k + (((long)b * S + kbase) * NKV + nkv) * H,
-                         (long)NKV * H, klen - kbase, k_lds);
-    stage_transposed<H, KT>(v + (((long)b * S + kbase) * NKV + nkv) * H,
-                            (long)NKV * H, klen - kbase, vt_lds);
+    stage_regular<H, KT, FWD_BLOCK>(
+        k + (((long)b * S + kbase) * NKV + nkv) * H, (long)NKV * H,
+        klen - kbase, k_lds);
+    stage_transposed<H, KT, FWD_BLOCK>(
+        v + (((long)b * S + kbase) * NKV + nkv) * H, (long)NKV * H,
+        klen - kbase, vt_lds);
     __syncthreads();
 
     // S strip: 16 q rows x KT keys, fp32.
     float s[4][4];  // [nf over keys][r]
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int nf = 0; nf < 4; ++nf) {
       f32x4 acc = {0.f, 0.f, 0.f, 0.f};
@@ -199,6 +207,7 @@ __global__ __launch_bounds__(BLOCK) void fa_fwd_kernel(
 #pragma unroll
       for (int r = 0; r < 4; ++r) s[nf][r] = acc[r];
     }
+    __builtin_amdgcn_s_setprio(0);
 
     // Scale + mask + bias.
 #pragma unroll
@@ -267,6 +276,7 @@ __global__ __launch_bounds__(BLOCK) void fa_fwd_kernel(
     // Same-wave DS ordering makes the reads below safe without a barrier.
 
     // O += P @ V : A = P (rows=q, k=keys), B from Vt (k=keys, col=h).
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int kk2 = 0; kk2 < KT / 32; ++kk2) {
       bf16x8 pa = lds_frag(pw, cl, KT * 2, (kk2 * 32 + g * 8) * 2);
@@ -277,6 +287,7 @@ __global__ __launch_bounds__(BLOCK) void fa_fwd_kernel(
         acc_o[hf] = mfma16x16x32_bf16(pa, vb, acc_o[hf]);
       }
     }
+    __builtin_amdgcn_s_setprio(0);
     __syncthreads();
   }
 
@@ -643,13 +654,15 @@ std::vector<torch::Tensor> fa_fwd(torch::Tensor q, torch::Tensor k,
   auto o = torch::empty_like(q);
   auto lse = torch::empty({B, N, T}, q.options().dtype(torch::kFloat32));
   auto stream = at::cuda::getCurrentCUDAStream();
-  dim3 grid((T + QT - 1) / QT, N, B);
+  dim3 grid((T + FQT - 1) / FQT, N, B);
   const int* klp = klen.has_value() ? klen->data_ptr<int>() : nullptr;
   const unsigned short* bp =
       bias.has_value() ? (const unsigned short*)bias->data_ptr() : nullptr;
-  size_t shmem = (size_t)KT * H * 2 + (size_t)H * KT * 2 + NW * 16 * KT * 2;
+  size_t shmem =
+      (size_t)KT * H * 2 + (size_t)H * KT * 2 + FWD_NW * 16 * KT * 2;
 #define FA_FWD(HH)                                                          \
-  hipLaunchKernelGGL((fa_fwd_kernel<HH>), grid, dim3(BLOCK), shmem, stream, \
+  hipLaunchKernelGGL((fa_fwd_kernel<HH>), grid, dim3(FWD_BLOCK), shmem,     \
+                     stream,                                                \
                      (const unsigned short*)q.data_ptr(),                   \
                      (const unsigned short*)k.data_ptr(),                   \
                      (const unsigned short*)v.data_ptr(), klp, bp,          \
