@@ -340,8 +340,8 @@ __global__ void fa_bwd_delta(const unsigned short* __restrict__ dout,
 // ---------------------------------------------------------------------------
 // Backward main
 // ---------------------------------------------------------------------------
-template <int H, bool BIAS_GRAD>
-__global__ __launch_bounds__(BLOCK) void fa_bwd_kernel(
+template <int H, bool BIAS_GRAD, int KTB, int NWB>
+__global__ __launch_bounds__(NWB * WAVE_SIZE) void fa_bwd_kernel(
     const unsigned short* __restrict__ dout,
     const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
     const unsigned short* __restrict__ v, const float* __restrict__ lse,
@@ -359,10 +359,10 @@ __global__ __launch_bounds__(BLOCK) void fa_bwd_kernel(
   char* qt_lds = q_lds + QT * ROWB;          // [H][QT] swz
   char* do_lds = qt_lds + H * QT * 2;        // [QT][H] swz
   char* dot_lds = do_lds + QT * ROWB;        // [H][QT] swz
-  char* kt_lds = dot_lds + H * QT * 2;       // [H][KT] swz
-  char* ds_lds = kt_lds + H * KT * 2;        // [QT][KT] swz
-  char* a_lds = ds_lds + QT * KT * 2;        // [NW][16][QT] swz
-  float* lse_s = (float*)(a_lds + NW * 16 * QT * 2);   // [QT]
+  char* kt_lds = dot_lds + H * QT * 2;       // [H][KTB] swz
+  char* ds_lds = kt_lds + H * KTB * 2;       // [QT][KTB] swz
+  char* a_lds = ds_lds + QT * KTB * 2;       // [NWB][16][QT] swz
+  float* lse_s = (float*)(a_lds + NWB * 16 * QT * 2);  // [QT]
   float* delta_s = lse_s + QT;                         // [QT]
   float* dbias_s = delta_s + QT;                       // [nbias] if BIAS_GRAD
 
@@ -375,12 +375,13 @@ __global__ __launch_bounds__(BLOCK) void fa_bwd_kernel(
   const int g = lane >> 4;
   const int cl = lane & 15;
   const int klen = klen_ptr ? klen_ptr[b] : S;
-  const int kbase = kt * KT;
+  const int kbase = kt * KTB;
   const int k0 = kbase + wid * 16;  // wave's first key
 
   // Stage K^T once (shared by all heads in the group).
-  stage_transposed<H, KT>(k + (((long)b * S + kbase) * NKV + nkv) * H,
-                          (long)NKV * H, klen - kbase, kt_lds);
+  stage_transposed<H, KTB, NWB * WAVE_SIZE>(
+      k + (((long)b * S + kbase) * NKV + nkv) * H, (long)NKV * H,
+      klen - kbase, kt_lds);
 
   // K, V fragments (A layout) from global: row = key (cl), k = h.
   bf16x8 kfrag[KH], vfrag[KH];
@@ -403,7 +404,7 @@ __global__ __launch_bounds__(BLOCK) void fa_bwd_kernel(
 
   // q-tile range for this key tile.
   int qlo = win_r < 0 ? 0 : max(0, kbase - win_r);
-  int qhi = win_l < 0 ? T - 1 : min(T - 1, kbase + KT - 1 + win_l);
+  int qhi = win_l < 0 ? T - 1 : min(T - 1, kbase + KTB - 1 + win_l);
   const int qt_lo = qlo / QT, qt_hi = max(qhi, 0) / QT;
 
   for (int head = 0; head < group; ++head) {
@@ -416,21 +417,26 @@ __global__ __launch_bounds__(BLOCK) void fa_bwd_kernel(
       acc_dv[hf] = {0.f, 0.f, 0.f, 0.f};
     }
     if (BIAS_GRAD) {
-      for (int i = threadIdx.x; i < nbias; i += BLOCK) dbias_s[i] = 0.f;
+      for (int i = threadIdx.x; i < nbias; i += NWB * WAVE_SIZE)
+        dbias_s[i] = 0.f;
     }
 
     for (int qt2 = qt_lo; qt2 <= qt_hi; ++qt2) {
       const int qb = qt2 * QT;
       __syncthreads();
-      stage_regular<H, QT>(q + (((long)b * T + qb) * N + n) * H, (long)N * H,
-                           T - qb, q_lds);
-      stage_transposed<H, QT>(q + (((long)b * T + qb) * N + n) * H,
-                              (long)N * H, T - qb, qt_lds);
-      stage_regular<H, QT>(dout + (((long)b * T + qb) * N + n) * H,
-                           (long)N * H, T - qb, do_lds);
-      stage_transposed<H, QT>(dout + (((long)b * T + qb) * N + n) * H,
-                              (long)N * H, T - qb, dot_lds);
-      for (int i = threadIdx.x; i < QT; i += BLOCK) {
+      stage_regular<H, QT, NWB * WAVE_SIZE>(
+          q + (((long)b * T + qb) * N + n) * H, (long)N * H, T - qb,
+          q_lds);
+      stage_transposed<H, QT, NWB * WAVE_SIZE>(
+          q + (((long)b * T + qb) * N + n) * H, (long)N * H, T - qb,
+          qt_lds);
+      stage_regular<H, QT, NWB * WAVE_SIZE>(
+          dout + (((long)b * T + qb) * N + n) * H, (long)N * H, T - qb,
+          do_lds);
+      stage_transposed<H, QT, NWB * WAVE_SIZE>(
+          dout + (((long)b * T + qb) * N + n) * H, (long)N * H, T - qb,
+          dot_lds);
+      for (int i = threadIdx.x; i < QT; i += NWB * WAVE_SIZE) {
         int qrow = qb + i;
         lse_s[i] = qrow < T ? lse[((long)b * N + n) * T + qrow] : NEG_INF;
         delta_s[i] = qrow < T ? delta[((long)b * N + n) * T + qrow] : 0.f;
@@ -548,38 +554,42 @@ __global__ __launch_bounds__(BLOCK) void fa_bwd_kernel(
           int qrow = nf * 16 + cl;
           int key = wid * 16 + g * 4 + r;
           *reinterpret_cast<unsigned short*>(
-              ds_lds + qrow * (KT * 2) + swz(qrow, key * 2)) =
+              ds_lds + qrow * (KTB * 2) + swz(qrow, key * 2)) =
               float_to_bf16_bits(dlg[nf][r] * scale);
         }
       }
       __syncthreads();
 
-      // dQ strip for this wave: rows q = qb + wid*16 + cl (A from ds_lds),
-      // B = K from kt_lds; atomic-accumulate fp32.
-      f32x4 acc_dq[HF];
+      // dQ strips: rows q = qb + dw*16 + .. (A from ds_lds), B = K from
+      // kt_lds; atomic-accumulate fp32. Only QT/16 strips exist, so with
+      // NWB > QT/16 the extra waves skip this phase.
+      if (wid < QT / 16) {
+        f32x4 acc_dq[HF];
 #pragma unroll
-      for (int hf = 0; hf < HF; ++hf) acc_dq[hf] = {0.f, 0.f, 0.f, 0.f};
+        for (int hf = 0; hf < HF; ++hf) acc_dq[hf] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-      for (int kk2 = 0; kk2 < KT / 32; ++kk2) {
-        bf16x8 da =
-            lds_frag(ds_lds, wid * 16 + cl, KT * 2, (kk2 * 32 + g * 8) * 2);
+        for (int kk2 = 0; kk2 < KTB / 32; ++kk2) {
+          bf16x8 da = lds_frag(ds_lds, wid * 16 + cl, KTB * 2,
+                               (kk2 * 32 + g * 8) * 2);
 #pragma unroll
-        for (int hf = 0; hf < HF; ++hf) {
-          bf16x8 bk =
-              lds_frag(kt_lds, hf * 16 + cl, KT * 2, (kk2 * 32 + g * 8) * 2);
-          acc_dq[hf] = mfma16x16x32_bf16(da, bk, acc_dq[hf]);
+          for (int hf = 0; hf < HF; ++hf) {
+            bf16x8 bk = lds_frag(kt_lds, hf * 16 + cl, KTB * 2,
+                                 (kk2 * 32 + g * 8) * 2);
+            acc_dq[hf] = mfma16x16x32_bf16(da, bk, acc_dq[hf]);
+          }
         }
-      }
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int qrow = qb + wid * 16 + g * 4 + r;
-        if (qrow >= T) continue;
+        for (int r = 0; r < 4; ++r) {
+          const int qrow = qb + wid * 16 + g * 4 + r;
+          if (qrow >= T) continue;
 #pragma unroll
-        for (int hf = 0; hf < HF; ++hf) {
-          if (acc_dq[hf][r] != 0.f) {
-            atomicAdd(
-                dq_acc + (((long)b * T + qrow) * N + n) * H + hf * 16 + cl,
-                acc_dq[hf][r]);
+          for (int hf = 0; hf < HF; ++hf) {
+            if (acc_dq[hf][r] != 0.f) {
+              atomicAdd(
+                  dq_acc + (((long)b * T + qrow) * N + n) * H + hf * 16 +
+                      cl,
+                  acc_dq[hf][r]);
+            }
           }
         }
       }
@@ -588,7 +598,7 @@ __global__ __launch_bounds__(BLOCK) void fa_bwd_kernel(
     // Flush dbias for this head.
     if (BIAS_GRAD) {
       __syncthreads();
-      for (int i = threadIdx.x; i < nbias; i += BLOCK) {
+      for (int i = threadIdx.x; i < nbias; i += NWB * WAVE_SIZE) {
         if (dbias_s[i] != 0.f)
           atomicAdd(dbias + (long)n * nbias + i, dbias_s[i]);
       }
@@ -727,16 +737,22 @@ std::vector<torch::Tensor> fa_bwd(torch::Tensor dout, torch::Tensor q,
       bias.has_value() ? (const unsigned short*)bias->data_ptr() : nullptr;
   bool bg = bias.has_value() && bias_grad;
 
-  size_t shmem = (size_t)QT * H * 2 * 2   // q_lds + do_lds
-                 + (size_t)H * QT * 2 * 2  // qt_lds + dot_lds
-                 + (size_t)H * KT * 2      // kt_lds
-                 + (size_t)QT * KT * 2     // ds_lds
-                 + (size_t)NW * 16 * QT * 2  // a_lds
+  // H=64 uses 128-key tiles with 8 waves (q-tile staging amortized 2x);
+  // H=128 keeps 64-key tiles (LDS budget).
+  const int ktb = H == 64 ? 128 : 64;
+  const int nwb = H == 64 ? 8 : 4;
+  size_t shmem = (size_t)QT * H * 2 * 2     // q_lds + do_lds
+                 + (size_t)H * QT * 2 * 2   // qt_lds + dot_lds
+                 + (size_t)H * ktb * 2      // kt_lds
+                 + (size_t)QT * ktb * 2     // ds_lds
+                 + (size_t)nwb * 16 * QT * 2  // a_lds
                  + 2 * QT * sizeof(float) + (bg ? nbias * sizeof(float) : 0);
-  dim3 grid((S + KT - 1) / KT, NKV, B);
+  dim3 grid((S + ktb - 1) / ktb, NKV, B);
 #define FA_BWD(HH, BG)                                                       \
   hipLaunchKernelGGL(                                                        \
-      (fa_bwd_kernel<HH, BG>), grid, dim3(BLOCK), shmem, stream,             \
+      (fa_bwd_kernel<HH, BG, (HH == 64 ? 128 : 64),                          \
+                     (HH == 64 ? 8 : 4)>),                                   \
+      grid, dim3((HH == 64 ? 8 : 4) * WAVE_SIZE), shmem, stream,             \
       (const unsigned short*)dout.data_ptr(),                                \
       (const unsigned short*)q.data_ptr(),                                   \
       (const unsigned short*)k.data_ptr(),                                   \
