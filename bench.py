@@ -26,7 +26,7 @@ def main():
   ap.add_argument('--gpus', type=int, default=1)
   ap.add_argument('--steps', type=int, default=20)
   ap.add_argument('--warmup', type=int, default=5)
-  ap.add_argument('--batch', type=int, default=64,
+  ap.add_argument('--batch', type=int, default=128,
                   help='Per-GPU batch size.')
   ap.add_argument('--no-graph', action='store_true',
                   help='Disable hipGraph step capture (eager steps).')

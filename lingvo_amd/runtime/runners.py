@@ -105,6 +105,13 @@ class Trainer(BaseRunner):
                = None, grad_sync=None, **kwargs):
     super().__init__(model_params, logdir, 'trainer', **kwargs)
     self._max_steps = max_steps
+    if grad_sync is None:
+      # Auto-enable DP grad sync when launched under torchrun.
+      import torch.distributed as dist
+      if dist.is_available() and dist.is_initialized() and \
+          dist.get_world_size() > 1:
+        from lingvo_amd.parallel.ddp import GradSync
+        grad_sync = GradSync(self.model.GetTask())
     self._grad_sync = grad_sync
     self._tracker = StepRateTracker()
     self._metrics_log = os.path.join(logdir, 'train', 'metrics.jsonl')

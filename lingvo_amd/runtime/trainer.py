@@ -116,6 +116,15 @@ def main(argv: Optional[List[str]] = None) -> None:
     print('error: --model is required', file=sys.stderr)
     sys.exit(2)
   os.makedirs(args.logdir, exist_ok=True)
+  # Under torchrun (WORLD_SIZE>1), initialize DP: one process per GPU,
+  # RCCL over xGMI (gloo on CPU).
+  from lingvo_amd.parallel import ddp
+  rank = ddp.InitDistributed()
+  if rank != 0 and args.job == 'trainer_client':
+    args.job = 'trainer'  # controller artifacts written by rank 0 only
+  if args.device is None and torch.cuda.is_available():
+    local_rank = int(os.environ.get('LOCAL_RANK', '0'))
+    args.device = f'cuda:{local_rank}'
   RunnerManager(args).Start()
 
 
