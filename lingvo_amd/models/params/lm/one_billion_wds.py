@@ -84,3 +84,25 @@ class WordLevelOneBwdsRnnLm(SingleTaskModelParams):
         optimizer=optimizer_lib.Adagrad.Params(),
         clip_gradient_norm_to_value=1.0)
     return p
+
+
+@registry.RegisterSingleTaskModel
+class OneBWdsGPipeTransformerWPM(OneBWdsTransformerLm):
+  """GPipe pipelined variant (reference one_billion_wds.py:181: 32
+  layers, 4 splits, 32 microbatches on 4x V100-16GB). On MI355X the
+  same model fits on one GPU (see OneBWdsTransformerLm); this config
+  carries the pipeline topology for multi-GPU PP runs via
+  lingvo_amd.parallel.gpipe_lm (one process per stage over RCCL P2P).
+  """
+
+  NUM_STAGES = 4
+  NUM_MICRO_BATCHES = 32
+
+  def Task(self):
+    p = super().Task().Set(name='1bwds_gpipe_transformer_lm')
+    # Pipeline topology consumed by the gpipe_lm launcher.
+    p.Define('pipeline_num_stages', self.NUM_STAGES, 'Pipeline stages.')
+    p.Define('pipeline_num_micro_batches', self.NUM_MICRO_BATCHES,
+             'Microbatches per step (>= 4x stages per reference '
+             'guidance, one_billion_wds.py:197).')
+    return p

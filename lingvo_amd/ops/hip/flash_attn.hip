@@ -737,10 +737,10 @@ std::vector<torch::Tensor> fa_bwd(torch::Tensor dout, torch::Tensor q,
       bias.has_value() ? (const unsigned short*)bias->data_ptr() : nullptr;
   bool bg = bias.has_value() && bias_grad;
 
-  // H=64 uses 128-key tiles with 8 waves (q-tile staging amortized 2x);
-  // H=128 keeps 64-key tiles (LDS budget).
-  const int ktb = H == 64 ? 128 : 64;
-  const int nwb = H == 64 ? 8 : 4;
+  // 128-key tiles with 8 waves (q-tile staging amortized 2x). H=128
+  // uses 144KB LDS -> 1 block/CU; staging savings outweigh occupancy.
+  const int ktb = 128;
+  const int nwb = 8;
   size_t shmem = (size_t)QT * H * 2 * 2     // q_lds + do_lds
                  + (size_t)H * QT * 2 * 2   // qt_lds + dot_lds
                  + (size_t)H * ktb * 2      // kt_lds
@@ -750,9 +750,8 @@ std::vector<torch::Tensor> fa_bwd(torch::Tensor dout, torch::Tensor q,
   dim3 grid((S + ktb - 1) / ktb, NKV, B);
 #define FA_BWD(HH, BG)                                                       \
   hipLaunchKernelGGL(                                                        \
-      (fa_bwd_kernel<HH, BG, (HH == 64 ? 128 : 64),                          \
-                     (HH == 64 ? 8 : 4)>),                                   \
-      grid, dim3((HH == 64 ? 8 : 4) * WAVE_SIZE), shmem, stream,             \
+      (fa_bwd_kernel<HH, BG, 128, 8>), grid, dim3(8 * WAVE_SIZE), shmem,     \
+      stream,                                                                \
       (const unsigned short*)dout.data_ptr(),                                \
       (const unsigned short*)q.data_ptr(),                                   \
       (const unsigned short*)k.data_ptr(),                                   \

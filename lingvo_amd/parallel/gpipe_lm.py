@@ -1,0 +1,122 @@
+"""Pipelined Transformer LM stages for GPipe
+(reference: lingvo/core/layers_with_gpipe.py:576 GPipeTransformerStack;
+config lm.one_billion_wds.OneBWdsGPipeTransformerWPM,
+one_billion_wds.py:181-198 — 32 layers over 4 splits, 32 microbatches).
+
+Each rank builds ONE stage: stage 0 owns the embedding + the first layer
+slice, the last stage owns the final LN + softmax; layers are partitioned
+contiguously (PartitionSequentialLayers). Driven by GPipeRunner's
+fill-drain schedule over torch.distributed P2P.
+"""
+
+from __future__ import annotations
+
+from typing import List, Optional
+
+import torch
+
+from lingvo_amd.core import py_utils
+from lingvo_amd.core.base_layer import BaseLayer
+from lingvo_amd.core.nested_map import NestedMap
+from lingvo_amd.layers import layers as lingvo_layers
+from lingvo_amd.layers import transformer as transformer_lib
+from lingvo_amd.parallel.pipeline import (GPipeRunner,
+                                          PartitionSequentialLayers)
+
+
+class TransformerLmStage(BaseLayer):
+  """One pipeline stage of a causal transformer LM."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('vocab_size', 32000, 'Vocab.')
+    p.Define('model_dim', 1024, 'Model dim.')
+    p.Define('num_layers_total', 12, 'Total layers in the LM.')
+    p.Define('num_heads', 16, 'Heads.')
+    p.Define('hidden_dim', 0, 'FFN hidden.')
+    p.Define('dropout_prob', 0.0, 'Dropout.')
+    p.Define('stage_idx', 0, 'This stage.')
+    p.Define('num_stages', 1, 'Total stages.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    parts = PartitionSequentialLayers(list(range(p.num_layers_total)),
+                                      p.num_stages)
+    self._my_layers = parts[p.stage_idx]
+    self.is_first = p.stage_idx == 0
+    self.is_last = p.stage_idx == p.num_stages - 1
+    if self.is_first:
+      self.CreateChild('emb', lingvo_layers.EmbeddingLayer.Params().Set(
+          vocab_size=p.vocab_size, embedding_dim=p.model_dim,
+          scale_sqrt_depth=True))
+      self.CreateChild('pos_emb',
+                       lingvo_layers.PositionalEmbeddingLayer.Params().Set(
+                           embedding_dim=p.model_dim))
+    layer_ps = []
+    for li in self._my_layers:
+      lp = transformer_lib.TransformerLayer.Params().Set(
+          name=f'layer_{li}', input_dim=p.model_dim,
+          num_heads=p.num_heads, mask_self_atten=True)
+      lp.tr_fflayer_tpl.hidden_dim = p.hidden_dim or 4 * p.model_dim
+      lp.tr_atten_tpl.residual_dropout_prob = p.dropout_prob
+      lp.tr_fflayer_tpl.residual_dropout_prob = p.dropout_prob
+      layer_ps.append(lp)
+    self.CreateChildren('layers', layer_ps)
+    if self.is_last:
+      self.CreateChild('final_ln', lingvo_layers.LayerNorm.Params().Set(
+          input_dim=p.model_dim))
+      self.CreateChild('softmax',
+                       lingvo_layers.SimpleFullSoftmax.Params().Set(
+                           input_dim=p.model_dim,
+                           num_classes=p.vocab_size))
+
+  def FProp(self, theta: NestedMap, nmap: NestedMap) -> NestedMap:
+    p = self.p
+    if self.is_first:
+      x = self.emb.EmbLookup(theta.emb, nmap.ids.long()).to(
+          self.fprop_dtype)
+      pos = self.pos_emb.FProp(theta.pos_emb, x.shape[1],
+                               device=x.device)
+      x = x + pos.unsqueeze(0).to(x.dtype)
+    else:
+      x = nmap.act
+    paddings = nmap.paddings
+    for i, layer in enumerate(self.layers):
+      x = layer.FProp(theta.layers[i], x, paddings)
+    if self.is_last:
+      x = self.final_ln.FProp(theta.final_ln, x)
+      return NestedMap(act=x, paddings=paddings)
+    return NestedMap(act=x, paddings=paddings)
+
+  def XentLoss(self, theta, act, labels, weights):
+    return self.softmax.XentLoss(theta.softmax, act,
+                                 class_weights=weights,
+                                 class_ids=labels)
+
+
+def RunGPipeLmStep(stage: TransformerLmStage, runner: GPipeRunner,
+                   batches: List[NestedMap]) -> Optional[torch.Tensor]:
+  """One pipelined train step over len(batches) microbatches.
+
+  Stage 0 feeds microbatches; the last stage computes the xent loss.
+  Returns mean loss on the last stage, None elsewhere. Gradients are
+  left accumulated in stage parameters (caller applies the optimizer).
+  """
+  theta = stage.theta
+
+  def fprop(nmap):
+    return stage.FProp(theta, nmap)
+
+  def input_fn(m):
+    b = batches[m]
+    return NestedMap(ids=b.ids, paddings=b.paddings.float())
+
+  def loss_fn(nmap, m):
+    b = batches[m]
+    xent = stage.XentLoss(theta, nmap.act, b.labels, b.weights)
+    return xent.avg_xent
+
+  return runner.RunStep(fprop, input_fn=input_fn, loss_fn=loss_fn)
