@@ -122,7 +122,7 @@ constexpr int FWD_NW = 8;                 // waves per fwd block
 constexpr int FWD_BLOCK = FWD_NW * WAVE_SIZE;
 constexpr int FQT = FWD_NW * 16;          // q rows per fwd block
 
-template <int H>
+template <int H, int KTF>
 __global__ __launch_bounds__(FWD_BLOCK) void fa_fwd_kernel(
     const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
     const unsigned short* __restrict__ v,
@@ -133,10 +133,11 @@ __global__ __launch_bounds__(FWD_BLOCK) void fa_fwd_kernel(
   constexpr int ROWB = H * 2;
   constexpr int KH = H / 32;   // mfma K-steps over head dim
   constexpr int HF = H / 16;   // output col frags
+  constexpr int NF = KTF / 16; // key frags per strip
   extern __shared__ char smem[];
-  char* k_lds = smem;                        // [KT][H] swz
-  char* vt_lds = k_lds + KT * ROWB;          // [H][KT] swz
-  char* p_lds = vt_lds + H * KT * 2;         // [FWD_NW][16][KT] swz
+  char* k_lds = smem;                        // [KTF][H] swz
+  char* vt_lds = k_lds + KTF * ROWB;         // [H][KTF] swz
+  char* p_lds = vt_lds + H * KTF * 2;        // [FWD_NW][16][KTF] swz
 
   const int qt = blockIdx.x;
   const int n = blockIdx.y;
@@ -179,25 +180,25 @@ __global__ __launch_bounds__(FWD_BLOCK) void fa_fwd_kernel(
   int kmax_excl =
       min(klen, win_r < 0 ? S : min(S, qt * FQT + FQT - 1 + win_r + 1));
   int kmin = win_l < 0 ? 0 : max(0, qt * FQT - win_l);
-  const int kt_lo = kmin / KT;
-  const int kt_hi = (max(kmax_excl, 1) - 1) / KT;
+  const int kt_lo = kmin / KTF;
+  const int kt_hi = (max(kmax_excl, 1) - 1) / KTF;
 
   for (int kt = kt_lo; kt <= kt_hi; ++kt) {
-    const int kbase = kt * KT;
+    const int kbase = kt * KTF;
     // Stage K and V^T.
-    stage_regular<H, KT, FWD_BLOCK>(
+    stage_regular<H, KTF, FWD_BLOCK>(
         k + (((long)b * S + kbase) * NKV + nkv) * H, (long)NKV * H,
         klen - kbase, k_lds);
-    stage_transposed<H, KT, FWD_BLOCK>(
+    stage_transposed<H, KTF, FWD_BLOCK>(
         v + (((long)b * S + kbase) * NKV + nkv) * H, (long)NKV * H,
         klen - kbase, vt_lds);
     __syncthreads();
 
-    // S strip: 16 q rows x KT keys, fp32.
-    float s[4][4];  // [nf over keys][r]
+    // S strip: 16 q rows x KTF keys, fp32.
+    float s[NF][4];  // [nf over keys][r]
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int nf = 0; nf < 4; ++nf) {
+    for (int nf = 0; nf < NF; ++nf) {
       f32x4 acc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int kk = 0; kk < KH; ++kk) {
@@ -211,7 +212,7 @@ __global__ __launch_bounds__(FWD_BLOCK) void fa_fwd_kernel(
 
     // Scale + mask + bias.
 #pragma unroll
-    for (int nf = 0; nf < 4; ++nf) {
+    for (int nf = 0; nf < NF; ++nf) {
       const int kcol = kbase + nf * 16 + cl;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
@@ -233,7 +234,9 @@ __global__ __launch_bounds__(FWD_BLOCK) void fa_fwd_kernel(
     float rowmax[4], rowsum[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      float mx = fmaxf(fmaxf(s[0][r], s[1][r]), fmaxf(s[2][r], s[3][r]));
+      float mx = s[0][r];
+#pragma unroll
+      for (int nf = 1; nf < NF; ++nf) mx = fmaxf(mx, s[nf][r]);
 #pragma unroll
       for (int off = 1; off < 16; off <<= 1) mx = fmaxf(mx, __shfl_xor(mx, off));
       rowmax[r] = mx;
@@ -246,7 +249,7 @@ __global__ __launch_bounds__(FWD_BLOCK) void fa_fwd_kernel(
       m_run[r] = m_new;
       float sum = 0.f;
 #pragma unroll
-      for (int nf = 0; nf < 4; ++nf) {
+      for (int nf = 0; nf < NF; ++nf) {
         float p = (s[nf][r] <= NEG_INF * 0.5f) ? 0.f
                                                : __expf(s[nf][r] - m_new);
         s[nf][r] = p;
@@ -260,16 +263,16 @@ __global__ __launch_bounds__(FWD_BLOCK) void fa_fwd_kernel(
       for (int hf = 0; hf < HF; ++hf) acc_o[hf][r] *= alpha[r];
     }
 
-    // P -> per-wave LDS (bf16, C layout -> row-major [16][KT], swizzled).
-    char* pw = p_lds + wid * 16 * (KT * 2);
+    // P -> per-wave LDS (bf16, C layout -> row-major [16][KTF], swizzled).
+    char* pw = p_lds + wid * 16 * (KTF * 2);
 #pragma unroll
-    for (int nf = 0; nf < 4; ++nf) {
+    for (int nf = 0; nf < NF; ++nf) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         int row = g * 4 + r;
         int col = nf * 16 + cl;
         *reinterpret_cast<unsigned short*>(
-            pw + row * (KT * 2) + swz(row, col * 2)) =
+            pw + row * (KTF * 2) + swz(row, col * 2)) =
             float_to_bf16_bits(s[nf][r]);
       }
     }
@@ -278,12 +281,12 @@ __global__ __launch_bounds__(FWD_BLOCK) void fa_fwd_kernel(
     // O += P @ V : A = P (rows=q, k=keys), B from Vt (k=keys, col=h).
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int kk2 = 0; kk2 < KT / 32; ++kk2) {
-      bf16x8 pa = lds_frag(pw, cl, KT * 2, (kk2 * 32 + g * 8) * 2);
+    for (int kk2 = 0; kk2 < KTF / 32; ++kk2) {
+      bf16x8 pa = lds_frag(pw, cl, KTF * 2, (kk2 * 32 + g * 8) * 2);
 #pragma unroll
       for (int hf = 0; hf < HF; ++hf) {
         bf16x8 vb =
-            lds_frag(vt_lds, hf * 16 + cl, KT * 2, (kk2 * 32 + g * 8) * 2);
+            lds_frag(vt_lds, hf * 16 + cl, KTF * 2, (kk2 * 32 + g * 8) * 2);
         acc_o[hf] = mfma16x16x32_bf16(pa, vb, acc_o[hf]);
       }
     }
@@ -668,11 +671,12 @@ std::vector<torch::Tensor> fa_fwd(torch::Tensor q, torch::Tensor k,
   const int* klp = klen.has_value() ? klen->data_ptr<int>() : nullptr;
   const unsigned short* bp =
       bias.has_value() ? (const unsigned short*)bias->data_ptr() : nullptr;
+  const int ktf = 128;
   size_t shmem =
-      (size_t)KT * H * 2 + (size_t)H * KT * 2 + FWD_NW * 16 * KT * 2;
+      (size_t)ktf * H * 2 + (size_t)H * ktf * 2 + FWD_NW * 16 * ktf * 2;
 #define FA_FWD(HH)                                                          \
-  hipLaunchKernelGGL((fa_fwd_kernel<HH>), grid, dim3(FWD_BLOCK), shmem,     \
-                     stream,                                                \
+  hipLaunchKernelGGL((fa_fwd_kernel<HH, 128>), grid, dim3(FWD_BLOCK),       \
+                     shmem, stream,                                         \
                      (const unsigned short*)q.data_ptr(),                   \
                      (const unsigned short*)k.data_ptr(),                   \
                      (const unsigned short*)v.data_ptr(), klp, bp,          \
