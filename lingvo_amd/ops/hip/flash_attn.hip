@@ -671,13 +671,24 @@ std::vector<torch::Tensor> fa_fwd(torch::Tensor q, torch::Tensor k,
   const int* klp = klen.has_value() ? klen->data_ptr<int>() : nullptr;
   const unsigned short* bp =
       bias.has_value() ? (const unsigned short*)bias->data_ptr() : nullptr;
-  const int ktf = 128;
+  // Big KV tiles amortize staging barriers, but waste work on short
+  // sequences (masked overhang): pick by S.
+  const int ktf = S >= 512 ? 128 : 64;
   size_t shmem =
       (size_t)ktf * H * 2 + (size_t)H * ktf * 2 + FWD_NW * 16 * ktf * 2;
 #define FA_FWD(HH)                                                          \
-  hipLaunchKernelGGL((fa_fwd_kernel<HH, 128>), grid, dim3(FWD_BLOCK),       \
-                     shmem, stream,                                         \
+  if (ktf == 128)                                                           \
+    hipLaunchKernelGGL((fa_fwd_kernel<HH, 128>), grid, dim3(FWD_BLOCK),     \
+                       shmem, stream,                                       \
                      (const unsigned short*)q.data_ptr(),                   \
+                     (const unsigned short*)k.data_ptr(),                   \
+                     (const unsigned short*)v.data_ptr(), klp, bp,          \
+                     (unsigned short*)o.data_ptr(), lse.data_ptr<float>(),  \
+                     B, T, S, N, NKV, (int)win_l, (int)win_r,               \
+                     (int)bias_clip, (float)scale)
+#define FA_FWD64(HH)                                                        \
+  hipLaunchKernelGGL((fa_fwd_kernel<HH, 64>), grid, dim3(FWD_BLOCK),        \
+                     shmem, stream, (const unsigned short*)q.data_ptr(),    \
                      (const unsigned short*)k.data_ptr(),                   \
                      (const unsigned short*)v.data_ptr(), klp, bp,          \
                      (unsigned short*)o.data_ptr(), lse.data_ptr<float>(),  \
@@ -685,10 +696,13 @@ std::vector<torch::Tensor> fa_fwd(torch::Tensor q, torch::Tensor k,
                      (int)bias_clip, (float)scale)
   if (H == 64) {
     FA_FWD(64);
+    else FA_FWD64(64);
   } else {
     FA_FWD(128);
+    else FA_FWD64(128);
   }
 #undef FA_FWD
+#undef FA_FWD64
   return {o, lse};
 }
 
