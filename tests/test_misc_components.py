@@ -148,3 +148,23 @@ def test_weighted_mix_yielder(tmp_path):
     counts[src] += 1
   assert counts[0] > counts[1] * 3
   mix.stop()
+
+
+def test_summary_writer_and_model_analysis(tmp_path):
+  from lingvo_amd.core import summary_utils
+  w = summary_utils.SummaryWriter(str(tmp_path))
+  w.scalar('loss', torch.tensor(1.5), step=3)
+  w.histogram('grads', torch.randn(100), step=3)
+  w.text('note', 'hello', step=3)
+  import json
+  recs = [json.loads(l) for l in open(tmp_path / 'events.jsonl')]
+  assert [r['kind'] for r in recs] == ['scalar', 'histogram', 'text']
+  model = torch.nn.Linear(4, 2)
+  report = summary_utils.ModelAnalysis(model)
+  assert 'total #params: 10' in report
+
+
+def test_datasets_introspection():
+  from lingvo_amd.runtime import datasets
+  ds = datasets.GetDatasets('image.mnist.LeNet5')
+  assert ds == ['Dev', 'Test', 'Train']
