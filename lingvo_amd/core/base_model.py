@@ -138,7 +138,10 @@ class BaseTask(BaseLayer):
     self.MaybeConvertBf16Weights()
     step = self.global_step
     with py_utils.StepSeedScope(self.p.random_seed or 1234, step):
-      metrics, _ = self.FProp(self.theta, input_batch)
+      theta = self.theta
+      if self.p.train.vn_std:
+        theta = py_utils.AddVn(theta, self.p.train.vn_std)
+      metrics, _ = self.FProp(theta, input_batch)
       loss_name = self.learners[0].p.loss_name
       loss = metrics[loss_name][0]
       for i, lrn in enumerate(self.learners):

@@ -351,6 +351,21 @@ def ToScalar(x) -> float:
   return float(x)
 
 
+def AddVn(theta: NestedMap, vn_std: float) -> NestedMap:
+  """Adds variational (Gaussian) noise to floating theta leaves
+  (reference py_utils.py:3738 AddVN); deterministic per step scope."""
+
+  def add_noise(t):
+    if isinstance(t, torch.Tensor) and t.is_floating_point() and \
+        t.requires_grad:
+      u = GraphSafeUniform(t.shape, t.device).clamp(1e-6, 1 - 1e-6)
+      normal = torch.erfinv(2 * u - 1) * math.sqrt(2.0)
+      return t + vn_std * normal.to(t.dtype)
+    return t
+
+  return theta.Transform(add_noise)
+
+
 def GraphSafeUniform(shape, device, op_seed: Optional[int] = None
                      ) -> torch.Tensor:
   """Uniform [0,1) tensor whose values vary per step via the device
