@@ -124,3 +124,31 @@ def test_bf16_master_weight_resume(tmp_path):
   w2 = torch.cat([p.detach().float().reshape(-1)
                   for p in t2.parameters()])
   assert torch.equal(w1, w2)
+
+
+def test_saver_gc_and_async(tmp_path):
+  """keep_latest_n GC + async save (reference saver.py:139-152)."""
+  import torch
+  from lingvo_amd.core.checkpointer import (LatestCheckpoint, Saver,
+                                            StepFromPath)
+  saver = Saver(str(tmp_path), keep_latest_n=2, async_save=True)
+  payload = {'model': {'w': torch.ones(3)}, 'step': 0}
+  for step in (1, 2, 3, 4):
+    payload['step'] = step
+    saver.Save(dict(payload), step)
+  saver.Sync()
+  saver._GC(); saver._WriteStateFile()
+  import glob
+  cks = sorted(glob.glob(str(tmp_path / 'ckpt-*.pt')))
+  assert len(cks) == 2
+  latest = LatestCheckpoint(str(tmp_path))
+  assert StepFromPath(latest) == 4
+
+
+def test_saver_rejects_nonfinite(tmp_path):
+  import torch
+  from lingvo_amd.core.checkpointer import Saver
+  saver = Saver(str(tmp_path))
+  bad = {'model': {'w': torch.tensor([1.0, float('nan')])}, 'step': 1}
+  with pytest.raises(FloatingPointError):
+    saver.Save(bad, 1)
