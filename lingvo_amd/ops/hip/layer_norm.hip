@@ -191,6 +191,156 @@ __global__ void ln_bwd_reduce(const float* __restrict__ partials,
   }
 }
 
+// ---- large-D path: one block (256 threads) per row, 8 cols per thread ----
+// The wave-per-row kernel's per-lane D/64-wide register arrays spill to
+// scratch for D >= 2048 (observed 388 B/lane); here each thread owns 8
+// columns per pass so register use is constant in D.
+template <bool RMS, int ITERS>
+__global__ __launch_bounds__(kBlock) void ln_fwd_big(
+    const unsigned short* __restrict__ x,
+    const unsigned short* __restrict__ scale,
+    const unsigned short* __restrict__ bias,
+    unsigned short* __restrict__ y, float* __restrict__ mean_out,
+    float* __restrict__ rstd_out, int rows, int D, float eps) {
+  __shared__ float scratch[kWavesPerBlock];
+  const int tid = threadIdx.x;
+  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+    const unsigned short* xr = x + (long)row * D;
+    float sum = 0.f, sumsq = 0.f;
+#pragma unroll
+    for (int it = 0; it < ITERS; ++it) {
+      int base = (it * kBlock + tid) * 8;
+      if (base < D) {
+        ushortx8 v = *reinterpret_cast<const ushortx8*>(xr + base);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float f = bf16_bits_to_float(v[j]);
+          sum += f;
+          sumsq += f * f;
+        }
+      }
+    }
+    sum = block_reduce_sum<kWavesPerBlock>(sum, scratch);
+    sumsq = block_reduce_sum<kWavesPerBlock>(sumsq, scratch);
+    float mu = RMS ? 0.f : sum / D;
+    float rstd = rsqrtf(sumsq / D - mu * mu + eps);
+    if (tid == 0) {
+      if (!RMS) mean_out[row] = mu;
+      rstd_out[row] = rstd;
+    }
+    unsigned short* yr = y + (long)row * D;
+#pragma unroll
+    for (int it = 0; it < ITERS; ++it) {
+      int base = (it * kBlock + tid) * 8;
+      if (base < D) {
+        ushortx8 v = *reinterpret_cast<const ushortx8*>(xr + base);
+        ushortx8 ws = *reinterpret_cast<const ushortx8*>(scale + base);
+        ushortx8 ov;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float xhat = (bf16_bits_to_float(v[j]) - mu) * rstd;
+          float w = 1.f + bf16_bits_to_float(ws[j]);
+          float bb =
+              RMS ? 0.f
+                  : bf16_bits_to_float(
+                        reinterpret_cast<const unsigned short*>(bias)[base +
+                                                                      j]);
+          ov[j] = float_to_bf16_bits(xhat * w + bb);
+        }
+        *reinterpret_cast<ushortx8*>(yr + base) = ov;
+      }
+    }
+    __syncthreads();
+  }
+}
+
+template <bool RMS, int ITERS>
+__global__ __launch_bounds__(kBlock) void ln_bwd_big(
+    const unsigned short* __restrict__ dy,
+    const unsigned short* __restrict__ x,
+    const unsigned short* __restrict__ scale,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    unsigned short* __restrict__ dx, float* __restrict__ partials,
+    int rows, int D) {
+  __shared__ float scratch[kWavesPerBlock];
+  const int tid = threadIdx.x;
+  float ds_acc[ITERS][8];
+  float db_acc[ITERS][8];
+#pragma unroll
+  for (int it = 0; it < ITERS; ++it)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      ds_acc[it][j] = 0.f;
+      db_acc[it][j] = 0.f;
+    }
+  const float inv_d = 1.f / (float)D;
+  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+    const unsigned short* xr = x + (long)row * D;
+    const unsigned short* dyr = dy + (long)row * D;
+    float mu = RMS ? 0.f : mean[row];
+    float rs = rstd[row];
+    float s1 = 0.f, s2 = 0.f;
+#pragma unroll
+    for (int it = 0; it < ITERS; ++it) {
+      int base = (it * kBlock + tid) * 8;
+      if (base < D) {
+        ushortx8 xv = *reinterpret_cast<const ushortx8*>(xr + base);
+        ushortx8 dv = *reinterpret_cast<const ushortx8*>(dyr + base);
+        ushortx8 ws = *reinterpret_cast<const ushortx8*>(scale + base);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float xh = (bf16_bits_to_float(xv[j]) - mu) * rs;
+          float dyf = bf16_bits_to_float(dv[j]);
+          float w = 1.f + bf16_bits_to_float(ws[j]);
+          float dxhat = dyf * w;
+          s1 += dxhat;
+          s2 += dxhat * xh;
+          ds_acc[it][j] += dyf * xh;
+          db_acc[it][j] += dyf;
+        }
+      }
+    }
+    s1 = block_reduce_sum<kWavesPerBlock>(s1, scratch) * inv_d;
+    s2 = block_reduce_sum<kWavesPerBlock>(s2, scratch) * inv_d;
+    unsigned short* dxr = dx + (long)row * D;
+#pragma unroll
+    for (int it = 0; it < ITERS; ++it) {
+      int base = (it * kBlock + tid) * 8;
+      if (base < D) {
+        ushortx8 xv = *reinterpret_cast<const ushortx8*>(xr + base);
+        ushortx8 dv = *reinterpret_cast<const ushortx8*>(dyr + base);
+        ushortx8 ws = *reinterpret_cast<const ushortx8*>(scale + base);
+        ushortx8 ov;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float xh = (bf16_bits_to_float(xv[j]) - mu) * rs;
+          float dxhat = bf16_bits_to_float(dv[j]) *
+                        (1.f + bf16_bits_to_float(ws[j]));
+          float val = RMS ? rs * (dxhat - xh * s2)
+                          : rs * (dxhat - s1 - xh * s2);
+          ov[j] = float_to_bf16_bits(val);
+        }
+        *reinterpret_cast<ushortx8*>(dxr + base) = ov;
+      }
+    }
+    __syncthreads();
+  }
+  // Column partials: [2, nblocks, D].
+  float* ds = partials + (long)blockIdx.x * D;
+  float* db = partials + (long)(gridDim.x + blockIdx.x) * D;
+#pragma unroll
+  for (int it = 0; it < ITERS; ++it) {
+    int base = (it * kBlock + tid) * 8;
+    if (base < D) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        ds[base + j] = ds_acc[it][j];
+        db[base + j] = db_acc[it][j];
+      }
+    }
+  }
+}
+
 // ---- generic-D fallback (scalar, one block per row) -----------------------
 template <bool RMS>
 __global__ void ln_fwd_generic(const unsigned short* __restrict__ x,
@@ -278,8 +428,15 @@ __global__ void ln_bwd_generic(const unsigned short* __restrict__ dy,
 
 inline bool use_vec_path(long D) {
   long nvec = D / WAVE_SIZE;
-  return D % (WAVE_SIZE * 8) == 0 &&
-         (nvec == 8 || nvec == 16 || nvec == 24 || nvec == 32 || nvec == 64);
+  // Larger NVEC spills to scratch (D/64-wide per-lane arrays); D >= 1536
+  // takes the block-per-row big path instead.
+  return D % (WAVE_SIZE * 8) == 0 && (nvec == 8 || nvec == 16);
+}
+
+inline int big_iters(long D) {  // 0 = not eligible
+  if (D % 8 != 0) return 0;
+  long iters = (D + 8 * 256 - 1) / (8 * 256);
+  return iters <= 4 ? (int)iters : 0;
 }
 
 }  // namespace
@@ -330,6 +487,26 @@ std::vector<torch::Tensor> layer_norm_fwd(torch::Tensor x, torch::Tensor scale,
         TORCH_CHECK(false, "unhandled NVEC");
     }
 #undef LN_FWD_CASE
+  } else if (big_iters(D)) {
+    int grid = (int)std::min<long>(rows, 2048);
+#define LN_FWD_BIG(IT)                                                     \
+  case IT:                                                                 \
+    if (rms)                                                               \
+      hipLaunchKernelGGL((ln_fwd_big<true, IT>), dim3(grid), dim3(kBlock), \
+                         0, stream, xp, sp, bp, yp, mp, rp, (int)rows,     \
+                         (int)D, (float)eps);                              \
+    else                                                                   \
+      hipLaunchKernelGGL((ln_fwd_big<false, IT>), dim3(grid),              \
+                         dim3(kBlock), 0, stream, xp, sp, bp, yp, mp, rp,  \
+                         (int)rows, (int)D, (float)eps);                   \
+    break;
+    switch (big_iters(D)) {
+      LN_FWD_BIG(1)
+      LN_FWD_BIG(2)
+      LN_FWD_BIG(3)
+      LN_FWD_BIG(4)
+    }
+#undef LN_FWD_BIG
   } else {
     int grid = memory_bound_grid(rows * kBlock, kBlock, 1024);
     if (rms)
@@ -397,6 +574,34 @@ std::vector<torch::Tensor> layer_norm_bwd(torch::Tensor dy, torch::Tensor x,
                        0, stream, partials.data_ptr<float>(),
                        dscale.data_ptr<float>(), dbias.data_ptr<float>(),
                        nwaves, (int)D);
+  } else if (big_iters(D)) {
+    int grid = (int)std::min<long>(rows, 1024);
+    auto partials = torch::empty({2L * grid, D}, opts);
+#define LN_BWD_BIG(IT)                                                    \
+  case IT:                                                                \
+    if (rms)                                                              \
+      hipLaunchKernelGGL((ln_bwd_big<true, IT>), dim3(grid),              \
+                         dim3(kBlock), 0, stream, dyp, xp, sp, mp, rp,    \
+                         dxp, partials.data_ptr<float>(), (int)rows,      \
+                         (int)D);                                         \
+    else                                                                  \
+      hipLaunchKernelGGL((ln_bwd_big<false, IT>), dim3(grid),             \
+                         dim3(kBlock), 0, stream, dyp, xp, sp, mp, rp,    \
+                         dxp, partials.data_ptr<float>(), (int)rows,      \
+                         (int)D);                                         \
+    break;
+    switch (big_iters(D)) {
+      LN_BWD_BIG(1)
+      LN_BWD_BIG(2)
+      LN_BWD_BIG(3)
+      LN_BWD_BIG(4)
+    }
+#undef LN_BWD_BIG
+    int splitk3 = (int)std::min<long>(64, std::max<long>(1, grid / 8));
+    hipLaunchKernelGGL(ln_bwd_reduce, dim3(cdiv(D, 256), splitk3),
+                       dim3(256), 0, stream, partials.data_ptr<float>(),
+                       dscale.data_ptr<float>(), dbias.data_ptr<float>(),
+                       grid, (int)D);
   } else {
     int grid = memory_bound_grid(rows, 1, 256);
     auto partials = torch::empty({2L * grid, D}, opts);

@@ -26,7 +26,7 @@ def _rms_ref(x, scale, eps=1e-6):
 
 @gpu
 @pytest.mark.parametrize('shape', [(4, 7, 512), (2, 3, 1024), (8, 2048),
-                                   (3, 5, 384), (2, 130)])
+                                   (3, 5, 384), (2, 130), (4, 4096)])
 def test_layer_norm_fwd_matches_ref(shape):
   from lingvo_amd.ops import layer_norm as ln
   torch.manual_seed(0)
@@ -40,7 +40,7 @@ def test_layer_norm_fwd_matches_ref(shape):
 
 
 @gpu
-@pytest.mark.parametrize('d', [512, 1024, 384])
+@pytest.mark.parametrize('d', [512, 1024, 384, 2048, 4096])
 def test_layer_norm_bwd_matches_ref(d):
   from lingvo_amd.ops import layer_norm as ln
   torch.manual_seed(1)
