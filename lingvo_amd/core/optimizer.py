@@ -275,3 +275,63 @@ class Accumulator(Base):
 
   def CreateTorchOptimizer(self, params, lr):
     return self.inner.CreateTorchOptimizer(params, lr)
+
+
+class AdaDelta(Base):
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('decay', 0.95, 'Rho.')
+    p.Define('epsilon', 1e-6, 'Epsilon.')
+    return p
+
+  def CreateTorchOptimizer(self, params, lr):
+    return torch.optim.Adadelta(params, lr=lr, rho=self.p.decay,
+                                eps=self.p.epsilon)
+
+
+class CompositeOptimizer(Base):
+  """Regex -> optimizer routing (reference optimizer.py:199): parameters
+  whose names match a pattern use that sub-optimizer."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('optimizer_map', [],
+             'List of (name_regex, optimizer_params, lr_mult).')
+    p.Define('default_optimizer', Adam.Params(), 'Fallback optimizer.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    subs = [op.Copy().Set(name=f'sub_{i}')
+            for i, (_, op, _) in enumerate(self.p.optimizer_map)]
+    subs.append(self.p.default_optimizer.Copy().Set(name='sub_default'))
+    self.CreateChildren('subs', subs)
+
+  def CreateTorchOptimizer(self, params, lr):
+    import re as _re
+    # params must be (name, param) pairs for routing; Learner passes raw
+    # params, so CompositeOptimizer users call CreateRoutedOptimizers.
+    return self.subs[-1].CreateTorchOptimizer(params, lr)
+
+  def CreateRoutedOptimizers(self, named_params, lr):
+    import re as _re
+    buckets = [[] for _ in self.p.optimizer_map]
+    default = []
+    for name, prm in named_params:
+      for i, (pat, _, _) in enumerate(self.p.optimizer_map):
+        if _re.search(pat, name):
+          buckets[i].append(prm)
+          break
+      else:
+        default.append(prm)
+    opts = []
+    for i, (_, _, mult) in enumerate(self.p.optimizer_map):
+      if buckets[i]:
+        opts.append(self.subs[i].CreateTorchOptimizer(buckets[i],
+                                                      lr * mult))
+    if default:
+      opts.append(self.subs[-1].CreateTorchOptimizer(default, lr))
+    return opts

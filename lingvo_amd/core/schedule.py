@@ -148,3 +148,39 @@ class PolynomialSchedule(BaseSchedule):
       return v1
     frac = (step - s0) / (s1 - s0)
     return v0 + (v1 - v0) * frac ** p.power
+
+
+class DevBasedSchedule(BaseSchedule):
+  """Plateau-based decay (reference schedule.py DevBasedSchedule): the
+  eval job reports the dev metric via ReportMetric; the multiplier decays
+  by `factor` when the metric has not improved for `window` steps."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('factor', 0.5, 'Decay factor on plateau.')
+    p.Define('window', 5000, 'Steps without improvement before decay.')
+    p.Define('min_factor', 0.01, 'Multiplier floor.')
+    p.Define('metric_minimize', True, 'Lower metric is better.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    self._mult = 1.0
+    self._best = None
+    self._best_step = 0
+
+  def ReportMetric(self, value: float, step: int) -> None:
+    p = self.p
+    better = (self._best is None or
+              (value < self._best if p.metric_minimize
+               else value > self._best))
+    if better:
+      self._best = value
+      self._best_step = step
+    elif step - self._best_step > p.window:
+      self._mult = max(p.min_factor, self._mult * p.factor)
+      self._best_step = step
+
+  def Value(self, step: int) -> float:
+    return self._mult

@@ -319,6 +319,10 @@ class SimpleFullSoftmax(BaseLayer):
     p.Define('input_dim', 0, 'Input dim.')
     p.Define('num_classes', 0, 'Number of classes.')
     p.Define('chunk_size', 0, 'If >0, compute xent in vocab chunks.')
+    p.Define('num_sampled', 0,
+             'If >0, sampled softmax with this many negatives during '
+             'training (reference layers.py:3697 sampled support; the '
+             'WordLevelOneBwdsSimpleSampledSoftmax baseline).')
     return p
 
   def __init__(self, params):
@@ -344,6 +348,22 @@ class SimpleFullSoftmax(BaseLayer):
                                reduction='none')
     return NestedMap(per_example_xent=per_example, log_probs=log_probs)
 
+  def _SampledXent(self, theta, inputs2d, class_ids):
+    """Uniform-negative sampled softmax (training only)."""
+    p = self.p
+    n = inputs2d.shape[0]
+    labels = class_ids.reshape(-1).long()
+    u = py_utils.GraphSafeUniform((p.num_sampled,), inputs2d.device)
+    neg = (u * p.num_classes).long().clamp_max(p.num_classes - 1)
+    cols = torch.cat([labels, neg])  # [N + S]
+    w_cols = theta.linear_w[:, cols]  # [D, N+S]
+    b_cols = theta.bias[cols]
+    logits = torch.matmul(inputs2d, w_cols) + b_cols  # [N, N+S]
+    # The true class of row i is column i.
+    tgt = torch.arange(n, device=inputs2d.device)
+    import torch.nn.functional as F2
+    return F2.cross_entropy(logits.float(), tgt, reduction='none')
+
   def XentLoss(self, theta: NestedMap, inputs: torch.Tensor,
                class_weights: torch.Tensor,
                class_ids: Optional[torch.Tensor] = None,
@@ -352,6 +372,14 @@ class SimpleFullSoftmax(BaseLayer):
     p = self.p
     inputs2d = inputs.reshape(-1, p.input_dim)
     w = class_weights.reshape(-1).float()
+    if p.num_sampled and self.training and class_ids is not None:
+      per_example = self._SampledXent(theta, inputs2d, class_ids)
+      total_weight = w.sum()
+      total_xent = (per_example * w).sum()
+      return NestedMap(
+          total_xent=total_xent, total_weight=total_weight,
+          avg_xent=total_xent / total_weight.clamp_min(1e-8),
+          per_example_xent=per_example)
     if (inputs2d.is_cuda and class_probabilities is None and
         p.chunk_size == 0):
       from lingvo_amd.ops import softmax_xent

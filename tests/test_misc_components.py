@@ -168,3 +168,71 @@ def test_datasets_introspection():
   from lingvo_amd.runtime import datasets
   ds = datasets.GetDatasets('image.mnist.LeNet5')
   assert ds == ['Dev', 'Test', 'Train']
+
+
+def test_sampled_softmax():
+  from lingvo_amd.layers import layers as lingvo_layers
+  p = lingvo_layers.SimpleFullSoftmax.Params().Set(
+      name='sm', input_dim=16, num_classes=1000, num_sampled=64,
+      random_seed=1)
+  sm = p.Instantiate()
+  x = torch.randn(8, 16)
+  ids = torch.randint(0, 1000, (8,))
+  with py_utils.StepSeedScope(1, 0):
+    xent = sm.XentLoss(sm.theta, x, class_weights=torch.ones(8),
+                       class_ids=ids)
+  assert xent.per_example_xent.shape == (8,)
+  assert torch.isfinite(xent.avg_xent)
+  # eval path uses the full softmax
+  sm.eval()
+  xent_full = sm.XentLoss(sm.theta, x, class_weights=torch.ones(8),
+                          class_ids=ids)
+  assert xent_full.per_example_xent.shape == (8,)
+
+
+def test_dev_based_schedule():
+  from lingvo_amd.core import schedule as schedule_lib
+  s = schedule_lib.DevBasedSchedule.Params().Set(
+      name='dev', factor=0.5, window=10).Instantiate()
+  s.ReportMetric(5.0, 0)
+  assert s.Value(5) == 1.0
+  s.ReportMetric(6.0, 20)  # worse, past window -> decay
+  assert s.Value(20) == 0.5
+  s.ReportMetric(4.0, 25)  # improvement resets
+  assert s.Value(25) == 0.5
+
+
+def test_composite_optimizer_routing():
+  from lingvo_amd.core import optimizer as optimizer_lib
+  p = optimizer_lib.CompositeOptimizer.Params().Set(
+      name='comp',
+      optimizer_map=[(r'bias', optimizer_lib.SGD.Params(), 0.1)],
+      default_optimizer=optimizer_lib.Adam.Params())
+  comp = p.Instantiate()
+  lin = torch.nn.Linear(4, 4)
+  opts = comp.CreateRoutedOptimizers(lin.named_parameters(), lr=0.1)
+  assert len(opts) == 2
+  assert isinstance(opts[0], torch.optim.SGD)
+  assert abs(opts[0].param_groups[0]['lr'] - 0.01) < 1e-9
+
+
+def test_datasources():
+  from lingvo_amd.core import datasource
+  from lingvo_amd.models import mnist as mnist_model
+  sp = datasource.SimpleDataSource.Params().Set(
+      name='s',
+      input_generator=mnist_model.FakeMnistData.Params().Set(batch_size=2))
+  src = sp.Instantiate()
+  b = src.GetNext()
+  assert b.data.shape[0] == 2
+  mix_p = datasource.CrossBatchMixingDataSource.Params().Set(
+      name='mix', sub=[sp.Copy(), sp.Copy()], weights=[0.9, 0.1],
+      random_seed=1)
+  mix = mix_p.Instantiate()
+  ids = [int(mix.GetNext().source_id) for _ in range(20)]
+  assert ids.count(0) > ids.count(1)
+  cur_p = datasource.CurriculumDataSource.Params().Set(
+      name='cur', sub=[sp.Copy(), sp.Copy()], boundaries=[5])
+  cur = cur_p.Instantiate()
+  cur.SetStep(0); cur.GetNext()
+  cur.SetStep(10); cur.GetNext()
