@@ -145,6 +145,47 @@ class MultiHeadedAttention(BaseLayer):
                           dtype=dtype),
         time_step=0)
 
+  def StreamStep(self, theta: NestedMap, x_chunk: torch.Tensor,
+                 paddings_chunk: torch.Tensor,
+                 state: NestedMap) -> tuple:
+    """Chunked streaming self-attention (reference StreamStep,
+    batch_major_attention.py / conformer_layer.py:390): attends causally
+    within the chunk and over up to `left_context` cached frames."""
+    p = self.p
+    n, nkv, h = self._n, self._nkv, self._h
+    b, c = x_chunk.shape[0], x_chunk.shape[1]
+    q, k, v = self._Project(theta, x_chunk)
+    t0 = state.time_step
+    L = p.left_context if p.left_context >= 0 else state.key.shape[1]
+    state.key[:, t0:t0 + c] = k.to(state.key.dtype)
+    state.value[:, t0:t0 + c] = v.to(state.value.dtype)
+    state.time_step = t0 + c
+    lo = max(0, t0 - L)
+    keys = state.key[:, lo:t0 + c]
+    values = state.value[:, lo:t0 + c]
+    group = n // nkv
+    kf = keys.float()
+    vf = values.float()
+    if group > 1:
+      kf = kf.repeat_interleave(group, dim=2)
+      vf = vf.repeat_interleave(group, dim=2)
+    logits = torch.einsum('bcnh,bsnh->bncs', q.float(), kf)
+    logits = logits / math.sqrt(h)
+    qpos = (t0 + torch.arange(c, device=x_chunk.device))[:, None]
+    kpos = (lo + torch.arange(kf.shape[1], device=x_chunk.device))[None, :]
+    mask = (kpos <= qpos) & (kpos >= qpos - L)
+    if p.rel_pos_bias:
+      d = (qpos - kpos).clamp(-p.rel_pos_clip, p.rel_pos_clip)           + p.rel_pos_clip
+      logits = logits + theta.rel_bias.float()[None, :, d]
+    logits = logits.masked_fill(~mask[None, None], -1e30)
+    probs = torch.softmax(logits, dim=-1)
+    ctx = torch.einsum('bncs,bsnh->bcnh', probs, vf)
+    ctx = ctx.reshape(b, c, n * h).to(x_chunk.dtype)
+    post = py_utils.MatmulBias(ctx, theta.post_w,
+                               theta.post_b if p.use_bias else None)
+    post = py_utils.ApplyPadding(paddings_chunk, post)
+    return post, state
+
   def ExtendStep(self, theta: NestedMap, query_vec: torch.Tensor,
                  cached_states: NestedMap,
                  per_step_padding=None) -> Tuple[torch.Tensor, NestedMap]:

@@ -91,6 +91,40 @@ class LConvLayer(BaseLayer):
                                               inputs)
     return inputs + x
 
+  def InitStreamState(self, batch: int, device, dtype) -> NestedMap:
+    p = self.p
+    return NestedMap(conv=torch.zeros(batch, p.kernel_size - 1,
+                                      p.input_dim, device=device,
+                                      dtype=dtype))
+
+  def StreamStep(self, theta: NestedMap, inputs: torch.Tensor,
+                 paddings: torch.Tensor, state: NestedMap):
+    """Causal streaming chunk (reference conformer_layer.py:390). Requires
+    is_causal=True and a stream-safe norm (conv_norm='layer')."""
+    p = self.p
+    assert p.is_causal, 'StreamStep requires a causal LConv'
+    x = self.ln.FProp(theta.ln, inputs)
+    x = py_utils.MatmulBias(x, theta.pw1_w, theta.pw1_b)
+    a, b = x.chunk(2, dim=-1)
+    x = a * torch.sigmoid(b)
+    x = py_utils.ApplyPadding(paddings, x)
+    # Depthwise causal conv over [state | chunk].
+    full = torch.cat([state.conv.to(x.dtype), x], dim=1)
+    y = conv1d_ops.depthwise_conv1d(full, theta.dw_w, theta.dw_b,
+                                    causal=True)
+    x = y[:, state.conv.shape[1]:]
+    new_state = NestedMap(conv=full[:, -(p.kernel_size - 1):]
+                          if p.kernel_size > 1 else state.conv)
+    if p.conv_norm == 'layer':
+      x = self.norm.FProp(theta.norm, x)
+      x = py_utils.ApplyPadding(paddings, x)
+    else:
+      x = self.norm.FProp(theta.norm, x, paddings)
+    x = F.silu(x)
+    x = py_utils.MatmulBias(x, theta.pw2_w, theta.pw2_b)
+    x = py_utils.ApplyPadding(paddings, x)
+    return inputs + x, new_state
+
 
 class ConformerLayer(BaseLayer):
   """½FFN -> MHSA -> LConv -> ½FFN -> LN (reference conformer_layer.py:471)."""

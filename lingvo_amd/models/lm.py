@@ -130,6 +130,7 @@ class RnnLm(BaseLayer):
     p.Define('rnn_dims', [2048, 2048], 'Per-layer LSTM dims.')
     p.Define('rnn_proj', 1024, 'LSTM projection dim (0 = none).')
     p.Define('dropout_prob', 0.1, 'Dropout.')
+    p.Define('num_sampled', 0, 'Sampled-softmax negatives (training).')
     return p
 
   def __init__(self, params):
@@ -148,7 +149,8 @@ class RnnLm(BaseLayer):
     self.CreateChild('rnns', rnn_layers.StackedFRNNLayerByLayer.Params().Set(
         cell_tpl=cells, skip_start=1))
     self.CreateChild('softmax', lingvo_layers.SimpleFullSoftmax.Params().Set(
-        input_dim=in_dim, num_classes=p.vocab_size))
+        input_dim=in_dim, num_classes=p.vocab_size,
+        num_sampled=p.num_sampled))
 
   def FProp(self, theta, ids, paddings):
     x = self.emb.EmbLookup(theta.emb, ids.long()).to(self.fprop_dtype)

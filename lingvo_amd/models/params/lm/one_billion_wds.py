@@ -78,7 +78,8 @@ class WordLevelOneBwdsRnnLm(SingleTaskModelParams):
     p.train.bf16_weights = True
     p.lm = lm_model.RnnLm.Params().Set(
         vocab_size=32000, emb_dim=1024, rnn_dims=[2048, 2048],
-        rnn_proj=1024, dropout_prob=0.1)
+        rnn_proj=1024, dropout_prob=0.1,
+        num_sampled=4096)  # reference: SimpleSampledSoftmax baseline
     p.train.learner = learner_lib.Learner.Params().Set(
         learning_rate=0.1,
         optimizer=optimizer_lib.Adagrad.Params(),

@@ -76,7 +76,101 @@ def RecvNestedMap(src: int, group=None, device='cpu') -> NestedMap:
   return out
 
 
-class GPipeRunner:
+def _ISendTensor(t: torch.Tensor, dst: int, group):
+  header = torch.zeros(_MAX_DIMS + 2, dtype=torch.int64)
+  header[0] = t.dim()
+  for i, d in enumerate(t.shape):
+    header[1 + i] = d
+  header[_MAX_DIMS + 1] = _DTYPE_CODES[t.dtype]
+  t = t.contiguous()
+  w1 = dist.isend(header, dst, group=group)
+  w2 = dist.isend(t, dst, group=group)
+  return [(w1, header), (w2, t)]
+
+
+def ISendNestedMap(nmap: NestedMap, dst: int, group=None):
+  """Non-blocking send; returns [(work, tensor)] to keep alive/wait."""
+  pending = []
+  flat = nmap.FlattenItems()
+  count = torch.tensor([len(flat)], dtype=torch.int64)
+  pending.append((dist.isend(count, dst, group=group), count))
+  for key, val in flat:
+    kb = key.encode()[:64].ljust(64)
+    kt = torch.frombuffer(bytearray(kb), dtype=torch.uint8).clone()
+    pending.append((dist.isend(kt, dst, group=group), kt))
+    pending.extend(_ISendTensor(val, dst, group))
+  return pending
+
+
+class _GPipe1F1BMixin:
+  """1F1B schedule implementation (Megatron-style ordering): at most
+  (num_stages - stage_idx) live activations instead of all M."""
+
+  def _Run1F1B(self, stage_fprop, input_fn, loss_fn):
+    saved_in = {}
+    saved_out = {}
+    losses = {}
+    pending = []
+
+    def forward(m):
+      if self.is_first:
+        inp = input_fn(m)
+      else:
+        inp = RecvNestedMap(self._prev, self.group, self.device)
+        inp = inp.Transform(
+            lambda t: t.requires_grad_(True)
+            if t.is_floating_point() else t)
+      out = stage_fprop(inp)
+      if self.is_last:
+        losses[m] = loss_fn(out, m)
+      else:
+        pending.extend(ISendNestedMap(
+            out.Transform(lambda t: t.detach()), self._next, self.group))
+        saved_out[m] = out
+      saved_in[m] = inp
+
+    def backward(m):
+      if self.is_last:
+        (losses[m] / self.num_micro).backward()
+      else:
+        grads = RecvNestedMap(self._next, self.group, self.device)
+        out = saved_out.pop(m)
+        otensors, gtensors = [], []
+        for key, val in out.FlattenItems():
+          g = grads.Get(key)
+          if g is not None and isinstance(val, torch.Tensor) and \
+              val.requires_grad:
+            otensors.append(val)
+            gtensors.append(g)
+        torch.autograd.backward(otensors, gtensors)
+      inp = saved_in.pop(m)
+      if not self.is_first:
+        gmap = NestedMap()
+        for key, val in inp.FlattenItems():
+          if isinstance(val, torch.Tensor) and val.requires_grad and \
+              val.grad is not None:
+            gmap.Set(key, val.grad)
+        pending.extend(ISendNestedMap(gmap, self._prev, self.group))
+
+    m_total = self.num_micro
+    warmup = min(self.num_stages - 1 - self.stage_idx, m_total)
+    for m in range(warmup):
+      forward(m)
+    for i in range(m_total - warmup):
+      forward(warmup + i)
+      backward(i)
+    for i in range(m_total - warmup, m_total):
+      backward(i)
+    for work, _ in pending:
+      work.wait()
+    if self.is_last and losses:
+      return torch.stack(
+          [losses[m].detach() for m in range(m_total)]).mean()
+    return None
+
+
+
+class GPipeRunner(_GPipe1F1BMixin):
   """Fill-drain microbatch schedule for one pipeline stage.
 
   stage_fprop(microbatch_nmap) -> nmap: this stage's forward (already
@@ -106,11 +200,18 @@ class GPipeRunner:
   def RunStep(self, stage_fprop: Callable[[NestedMap], NestedMap],
               input_fn: Optional[Callable[[int], NestedMap]] = None,
               loss_fn: Optional[Callable[[NestedMap, int], torch.Tensor]]
-              = None) -> Optional[torch.Tensor]:
-    """One full train step = M forward microbatches then M backwards.
+              = None, schedule: str = 'fill_drain'
+              ) -> Optional[torch.Tensor]:
+    """One full train step over M microbatches.
 
-    Returns the mean loss on the last stage, None elsewhere.
+    schedule='fill_drain' (GPipe) or '1f1b' (one-forward-one-backward:
+    at most `num_stages - stage_idx` activations live at a time instead
+    of all M — the standard memory-efficient PP schedule). Gradients are
+    identical between schedules (tested); returns the mean loss on the
+    last stage, None elsewhere.
     """
+    if schedule == '1f1b':
+      return self._Run1F1B(stage_fprop, input_fn, loss_fn)
     saved_in: List[NestedMap] = []
     saved_out: List[NestedMap] = []
     losses: List[torch.Tensor] = []

@@ -224,3 +224,41 @@ def test_conv_layers_with_time_padding():
   pooled = gp.FProp(gp.theta, x2,
                     py_utils.PaddingsFromLengths(torch.tensor([10, 4]), 10))
   assert pooled.shape == (2, 8)
+
+
+def test_streaming_matches_full_causal():
+  """Chunked StreamStep == full causal FProp (MHA and LConv)."""
+  # MHA streaming
+  p = attention_lib.MultiHeadedAttention.Params().Set(
+      name='mha', input_dim=64, hidden_dim=64, num_heads=1, causal=True,
+      left_context=6, random_seed=5)
+  mha = p.Instantiate()
+  mha.eval()
+  x = torch.randn(2, 12, 64)
+  pad = torch.zeros(2, 12)
+  full = mha.FProp(mha.theta, x, pad)
+  state = mha.InitStates(mha.theta, 2, 12, 'cpu', torch.float32)
+  outs = []
+  for c0 in range(0, 12, 4):
+    o, state = mha.StreamStep(mha.theta, x[:, c0:c0 + 4],
+                              pad[:, c0:c0 + 4], state)
+    outs.append(o)
+  stream = torch.cat(outs, dim=1)
+  assert (full - stream).abs().max() < 1e-3
+
+  # LConv streaming
+  lp = conformer_lib.LConvLayer.Params().Set(
+      name='lconv', input_dim=16, kernel_size=4, is_causal=True,
+      conv_norm='layer', random_seed=5)
+  lconv = lp.Instantiate()
+  lconv.eval()
+  x2 = torch.randn(2, 12, 16)
+  full2 = lconv.FProp(lconv.theta, x2, pad)
+  st = lconv.InitStreamState(2, 'cpu', torch.float32)
+  outs2 = []
+  for c0 in range(0, 12, 3):
+    o, st = lconv.StreamStep(lconv.theta, x2[:, c0:c0 + 3],
+                             pad[:, c0:c0 + 3], st)
+    outs2.append(o)
+  stream2 = torch.cat(outs2, dim=1)
+  assert (full2 - stream2).abs().max() < 1e-4

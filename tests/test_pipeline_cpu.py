@@ -161,3 +161,65 @@ def test_gpipe_transformer_lm_two_stages():
     losses.append(xent.avg_xent)
   ref_loss = float(torch.stack(losses).mean())
   assert abs(loss - ref_loss) < 1e-4, (loss, ref_loss)
+
+
+def _run_stage_1f1b(rank, world, port, num_micro, results):
+  os.environ['MASTER_ADDR'] = '127.0.0.1'
+  os.environ['MASTER_PORT'] = str(port)
+  dist.init_process_group('gloo', rank=rank, world_size=world)
+  from lingvo_amd.core.nested_map import NestedMap
+  from lingvo_amd.parallel.pipeline import GPipeRunner
+
+  stage = _make_stage(rank)
+  runner = GPipeRunner(rank, world, num_micro)
+
+  def fprop(nmap):
+    return NestedMap(act=stage(nmap.act))
+
+  def input_fn(m):
+    return NestedMap(act=_inputs(m))
+
+  def loss_fn(nmap, m):
+    return ((nmap.act - _targets(m)) ** 2).mean()
+
+  loss = runner.RunStep(fprop, input_fn=input_fn, loss_fn=loss_fn,
+                        schedule='1f1b')
+  results[f'loss{rank}'] = None if loss is None else float(loss)
+  results[f'grads{rank}'] = torch.cat(
+      [p.grad.reshape(-1) for p in stage.parameters()])
+  dist.destroy_process_group()
+
+
+@pytest.mark.parametrize('world', [2, 3])
+def test_1f1b_matches_fill_drain_reference(world):
+  num_micro = 5
+  ctx = mp.get_context('spawn')
+  with ctx.Manager() as mgr:
+    results = mgr.dict()
+    procs = [ctx.Process(target=_run_stage_1f1b,
+                         args=(r, world, 29537 + world, num_micro,
+                               results))
+             for r in range(world)]
+    for p in procs:
+      p.start()
+    for p in procs:
+      p.join(180)
+      assert p.exitcode == 0
+    loss_pipe = results[f'loss{world - 1}']
+    grads = [results[f'grads{r}'] for r in range(world)]
+
+  # Single-process reference.
+  stages = [_make_stage(r) for r in range(world)]
+  losses = []
+  for m in range(num_micro):
+    x = _inputs(m)
+    for st in stages:
+      x = st(x)
+    losses.append(((x - _targets(m)) ** 2).mean())
+  total = torch.stack(losses).mean()
+  total.backward()
+  assert abs(loss_pipe - float(total)) < 1e-5
+  for r in range(world):
+    ref = torch.cat([p.grad.reshape(-1)
+                     for p in stages[r].parameters()])
+    assert torch.allclose(grads[r], ref, atol=1e-6), r
