@@ -238,6 +238,18 @@ class TransformerModel(BaseTask):
     out.target_ids = input_batch.tgt.ids
     return out
 
+  def Inference(self) -> NestedMap:
+    """Beam-search translate subgraph (reference base_model.py:943)."""
+
+    def default(src_ids, src_paddings):
+      enc = self.encoder.FProp(self.theta.encoder, src_ids, src_paddings)
+      out = self.decoder.BeamSearchDecode(self.theta.decoder, enc,
+                                          src_paddings)
+      return NestedMap(topk_ids=out.topk_ids, topk_lens=out.topk_lens,
+                       topk_scores=out.topk_scores)
+
+    return NestedMap(default=default)
+
   def CreateDecoderMetrics(self) -> NestedMap:
     return NestedMap(corpus_bleu=metrics_lib.CorpusBleuMetric(),
                      num_samples_in_batch=metrics_lib.AverageMetric())
@@ -250,22 +262,3 @@ class TransformerModel(BaseTask):
       ref = ' '.join(str(int(x)) for x in refs[i] if int(x) > 2)
       decode_metrics.corpus_bleu.Update(ref, hyp)
     decode_metrics.num_samples_in_batch.Update(float(hyps.shape[0]))
-
-
-def _AddMtInference():
-  def Inference(self):
-    """Beam-search translate subgraph (reference base_model.py:943)."""
-
-    def default(src_ids, src_paddings):
-      enc = self.encoder.FProp(self.theta.encoder, src_ids, src_paddings)
-      out = self.decoder.BeamSearchDecode(self.theta.decoder, enc,
-                                          src_paddings)
-      return NestedMap(topk_ids=out.topk_ids, topk_lens=out.topk_lens,
-                       topk_scores=out.topk_scores)
-
-    return NestedMap(default=default)
-
-  TransformerModel.Inference = Inference
-
-
-_AddMtInference()
