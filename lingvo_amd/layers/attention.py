@@ -81,17 +81,20 @@ class MultiHeadedAttention(BaseLayer):
 
   def FProp(self, theta: NestedMap, query_vec: torch.Tensor,
             paddings: Optional[torch.Tensor] = None,
-            segment_mask=None) -> torch.Tensor:
-    """Self-attention over [B, T, D] with [B, T] paddings."""
+            segment_ids: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Self-attention over [B, T, D] with [B, T] paddings. Optional
+    packed-input segment_ids [B, T] block cross-segment attention
+    (reference packed-input segment mask, batch_major_attention.py)."""
     p = self.p
     q, k, v = self._Project(theta, query_vec)
     klen = None
-    if paddings is not None:
+    if paddings is not None and segment_ids is None:
       klen = py_utils.LengthsFromPaddings(paddings).to(torch.int32)
     bias = theta.rel_bias if p.rel_pos_bias else None
     win_r = 0 if p.causal else p.right_context
     out = flash_attn.flash_attention(
-        q, k, v, klen, bias, p.left_context, win_r, p.rel_pos_clip)
+        q, k, v, klen, bias, p.left_context, win_r, p.rel_pos_clip,
+        q_segment_ids=segment_ids, k_segment_ids=segment_ids)
     if p.atten_dropout_prob and not self.do_eval:
       # Dropout on the context vectors (prob-dropout approximation; the
       # reference drops attention probs, batch_major_attention.py:1045).

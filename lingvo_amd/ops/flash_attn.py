@@ -22,7 +22,8 @@ import torch
 from lingvo_amd.ops import _loader
 
 
-def _ref_attention(q, k, v, klen, bias, win_l, win_r, bias_clip, scale):
+def _ref_attention(q, k, v, klen, bias, win_l, win_r, bias_clip, scale,
+                   q_segment_ids=None, k_segment_ids=None):
   """fp32 reference with identical semantics (any backend)."""
   B, T, N, H = q.shape
   S, NKV = k.shape[1], k.shape[2]
@@ -47,6 +48,9 @@ def _ref_attention(q, k, v, klen, bias, win_l, win_r, bias_clip, scale):
   mask = mask[None, None]
   if klen is not None:
     mask = mask & (kpos[None, None] < klen[:, None, None, None])
+  if q_segment_ids is not None:
+    seg = (q_segment_ids[:, :, None] == k_segment_ids[:, None, :])
+    mask = mask & seg[:, None]
   logits = logits.masked_fill(~mask, -1e30)
   probs = torch.softmax(logits, dim=-1)
   # fully-masked rows -> 0
@@ -59,14 +63,18 @@ def _ref_attention(q, k, v, klen, bias, win_l, win_r, bias_clip, scale):
 class _FlashAttnFn(torch.autograd.Function):
 
   @staticmethod
-  def forward(ctx, q, k, v, klen, bias, win_l, win_r, bias_clip, scale):
+  def forward(ctx, q, k, v, klen, bias, qseg, kseg, win_l, win_r,
+              bias_clip, scale):
     ext = _loader.get_ext(required=True)
     bias_b = None if bias is None else bias.to(torch.bfloat16).contiguous()
-    o, lse = ext.fa_fwd(q, k, v, klen, bias_b, win_l, win_r, bias_clip,
-                        scale)
+    o, lse = ext.fa_fwd(q, k, v, klen, bias_b, qseg, kseg, win_l, win_r,
+                        bias_clip, scale)
+    empty = torch.empty(0)
     ctx.save_for_backward(q, k, v, o, lse,
-                          klen if klen is not None else torch.empty(0),
-                          bias_b if bias_b is not None else torch.empty(0))
+                          klen if klen is not None else empty,
+                          bias_b if bias_b is not None else empty,
+                          qseg if qseg is not None else empty,
+                          kseg if kseg is not None else empty)
     ctx.cfg = (win_l, win_r, bias_clip, scale, bias is not None and
                bias.requires_grad, None if bias is None else bias.dtype)
     return o
@@ -74,15 +82,18 @@ class _FlashAttnFn(torch.autograd.Function):
   @staticmethod
   def backward(ctx, dout):
     ext = _loader.get_ext(required=True)
-    q, k, v, o, lse, klen, bias_b = ctx.saved_tensors
+    q, k, v, o, lse, klen, bias_b, qseg, kseg = ctx.saved_tensors
     win_l, win_r, bias_clip, scale, bias_grad, bias_dtype = ctx.cfg
     klen = klen if klen.numel() else None
     bias_b = bias_b if bias_b.numel() else None
+    qseg = qseg if qseg.numel() else None
+    kseg = kseg if kseg.numel() else None
     dq, dk, dv, dbias = ext.fa_bwd(
-        dout.contiguous(), q, k, v, o, lse, klen, bias_b, bias_grad,
-        win_l, win_r, bias_clip, scale)
+        dout.contiguous(), q, k, v, o, lse, klen, bias_b, qseg, kseg,
+        bias_grad, win_l, win_r, bias_clip, scale)
     dbias_out = dbias.to(bias_dtype) if bias_grad else None
-    return dq, dk, dv, None, dbias_out, None, None, None, None
+    return (dq, dk, dv, None, dbias_out, None, None, None, None, None,
+            None)
 
 
 def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
@@ -90,18 +101,27 @@ def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                     bias: Optional[torch.Tensor] = None,
                     win_l: int = -1, win_r: int = -1,
                     bias_clip: int = 127,
-                    scale: Optional[float] = None) -> torch.Tensor:
-  """q [B,T,N,H], k/v [B,S,NKV,H] -> [B,T,N,H]."""
+                    scale: Optional[float] = None,
+                    q_segment_ids: Optional[torch.Tensor] = None,
+                    k_segment_ids: Optional[torch.Tensor] = None
+                    ) -> torch.Tensor:
+  """q [B,T,N,H], k/v [B,S,NKV,H] -> [B,T,N,H]. Optional packed-input
+  segment ids [B,T]/[B,S]: attention is blocked across segments
+  (reference PackSequences segment_ids consumed by the mask path)."""
   if scale is None:
     scale = 1.0 / math.sqrt(q.shape[-1])
   if klen is not None:
     klen = klen.to(torch.int32).contiguous()
+  if q_segment_ids is not None:
+    q_segment_ids = q_segment_ids.to(torch.int32).contiguous()
+    k_segment_ids = k_segment_ids.to(torch.int32).contiguous()
   if q.is_cuda:
     orig = q.dtype
     out = _FlashAttnFn.apply(
         q.to(torch.bfloat16).contiguous(), k.to(torch.bfloat16).contiguous(),
-        v.to(torch.bfloat16).contiguous(), klen, bias, win_l, win_r,
-        bias_clip, scale)
+        v.to(torch.bfloat16).contiguous(), klen, bias, q_segment_ids,
+        k_segment_ids, win_l, win_r, bias_clip, scale)
     return out.to(orig) if orig != torch.bfloat16 else out
-  out = _ref_attention(q, k, v, klen, bias, win_l, win_r, bias_clip, scale)
+  out = _ref_attention(q, k, v, klen, bias, win_l, win_r, bias_clip, scale,
+                       q_segment_ids, k_segment_ids)
   return out.to(q.dtype)

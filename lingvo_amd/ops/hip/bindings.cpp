@@ -21,6 +21,8 @@ std::vector<torch::Tensor> fa_fwd(torch::Tensor q, torch::Tensor k,
                                   torch::Tensor v,
                                   c10::optional<torch::Tensor> klen,
                                   c10::optional<torch::Tensor> bias,
+                                  c10::optional<torch::Tensor> qseg,
+                                  c10::optional<torch::Tensor> kseg,
                                   int64_t win_l, int64_t win_r,
                                   int64_t bias_clip, double scale);
 std::vector<torch::Tensor> fa_bwd(torch::Tensor dout, torch::Tensor q,
@@ -28,6 +30,8 @@ std::vector<torch::Tensor> fa_bwd(torch::Tensor dout, torch::Tensor q,
                                   torch::Tensor o, torch::Tensor lse,
                                   c10::optional<torch::Tensor> klen,
                                   c10::optional<torch::Tensor> bias,
+                                  c10::optional<torch::Tensor> qseg,
+                                  c10::optional<torch::Tensor> kseg,
                                   bool bias_grad, int64_t win_l,
                                   int64_t win_r, int64_t bias_clip,
                                   double scale);

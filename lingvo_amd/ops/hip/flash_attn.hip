@@ -128,6 +128,8 @@ __global__ __launch_bounds__(FWD_BLOCK) void fa_fwd_kernel(
     const unsigned short* __restrict__ v,
     const int* __restrict__ klen_ptr,          // [B] or null
     const unsigned short* __restrict__ bias,   // [N][2C+1] or null
+    const int* __restrict__ qseg,              // [B][T] or null (packed)
+    const int* __restrict__ kseg,              // [B][S] or null
     unsigned short* __restrict__ o, float* __restrict__ lse, int B, int T,
     int S, int N, int NKV, int win_l, int win_r, int bias_clip, float scale) {
   constexpr int ROWB = H * 2;
@@ -225,6 +227,9 @@ __global__ __launch_bounds__(FWD_BLOCK) void fa_fwd_kernel(
               bias[(long)n * (2 * bias_clip + 1) + d + bias_clip]);
         }
         if (!visible(qrow, kcol, klen, win_l, win_r) || qrow >= T)
+          val = NEG_INF;
+        if (qseg && kcol < S && qrow < T &&
+            qseg[(long)b * T + qrow] != kseg[(long)b * S + kcol])
           val = NEG_INF;
         s[nf][r] = val;
       }
@@ -349,7 +354,9 @@ __global__ __launch_bounds__(NWB * WAVE_SIZE) void fa_bwd_kernel(
     const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
     const unsigned short* __restrict__ v, const float* __restrict__ lse,
     const float* __restrict__ delta, const int* __restrict__ klen_ptr,
-    const unsigned short* __restrict__ bias, float* __restrict__ dq_acc,
+    const unsigned short* __restrict__ bias,
+    const int* __restrict__ qseg, const int* __restrict__ kseg,
+    float* __restrict__ dq_acc,
     unsigned short* __restrict__ dk, unsigned short* __restrict__ dv,
     float* __restrict__ dbias, int B, int T, int S, int N, int NKV,
     int win_l, int win_r, int bias_clip, float scale) {
@@ -469,6 +476,9 @@ __global__ __launch_bounds__(NWB * WAVE_SIZE) void fa_bwd_kernel(
             val += bf16_bits_to_float(bias[(long)n * nbias + d + bias_clip]);
           }
           bool vis = visible(qcol, key, klen, win_l, win_r) && qcol < T;
+          if (vis && qseg &&
+              qseg[(long)b * T + qcol] != kseg[(long)b * S + key])
+            vis = false;
           float l = lse_s[nf * 16 + cl];
           pt[nf][r] =
               (vis && l > NEG_INF * 0.5f) ? __expf(val - l) : 0.f;
@@ -655,6 +665,8 @@ std::vector<torch::Tensor> fa_fwd(torch::Tensor q, torch::Tensor k,
                                   torch::Tensor v,
                                   c10::optional<torch::Tensor> klen,
                                   c10::optional<torch::Tensor> bias,
+                                  c10::optional<torch::Tensor> qseg,
+                                  c10::optional<torch::Tensor> kseg,
                                   int64_t win_l, int64_t win_r,
                                   int64_t bias_clip, double scale) {
   check_btnh(q, "q");
@@ -671,6 +683,8 @@ std::vector<torch::Tensor> fa_fwd(torch::Tensor q, torch::Tensor k,
   const int* klp = klen.has_value() ? klen->data_ptr<int>() : nullptr;
   const unsigned short* bp =
       bias.has_value() ? (const unsigned short*)bias->data_ptr() : nullptr;
+  const int* qsp = qseg.has_value() ? qseg->data_ptr<int>() : nullptr;
+  const int* ksp = kseg.has_value() ? kseg->data_ptr<int>() : nullptr;
   // Big KV tiles amortize staging barriers, but waste work on short
   // sequences (masked overhang): pick by S.
   const int ktf = S >= 512 ? 128 : 64;
@@ -682,18 +696,18 @@ std::vector<torch::Tensor> fa_fwd(torch::Tensor q, torch::Tensor k,
                        shmem, stream,                                       \
                      (const unsigned short*)q.data_ptr(),                   \
                      (const unsigned short*)k.data_ptr(),                   \
-                     (const unsigned short*)v.data_ptr(), klp, bp,          \
-                     (unsigned short*)o.data_ptr(), lse.data_ptr<float>(),  \
-                     B, T, S, N, NKV, (int)win_l, (int)win_r,               \
-                     (int)bias_clip, (float)scale)
+                     (const unsigned short*)v.data_ptr(), klp, bp, qsp,     \
+                     ksp, (unsigned short*)o.data_ptr(),                    \
+                     lse.data_ptr<float>(), B, T, S, N, NKV, (int)win_l,    \
+                     (int)win_r, (int)bias_clip, (float)scale)
 #define FA_FWD64(HH)                                                        \
   hipLaunchKernelGGL((fa_fwd_kernel<HH, 64>), grid, dim3(FWD_BLOCK),        \
                      shmem, stream, (const unsigned short*)q.data_ptr(),    \
                      (const unsigned short*)k.data_ptr(),                   \
-                     (const unsigned short*)v.data_ptr(), klp, bp,          \
-                     (unsigned short*)o.data_ptr(), lse.data_ptr<float>(),  \
-                     B, T, S, N, NKV, (int)win_l, (int)win_r,               \
-                     (int)bias_clip, (float)scale)
+                     (const unsigned short*)v.data_ptr(), klp, bp, qsp,     \
+                     ksp, (unsigned short*)o.data_ptr(),                    \
+                     lse.data_ptr<float>(), B, T, S, N, NKV, (int)win_l,    \
+                     (int)win_r, (int)bias_clip, (float)scale)
   if (H == 64) {
     FA_FWD(64);
     else FA_FWD64(64);
@@ -711,6 +725,8 @@ std::vector<torch::Tensor> fa_bwd(torch::Tensor dout, torch::Tensor q,
                                   torch::Tensor o, torch::Tensor lse,
                                   c10::optional<torch::Tensor> klen,
                                   c10::optional<torch::Tensor> bias,
+                                  c10::optional<torch::Tensor> qseg,
+                                  c10::optional<torch::Tensor> kseg,
                                   bool bias_grad, int64_t win_l,
                                   int64_t win_r, int64_t bias_clip,
                                   double scale) {
@@ -753,6 +769,8 @@ std::vector<torch::Tensor> fa_bwd(torch::Tensor dout, torch::Tensor q,
   const int* klp = klen.has_value() ? klen->data_ptr<int>() : nullptr;
   const unsigned short* bp =
       bias.has_value() ? (const unsigned short*)bias->data_ptr() : nullptr;
+  const int* qsp = qseg.has_value() ? qseg->data_ptr<int>() : nullptr;
+  const int* ksp = kseg.has_value() ? kseg->data_ptr<int>() : nullptr;
   bool bg = bias.has_value() && bias_grad;
 
   // 128-key tiles with 8 waves (q-tile staging amortized 2x). H=128
@@ -774,7 +792,8 @@ std::vector<torch::Tensor> fa_bwd(torch::Tensor dout, torch::Tensor q,
       (const unsigned short*)q.data_ptr(),                                   \
       (const unsigned short*)k.data_ptr(),                                   \
       (const unsigned short*)v.data_ptr(), lse.data_ptr<float>(),            \
-      delta.data_ptr<float>(), klp, bp, dq_acc.data_ptr<float>(),            \
+      delta.data_ptr<float>(), klp, bp, qsp, ksp,                            \
+      dq_acc.data_ptr<float>(),                                              \
       (unsigned short*)dk_t.data_ptr(), (unsigned short*)dv_t.data_ptr(),    \
       dbias_t.numel() ? dbias_t.data_ptr<float>() : nullptr, B, T, S, N,     \
       NKV, (int)win_l, (int)win_r, (int)bias_clip, (float)scale)
