@@ -249,7 +249,13 @@ __global__ __launch_bounds__(FWD_BLOCK) void fa_fwd_kernel(
     float alpha[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
+      // defer-max (guide T13): keep the old running max when growth is
+      // < 8 — P is then bounded by e^8, fine in fp32 accumulation, and
+      // the O-rescale + exp updates are skipped.
       float m_new = fmaxf(m_run[r], rowmax[r]);
+      if (m_run[r] > NEG_INF * 0.5f && rowmax[r] <= m_run[r] + 8.f) {
+        m_new = m_run[r];
+      }
       alpha[r] = (m_run[r] == NEG_INF) ? 0.f : __expf(m_run[r] - m_new);
       m_run[r] = m_new;
       float sum = 0.f;
@@ -263,9 +269,13 @@ __global__ __launch_bounds__(FWD_BLOCK) void fa_fwd_kernel(
 #pragma unroll
       for (int off = 1; off < 16; off <<= 1) sum += __shfl_xor(sum, off);
       rowsum[r] = sum;
-      l_run[r] = l_run[r] * alpha[r] + sum;
+      if (alpha[r] != 1.f) {
+        l_run[r] = l_run[r] * alpha[r] + sum;
 #pragma unroll
-      for (int hf = 0; hf < HF; ++hf) acc_o[hf][r] *= alpha[r];
+        for (int hf = 0; hf < HF; ++hf) acc_o[hf][r] *= alpha[r];
+      } else {
+        l_run[r] += sum;
+      }
     }
 
     // P -> per-wave LDS (bf16, C layout -> row-major [16][KTF], swizzled).
