@@ -122,7 +122,7 @@ constexpr int FWD_NW = 8;                 // waves per fwd block
 constexpr int FWD_BLOCK = FWD_NW * WAVE_SIZE;
 constexpr int FQT = FWD_NW * 16;          // q rows per fwd block
 
-template <int H, int KTF>
+template <int H, int KTF, bool SEG>
 __global__ __launch_bounds__(FWD_BLOCK) void fa_fwd_kernel(
     const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
     const unsigned short* __restrict__ v,
@@ -228,9 +228,11 @@ __global__ __launch_bounds__(FWD_BLOCK) void fa_fwd_kernel(
         }
         if (!visible(qrow, kcol, klen, win_l, win_r) || qrow >= T)
           val = NEG_INF;
-        if (qseg && kcol < S && qrow < T &&
-            qseg[(long)b * T + qrow] != kseg[(long)b * S + kcol])
-          val = NEG_INF;
+        if (SEG) {
+          if (kcol < S && qrow < T &&
+              qseg[(long)b * T + qrow] != kseg[(long)b * S + kcol])
+            val = NEG_INF;
+        }
         s[nf][r] = val;
       }
     }
@@ -249,13 +251,7 @@ __global__ __launch_bounds__(FWD_BLOCK) void fa_fwd_kernel(
     float alpha[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      // defer-max (guide T13): keep the old running max when growth is
-      // < 8 — P is then bounded by e^8, fine in fp32 accumulation, and
-      // the O-rescale + exp updates are skipped.
       float m_new = fmaxf(m_run[r], rowmax[r]);
-      if (m_run[r] > NEG_INF * 0.5f && rowmax[r] <= m_run[r] + 8.f) {
-        m_new = m_run[r];
-      }
       alpha[r] = (m_run[r] == NEG_INF) ? 0.f : __expf(m_run[r] - m_new);
       m_run[r] = m_new;
       float sum = 0.f;
@@ -269,13 +265,9 @@ __global__ __launch_bounds__(FWD_BLOCK) void fa_fwd_kernel(
 #pragma unroll
       for (int off = 1; off < 16; off <<= 1) sum += __shfl_xor(sum, off);
       rowsum[r] = sum;
-      if (alpha[r] != 1.f) {
-        l_run[r] = l_run[r] * alpha[r] + sum;
+      l_run[r] = l_run[r] * alpha[r] + sum;
 #pragma unroll
-        for (int hf = 0; hf < HF; ++hf) acc_o[hf][r] *= alpha[r];
-      } else {
-        l_run[r] += sum;
-      }
+      for (int hf = 0; hf < HF; ++hf) acc_o[hf][r] *= alpha[r];
     }
 
     // P -> per-wave LDS (bf16, C layout -> row-major [16][KTF], swizzled).
@@ -702,7 +694,8 @@ std::vector<torch::Tensor> fa_fwd(torch::Tensor q, torch::Tensor k,
       (size_t)ktf * H * 2 + (size_t)H * ktf * 2 + FWD_NW * 16 * ktf * 2;
 #define FA_FWD(HH)                                                          \
   if (ktf == 128)                                                           \
-    hipLaunchKernelGGL((fa_fwd_kernel<HH, 128>), grid, dim3(FWD_BLOCK),     \
+    hipLaunchKernelGGL((fa_fwd_kernel<HH, 128, false>), grid,               \
+                       dim3(FWD_BLOCK),                                     \
                        shmem, stream,                                       \
                      (const unsigned short*)q.data_ptr(),                   \
                      (const unsigned short*)k.data_ptr(),                   \
@@ -711,14 +704,40 @@ std::vector<torch::Tensor> fa_fwd(torch::Tensor q, torch::Tensor k,
                      lse.data_ptr<float>(), B, T, S, N, NKV, (int)win_l,    \
                      (int)win_r, (int)bias_clip, (float)scale)
 #define FA_FWD64(HH)                                                        \
-  hipLaunchKernelGGL((fa_fwd_kernel<HH, 64>), grid, dim3(FWD_BLOCK),        \
+  hipLaunchKernelGGL((fa_fwd_kernel<HH, 64, false>), grid,                  \
+                     dim3(FWD_BLOCK),                                       \
                      shmem, stream, (const unsigned short*)q.data_ptr(),    \
                      (const unsigned short*)k.data_ptr(),                   \
                      (const unsigned short*)v.data_ptr(), klp, bp, qsp,     \
                      ksp, (unsigned short*)o.data_ptr(),                    \
                      lse.data_ptr<float>(), B, T, S, N, NKV, (int)win_l,    \
                      (int)win_r, (int)bias_clip, (float)scale)
-  if (H == 64) {
+#define FA_FWD_SEG(HH)                                                      \
+  if (ktf == 128)                                                           \
+    hipLaunchKernelGGL((fa_fwd_kernel<HH, 128, true>), grid,                \
+                       dim3(FWD_BLOCK), shmem, stream,                      \
+                       (const unsigned short*)q.data_ptr(),                 \
+                       (const unsigned short*)k.data_ptr(),                 \
+                       (const unsigned short*)v.data_ptr(), klp, bp, qsp,   \
+                       ksp, (unsigned short*)o.data_ptr(),                  \
+                       lse.data_ptr<float>(), B, T, S, N, NKV, (int)win_l,  \
+                       (int)win_r, (int)bias_clip, (float)scale);           \
+  else                                                                      \
+    hipLaunchKernelGGL((fa_fwd_kernel<HH, 64, true>), grid,                 \
+                       dim3(FWD_BLOCK), shmem, stream,                      \
+                       (const unsigned short*)q.data_ptr(),                 \
+                       (const unsigned short*)k.data_ptr(),                 \
+                       (const unsigned short*)v.data_ptr(), klp, bp, qsp,   \
+                       ksp, (unsigned short*)o.data_ptr(),                  \
+                       lse.data_ptr<float>(), B, T, S, N, NKV, (int)win_l,  \
+                       (int)win_r, (int)bias_clip, (float)scale)
+  if (qsp != nullptr) {
+    if (H == 64) {
+      FA_FWD_SEG(64);
+    } else {
+      FA_FWD_SEG(128);
+    }
+  } else if (H == 64) {
     FA_FWD(64);
     else FA_FWD64(64);
   } else {
@@ -727,6 +746,7 @@ std::vector<torch::Tensor> fa_fwd(torch::Tensor q, torch::Tensor k,
   }
 #undef FA_FWD
 #undef FA_FWD64
+#undef FA_FWD_SEG
   return {o, lse};
 }
 
