@@ -436,3 +436,112 @@ class UniformLabelSmoother(BaseLayer):
         dtype=torch.float32, device=target_ids.device)
     probs.scatter_(-1, target_ids.long().unsqueeze(-1), 1.0 - p.uncertainty)
     return probs
+
+
+class HighwaySkipLayer(BaseLayer):
+  """Highway connection y = g*x_trans + (1-g)*x (reference layers.py:5461)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('input_dim', 0, 'Dim.')
+    p.Define('batch_norm', False, 'Unused (parity).')
+    p.Define('carry_bias_init', -1.0, 'Carry gate bias init.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    self.CreateVariable('w_t', py_utils.WeightParams(
+        [p.input_dim, p.input_dim], p.params_init, p.dtype))
+    self.CreateVariable('b_t', py_utils.WeightParams(
+        [p.input_dim], py_utils.WeightInit.Constant(p.carry_bias_init),
+        p.dtype))
+    self.CreateVariable('w_h', py_utils.WeightParams(
+        [p.input_dim, p.input_dim], p.params_init, p.dtype))
+    self.CreateVariable('b_h', py_utils.WeightParams(
+        [p.input_dim], py_utils.WeightInit.Constant(0.0), p.dtype))
+
+  def FProp(self, theta: NestedMap, x: torch.Tensor,
+            transformed: Optional[torch.Tensor] = None) -> torch.Tensor:
+    if transformed is None:
+      transformed = torch.relu(py_utils.MatmulBias(x, theta.w_h, theta.b_h))
+    gate = torch.sigmoid(py_utils.MatmulBias(x, theta.w_t, theta.b_t))
+    return gate * transformed + (1.0 - gate) * x
+
+
+class GluLayer(BaseLayer):
+  """Gated linear unit block with residual (reference layers.py:6124)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('input_dim', 0, 'Dim.')
+    p.Define('output_dim', 0, 'Output (0 = input_dim).')
+    p.Define('dropout_prob', 0.0, 'Dropout.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    out = p.output_dim or p.input_dim
+    self.CreateChild('ln', LayerNorm.Params().Set(input_dim=p.input_dim))
+    self.CreateVariable('w', py_utils.WeightParams(
+        [p.input_dim, 2 * out], p.params_init, p.dtype))
+    self.CreateVariable('b', py_utils.WeightParams(
+        [2 * out], py_utils.WeightInit.Constant(0.0), p.dtype))
+
+  def FProp(self, theta: NestedMap, x: torch.Tensor,
+            paddings: Optional[torch.Tensor] = None) -> torch.Tensor:
+    p = self.p
+    h = self.ln.FProp(theta.ln, x)
+    h = py_utils.MatmulBias(h, theta.w, theta.b)
+    a, g = h.chunk(2, dim=-1)
+    out = a * torch.sigmoid(g)
+    if p.dropout_prob and not self.do_eval:
+      out = py_utils.DeterministicDropout(out, 1.0 - p.dropout_prob)
+    if (p.output_dim or p.input_dim) == p.input_dim:
+      out = out + x
+    if paddings is not None:
+      out = py_utils.ApplyPadding(paddings, out)
+    return out
+
+
+class GradNormTracker(BaseLayer):
+  """Tracks a running log-grad-norm and flags outlier steps
+  (reference layers.py:5590; the Learner consumes the verdict to skip)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('decay', 0.995, 'EMA decay of log grad norm stats.')
+    p.Define('clip_threshold', 4.0, 'Allowed stds above the mean.')
+    p.Define('grad_norm_clip_cap_min', 0.0, 'Floor for the cap.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    self.register_buffer('log_mean', torch.zeros(()))
+    self.register_buffer('log_var', torch.ones(()))
+    self.register_buffer('count', torch.zeros(()))
+
+  def FProp(self, theta: NestedMap, grad_norm: torch.Tensor) -> bool:
+    """Returns True if this step's grad norm is acceptable; updates
+    running stats only on accepted steps."""
+    p = self.p
+    log_n = torch.log(grad_norm.detach().float().clamp_min(1e-10))
+    if float(self.count) < 10:
+      ok = True
+    else:
+      std = self.log_var.clamp_min(1e-6).sqrt()
+      cap = self.log_mean + p.clip_threshold * std
+      ok = bool(log_n <= torch.maximum(
+          cap, torch.tensor(math.log(max(p.grad_norm_clip_cap_min,
+                                         1e-10)))))
+    if ok:
+      d = p.decay
+      delta = log_n - self.log_mean
+      self.log_mean.mul_(d).add_((1 - d) * log_n)
+      self.log_var.mul_(d).add_((1 - d) * delta * delta)
+      self.count += 1
+    return ok

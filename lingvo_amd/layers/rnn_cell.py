@@ -179,3 +179,43 @@ class GRUCell(RNNCell):
     if 'padding' in inputs and inputs.padding is not None:
       m1 = m1 * (1 - inputs.padding) + state0.m * inputs.padding
     return NestedMap(m=m1)
+
+
+class SRUCell(RNNCell):
+  """Simple Recurrent Unit (reference rnn_cell.py:2174): the matmuls
+  depend only on the input, so they batch across time; only cheap
+  elementwise recurrences remain sequential."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    h = p.num_output_nodes
+    self.CreateVariable('w', py_utils.WeightParams(
+        [p.num_input_nodes, 4 * h], p.params_init, p.dtype))
+    self.CreateVariable('b', py_utils.WeightParams(
+        [4 * h], py_utils.WeightInit.Constant(0.0), p.dtype))
+
+  def InitState(self, batch, device, dtype) -> NestedMap:
+    h = self.p.num_output_nodes
+    return NestedMap(c=torch.zeros(batch, h, device=device, dtype=dtype),
+                     m=torch.zeros(batch, h, device=device, dtype=dtype))
+
+  def FProp(self, theta: NestedMap, state0: NestedMap,
+            inputs: NestedMap) -> NestedMap:
+    h = self.p.num_output_nodes
+    proj = torch.matmul(inputs.act, theta.w) + theta.b
+    x_t, f_t, r_t, x2_t = proj.split([h, h, h, h], dim=-1)
+    f = torch.sigmoid(f_t)
+    r = torch.sigmoid(r_t)
+    c1 = f * state0.c + (1 - f) * x_t
+    m1 = r * torch.tanh(c1) + (1 - r) * x2_t
+    if 'padding' in inputs and inputs.padding is not None:
+      pad = inputs.padding
+      c1 = c1 * (1 - pad) + state0.c * pad
+      m1 = m1 * (1 - pad) + state0.m * pad
+    return NestedMap(c=c1, m=m1)
