@@ -58,8 +58,10 @@ def main():
   sync = ddp.GradSync(task) if world > 1 else None
   finalize = sync.Finalize if sync else None
 
-  # Pre-generate a handful of synthetic batches on device.
+  # Pre-generate a handful of synthetic batches on device. Each DP rank
+  # draws a distinct stream (weak scaling: different data per rank).
   gen = task.input_generator
+  gen._batch_count = rank * 1009
   batches = []
   for _ in range(4):
     b = gen.GetPreprocessedInputBatch()
