@@ -23,6 +23,7 @@ _PARAM_MODULES = [
     'lingvo_amd.models.params.mt.wmt14_en_de',
     'lingvo_amd.models.params.punctuator.codelab',
     'lingvo_amd.models.params.milan.cxc',
+    'lingvo_amd.models.params.car.kitti',
 ]
 
 

@@ -1,0 +1,35 @@
+"""Registered car params (reference lingvo/tasks/car/params — StarNet/
+pillars on KITTI; synthetic scenes here, no network for the dataset)."""
+
+from __future__ import annotations
+
+import torch
+
+from lingvo_amd.core import learner as learner_lib
+from lingvo_amd.core import optimizer as optimizer_lib
+from lingvo_amd.core import registry
+from lingvo_amd.core.base_model_params import SingleTaskModelParams
+from lingvo_amd.models import car as car_model
+
+
+@registry.RegisterSingleTaskModel
+class StarNetPillars(SingleTaskModelParams):
+
+  def Train(self):
+    return car_model.SyntheticPointCloudInput.Params().Set(
+        name='train', batch_size=4)
+
+  def Dev(self):
+    return self.Train().Set(name='dev')
+
+  def Test(self):
+    return self.Train().Set(name='test')
+
+  def Task(self):
+    p = car_model.PillarsModel.Params().Set(name='pillars')
+    p.fprop_dtype = torch.float32  # small model; fp32 everywhere
+    p.train.learner = learner_lib.Learner.Params().Set(
+        learning_rate=1e-3,
+        optimizer=optimizer_lib.Adam.Params(),
+        clip_gradient_norm_to_value=5.0)
+    return p

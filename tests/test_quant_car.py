@@ -67,3 +67,23 @@ def test_furthest_point_sampling():
   assert len(set(idx.tolist())) == 3
   # the far points should be picked
   assert 2 in idx.tolist() and 3 in idx.tolist()
+
+
+def test_pillars_model_train_and_decode():
+  from lingvo_amd.core import registry
+  p = registry.GetParams('car.kitti.StarNetPillars', 'Train')
+  p.task.random_seed = 6
+  p.task.grid_size = 32
+  p.input.Set(batch_size=2, num_points=512)
+  task = p.Instantiate().GetTask()
+  losses = []
+  for _ in range(3):
+    m = task.TrainStep(task.GetInputBatch())
+    losses.append(float(m['loss'][0]))
+  assert all(l == l for l in losses)
+  task.eval()
+  out = task.Decode(task.GetInputBatch())
+  assert out.ap.shape == (2,)
+  dm = task.CreateDecoderMetrics()
+  task.PostProcessDecodeOut(out, dm)
+  assert 0.0 <= dm.ap3d.value <= 1.0
