@@ -39,6 +39,9 @@ def main():
   from lingvo_amd.parallel import ddp
 
   world = int(os.environ.get('WORLD_SIZE', '1'))
+  # Keep the NCCL watchdog from probing streams during hipGraph capture
+  # (capture falls back to eager if it still objects).
+  os.environ.setdefault('TORCH_NCCL_ASYNC_ERROR_HANDLING', '0')
   rank = ddp.InitDistributed()
   local_rank = int(os.environ.get('LOCAL_RANK', '0'))
   has_gpu = torch.cuda.is_available()

@@ -65,3 +65,46 @@ def test_inference_export_and_predictor(tmp_path):
   assert out.hyps.shape[0] == 2
   enc = pred.Run('encode', src_inputs=src, paddings=pad)
   assert enc.encoded.shape[-1] == 64
+
+
+def test_runner_retries_transient_errors(tmp_path):
+  """_RunLoop retry policy (reference base_runner.py:399-527)."""
+  from lingvo_amd.runtime.runners import BaseRunner
+  from lingvo_amd.core import registry
+
+  model_p = registry.GetParams('image.mnist.LeNet5', 'Train')
+  r = BaseRunner(model_p, str(tmp_path), 'test', device='cpu',
+                 max_retries=3)
+  calls = []
+
+  def flaky():
+    calls.append(1)
+    if len(calls) < 3:
+      raise ConnectionError('transient')
+
+  import time as _time
+  orig_sleep = _time.sleep
+  _time.sleep = lambda s: None  # no backoff wait in tests
+  try:
+    r._RunLoop(flaky)
+  finally:
+    _time.sleep = orig_sleep
+  assert len(calls) == 3
+
+  # fatal errors are not retried
+  def fatal():
+    raise FloatingPointError('nan loss')
+
+  with pytest.raises(FloatingPointError):
+    r._RunLoop(fatal)
+
+  # retry budget exhausts
+  def always():
+    raise TimeoutError('down')
+
+  _time.sleep = lambda s: None
+  try:
+    with pytest.raises(TimeoutError):
+      r._RunLoop(always)
+  finally:
+    _time.sleep = orig_sleep
