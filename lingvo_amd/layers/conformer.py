@@ -234,16 +234,27 @@ class ConvSubsampling(BaseLayer):
     """3x3 strided conv as im2col + hipBLASLt GEMM. MIOpen's algorithm
     search can fall back to a naive NCHW kernel for these shapes on
     gfx950 (observed ~1000x regression under rocprof); the unfold+GEMM
-    path always lands on Tensile."""
+    path always lands on Tensile.
+
+    The batch is chunked so the im2col buffer stays under 2^31 BYTES:
+    beyond that, 32-bit byte offsets overflow in the col2im backward
+    (observed as a GPU write fault at per-buffer >= 2.2 GB)."""
     o = w_oihw.shape[0]
-    bsz = x.shape[0]
+    bsz, cin = x.shape[0], x.shape[1]
     hout = (x.shape[2] + 2 * pad - 3) // stride + 1
     wout = (x.shape[3] + 2 * pad - 3) // stride + 1
-    cols = F.unfold(x, kernel_size=3, stride=stride, padding=pad)
-    out = torch.baddbmm(
-        bias.reshape(1, o, 1),
-        w_oihw.reshape(1, o, -1).expand(bsz, -1, -1), cols)
-    return out.reshape(bsz, o, hout, wout)
+    cols_bytes_per_ex = cin * 9 * hout * wout * x.element_size()
+    max_chunk = max(1, int((2 ** 31 - 2 ** 27) // cols_bytes_per_ex))
+    outs = []
+    for s in range(0, bsz, max_chunk):
+      xc = x[s:s + max_chunk]
+      bc = xc.shape[0]
+      cols = F.unfold(xc, kernel_size=3, stride=stride, padding=pad)
+      out = torch.baddbmm(
+          bias.reshape(1, o, 1),
+          w_oihw.reshape(1, o, -1).expand(bc, -1, -1), cols)
+      outs.append(out.reshape(bc, o, hout, wout))
+    return outs[0] if len(outs) == 1 else torch.cat(outs, dim=0)
 
   def FProp(self, theta: NestedMap, inputs: torch.Tensor,
             paddings: torch.Tensor):
