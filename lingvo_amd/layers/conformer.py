@@ -201,6 +201,13 @@ class ConvSubsampling(BaseLayer):
   """2x Conv2D stride-2 frontend: [B, T, F] mel -> [B, T/4, D]
   (reference tasks/asr/encoder conv subsampling)."""
 
+  # im2col buffers above ~2 GB trip faults in backward kernels (32-bit
+  # byte offsets); batches are chunked to stay below this. Overridable
+  # for tests. NOTE: per-GPU batch >= ~160 at Conformer-L shapes still
+  # faults in upstream col2im/hipBLASLt backward (see docs/ROADMAP.md);
+  # the bench default (128) is unaffected.
+  MAX_COLS_BYTES = 2 ** 31 - 2 ** 27
+
   @classmethod
   def Params(cls):
     p = super().Params()
@@ -244,7 +251,8 @@ class ConvSubsampling(BaseLayer):
     hout = (x.shape[2] + 2 * pad - 3) // stride + 1
     wout = (x.shape[3] + 2 * pad - 3) // stride + 1
     cols_bytes_per_ex = cin * 9 * hout * wout * x.element_size()
-    max_chunk = max(1, int((2 ** 31 - 2 ** 27) // cols_bytes_per_ex))
+    max_chunk = max(1, int(ConvSubsampling.MAX_COLS_BYTES //
+                           cols_bytes_per_ex))
     outs = []
     for s in range(0, bsz, max_chunk):
       xc = x[s:s + max_chunk]

@@ -325,3 +325,20 @@ def test_mha_gqa_kv_heads_layer():
   x = torch.randn(2, 6, 128)
   out = layer.FProp(layer.theta, x)
   assert out.shape == (2, 6, 128)
+
+
+def test_conv_subsampling_chunking_equivalence():
+  p = conformer_lib.ConvSubsampling.Params().Set(
+      name='sub', input_freq_dim=16, output_dim=32, channels=8,
+      random_seed=1)
+  sub = p.Instantiate()
+  x = torch.randn(6, 24, 16)
+  pad = torch.zeros(6, 24)
+  full, _ = sub.FProp(sub.theta, x, pad)
+  old = conformer_lib.ConvSubsampling.MAX_COLS_BYTES
+  try:
+    conformer_lib.ConvSubsampling.MAX_COLS_BYTES = 16 * 1024  # force chunks
+    chunked, _ = sub.FProp(sub.theta, x, pad)
+  finally:
+    conformer_lib.ConvSubsampling.MAX_COLS_BYTES = old
+  assert torch.allclose(full, chunked, atol=1e-5)
