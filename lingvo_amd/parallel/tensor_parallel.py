@@ -206,3 +206,120 @@ def ShardTransformerStackForTp(stack_params, tp_group=None):
       residual_weight=ff.residual_weight, tp_group=tp_group)
   tpl.tr_fflayer_tpl = new_ff
   return stack_params
+
+
+class TpMultiHeadedAttention(BaseLayer):
+  """Head-sharded self-attention: each TP rank owns N/W query heads
+  (and their KV heads), runs the flash kernel on its local heads, and
+  the row-parallel output projection all-reduces the result — the
+  explicit lowering of the reference's attention sharding
+  `wq:[M,N,H] -> (M,N)` over the mesh (gshard_builder.py DenseBuilder;
+  tasks/lm/README.md:100-115). Drop-in for MultiHeadedAttention.FProp.
+  """
+
+  @classmethod
+  def Params(cls):
+    from lingvo_amd.layers.attention import MultiHeadedAttention
+    p = MultiHeadedAttention.Params()
+    p.cls = cls
+    p.Define('tp_group', None, 'TP process group.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    world, rank = _TpInfo(p.tp_group)
+    n = p.num_heads
+    assert n % world == 0, 'num_heads must divide TP world size'
+    nkv_full = p.num_kv_heads or n
+    assert nkv_full % world == 0
+    h = p.dim_per_head or (p.hidden_dim or p.input_dim) // n
+    self._n = n // world
+    self._nkv = nkv_full // world
+    self._h = h
+    self._world, self._rank = world, rank
+    d = p.input_dim
+    # Full-matrix seeded init, sliced by head block, so TP=k matches
+    # TP=1 numerics exactly.
+    self.CreateVariable('qkv_w_full_seeded', py_utils.WeightParams(
+        [d, (n + 2 * nkv_full) * h], p.params_init, p.dtype))
+    with torch.no_grad():
+      w = self.qkv_w_full_seeded
+      q_w = w[:, :n * h].reshape(d, world, self._n * h)[:, rank]
+      k_w = w[:, n * h:(n + nkv_full) * h].reshape(
+          d, world, self._nkv * h)[:, rank]
+      v_w = w[:, (n + nkv_full) * h:].reshape(
+          d, world, self._nkv * h)[:, rank]
+      shard = torch.cat([q_w, k_w, v_w], dim=1).clone()
+    del self._parameters['qkv_w_full_seeded']
+    self.register_parameter('qkv_w', torch.nn.Parameter(shard))
+    if p.use_bias:
+      self.CreateVariable('qkv_b', py_utils.WeightParams(
+          [(self._n + 2 * self._nkv) * h],
+          py_utils.WeightInit.Constant(0.0), p.dtype))
+    self.CreateVariable('post_w_full_seeded', py_utils.WeightParams(
+        [n * h, d], p.params_init, p.dtype))
+    with torch.no_grad():
+      pw = self.post_w_full_seeded.reshape(world, self._n * h, d)[rank]
+      pshard = pw.clone()
+    del self._parameters['post_w_full_seeded']
+    self.register_parameter('post_w', torch.nn.Parameter(pshard))
+    if p.use_bias:
+      # Bias added once after the all-reduce (owned by every rank but
+      # scaled so the sum is applied once).
+      self.CreateVariable('post_b', py_utils.WeightParams(
+          [d], py_utils.WeightInit.Constant(0.0), p.dtype))
+    if p.rel_pos_bias:
+      self.CreateVariable('rel_bias_full_seeded', py_utils.WeightParams(
+          [n, 2 * p.rel_pos_clip + 1], py_utils.WeightInit.Constant(0.0),
+          p.dtype))
+      with torch.no_grad():
+        rb = self.rel_bias_full_seeded.reshape(
+            world, self._n, -1)[rank].clone()
+      del self._parameters['rel_bias_full_seeded']
+      self.register_parameter('rel_bias', torch.nn.Parameter(rb))
+
+  def FProp(self, theta: NestedMap, query_vec: torch.Tensor,
+            paddings=None, segment_ids=None) -> torch.Tensor:
+    from lingvo_amd.ops import flash_attn
+    p = self.p
+    n, nkv, h = self._n, self._nkv, self._h
+    b, t = query_vec.shape[0], query_vec.shape[1]
+    x = _CopyToTp.apply(query_vec, p.tp_group)
+    qkv = py_utils.MatmulBias(x, theta.qkv_w,
+                              theta.qkv_b if p.use_bias else None)
+    q, k, v = qkv.split([n * h, nkv * h, nkv * h], dim=-1)
+    q = q.reshape(b, t, n, h)
+    k = k.reshape(b, t, nkv, h)
+    v = v.reshape(b, t, nkv, h)
+    klen = None
+    if paddings is not None and segment_ids is None:
+      klen = py_utils.LengthsFromPaddings(paddings).to(torch.int32)
+    bias = theta.rel_bias if p.rel_pos_bias else None
+    win_r = 0 if p.causal else p.right_context
+    out = flash_attn.flash_attention(
+        q, k, v, klen, bias, p.left_context, win_r, p.rel_pos_clip,
+        q_segment_ids=segment_ids, k_segment_ids=segment_ids)
+    ctx = out.reshape(b, t, n * h)
+    post = torch.matmul(ctx, theta.post_w)
+    post = _ReduceFromTp.apply(post, p.tp_group)
+    if p.use_bias:
+      post = post + theta.post_b
+    if paddings is not None:
+      post = py_utils.ApplyPadding(paddings, post)
+    return post
+
+
+def ShardAttentionForTp(stack_params, tp_group=None):
+  """Planner: swaps the stack's attention template for the TP variant
+  (composes with ShardTransformerStackForTp for the FFN)."""
+  atten = stack_params.transformer_tpl.tr_atten_tpl.atten_tpl
+  new_p = TpMultiHeadedAttention.Params()
+  for name, _ in new_p.IterParams():
+    if name in ('cls', 'tp_group'):
+      continue
+    if name in atten:
+      setattr(new_p, name, atten.Get(name))
+  new_p.tp_group = tp_group
+  stack_params.transformer_tpl.tr_atten_tpl.atten_tpl = new_p
+  return stack_params

@@ -77,3 +77,68 @@ def test_shard_planner_rewrites_ffn():
   x = torch.randn(2, 4, 16)
   out = stack.FProp(stack.theta, x, torch.zeros(2, 4))
   assert out.shape == x.shape
+
+
+def _tp_attn_params(seed=13):
+  return tp.TpMultiHeadedAttention.Params().Set(
+      name='tpa', input_dim=128, hidden_dim=128, num_heads=2, causal=True,
+      rel_pos_bias=True, random_seed=seed)
+
+
+def _run_tp_attn(rank, world, port, results):
+  os.environ['MASTER_ADDR'] = '127.0.0.1'
+  os.environ['MASTER_PORT'] = str(port)
+  dist.init_process_group('gloo', rank=rank, world_size=world)
+  layer = _tp_attn_params().Instantiate()
+  g = torch.Generator().manual_seed(55)
+  x = torch.randn(2, 8, 128, generator=g, requires_grad=True)
+  out = layer.FProp(layer.theta, x, torch.zeros(2, 8))
+  out.sum().backward()
+  results[f'out{rank}'] = out.detach()
+  results[f'dx{rank}'] = x.grad.clone()
+  results[f'qkv_shape{rank}'] = tuple(layer.qkv_w.shape)
+  dist.destroy_process_group()
+
+
+def test_tp_attention_matches_single_process():
+  ctx = mp.get_context('spawn')
+  with ctx.Manager() as mgr:
+    results = mgr.dict()
+    procs = [ctx.Process(target=_run_tp_attn, args=(r, 2, 29545, results))
+             for r in range(2)]
+    for p in procs:
+      p.start()
+    for p in procs:
+      p.join(120)
+      assert p.exitcode == 0
+    out0, out1 = results['out0'], results['out1']
+    dx0 = results['dx0']
+    shard_shape = results['qkv_shape0']
+
+  # each rank holds 1 of 2 heads: qkv shard [(1+2)*64] cols
+  assert shard_shape == (128, 3 * 64)
+  # TP=1 reference
+  layer = _tp_attn_params().Instantiate()
+  g = torch.Generator().manual_seed(55)
+  x = torch.randn(2, 8, 128, generator=g, requires_grad=True)
+  ref = layer.FProp(layer.theta, x, torch.zeros(2, 8))
+  ref.sum().backward()
+  assert torch.allclose(out0, out1, atol=1e-5)
+  assert torch.allclose(out0, ref.detach(), atol=1e-4), \
+      (out0 - ref.detach()).abs().max()
+  assert torch.allclose(dx0, x.grad, atol=1e-4)
+
+
+def test_shard_attention_planner():
+  from lingvo_amd.layers import transformer as transformer_lib
+  sp = transformer_lib.StackedTransformerLayers.Params().Set(
+      name='s', model_dim=128, num_layers=1, num_heads=2, hidden_dim=256,
+      random_seed=2)
+  tp.ShardAttentionForTp(sp)
+  tp.ShardTransformerStackForTp(sp)
+  sp.transformer_tpl.tr_fflayer_tpl.input_dim = 128
+  sp.transformer_tpl.tr_fflayer_tpl.hidden_dim = 256
+  stack = sp.Instantiate()
+  x = torch.randn(2, 6, 128)
+  out = stack.FProp(stack.theta, x, torch.zeros(2, 6))
+  assert out.shape == x.shape
