@@ -125,9 +125,32 @@ class VocabFileTokenizer(BaseTokenizer):
 class WpmTokenizer(VocabFileTokenizer):
   """Greedy longest-match wordpiece (reference core/wpm_encoder.py).
   Word-internal continuation pieces carry no marker; word starts are
-  prefixed with '▁' (sentencepiece-style)."""
+  prefixed with '▁' (sentencepiece-style).
+
+  When the native extension is built, encoding runs in the C++
+  `WpmEncoder` (hip/wpm_tokenizer.cpp — GIL-released, multi-threaded
+  batch path, the MI355X-native stand-in for the reference's
+  tokenizer_ops_kernels.cc); the Python scan below is the fallback and
+  the numerics oracle for its tests."""
 
   WORD_MARK = '▁'
+
+  def __init__(self, params):
+    super().__init__(params)
+    self._native = None
+    from lingvo_amd.ops import _loader
+    ext = _loader.get_ext()
+    if ext is not None and hasattr(ext, 'WpmEncoder'):
+      pieces = [self._inv.get(i, '\0<gap>') for i in
+                range(max(self._inv) + 1)] if self._inv else []
+      self._native = ext.WpmEncoder(pieces, self.p.target_unk_id)
+
+  def EncodeBatch(self, lines: Sequence[str],
+                  num_threads: int = 4) -> List[List[int]]:
+    """Batch encode; C++ multi-threaded when the extension is built."""
+    if self._native is not None:
+      return self._native.encode_batch(list(lines), num_threads)
+    return [self._TokensToIds(l) for l in lines]
 
   def _EncodeWord(self, word: str) -> List[int]:
     p = self.p
@@ -148,6 +171,8 @@ class WpmTokenizer(VocabFileTokenizer):
     return pieces
 
   def _TokensToIds(self, text: str) -> List[int]:
+    if self._native is not None:
+      return list(self._native.encode(text))
     out = []
     for w in text.split():
       out.extend(self._EncodeWord(w))

@@ -118,3 +118,28 @@ def test_pack_sequences():
   sid = packed.src_segment_ids[0]
   if sid[0] != sid[-1] and sid[-1] > 0:
     assert not mask[0, 0, -1]
+
+
+def test_wpm_native_matches_python():
+  """C++ WpmEncoder == Python greedy longest-match (incl. UTF-8/unk)."""
+  from lingvo_amd.ops import _loader
+  ext = _loader.get_ext()
+  if ext is None or not hasattr(ext, 'WpmEncoder'):
+    pytest.skip('native extension not built')
+  vocab = ['<unk>', '<s>', '</s>', '▁the', '▁cat', '▁', '▁é', 'é',
+           'c', 'a', 't', 's', '日', '▁日本']
+  tok = tokenizers.WpmTokenizer.Params().Set(
+      name='w', tokens=vocab).Instantiate()
+  assert tok._native is not None
+  texts = ['the cats', 'écat 日本日 xyz', '  the\t日本 ', '', 'Ωcats']
+  py = tokenizers.WpmTokenizer.Params().Set(
+      name='w2', tokens=vocab).Instantiate()
+  py._native = None  # force the Python scan
+  for t in texts:
+    assert tok._TokensToIds(t) == py._TokensToIds(t), t
+  # batch path (threads) equals per-line
+  batch = tok.EncodeBatch(texts, num_threads=3)
+  assert [list(b) for b in batch] == [py._TokensToIds(t) for t in texts]
+  # native decode round-trip
+  ids = tok._TokensToIds('the cats')
+  assert tok._native.decode(ids) == 'the cats'
