@@ -229,6 +229,47 @@ class PositionalEmbeddingLayer(BaseLayer):
     return emb.to(self.fprop_dtype)
 
 
+class RotaryPositionalEmbeddingLayer(BaseLayer):
+  """Rotary position embedding (reference layers.py:3476
+  RotaryPositionalEmbeddingLayer; RoFormer arXiv:2104.09864).
+
+  Rotates each head-dim pair (x[2i], x[2i+1]) by pos * theta_i where
+  theta_i spans [1/min_timescale, 1/max_timescale] geometrically, so
+  q·k after rotation depends only on relative position.
+  """
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('embedding_dim', 0, 'Per-head dim H (must be even).')
+    p.Define('min_timescale', 1, 'Min timescale.')
+    p.Define('max_timescale', 10_000, 'Max timescale.')
+    return p
+
+  def _SinCos(self, seq_length, device, position=None):
+    p = self.p
+    half = p.embedding_dim // 2
+    frac = torch.arange(half, dtype=torch.float32, device=device) / half
+    timescale = p.min_timescale * (p.max_timescale / p.min_timescale) ** frac
+    if position is None:
+      position = torch.arange(seq_length, dtype=torch.float32,
+                              device=device)[None, :]
+    angles = position.float()[:, :, None] / timescale[None, None, :]
+    return torch.sin(angles), torch.cos(angles)  # [B|1, T, H/2]
+
+  def FProp(self, theta: NestedMap, inputs: torch.Tensor,
+            position: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """inputs [B, T, N, H]; optional position [B, T] (decode offsets)."""
+    h = inputs.shape[-1]
+    assert h == self.p.embedding_dim and h % 2 == 0, h
+    sin, cos = self._SinCos(inputs.shape[1], inputs.device, position)
+    sin = sin[:, :, None, :]  # broadcast over heads
+    cos = cos[:, :, None, :]
+    x1, x2 = inputs.float().chunk(2, dim=-1)
+    out = torch.cat([x1 * cos - x2 * sin, x2 * cos + x1 * sin], dim=-1)
+    return out.to(inputs.dtype)
+
+
 class LayerNorm(BaseLayer):
   """Layer normalization over the last dim (reference layers.py:4927).
 
