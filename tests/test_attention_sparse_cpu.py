@@ -1,0 +1,132 @@
+"""Chunkwise / routing / Performer attention tests."""
+
+import math
+
+import torch
+
+from lingvo_amd.layers import attention as attention_lib
+from lingvo_amd.layers import attention_sparse as asp
+
+
+def _mk(cls, **kw):
+  kwargs = dict(name='a', input_dim=32, hidden_dim=32, num_heads=2,
+                random_seed=21)
+  kwargs.update(kw)
+  layer = cls.Params().Set(**kwargs).Instantiate()
+  layer.eval()
+  return layer
+
+
+def test_chunkwise_matches_full_within_chunk():
+  """chunk_size >= S, left_chunks=0 == dense attention."""
+  layer = _mk(asp.ChunkwiseSelfAttention, chunk_size=16)
+  ref = _mk(attention_lib.MultiHeadedAttention)
+  ref.load_state_dict(layer.state_dict(), strict=False)
+  g = torch.Generator().manual_seed(1)
+  x = torch.randn(2, 10, 32, generator=g)
+  pad = torch.zeros(2, 10)
+  pad[1, 8:] = 1.0
+  out = layer.FProp(layer.theta, x, pad)
+  full = ref.FProp(ref.theta, x, pad)
+  assert (out - full).abs().max() < 1e-4
+
+
+def test_chunkwise_blocks_cross_chunk():
+  """With W=4, left_chunks=0, tokens in different chunks don't mix."""
+  layer = _mk(asp.ChunkwiseSelfAttention, chunk_size=4)
+  g = torch.Generator().manual_seed(2)
+  x = torch.randn(1, 8, 32, generator=g)
+  pad = torch.zeros(1, 8)
+  out1 = layer.FProp(layer.theta, x, pad)
+  x2 = x.clone()
+  x2[:, 6] += 10.0  # second chunk change
+  out2 = layer.FProp(layer.theta, x2, pad)
+  assert (out1[:, :4] - out2[:, :4]).abs().max() < 1e-5
+  assert (out1[:, 4:] - out2[:, 4:]).abs().max() > 1e-3
+
+
+def test_chunkwise_causal_with_left_chunks_matches_local():
+  """chunkwise causal with left_chunks=1 == local attention with
+  window [W + (pos within chunk)] — check vs dense masked reference."""
+  layer = _mk(asp.ChunkwiseSelfAttention, chunk_size=4, left_chunks=1,
+              causal=True)
+  g = torch.Generator().manual_seed(3)
+  x = torch.randn(2, 12, 32, generator=g)
+  pad = torch.zeros(2, 12)
+  out = layer.FProp(layer.theta, x, pad)
+
+  # dense reference with the exact chunkwise-causal mask
+  q, k, v = layer._Project(layer.theta, x)
+  qf, kf, vf = q.float(), k.float(), v.float()
+  logits = torch.einsum('btnh,bsnh->bnts', qf, kf) / math.sqrt(16)
+  tpos = torch.arange(12)
+  qc, kc = tpos[:, None] // 4, tpos[None, :] // 4
+  mask = (kc >= qc - 1) & (kc <= qc) & (tpos[None, :] <= tpos[:, None])
+  logits = logits.masked_fill(~mask[None, None], -1e30)
+  probs = torch.softmax(logits, -1)
+  ctx = torch.einsum('bnts,bsnh->btnh', probs, vf).reshape(2, 12, 32)
+  want = ctx @ layer.theta.post_w.float() + layer.theta.post_b.float()
+  assert (out - want.to(out.dtype)).abs().max() < 1e-3
+
+
+def test_routing_attention_runs_and_respects_padding():
+  layer = _mk(asp.RoutingAttention, num_clusters=2, atten_window=8)
+  g = torch.Generator().manual_seed(4)
+  x = torch.randn(2, 12, 32, generator=g, requires_grad=True)
+  pad = torch.zeros(2, 12)
+  pad[0, 10:] = 1.0
+  out = layer.FProp(layer.theta, x, pad)
+  assert out.shape == x.shape
+  assert out[0, 10:].abs().max() < 1e-6
+  out.sum().backward()
+  assert x.grad is not None
+  # window >= S and 1 cluster == dense attention
+  dense = _mk(asp.RoutingAttention, num_clusters=1, atten_window=12)
+  ref = _mk(attention_lib.MultiHeadedAttention)
+  ref.load_state_dict(
+      {k: v for k, v in dense.state_dict().items() if 'centroids' not in k},
+      strict=False)
+  o1 = dense.FProp(dense.theta, x.detach(), pad)
+  o2 = ref.FProp(ref.theta, x.detach(), pad)
+  assert (o1 - o2).abs().max() < 1e-4
+
+
+def test_routing_centroid_ema_updates():
+  layer = _mk(asp.RoutingAttention, num_clusters=2, atten_window=8,
+              decay=0.5)
+  layer.train()
+  before = layer.centroids.clone()
+  g = torch.Generator().manual_seed(5)
+  x = torch.randn(2, 12, 32, generator=g)
+  layer.FProp(layer.theta, x, torch.zeros(2, 12))
+  assert (layer.centroids - before).abs().max() > 1e-4
+
+
+def test_performer_approximates_softmax():
+  """With many features, FAVOR+ approaches exact softmax attention."""
+  layer = _mk(asp.PerformerAttention, num_random_features=2048)
+  ref = _mk(attention_lib.MultiHeadedAttention)
+  ref.load_state_dict(
+      {k: v for k, v in layer.state_dict().items() if 'proj_mat' not in k},
+      strict=False)
+  g = torch.Generator().manual_seed(6)
+  x = torch.randn(2, 8, 32, generator=g) * 0.3  # mild logits
+  pad = torch.zeros(2, 8)
+  out = layer.FProp(layer.theta, x, pad)
+  full = ref.FProp(ref.theta, x, pad)
+  rel = (out - full).norm() / full.norm()
+  assert rel < 0.08, rel.item()
+
+
+def test_performer_causal_prefix_property():
+  """Causal Performer outputs at t depend only on inputs <= t."""
+  layer = _mk(asp.PerformerAttention, num_random_features=64, causal=True)
+  g = torch.Generator().manual_seed(7)
+  x = torch.randn(1, 8, 32, generator=g)
+  pad = torch.zeros(1, 8)
+  out1 = layer.FProp(layer.theta, x, pad)
+  x2 = x.clone()
+  x2[:, 5] += 3.0
+  out2 = layer.FProp(layer.theta, x2, pad)
+  assert (out1[:, :5] - out2[:, :5]).abs().max() < 1e-5
+  assert (out1[:, 5:] - out2[:, 5:]).abs().max() > 1e-4
