@@ -180,14 +180,26 @@ class GPipeRunner(_GPipe1F1BMixin):
 
   def __init__(self, stage_idx: int, num_stages: int,
                num_micro_batches: int, group=None,
-               device: str = 'cpu'):
+               device: str = 'cpu', prev_rank: Optional[int] = None,
+               next_rank: Optional[int] = None):
+    """prev_rank/next_rank are GLOBAL ranks of the neighboring stages
+    (torch.distributed P2P addresses globally even inside a group);
+    they default to stage_idx -/+ 1, which is correct when the pipeline
+    group is the whole world. PP x DP composition passes the grid
+    neighbors from PpDpTopology."""
     self.stage_idx = stage_idx
     self.num_stages = num_stages
     self.num_micro = num_micro_batches
     self.group = group
     self.device = device
-    self._prev = stage_idx - 1 if stage_idx > 0 else None
-    self._next = stage_idx + 1 if stage_idx < num_stages - 1 else None
+    if stage_idx > 0:
+      self._prev = prev_rank if prev_rank is not None else stage_idx - 1
+    else:
+      self._prev = None
+    if stage_idx < num_stages - 1:
+      self._next = next_rank if next_rank is not None else stage_idx + 1
+    else:
+      self._next = None
 
   @property
   def is_first(self) -> bool:
