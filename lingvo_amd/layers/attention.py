@@ -250,3 +250,96 @@ class MultiHeadedAttention(BaseLayer):
     if p.use_bias:
       post = post + theta.post_b
     return post, cached_states
+
+
+class TransformerXLAttention(MultiHeadedAttention):
+  """Transformer-XL relative attention with segment memory (reference
+  batch_major_attention.py:2233 MultiHeadedAttentionXL +
+  attention_util.py:384 PositionalAttenLogits; Dai et al. 2019).
+
+  logits_ij = (q_i + u) . k_j  +  (q_i + v) . r_{d(i,j)}
+  with learned per-head biases u, v and sinusoidal relative embeddings
+  r projected per head. `memory` (the previous segment's input hidden
+  states, no grad) is prepended on the key/value side, giving XL-style
+  segment recurrence. The O(T^2) gather below is the CPU oracle; the
+  rel-shift fused form inside the flash kernel is the round-2 GPU path.
+  """
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('max_rel_dist', 512, 'Max relative distance embedded.')
+    p.cls = cls
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    n, h = self._n, self._h
+    assert self._nkv == n, 'XL attention uses full KV heads'
+    self.CreateVariable('u_var', py_utils.WeightParams(
+        [n, h], py_utils.WeightInit.Constant(0.0), p.dtype))
+    self.CreateVariable('v_var', py_utils.WeightParams(
+        [n, h], py_utils.WeightInit.Constant(0.0), p.dtype))
+    self.CreateVariable('pos_proj', py_utils.WeightParams(
+        [p.input_dim, n * h], p.params_init, p.dtype))
+
+  def _RelEmb(self, dists: torch.Tensor, theta) -> torch.Tensor:
+    """Sinusoidal embedding of signed distances -> [L, N, H]."""
+    p = self.p
+    d = dists.float().clamp(-p.max_rel_dist, p.max_rel_dist)
+    half = p.input_dim // 2
+    inv = torch.exp(torch.arange(half, device=d.device, dtype=torch.float32)
+                    * -(math.log(10000.0) / max(1, half - 1)))
+    ang = d[:, None] * inv[None, :]
+    emb = torch.cat([torch.sin(ang), torch.cos(ang)], dim=-1)
+    if p.input_dim % 2:
+      emb = torch.nn.functional.pad(emb, (0, 1))
+    r = torch.matmul(emb.to(theta.pos_proj.dtype), theta.pos_proj)
+    return r.reshape(-1, self._n, self._h)
+
+  def FProp(self, theta: NestedMap, query_vec: torch.Tensor,
+            paddings: Optional[torch.Tensor] = None,
+            memory: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """query_vec [B,T,D]; memory [B,M,D] previous-segment hidden states
+    (detached inside). Returns [B,T,D]."""
+    p = self.p
+    n, h = self._n, self._h
+    b, t, _ = query_vec.shape
+    m = 0 if memory is None else memory.shape[1]
+    kv_in = query_vec if memory is None else torch.cat(
+        [memory.detach(), query_vec], dim=1)
+    q, _, _ = self._Project(theta, query_vec)
+    _, k, v = self._Project(theta, kv_in)
+    qf, kf, vf = q.float(), k.float(), v.float()
+    u = theta.u_var.float()[None, None]   # [1,1,N,H]
+    vv = theta.v_var.float()[None, None]
+    scale = 1.0 / math.sqrt(h)
+    ac = torch.einsum('btnh,bsnh->bnts', qf + u, kf)
+    # distances d(i, j) = (m + i) - j for key index j in [0, m+t)
+    qpos = torch.arange(t, device=q.device) + m
+    kpos = torch.arange(m + t, device=q.device)
+    dmat = qpos[:, None] - kpos[None, :]                   # [T, S]
+    uniq = torch.arange(-(m + t) + 1, m + t, device=q.device)
+    r = self._RelEmb(uniq, theta).float()                  # [L,N,H]
+    bd_all = torch.einsum('btnh,lnh->bntl', qf + vv, r)
+    idx = (dmat + (m + t) - 1).reshape(-1)
+    bd = bd_all.reshape(b, n, t, -1).gather(
+        3, idx.reshape(1, 1, t, m + t).expand(b, n, t, m + t))
+    logits = (ac + bd) * scale
+    mask = kpos[None, :] <= qpos[:, None] if p.causal else \
+        torch.ones(t, m + t, dtype=torch.bool, device=q.device)
+    mask = mask[None, None].expand(b, 1, t, m + t).clone()
+    if paddings is not None:
+      kpad = paddings if memory is None else torch.cat(
+          [torch.zeros(b, m, device=q.device), paddings], dim=1)
+      mask = mask & (kpad[:, None, None, :] < 0.5)
+    logits = logits.masked_fill(~mask, -1e30)
+    probs = torch.softmax(logits, dim=-1)
+    ctx = torch.einsum('bnts,bsnh->btnh', probs, vf)
+    ctx = ctx.reshape(b, t, n * h).to(query_vec.dtype)
+    post = py_utils.MatmulBias(ctx, theta.post_w,
+                               theta.post_b if p.use_bias else None)
+    if paddings is not None:
+      post = py_utils.ApplyPadding(paddings, post)
+    return post

@@ -390,3 +390,39 @@ def test_rope_attention_decode_matches_fprop():
     o, st = sl.StreamStep(sl.theta, x[:, c0:c0 + 4], pad[:, c0:c0 + 4], st)
     outs.append(o)
   assert (full_s - torch.cat(outs, dim=1)).abs().max() < 1e-3
+
+
+def test_xl_attention_reduces_to_plain_and_memory_consistency():
+  # u=v=0, pos_proj=0 => exactly plain causal MHA with shared weights
+  p = attention_lib.TransformerXLAttention.Params().Set(
+      name='xl', input_dim=32, hidden_dim=32, num_heads=2, causal=True,
+      random_seed=17)
+  xl = p.Instantiate()
+  xl.eval()
+  with torch.no_grad():
+    xl.pos_proj.zero_()
+  ref = attention_lib.MultiHeadedAttention.Params().Set(
+      name='xl', input_dim=32, hidden_dim=32, num_heads=2, causal=True,
+      random_seed=17).Instantiate()
+  ref.eval()
+  ref.load_state_dict(xl.state_dict(), strict=False)
+  g = torch.Generator().manual_seed(8)
+  x = torch.randn(2, 10, 32, generator=g)
+  pad = torch.zeros(2, 10)
+  assert (xl.FProp(xl.theta, x, pad) -
+          ref.FProp(ref.theta, x, pad)).abs().max() < 1e-4
+
+  # content-dependent position term changes the output
+  with torch.no_grad():
+    xl.pos_proj.normal_(std=0.1)
+    xl.u_var.normal_(std=0.1)
+    xl.v_var.normal_(std=0.1)
+  out_pos = xl.FProp(xl.theta, x, pad)
+  assert (out_pos - ref.FProp(ref.theta, x, pad)).abs().max() > 1e-3
+
+  # segment recurrence: attending x2 with memory x1 == the x2 slice of
+  # full attention over [x1; x2]
+  x1, x2 = x[:, :6], x[:, 6:]
+  full = xl.FProp(xl.theta, x, pad)
+  with_mem = xl.FProp(xl.theta, x2, pad[:, 6:], memory=x1)
+  assert (full[:, 6:] - with_mem).abs().max() < 1e-4
