@@ -196,6 +196,28 @@ class ConformerLayer(BaseLayer):
           use_reentrant=False)
     return self._Body(theta, inputs, paddings)
 
+  # ---- streaming (reference conformer_layer.py:390 StreamStep) --------
+  def InitStreamState(self, theta: NestedMap, batch: int, max_len: int,
+                      device, dtype=torch.float32) -> NestedMap:
+    """Requires is_causal=True, conv_norm='layer' and a finite
+    atten_left_context (or max_len-bounded history)."""
+    assert self.p.is_causal, 'streaming requires a causal block'
+    return NestedMap(
+        atten=self.trans_atten.InitStates(theta.trans_atten, batch,
+                                          max_len, device, dtype),
+        lconv=self.lconv.InitStreamState(batch, device, dtype))
+
+  def StreamStep(self, theta: NestedMap, x_chunk: torch.Tensor,
+                 paddings_chunk: torch.Tensor, state: NestedMap):
+    x = self.fflayer_start.FProp(theta.fflayer_start, x_chunk,
+                                 paddings_chunk)
+    x, state.atten = self.trans_atten.StreamStep(
+        theta.trans_atten, x, paddings_chunk, state.atten)
+    x, state.lconv = self.lconv.StreamStep(theta.lconv, x,
+                                           paddings_chunk, state.lconv)
+    x = self.fflayer_end.FProp(theta.fflayer_end, x, paddings_chunk)
+    return self.final_ln.FProp(theta.final_ln, x), state
+
 
 class ConvSubsampling(BaseLayer):
   """2x Conv2D stride-2 frontend: [B, T, F] mel -> [B, T/4, D]

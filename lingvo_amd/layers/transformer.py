@@ -69,6 +69,13 @@ class TransformerAttentionLayer(BaseLayer):
     ctx, states = self.atten.ExtendStep(theta.atten, x, cached_states)
     return query_vec + ctx, states
 
+  def StreamStep(self, theta, x_chunk, paddings_chunk, state):
+    """Chunked streaming (causal; left-context-bounded KV history)."""
+    x = self.layer_norm.FProp(theta.layer_norm, x_chunk)
+    ctx, state = self.atten.StreamStep(theta.atten, x, paddings_chunk,
+                                       state)
+    return x_chunk + ctx, state
+
 
 class TransformerFeedForwardLayer(BaseLayer):
   """Pre-LN FFN: LN -> FC(hidden) -> act -> dropout -> FC(out) -> residual

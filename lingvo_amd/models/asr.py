@@ -110,6 +110,25 @@ class ConformerEncoder(BaseLayer):
       x = block.FProp(theta.blocks[i], x, out_pad)
     return x, out_pad
 
+  # ---- streaming over the conformer stack (post-frontend chunks;
+  # reference conformer StreamStep composition). The frontend runs
+  # per-chunk upstream (ConvSubsampling is local: receptive field 7
+  # input frames), streaming frontend fusion is a round-2 item.
+  def InitStreamState(self, theta: NestedMap, batch: int, max_len: int,
+                      device, dtype=torch.float32) -> NestedMap:
+    return NestedMap(blocks=[
+        b.InitStreamState(theta.blocks[i], batch, max_len, device, dtype)
+        for i, b in enumerate(self.blocks)])
+
+  def StreamStep(self, theta: NestedMap, feat_chunk: torch.Tensor,
+                 paddings_chunk: torch.Tensor, state: NestedMap):
+    """feat_chunk [B, C, model_dim]: already-subsampled features."""
+    x = feat_chunk
+    for i, block in enumerate(self.blocks):
+      x, state.blocks[i] = block.StreamStep(
+          theta.blocks[i], x, paddings_chunk, state.blocks[i])
+    return x, state
+
 
 class AsrDecoder(BaseLayer):
   """Teacher-forced attention LSTM decoder (LAS-style,
