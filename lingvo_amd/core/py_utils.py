@@ -433,3 +433,20 @@ class Timer:
 
   def __exit__(self, *a):
     self.Stop()
+
+
+def SinkhornAssignment(scores: torch.Tensor, tau: float = 0.1,
+                       n_iters: int = 20) -> torch.Tensor:
+  """Differentiable (soft) assignment via Sinkhorn normalization
+  (reference lingvo/core/differentiable_assignment.py): iterated
+  log-space row/column normalization of scores/tau converges to a
+  doubly-stochastic matrix that approaches the argmax permutation as
+  tau -> 0. scores [..., N, M] -> soft assignment of the same shape.
+  """
+  log_alpha = scores / tau
+  for _ in range(n_iters):
+    log_alpha = log_alpha - torch.logsumexp(log_alpha, dim=-1,
+                                            keepdim=True)
+    log_alpha = log_alpha - torch.logsumexp(log_alpha, dim=-2,
+                                            keepdim=True)
+  return torch.exp(log_alpha)

@@ -426,3 +426,51 @@ def test_xl_attention_reduces_to_plain_and_memory_consistency():
   full = xl.FProp(xl.theta, x, pad)
   with_mem = xl.FProp(xl.theta, x2, pad[:, 6:], memory=x1)
   assert (full[:, 6:] - with_mem).abs().max() < 1e-4
+
+
+def test_funnel_pool_and_upsample():
+  from lingvo_amd.layers import funnel
+  g = torch.Generator().manual_seed(6)
+  x = torch.randn(2, 7, 4, generator=g)
+  pad = torch.zeros(2, 7)
+  pad[1, 5:] = 1.0
+  pool = funnel.FunnelPoolingLayer.Params().Set(
+      name='p', stride=2).Instantiate()
+  y, ypad = pool.FProp(pool.theta, x, pad)
+  assert y.shape == (2, 4, 4)
+  # window [4,5] for b=1 has one real frame -> avg over the real one
+  assert torch.allclose(y[1, 2], x[1, 4], atol=1e-6)
+  assert ypad[1].tolist() == [0, 0, 0, 1]  # window [6(pad),7(pad-fill)]
+  # max pooling ignores padded frames
+  pmax = funnel.FunnelPoolingLayer.Params().Set(
+      name='m', stride=2, pooling_type='MAX').Instantiate()
+  ym, _ = pmax.FProp(pmax.theta, x, pad)
+  assert torch.allclose(ym[0, 0], torch.maximum(x[0, 0], x[0, 1]))
+  assert torch.allclose(ym[1, 2], x[1, 4], atol=1e-6)
+
+  up = funnel.FunnelUpsampleLayer.Params().Set(
+      name='u', stride=2, input_dim=4, random_seed=3).Instantiate()
+  z = up.FProp(up.theta, y, target_len=7)
+  assert z.shape == (2, 7, 4)
+  nop = funnel.FunnelUpsampleLayer.Params().Set(
+      name='n', stride=2, input_dim=4, use_projection=False).Instantiate()
+  z2 = nop.FProp(nop.theta, y, target_len=7)
+  assert torch.allclose(z2[:, 0], y[:, 0]) and \
+      torch.allclose(z2[:, 1], y[:, 0])
+
+
+def test_sinkhorn_assignment():
+  from lingvo_amd.core import py_utils as pu
+  scores = torch.tensor([[9.0, 0.1, 0.2],
+                         [0.3, 8.0, 0.1],
+                         [0.2, 0.4, 7.0]])
+  a = pu.SinkhornAssignment(scores, tau=0.3, n_iters=50)
+  # doubly stochastic
+  assert torch.allclose(a.sum(-1), torch.ones(3), atol=1e-3)
+  assert torch.allclose(a.sum(-2), torch.ones(3), atol=1e-3)
+  # approaches the identity permutation
+  assert a.diag().min() > 0.95
+  # differentiable
+  s = scores.clone().requires_grad_(True)
+  pu.SinkhornAssignment(s, tau=0.5, n_iters=10).trace().backward()
+  assert s.grad is not None and torch.isfinite(s.grad).all()
