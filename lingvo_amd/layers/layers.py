@@ -586,3 +586,54 @@ class GradNormTracker(BaseLayer):
       self.log_var.mul_(d).add_((1 - d) * delta * delta)
       self.count += 1
     return ok
+
+
+class MultitaskAdapterLayer(BaseLayer):
+  """Per-task residual bottleneck adapters (reference layers.py:6205
+  MultitaskAdapterLayer; Houlsby et al. 2019). One [num_tasks, ...]
+  weight stack; FProp gathers each example's task adapter:
+      y = x + Wup_t * act(Wdown_t * LN(x) + b_t) + bup_t.
+  """
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('num_tasks', 1, 'Task count.')
+    p.Define('input_dim', 0, 'Model dim D.')
+    p.Define('bottleneck_dim', 0, 'Adapter bottleneck.')
+    p.Define('activation', 'RELU', 'Bottleneck activation.')
+    p.Define('layer_norm', True, 'LN before the adapter.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    self.CreateVariable('down_w', py_utils.WeightParams(
+        [p.num_tasks, p.input_dim, p.bottleneck_dim], p.params_init,
+        p.dtype))
+    self.CreateVariable('down_b', py_utils.WeightParams(
+        [p.num_tasks, p.bottleneck_dim],
+        py_utils.WeightInit.Constant(0.0), p.dtype))
+    self.CreateVariable('up_w', py_utils.WeightParams(
+        [p.num_tasks, p.bottleneck_dim, p.input_dim], p.params_init,
+        p.dtype))
+    self.CreateVariable('up_b', py_utils.WeightParams(
+        [p.num_tasks, p.input_dim],
+        py_utils.WeightInit.Constant(0.0), p.dtype))
+    if p.layer_norm:
+      self.CreateChild('ln', LayerNorm.Params().Set(
+          input_dim=p.input_dim))
+
+  def FProp(self, theta: NestedMap, inputs: torch.Tensor,
+            tasks: torch.Tensor) -> torch.Tensor:
+    """inputs [B, T, D]; tasks [B] int task ids."""
+    p = self.p
+    x = self.ln.FProp(theta.ln, inputs) if p.layer_norm else inputs
+    t = tasks.long()
+    dw = theta.down_w[t]       # [B, D, bottleneck]
+    db = theta.down_b[t]       # [B, bottleneck]
+    uw = theta.up_w[t]
+    ub = theta.up_b[t]
+    act = activations.GetFn(p.activation)
+    h = act(torch.baddbmm(db.unsqueeze(1), x, dw))
+    return inputs + torch.baddbmm(ub.unsqueeze(1), h, uw)
