@@ -119,3 +119,83 @@ class QuantizedProjectionLayer(QuantizableLayer):
   def FProp(self, theta, x):
     w = self.QWeight(theta.w)
     return self.QAct('out', torch.matmul(x, w) + theta.b)
+
+
+class FakeQuantizationSchedule(BaseLayer):
+  """Ramped clipping-cap schedule (reference quant_utils.py:1316
+  FakeQuantizationSchedule): between clip_start_step and clip_end_step
+  the clipping cap interpolates from start_cap to end_cap; fake
+  quantization itself switches on at quant_start_step."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('clip_start_step', 0, 'Begin ramping the cap.')
+    p.Define('clip_end_step', 0, 'Cap fully ramped.')
+    p.Define('quant_start_step', 0, 'Fake-quant active from here.')
+    p.Define('start_cap', 8.0, 'Initial clipping cap.')
+    p.Define('end_cap', 1.0, 'Final clipping cap.')
+    return p
+
+  def CurrentCap(self, step: int) -> float:
+    p = self.p
+    if step <= p.clip_start_step:
+      return p.start_cap
+    if step >= p.clip_end_step:
+      return p.end_cap
+    frac = (step - p.clip_start_step) / max(
+        1, p.clip_end_step - p.clip_start_step)
+    return p.start_cap + frac * (p.end_cap - p.start_cap)
+
+  def ShouldQuantize(self, step: int) -> bool:
+    return step >= self.p.quant_start_step
+
+
+class _FakeQuantAsymFn(torch.autograd.Function):
+  """Straight-through asymmetric fake quantization."""
+
+  @staticmethod
+  def forward(ctx, x, scale, zero_point, bits):
+    qmax = 2.0 ** bits - 1
+    q = torch.clamp(torch.round(x / scale + zero_point), 0, qmax)
+    return (q - zero_point) * scale
+
+  @staticmethod
+  def backward(ctx, g):
+    return g, None, None, None
+
+
+class PassiveAsymQDomain(QDomain):
+  """Asymmetric per-name calibration domain (reference
+  quant_utils.py:1606 PassiveAsymQDomain): tracks running min/max per
+  named activation and fake-quantizes into an asymmetric [min, max]
+  range with a zero point — the domain used for post-training-style
+  passive calibration."""
+
+  def __init__(self, params):
+    super().__init__(params)
+    self._ranges = {}
+
+  def QuantizeNamedTensor(self, name: str, x: torch.Tensor,
+                          calibrate: bool = True) -> torch.Tensor:
+    p = self.p
+    if name not in self._ranges:
+      self._ranges[name] = [torch.zeros(()), torch.ones(())]
+    lo, hi = self._ranges[name]
+    if calibrate and self.training:
+      with torch.no_grad():
+        lo.mul_(p.decay).add_(x.detach().min().float() * (1 - p.decay))
+        hi.mul_(p.decay).add_(x.detach().max().float() * (1 - p.decay))
+    if self._step < p.start_step:
+      return x
+    qmax = 2.0 ** p.bits - 1
+    scale = ((hi - lo).clamp_min(1e-6) / qmax).to(x.dtype)
+    zp = torch.round(-lo / scale.float()).to(x.dtype)
+    return _FakeQuantAsymFn.apply(x, scale, zp, p.bits)
+
+  def QuantizeTensor(self, x, calibrate: bool = True):
+    return self.QuantizeNamedTensor('default', x, calibrate)
+
+  def state_dict_ranges(self):
+    return {k: (float(v[0]), float(v[1]))
+            for k, v in self._ranges.items()}

@@ -87,3 +87,30 @@ def test_pillars_model_train_and_decode():
   dm = task.CreateDecoderMetrics()
   task.PostProcessDecodeOut(out, dm)
   assert 0.0 <= dm.ap3d.value <= 1.0
+
+
+def test_fake_quant_schedule_and_asym_domain():
+  import torch
+  from lingvo_amd.core import quant_utils
+  sched = quant_utils.FakeQuantizationSchedule.Params().Set(
+      name='s', clip_start_step=10, clip_end_step=20,
+      quant_start_step=15, start_cap=8.0, end_cap=1.0).Instantiate()
+  assert sched.CurrentCap(0) == 8.0
+  assert sched.CurrentCap(25) == 1.0
+  assert 1.0 < sched.CurrentCap(15) < 8.0
+  assert not sched.ShouldQuantize(14) and sched.ShouldQuantize(15)
+
+  dom = quant_utils.PassiveAsymQDomain.Params().Set(
+      name='d', bits=8, decay=0.0).Instantiate()
+  dom.train()
+  dom.SetStep(1)
+  x = torch.linspace(-1.0, 3.0, 101)
+  q = dom.QuantizeNamedTensor('act', x)
+  # asymmetric range covers [-1, 3]; quantization error <= scale/2
+  scale = 4.0 / 255
+  assert (q - x).abs().max() <= scale * 0.51 + 1e-6
+  # straight-through grad
+  x2 = x.clone().requires_grad_(True)
+  dom.QuantizeNamedTensor('act', x2).sum().backward()
+  assert torch.allclose(x2.grad, torch.ones_like(x2))
+  assert 'act' in dom.state_dict_ranges()
