@@ -341,3 +341,30 @@ def test_step_api_embedding_and_rnn_stack_residual():
   st = rs.ZeroState(rs.theta, NestedMap(), 3, 'cpu', torch.float32)
   y, st = rs.FProp(rs.theta, NestedMap(), out, torch.zeros(3, 1), st)
   assert y.output.shape == (3, 6)
+
+
+def test_symbol_insertion_roundtrip():
+  import torch
+  from lingvo_amd.core import insertion
+  from lingvo_amd.core import py_utils
+  layer = insertion.SymbolInsertionLayer.Params().Set(
+      name='ins').Instantiate()
+  layer.train()
+  g = torch.Generator().manual_seed(4)
+  x = torch.randint(3, 50, (3, 10), generator=g)
+  pad = torch.zeros(3, 10)
+  pad[1, 7:] = 1.0
+  pad[2, 4:] = 1.0
+  with py_utils.StepSeedScope(global_seed=5, step=1):
+    out = layer.FProp(layer.theta, x, pad)
+  # canvas rows are subsequences of x
+  for i in range(3):
+    canv = out.canvas[i][out.canvas_paddings[i] < 0.5].tolist()
+    full = x[i][pad[i] < 0.5].tolist()
+    it = iter(full)
+    assert all(tok in it for tok in canv), (canv, full)  # subsequence
+  # canvas + targets reconstruct the original sequences exactly
+  rec = insertion.ReconstructFromCanvas(out.canvas, out.canvas_paddings,
+                                        out.target_indices)
+  for i in range(3):
+    assert rec[i] == x[i][pad[i] < 0.5].tolist(), i
