@@ -518,3 +518,31 @@ def test_multitask_adapter_routes_by_task():
   g = ad.down_w.grad
   assert g[0].abs().sum() > 0 and g[1].abs().sum() > 0
   assert g[2].abs().sum() == 0  # task 2 unused
+
+
+def test_evolved_transformer_layers():
+  from lingvo_amd.layers import evolved_transformer as evt
+  g = torch.Generator().manual_seed(9)
+  x = torch.randn(2, 10, 8, generator=g)
+  pad = torch.zeros(2, 10)
+  pad[1, 7:] = 1.0
+
+  enc = evt.EvolvedTransformerEncoderLayer.Params().Set(
+      name='e', input_dim=8, num_heads=2, random_seed=3).Instantiate()
+  enc.eval()
+  out = enc.FProp(enc.theta, x, pad)
+  assert out.shape == x.shape
+  assert out[1, 7:].abs().max() < 1e-6
+  out.sum().backward()
+
+  dec = evt.EvolvedTransformerDecoderBranchedConvsLayer.Params().Set(
+      name='d', input_dim=8, random_seed=4).Instantiate()
+  dec.eval()
+  o1 = dec.FProp(dec.theta, x, pad)
+  assert o1.shape == x.shape
+  # causality: perturbing a later frame leaves earlier outputs unchanged
+  x2 = x.clone()
+  x2[:, 6] += 5.0
+  o2 = dec.FProp(dec.theta, x2, pad)
+  assert (o1[:, :6] - o2[:, :6]).abs().max() < 1e-5
+  assert (o1[0, 6:] - o2[0, 6:]).abs().max() > 1e-4
