@@ -32,6 +32,9 @@ class BaseProgram:
     p.Define('name', '', 'Program name.')
     p.Define('steps_per_loop', 100, 'Steps per Run() call.')
     p.Define('dataset_name', 'Train', 'Input dataset.')
+    p.Define('ml_perf_log', False,
+             'Emit MLPerf structured log lines (reference '
+             'ml_perf_log.py via program.py mlperf hooks).')
     return p
 
   def __init__(self, params: Params, task, logdir: str, device: str):
@@ -85,6 +88,11 @@ class TrainProgram(BaseProgram):
         metrics = task.TrainStep(batch, grad_sync_finalize=finalize)
     loss = py_utils.ToScalar(metrics[task.learners[0].p.loss_name][0])
     self._Log({'step': task.global_step, 'loss': loss})
+    if self.p.ml_perf_log:
+      from lingvo_amd.utils import helpers
+      helpers.mlperf_print('block_stop', metadata={
+          'first_step': task.global_step - self.p.steps_per_loop + 1,
+          'step': task.global_step, 'loss': loss})
     return NestedMap(loss=loss, step=task.global_step)
 
 
@@ -178,6 +186,10 @@ class Executor:
         [l.EnsureOptimizer(self.task) for l in self.task.learners])
 
   def Start(self) -> None:
+    if self.schedule.p.train_program.ml_perf_log:
+      from lingvo_amd.utils import helpers
+      helpers.mlperf_print('init_start')
+      helpers.mlperf_print('run_start')
     restored = self.ckpt.Restore()
     while True:
       t0 = time.perf_counter()
@@ -190,3 +202,7 @@ class Executor:
         break
     self.ckpt.Save()
     self.ckpt.Sync()
+    if self.schedule.p.train_program.ml_perf_log:
+      from lingvo_amd.utils import helpers
+      helpers.mlperf_print('run_stop',
+                           metadata={'step': self.task.global_step})

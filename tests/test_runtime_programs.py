@@ -108,3 +108,21 @@ def test_runner_retries_transient_errors(tmp_path):
       r._RunLoop(always)
   finally:
     _time.sleep = orig_sleep
+
+
+def test_mlperf_logging_emitted(tmp_path, capsys):
+  """ml_perf_log=True emits :::MLLOG lines from the executor loop."""
+  from lingvo_amd.runtime import program as program_lib
+  mp = registry.GetParams('image.mnist.LeNet5', 'Train')
+  mp.task.random_seed = 3
+  sched = program_lib.SimpleProgramSchedule.Params()
+  sched.train_program.steps_per_loop = 2
+  sched.train_program.ml_perf_log = True
+  ex = program_lib.Executor(mp, str(tmp_path), sched, device='cpu',
+                            max_steps=2)
+  ex.Start()
+  out = capsys.readouterr().out
+  lines = [l for l in out.splitlines() if l.startswith(':::MLLOG')]
+  keys = [json.loads(l.split(' ', 1)[1])['key'] for l in lines]
+  assert 'run_start' in keys and 'run_stop' in keys
+  assert 'block_stop' in keys
