@@ -236,3 +236,103 @@ class Checkpointer:
           dst_name = re.sub(pattern, repl, src_name)
           if dst_name in own_sd and own_sd[dst_name].shape == tensor.shape:
             own_sd[dst_name].copy_(tensor)
+
+
+# ---- sharded (per-rank) checkpoints -----------------------------------
+#
+# For TP/PP-partitioned models every rank owns DIFFERENT parameters, so
+# each rank writes its own shard file (reference: the sharded-save path
+# of checkpointer/saver used under SPMD; here it is explicit, one file
+# per rank over the shared filesystem, with rank 0 owning the text
+# state file).
+
+def ShardedCheckpointPath(train_dir: str, step: int, shard: int,
+                          num_shards: int) -> str:
+  return os.path.join(
+      train_dir,
+      f'ckpt-{step:08d}.shard-{shard:05d}-of-{num_shards:05d}.pt')
+
+
+class ShardedCheckpointer(Checkpointer):
+  """Per-rank shard save/restore; pass the TP/PP rank and shard count
+  (defaults from torch.distributed when initialized)."""
+
+  def __init__(self, params: Params, train_dir: str, model,
+               optimizers: Optional[List[torch.optim.Optimizer]] = None,
+               shard_id: Optional[int] = None,
+               num_shards: Optional[int] = None):
+    super().__init__(params, train_dir, model, optimizers)
+    import torch.distributed as dist
+    if shard_id is None:
+      shard_id = dist.get_rank() if dist.is_initialized() else 0
+    if num_shards is None:
+      num_shards = dist.get_world_size() if dist.is_initialized() else 1
+    self._shard = shard_id
+    self._num_shards = num_shards
+
+  def Save(self, step: Optional[int] = None) -> str:
+    import torch.distributed as dist
+    step = int(step if step is not None else self._model.global_step)
+    payload = self._Payload()
+    cpu_payload = {
+        k: ({kk: (vv.detach().cpu() if isinstance(vv, torch.Tensor)
+                  else vv) for kk, vv in v.items()}
+            if isinstance(v, dict) else v)
+        for k, v in payload.items()
+    }
+    self._saver._SanityCheck(cpu_payload)
+    path = ShardedCheckpointPath(self._dir, step, self._shard,
+                                 self._num_shards)
+    tmp = path + '.tmp'
+    torch.save(cpu_payload, tmp)
+    os.replace(tmp, path)
+    if dist.is_initialized():
+      dist.barrier()  # all shards on disk before the state file commits
+    if self._shard == 0:
+      state = os.path.join(self._dir, 'checkpoint')
+      tmp = state + '.tmp'
+      with open(tmp, 'w') as f:
+        f.write(f'model_checkpoint_path: "ckpt-{step:08d}"\n')
+        f.write(f'num_shards: {self._num_shards}\n')
+      os.replace(tmp, state)
+      self._GCShards(step)
+    self._last_save_time = time.time()
+    self._last_save_step = step
+    return path
+
+  def _GCShards(self, newest_step: int) -> None:
+    steps = sorted({StepFromPath(p) for p in glob.glob(
+        os.path.join(self._dir, 'ckpt-????????.shard-*.pt'))})
+    drop = steps[:-self.p.keep_latest_n] if self.p.keep_latest_n else []
+    for s in drop:
+      for p in glob.glob(os.path.join(
+          self._dir, f'ckpt-{s:08d}.shard-*.pt')):
+        os.remove(p)
+
+  def LatestStep(self) -> Optional[int]:
+    state = os.path.join(self._dir, 'checkpoint')
+    if os.path.exists(state):
+      with open(state) as f:
+        for line in f:
+          m = re.match(r'model_checkpoint_path:\s*"ckpt-(\d+)"',
+                       line.strip())
+          if m:
+            return int(m.group(1))
+    steps = sorted({StepFromPath(p) for p in glob.glob(
+        os.path.join(self._dir, 'ckpt-????????.shard-*.pt'))})
+    return steps[-1] if steps else None
+
+  def Restore(self, path: Optional[str] = None) -> Optional[int]:
+    step = self.LatestStep()
+    if step is None:
+      self._MaybeWarmStart()
+      return None
+    shard_path = ShardedCheckpointPath(self._dir, step, self._shard,
+                                       self._num_shards)
+    payload = torch.load(shard_path, map_location='cpu',
+                         weights_only=False)
+    self._model.load_state_dict(payload['model'], strict=False)
+    for i, opt in enumerate(self._optimizers):
+      if f'optimizer_{i}' in payload:
+        opt.load_state_dict(payload[f'optimizer_{i}'])
+    return int(payload.get('step', step))

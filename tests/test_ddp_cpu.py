@@ -96,3 +96,59 @@ def test_full_task_dp_training_keeps_replicas_in_sync():
       p.join(240)
       assert p.exitcode == 0
     assert torch.allclose(results[0], results[1], atol=1e-6)
+
+
+def _run_sharded_ckpt(rank, world, port, tmpdir, results):
+  os.environ['MASTER_ADDR'] = '127.0.0.1'
+  os.environ['MASTER_PORT'] = str(port)
+  dist.init_process_group('gloo', rank=rank, world_size=world)
+  from lingvo_amd.core import checkpointer as ckpt_lib
+
+  class Shard(torch.nn.Module):
+    def __init__(self):
+      super().__init__()
+      # per-rank distinct parameter (like a TP/PP shard)
+      self.w = torch.nn.Parameter(
+          torch.full((4,), float(rank + 1)))
+      self.global_step = 7
+
+  model = Shard()
+  opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9)
+  (model.w ** 2).sum().backward()
+  opt.step()
+  ck = ckpt_lib.ShardedCheckpointer(
+      ckpt_lib.Checkpointer.Params().Set(keep_latest_n=2),
+      tmpdir, model, [opt])
+  ck.Save(step=7)
+  saved_w = model.w.detach().clone()
+  with torch.no_grad():
+    model.w.fill_(-99.0)
+  opt2 = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9)
+  ck2 = ckpt_lib.ShardedCheckpointer(
+      ckpt_lib.Checkpointer.Params(), tmpdir, model, [opt2])
+  step = ck2.Restore()
+  results[f'ok{rank}'] = bool(
+      step == 7 and torch.allclose(model.w.detach(), saved_w) and
+      'momentum_buffer' in list(opt2.state.values())[0])
+  dist.destroy_process_group()
+
+
+def test_sharded_checkpoint_roundtrip(tmp_path):
+  ctx = mp.get_context('spawn')
+  with ctx.Manager() as mgr:
+    results = mgr.dict()
+    procs = [ctx.Process(target=_run_sharded_ckpt,
+                         args=(r, 2, 29563, str(tmp_path), results))
+             for r in range(2)]
+    for p in procs:
+      p.start()
+    for p in procs:
+      p.join(120)
+      assert p.exitcode == 0
+    assert results['ok0'] and results['ok1']
+  import glob as globlib
+  shards = globlib.glob(str(tmp_path / 'ckpt-00000007.shard-*.pt'))
+  assert len(shards) == 2
+  with open(tmp_path / 'checkpoint') as f:
+    txt = f.read()
+  assert 'ckpt-00000007' in txt and 'num_shards: 2' in txt
