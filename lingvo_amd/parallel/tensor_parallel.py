@@ -323,3 +323,46 @@ def ShardAttentionForTp(stack_params, tp_group=None):
   new_p.tp_group = tp_group
   stack_params.transformer_tpl.tr_atten_tpl.atten_tpl = new_p
   return stack_params
+
+
+def LowerShardingAnnotations(params, tp_group=None):
+  """Annotation-driven planner (the explicit-RCCL lowering of GShard's
+  device_mesh / weight_split_dims_mapping annotations, reference
+  gshard_utils.py:39-137 + base_layer.py:262-280).
+
+  Recursively walks a Params tree; any StackedTransformerLayers whose
+  transformer_tpl carries weight_split_dims_mapping (or whose sub
+  templates do) is rewritten to the explicit TP FFN + TP attention
+  classes. Model code keeps the same annotation surface as the
+  reference; this pass inserts the collectives."""
+  from lingvo_amd.core.hyperparams import Params as _Params
+  from lingvo_amd.layers import transformer as transformer_lib
+
+  def annotated(p):
+    try:
+      return (p.Get('weight_split_dims_mapping') is not None or
+              p.Get('device_mesh') is not None)
+    except Exception:
+      return False
+
+  def visit(p):
+    if not isinstance(p, _Params):
+      return
+    cls = p.Get('cls') if 'cls' in p else None
+    if cls is transformer_lib.StackedTransformerLayers:
+      tpl = p.transformer_tpl
+      if annotated(p) or annotated(tpl) or \
+          annotated(tpl.tr_fflayer_tpl) or \
+          annotated(tpl.tr_atten_tpl.atten_tpl):
+        ShardTransformerStackForTp(p, tp_group)
+        ShardAttentionForTp(p, tp_group)
+      return
+    for name, val in p.IterParams():
+      if isinstance(val, _Params):
+        visit(val)
+      elif isinstance(val, (list, tuple)):
+        for v in val:
+          visit(v) if isinstance(v, _Params) else None
+
+  visit(params)
+  return params

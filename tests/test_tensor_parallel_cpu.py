@@ -142,3 +142,27 @@ def test_shard_attention_planner():
   x = torch.randn(2, 6, 128)
   out = stack.FProp(stack.theta, x, torch.zeros(2, 6))
   assert out.shape == x.shape
+
+
+def test_lower_sharding_annotations():
+  """Annotated stacks rewrite to TP classes; unannotated stay."""
+  from lingvo_amd.layers import transformer as transformer_lib
+  sp = transformer_lib.StackedTransformerLayers.Params().Set(
+      name='s', model_dim=16, num_layers=1, num_heads=1, hidden_dim=32,
+      random_seed=2)
+  sp.transformer_tpl.tr_fflayer_tpl.Set(
+      input_dim=16, hidden_dim=32, weight_split_dims_mapping=[-1, 0])
+  outer = tp.TpFeedForwardLayer.Params()  # host tree containing the stack
+  from lingvo_amd.core.hyperparams import Params
+  host = Params()
+  host.Define('stack', sp, 'nested')
+  tp.LowerShardingAnnotations(host)
+  assert sp.transformer_tpl.tr_fflayer_tpl.cls is tp.TpFeedForwardLayer
+  assert sp.transformer_tpl.tr_atten_tpl.atten_tpl.cls is \
+      tp.TpMultiHeadedAttention
+
+  sp2 = transformer_lib.StackedTransformerLayers.Params().Set(
+      name='s2', model_dim=16, num_layers=1, num_heads=1, hidden_dim=32)
+  tp.LowerShardingAnnotations(sp2)
+  assert sp2.transformer_tpl.tr_fflayer_tpl.cls is not \
+      tp.TpFeedForwardLayer
