@@ -247,6 +247,111 @@ class Params:
         continue
     return self
 
+  def ToProto(self):
+    """Serializes to a google.protobuf Struct (reference
+    hyperparams.py:529 ToProto — the MI355X-native equivalent uses the
+    well-known Struct type instead of a custom Hyperparam schema). Each
+    leaf path maps to {'t': tag, 'v': value}; classes/functions/dtypes
+    are stored by qualified name and restored by import in FromProto."""
+    from google.protobuf import struct_pb2
+
+    def encode(val):
+      if val is None:
+        return 'none', 0
+      if isinstance(val, bool):
+        return 'bool', val
+      if isinstance(val, int):
+        return 'int', val
+      if isinstance(val, float):
+        return 'float', val
+      if isinstance(val, str):
+        return 'str', val
+      if isinstance(val, enum.Enum):
+        return 'enum', (f'{type(val).__module__}/'
+                        f'{type(val).__qualname__}/{val.name}')
+      if isinstance(val, type):
+        return 'type', f'{val.__module__}/{val.__qualname__}'
+      try:
+        import torch
+        if isinstance(val, torch.dtype):
+          return 'dtype', str(val)
+      except ImportError:
+        pass
+      if callable(val) and hasattr(val, '__qualname__'):
+        return 'fn', (f'{getattr(val, "__module__", "?")}/'
+                      f'{val.__qualname__}')
+      return 'literal', repr(val)
+
+    proto = struct_pb2.Struct()
+
+    def recurse(p, pref):
+      if isinstance(p, Params):
+        for name in sorted(p._params):
+          recurse(p._params[name].value, f'{pref}{name}.')
+      elif isinstance(p, (list, tuple)) and any(
+          isinstance(v, Params) for v in p):
+        for i, v in enumerate(p):
+          recurse(v, f'{pref[:-1]}[{i}].')
+      else:
+        tag, enc = encode(p)
+        entry = struct_pb2.Struct()
+        entry.fields['t'].string_value = tag
+        if tag in ('int', 'float'):
+          entry.fields['v'].number_value = enc
+        elif tag == 'bool':
+          entry.fields['v'].bool_value = enc
+        elif tag == 'none':
+          entry.fields['v'].null_value = 0
+        else:
+          entry.fields['v'].string_value = enc
+        proto.fields[pref[:-1]].struct_value.CopyFrom(entry)
+
+    recurse(self, '')
+    return proto
+
+  def FromProto(self, proto) -> 'Params':
+    """Applies a ToProto() Struct to this tree in place; classes,
+    functions, dtypes and enums are restored by importing their
+    qualified names (reference hyperparams.py:611 FromProto)."""
+    import importlib
+
+    def resolve(qual):
+      mod, _, rest = qual.partition('/')
+      obj = importlib.import_module(mod)
+      for part in rest.split('.'):
+        obj = getattr(obj, part)
+      return obj
+
+    for key, entry in proto.fields.items():
+      s = entry.struct_value
+      tag = s.fields['t'].string_value
+      v = s.fields['v']
+      if tag == 'none':
+        val = None
+      elif tag == 'bool':
+        val = v.bool_value
+      elif tag == 'int':
+        val = int(v.number_value)
+      elif tag == 'float':
+        val = v.number_value
+      elif tag == 'str':
+        val = v.string_value
+      elif tag == 'dtype':
+        import torch
+        val = getattr(torch, v.string_value.split('.')[-1])
+      elif tag in ('type', 'fn'):
+        val = resolve(v.string_value)
+      elif tag == 'enum':
+        mod_qual, _, member = v.string_value.rpartition('/')
+        val = getattr(resolve(mod_qual), member)
+      else:  # literal
+        val = ast.literal_eval(v.string_value)
+      try:
+        self.SetPath(key, val)
+      except AttributeError:
+        continue
+    return self
+
   def TextDiff(self, other: 'Params') -> str:
     """Returns a unified human-readable diff of two Params trees."""
     mine = dict(

@@ -122,3 +122,47 @@ def test_list_of_params_totext():
   text = p.ToText()
   assert 'blocks[0].d : 1' in text
   assert 'blocks[1].d : 1' in text
+
+
+def test_proto_roundtrip():
+  import torch
+  from lingvo_amd.core.hyperparams import Params
+  from lingvo_amd.core import py_utils
+  p = Params()
+  p.Define('lr', 0.1, 'float')
+  p.Define('steps', 100, 'int')
+  p.Define('name', 'conformer', 'str')
+  p.Define('flag', True, 'bool')
+  p.Define('nothing', None, 'none')
+  p.Define('dims', [1, 2, 3], 'list literal')
+  p.Define('dtype', torch.bfloat16, 'torch dtype')
+  p.Define('cls_val', py_utils.WeightInit, 'a class')
+  sub = Params()
+  sub.Define('inner', 7, 'nested')
+  p.Define('sub', sub, 'subtree')
+
+  proto = p.ToProto()
+  q = p.Copy()
+  q.lr = 9.9
+  q.steps = 1
+  q.name = 'x'
+  q.flag = False
+  q.dims = []
+  q.dtype = torch.float32
+  q.sub.inner = -1
+  q.FromProto(proto)
+  assert q.lr == 0.1 and q.steps == 100 and q.name == 'conformer'
+  assert q.flag is True and q.nothing is None
+  assert q.dims == [1, 2, 3]
+  assert q.dtype is torch.bfloat16
+  assert q.cls_val is py_utils.WeightInit
+  assert q.sub.inner == 7
+  # serialized form survives the wire
+  blob = proto.SerializeToString()
+  from google.protobuf import struct_pb2
+  back = struct_pb2.Struct()
+  back.ParseFromString(blob)
+  q2 = p.Copy()
+  q2.lr = 0.0
+  q2.FromProto(back)
+  assert q2.lr == 0.1
