@@ -180,3 +180,50 @@ class GreedySearchHelper:
     lens = (out != self.eos_id).long().sum(-1) + 1
     return NestedMap(topk_ids=out.unsqueeze(1),
                      topk_lens=lens.clamp_max(out.shape[1]).unsqueeze(1))
+
+
+def MergeBeamSearchOutputs(max_hyps_per_beam: int,
+                           beam_search_outputs) -> NestedMap:
+  """Merges multiple BeamSearchDecode outputs into one ranked set
+  (reference beam_search_helper.py:681): concatenate the hypothesis
+  axes, drop exact-duplicate token sequences (keeping the best score),
+  and return the top `max_hyps_per_beam` per beam."""
+  max_len = max(o.topk_ids.shape[-1] for o in beam_search_outputs)
+
+  def pad_ids(ids):
+    if ids.shape[-1] < max_len:
+      ids = torch.nn.functional.pad(ids, (0, max_len - ids.shape[-1]))
+    return ids
+
+  ids = torch.cat([pad_ids(o.topk_ids) for o in beam_search_outputs],
+                  dim=1)                                   # [B, K', L]
+  lens = torch.cat([o.topk_lens for o in beam_search_outputs], dim=1)
+  scores = torch.cat([o.topk_scores for o in beam_search_outputs], dim=1)
+  b, ktot, _ = ids.shape
+  # mask tokens beyond each hyp's length so equal-content hyps compare
+  # equal regardless of stale tail tokens
+  arange = torch.arange(max_len, device=ids.device)
+  ids = ids * (arange[None, None, :] < lens[:, :, None]).long()
+  out_ids = torch.zeros(b, max_hyps_per_beam, max_len, dtype=ids.dtype,
+                        device=ids.device)
+  out_lens = torch.zeros(b, max_hyps_per_beam, dtype=lens.dtype,
+                         device=ids.device)
+  out_scores = torch.full((b, max_hyps_per_beam), -1e30,
+                          device=ids.device)
+  for i in range(b):
+    order = scores[i].argsort(descending=True)
+    seen = set()
+    slot = 0
+    for h in order.tolist():
+      key = tuple(ids[i, h, :int(lens[i, h])].tolist())
+      if key in seen:
+        continue
+      seen.add(key)
+      out_ids[i, slot] = ids[i, h]
+      out_lens[i, slot] = lens[i, h]
+      out_scores[i, slot] = scores[i, h]
+      slot += 1
+      if slot == max_hyps_per_beam:
+        break
+  return NestedMap(topk_ids=out_ids, topk_lens=out_lens,
+                   topk_scores=out_scores)

@@ -368,3 +368,23 @@ def test_symbol_insertion_roundtrip():
                                         out.target_indices)
   for i in range(3):
     assert rec[i] == x[i][pad[i] < 0.5].tolist(), i
+
+
+def test_merge_beam_search_outputs():
+  import torch
+  from lingvo_amd.core import beam_search_helper as bsh
+  from lingvo_amd.core.nested_map import NestedMap
+  a = NestedMap(
+      topk_ids=torch.tensor([[[5, 6, 2, 0], [5, 2, 0, 0]]]),
+      topk_lens=torch.tensor([[3, 2]]),
+      topk_scores=torch.tensor([[-1.0, -2.0]]))
+  b = NestedMap(
+      topk_ids=torch.tensor([[[5, 6, 2], [7, 2, 9]]]),  # dup of a[0] + new
+      topk_lens=torch.tensor([[3, 2]]),
+      topk_scores=torch.tensor([[-0.5, -1.5]]))
+  merged = bsh.MergeBeamSearchOutputs(3, [a, b])
+  # duplicate (5,6,2) keeps the better score -0.5
+  assert merged.topk_scores[0].tolist() == [-0.5, -1.5, -2.0]
+  assert merged.topk_ids[0, 0, :3].tolist() == [5, 6, 2]
+  assert merged.topk_ids[0, 1, :2].tolist() == [7, 2]
+  assert merged.topk_ids[0, 2, :2].tolist() == [5, 2]
