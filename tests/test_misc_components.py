@@ -388,3 +388,55 @@ def test_merge_beam_search_outputs():
   assert merged.topk_ids[0, 0, :3].tolist() == [5, 6, 2]
   assert merged.topk_ids[0, 1, :2].tolist() == [7, 2]
   assert merged.topk_ids[0, 2, :2].tolist() == [5, 2]
+
+
+def _toy_search_fns(vocab=6, seed=3):
+  import torch
+  from lingvo_amd.core.nested_map import NestedMap
+  g = torch.Generator().manual_seed(seed)
+  table = torch.log_softmax(torch.randn(vocab, vocab, generator=g) * 2,
+                            dim=-1)
+
+  def init_fn(batch, k):
+    return NestedMap(dummy=torch.zeros(batch * k))
+
+  def step_fn(state, prev_ids):
+    return table[prev_ids], state
+
+  def reorder_fn(state, gather):
+    state.dummy = state.dummy[gather]
+    return state
+
+  return init_fn, step_fn, reorder_fn
+
+
+def test_flat_beam_search_matches_reference_helper():
+  from lingvo_amd.core import beam_search_helper as bsh
+  from lingvo_amd.core import flat_beam_search_helper as fbsh
+  init_fn, step_fn, reorder_fn = _toy_search_fns()
+  ref = bsh.BeamSearchHelper(bsh.BeamSearchHelper.Params().Set(
+      num_hyps_per_beam=4, max_steps=8))
+  flat = fbsh.FlatBeamSearchHelper(fbsh.FlatBeamSearchHelper.Params().Set(
+      num_hyps_per_beam=4, max_steps=8, length_norm_alpha=0.0))
+  out_r = ref.BeamSearchDecode(2, init_fn, step_fn, reorder_fn)
+  out_f = flat.BeamSearchDecode(2, init_fn, step_fn, reorder_fn)
+  # both helpers agree on the best hypothesis per beam
+  for b in range(2):
+    lr = int(out_r.topk_lens[b, 0])
+    lf = int(out_f.topk_lens[b, 0])
+    assert out_r.topk_ids[b, 0, :lr].tolist() == \
+        out_f.topk_ids[b, 0, :lf].tolist(), b
+    assert abs(float(out_r.topk_scores[b, 0]) -
+               float(out_f.topk_scores[b, 0])) < 1e-4
+
+
+def test_flat_beam_search_nbest_sorted_and_unique_pool():
+  import torch
+  from lingvo_amd.core import flat_beam_search_helper as fbsh
+  init_fn, step_fn, reorder_fn = _toy_search_fns(seed=11)
+  flat = fbsh.FlatBeamSearchHelper(fbsh.FlatBeamSearchHelper.Params().Set(
+      num_hyps_per_beam=3, max_steps=6))
+  out = flat.BeamSearchDecode(1, init_fn, step_fn, reorder_fn)
+  s = out.topk_scores[0]
+  assert bool((s[:-1] >= s[1:]).all())  # descending
+  assert bool((s > -1e29).all())        # all slots filled
