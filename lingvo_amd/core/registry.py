@@ -24,6 +24,7 @@ _PARAM_MODULES = [
     'lingvo_amd.models.params.punctuator.codelab',
     'lingvo_amd.models.params.milan.cxc',
     'lingvo_amd.models.params.car.kitti',
+    'lingvo_amd.models.params.car.waymo',
 ]
 
 

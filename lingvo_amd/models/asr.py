@@ -130,6 +130,51 @@ class ConformerEncoder(BaseLayer):
     return x, state
 
 
+class LasEncoder(BaseLayer):
+  """LAS-style ASR encoder: conv subsampling frontend + stacked
+  bidirectional LSTMs (reference tasks/asr/encoder.py:32 AsrEncoder and
+  the Librispeech960Base config, librispeech.py:106-117: conv + 4x
+  biLSTM-1024 with per-layer projection back to model_dim)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('input_dim', 80, 'Mel bins.')
+    p.Define('model_dim', 1024, 'Per-direction LSTM dim * 2.')
+    p.Define('num_lstm_layers', 4, 'biLSTM layers.')
+    p.Define('dropout_prob', 0.0, 'Inter-layer dropout.')
+    p.Define('subsample_channels', 32, 'Frontend conv channels.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    from lingvo_amd.layers import rnn_cell
+    from lingvo_amd.layers import rnn_layers
+    self.CreateChild('sub', conformer_lib.ConvSubsampling.Params().Set(
+        input_freq_dim=p.input_dim, output_dim=p.model_dim,
+        channels=p.subsample_channels))
+    half = p.model_dim // 2
+    layer_ps = []
+    for i in range(p.num_lstm_layers):
+      cell = rnn_cell.LSTMCellSimple.Params().Set(
+          num_input_nodes=p.model_dim, num_output_nodes=half)
+      layer_ps.append(rnn_layers.BidirectionalFRNN.Params().Set(
+          name=f'blstm_{i}', fwd=cell.Copy(), bak=cell.Copy()))
+    self.CreateChildren('rnn', layer_ps)
+
+  def FProp(self, theta: NestedMap, src_inputs: torch.Tensor,
+            paddings: torch.Tensor):
+    x = src_inputs.to(self.fprop_dtype)
+    x, out_pad = self.sub.FProp(theta.sub, x, paddings)
+    for i, layer in enumerate(self.rnn):
+      y = layer.FProp(theta.rnn[i], x, out_pad)
+      if self.p.dropout_prob and not self.do_eval:
+        y = py_utils.DeterministicDropout(y, 1.0 - self.p.dropout_prob)
+      x = y
+    return x, out_pad
+
+
 class AsrDecoder(BaseLayer):
   """Teacher-forced attention LSTM decoder (LAS-style,
   reference tasks/asr/decoder.py:48). Dot-product attention over

@@ -86,6 +86,11 @@ class TransformerLm(BaseLayer):
         model_dim=p.model_dim, num_layers=p.num_layers,
         num_heads=p.num_heads, hidden_dim=p.hidden_dim,
         mask_self_atten=True, remat=p.remat)
+    # propagate GShard-style sharding annotations to the stack so
+    # LowerShardingAnnotations can rewrite it to explicit TP layers
+    if p.weight_split_dims_mapping is not None:
+      stack_p.weight_split_dims_mapping = p.weight_split_dims_mapping
+      stack_p.device_mesh = p.device_mesh
     stack_p.transformer_tpl.tr_atten_tpl.residual_dropout_prob = \
         p.dropout_prob
     stack_p.transformer_tpl.tr_fflayer_tpl.residual_dropout_prob = \

@@ -74,3 +74,34 @@ class Librispeech960ConformerS(Librispeech960WpmConformerL):
     p.encoder.Set(model_dim=256, num_layers=4, num_heads=4)
     p.decoder.Set(rnn_cell_dim=320, source_dim=256)
     return p
+
+
+@registry.RegisterSingleTaskModel
+class Librispeech960Base(Librispeech960WpmConformerL):
+  """LAS baseline (reference librispeech.py:28 Librispeech960Base):
+  conv frontend + 4x biLSTM-1024 encoder, attention LSTM decoder."""
+
+  def Task(self):
+    p = asr_model.AsrModel.Params().Set(name='librispeech_las')
+    p.encoder = asr_model.LasEncoder.Params().Set(
+        input_dim=80, model_dim=1024, num_lstm_layers=4,
+        dropout_prob=0.2)
+    p.fprop_dtype = torch.bfloat16
+    p.train.bf16_weights = True
+    p.decoder.Set(vocab_size=self.VOCAB, emb_dim=128, rnn_cell_dim=1024,
+                  num_lstm_layers=2, source_dim=1024, dropout_prob=0.2)
+    p.train.learner = learner_lib.Learner.Params().Set(
+        learning_rate=2.5e-4,
+        optimizer=optimizer_lib.Adam.Params().Set(
+            beta1=0.9, beta2=0.999, epsilon=1e-6),
+        lr_schedule=schedule_lib.ContinuousSchedule.Params().Set(
+            start_step=50_000, half_life_steps=100_000),
+        clip_gradient_norm_to_value=1.0)
+    return p
+
+
+@registry.RegisterSingleTaskModel
+class Librispeech960BaseGrapheme(Librispeech960Base):
+  """Grapheme-target LAS (reference librispeech.py:156)."""
+
+  VOCAB = 76

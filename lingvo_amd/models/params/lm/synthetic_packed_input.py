@@ -73,3 +73,40 @@ class MoELm64E(DenseLm1B):
              moe_every_n=2, num_experts=self.NUM_EXPERTS,
              expert_capacity_factor=2.0)
     return p
+
+
+@registry.RegisterSingleTaskModel
+class DenseLm8B(DenseLm1B):
+  """~8B-param dense LM (reference synthetic_packed_input.py:53
+  DenseLm8B shape: 32 layers, d=4096). Fits one MI355X (288 GB HBM);
+  for TP runs the stack carries sharding annotations lowered by
+  LowerShardingAnnotations."""
+
+  LAYERS = 32
+  DIM = 4096
+
+  def Task(self):
+    p = super().Task()
+    p.lm.name = 'dense_lm_8b'
+    return p
+
+
+@registry.RegisterSingleTaskModel
+class DenseLm128B8x8(DenseLm8B):
+  """~128B dense LM for 8-way TP x 8-way DP (the scaled GShard family,
+  reference synthetic_packed_input.py:330 DenseLm1T16x16 pattern —
+  sized for one 8-GPU MI355X node instead of a TPU pod: 2.3 TB HBM
+  holds the fp32 masters + bf16 weights sharded 8 ways)."""
+
+  LAYERS = 64
+  DIM = 12288
+  BATCH = 1
+
+  def Task(self):
+    p = super().Task()
+    p.lm.name = 'dense_lm_128b'
+    # GShard-style annotation; propagated to the transformer stack and
+    # lowered to explicit column/row-parallel layers + RCCL collectives
+    # by parallel.tensor_parallel.LowerShardingAnnotations.
+    p.lm.weight_split_dims_mapping = [-1, 0]
+    return p

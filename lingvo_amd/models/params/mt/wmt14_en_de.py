@@ -63,3 +63,14 @@ class WmtEnDeTransformerBig(WmtEnDeTransformerBase):
   DIM = 1024
   FF = 4096
   HEADS = 16
+
+
+@registry.RegisterSingleTaskModel
+class WmtEnDeTransformerSmall(WmtEnDeTransformerBase):
+  """Transformer small (reference wmt14_en_de.py:100): d=256, ff=1024,
+  4 heads, 2+2 layers — debugging config."""
+
+  DIM = 256
+  FF = 1024
+  HEADS = 4
+  LAYERS = 2

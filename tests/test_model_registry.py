@@ -46,3 +46,48 @@ def test_small_models_train_step(key):
   task = model.GetTask()
   m = task.TrainStep(task.GetInputBatch())
   assert torch.isfinite(m['loss'][0])
+
+
+def test_las_encoder_small_train_step():
+  """LAS (conv + biLSTM) ASR variant instantiates and steps on CPU."""
+  model_p = registry.GetParams('asr.librispeech.Librispeech960Base',
+                               'Train')
+  model_p.task.fprop_dtype = torch.float32
+  model_p.task.train.bf16_weights = False
+  model_p.task.random_seed = 5
+  model_p.task.encoder.Set(model_dim=64, num_lstm_layers=2,
+                           subsample_channels=8)
+  model_p.task.decoder.Set(rnn_cell_dim=32, source_dim=64, emb_dim=16,
+                           vocab_size=32)
+  model_p.input.Set(batch_size=2, frame_len=32, target_len=6,
+                    vocab_size=32)
+  model = model_p.Instantiate()
+  task = model.GetTask()
+  m = task.TrainStep(task.GetInputBatch())
+  assert torch.isfinite(m['loss'][0])
+
+
+def test_dense_lm_family_and_waymo_registered():
+  keys = registry.GetAllRegisteredClasses()
+  for k in ['lm.synthetic_packed_input.DenseLm8B',
+            'lm.synthetic_packed_input.DenseLm128B8x8',
+            'mt.wmt14_en_de.WmtEnDeTransformerSmall',
+            'asr.librispeech.Librispeech960Base',
+            'car.waymo.WaymoPillars']:
+    assert k in keys, k
+  # the 128B config carries TP sharding annotations
+  p = registry.GetParams('lm.synthetic_packed_input.DenseLm128B8x8',
+                         'Train')
+  assert p.task.lm.weight_split_dims_mapping == [-1, 0]
+
+
+def test_waymo_pillars_small_train_step():
+  model_p = registry.GetParams('car.waymo.WaymoPillars', 'Train')
+  model_p.task.random_seed = 2
+  model_p.task.Set(grid_size=16, backbone_channels=[8, 16],
+                   point_feat_dim=8)
+  model_p.input.Set(batch_size=2, num_points=128)
+  model = model_p.Instantiate()
+  task = model.GetTask()
+  m = task.TrainStep(task.GetInputBatch())
+  assert torch.isfinite(m['loss'][0])
