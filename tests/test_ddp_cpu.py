@@ -135,10 +135,11 @@ def _run_sharded_ckpt(rank, world, port, tmpdir, results):
 
 def test_sharded_checkpoint_roundtrip(tmp_path):
   ctx = mp.get_context('spawn')
+  port = 29563 + os.getpid() % 997  # avoid TIME_WAIT collisions
   with ctx.Manager() as mgr:
     results = mgr.dict()
     procs = [ctx.Process(target=_run_sharded_ckpt,
-                         args=(r, 2, 29563, str(tmp_path), results))
+                         args=(r, 2, port, str(tmp_path), results))
              for r in range(2)]
     for p in procs:
       p.start()
