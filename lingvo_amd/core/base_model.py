@@ -70,6 +70,10 @@ class BaseTask(BaseLayer):
     tp.Define('start_up_delay_steps', 200, 'Unused on MI355X; kept for '
               'config parity.')
     tp.Define('vn_std', 0.0, 'Variational noise std (0 disables).')
+    tp.Define('pruner_hparams', None,
+              'dict of MagnitudePruner kwargs (reference '
+              'base_model.py:1105 _GetMaskUpdateOp model_pruning hook); '
+              'None disables magnitude pruning.')
     p.Define('train', tp, 'Training hyperparameters subtree.')
     ep = Params()
     ep.Define('samples_per_summary', 1000, 'Eval samples per summary.')
@@ -151,6 +155,12 @@ class BaseTask(BaseLayer):
         for k, v in lmetrics.items():
           metrics[k if i == 0 else f'{k}_{i}'] = v
     self.global_step_var += 1
+    if self.p.train.pruner_hparams is not None:
+      if not hasattr(self, '_pruner'):
+        from lingvo_amd.core.pruning_utils import MagnitudePruner
+        self._pruner = MagnitudePruner(self,
+                                       **self.p.train.pruner_hparams)
+      self._pruner.Prune(int(self.global_step))
     self.PostTrainingStepUpdate(self.global_step)
     if self._ema is not None:
       self._ema.Update(self.named_parameters())

@@ -152,3 +152,19 @@ def test_saver_rejects_nonfinite(tmp_path):
   bad = {'model': {'w': torch.tensor([1.0, float('nan')])}, 'step': 1}
   with pytest.raises(FloatingPointError):
     saver.Save(bad, 1)
+
+
+def test_pruning_hook_in_train_loop():
+  """train.pruner_hparams wires MagnitudePruner into TrainStep."""
+  from lingvo_amd.core import registry
+  model_p = registry.GetParams('image.mnist.LeNet5', 'Train')
+  model_p.task.random_seed = 6
+  model_p.task.train.pruner_hparams = dict(
+      weight_regex='fc', final_sparsity=0.6, begin_step=0, end_step=4,
+      frequency=1, min_numel=64)
+  model = model_p.Instantiate()
+  task = model.GetTask()
+  for _ in range(5):
+    task.TrainStep(task.GetInputBatch())
+  sp = task._pruner.MeasuredSparsity()
+  assert abs(sp - 0.6) < 0.05, sp
