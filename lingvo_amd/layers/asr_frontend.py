@@ -86,3 +86,39 @@ class MelAsrFrontend(BaseLayer):
         (sample_lens - self._win) // self._hop + 1, min=0, max=t)
     out_paddings = py_utils.PaddingsFromLengths(frame_lens, t)
     return py_utils.ApplyPadding(out_paddings, logmel), out_paddings
+
+  # ---- streaming (reference frontend.py:413 chunk mode) ---------------
+  def InitStreamState(self, batch: int, device=None) -> NestedMap:
+    return NestedMap(
+        buf=torch.zeros(batch, 0, device=device),
+        last=torch.zeros(batch, 1, device=device))
+
+  def StreamStep(self, theta: NestedMap, wav_chunk: torch.Tensor,
+                 state: NestedMap):
+    """Exact chunked framing: emits every frame whose full window is
+    available, carrying the sub-window tail and the preemphasis
+    history sample. Concatenating StreamStep outputs equals FProp on
+    the whole waveform (no-padding case)."""
+    p = self.p
+    x = wav_chunk.float()
+    prev = torch.cat([state.last, x[:, :-1]], dim=1)
+    xp = x - p.preemph * prev if p.preemph else x
+    state.last = x[:, -1:]
+    buf = torch.cat([state.buf, xp], dim=1)
+    n = buf.shape[1]
+    # torch.stft with center=False frames by n_fft (the window is
+    # zero-padded up to it), so completeness is in n_fft units.
+    nframes = max(0, (n - self._fft) // self._hop + 1)
+    if nframes == 0:
+      state.buf = buf
+      return torch.zeros(x.shape[0], 0, p.num_bins, device=x.device), \
+          state
+    used = buf[:, :self._fft + (nframes - 1) * self._hop]
+    spec = torch.stft(used, n_fft=self._fft, hop_length=self._hop,
+                      win_length=self._win, window=self.window,
+                      center=False, return_complex=True)
+    power = spec.abs() ** 2
+    mel = torch.matmul(power.transpose(1, 2), self.mel_fb)
+    logmel = torch.log(mel.clamp_min(p.mel_floor))
+    state.buf = buf[:, nframes * self._hop:]
+    return logmel, state

@@ -412,3 +412,68 @@ class AsrModel(BaseTask):
       ref = ' '.join(str(int(x)) for x in refs[i] if int(x) > 2)
       decode_metrics.wer.Update(ref, hyp)
     decode_metrics.num_samples_in_batch.Update(float(hyps.shape[0]))
+
+
+class StreamingRecognizer:
+  """End-to-end streaming ASR driver (reference: the streaming chunk
+  mode of frontend.py:413 + conformer StreamStep composition).
+
+  Audio chunks stream through the exact chunked Mel frontend; the
+  accumulated mel features run through the (local) conv subsampling
+  with the last output frame held back — 'same' right-edge padding only
+  affects the final frame, so every released frame is final — and new
+  subsampled frames stream through the conformer stack's StreamStep.
+  Finish() flushes the held-back frame and greedy-decodes.
+
+  Requires a causal encoder config (is_causal, conv_norm='layer').
+  """
+
+  def __init__(self, model: AsrModel, frontend, batch: int,
+               max_enc_frames: int = 4096, device='cpu'):
+    self.model = model
+    self.frontend = frontend
+    self.batch = batch
+    self.device = device
+    self.fe_state = frontend.InitStreamState(batch, device)
+    self.enc_state = model.encoder.InitStreamState(
+        model.theta.encoder, batch, max_enc_frames, device,
+        torch.float32)
+    self.mel = torch.zeros(batch, 0, frontend.p.num_bins, device=device)
+    self.emitted = 0          # subsampled frames already streamed
+    self.enc_frames = []
+
+  def _SubAll(self, final: bool):
+    enc = self.model.encoder
+    t = self.mel.shape[1]
+    if t == 0:
+      return
+    feats, out_pad = enc.sub.FProp(self.model.theta.encoder.sub,
+                                   self.mel, torch.zeros(
+                                       self.batch, t, device=self.device))
+    avail = feats.shape[1] if final else max(0, feats.shape[1] - 1)
+    if avail > self.emitted:
+      new = feats[:, self.emitted:avail]
+      out, self.enc_state = enc.StreamStep(
+          self.model.theta.encoder, new,
+          torch.zeros(self.batch, new.shape[1], device=self.device),
+          self.enc_state)
+      self.enc_frames.append(out)
+      self.emitted = avail
+
+  @torch.no_grad()
+  def Push(self, wav_chunk: torch.Tensor) -> None:
+    frames, self.fe_state = self.frontend.StreamStep(
+        self.frontend.theta, wav_chunk, self.fe_state)
+    if frames.shape[1]:
+      self.mel = torch.cat([self.mel, frames], dim=1)
+      self._SubAll(final=False)
+
+  @torch.no_grad()
+  def Finish(self) -> NestedMap:
+    self._SubAll(final=True)
+    enc = torch.cat(self.enc_frames, dim=1) if self.enc_frames else \
+        torch.zeros(self.batch, 0, self.model.encoder.p.model_dim)
+    pad = torch.zeros(self.batch, enc.shape[1], device=self.device)
+    hyps = self.model.decoder.GreedyDecode(self.model.theta.decoder,
+                                           enc, pad)
+    return NestedMap(encoded=enc, hyps=hyps)

@@ -130,3 +130,59 @@ def test_conformer_encoder_stack_streaming():
     outs.append(o)
   stream = torch.cat(outs, dim=1)
   assert (x - stream).abs().max() < 1e-3
+
+
+def test_mel_frontend_streaming_matches_full():
+  import torch
+  from lingvo_amd.layers import asr_frontend
+  fe = asr_frontend.MelAsrFrontend.Params().Set(name='fe').Instantiate()
+  g = torch.Generator().manual_seed(8)
+  wav = torch.randn(2, 4000, generator=g)
+  full, _ = fe.FProp(fe.theta, wav, torch.zeros(2, 4000))
+  st = fe.InitStreamState(2)
+  outs = []
+  for c0 in range(0, 4000, 700):  # uneven chunks vs hop=160
+    o, st = fe.StreamStep(fe.theta, wav[:, c0:c0 + 700], st)
+    if o.shape[1]:
+      outs.append(o)
+  stream = torch.cat(outs, dim=1)
+  assert stream.shape[1] == full.shape[1], (stream.shape, full.shape)
+  assert (stream - full).abs().max() < 1e-4
+
+
+def test_streaming_recognizer_matches_offline():
+  """Audio-chunk streaming == offline encoder + greedy decode."""
+  import torch
+  from lingvo_amd.core.nested_map import NestedMap
+  from lingvo_amd.layers import asr_frontend
+  from lingvo_amd.models import asr as asr_lib
+  ep = asr_lib.ConformerEncoder.Params().Set(
+      name='enc', input_dim=80, model_dim=32, num_layers=2, num_heads=2,
+      kernel_size=4, dropout_prob=0.0, specaug_tpl=None, random_seed=13)
+  ep.conformer_tpl.is_causal = True
+  ep.conformer_tpl.conv_norm = 'layer'
+  ep.conformer_tpl.atten_left_context = 64
+  mp = asr_lib.AsrModel.Params().Set(name='m', encoder=ep,
+                                     random_seed=13)
+  mp.decoder.Set(vocab_size=16, emb_dim=8, rnn_cell_dim=16,
+                 source_dim=32, dropout_prob=0.0)
+  model = mp.Instantiate()
+  model.eval()
+  fe = asr_frontend.MelAsrFrontend.Params().Set(name='fe').Instantiate()
+  g = torch.Generator().manual_seed(3)
+  wav = torch.randn(2, 16000, generator=g)
+
+  # offline: frontend -> encoder -> greedy
+  with torch.no_grad():
+    mel, mel_pad = fe.FProp(fe.theta, wav, torch.zeros(2, 16000))
+    enc, enc_pad = model.encoder.FProp(model.theta.encoder, mel, mel_pad)
+    hyps_off = model.decoder.GreedyDecode(model.theta.decoder, enc,
+                                          enc_pad)
+
+  rec = asr_lib.StreamingRecognizer(model, fe, batch=2)
+  for c0 in range(0, 16000, 3000):
+    rec.Push(wav[:, c0:c0 + 3000])
+  out = rec.Finish()
+  assert out.encoded.shape == enc.shape, (out.encoded.shape, enc.shape)
+  assert (out.encoded - enc).abs().max() < 1e-3
+  assert torch.equal(out.hyps, hyps_off)
