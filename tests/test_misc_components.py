@@ -440,3 +440,19 @@ def test_flat_beam_search_nbest_sorted_and_unique_pool():
   s = out.topk_scores[0]
   assert bool((s[:-1] >= s[1:]).all())  # descending
   assert bool((s > -1e29).all())        # all slots filled
+
+
+def test_matplotlib_figure_summary():
+  import torch
+  from lingvo_amd.core import plot
+  probs = torch.rand(3, 10, 12)
+  spec = torch.rand(3, 80, 50)
+  fig = plot.MatplotlibFigureSummary('diag', max_outputs=2)
+  fig.AddSubplot([probs], title='atten', xlabel='src', ylabel='tgt')
+  fig.AddSubplot([spec], title='spectrogram')
+  imgs = fig.Finalize()
+  assert imgs.shape[0] == 2 and imgs.shape[-1] == 3
+  assert imgs.dtype == torch.uint8
+  assert int(imgs.float().std()) >= 0  # rendered, non-degenerate
+  one = plot.AttentionSummary('a', probs, max_outputs=1)
+  assert one.shape[0] == 1
