@@ -58,13 +58,19 @@ class GradSync:
 
   def __init__(self, module: torch.nn.Module, bucket_cap_mb: float = 25,
                grad_dtype: Optional[torch.dtype] = None,
-               process_group=None):
+               process_group=None, compressor=None):
+    """compressor: optional GradDropCompressor-like object whose
+    .compress(name, grad) sparsifies each gradient (with local error
+    feedback) before it enters the all-reduce bucket (reference
+    graddrop.py wired at the CRS boundary)."""
     self._pg = process_group
     self._world = dist.get_world_size(process_group) if \
         dist.is_initialized() else 1
     self._hooks = []
     self._buckets: List[_Bucket] = []
     self._param_bucket = {}
+    self._comp = compressor
+    self._names = {id(p): n for n, p in module.named_parameters()}
     if self._world <= 1:
       return
     params = [p for p in module.parameters() if p.requires_grad]
@@ -100,8 +106,11 @@ class GradSync:
     def hook(param):
       off = bkt.offsets[id(param)]
       if param.grad is not None:
+        g = param.grad.detach()
+        if self._comp is not None:
+          g = self._comp.compress(self._names[id(param)], g.float())
         bkt.buffer[off:off + param.numel()].copy_(
-            param.grad.detach().reshape(-1).to(self._dtype))
+            g.reshape(-1).to(self._dtype))
         bkt.ready.add(id(param))
       if len(bkt.ready) == len(bkt.params) and bkt.work is None:
         bkt.work = dist.all_reduce(bkt.buffer, op=dist.ReduceOp.SUM,
@@ -122,8 +131,11 @@ class GradSync:
           if id(p) not in bkt.ready:
             off = bkt.offsets[id(p)]
             if p.grad is not None:
+              g = p.grad.detach()
+              if self._comp is not None:
+                g = self._comp.compress(self._names[id(p)], g.float())
               bkt.buffer[off:off + p.numel()].copy_(
-                  p.grad.detach().reshape(-1).to(self._dtype))
+                  g.reshape(-1).to(self._dtype))
             else:
               bkt.buffer[off:off + p.numel()].zero_()
         bkt.work = dist.all_reduce(bkt.buffer, op=dist.ReduceOp.SUM,

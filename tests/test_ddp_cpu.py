@@ -153,3 +153,50 @@ def test_sharded_checkpoint_roundtrip(tmp_path):
   with open(tmp_path / 'checkpoint') as f:
     txt = f.read()
   assert 'ckpt-00000007' in txt and 'num_shards: 2' in txt
+
+
+def _run_graddrop_sync(rank, world, port, results):
+  os.environ['MASTER_ADDR'] = '127.0.0.1'
+  os.environ['MASTER_PORT'] = str(port)
+  dist.init_process_group('gloo', rank=rank, world_size=world)
+  from lingvo_amd.core.optimizer_experiments import GradDropCompressor
+  from lingvo_amd.parallel.ddp import GradSync
+  lin = torch.nn.Linear(4, 1, bias=False)
+  with torch.no_grad():
+    lin.weight.fill_(0.0)
+  comp = GradDropCompressor(keep_frac=0.25)
+  sync = GradSync(lin, compressor=comp)
+  # per-rank distinct grads: rank r has one large element at index r.
+  # Set .grad directly (no backward) to exercise the Finalize pull
+  # path, which also runs the compressor.
+  g = torch.full((1, 4), 0.1)
+  g[0, rank] = 4.0
+  lin.weight.grad = g.clone()
+  sync.Finalize()
+  results[f'grad{rank}'] = lin.weight.grad.clone()
+  results[f'res{rank}'] = comp._residual['weight'].clone()
+  dist.destroy_process_group()
+
+
+def test_gradsync_with_graddrop_compressor():
+  ctx = mp.get_context('spawn')
+  port = 29570 + os.getpid() % 997
+  with ctx.Manager() as mgr:
+    results = mgr.dict()
+    procs = [ctx.Process(target=_run_graddrop_sync,
+                         args=(r, 2, port, results))
+             for r in range(2)]
+    for p in procs:
+      p.start()
+    for p in procs:
+      p.join(120)
+      assert p.exitcode == 0
+    g0, g1 = results['grad0'], results['grad1']
+    res0 = results['res0']
+  # both ranks see the same averaged compressed grad: rank r kept only
+  # its 4.0 element -> average [2.0, 2.0, 0, 0]
+  assert torch.allclose(g0, g1)
+  assert torch.allclose(g0, torch.tensor([[2.0, 2.0, 0.0, 0.0]]),
+                        atol=1e-5)
+  # dropped 0.1 elements live in the residual for the next step
+  assert abs(float(res0[0, 2]) - 0.1) < 1e-6
