@@ -290,3 +290,71 @@ class PerformerAttention(attention_lib.MultiHeadedAttention):
     if paddings is not None:
       post = py_utils.ApplyPadding(paddings, post)
     return post
+
+
+class BlockSparseAttention(attention_lib.MultiHeadedAttention):
+  """Attention under a fixed block-level visibility pattern (reference
+  self_attention_layer.py:30 BlockSparseAttention): queries in block i
+  attend keys in block j iff block_mask[i][j]. Patterns like
+  local+global land as small boolean matrices; on GPU the masked
+  blocks are skipped by the flash kernel's tile loop (round 2), the
+  dense-mask fp32 form below is the oracle."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('block_size', 64, 'Block width in tokens.')
+    p.Define('block_mask', None,
+             'Nested list / tensor [nq_blocks, nk_blocks] of 0/1; '
+             'None = full attention.')
+    p.cls = cls
+    return p
+
+  def FProp(self, theta: NestedMap, query_vec: torch.Tensor,
+            paddings: Optional[torch.Tensor] = None,
+            segment_ids: Optional[torch.Tensor] = None) -> torch.Tensor:
+    p = self.p
+    assert segment_ids is None
+    b, t, _ = query_vec.shape
+    n, h = self._n, self._h
+    assert self._nkv == n
+    q, k, v = self._Project(theta, query_vec)
+    scale = 1.0 / math.sqrt(h)
+    logits = torch.einsum('btnh,bsnh->bnts', q.float(),
+                          k.float()) * scale
+    tpos = torch.arange(t, device=q.device)
+    mask = torch.ones(t, t, dtype=torch.bool, device=q.device)
+    if p.block_mask is not None:
+      bm = torch.as_tensor(p.block_mask, dtype=torch.bool,
+                           device=q.device)
+      qb = (tpos // p.block_size).clamp(max=bm.shape[0] - 1)
+      kb = (tpos // p.block_size).clamp(max=bm.shape[1] - 1)
+      mask = bm[qb[:, None], kb[None, :]]
+    if p.causal:
+      mask = mask & (tpos[None, :] <= tpos[:, None])
+    mask4 = mask[None, None].expand(b, 1, t, t).clone()
+    if paddings is not None:
+      mask4 = mask4 & (paddings[:, None, None, :] < 0.5)
+    logits = logits.masked_fill(~mask4, -1e30)
+    probs = torch.softmax(logits, dim=-1)
+    probs = torch.where(mask4.any(-1, keepdim=True), probs,
+                        torch.zeros_like(probs))
+    ctx = torch.einsum('bnts,bsnh->btnh', probs, v.float())
+    ctx = ctx.reshape(b, t, n * h).to(query_vec.dtype)
+    post = py_utils.MatmulBias(ctx, theta.post_w,
+                               theta.post_b if p.use_bias else None)
+    if paddings is not None:
+      post = py_utils.ApplyPadding(paddings, post)
+    return post
+
+
+def LocalGlobalBlockMask(num_blocks: int, num_global: int = 1,
+                         local_width: int = 1) -> list:
+  """Common block pattern: the first `num_global` blocks see/are seen
+  by everything; other blocks see +-local_width neighbors."""
+  m = [[0] * num_blocks for _ in range(num_blocks)]
+  for i in range(num_blocks):
+    for j in range(num_blocks):
+      if i < num_global or j < num_global or abs(i - j) <= local_width:
+        m[i][j] = 1
+  return m

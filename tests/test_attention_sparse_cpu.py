@@ -130,3 +130,29 @@ def test_performer_causal_prefix_property():
   out2 = layer.FProp(layer.theta, x2, pad)
   assert (out1[:, :5] - out2[:, :5]).abs().max() < 1e-5
   assert (out1[:, 5:] - out2[:, 5:]).abs().max() > 1e-4
+
+
+def test_block_sparse_attention():
+  full_mask = [[1, 1], [1, 1]]
+  layer = _mk(asp.BlockSparseAttention, block_size=4,
+              block_mask=full_mask)
+  ref = _mk(attention_lib.MultiHeadedAttention)
+  ref.load_state_dict(layer.state_dict(), strict=False)
+  g = torch.Generator().manual_seed(12)
+  x = torch.randn(2, 8, 32, generator=g)
+  pad = torch.zeros(2, 8)
+  # full mask == dense attention
+  assert (layer.FProp(layer.theta, x, pad) -
+          ref.FProp(ref.theta, x, pad)).abs().max() < 1e-4
+  # block-diagonal mask: cross-block influence is zero
+  diag = _mk(asp.BlockSparseAttention, block_size=4,
+             block_mask=[[1, 0], [0, 1]])
+  o1 = diag.FProp(diag.theta, x, pad)
+  x2 = x.clone()
+  x2[:, 6] += 9.0
+  o2 = diag.FProp(diag.theta, x2, pad)
+  assert (o1[:, :4] - o2[:, :4]).abs().max() < 1e-5
+  assert (o1[:, 4:] - o2[:, 4:]).abs().max() > 1e-3
+  # local+global pattern helper
+  m = asp.LocalGlobalBlockMask(4, num_global=1, local_width=1)
+  assert m[0] == [1, 1, 1, 1] and m[3] == [1, 0, 1, 1]
