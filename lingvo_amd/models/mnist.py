@@ -120,3 +120,15 @@ class ModelV1(BaseClassifier):
       preds = self.ComputePredictions(self.theta, input_batch)
     return NestedMap(correct_top1=(
         preds.logits.argmax(-1) == input_batch.label.long()).float())
+
+  def Inference(self) -> NestedMap:
+    """Named inference subgraphs (reference base_model.py:943)."""
+
+    def default(images):
+      preds = self.ComputePredictions(self.theta,
+                                      NestedMap(data=images))
+      return NestedMap(logits=preds.logits,
+                       probs=torch.softmax(preds.logits.float(), -1),
+                       label=preds.logits.argmax(-1))
+
+    return NestedMap(default=default)

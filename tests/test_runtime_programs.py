@@ -126,3 +126,32 @@ def test_mlperf_logging_emitted(tmp_path, capsys):
   keys = [json.loads(l.split(' ', 1)[1])['key'] for l in lines]
   assert 'run_start' in keys and 'run_stop' in keys
   assert 'block_stop' in keys
+
+
+def test_inference_server_endpoints(tmp_path):
+  """ASGI in-process test of the serving front-end (no sockets)."""
+  from lingvo_amd.runtime.inference import InferenceGraphExporter, Predictor
+  from lingvo_amd.runtime.server import MakeApp
+  from fastapi.testclient import TestClient
+
+  model_p = registry.GetParams('image.mnist.LeNet5', 'Train')
+  model_p.task.random_seed = 9
+  path = str(tmp_path / 'inference.pt')
+  InferenceGraphExporter.Export(model_p, path)
+  app = MakeApp(Predictor(path, device='cpu'))
+  client = TestClient(app)
+
+  r = client.get('/health')
+  assert r.status_code == 200 and r.json()['status'] == 'ok'
+  subgraphs = r.json()['subgraphs']
+  assert 'default' in subgraphs
+
+  imgs = torch.zeros(2, 28, 28, 1).tolist()
+  r = client.post('/predict/default', json={'images': imgs})
+  assert r.status_code == 200, r.text
+  out = r.json()
+  key = 'logits' if 'logits' in out else sorted(out)[0]
+  assert len(out[key]) == 2
+
+  r = client.post('/predict/nope', json={})
+  assert r.status_code == 404
