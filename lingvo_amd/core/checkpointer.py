@@ -336,3 +336,21 @@ class ShardedCheckpointer(Checkpointer):
       if f'optimizer_{i}' in payload:
         opt.load_state_dict(payload[f'optimizer_{i}'])
     return int(payload.get('step', step))
+
+
+def WriteNpArrays(path_prefix: str, arrays: Dict) -> None:
+  """Numpy checkpoint IO (reference saver.py:574 WriteNpArrays): saves
+  a {name: tensor/ndarray} dict as an .npz alongside the torch
+  checkpoints (interchange format for non-torch consumers)."""
+  import numpy as np
+  np.savez(path_prefix + '.npz',
+           **{k: (v.detach().cpu().numpy()
+                  if isinstance(v, torch.Tensor) else v)
+              for k, v in arrays.items()})
+
+
+def ReadNpArrays(path_prefix: str) -> Dict:
+  """Inverse of WriteNpArrays; returns {name: torch.Tensor}."""
+  import numpy as np
+  with np.load(path_prefix + '.npz') as data:
+    return {k: torch.from_numpy(data[k].copy()) for k in data.files}

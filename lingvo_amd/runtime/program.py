@@ -206,3 +206,44 @@ class Executor:
       from lingvo_amd.utils import helpers
       helpers.mlperf_print('run_stop',
                            metadata={'step': self.task.global_step})
+
+
+class InputBenchmark(BaseProgram):
+  """Times the input pipeline alone (reference program.py:2249
+  InputBenchmark): steps_per_loop GetInputBatch+ToDevice calls,
+  reporting batches/sec."""
+
+  def Run(self) -> NestedMap:
+    task = self.task
+    t0 = time.perf_counter()
+    n = 0
+    for _ in range(self.p.steps_per_loop):
+      batch = task.GetInputBatch()
+      task.input_generator.ToDevice(batch, self.device)
+      n += 1
+    dt = time.perf_counter() - t0
+    rate = n / max(1e-9, dt)
+    self._Log({'batches_per_sec': rate, 'steps': n})
+    return NestedMap(batches_per_sec=rate)
+
+
+class MultiTaskProgramSchedule:
+  """Per-step task sampling over a MultiTaskModel (reference
+  program.py:2319): each Run samples a task via the model's
+  task_scheduler and executes that task's train program."""
+
+  def __init__(self, model, logdir: str, device: str,
+               steps_per_loop: int = 1, grad_syncs=None):
+    self.model = model
+    self.programs = {}
+    for name in model.task_names:
+      tp = TrainProgram.Params().Set(name=f'train_{name}',
+                                     steps_per_loop=steps_per_loop)
+      self.programs[name] = TrainProgram(
+          tp, model.GetTask(name), logdir, device,
+          (grad_syncs or {}).get(name))
+
+  def Run(self) -> NestedMap:
+    name = self.model.SampleTask()
+    out = self.programs[name].Run()
+    return NestedMap(task=name, **out)

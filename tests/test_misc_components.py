@@ -456,3 +456,41 @@ def test_matplotlib_figure_summary():
   assert int(imgs.float().std()) >= 0  # rendered, non-degenerate
   one = plot.AttentionSummary('a', probs, max_outputs=1)
   assert one.shape[0] == 1
+
+
+def test_adaptive_and_piecewise_schedulers():
+  from lingvo_amd.core import task_scheduler as ts
+  a = ts.AdaptiveScheduler.Params().Set(
+      name='a', tasks=['x', 'y'], targets=[1.0, 1.0],
+      random_seed=3).Instantiate()
+  a.ReportMetric('x', 1.01)   # nearly converged
+  a.ReportMetric('y', 10.0)   # far from target
+  picks = [a.Sample(0) for _ in range(200)]
+  assert picks.count('y') > picks.count('x') * 3
+
+  const = ts.ConstantScheduler.Params().Set(task_probs=[('x', 1.0)])
+  const2 = ts.ConstantScheduler.Params().Set(task_probs=[('y', 1.0)])
+  pw = ts.PieceWiseScheduler.Params().Set(
+      name='pw', schedule_steps=[(const, 10), (const2, 10**9)]
+  ).Instantiate()
+  assert pw.Sample(5) == 'x' and pw.Sample(50) == 'y'
+
+
+def test_input_benchmark_and_np_arrays(tmp_path):
+  import torch
+  from lingvo_amd.core import registry
+  from lingvo_amd.runtime import program as program_lib
+  model_p = registry.GetParams('image.mnist.LeNet5', 'Train')
+  model_p.task.random_seed = 3
+  model = model_p.Instantiate()
+  prog = program_lib.InputBenchmark(
+      program_lib.InputBenchmark.Params().Set(steps_per_loop=3),
+      model.GetTask(), str(tmp_path), 'cpu')
+  out = prog.Run()
+  assert out.batches_per_sec > 0
+
+  from lingvo_amd.core import checkpointer as ckpt_lib
+  ckpt_lib.WriteNpArrays(str(tmp_path / 'arrays'),
+                         {'w': torch.arange(6).reshape(2, 3)})
+  back = ckpt_lib.ReadNpArrays(str(tmp_path / 'arrays'))
+  assert torch.equal(back['w'], torch.arange(6).reshape(2, 3))
