@@ -160,6 +160,7 @@ class SimpleProgramSchedule:
       out.train = self.train_program.Run()
     for prog in self.eval_programs:
       out[prog.p.name or 'eval'] = prog.Run()
+    self.last_result = out
     return out
 
 
@@ -170,7 +171,10 @@ class Executor:
   def __init__(self, model_params: InstantiableParams, logdir: str,
                schedule_params: Optional[Params] = None,
                device: Optional[str] = None, max_steps: Optional[int]
-               = None, grad_sync=None):
+               = None, grad_sync=None, trial=None):
+    """trial: optional utils.helpers.Trial for hyperparameter search —
+    eval results are reported and its ShouldStop() ends the run early
+    (reference base_trial.py hooks in runners/executor)."""
     self.device = device or ('cuda:0' if torch.cuda.is_available()
                              else 'cpu')
     self.model = model_params.Instantiate().to(self.device)
@@ -180,6 +184,8 @@ class Executor:
     sched_p = schedule_params or SimpleProgramSchedule.Params()
     self.schedule = SimpleProgramSchedule(sched_p, self.task, logdir,
                                           self.device, grad_sync)
+    from lingvo_amd.utils import helpers
+    self.trial = trial or helpers.NoOpTrial()
     self.ckpt = Checkpointer(
         Checkpointer.Params().Set(save_interval_seconds=600),
         os.path.join(logdir, 'train'), self.model,
@@ -197,6 +203,12 @@ class Executor:
       cycle_secs = time.perf_counter() - t0
       self.ckpt.MaybeSave()
       step = self.task.global_step
+      out = getattr(self.schedule, 'last_result', None)
+      if out is not None and 'train' in out:
+        self.trial.ReportEvalMeasure(step,
+                                     {'loss': float(out.train.loss)}, '')
+      if self.trial.ShouldStop():
+        break
       limit = self.max_steps or self.task.p.train.max_steps
       if limit is not None and step >= limit:
         break

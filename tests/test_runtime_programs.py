@@ -155,3 +155,29 @@ def test_inference_server_endpoints(tmp_path):
 
   r = client.post('/predict/nope', json={})
   assert r.status_code == 404
+
+
+def test_trial_early_stop_in_executor(tmp_path):
+  from lingvo_amd.runtime import program as program_lib
+  from lingvo_amd.utils import helpers
+
+  class StopAfterOne(helpers.Trial):
+    def __init__(self):
+      self.reports = []
+    def ReportEvalMeasure(self, step, metrics, ckpt):
+      self.reports.append((step, metrics))
+      return False
+    def ShouldStop(self):
+      return len(self.reports) >= 1
+
+  mp = registry.GetParams('image.mnist.LeNet5', 'Train')
+  mp.task.random_seed = 3
+  sched = program_lib.SimpleProgramSchedule.Params()
+  sched.train_program.steps_per_loop = 2
+  trial = StopAfterOne()
+  ex = program_lib.Executor(mp, str(tmp_path), sched, device='cpu',
+                            max_steps=100, trial=trial)
+  ex.Start()
+  # stopped by the trial after one loop, far before max_steps
+  assert ex.task.global_step == 2
+  assert trial.reports and 'loss' in trial.reports[0][1]
