@@ -25,7 +25,9 @@ from lingvo_amd.layers import activations
 
 
 class ProjectionLayer(BaseLayer):
-  """Linear projection + optional bias + activation."""
+  """Linear projection + optional bias + activation. Optional
+  quantization-aware training via qdomain_tpl (reference
+  layers.py:845 ProjectionLayer QWeight/QAct wiring)."""
 
   @classmethod
   def Params(cls):
@@ -34,6 +36,8 @@ class ProjectionLayer(BaseLayer):
     p.Define('output_dim', 0, 'Output dimension.')
     p.Define('has_bias', False, 'Add bias.')
     p.Define('activation', 'NONE', 'Activation name.')
+    p.Define('qdomain_tpl', None,
+             'QDomain params; None disables fake quantization.')
     return p
 
   def __init__(self, params):
@@ -45,16 +49,28 @@ class ProjectionLayer(BaseLayer):
     if p.has_bias:
       self.CreateVariable('b', py_utils.WeightParams(
           [p.output_dim], py_utils.WeightInit.Constant(0.0), p.dtype))
+    if p.qdomain_tpl is not None:
+      self.CreateChild('qdomain', p.qdomain_tpl)
 
   def FProp(self, theta: NestedMap, inputs: torch.Tensor,
             paddings: Optional[torch.Tensor] = None) -> torch.Tensor:
     p = self.p
-    out = py_utils.MatmulBias(inputs, theta.w,
+    w = theta.w
+    if p.qdomain_tpl is not None:
+      w = self.qdomain.QuantizeTensor(w, calibrate=False)
+    out = py_utils.MatmulBias(inputs, w,
                               theta.b if p.has_bias else None)
     out = activations.GetFn(p.activation)(out)
+    if p.qdomain_tpl is not None:
+      out = self.qdomain.QuantizeTensor(out, calibrate=True)
     if paddings is not None:
       out = py_utils.ApplyPadding(paddings, out)
     return out
+
+  def PostTrainingStepUpdate(self, global_step: int) -> None:
+    if self.p.qdomain_tpl is not None:
+      self.qdomain.SetStep(global_step)
+    super().PostTrainingStepUpdate(global_step)
 
 
 class FCLayer(ProjectionLayer):

@@ -114,3 +114,27 @@ def test_fake_quant_schedule_and_asym_domain():
   dom.QuantizeNamedTensor('act', x2).sum().backward()
   assert torch.allclose(x2.grad, torch.ones_like(x2))
   assert 'act' in dom.state_dict_ranges()
+
+
+def test_projection_layer_qdomain():
+  import torch
+  from lingvo_amd.core import quant_utils
+  from lingvo_amd.layers import layers as lingvo_layers
+  p = lingvo_layers.ProjectionLayer.Params().Set(
+      name='p', input_dim=8, output_dim=8, has_bias=True,
+      random_seed=3,
+      qdomain_tpl=quant_utils.QDomain.Params().Set(bits=8, decay=0.0))
+  layer = p.Instantiate()
+  layer.train()
+  x = torch.randn(4, 8)
+  out_q = layer.FProp(layer.theta, x)
+  ref = lingvo_layers.ProjectionLayer.Params().Set(
+      name='p', input_dim=8, output_dim=8, has_bias=True,
+      random_seed=3).Instantiate()
+  out_f = ref.FProp(ref.theta, x)
+  # quantized output close but not identical to the fp path
+  assert (out_q - out_f).abs().max() < 0.2
+  assert (out_q - out_f).abs().max() > 0
+  # straight-through grads flow
+  (out_q.sum()).backward()
+  assert layer.w.grad is not None
