@@ -10,6 +10,8 @@
 
 #include <torch/extension.h>
 
+#include "wpm_encoder.h"
+
 #include <atomic>
 #include <condition_variable>
 #include <cstdint>
@@ -18,6 +20,7 @@
 #include <mutex>
 #include <random>
 #include <string>
+#include <deque>
 #include <thread>
 #include <vector>
 
@@ -178,6 +181,123 @@ class RecordYielder {
   std::vector<std::thread> threads_;
 };
 
+
+// Native text->LM-batch pipeline: reader threads feed worker threads
+// that tokenize (WPM), bucket by token length, and emit padded int64
+// batches — the C++ composition of RecordYielder + tokenizer +
+// RecordBatcher (reference record_batcher.h:89), fully GIL-free.
+class TextLmBatcher {
+ public:
+  TextLmBatcher(std::vector<std::string> files,
+                std::vector<std::string> pieces, int64_t unk_id,
+                int64_t sos_id, int64_t eos_id,
+                std::vector<int64_t> bucket_bounds,
+                std::vector<int64_t> bucket_limits, int64_t seed,
+                int num_threads, int64_t buffer_size, bool repeat)
+      : yielder_(std::move(files), "text", seed, buffer_size, 2, repeat),
+        encoder_(std::move(pieces), unk_id),
+        sos_id_(sos_id),
+        eos_id_(eos_id),
+        bounds_(std::move(bucket_bounds)),
+        limits_(std::move(bucket_limits)) {
+    TORCH_CHECK(bounds_.size() == limits_.size(),
+                "bucket bounds/limits mismatch");
+    buckets_.resize(bounds_.size());
+    for (int i = 0; i < std::max(1, num_threads); ++i) {
+      workers_.emplace_back([this] { WorkerLoop(); });
+    }
+  }
+
+  ~TextLmBatcher() { Stop(); }
+
+  void Stop() {
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      stop_ = true;
+    }
+    cv_batch_.notify_all();
+    yielder_.Stop();
+    for (auto& t : workers_) {
+      if (t.joinable()) t.join();
+    }
+    workers_.clear();
+  }
+
+  // Returns (ids [B, L+1] with SOS, labels [B, L+1] with EOS,
+  // paddings [B, L+1]) for one ready bucket.
+  std::vector<torch::Tensor> GetBatch() {
+    std::unique_lock<std::mutex> lk(mu_);
+    cv_batch_.wait(lk, [this] { return stop_ || !ready_.empty(); });
+    TORCH_CHECK(!stop_, "TextLmBatcher stopped");
+    auto batch = std::move(ready_.front());
+    ready_.pop_front();
+    lk.unlock();
+    const int64_t bound = bounds_[batch.bucket] + 1;  // +1 for SOS/EOS
+    const int64_t b = (int64_t)batch.examples.size();
+    auto ids = torch::full({b, bound}, eos_id_, torch::kInt64);
+    auto labels = torch::full({b, bound}, eos_id_, torch::kInt64);
+    auto pad = torch::ones({b, bound}, torch::kFloat32);
+    auto ids_a = ids.accessor<int64_t, 2>();
+    auto lab_a = labels.accessor<int64_t, 2>();
+    auto pad_a = pad.accessor<float, 2>();
+    for (int64_t i = 0; i < b; ++i) {
+      const auto& toks = batch.examples[i];
+      int64_t n = (int64_t)toks.size();
+      ids_a[i][0] = sos_id_;
+      for (int64_t j = 0; j < n; ++j) {
+        ids_a[i][j + 1] = toks[j];
+        lab_a[i][j] = toks[j];
+      }
+      lab_a[i][n] = eos_id_;
+      for (int64_t j = 0; j <= n; ++j) pad_a[i][j] = 0.0f;
+    }
+    return {ids, labels, pad};
+  }
+
+ private:
+  struct PendingBatch {
+    size_t bucket;
+    std::vector<std::vector<int64_t>> examples;
+  };
+
+  void WorkerLoop() {
+    while (!stop_) {
+      std::pair<std::string, int32_t> rec;
+      try {
+        rec = yielder_.Yield();
+      } catch (...) {
+        break;  // stopped or exhausted
+      }
+      auto toks = encoder_.Encode(rec.first);
+      int64_t len = (int64_t)toks.size();
+      std::lock_guard<std::mutex> lk(mu_);
+      for (size_t bi = 0; bi < bounds_.size(); ++bi) {
+        if (len <= bounds_[bi]) {
+          buckets_[bi].push_back(std::move(toks));
+          if ((int64_t)buckets_[bi].size() >= limits_[bi]) {
+            ready_.push_back({bi, std::move(buckets_[bi])});
+            buckets_[bi].clear();
+            cv_batch_.notify_one();
+          }
+          break;  // too-long records (> last bound) are dropped
+        }
+      }
+    }
+  }
+
+  RecordYielder yielder_;
+  lingvo_amd::WpmEncoder encoder_;
+  int64_t sos_id_, eos_id_;
+  std::vector<int64_t> bounds_, limits_;
+
+  std::mutex mu_;
+  std::condition_variable cv_batch_;
+  std::vector<std::vector<std::vector<int64_t>>> buckets_;
+  std::deque<PendingBatch> ready_;
+  std::atomic<bool> stop_{false};
+  std::vector<std::thread> workers_;
+};
+
 }  // namespace
 
 void RegisterInputPipeline(py::module_& m) {
@@ -198,4 +318,24 @@ void RegisterInputPipeline(py::module_& m) {
            })
       .def("current_epoch", &RecordYielder::current_epoch)
       .def("stop", &RecordYielder::Stop);
+  py::class_<TextLmBatcher>(m, "TextLmBatcher")
+      .def(py::init<std::vector<std::string>, std::vector<std::string>,
+                    int64_t, int64_t, int64_t, std::vector<int64_t>,
+                    std::vector<int64_t>, int64_t, int, int64_t, bool>(),
+           py::arg("files"), py::arg("pieces"), py::arg("unk_id") = 0,
+           py::arg("sos_id") = 1, py::arg("eos_id") = 2,
+           py::arg("bucket_bounds") = std::vector<int64_t>{64},
+           py::arg("bucket_limits") = std::vector<int64_t>{16},
+           py::arg("seed") = 0, py::arg("num_threads") = 2,
+           py::arg("buffer_size") = 10000, py::arg("repeat") = true)
+      .def("get_batch",
+           [](TextLmBatcher& self) {
+             std::vector<torch::Tensor> out;
+             {
+               py::gil_scoped_release release;
+               out = self.GetBatch();
+             }
+             return py::make_tuple(out[0], out[1], out[2]);
+           })
+      .def("stop", &TextLmBatcher::Stop);
 }

@@ -143,3 +143,35 @@ def test_wpm_native_matches_python():
   # native decode round-trip
   ids = tok._TokensToIds('the cats')
   assert tok._native.decode(ids) == 'the cats'
+
+
+def test_native_text_lm_batcher(tmp_path):
+  """C++ yield->tokenize->bucket->pad pipeline end to end."""
+  from lingvo_amd.ops import _loader
+  ext = _loader.get_ext()
+  if ext is None or not hasattr(ext, 'TextLmBatcher'):
+    pytest.skip('native extension not built')
+  vocab = ['<unk>', '<s>', '</s>', '▁a', '▁b', '▁c', '▁d', 'x']
+  f = tmp_path / 'corpus.txt'
+  with open(f, 'w') as fh:
+    for _ in range(50):
+      fh.write('a b\n')            # 2 tokens -> bucket 0 (bound 3)
+      fh.write('a b c d axx\n')    # 7 tokens -> bucket 1 (bound 8)
+  batcher = ext.TextLmBatcher(
+      [str(f)], vocab, unk_id=0, sos_id=1, eos_id=2,
+      bucket_bounds=[3, 8], bucket_limits=[4, 2], seed=7,
+      num_threads=2, repeat=True)
+  import torch
+  seen_shapes = set()
+  for _ in range(8):
+    ids, labels, pad = batcher.get_batch()
+    seen_shapes.add(tuple(ids.shape))
+    b, L = ids.shape
+    assert ids[:, 0].eq(1).all()               # SOS first
+    lens = (1 - pad).sum(1).long()
+    for i in range(b):
+      n = int(lens[i]) - 1                     # token count
+      assert labels[i, n] == 2                 # EOS terminates labels
+      assert (ids[i, 1:n + 1] == labels[i, :n]).all()  # shift property
+  assert (4, 4) in seen_shapes and (2, 9) in seen_shapes, seen_shapes
+  batcher.stop()
