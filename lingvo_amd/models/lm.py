@@ -45,6 +45,45 @@ class SyntheticLmInput(BaseSequenceInputGenerator):
                      weights=weights)
 
 
+class TextFileLmInput(BaseSequenceInputGenerator):
+  """Real-corpus LM input over the native C++ pipeline: TextLmBatcher
+  (yield -> WPM tokenize -> bucket -> pad) feeds token batches without
+  touching the GIL (reference BaseInputGeneratorFromFiles + the
+  record_batcher/tokenizer op chain)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.batch_size = 16  # used as the per-bucket limit default
+    p.Define('files', [], 'Text files, one example per line.')
+    p.Define('tokens', [], 'WPM vocab pieces (index = id).')
+    p.Define('num_threads', 2, 'Tokenizer worker threads.')
+    p.Define('input_seed', 301, 'Shuffle seed.')
+    p.bucket_upper_bound = [64]  # inherited bucketing params
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    from lingvo_amd.ops import _loader
+    ext = _loader.get_ext(required=True)
+    limits = list(p.bucket_batch_limit) or \
+        [p.batch_size] * len(p.bucket_upper_bound)
+    self._batcher = ext.TextLmBatcher(
+        list(p.files), list(p.tokens), 0, 1, 2,
+        list(p.bucket_upper_bound), limits, p.input_seed,
+        p.num_threads, 10000, True)
+
+  def _InputBatch(self) -> NestedMap:
+    ids, labels, paddings = self._batcher.get_batch()
+    return NestedMap(ids=ids, labels=labels, paddings=paddings,
+                     weights=1.0 - paddings)
+
+  def __del__(self):
+    if hasattr(self, '_batcher'):
+      self._batcher.stop()
+
+
 class TransformerLm(BaseLayer):
   """Causal transformer LM (reference tasks/lm/layers.py TransformerLm /
   GPipeTransformerLm at one_billion_wds.py:181)."""
