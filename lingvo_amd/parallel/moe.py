@@ -89,6 +89,19 @@ def Top2Gating(logits: torch.Tensor, capacity: int,
                    pos2=pos2, keep1=keep1, keep2=keep2, aux_loss=aux_loss)
 
 
+def ExpertChoiceGating(logits: torch.Tensor, capacity: int):
+  """Expert-choice routing (reference gshard_layers.py:2367-2987
+  expert-choice variants; Zhou et al. 2022): each expert picks its
+  top-`capacity` tokens by affinity, so load is balanced by
+  construction (no aux loss, no dropping asymmetry).
+
+  logits [N, E] -> NestedMap(idx [E, C] token ids, gates [E, C]).
+  """
+  probs = torch.softmax(logits, dim=-1)      # over experts per token
+  gates, idx = probs.t().topk(capacity, dim=-1)  # per-expert top-C
+  return NestedMap(idx=idx, gates=gates)
+
+
 class MoEFeedForwardLayer(BaseLayer):
   """Drop-in FFN replacement with E experts and top-2 routing
   (reference MoEBuilder gshard_builder.py:55; conformer MoE option
@@ -103,6 +116,7 @@ class MoEFeedForwardLayer(BaseLayer):
     p.Define('expert_capacity_factor', 2.0, 'Capacity factor c.')
     p.Define('activation', 'RELU', 'Expert activation.')
     p.Define('aux_loss_weight', 0.01, 'Load-balancing loss weight.')
+    p.Define('gating', 'top2', "'top2' | 'expert_choice'.")
     p.Define('moe_group', None,
              'torch.distributed group for EP (None = default group when '
              'initialized).')
@@ -140,6 +154,9 @@ class MoEFeedForwardLayer(BaseLayer):
       logits = logits.masked_fill(mask.unsqueeze(1), -1e30)
 
     capacity = max(4, int(p.expert_capacity_factor * n / e))
+    if p.gating == 'expert_choice':
+      return self._FPropExpertChoice(theta, inputs, x, logits, capacity,
+                                     paddings)
     gating = Top2Gating(logits, capacity)
     self._last_aux_loss = gating.aux_loss * p.aux_loss_weight
 
@@ -180,6 +197,29 @@ class MoEFeedForwardLayer(BaseLayer):
     for sel, top, pos, gate in combine_idx:
       out.index_add_(0, sel,
                      expert_out[top, pos] * gate.unsqueeze(1).to(x.dtype))
+    out = out.reshape(b, t, d)
+    if paddings is not None:
+      out = py_utils.ApplyPadding(paddings, out)
+    return out
+
+  def _FPropExpertChoice(self, theta, inputs, x, logits, capacity,
+                         paddings):
+    """Expert-choice path (EP world=1; the all-to-all form needs a
+    global token view and lands with the round-2 gating kernel)."""
+    p = self.p
+    b, t, d = inputs.shape
+    n = b * t
+    world, _, _ = self._EpWorld()
+    assert world == 1, 'expert_choice gating is single-rank for now'
+    gating = ExpertChoiceGating(logits, min(capacity, n))
+    self._last_aux_loss = logits.new_zeros(())  # balanced by design
+    xe = x[gating.idx]                          # [E, C, D]
+    act_fn = activations.GetFn(p.activation)
+    ye = torch.bmm(act_fn(torch.bmm(xe, theta.wi)), theta.wo)
+    out = x.new_zeros(n, d)
+    gates = gating.gates.to(x.dtype).unsqueeze(-1)
+    out.index_add_(0, gating.idx.reshape(-1),
+                   (ye * gates).reshape(-1, d))
     out = out.reshape(b, t, d)
     if paddings is not None:
       out = py_utils.ApplyPadding(paddings, out)

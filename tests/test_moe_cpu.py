@@ -109,3 +109,33 @@ def test_moe_lm_train_step():
           if isinstance(mm, MoEFeedForwardLayer)]
   assert len(moes) == 1
   assert moes[0].AuxLoss() is not None
+
+
+def test_expert_choice_gating_balanced():
+  import torch
+  from lingvo_amd.parallel import moe
+  g = torch.Generator().manual_seed(5)
+  logits = torch.randn(32, 4, generator=g)
+  out = moe.ExpertChoiceGating(logits, capacity=8)
+  assert out.idx.shape == (4, 8) and out.gates.shape == (4, 8)
+  # every expert processes exactly `capacity` tokens: balanced
+  assert out.idx.max() < 32
+  assert (out.gates >= 0).all() and (out.gates <= 1).all()
+
+
+def test_moe_layer_expert_choice_path():
+  import torch
+  from lingvo_amd.parallel import moe
+  p = moe.MoEFeedForwardLayer.Params().Set(
+      name='moe', input_dim=16, hidden_dim=32, num_experts=4,
+      gating='expert_choice', expert_capacity_factor=2.0, random_seed=3)
+  layer = p.Instantiate()
+  x = torch.randn(2, 12, 16, requires_grad=True)
+  pad = torch.zeros(2, 12)
+  pad[1, 10:] = 1.0
+  out = layer.FProp(layer.theta, x, pad)
+  assert out.shape == x.shape
+  assert out[1, 10:].abs().max() < 1e-6
+  out.sum().backward()
+  assert layer.wi.grad is not None and x.grad is not None
+  assert float(layer.AuxLoss()) == 0.0  # balanced by construction
