@@ -48,11 +48,13 @@ class TransformerAttentionLayer(BaseLayer):
   def FProp(self, theta: NestedMap, query_vec: torch.Tensor,
             paddings: Optional[torch.Tensor] = None,
             source_vecs: Optional[torch.Tensor] = None,
-            source_paddings: Optional[torch.Tensor] = None) -> torch.Tensor:
+            source_paddings: Optional[torch.Tensor] = None,
+            segment_ids: Optional[torch.Tensor] = None) -> torch.Tensor:
     p = self.p
     x = self.layer_norm.FProp(theta.layer_norm, query_vec)
     if source_vecs is None:
-      ctx = self.atten.FProp(theta.atten, x, paddings)
+      ctx = self.atten.FProp(theta.atten, x, paddings,
+                             segment_ids=segment_ids)
     else:
       ctx = self.atten.FPropCross(theta.atten, x, source_vecs, source_vecs,
                                   source_paddings)
@@ -206,8 +208,10 @@ class TransformerLayer(BaseLayer):
   def FProp(self, theta: NestedMap, inputs: torch.Tensor,
             paddings: Optional[torch.Tensor] = None,
             aux_vecs: Optional[torch.Tensor] = None,
-            aux_paddings: Optional[torch.Tensor] = None) -> torch.Tensor:
-    x = self.self_atten.FProp(theta.self_atten, inputs, paddings)
+            aux_paddings: Optional[torch.Tensor] = None,
+            segment_ids: Optional[torch.Tensor] = None) -> torch.Tensor:
+    x = self.self_atten.FProp(theta.self_atten, inputs, paddings,
+                              segment_ids=segment_ids)
     if self.p.has_aux_atten:
       x = self.cross_atten.FProp(theta.cross_atten, x, paddings,
                                  source_vecs=aux_vecs,
@@ -276,18 +280,23 @@ class StackedTransformerLayers(BaseLayer):
 
   def FProp(self, theta: NestedMap, inputs: torch.Tensor,
             paddings: Optional[torch.Tensor] = None,
-            aux_vecs=None, aux_paddings=None) -> torch.Tensor:
+            aux_vecs=None, aux_paddings=None,
+            segment_ids: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Optional segment_ids [B, T] run packed-input training (the
+    reference's packed segment masks, PackedBatchMajorLanguageModel
+    model.py:408): cross-segment attention is blocked in-kernel."""
     p = self.p
     x = inputs
     for i, layer in enumerate(self.x_layers):
       if p.remat and self.training:
         x = torch.utils.checkpoint.checkpoint(
             lambda x_, i_=i: self.x_layers[i_].FProp(
-                theta.x_layers[i_], x_, paddings, aux_vecs, aux_paddings),
+                theta.x_layers[i_], x_, paddings, aux_vecs, aux_paddings,
+                segment_ids),
             x, use_reentrant=False)
       else:
         x = layer.FProp(theta.x_layers[i], x, paddings, aux_vecs,
-                        aux_paddings)
+                        aux_paddings, segment_ids)
     if p.final_ln:
       x = self.final_layer_norm.FProp(theta.final_layer_norm, x)
     return x

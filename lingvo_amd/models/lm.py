@@ -149,13 +149,22 @@ class TransformerLm(BaseLayer):
     return self.emb.EmbLookup(theta.emb, ids)
 
   def FProp(self, theta: NestedMap, ids: torch.Tensor,
-            paddings: torch.Tensor) -> torch.Tensor:
+            paddings: torch.Tensor,
+            segment_ids: Optional[torch.Tensor] = None,
+            segment_pos: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Packed inputs: segment_ids [B,T] block cross-segment attention,
+    segment_pos [B,T] restarts positions per segment (reference
+    PackedBatchMajorLanguageModel, tasks/lm/model.py:408)."""
     x = self._Emb(theta, ids.long()).to(self.fprop_dtype)
     pos = self.pos_emb.FProp(theta.pos_emb, ids.shape[1], device=ids.device)
-    x = x + pos.unsqueeze(0).to(x.dtype)
+    if segment_pos is not None:
+      x = x + pos[segment_pos.long()].to(x.dtype)
+    else:
+      x = x + pos.unsqueeze(0).to(x.dtype)
     if self.p.dropout_prob and not self.do_eval:
       x = py_utils.DeterministicDropout(x, 1.0 - self.p.dropout_prob)
-    return self.stack.FProp(theta.stack, x, paddings)
+    return self.stack.FProp(theta.stack, x, paddings,
+                            segment_ids=segment_ids)
 
   def XentLoss(self, theta, act, labels, weights):
     return self.softmax.XentLoss(theta.softmax, act, class_weights=weights,
@@ -221,7 +230,12 @@ class LanguageModel(BaseTask):
     self.CreateChild('lm', self.p.lm)
 
   def ComputePredictions(self, theta, input_batch):
-    act = self.lm.FProp(theta.lm, input_batch.ids, input_batch.paddings)
+    kwargs = {}
+    if input_batch.Get('segment_ids') is not None:
+      kwargs = dict(segment_ids=input_batch.segment_ids,
+                    segment_pos=input_batch.Get('segment_pos'))
+    act = self.lm.FProp(theta.lm, input_batch.ids, input_batch.paddings,
+                        **kwargs)
     return NestedMap(activations=act)
 
   def ComputeLoss(self, theta, predictions, input_batch):

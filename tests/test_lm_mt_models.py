@@ -119,3 +119,34 @@ def test_text_file_lm_input_trains(tmp_path):
   assert batch.ids[:, 0].eq(1).all()
   m = task.TrainStep(batch)
   assert torch.isfinite(m['loss'][0])
+
+
+def test_packed_lm_matches_unpacked():
+  """Two sequences packed into one row == two separate rows (exact)."""
+  import torch
+  from lingvo_amd.models import lm as lm_model
+  p = lm_model.TransformerLm.Params().Set(
+      name='lm', vocab_size=32, model_dim=16, num_layers=2, num_heads=1,
+      hidden_dim=32, dropout_prob=0.0, random_seed=7)
+  lm = p.Instantiate()
+  lm.eval()
+  g = torch.Generator().manual_seed(2)
+  a = torch.randint(3, 32, (1, 5), generator=g)
+  b = torch.randint(3, 32, (1, 3), generator=g)
+
+  # unpacked: two rows, padded to 5
+  ids = torch.zeros(2, 5, dtype=torch.long)
+  ids[0] = a[0]
+  ids[1, :3] = b[0]
+  pad = torch.zeros(2, 5)
+  pad[1, 3:] = 1.0
+  ref = lm.FProp(lm.theta, ids, pad)
+
+  # packed: one row [a; b] with segment ids/positions
+  packed = torch.cat([a, b], dim=1)          # [1, 8]
+  seg = torch.tensor([[1, 1, 1, 1, 1, 2, 2, 2]])
+  pos = torch.tensor([[0, 1, 2, 3, 4, 0, 1, 2]])
+  out = lm.FProp(lm.theta, packed, torch.zeros(1, 8),
+                 segment_ids=seg, segment_pos=pos)
+  assert (out[0, :5] - ref[0]).abs().max() < 1e-4
+  assert (out[0, 5:] - ref[1, :3]).abs().max() < 1e-4
