@@ -205,3 +205,110 @@ def ShardSequence(x: torch.Tensor, rank: int, world: int,
   s = x.shape[dim]
   assert s % world == 0, f'seq len {s} not divisible by cp={world}'
   return x.narrow(dim, rank * (s // world), s // world)
+
+
+class _SeqHeadAllToAll(torch.autograd.Function):
+  """All-to-all along dim 0; backward is the inverse all-to-all."""
+
+  @staticmethod
+  def forward(ctx, x, group):
+    ctx.group = group
+    x = x.contiguous()
+    out = torch.empty_like(x)
+    dist.all_to_all_single(out, x, group=group)
+    return out
+
+  @staticmethod
+  def backward(ctx, g):
+    g = g.contiguous()
+    out = torch.empty_like(g)
+    dist.all_to_all_single(out, g, group=ctx.group)
+    return out, None
+
+
+def _ScatterHeadsGatherSeq(x: torch.Tensor, world: int, group):
+  """[B, s, N, H] seq-sharded -> [B, s*world, N/world, H]: each rank
+  gets the FULL sequence for its N/world heads (Ulysses layout)."""
+  b, s, n, h = x.shape
+  nl = n // world
+  xs = x.reshape(b, s, world, nl, h).permute(2, 0, 1, 3, 4).contiguous()
+  recv = _SeqHeadAllToAll.apply(xs, group)          # [P, B, s, nl, H]
+  return recv.permute(1, 0, 2, 3, 4).reshape(b, world * s, nl, h)
+
+
+def _GatherHeadsScatterSeq(x: torch.Tensor, world: int, group):
+  """Inverse of _ScatterHeadsGatherSeq: [B, S, nl, H] -> [B, S/world,
+  nl*world, H]."""
+  b, stot, nl, h = x.shape
+  s = stot // world
+  xs = x.reshape(b, world, s, nl, h).permute(1, 0, 2, 3, 4).contiguous()
+  recv = _SeqHeadAllToAll.apply(xs, group)          # [P, B, s, nl, H]
+  return recv.permute(1, 2, 0, 3, 4).reshape(b, s, world * nl, h)
+
+
+class UlyssesMultiHeadedAttention(attention_lib.MultiHeadedAttention):
+  """DeepSpeed-Ulysses-style sequence parallelism: instead of rotating
+  K/V around a ring, ONE all-to-all re-shards activations from
+  sequence-split to head-split, the full-sequence attention runs
+  locally on N/P heads (the dense flash kernel, unchanged), and a
+  second all-to-all restores the sequence split. Two a2a pairs per
+  layer vs cp ring hops — on 8-way xGMI the a2a moves each byte once
+  over one link, so Ulysses wins when N >= P and the ring wins for
+  GQA-heavy configs (N/P < 1). Both live behind the same atten_tpl
+  surface."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('cp_group', None, 'Sequence-parallel group; None = WORLD.')
+    p.cls = cls
+    return p
+
+  def FProp(self, theta: NestedMap, query_vec: torch.Tensor,
+            paddings: Optional[torch.Tensor] = None,
+            segment_ids: Optional[torch.Tensor] = None) -> torch.Tensor:
+    p = self.p
+    if segment_ids is not None:
+      raise NotImplementedError('packed inputs under Ulysses SP')
+    world = dist.get_world_size(p.cp_group) if dist.is_initialized() \
+        else 1
+    if world == 1:
+      return super().FProp(theta, query_vec, paddings)
+    assert self._n % world == 0, (self._n, world)
+    assert self._nkv % world == 0, 'GQA heads must divide the SP degree'
+    q, k, v = self._Project(theta, query_vec)
+    if p.use_rope:
+      # positions are GLOBAL: offset by this rank's sequence start
+      rank = dist.get_rank(p.cp_group)
+      s = q.shape[1]
+      pos = (rank * s + torch.arange(
+          s, device=q.device, dtype=torch.float32)).expand(q.shape[0], s)
+      q = self.rope.FProp(theta.rope, q, position=pos)
+      k = self.rope.FProp(theta.rope, k, position=pos)
+    qg = _ScatterHeadsGatherSeq(q, world, p.cp_group)
+    kg = _ScatterHeadsGatherSeq(k, world, p.cp_group)
+    vg = _ScatterHeadsGatherSeq(v, world, p.cp_group)
+    klen = None
+    if paddings is not None:
+      klen = py_utils.LengthsFromPaddings(paddings).clone()
+      dist.all_reduce(klen, group=p.cp_group)
+      klen = klen.to(torch.int32)
+    bias = None
+    if p.rel_pos_bias:
+      # local head slice of the bias table (chunk r lands on rank r)
+      rank = dist.get_rank(p.cp_group)
+      nl = self._n // world
+      bias = theta.rel_bias[rank * nl:(rank + 1) * nl]
+    from lingvo_amd.ops import flash_attn
+    out = flash_attn.flash_attention(
+        qg, kg, vg, klen, bias,
+        p.left_context, 0 if p.causal else p.right_context,
+        p.rel_pos_clip)
+    out = _GatherHeadsScatterSeq(out, world, p.cp_group)
+    b, t = out.shape[0], out.shape[1]
+    ctx = out.reshape(b, t, self._n * self._h)
+    post = py_utils.MatmulBias(ctx, theta.post_w,
+                               theta.post_b if p.use_bias else None)
+    if paddings is not None:
+      post = py_utils.ApplyPadding(paddings, post)
+    return post

@@ -128,3 +128,63 @@ def test_cp_layer_matches_full():
   # the DP all-reduce in training).
   assert torch.allclose(results['dw0'] + results['dw1'],
                         layer.qkv_w.grad, atol=1e-4)
+
+
+def _run_ulysses(rank, world, port, results):
+  os.environ['MASTER_ADDR'] = '127.0.0.1'
+  os.environ['MASTER_PORT'] = str(port)
+  dist.init_process_group('gloo', rank=rank, world_size=world)
+  layer = cp.UlyssesMultiHeadedAttention.Params().Set(
+      name='u', input_dim=32, hidden_dim=32, num_heads=4, causal=True,
+      rel_pos_bias=True, random_seed=19).Instantiate()
+  gb = torch.Generator().manual_seed(77)
+  with torch.no_grad():
+    layer.rel_bias.copy_(
+        torch.randn(layer.rel_bias.shape, generator=gb) * 0.1)
+  g = torch.Generator().manual_seed(44)
+  x = torch.randn(2, 12, 32, generator=g)
+  pad = torch.zeros(2, 12)
+  pad[1, 9:] = 1.0
+  xl = cp.ShardSequence(x, rank, world).detach().requires_grad_(True)
+  out = layer.FProp(layer.theta, xl, cp.ShardSequence(pad, rank, world))
+  out.square().sum().backward()
+  results[f'out{rank}'] = out.detach()
+  results[f'dx{rank}'] = xl.grad.clone()
+  results[f'dw{rank}'] = layer.qkv_w.grad.clone()
+  results[f'bias{rank}'] = layer.rel_bias.detach().clone()
+  dist.destroy_process_group()
+
+
+def test_ulysses_attention_matches_full():
+  ctx = mp.get_context('spawn')
+  with ctx.Manager() as mgr:
+    results = mgr.dict()
+    procs = [ctx.Process(target=_run_ulysses, args=(r, 2, 29575, results))
+             for r in range(2)]
+    for p in procs:
+      p.start()
+    for p in procs:
+      p.join(120)
+      assert p.exitcode == 0
+    results = dict(results)
+
+  from lingvo_amd.layers import attention as attention_lib
+  ref = attention_lib.MultiHeadedAttention.Params().Set(
+      name='u', input_dim=32, hidden_dim=32, num_heads=4, causal=True,
+      rel_pos_bias=True, random_seed=19).Instantiate()
+  with torch.no_grad():
+    ref.rel_bias.copy_(results['bias0'])
+  g = torch.Generator().manual_seed(44)
+  x = torch.randn(2, 12, 32, generator=g, requires_grad=True)
+  pad = torch.zeros(2, 12)
+  pad[1, 9:] = 1.0
+  full = ref.FProp(ref.theta, x, pad)
+  full.square().sum().backward()
+  for r in range(2):
+    sl = slice(r * 6, (r + 1) * 6)
+    assert torch.allclose(results[f'out{r}'], full[:, sl].detach(),
+                          atol=1e-4), r
+    assert torch.allclose(results[f'dx{r}'], x.grad[:, sl], atol=1e-4)
+  # replicated qkv_w: per-rank grads SUM to the full grad
+  assert torch.allclose(results['dw0'] + results['dw1'],
+                        ref.qkv_w.grad, atol=1e-4)
