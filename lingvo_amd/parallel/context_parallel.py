@@ -312,3 +312,93 @@ class UlyssesMultiHeadedAttention(attention_lib.MultiHeadedAttention):
     if paddings is not None:
       post = py_utils.ApplyPadding(paddings, post)
     return post
+
+
+def ZigzagShard(x: torch.Tensor, rank: int, world: int,
+                dim: int = 1) -> torch.Tensor:
+  """Zigzag CP sharding for causal load balance: the sequence splits
+  into 2P chunks and rank r holds chunks (r, 2P-1-r), so every rank
+  sees the same mix of early (cheap) and late (expensive) causal
+  positions. Returns concat of the two chunks along `dim`."""
+  s = x.shape[dim]
+  assert s % (2 * world) == 0, (s, world)
+  L = s // (2 * world)
+  lo = x.narrow(dim, rank * L, L)
+  hi = x.narrow(dim, (2 * world - 1 - rank) * L, L)
+  return torch.cat([lo, hi], dim=dim)
+
+
+def ZigzagPositions(rank: int, world: int, total_len: int,
+                    device=None) -> torch.Tensor:
+  """Global positions of rank r's zigzag shard, [total_len/P]."""
+  L = total_len // (2 * world)
+  lo = torch.arange(rank * L, (rank + 1) * L, device=device)
+  hi = torch.arange((2 * world - 1 - rank) * L,
+                    (2 * world - rank) * L, device=device)
+  return torch.cat([lo, hi])
+
+
+def _BlockAttentionPos(q, k, v, q_pos, k_pos, causal, scale, klen):
+  """_BlockAttention with explicit (possibly non-contiguous) global
+  positions per query/key — the zigzag form."""
+  b, tq, n, h = q.shape
+  sk, nkv = k.shape[1], k.shape[2]
+  qf = q.permute(0, 2, 1, 3)
+  kf = k.permute(0, 2, 1, 3)
+  vf = v.permute(0, 2, 1, 3)
+  if n != nkv:
+    kf = kf.repeat_interleave(n // nkv, dim=1)
+    vf = vf.repeat_interleave(n // nkv, dim=1)
+  logits = torch.einsum('bnth,bnsh->bnts', qf, kf) * scale
+  mask = torch.ones(tq, sk, dtype=torch.bool, device=q.device)
+  if causal:
+    mask &= k_pos[None, :] <= q_pos[:, None]
+  mask = mask[None, None]
+  if klen is not None:
+    mask = mask & (k_pos[None, None, None, :] <
+                   klen[:, None, None, None])
+  logits = logits.masked_fill(~mask, -1e30)
+  lse = torch.logsumexp(logits, dim=-1)
+  probs = torch.exp(logits - lse.unsqueeze(-1))
+  out = torch.einsum('bnts,bnsh->bnth', probs, vf)
+  return out.permute(0, 2, 1, 3), lse
+
+
+def RingAttentionZigzag(q: torch.Tensor, k: torch.Tensor,
+                        v: torch.Tensor,
+                        klen: Optional[torch.Tensor] = None,
+                        causal: bool = True,
+                        scale: Optional[float] = None,
+                        group=None) -> torch.Tensor:
+  """Ring attention over ZigzagShard-ed q/k/v: same exact LSE merge as
+  RingAttention, but every rank does ~the same causal work per block
+  (the round-2 balance fix, landed early since it is exactness-testable
+  on CPU). Inputs/outputs are in zigzag-local layout."""
+  if scale is None:
+    scale = 1.0 / math.sqrt(q.shape[-1])
+  world = dist.get_world_size(group) if dist.is_initialized() else 1
+  rank = dist.get_rank(group) if world > 1 else 0
+  t_local = q.shape[1]
+  total = t_local * world
+  q_pos = ZigzagPositions(rank, world, total, device=q.device)
+  qf = q.float()
+  # kv travels with its source rank id so positions can be derived
+  kv = torch.stack([k.float(), v.float()])
+  outs, lses = [], []
+  for i in range(world):
+    src = (rank - i) % world
+    k_pos = ZigzagPositions(src, world, total, device=q.device)
+    out_b, lse_b = _BlockAttentionPos(qf, kv[0], kv[1], q_pos, k_pos,
+                                      causal, scale, klen)
+    outs.append(out_b)
+    lses.append(lse_b)
+    if i < world - 1:
+      kv = _RingExchange.apply(kv, group)
+  lse_all = torch.stack(lses, dim=-1)
+  lse_tot = torch.logsumexp(lse_all, dim=-1)
+  w = torch.exp(lse_all - lse_tot.unsqueeze(-1))
+  out = sum(w[..., i, None] * o.permute(0, 2, 1, 3)
+            for i, o in enumerate(outs))
+  out = torch.where(lse_tot.unsqueeze(-1) > -1e29, out,
+                    torch.zeros_like(out))
+  return out.permute(0, 2, 1, 3).to(q.dtype)

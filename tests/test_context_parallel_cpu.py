@@ -188,3 +188,45 @@ def test_ulysses_attention_matches_full():
   # replicated qkv_w: per-rank grads SUM to the full grad
   assert torch.allclose(results['dw0'] + results['dw1'],
                         ref.qkv_w.grad, atol=1e-4)
+
+
+def _run_zigzag(rank, world, port, results):
+  os.environ['MASTER_ADDR'] = '127.0.0.1'
+  os.environ['MASTER_PORT'] = str(port)
+  dist.init_process_group('gloo', rank=rank, world_size=world)
+  q, k, v, klen = _make_inputs(causal=True)
+  ql = cp.ZigzagShard(q, rank, world).detach().requires_grad_(True)
+  kl = cp.ZigzagShard(k, rank, world).detach().requires_grad_(True)
+  vl = cp.ZigzagShard(v, rank, world).detach().requires_grad_(True)
+  out = cp.RingAttentionZigzag(ql, kl, vl, klen=klen, causal=True)
+  out.square().sum().backward()
+  results[f'out{rank}'] = out.detach()
+  results[f'dq{rank}'] = ql.grad.clone()
+  results[f'dk{rank}'] = kl.grad.clone()
+  dist.destroy_process_group()
+
+
+def test_zigzag_ring_attention_exact():
+  ctx = mp.get_context('spawn')
+  with ctx.Manager() as mgr:
+    results = mgr.dict()
+    procs = [ctx.Process(target=_run_zigzag, args=(r, 2, 29581, results))
+             for r in range(2)]
+    for p in procs:
+      p.start()
+    for p in procs:
+      p.join(120)
+      assert p.exitcode == 0
+    results = dict(results)
+
+  q, k, v, klen = _make_inputs(causal=True)
+  q, k, v = (t.requires_grad_(True) for t in (q, k, v))
+  ref = flash_attn.flash_attention(q, k, v, klen.to(torch.int32), None,
+                                   -1, 0)
+  ref.square().sum().backward()
+  for r in range(2):
+    pos = cp.ZigzagPositions(r, 2, q.shape[1])
+    assert torch.allclose(results[f'out{r}'],
+                          ref[:, pos].detach(), atol=1e-4), r
+    assert torch.allclose(results[f'dq{r}'], q.grad[:, pos], atol=1e-4)
+    assert torch.allclose(results[f'dk{r}'], k.grad[:, pos], atol=1e-4)
