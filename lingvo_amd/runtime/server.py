@@ -14,14 +14,17 @@ Endpoints:
 
 Requests are served under a lock per predictor (one model instance per
 GPU; scale-out is one server process per GPU behind a round-robin
-proxy, matching the one-process-per-GPU training topology). Batching
-across requests is the round-2 item (micro-batching queue).
+proxy, matching the one-process-per-GPU training topology). With
+micro_batch=True concurrent single-example requests coalesce into one
+stacked model call (MicroBatcher).
 """
 
 from __future__ import annotations
 
 import argparse
+import queue as queue_lib
 import threading
+import time
 from typing import Optional
 
 import torch
@@ -40,12 +43,86 @@ def _ToJsonable(value):
   return value
 
 
-def MakeApp(predictor: Predictor):
+class MicroBatcher:
+  """Dynamic request batching: concurrent single-example requests to
+  the same subgraph coalesce (within max_wait_ms, up to max_batch) into
+  one stacked model call — the serving-throughput pattern the reference
+  relies on TF-Serving for. Requests are grouped by (subgraph, feed
+  keys, per-example shapes); results split back per caller."""
+
+  def __init__(self, run_fn, max_batch: int = 8, max_wait_ms: float = 3.0):
+    self._run = run_fn
+    self.max_batch = max_batch
+    self.max_wait = max_wait_ms / 1000.0
+    self._q: queue_lib.Queue = queue_lib.Queue()
+    self.batches_run = 0
+    self.examples_run = 0
+    self._thread = threading.Thread(target=self._Loop, daemon=True)
+    self._thread.start()
+
+  @staticmethod
+  def _Sig(subgraph, feeds):
+    return (subgraph,
+            tuple(sorted((k, tuple(v.shape[1:])) for k, v in
+                         feeds.items())))
+
+  def Submit(self, subgraph: str, feeds: dict):
+    """feeds: {name: tensor [1, ...]} single-example request; blocks
+    until the coalesced batch runs. Returns {name: tensor [1, ...]}."""
+    ev = threading.Event()
+    slot = {}
+    self._q.put((self._Sig(subgraph, feeds), subgraph, feeds, ev, slot))
+    ev.wait()
+    if 'error' in slot:
+      raise slot['error']
+    return slot['out']
+
+  def _Loop(self):
+    while True:
+      sig, subgraph, feeds, ev, slot = self._q.get()
+      group = [(feeds, ev, slot)]
+      deadline = time.monotonic() + self.max_wait
+      pending = []
+      while len(group) < self.max_batch:
+        timeout = deadline - time.monotonic()
+        if timeout <= 0:
+          break
+        try:
+          item = self._q.get(timeout=timeout)
+        except queue_lib.Empty:
+          break
+        if item[0] == sig:
+          group.append(item[2:])
+        else:
+          pending.append(item)  # different shape/subgraph: next round
+      for item in pending:
+        self._q.put(item)
+      try:
+        stacked = {k: torch.cat([g[0][k] for g in group], dim=0)
+                   for k in group[0][0]}
+        out = self._run(subgraph, **stacked)
+        self.batches_run += 1
+        self.examples_run += len(group)
+        for i, (_, ev_i, slot_i) in enumerate(group):
+          slot_i['out'] = {k: (v[i:i + 1] if isinstance(v, torch.Tensor)
+                               else v) for k, v in out.items()}
+          ev_i.set()
+      except Exception as e:  # surface to every caller in the batch
+        for _, ev_i, slot_i in group:
+          slot_i['error'] = e
+          ev_i.set()
+
+
+def MakeApp(predictor: Predictor, micro_batch: bool = False,
+            max_batch: int = 8, max_wait_ms: float = 3.0):
   """Builds the FastAPI app around an already-loaded Predictor."""
   from fastapi import FastAPI, HTTPException
 
   app = FastAPI(title='lingvo_amd inference')
   lock = threading.Lock()
+  batcher = MicroBatcher(predictor.Run, max_batch, max_wait_ms) \
+      if micro_batch else None
+  app.state.batcher = batcher
 
   @app.get('/health')
   def health():
@@ -62,11 +139,15 @@ def MakeApp(predictor: Predictor):
         tensors[k] = torch.as_tensor(v)
       except Exception as e:
         raise HTTPException(400, f'feed {k!r} not tensor-like: {e}')
-    with lock:
-      try:
-        out = predictor.Run(subgraph, **tensors)
-      except TypeError as e:
-        raise HTTPException(400, str(e))
+    try:
+      if batcher is not None and all(
+          v.dim() > 0 and v.shape[0] == 1 for v in tensors.values()):
+        out = batcher.Submit(subgraph, tensors)
+      else:
+        with lock:
+          out = predictor.Run(subgraph, **tensors)
+    except TypeError as e:
+      raise HTTPException(400, str(e))
     return _ToJsonable(out)
 
   return app

@@ -181,3 +181,37 @@ def test_trial_early_stop_in_executor(tmp_path):
   # stopped by the trial after one loop, far before max_steps
   assert ex.task.global_step == 2
   assert trial.reports and 'loss' in trial.reports[0][1]
+
+
+def test_server_micro_batching(tmp_path):
+  """Concurrent single-example requests coalesce into stacked calls."""
+  import threading
+  from lingvo_amd.runtime.inference import InferenceGraphExporter, Predictor
+  from lingvo_amd.runtime.server import MakeApp
+  from fastapi.testclient import TestClient
+
+  model_p = registry.GetParams('image.mnist.LeNet5', 'Train')
+  model_p.task.random_seed = 9
+  path = str(tmp_path / 'inference.pt')
+  InferenceGraphExporter.Export(model_p, path)
+  app = MakeApp(Predictor(path, device='cpu'), micro_batch=True,
+                max_wait_ms=30.0)
+  client = TestClient(app)
+
+  img = torch.zeros(1, 28, 28, 1).tolist()
+  ref = client.post('/predict/default', json={'images': img}).json()
+
+  results = [None] * 6
+  def call(i):
+    results[i] = client.post('/predict/default',
+                             json={'images': img}).json()
+  threads = [threading.Thread(target=call, args=(i,)) for i in range(6)]
+  for t in threads:
+    t.start()
+  for t in threads:
+    t.join(30)
+  for r in results:
+    assert r is not None and r['label'] == ref['label']
+  b = app.state.batcher
+  assert b.examples_run >= 7
+  assert b.batches_run < b.examples_run  # at least one coalesced batch
