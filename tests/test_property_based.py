@@ -1,0 +1,109 @@
+"""Property-based tests (hypothesis) for foundation invariants:
+NestedMap structure ops, Params text round-trip, packing, entmax."""
+
+import string
+
+import torch
+from hypothesis import given, settings, strategies as st
+
+from lingvo_amd.core.hyperparams import Params
+from lingvo_amd.core.nested_map import NestedMap
+
+_IDENT = st.text(alphabet=string.ascii_lowercase, min_size=1,
+                 max_size=6)
+_SCALARS = st.one_of(
+    st.integers(-10**6, 10**6),
+    st.floats(-1e6, 1e6, allow_nan=False),
+    st.booleans(),
+    st.text(alphabet=string.printable, max_size=12),
+    st.none())
+
+
+@settings(max_examples=50, deadline=None)
+@given(st.dictionaries(_IDENT, _SCALARS, min_size=1, max_size=6),
+       st.dictionaries(_IDENT, _SCALARS, max_size=4))
+def test_params_text_roundtrip(top, nested):
+  p = Params()
+  for k, v in top.items():
+    p.Define(k, v, 'x')
+  sub = Params()
+  for k, v in nested.items():
+    sub.Define(k, v, 'x')
+  p.Define('zsub', sub, 'sub')
+  text = p.ToText()
+  q = p.Copy()
+  # perturb every scalar then restore from text
+  for k, v in top.items():
+    setattr(q, k, None if v is not None else 0)
+  q.FromText(text)
+  for k, v in top.items():
+    got = q.Get(k)
+    if isinstance(v, float):
+      assert got == v or abs(got - v) < 1e-9, (k, v, got)
+    else:
+      assert got == v, (k, v, got)
+
+
+@settings(max_examples=50, deadline=None)
+@given(st.lists(st.integers(0, 100), min_size=1, max_size=8),
+       st.lists(st.integers(0, 100), min_size=1, max_size=8))
+def test_nested_map_flatten_pack_inverse(a, b):
+  nmap = NestedMap(x=torch.tensor(a), sub=NestedMap(y=torch.tensor(b)),
+                   seq=[torch.tensor(a), torch.tensor(b)])
+  flat = nmap.Flatten()
+  packed = nmap.Pack(flat)
+  assert packed.IsCompatible(nmap)
+  for t1, t2 in zip(packed.Flatten(), flat):
+    assert torch.equal(t1, t2)
+  # Transform preserves structure
+  doubled = nmap.Transform(lambda t: t * 2)
+  for t1, t2 in zip(doubled.Flatten(), flat):
+    assert torch.equal(t1, t2 * 2)
+
+
+@settings(max_examples=30, deadline=None)
+@given(st.integers(2, 6), st.integers(2, 16))
+def test_entmax_simplex_properties(rows, cols):
+  from lingvo_amd.layers import activations
+  torch.manual_seed(rows * 100 + cols)
+  x = torch.randn(rows, cols) * 3
+  p = activations.Entmax15(x)
+  assert torch.allclose(p.sum(-1), torch.ones(rows), atol=1e-4)
+  assert (p >= -1e-7).all()
+  # permutation equivariance
+  perm = torch.randperm(cols)
+  p2 = activations.Entmax15(x[:, perm])
+  assert torch.allclose(p2, p[:, perm], atol=1e-5)
+  # monotone: raising one logit never lowers its probability
+  x3 = x.clone()
+  x3[:, 0] += 1.0
+  p3 = activations.Entmax15(x3)
+  assert (p3[:, 0] >= p[:, 0] - 1e-6).all()
+
+
+@settings(max_examples=30, deadline=None)
+@given(st.lists(st.integers(1, 6), min_size=2, max_size=5))
+def test_pack_unpack_roundtrip(lens):
+  """PackSequences + ApplyPacking reconstruct the original tokens."""
+  from lingvo_amd.core import pack_ops
+  b = len(lens)
+  g = torch.Generator().manual_seed(sum(lens) + b)
+  max_len = max(lens)
+  ids = torch.zeros(b, max_len, dtype=torch.long)
+  for i, L in enumerate(lens):
+    ids[i, :L] = torch.randint(3, 50, (L,), generator=g)
+  src_lens = torch.tensor(lens)
+  # packed budget large enough that nothing is dropped
+  out = pack_ops.PackSequences(src_lens, src_lens, packed_batch=b,
+                               src_time=sum(lens), tgt_time=sum(lens))
+  packed = pack_ops.ApplyPacking(ids, 0, out.src_segment_ids,
+                                 out.src_indices_in_input)
+  orig = sorted(t for i, L in enumerate(lens)
+                for t in ids[i, :L].tolist())
+  got = sorted(packed[out.src_segment_ids > 0].tolist())
+  assert got == orig
+  # positions restart at each segment
+  pos = out.src_segment_pos
+  seg = out.src_segment_ids
+  starts = (seg != torch.roll(seg, 1, dims=1)) & (seg > 0)
+  assert (pos[starts] == 0).all()
