@@ -16,7 +16,6 @@ Same callback contract as BeamSearchHelper:
 
 from __future__ import annotations
 
-import math
 from typing import Callable
 
 import torch

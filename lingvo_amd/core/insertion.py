@@ -9,8 +9,6 @@ in parallel until every slot emits end-of-slot.
 
 from __future__ import annotations
 
-from typing import Tuple
-
 import torch
 
 from lingvo_amd.core import py_utils

@@ -9,9 +9,6 @@ blocks), so Shampoo is just another torch.optim.Optimizer here.
 
 from __future__ import annotations
 
-import math
-from typing import List, Optional
-
 import torch
 
 from lingvo_amd.core import optimizer as optimizer_lib

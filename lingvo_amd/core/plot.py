@@ -9,7 +9,7 @@ the Agg backend — no display needed on training nodes.
 from __future__ import annotations
 
 import io
-from typing import Callable, List, Optional, Sequence, Tuple
+from typing import Callable, List, Sequence, Tuple
 
 import numpy as np
 import torch

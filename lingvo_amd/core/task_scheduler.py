@@ -3,8 +3,6 @@
 from __future__ import annotations
 
 import random
-from typing import List, Tuple
-
 from lingvo_amd.core.base_layer import BaseLayer
 
 

@@ -18,7 +18,7 @@ causal) — all GEMM-shaped work lands on MFMA via hipBLASLt.
 from __future__ import annotations
 
 import math
-from typing import Optional, Tuple
+from typing import Optional
 
 import torch
 import torch.nn.functional as F

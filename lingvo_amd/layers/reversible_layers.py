@@ -12,11 +12,10 @@ goes straight into bigger per-GPU batches.
 
 from __future__ import annotations
 
-from typing import List, Tuple
+from typing import Tuple
 
 import torch
 
-from lingvo_amd.core import py_utils
 from lingvo_amd.core.base_layer import BaseLayer
 from lingvo_amd.core.nested_map import NestedMap
 

@@ -6,13 +6,7 @@ network access)."""
 
 from __future__ import annotations
 
-import torch
-
-from lingvo_amd.core import learner as learner_lib
-from lingvo_amd.core import optimizer as optimizer_lib
 from lingvo_amd.core import registry
-from lingvo_amd.core.base_model_params import SingleTaskModelParams
-from lingvo_amd.models import car as car_model
 from lingvo_amd.models.params.car.kitti import StarNetPillars
 
 

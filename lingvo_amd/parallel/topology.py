@@ -12,7 +12,7 @@ Layout (num_stages contiguous so pipeline neighbors share an xGMI hop):
 
 from __future__ import annotations
 
-from typing import List, Optional
+from typing import Optional
 
 import torch.distributed as dist
 
