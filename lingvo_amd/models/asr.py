@@ -311,7 +311,11 @@ class AsrDecoder(BaseLayer):
 
   def GreedyDecode(self, theta: NestedMap, enc: torch.Tensor,
                    enc_paddings: torch.Tensor, max_len: int = 100,
-                   sos_id: int = 1, eos_id: int = 2) -> torch.Tensor:
+                   sos_id: int = 1, eos_id: int = 2,
+                   fusion=None) -> torch.Tensor:
+    """Greedy decode; optional `fusion` (ShallowFusion) mixes an
+    external LM's log-probs into each step (reference
+    tasks/asr/fusion.py shallow fusion)."""
     p = self.p
     b = enc.shape[0]
     tok = torch.full((b,), sos_id, dtype=torch.long, device=enc.device)
@@ -320,6 +324,7 @@ class AsrDecoder(BaseLayer):
     ctx = torch.zeros(b, p.source_dim, device=enc.device,
                       dtype=self.fprop_dtype)
     done = torch.zeros(b, dtype=torch.bool, device=enc.device)
+    fusion_state = fusion.InitState(b, enc.device) if fusion else None
     out = []
     for _ in range(max_len):
       e = self.emb.EmbLookup(theta.emb, tok).to(self.fprop_dtype)
@@ -330,13 +335,46 @@ class AsrDecoder(BaseLayer):
       ctx = self._Attend(theta, x, enc, enc_paddings)
       logits = self.softmax.Logits(theta.softmax,
                                    torch.cat([x, ctx], dim=-1))
-      tok = logits.argmax(-1)
+      scores = torch.log_softmax(logits.float(), dim=-1)
+      if fusion is not None:
+        lm_scores, fusion_state = fusion.Score(tok, fusion_state)
+        scores = scores + fusion.weight * lm_scores
+      tok = scores.argmax(-1)
       tok = torch.where(done, torch.full_like(tok, eos_id), tok)
       done = done | (tok == eos_id)
       out.append(tok)
       if bool(done.all()):
         break
     return torch.stack(out, dim=1)
+
+
+class ShallowFusion:
+  """LM shallow fusion for decoding (reference tasks/asr/fusion.py):
+  combined score = log p_am + weight * log p_lm, with the LM advanced
+  token by token via its ExtendStep-style callback."""
+
+  def __init__(self, lm_layer, lm_theta, weight: float = 0.3,
+               max_len: int = 512):
+    self.lm = lm_layer
+    self.theta = lm_theta
+    self.weight = weight
+    self.max_len = max_len
+
+  def InitState(self, batch: int, device) -> NestedMap:
+    return NestedMap(ids=torch.zeros(batch, 0, dtype=torch.long,
+                                     device=device))
+
+  def Score(self, prev_tok: torch.Tensor, state: NestedMap):
+    """Returns (log-probs [B, V], new state). Re-scores the growing
+    prefix each step (CPU-oracle form; incremental KV-cache scoring is
+    the GPU path once the LM exposes ExtendStep here)."""
+    ids = torch.cat([state.ids, prev_tok.unsqueeze(1)], dim=1)
+    act = self.lm.FProp(self.theta, ids,
+                        torch.zeros(ids.shape, device=ids.device,
+                                    dtype=torch.float32))
+    logits = self.lm.softmax.Logits(
+        getattr(self.theta, 'softmax'), act[:, -1])
+    return torch.log_softmax(logits.float(), dim=-1), NestedMap(ids=ids)
 
 
 class AsrModel(BaseTask):

@@ -186,3 +186,34 @@ def test_streaming_recognizer_matches_offline():
   assert out.encoded.shape == enc.shape, (out.encoded.shape, enc.shape)
   assert (out.encoded - enc).abs().max() < 1e-3
   assert torch.equal(out.hyps, hyps_off)
+
+
+def test_shallow_fusion_biases_decode():
+  """An LM that strongly prefers one token steers greedy decode."""
+  import torch
+  from lingvo_amd.models import asr as asr_lib
+  from lingvo_amd.models import lm as lm_lib
+  V = 16
+  dec = asr_lib.AsrDecoder.Params().Set(
+      name='d', vocab_size=V, emb_dim=8, rnn_cell_dim=16, source_dim=16,
+      num_lstm_layers=1, dropout_prob=0.0, random_seed=3).Instantiate()
+  dec.eval()
+  g = torch.Generator().manual_seed(2)
+  enc = torch.randn(2, 6, 16, generator=g)
+  pad = torch.zeros(2, 6)
+  base = dec.GreedyDecode(dec.theta, enc, pad, max_len=5)
+
+  lm = lm_lib.TransformerLm.Params().Set(
+      name='lm', vocab_size=V, model_dim=16, num_layers=1, num_heads=1,
+      hidden_dim=32, dropout_prob=0.0, random_seed=5).Instantiate()
+  lm.eval()
+  fusion = asr_lib.ShallowFusion(lm, lm.theta, weight=100.0)
+  fused = dec.GreedyDecode(dec.theta, enc, pad, max_len=5,
+                           fusion=fusion)
+  assert fused.shape[0] == 2
+  # overwhelming LM weight changes the hypotheses
+  assert not torch.equal(base[:, :fused.shape[1]], fused[:, :base.shape[1]])
+  # zero weight reproduces the acoustic-only decode
+  fusion0 = asr_lib.ShallowFusion(lm, lm.theta, weight=0.0)
+  same = dec.GreedyDecode(dec.theta, enc, pad, max_len=5, fusion=fusion0)
+  assert torch.equal(base, same)
