@@ -224,3 +224,124 @@ class PillarsModel(BaseTask):
       decode_metrics.ap3d.Update(ap)
     decode_metrics.num_samples_in_batch.Update(
         float(decode_out.ap.shape[0]))
+
+
+class StarNetModel(BaseTask):
+  """Point-based 3D detector (reference tasks/car StarNet,
+  starnet.py: sampled anchor centers + per-cell featurizer): furthest-
+  point-sample C centers, gather the K nearest points per center,
+  featurize each cell with a shared MLP + max-pool, and predict an
+  occupancy logit + box residuals per center. Contrast with
+  PillarsModel's dense BEV grid: compute follows the points, not the
+  scene area."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('num_centers', 64, 'Sampled anchor centers C.')
+    p.Define('num_neighbors', 64, 'Points gathered per center K.')
+    p.Define('feat_dim', 64, 'Cell feature dim.')
+    p.Define('match_radius', 3.0, 'Center-to-gt match distance (m).')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    self.CreateChild('cell_mlp', lingvo_layers.FeedForwardNet.Params().Set(
+        input_dim=3, hidden_layer_dims=[32, p.feat_dim]))
+    self.CreateChild('head', lingvo_layers.FeedForwardNet.Params().Set(
+        input_dim=p.feat_dim, hidden_layer_dims=[p.feat_dim, 9],
+        activation=['RELU', 'NONE']))
+
+  def _SampleCenters(self, points):
+    from lingvo_amd.models import car_ops
+    b = points.shape[0]
+    centers = []
+    for i in range(b):
+      idx = car_ops.SamplePoints(points[i], self.p.num_centers,
+                                 seed=1234 + i)
+      centers.append(points[i, idx])
+    return torch.stack(centers)  # [B, C, 3]
+
+  def _Forward(self, theta, points):
+    p = self.p
+    centers = self._SampleCenters(points)              # [B, C, 3]
+    d = torch.cdist(centers, points)                   # [B, C, N]
+    knn = d.topk(p.num_neighbors, largest=False).indices
+    gathered = torch.gather(
+        points.unsqueeze(1).expand(-1, centers.shape[1], -1, -1), 2,
+        knn.unsqueeze(-1).expand(-1, -1, -1, 3))       # [B, C, K, 3]
+    rel = gathered - centers.unsqueeze(2)
+    feats = self.cell_mlp.FProp(theta.cell_mlp, rel).max(dim=2).values
+    out = self.head.FProp(theta.head, feats)           # [B, C, 9]
+    return centers, out
+
+  def _CenterTargets(self, centers, gt_boxes, num_boxes):
+    p = self.p
+    b, c, _ = centers.shape
+    occ = torch.zeros(b, c, device=centers.device)
+    reg = torch.zeros(b, c, 8, device=centers.device)
+    for i in range(b):
+      k = int(num_boxes[i])
+      if k == 0:
+        continue
+      boxes = gt_boxes[i, :k]
+      d = torch.cdist(centers[i, :, :2], boxes[:, :2])  # [C, k]
+      best = d.argmin(dim=1)
+      near = d.gather(1, best.unsqueeze(1)).squeeze(1) < p.match_radius
+      for ci in torch.nonzero(near, as_tuple=True)[0].tolist():
+        box = boxes[best[ci]]
+        occ[i, ci] = 1.0
+        reg[i, ci] = torch.stack([
+            box[0] - centers[i, ci, 0], box[1] - centers[i, ci, 1],
+            box[2], box[3].log(), box[4].log(), box[5].log(),
+            box[6].sin(), box[6].cos()])
+    return occ, reg
+
+  def ComputePredictions(self, theta, input_batch):
+    centers, out = self._Forward(theta, input_batch.points)
+    return NestedMap(centers=centers, head_out=out)
+
+  def ComputeLoss(self, theta, predictions, input_batch):
+    out = predictions.head_out.float()
+    occ_logit = out[..., 0]
+    reg_pred = out[..., 1:]
+    occ, reg = self._CenterTargets(predictions.centers,
+                                   input_batch.gt_boxes,
+                                   input_batch.num_boxes)
+    occ_loss = F.binary_cross_entropy_with_logits(occ_logit, occ)
+    mask = occ.unsqueeze(-1)
+    denom = mask.sum().clamp_min(1.0)
+    reg_loss = (F.smooth_l1_loss(reg_pred * mask, reg * mask,
+                                 reduction='sum') / denom)
+    loss = occ_loss + reg_loss
+    w = torch.tensor(float(out.shape[0]))
+    metrics = NestedMap(loss=(loss, w),
+                        occ_loss=(occ_loss.detach(), w),
+                        reg_loss=(reg_loss.detach(), w),
+                        num_samples_in_batch=(w, torch.ones(())))
+    return metrics, NestedMap()
+
+  def Decode(self, input_batch) -> NestedMap:
+    from lingvo_amd.models import car_ops
+    with torch.no_grad():
+      preds = self.ComputePredictions(self.theta, input_batch)
+    out = preds.head_out.float()
+    scores = torch.sigmoid(out[..., 0])
+    boxes_all, scores_all = [], []
+    for i in range(out.shape[0]):
+      ctr = preds.centers[i]
+      r = out[i, :, 1:]
+      boxes = torch.stack([
+          ctr[:, 0] + r[:, 0], ctr[:, 1] + r[:, 1], r[:, 2],
+          r[:, 3].exp(), r[:, 4].exp(), r[:, 5].exp(),
+          torch.atan2(r[:, 6], r[:, 7])], dim=1)
+      conf = scores[i] > 0.3
+      boxes_c, scores_c = boxes[conf], scores[i][conf]
+      keep = car_ops.NonMaxSuppression3D(boxes_c, scores_c,
+                                         iou_threshold=0.3)
+      boxes_all.append(boxes_c[keep])
+      scores_all.append(scores_c[keep])
+    return NestedMap(boxes=boxes_all, scores=scores_all,
+                     gt_boxes=input_batch.gt_boxes,
+                     num_boxes=input_batch.num_boxes)

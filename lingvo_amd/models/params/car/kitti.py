@@ -33,3 +33,27 @@ class StarNetPillars(SingleTaskModelParams):
         optimizer=optimizer_lib.Adam.Params(),
         clip_gradient_norm_to_value=5.0)
     return p
+
+
+@registry.RegisterSingleTaskModel
+class StarNet(SingleTaskModelParams):
+  """Point-based StarNet detector (reference car.kitti.StarNet*)."""
+
+  def Train(self):
+    return car_model.SyntheticPointCloudInput.Params().Set(
+        name='train', batch_size=4)
+
+  def Dev(self):
+    return self.Train().Set(name='dev')
+
+  def Test(self):
+    return self.Train().Set(name='test')
+
+  def Task(self):
+    p = car_model.StarNetModel.Params().Set(name='starnet')
+    p.fprop_dtype = torch.float32
+    p.train.learner = learner_lib.Learner.Params().Set(
+        learning_rate=1e-3,
+        optimizer=optimizer_lib.Adam.Params(),
+        clip_gradient_norm_to_value=5.0)
+    return p

@@ -138,3 +138,21 @@ def test_projection_layer_qdomain():
   # straight-through grads flow
   (out_q.sum()).backward()
   assert layer.w.grad is not None
+
+
+def test_starnet_train_and_decode():
+  import torch
+  from lingvo_amd.core import registry
+  model_p = registry.GetParams('car.kitti.StarNet', 'Train')
+  model_p.task.random_seed = 4
+  model_p.task.Set(num_centers=16, num_neighbors=16, feat_dim=16)
+  model_p.input.Set(batch_size=2, num_points=256)
+  model = model_p.Instantiate()
+  task = model.GetTask()
+  batch = task.GetInputBatch()
+  m = task.TrainStep(batch)
+  assert torch.isfinite(m['loss'][0])
+  out = task.Decode(batch)
+  assert len(out.boxes) == 2
+  for bx, sc in zip(out.boxes, out.scores):
+    assert bx.shape[1] == 7 and bx.shape[0] == sc.shape[0]
