@@ -19,3 +19,10 @@ def pytest_collection_modifyitems(config, items):
   for item in items:
     if 'gpu' in item.keywords:
       item.add_marker(skip_gpu)
+
+
+def dist_port(base: int) -> int:
+  """Deterministic per-test-run rendezvous port: distinct bases keep
+  tests apart; the pid offset avoids TIME_WAIT collisions across runs."""
+  import os
+  return 20000 + (base - 29500) * 131 % 9000 + os.getpid() % 997

@@ -2,6 +2,7 @@
 
 import os
 
+from conftest import dist_port
 import torch
 import torch.distributed as dist
 import torch.multiprocessing as mp
@@ -69,11 +70,11 @@ def _check_ring(causal, port):
 
 
 def test_ring_attention_full():
-  _check_ring(causal=False, port=29555)
+  _check_ring(causal=False, port=dist_port(29555))
 
 
 def test_ring_attention_causal():
-  _check_ring(causal=True, port=29556)
+  _check_ring(causal=True, port=dist_port(29556))
 
 
 def _run_cp_layer(rank, world, port, results):
@@ -100,7 +101,7 @@ def test_cp_layer_matches_full():
   ctx = mp.get_context('spawn')
   with ctx.Manager() as mgr:
     results = mgr.dict()
-    procs = [ctx.Process(target=_run_cp_layer, args=(r, 2, 29557, results))
+    procs = [ctx.Process(target=_run_cp_layer, args=(r, 2, dist_port(29557), results))
              for r in range(2)]
     for p in procs:
       p.start()
@@ -159,7 +160,7 @@ def test_ulysses_attention_matches_full():
   ctx = mp.get_context('spawn')
   with ctx.Manager() as mgr:
     results = mgr.dict()
-    procs = [ctx.Process(target=_run_ulysses, args=(r, 2, 29575, results))
+    procs = [ctx.Process(target=_run_ulysses, args=(r, 2, dist_port(29575), results))
              for r in range(2)]
     for p in procs:
       p.start()
@@ -210,7 +211,7 @@ def test_zigzag_ring_attention_exact():
   ctx = mp.get_context('spawn')
   with ctx.Manager() as mgr:
     results = mgr.dict()
-    procs = [ctx.Process(target=_run_zigzag, args=(r, 2, 29581, results))
+    procs = [ctx.Process(target=_run_zigzag, args=(r, 2, dist_port(29581), results))
              for r in range(2)]
     for p in procs:
       p.start()

@@ -4,6 +4,7 @@ path the driver's 8-GPU scaling bench exercises with RCCL."""
 import os
 
 import pytest
+from conftest import dist_port
 import torch
 import torch.distributed as dist
 import torch.multiprocessing as mp
@@ -31,7 +32,7 @@ def _run_gradsync(rank, world, port, results):
 
 
 def test_gradsync_averages_across_ranks(tmp_path):
-  port = 29531
+  port = dist_port(29531)
   ctx = mp.get_context('spawn')
   with ctx.Manager() as mgr:
     results = mgr.dict()
@@ -84,7 +85,7 @@ def _run_bench_dp(rank, world, port, results):
 
 
 def test_full_task_dp_training_keeps_replicas_in_sync():
-  port = 29532
+  port = dist_port(29532)
   ctx = mp.get_context('spawn')
   with ctx.Manager() as mgr:
     results = mgr.dict()

@@ -4,6 +4,7 @@ MoE LM train step."""
 import os
 
 import pytest
+from conftest import dist_port
 import torch
 import torch.distributed as dist
 import torch.multiprocessing as mp
@@ -65,7 +66,7 @@ def test_moe_ep2_matches_local():
   ctx = mp.get_context('spawn')
   with ctx.Manager() as mgr:
     results = mgr.dict()
-    procs = [ctx.Process(target=_run_ep, args=(r, 2, 29534, results))
+    procs = [ctx.Process(target=_run_ep, args=(r, 2, dist_port(29534), results))
              for r in range(2)]
     for p in procs:
       p.start()

@@ -4,6 +4,7 @@
 import os
 
 import pytest
+from conftest import dist_port
 import torch
 import torch.distributed as dist
 import torch.multiprocessing as mp
@@ -57,7 +58,7 @@ def test_gpipe_two_stage_matches_reference():
   with ctx.Manager() as mgr:
     results = mgr.dict()
     procs = [ctx.Process(target=_run_stage,
-                         args=(r, 2, 29533, num_micro, results))
+                         args=(r, 2, dist_port(29533), num_micro, results))
              for r in range(2)]
     for p in procs:
       p.start()
@@ -127,7 +128,7 @@ def test_gpipe_transformer_lm_two_stages():
   with ctx.Manager() as mgr:
     results = mgr.dict()
     procs = [ctx.Process(target=_run_lm_stage,
-                         args=(r, 2, 29536, num_micro, results))
+                         args=(r, 2, dist_port(29536), num_micro, results))
              for r in range(2)]
     for p in procs:
       p.start()
@@ -197,7 +198,7 @@ def test_1f1b_matches_fill_drain_reference(world):
   with ctx.Manager() as mgr:
     results = mgr.dict()
     procs = [ctx.Process(target=_run_stage_1f1b,
-                         args=(r, world, 29537 + world, num_micro,
+                         args=(r, world, dist_port(29537) + world, num_micro,
                                results))
              for r in range(world)]
     for p in procs:
@@ -271,7 +272,7 @@ def test_gpipe_mt_two_stages(schedule):
   with ctx.Manager() as mgr:
     results = mgr.dict()
     procs = [ctx.Process(target=_run_mt_stage,
-                         args=(r, 2, 29541 + (schedule == '1f1b'),
+                         args=(r, 2, dist_port(29541) + (schedule == '1f1b'),
                                num_micro, schedule, results))
              for r in range(2)]
     for p in procs:
@@ -355,7 +356,7 @@ def test_ppdp_grid_2x2_matches_single_process():
   with ctx.Manager() as mgr:
     results = mgr.dict()
     procs = [ctx.Process(target=_run_ppdp,
-                         args=(r, 4, 29561, num_micro, results))
+                         args=(r, 4, dist_port(29561), num_micro, results))
              for r in range(4)]
     for p in procs:
       p.start()

@@ -3,6 +3,7 @@
 import os
 
 import pytest
+from conftest import dist_port
 import torch
 import torch.distributed as dist
 import torch.multiprocessing as mp
@@ -35,7 +36,7 @@ def test_tp2_matches_single_process():
   ctx = mp.get_context('spawn')
   with ctx.Manager() as mgr:
     results = mgr.dict()
-    procs = [ctx.Process(target=_run_tp, args=(r, 2, 29535, results))
+    procs = [ctx.Process(target=_run_tp, args=(r, 2, dist_port(29535), results))
              for r in range(2)]
     for p in procs:
       p.start()
@@ -104,7 +105,7 @@ def test_tp_attention_matches_single_process():
   ctx = mp.get_context('spawn')
   with ctx.Manager() as mgr:
     results = mgr.dict()
-    procs = [ctx.Process(target=_run_tp_attn, args=(r, 2, 29545, results))
+    procs = [ctx.Process(target=_run_tp_attn, args=(r, 2, dist_port(29545), results))
              for r in range(2)]
     for p in procs:
       p.start()
