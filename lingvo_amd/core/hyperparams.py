@@ -37,7 +37,14 @@ class _Param:
     self.description = description
 
   def __deepcopy__(self, memo):
-    p = _Param(self.name, _copy.deepcopy(self.value, memo), self.description)
+    try:
+      value = _copy.deepcopy(self.value, memo)
+    except TypeError:
+      # Runtime handles (e.g. torch.distributed ProcessGroup) are not
+      # copyable — share them by reference (reference hyperparams
+      # behavior for non-copyable leaf objects).
+      value = self.value
+    p = _Param(self.name, value, self.description)
     memo[id(self)] = p
     return p
 

@@ -63,3 +63,35 @@ class PpDpTopology:
       if param.grad is not None:
         dist.all_reduce(param.grad, group=self.dp_group)
         param.grad /= self.dp_degree
+
+
+class TpDpTopology:
+  """Tensor x data-parallel grid: TP groups are CONTIGUOUS ranks (same
+  node / max xGMI locality, where the per-block all-reduce lives), DP
+  groups stride across them.
+
+      rank = dp_idx * tp_degree + tp_idx
+  """
+
+  def __init__(self, tp_degree: int, world: Optional[int] = None,
+               rank: Optional[int] = None):
+    world = world if world is not None else dist.get_world_size()
+    rank = rank if rank is not None else dist.get_rank()
+    assert world % tp_degree == 0, (world, tp_degree)
+    self.tp_degree = tp_degree
+    self.dp_degree = world // tp_degree
+    self.tp_idx = rank % tp_degree
+    self.dp_idx = rank // tp_degree
+    tp_groups = [dist.new_group(list(range(d * tp_degree,
+                                           (d + 1) * tp_degree)))
+                 for d in range(self.dp_degree)]
+    dp_groups = [dist.new_group(list(range(t, world, tp_degree)))
+                 for t in range(tp_degree)]
+    self.tp_group = tp_groups[self.dp_idx]
+    self.dp_group = dp_groups[self.tp_idx]
+
+  def MakeGradSync(self, module, **kwargs):
+    """DP gradient sync over the replica axis only (TP-sharded weights
+    differ per tp rank; their replicas live across the DP axis)."""
+    from lingvo_amd.parallel.ddp import GradSync
+    return GradSync(module, process_group=self.dp_group, **kwargs)

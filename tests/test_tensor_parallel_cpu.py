@@ -167,3 +167,59 @@ def test_lower_sharding_annotations():
   tp.LowerShardingAnnotations(sp2)
   assert sp2.transformer_tpl.tr_fflayer_tpl.cls is not \
       tp.TpFeedForwardLayer
+
+
+def _run_tpdp(rank, world, port, results):
+  os.environ['MASTER_ADDR'] = '127.0.0.1'
+  os.environ['MASTER_PORT'] = str(port)
+  dist.init_process_group('gloo', rank=rank, world_size=world)
+  from lingvo_amd.parallel.topology import TpDpTopology
+  topo = TpDpTopology(tp_degree=2)
+  layer = _ffn_params(tp_group=topo.tp_group).Instantiate()
+  sync = topo.MakeGradSync(layer)
+  g = torch.Generator().manual_seed(600 + topo.dp_idx)  # per-replica data
+  x = torch.randn(2, 6, 16, generator=g)
+  out = layer.FProp(layer.theta, x)
+  out.sum().backward()
+  sync.Finalize()
+  results[f'coord{rank}'] = (topo.tp_idx, topo.dp_idx)
+  results[f'wi{rank}'] = layer.wi.w.grad.clone()
+  results[f'out{rank}'] = out.detach()
+  dist.destroy_process_group()
+
+
+def test_tpdp_grid_2x2():
+  """TP=2 x DP=2: shard grads DP-average to the single-process ref."""
+  ctx = mp.get_context('spawn')
+  with ctx.Manager() as mgr:
+    results = mgr.dict()
+    procs = [ctx.Process(target=_run_tpdp,
+                         args=(r, 4, dist_port(29591), results))
+             for r in range(4)]
+    for p in procs:
+      p.start()
+    for p in procs:
+      p.join(180)
+      assert p.exitcode == 0
+    results = dict(results)
+  coords = {results[f'coord{r}']: r for r in range(4)}
+  assert set(coords) == {(0, 0), (0, 1), (1, 0), (1, 1)}
+
+  # single-process reference over BOTH replicas' data, averaged
+  layer = _ffn_params().Instantiate()
+  for dp_idx in range(2):
+    g = torch.Generator().manual_seed(600 + dp_idx)
+    x = torch.randn(2, 6, 16, generator=g)
+    out = layer.FProp(layer.theta, x)
+    (out.sum() / 2).backward()
+    # replicas agree on their outputs with the full layer
+    for t in range(2):
+      r = coords[(t, dp_idx)]
+      assert torch.allclose(results[f'out{r}'], out.detach(), atol=1e-5)
+  # each tp shard's DP-averaged grad == matching slice of full grad
+  for t in range(2):
+    sl = slice(t * 16, (t + 1) * 16)
+    for dp_idx in range(2):
+      r = coords[(t, dp_idx)]
+      assert torch.allclose(results[f'wi{r}'],
+                            layer.wi.w.grad[:, sl], atol=1e-5), (t, dp_idx)
