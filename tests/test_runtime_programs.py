@@ -215,3 +215,34 @@ def test_server_micro_batching(tmp_path):
   b = app.state.batcher
   assert b.examples_run >= 7
   assert b.batches_run < b.examples_run  # at least one coalesced batch
+
+
+def test_speculative_decoding_exact_and_fewer_target_calls():
+  from lingvo_amd.models import lm as lm_lib
+  from lingvo_amd.runtime import speculative
+  V = 24
+  def mk(seed, layers):
+    p = lm_lib.TransformerLm.Params().Set(
+        name='lm', vocab_size=V, model_dim=16, num_layers=layers,
+        num_heads=1, hidden_dim=32, dropout_prob=0.0, random_seed=seed)
+    m = p.Instantiate()
+    m.eval()
+    return m
+  target = mk(3, 2)
+  draft = mk(9, 1)
+  g = torch.Generator().manual_seed(1)
+  prefix = torch.randint(3, V, (2, 4), generator=g)
+  spec = speculative.SpeculativeDecoder(target, target.theta,
+                                        draft, draft.theta, lookahead=4)
+  out = spec.Generate(prefix, max_new=16)
+  ref = speculative.GreedyReference(target, target.theta, prefix, 16)
+  n = min(out.ids.shape[1], ref.shape[1])
+  assert torch.equal(out.ids[:, :n], ref[:, :n])
+  # batched verification: fewer target calls than generated tokens
+  assert out.stats['target_calls'] < out.new_tokens
+  # a draft equal to the target accepts (nearly) everything
+  spec2 = speculative.SpeculativeDecoder(target, target.theta,
+                                         target, target.theta,
+                                         lookahead=4)
+  out2 = spec2.Generate(prefix, max_new=12)
+  assert out2.stats['accepted'] == out2.stats['proposed']
