@@ -546,3 +546,44 @@ def test_evolved_transformer_layers():
   o2 = dec.FProp(dec.theta, x2, pad)
   assert (o1[:, :6] - o2[:, :6]).abs().max() < 1e-5
   assert (o1[0, 6:] - o2[0, 6:]).abs().max() > 1e-4
+
+
+def test_lora_adaptation_and_merge():
+  from lingvo_amd.layers import lora
+  from lingvo_amd.models import lm as lm_lib
+  lm = lm_lib.TransformerLm.Params().Set(
+      name='lm', vocab_size=32, model_dim=16, num_layers=1, num_heads=1,
+      hidden_dim=32, dropout_prob=0.0, random_seed=4).Instantiate()
+  lm.eval()
+  ids = torch.randint(3, 32, (2, 6))
+  pads = torch.zeros(2, 6)
+  base_out = lm.FProp(lm.theta, ids, pads).detach()
+
+  trainable = lora.ApplyLora(lm, rank=4, alpha=8.0, seed=1)
+  assert trainable
+  # B zero-init: output unchanged at attach time
+  assert torch.allclose(lm.FProp(lm.theta, ids, pads).detach(),
+                        base_out, atol=1e-5)
+  # only LoRA params are trainable
+  n_train = sum(1 for p in lm.parameters() if p.requires_grad)
+  assert n_train == len(trainable)
+  # a training step moves the output through the deltas only
+  opt = torch.optim.SGD(trainable, lr=0.5)
+  loss = lm.FProp(lm.theta, ids, pads).square().sum()
+  loss.backward()
+  base_snapshot = {n: p.detach().clone()
+                   for n, p in lm.named_parameters()
+                   if not p.requires_grad and 'lora' not in n}
+  opt.step()
+  tuned_out = lm.FProp(lm.theta, ids, pads).detach()
+  assert (tuned_out - base_out).abs().max() > 1e-5
+  for n, p in lm.named_parameters():
+    if n in base_snapshot:
+      assert torch.equal(p.detach(), base_snapshot[n]), n
+
+  # merge folds deltas into the base weights: same output, no pairs
+  merged = lora.MergeLora(lm)
+  assert merged > 0
+  assert not any('lora' in n for n, _ in lm.named_parameters())
+  assert torch.allclose(lm.FProp(lm.theta, ids, pads).detach(),
+                        tuned_out, atol=1e-4)
