@@ -170,6 +170,37 @@ class TransformerLm(BaseLayer):
     return self.softmax.XentLoss(theta.softmax, act, class_weights=weights,
                                  class_ids=labels)
 
+  @torch.no_grad()
+  def Generate(self, theta: NestedMap, prefix: torch.Tensor,
+               max_new: int, eos_id: int = 2) -> torch.Tensor:
+    """Greedy incremental decode over the stack's KV cache
+    (ExtendStep): each new token costs one single-position pass instead
+    of re-running the whole prefix. Matches full-FProp greedy exactly.
+    """
+    b, t0 = prefix.shape
+    total = t0 + max_new
+    states = self.stack.InitStates(theta.stack, b, total,
+                                   prefix.device, self.fprop_dtype)
+    pos = self.pos_emb.FProp(theta.pos_emb, total, device=prefix.device)
+    ids = prefix.clone()
+    done = torch.zeros(b, dtype=torch.bool, device=prefix.device)
+    tok = None
+    for t in range(total - 1):
+      cur = prefix[:, t] if t < t0 else tok
+      x = self._Emb(theta, cur.long()).to(self.fprop_dtype)
+      x = (x + pos[t].to(x.dtype)).unsqueeze(1)          # [B, 1, D]
+      act, states = self.stack.ExtendStep(theta.stack, x, states)
+      if t < t0 - 1:
+        continue  # prefill: just populate the cache
+      logits = self.softmax.Logits(theta.softmax, act[:, 0])
+      tok = logits.argmax(-1)
+      tok = torch.where(done, torch.full_like(tok, eos_id), tok)
+      done = done | (tok == eos_id)
+      ids = torch.cat([ids, tok.unsqueeze(1)], dim=1)
+      if bool(done.all()):
+        break
+    return ids
+
 
 class RnnLm(BaseLayer):
   """LSTM LM (reference tasks/lm/layers.py:495 RnnLm; the

@@ -150,3 +150,19 @@ def test_packed_lm_matches_unpacked():
                  segment_ids=seg, segment_pos=pos)
   assert (out[0, :5] - ref[0]).abs().max() < 1e-4
   assert (out[0, 5:] - ref[1, :3]).abs().max() < 1e-4
+
+
+def test_lm_kv_cache_generate_matches_full_greedy():
+  import torch
+  from lingvo_amd.models import lm as lm_lib
+  from lingvo_amd.runtime import speculative
+  lm = lm_lib.TransformerLm.Params().Set(
+      name='lm', vocab_size=32, model_dim=16, num_layers=2, num_heads=1,
+      hidden_dim=32, dropout_prob=0.0, random_seed=6).Instantiate()
+  lm.eval()
+  g = torch.Generator().manual_seed(3)
+  prefix = torch.randint(3, 32, (2, 5), generator=g)
+  fast = lm.Generate(lm.theta, prefix, max_new=10)
+  ref = speculative.GreedyReference(lm, lm.theta, prefix, 10)
+  n = min(fast.shape[1], ref.shape[1])
+  assert torch.equal(fast[:, :n], ref[:, :n])
