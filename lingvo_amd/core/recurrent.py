@@ -25,11 +25,20 @@ def Recurrent(theta: NestedMap, state0: NestedMap, inputs: NestedMap,
               cell_fn: Callable[[NestedMap, NestedMap, NestedMap],
                                 Tuple[NestedMap, NestedMap]],
               remat: bool = False,
-              stop_fn=None) -> Tuple[NestedMap, NestedMap]:
+              stop_fn=None,
+              check_stateful_ops: bool = False
+              ) -> Tuple[NestedMap, NestedMap]:
   """Returns (acc_states, final_state).
 
   cell_fn(theta, state, inputs_t) -> (new_state, extras). acc_states
   stacks every new_state along a leading time dim.
+
+  check_stateful_ops (reference recurrent.py:1046 /
+  StatefulRandomOpsInDefun py_utils.py:5106): the remat backward
+  RE-RUNS cell_fn, so hidden statefulness (global RNG, in-place
+  mutation of captured state) silently corrupts gradients. With the
+  flag on, the first step runs cell_fn twice and asserts identical
+  outputs — a runtime stand-in for the reference's graph-time scan.
   """
   t_max = None
   for v in inputs.Flatten():
@@ -37,6 +46,17 @@ def Recurrent(theta: NestedMap, state0: NestedMap, inputs: NestedMap,
       t_max = v.shape[0] if t_max is None else min(t_max, v.shape[0])
   assert t_max is not None, 'Recurrent needs at least one tensor input'
 
+  if check_stateful_ops and t_max > 0:
+    probe_in = _SliceT(inputs, 0)
+    out_a, _ = cell_fn(theta, state0, probe_in)
+    out_b, _ = cell_fn(theta, state0, probe_in)
+    for va, vb in zip(out_a.Flatten(), out_b.Flatten()):
+      if isinstance(va, torch.Tensor) and not torch.equal(va, vb):
+        raise RuntimeError(
+            'Recurrent cell_fn is stateful (two runs differ) — '
+            'recompute-based backward would be wrong. Use the '
+            'deterministic step-seed RNG (py_utils.GraphSafeUniform) '
+            'inside scanned cells.')
   state = state0
   acc = []
   for t in range(t_max):

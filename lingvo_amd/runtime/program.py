@@ -203,6 +203,12 @@ class Executor:
       cycle_secs = time.perf_counter() - t0
       self.ckpt.MaybeSave()
       step = self.task.global_step
+      # per-cycle wall time (reference executor_cycle_secs export,
+      # executor.py:584-590)
+      with open(os.path.join(self.logdir, 'executor_metrics.jsonl'),
+                'a') as f:
+        f.write(json.dumps({'step': int(step),
+                            'executor_cycle_secs': cycle_secs}) + '\n')
       out = getattr(self.schedule, 'last_result', None)
       if out is not None and 'train' in out:
         self.trial.ReportEvalMeasure(step,

@@ -494,3 +494,43 @@ def test_input_benchmark_and_np_arrays(tmp_path):
                          {'w': torch.arange(6).reshape(2, 3)})
   back = ckpt_lib.ReadNpArrays(str(tmp_path / 'arrays'))
   assert torch.equal(back['w'], torch.arange(6).reshape(2, 3))
+
+
+def test_recurrent_stateful_op_detection():
+  import torch
+  from lingvo_amd.core import recurrent
+  from lingvo_amd.core.nested_map import NestedMap
+
+  def good_cell(theta, state, inp):
+    return NestedMap(h=state.h + inp.x), NestedMap()
+
+  inputs = NestedMap(x=torch.randn(4, 3))
+  out, final = recurrent.Recurrent(
+      NestedMap(), NestedMap(h=torch.zeros(3)), inputs, good_cell,
+      check_stateful_ops=True)
+  assert torch.allclose(final.h, inputs.x.sum(0))
+
+  def bad_cell(theta, state, inp):
+    return NestedMap(h=state.h + torch.rand_like(state.h)), NestedMap()
+
+  import pytest
+  with pytest.raises(RuntimeError, match='stateful'):
+    recurrent.Recurrent(NestedMap(), NestedMap(h=torch.zeros(3)),
+                        inputs, bad_cell, check_stateful_ops=True)
+
+
+def test_executor_cycle_metrics(tmp_path):
+  import json
+  from lingvo_amd.core import registry
+  from lingvo_amd.runtime import program as program_lib
+  mp2 = registry.GetParams('image.mnist.LeNet5', 'Train')
+  mp2.task.random_seed = 3
+  sched = program_lib.SimpleProgramSchedule.Params()
+  sched.train_program.steps_per_loop = 1
+  ex = program_lib.Executor(mp2, str(tmp_path), sched, device='cpu',
+                            max_steps=2)
+  ex.Start()
+  recs = [json.loads(l)
+          for l in open(tmp_path / 'executor_metrics.jsonl')]
+  assert len(recs) >= 2
+  assert all(r['executor_cycle_secs'] > 0 for r in recs)
