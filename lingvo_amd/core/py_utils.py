@@ -450,3 +450,30 @@ def SinkhornAssignment(scores: torch.Tensor, tau: float = 0.1,
     log_alpha = log_alpha - torch.logsumexp(log_alpha, dim=-2,
                                             keepdim=True)
   return torch.exp(log_alpha)
+
+
+def AssertShapeMatch(tensor: torch.Tensor, pattern) -> torch.Tensor:
+  """Shape assert (reference x_ops.cc:26 AssertShapeMatch): -1 entries
+  are wildcards. Returns the tensor for chaining."""
+  shape = tuple(tensor.shape)
+  assert len(shape) == len(pattern), (shape, pattern)
+  for got, want in zip(shape, pattern):
+    assert want == -1 or got == want, (shape, pattern)
+  return tensor
+
+
+def AssertIdShape(*tensors: torch.Tensor) -> None:
+  """All tensors share one shape (reference assert_kernels.cc)."""
+  shapes = {tuple(t.shape) for t in tensors}
+  assert len(shapes) == 1, shapes
+
+
+def EstimateFlops(fn, *args) -> int:
+  """Measured flop count of one call via the torch profiler
+  (the runtime stand-in for the reference's symbolic FPropMeta /
+  computation_cost.py estimates — counts real matmul/conv flops)."""
+  from torch.profiler import profile, ProfilerActivity
+  with profile(activities=[ProfilerActivity.CPU],
+               with_flops=True) as prof:
+    fn(*args)
+  return int(sum(e.flops for e in prof.key_averages() if e.flops))

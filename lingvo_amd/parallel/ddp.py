@@ -164,3 +164,25 @@ class GradSync:
     for h in self._hooks:
       h.remove()
     self._hooks = []
+
+
+def TwoShotAllReduce(t: torch.Tensor, group=None) -> torch.Tensor:
+  """Sum all-reduce as reduce-scatter (via all-to-all) + all-gather
+  (SURVEY §5 comm note: on 8-way xGMI every GPU has 7 point-to-point
+  links and NO switch, so a ring all-reduce is bound by ONE link while
+  the two-shot form drives all 7 simultaneously — better for the
+  latency/medium-size buckets the backward produces). Exact for any
+  1-D float tensor; pads to a world multiple internally."""
+  world = dist.get_world_size(group)
+  if world == 1:
+    return t
+  n = t.numel()
+  chunk = -(-n // world)
+  buf = torch.zeros(world * chunk, dtype=t.dtype, device=t.device)
+  buf[:n] = t.reshape(-1)
+  recv = torch.empty_like(buf)
+  dist.all_to_all_single(recv, buf, group=group)
+  mine = recv.reshape(world, chunk).sum(dim=0)         # my reduced chunk
+  out = torch.empty_like(buf)
+  dist.all_gather_into_tensor(out, mine.contiguous(), group=group)
+  return out[:n].reshape(t.shape)

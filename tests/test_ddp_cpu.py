@@ -201,3 +201,32 @@ def test_gradsync_with_graddrop_compressor():
                         atol=1e-5)
   # dropped 0.1 elements live in the residual for the next step
   assert abs(float(res0[0, 2]) - 0.1) < 1e-6
+
+
+def _run_two_shot(rank, world, port, results):
+  os.environ['MASTER_ADDR'] = '127.0.0.1'
+  os.environ['MASTER_PORT'] = str(port)
+  dist.init_process_group('gloo', rank=rank, world_size=world)
+  from lingvo_amd.parallel.ddp import TwoShotAllReduce
+  g = torch.Generator().manual_seed(100 + rank)
+  t = torch.randn(37, generator=g)  # odd size: exercises padding
+  ref = t.clone()
+  dist.all_reduce(ref)
+  out = TwoShotAllReduce(t)
+  results[f'ok{rank}'] = bool(torch.allclose(out, ref, atol=1e-6))
+  dist.destroy_process_group()
+
+
+def test_two_shot_all_reduce_matches_ring():
+  ctx = mp.get_context('spawn')
+  port = dist_port(29597)
+  with ctx.Manager() as mgr:
+    results = mgr.dict()
+    procs = [ctx.Process(target=_run_two_shot, args=(r, 2, port, results))
+             for r in range(2)]
+    for p in procs:
+      p.start()
+    for p in procs:
+      p.join(120)
+      assert p.exitcode == 0
+    assert results['ok0'] and results['ok1']

@@ -534,3 +534,16 @@ def test_executor_cycle_metrics(tmp_path):
           for l in open(tmp_path / 'executor_metrics.jsonl')]
   assert len(recs) >= 2
   assert all(r['executor_cycle_secs'] > 0 for r in recs)
+
+
+def test_shape_asserts_and_flop_estimate():
+  import torch
+  from lingvo_amd.core import py_utils as pu
+  x = torch.randn(4, 8)
+  pu.AssertShapeMatch(x, (4, -1))
+  pu.AssertIdShape(x, torch.zeros(4, 8))
+  with pytest.raises(AssertionError):
+    pu.AssertShapeMatch(x, (4, 9))
+  flops = pu.EstimateFlops(lambda: torch.randn(32, 64) @
+                           torch.randn(64, 16))
+  assert abs(flops - 2 * 32 * 64 * 16) / (2 * 32 * 64 * 16) < 0.2
