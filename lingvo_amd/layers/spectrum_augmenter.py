@@ -52,6 +52,38 @@ class SpectrumAugmenter(BaseLayer):
     shape[dim] = size
     return x * mask.reshape(shape).to(x.dtype)
 
+  def _TimeWarp(self, x, lengths):
+    """SpecAugment time warping (reference spectrum_augmenter.py
+    _TimeWarp): a random anchor frame w in [W, len-W] shifts by a random
+    distance d in [-W, W]; frames re-sample linearly on both sides.
+    Implemented as a per-example piecewise-linear index map + gather
+    with linear interpolation (graph-safe RNG, no host scalars)."""
+    p = self.p
+    b, t, f = x.shape
+    W = float(p.time_warp_max_frames)
+    lens = lengths.to(x.device).float().clamp(min=2 * W + 2)
+    u1 = py_utils.GraphSafeUniform((b,), x.device)
+    u2 = py_utils.GraphSafeUniform((b,), x.device)
+    anchor = W + u1 * (lens - 2 * W)                  # [B]
+    shift = (u2 * 2.0 - 1.0) * W
+    src_anchor = anchor + shift                       # sample source
+    pos = torch.arange(t, device=x.device).float()[None, :]  # [1, T]
+    lens_b = lens[:, None]
+    a = anchor[:, None]
+    sa = src_anchor[:, None]
+    left = pos / a.clamp_min(1.0) * sa
+    right = sa + (pos - a) / (lens_b - a).clamp_min(1.0) * (lens_b - sa)
+    src = torch.where(pos < a, left, right).clamp(0, t - 1 - 1e-4)
+    lo = src.floor().long()
+    frac = (src - lo.float()).unsqueeze(-1).to(x.dtype)
+    x_lo = torch.gather(x, 1, lo.unsqueeze(-1).expand(-1, -1, f))
+    x_hi = torch.gather(x, 1, (lo + 1).clamp(max=t - 1)
+                        .unsqueeze(-1).expand(-1, -1, f))
+    warped = x_lo * (1 - frac) + x_hi * frac
+    # only warp within the unpadded region
+    keep = pos >= lens_b
+    return torch.where(keep.unsqueeze(-1), x, warped)
+
   def FProp(self, theta: NestedMap, inputs: torch.Tensor,
             paddings: torch.Tensor) -> torch.Tensor:
     """inputs [B, T, F]; masking only in training."""
@@ -60,6 +92,8 @@ class SpectrumAugmenter(BaseLayer):
       return inputs
     lengths = py_utils.LengthsFromPaddings(paddings)
     x = inputs
+    if p.time_warp_max_frames:
+      x = self._TimeWarp(x, lengths)
     if p.time_mask_count:
       max_frames = p.time_mask_max_frames
       x = self._MaskDim(x, lengths, max_frames, p.time_mask_count, 1)
