@@ -246,3 +246,62 @@ def test_speculative_decoding_exact_and_fewer_target_calls():
                                          lookahead=4)
   out2 = spec2.Generate(prefix, max_new=12)
   assert out2.stats['accepted'] == out2.stats['proposed']
+
+
+def test_runner_retry_policy(tmp_path, monkeypatch):
+  """Transient errors retry with backoff; fatal errors re-raise."""
+  import time as time_mod
+  from lingvo_amd.runtime import runners
+  monkeypatch.setattr(time_mod, 'sleep', lambda s: None)
+  mp2 = registry.GetParams('image.mnist.LeNet5', 'Train')
+  r = runners.BaseRunner(mp2, str(tmp_path), 'test', max_retries=3)
+
+  calls = {'n': 0}
+  def flaky():
+    calls['n'] += 1
+    if calls['n'] < 3:
+      raise ConnectionError('transient')
+  r._RunLoop(flaky)
+  assert calls['n'] == 3
+
+  def always_broken():
+    raise ConnectionError('never recovers')
+  with pytest.raises(ConnectionError):
+    r._RunLoop(always_broken)
+
+  def fatal():
+    raise FloatingPointError('nan loss')
+  with pytest.raises(FloatingPointError):
+    r._RunLoop(fatal)
+  # status message recorded the retries
+  with open(tmp_path / 'test_status.txt') as f:
+    assert 'transient error' in f.read()
+
+
+def test_checkpoint_gc_keeps_latest_n(tmp_path):
+  from lingvo_amd.core import checkpointer as ckpt_lib
+  import glob as globlib
+  saver = ckpt_lib.Saver(str(tmp_path), keep_latest_n=3)
+  for step in range(1, 8):
+    saver.Save({'model': {'w': torch.ones(2)}, 'step': step}, step)
+  kept = sorted(globlib.glob(str(tmp_path / 'ckpt-*.pt')))
+  assert len(kept) == 3
+  assert kept[-1].endswith('ckpt-00000007.pt')
+  # the state file lists only surviving checkpoints
+  with open(tmp_path / 'checkpoint') as f:
+    txt = f.read()
+  assert 'ckpt-00000007.pt' in txt and 'ckpt-00000001.pt' not in txt
+  # non-finite payloads refuse to save
+  with pytest.raises(FloatingPointError):
+    saver.Save({'model': {'w': torch.tensor([float('nan')])},
+                'step': 9}, 9)
+
+
+def test_async_save_completes(tmp_path):
+  from lingvo_amd.core import checkpointer as ckpt_lib
+  saver = ckpt_lib.Saver(str(tmp_path), async_save=True)
+  saver.Save({'model': {'w': torch.arange(4.0)}, 'step': 1}, 1)
+  saver.Sync()
+  payload = torch.load(tmp_path / 'ckpt-00000001.pt',
+                       weights_only=False)
+  assert torch.equal(payload['model']['w'], torch.arange(4.0))
