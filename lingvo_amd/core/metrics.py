@@ -164,15 +164,26 @@ class AUCMetric(BaseMetric):
 
   @property
   def value(self) -> float:
-    pairs = sorted(zip(self._scores, self._labels))
     n_pos = sum(self._labels)
     n_neg = len(self._labels) - n_pos
     if not n_pos or not n_neg:
       return 0.0
-    rank_sum = 0.0
-    for rank, (_, label) in enumerate(pairs, start=1):
-      if label:
-        rank_sum += rank
+    # Mann-Whitney with AVERAGE ranks for tied scores (a plain sort
+    # would tie-break by label and bias the statistic).
+    order = sorted(range(len(self._scores)),
+                   key=lambda i: self._scores[i])
+    ranks = [0.0] * len(order)
+    i = 0
+    while i < len(order):
+      j = i
+      while j + 1 < len(order) and \
+          self._scores[order[j + 1]] == self._scores[order[i]]:
+        j += 1
+      avg = (i + j) / 2.0 + 1.0
+      for k in range(i, j + 1):
+        ranks[order[k]] = avg
+      i = j + 1
+    rank_sum = sum(r for r, l in zip(ranks, self._labels) if l)
     return (rank_sum - n_pos * (n_pos + 1) / 2) / (n_pos * n_neg)
 
 
