@@ -56,11 +56,16 @@ class BatchNormLayer(BaseLayer):
     if self.p.enable_cross_replica_sum_on_tpu:
       import torch.distributed as dist
       if dist.is_available() and dist.is_initialized():
-        stats = torch.cat([mean * count, var * count, count.reshape(1)])
+        # Combine via sufficient statistics (sum x, sum x^2, count):
+        # averaging per-replica variances alone drops the between-
+        # replica mean spread and understates the true variance.
+        sum_x = mean * count
+        sum_x2 = (var + mean * mean) * count
+        stats = torch.cat([sum_x, sum_x2, count.reshape(1)])
         dist.all_reduce(stats)
         n = stats[-1].clamp_min(1.0)
         mean = stats[:self.p.dim] / n
-        var = stats[self.p.dim:2 * self.p.dim] / n
+        var = stats[self.p.dim:2 * self.p.dim] / n - mean * mean
     return mean, var
 
   def FProp(self, theta: NestedMap, inputs: torch.Tensor,

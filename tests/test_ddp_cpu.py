@@ -230,3 +230,45 @@ def test_two_shot_all_reduce_matches_ring():
       p.join(120)
       assert p.exitcode == 0
     assert results['ok0'] and results['ok1']
+
+
+def _run_sync_bn(rank, world, port, results):
+  os.environ['MASTER_ADDR'] = '127.0.0.1'
+  os.environ['MASTER_PORT'] = str(port)
+  dist.init_process_group('gloo', rank=rank, world_size=world)
+  from lingvo_amd.layers import bn_layers
+  bn = bn_layers.BatchNormLayer.Params().Set(
+      name='bn', dim=4, enable_cross_replica_sum_on_tpu=True,
+      random_seed=3).Instantiate()
+  bn.train()
+  # replicas see shards with DIFFERENT means
+  g = torch.Generator().manual_seed(500 + rank)
+  x = torch.randn(3, 6, 4, generator=g) + 3.0 * rank
+  out = bn.FProp(bn.theta, x, torch.zeros(3, 6))
+  results[f'out{rank}'] = out.detach()
+  results[f'x{rank}'] = x
+  dist.destroy_process_group()
+
+
+def test_sync_batch_norm_matches_global_moments():
+  ctx = mp.get_context('spawn')
+  port = dist_port(29599)
+  with ctx.Manager() as mgr:
+    results = mgr.dict()
+    procs = [ctx.Process(target=_run_sync_bn, args=(r, 2, port, results))
+             for r in range(2)]
+    for p in procs:
+      p.start()
+    for p in procs:
+      p.join(120)
+      assert p.exitcode == 0
+    results = dict(results)
+  # single-process reference over the CONCATENATED batch
+  from lingvo_amd.layers import bn_layers
+  bn = bn_layers.BatchNormLayer.Params().Set(
+      name='bn', dim=4, random_seed=3).Instantiate()
+  bn.train()
+  x_all = torch.cat([results['x0'], results['x1']], dim=0)
+  ref = bn.FProp(bn.theta, x_all, torch.zeros(6, 6))
+  assert torch.allclose(results['out0'], ref[:3], atol=1e-4)
+  assert torch.allclose(results['out1'], ref[3:], atol=1e-4)
