@@ -62,6 +62,16 @@ class QDomain(BaseLayer):
     scale = (self.running_max / qmax).to(x.dtype)
     return _FakeQuantFn.apply(x, scale, p.bits)
 
+  def QuantizeWeight(self, w: torch.Tensor) -> torch.Tensor:
+    """Weights quantize against their OWN max (static per call) — the
+    activation running-max is the wrong scale for parameters."""
+    p = self.p
+    if self._step < p.start_step:
+      return w
+    qmax = 2.0 ** (p.bits - 1) - 1
+    scale = (w.detach().abs().max().clamp_min(1e-6) / qmax).to(w.dtype)
+    return _FakeQuantFn.apply(w, scale, p.bits)
+
 
 class QuantizableLayer(BaseLayer):
   """Layers subclass this and wrap tensors with QWeight/QAct
@@ -85,7 +95,9 @@ class QuantizableLayer(BaseLayer):
     return self.qdomain.QuantizeTensor(x, calibrate=calibrate)
 
   def QWeight(self, w: torch.Tensor) -> torch.Tensor:
-    return self._Q(w, calibrate=False)
+    if self.p.qdomain_default is None:
+      return w
+    return self.qdomain.QuantizeWeight(w)
 
   def QAct(self, name: str, x: torch.Tensor) -> torch.Tensor:
     return self._Q(x, calibrate=True)

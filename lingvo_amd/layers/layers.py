@@ -57,7 +57,7 @@ class ProjectionLayer(BaseLayer):
     p = self.p
     w = theta.w
     if p.qdomain_tpl is not None:
-      w = self.qdomain.QuantizeTensor(w, calibrate=False)
+      w = self.qdomain.QuantizeWeight(w)
     out = py_utils.MatmulBias(inputs, w,
                               theta.b if p.has_bias else None)
     out = activations.GetFn(p.activation)(out)

@@ -156,3 +156,19 @@ def test_starnet_train_and_decode():
   assert len(out.boxes) == 2
   for bx, sc in zip(out.boxes, out.scores):
     assert bx.shape[1] == 7 and bx.shape[0] == sc.shape[0]
+
+
+def test_weight_quant_uses_own_scale():
+  import torch
+  from lingvo_amd.core import quant_utils
+  dom = quant_utils.QDomain.Params().Set(
+      name='d', bits=8, decay=0.0).Instantiate()
+  dom.train()
+  dom.SetStep(1)
+  # calibrate activations at a LARGE range
+  dom.QuantizeTensor(torch.linspace(-100, 100, 11))
+  # a small weight must still quantize finely (own-max scale), not to
+  # the ~0.8-wide bins the activation range would imply
+  w = torch.linspace(-0.1, 0.1, 101)
+  qw = dom.QuantizeWeight(w)
+  assert (qw - w).abs().max() < 0.1 / 127 + 1e-6
