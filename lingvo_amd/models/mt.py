@@ -262,3 +262,88 @@ class TransformerModel(BaseTask):
       ref = ' '.join(str(int(x)) for x in refs[i] if int(x) > 2)
       decode_metrics.corpus_bleu.Update(ref, hyp)
     decode_metrics.num_samples_in_batch.Update(float(hyps.shape[0]))
+
+
+class RnmtEncoder(BaseLayer):
+  """RNMT+ text encoder: embedding + stacked biLSTM layers (reference
+  tasks/mt/encoder.py MTEncoderBiRNN / the WmtEnDeRNMT config,
+  wmt14_en_de.py:141)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('vocab_size', 32000, 'Vocab.')
+    p.Define('model_dim', 1024, 'Output dim (= 2 * per-dir LSTM).')
+    p.Define('num_lstm_layers', 4, 'biLSTM layers.')
+    p.Define('dropout_prob', 0.0, 'Inter-layer dropout.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    from lingvo_amd.layers import rnn_cell
+    from lingvo_amd.layers import rnn_layers
+    self.CreateChild('emb', lingvo_layers.EmbeddingLayer.Params().Set(
+        vocab_size=p.vocab_size, embedding_dim=p.model_dim,
+        scale_sqrt_depth=True))
+    half = p.model_dim // 2
+    layer_ps = []
+    for i in range(p.num_lstm_layers):
+      cell = rnn_cell.LSTMCellSimple.Params().Set(
+          num_input_nodes=p.model_dim, num_output_nodes=half)
+      layer_ps.append(rnn_layers.BidirectionalFRNN.Params().Set(
+          name=f'blstm_{i}', fwd=cell.Copy(), bak=cell.Copy()))
+    self.CreateChildren('rnn', layer_ps)
+
+  def FProp(self, theta: NestedMap, ids: torch.Tensor,
+            paddings: torch.Tensor) -> torch.Tensor:
+    x = self.emb.EmbLookup(theta.emb, ids.long()).to(self.fprop_dtype)
+    for i, layer in enumerate(self.rnn):
+      y = layer.FProp(theta.rnn[i], x, paddings)
+      if self.p.dropout_prob and not self.do_eval:
+        y = py_utils.DeterministicDropout(y, 1.0 - self.p.dropout_prob)
+      x = y + x if y.shape == x.shape else y  # residual from layer 2 on
+    return py_utils.ApplyPadding(paddings, x)
+
+
+class RnmtModel(BaseTask):
+  """RNN-based MT (reference WmtEnDeRNMT, wmt14_en_de.py:141): biLSTM
+  encoder + the attention LSTM decoder family. The decoder reuses the
+  LAS-style attention LSTM (tasks/asr/decoder.py:48 and the RNMT
+  decoder share that shape)."""
+
+  @classmethod
+  def Params(cls):
+    from lingvo_amd.models import asr as asr_model
+    p = super().Params()
+    p.Define('encoder', RnmtEncoder.Params(), 'Encoder.')
+    p.Define('decoder', asr_model.AsrDecoder.Params(), 'Decoder.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    self.CreateChild('encoder', self.p.encoder)
+    self.CreateChild('decoder', self.p.decoder)
+
+  def ComputePredictions(self, theta, input_batch):
+    enc = self.encoder.FProp(theta.encoder, input_batch.src.ids,
+                             input_batch.src.paddings)
+    preds = self.decoder.ComputePredictions(
+        theta.decoder, enc, input_batch.src.paddings, input_batch.tgt)
+    return preds
+
+  def ComputeLoss(self, theta, predictions, input_batch):
+    metrics, per_example = self.decoder.ComputeLoss(
+        theta.decoder, predictions, input_batch.tgt)
+    b = input_batch.src.ids.shape[0]
+    metrics.num_samples_in_batch = (torch.tensor(float(b)),
+                                    torch.ones(()))
+    return metrics, per_example
+
+  def Decode(self, input_batch) -> NestedMap:
+    with torch.no_grad():
+      enc = self.encoder.FProp(self.theta.encoder, input_batch.src.ids,
+                               input_batch.src.paddings)
+      hyps = self.decoder.GreedyDecode(self.theta.decoder, enc,
+                                       input_batch.src.paddings)
+    return NestedMap(topk_decoded=hyps, target_ids=input_batch.tgt.ids)

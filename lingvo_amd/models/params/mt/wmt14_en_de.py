@@ -74,3 +74,24 @@ class WmtEnDeTransformerSmall(WmtEnDeTransformerBase):
   FF = 1024
   HEADS = 4
   LAYERS = 2
+
+
+@registry.RegisterSingleTaskModel
+class WmtEnDeRNMT(WmtEnDeTransformerBase):
+  """RNMT+ (reference wmt14_en_de.py:141 WmtEnDeRNMT): biLSTM encoder
+  + attention LSTM decoder."""
+
+  def Task(self):
+    p = mt_model.RnmtModel.Params().Set(name='wmt14_en_de_rnmt')
+    p.fprop_dtype = torch.bfloat16
+    p.train.bf16_weights = True
+    p.encoder.Set(vocab_size=self.VOCAB, model_dim=1024,
+                  num_lstm_layers=4, dropout_prob=0.2)
+    p.decoder.Set(vocab_size=self.VOCAB, emb_dim=1024,
+                  rnn_cell_dim=1024, num_lstm_layers=2,
+                  source_dim=1024, dropout_prob=0.2)
+    p.train.learner = learner_lib.Learner.Params().Set(
+        learning_rate=1e-4,
+        optimizer=optimizer_lib.Adam.Params().Set(beta2=0.999),
+        clip_gradient_norm_to_value=5.0)
+    return p

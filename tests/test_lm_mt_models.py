@@ -166,3 +166,22 @@ def test_lm_kv_cache_generate_matches_full_greedy():
   ref = speculative.GreedyReference(lm, lm.theta, prefix, 10)
   n = min(fast.shape[1], ref.shape[1])
   assert torch.equal(fast[:, :n], ref[:, :n])
+
+
+def test_rnmt_model_train_and_decode():
+  import torch
+  from lingvo_amd.core import registry
+  model_p = registry.GetParams('mt.wmt14_en_de.WmtEnDeRNMT', 'Train')
+  model_p.task.fprop_dtype = torch.float32
+  model_p.task.train.bf16_weights = False
+  model_p.task.random_seed = 8
+  model_p.task.encoder.Set(model_dim=32, num_lstm_layers=2,
+                           vocab_size=64, dropout_prob=0.0)
+  model_p.task.decoder.Set(vocab_size=64, emb_dim=16, rnn_cell_dim=32,
+                           source_dim=32, dropout_prob=0.0)
+  model_p.input.Set(batch_size=2, src_len=10, tgt_len=8, vocab_size=64)
+  task = model_p.Instantiate().GetTask()
+  m = task.TrainStep(task.GetInputBatch())
+  assert torch.isfinite(m['loss'][0])
+  out = task.Decode(task.GetInputBatch())
+  assert out.topk_decoded.shape[0] == 2
