@@ -221,9 +221,16 @@ class AsrDecoder(BaseLayer):
   def ComputePredictions(self, theta: NestedMap, enc: torch.Tensor,
                          enc_paddings: torch.Tensor,
                          targets: NestedMap) -> NestedMap:
-    p = self.p
-    if p.num_lstm_layers == 2:
+    if self.p.num_lstm_layers == 2:
       return self._FastPredictions(theta, enc, enc_paddings, targets)
+    return self._LoopPredictions(theta, enc, enc_paddings, targets)
+
+  def _LoopPredictions(self, theta: NestedMap, enc: torch.Tensor,
+                       enc_paddings: torch.Tensor,
+                       targets: NestedMap) -> NestedMap:
+    """Generic per-cell teacher-forced loop (any layer count); the
+    numerics oracle for _FastPredictions."""
+    p = self.p
     b, l = targets.ids.shape
     emb_all = self.emb.EmbLookup(theta.emb, targets.ids.long()).to(
         self.fprop_dtype)

@@ -243,3 +243,24 @@ def test_specaugment_time_warp():
   with py_utils.StepSeedScope(7, 1):
     out2 = aug.FProp(aug.theta, x, pad)
   assert torch.equal(out, out2)
+
+
+def test_decoder_fast_path_matches_loop_path():
+  """_FastPredictions (fused GEMM loop) == generic per-cell loop."""
+  import torch
+  from lingvo_amd.models import asr as asr_lib
+  from lingvo_amd.core.nested_map import NestedMap
+  dec = asr_lib.AsrDecoder.Params().Set(
+      name='d', vocab_size=24, emb_dim=12, rnn_cell_dim=16,
+      source_dim=20, num_lstm_layers=2, dropout_prob=0.0,
+      random_seed=5).Instantiate()
+  dec.eval()
+  g = torch.Generator().manual_seed(1)
+  enc = torch.randn(2, 7, 20, generator=g)
+  pad = torch.zeros(2, 7)
+  pad[1, 5:] = 1.0
+  tgt = NestedMap(ids=torch.randint(3, 24, (2, 6), generator=g),
+                  paddings=torch.zeros(2, 6))
+  fast = dec._FastPredictions(dec.theta, enc, pad, tgt)
+  slow = dec._LoopPredictions(dec.theta, enc, pad, tgt)
+  assert (fast.atten_vecs - slow.atten_vecs).abs().max() < 1e-4
