@@ -145,6 +145,10 @@ class ConformerLayer(BaseLayer):
     p.Define('dropout_prob', 0.0, 'Dropout throughout.')
     p.Define('remat', False, 'Gradient-checkpoint this layer '
              '(reference conformer_layer.py:548 p.remat).')
+    p.Define('moe_num_experts', 0,
+             'If > 0, the ENDING half-FFN becomes a top-2 MoE FFN '
+             '(reference conformer_layer.py:1006 MoE-FFN option).')
+    p.Define('moe_capacity_factor', 2.0, 'MoE capacity factor.')
     return p
 
   @classmethod
@@ -165,7 +169,16 @@ class ConformerLayer(BaseLayer):
         residual_weight=0.5, residual_dropout_prob=p.dropout_prob,
         relu_dropout_prob=p.dropout_prob)
     self.CreateChild('fflayer_start', ff.Copy())
-    self.CreateChild('fflayer_end', ff.Copy())
+    if p.moe_num_experts:
+      self.CreateChild(
+          'fflayer_end',
+          transformer_lib.MoETransformerFeedForwardLayer.Params().Set(
+              input_dim=d, hidden_dim=hidden,
+              num_experts=p.moe_num_experts,
+              expert_capacity_factor=p.moe_capacity_factor,
+              residual_dropout_prob=p.dropout_prob))
+    else:
+      self.CreateChild('fflayer_end', ff.Copy())
     atten = transformer_lib.TransformerAttentionLayer.Params().Set(
         input_dim=d, num_heads=p.atten_num_heads, is_masked=p.is_causal,
         residual_dropout_prob=p.dropout_prob)

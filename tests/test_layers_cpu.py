@@ -587,3 +587,20 @@ def test_lora_adaptation_and_merge():
   assert not any('lora' in n for n, _ in lm.named_parameters())
   assert torch.allclose(lm.FProp(lm.theta, ids, pads).detach(),
                         tuned_out, atol=1e-4)
+
+
+def test_conformer_moe_ffn_option():
+  from lingvo_amd.layers import conformer as conformer_lib
+  p = conformer_lib.ConformerLayer.Params().Set(
+      name='c', input_dim=16, atten_num_heads=2, kernel_size=4,
+      conv_norm='layer', moe_num_experts=4, random_seed=3,
+      dropout_prob=0.0)
+  layer = p.Instantiate()
+  layer.eval()
+  x = torch.randn(2, 10, 16)
+  pad = torch.zeros(2, 10)
+  out = layer.FProp(layer.theta, x, pad)
+  assert out.shape == x.shape
+  out.sum().backward()
+  assert layer.fflayer_end.moe.wi.grad is not None
+  assert float(layer.fflayer_end.AuxLoss()) >= 0
