@@ -485,3 +485,29 @@ def EstimateFlops(fn, *args) -> int:
                with_flops=True) as prof:
     fn(*args)
   return int(sum(e.flops for e in prof.key_averages() if e.flops))
+
+
+def RandomPermutationSequence(num: int, batch: int,
+                              op_seed: Optional[int] = None
+                              ) -> torch.Tensor:
+  """Deterministic batch of random permutations of range(num)
+  (reference x_ops random permutation sequence op): [batch, num] longs,
+  reproducible under StepSeedScope."""
+  keys = GraphSafeUniform((batch, num), 'cpu', op_seed)
+  return keys.argsort(dim=1)
+
+
+class CachedCall:
+  """Caches a zero-arg callable's result (reference CachedCall op):
+  the fn runs once; later calls return the cached tensor/value."""
+
+  def __init__(self, fn):
+    self._fn = fn
+    self._has = False
+    self._val = None
+
+  def __call__(self):
+    if not self._has:
+      self._val = self._fn()
+      self._has = True
+    return self._val

@@ -547,3 +547,24 @@ def test_shape_asserts_and_flop_estimate():
   flops = pu.EstimateFlops(lambda: torch.randn(32, 64) @
                            torch.randn(64, 16))
   assert abs(flops - 2 * 32 * 64 * 16) / (2 * 32 * 64 * 16) < 0.2
+
+
+def test_random_permutation_and_cached_call():
+  import torch
+  from lingvo_amd.core import py_utils as pu
+  with pu.StepSeedScope(3, 1):
+    perms = pu.RandomPermutationSequence(8, 4)
+  assert perms.shape == (4, 8)
+  for row in perms:
+    assert sorted(row.tolist()) == list(range(8))
+  with pu.StepSeedScope(3, 1):
+    again = pu.RandomPermutationSequence(8, 4)
+  assert torch.equal(perms, again)
+
+  calls = {'n': 0}
+  def fn():
+    calls['n'] += 1
+    return torch.ones(3)
+  cc = pu.CachedCall(fn)
+  a, b = cc(), cc()
+  assert calls['n'] == 1 and torch.equal(a, b)
