@@ -305,6 +305,8 @@ class UlyssesMultiHeadedAttention(attention_lib.MultiHeadedAttention):
         p.left_context, 0 if p.causal else p.right_context,
         p.rel_pos_clip)
     out = _GatherHeadsScatterSeq(out, world, p.cp_group)
+    if p.atten_dropout_prob and not self.do_eval:
+      out = py_utils.DeterministicDropout(out, 1.0 - p.atten_dropout_prob)
     b, t = out.shape[0], out.shape[1]
     ctx = out.reshape(b, t, self._n * self._h)
     post = py_utils.MatmulBias(ctx, theta.post_w,

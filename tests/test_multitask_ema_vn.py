@@ -106,3 +106,25 @@ def test_variational_noise_changes_loss_but_deterministic():
   t3 = m3.GetTask()
   metrics3 = t3.TrainStep(t3.GetInputBatch())
   assert float(metrics1['loss'][0]) == float(metrics3['loss'][0])
+
+
+def test_multitask_program_schedule(tmp_path):
+  from lingvo_amd.runtime.program import MultiTaskProgramSchedule
+  mp_ = MultiTaskModel.Params().Set(name='multi')
+  mp_.task_params = Params()
+  mp_.task_params.Define('a', _task_params(3), '')
+  mp_.task_params.Define('b', _task_params(4), '')
+  mp_.task_probs = Params()
+  mp_.task_probs.Define('a', 0.5, '')
+  mp_.task_probs.Define('b', 0.5, '')
+  model = mp_.Instantiate()
+  sched = MultiTaskProgramSchedule(model, str(tmp_path), 'cpu',
+                                   steps_per_loop=1)
+  seen = set()
+  for _ in range(6):
+    out = sched.Run()
+    seen.add(out.task)
+    assert out.loss == out.loss
+  assert seen <= {'a', 'b'} and seen
+  total = sum(t.global_step for t in model.tasks)
+  assert total == 6
