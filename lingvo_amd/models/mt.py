@@ -347,3 +347,71 @@ class RnmtModel(BaseTask):
       hyps = self.decoder.GreedyDecode(self.theta.decoder, enc,
                                        input_batch.src.paddings)
     return NestedMap(topk_decoded=hyps, target_ids=input_batch.tgt.ids)
+
+
+class NmtTfRecordInput(BaseSequenceInputGenerator):
+  """Real-data MT input: tfrecord Examples with 'src_ids'/'tgt_ids'
+  int64 features (the NmtInput export shape, reference
+  tasks/mt/input_generator.py). C++ yielder + TF-free codec + length
+  bucketing by max(src, tgt) tokens."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.batch_size = 16
+    p.Define('files', [], 'TFRecord shards.')
+    p.Define('max_len', 128, 'Crop length.')
+    p.Define('input_seed', 301, 'Shuffle seed.')
+    p.bucket_upper_bound = [128]
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    from lingvo_amd.core import tf_example
+    from lingvo_amd.core.generic_input import RecordBatcher
+    from lingvo_amd.ops import _loader
+    ext = _loader.get_ext(required=True)
+    self._yielder = ext.RecordYielder(list(p.files), 'tfrecord',
+                                      p.input_seed, 1000, 2, True)
+    limits = list(p.bucket_batch_limit) or \
+        [p.batch_size] * len(p.bucket_upper_bound)
+
+    def proc(rec):
+      ex = tf_example.ParseExample(rec)
+      src = torch.tensor(ex['src_ids'][:p.max_len], dtype=torch.long)
+      tgt = torch.tensor(ex['tgt_ids'][:p.max_len - 1], dtype=torch.long)
+      return NestedMap(src=src, tgt=tgt,
+                       src_len=torch.tensor([src.numel()]),
+                       tgt_len=torch.tensor([tgt.numel()])), \
+          max(src.numel(), tgt.numel() + 1)
+
+    self._batcher = RecordBatcher(self._yielder, proc,
+                                  p.bucket_upper_bound, limits,
+                                  num_threads=2)
+
+  def _InputBatch(self) -> NestedMap:
+    batch = self._batcher.GetNext()
+    assert batch is not None, 'input exhausted'
+    b = batch.src.shape[0]
+    src_pad = py_utils.PaddingsFromLengths(batch.src_len.reshape(-1),
+                                           batch.src.shape[1])
+    lmax = int(batch.tgt_len.max()) + 1
+    ids = torch.full((b, lmax), 2, dtype=torch.long)
+    tgt_pad = torch.ones(b, lmax)
+    for i in range(b):
+      n = int(batch.tgt_len[i])
+      ids[i, 0] = 1
+      ids[i, 1:n + 1] = batch.tgt[i, :n]
+      tgt_pad[i, :n + 1] = 0.0
+    labels = ids.roll(-1, dims=1)
+    labels[:, -1] = 2
+    return NestedMap(
+        src=NestedMap(ids=batch.src, paddings=src_pad),
+        tgt=NestedMap(ids=ids, paddings=tgt_pad,
+                      labels=labels * (1 - tgt_pad).long(),
+                      weights=1.0 - tgt_pad))
+
+  def Stop(self):
+    self._batcher.Stop()
+    self._yielder.stop()

@@ -185,3 +185,27 @@ def test_rnmt_model_train_and_decode():
   assert torch.isfinite(m['loss'][0])
   out = task.Decode(task.GetInputBatch())
   assert out.topk_decoded.shape[0] == 2
+
+
+def test_mt_tfrecord_input_end_to_end(tmp_path):
+  import torch
+  from lingvo_amd.core import tf_example
+  from lingvo_amd.models import mt as mt_lib
+  g = torch.Generator().manual_seed(6)
+  recs = []
+  for i in range(16):
+    s = torch.randint(3, 40, (int(torch.randint(4, 10, (1,), generator=g)),),
+                      generator=g).tolist()
+    t = torch.randint(3, 40, (int(torch.randint(4, 10, (1,), generator=g)),),
+                      generator=g).tolist()
+    recs.append(tf_example.EncodeExample({'src_ids': s, 'tgt_ids': t}))
+  shard = tmp_path / 'mt.tfrecord'
+  tf_example.WriteTfRecord(str(shard), recs)
+  ip = mt_lib.NmtTfRecordInput.Params().Set(
+      name='in', files=[str(shard)], batch_size=4,
+      bucket_upper_bound=[32]).Instantiate()
+  batch = ip.GetPreprocessedInputBatch()
+  assert batch.src.ids.shape[0] == 4
+  assert batch.tgt.ids[:, 0].eq(1).all()
+  assert (batch.tgt.weights.sum(1) > 0).all()
+  ip.Stop()
