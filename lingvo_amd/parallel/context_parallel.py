@@ -50,18 +50,20 @@ class _RingExchange(torch.autograd.Function):
     nxt = _GlobalRank(group, (rank + 1) % world)
     prv = _GlobalRank(group, (rank - 1) % world)
     ctx.group, ctx.nxt, ctx.prv = group, nxt, prv
+    x = x.contiguous()
     buf = torch.empty_like(x)
     reqs = [dist.irecv(buf, prv, group=group),
-            dist.isend(x.contiguous(), nxt, group=group)]
+            dist.isend(x, nxt, group=group)]
     for r in reqs:
       r.wait()
     return buf
 
   @staticmethod
   def backward(ctx, g):
-    buf = torch.empty_like(g)
+    g = g.contiguous()  # recv buffer must be contiguous (empty_like
+    buf = torch.empty_like(g)  # copies the input's strides)
     reqs = [dist.irecv(buf, ctx.nxt, group=ctx.group),
-            dist.isend(g.contiguous(), ctx.prv, group=ctx.group)]
+            dist.isend(g, ctx.prv, group=ctx.group)]
     for r in reqs:
       r.wait()
     return buf, None

@@ -34,14 +34,19 @@ class _AllToAllFn(torch.autograd.Function):
   @staticmethod
   def forward(ctx, x, group):
     ctx.group = group
+    x = x.contiguous()
     out = torch.empty_like(x)
-    dist.all_to_all_single(out, x.contiguous(), group=group)
+    dist.all_to_all_single(out, x, group=group)
     return out
 
   @staticmethod
   def backward(ctx, grad):
+    # contiguous FIRST: empty_like of a permuted/strided grad would
+    # give a non-contiguous recv buffer and scramble the exchange
+    # (same bug class as the Ulysses scatter fix).
+    grad = grad.contiguous()
     gin = torch.empty_like(grad)
-    dist.all_to_all_single(gin, grad.contiguous(), group=ctx.group)
+    dist.all_to_all_single(gin, grad, group=ctx.group)
     return gin, None
 
 
