@@ -26,3 +26,17 @@ def dist_port(base: int) -> int:
   tests apart; the pid offset avoids TIME_WAIT collisions across runs."""
   import os
   return 20000 + (base - 29500) * 131 % 9000 + os.getpid() % 997
+
+
+import pytest
+
+
+@pytest.fixture(autouse=True)
+def _reap_child_processes():
+  """Terminate any leaked spawned children so a failed multi-process
+  test cannot hang the suite at interpreter exit."""
+  yield
+  import multiprocessing
+  for child in multiprocessing.active_children():
+    child.terminate()
+    child.join(5)
