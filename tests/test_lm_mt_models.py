@@ -209,3 +209,35 @@ def test_mt_tfrecord_input_end_to_end(tmp_path):
   assert batch.tgt.ids[:, 0].eq(1).all()
   assert (batch.tgt.weights.sum(1) > 0).all()
   ip.Stop()
+
+
+def test_mt_flat_beam_search_option():
+  import torch
+  from lingvo_amd.core import registry
+  model_p = registry.GetParams('mt.wmt14_en_de.WmtEnDeTransformerSmall',
+                               'Train')
+  model_p.task.fprop_dtype = torch.float32
+  model_p.task.train.bf16_weights = False
+  model_p.task.random_seed = 4
+  model_p.task.encoder.Set(model_dim=32, num_layers=1, num_heads=1,
+                           hidden_dim=64, vocab_size=48)
+  model_p.task.decoder.Set(model_dim=32, num_layers=1, num_heads=1,
+                           hidden_dim=64, vocab_size=48)
+  model_p.task.decoder.beam_search.Set(num_hyps_per_beam=3, max_steps=6)
+  model_p.input.Set(batch_size=2, src_len=8, tgt_len=6, vocab_size=48)
+
+  task = model_p.Instantiate().GetTask()
+  task.eval()
+  batch = task.GetInputBatch()
+  out_ref = task.Decode(batch)
+
+  model_p.task.decoder.use_flat_beam_search = True
+  task2 = model_p.Instantiate().GetTask()
+  task2.eval()
+  out_flat = task2.Decode(batch)
+  # same model weights (same seed): both searches return hyps of the
+  # contract shape; top-1 ids agree (alpha=0 both)
+  assert out_flat.topk_ids.shape[:2] == out_ref.topk_ids.shape[:2]
+  n = min(int(out_ref.topk_lens[0, 0]), int(out_flat.topk_lens[0, 0]))
+  assert torch.equal(out_ref.topk_ids[0, 0, :n],
+                     out_flat.topk_ids[0, 0, :n])
