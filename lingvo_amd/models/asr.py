@@ -438,6 +438,8 @@ class AsrModel(BaseTask):
     p = super().Params()
     p.Define('encoder', ConformerEncoder.Params(), 'Encoder params.')
     p.Define('decoder', AsrDecoder.Params(), 'Decoder params.')
+    p.Define('decode_num_hyps', 1,
+             'Beam width for Decode(); 1 = greedy.')
     return p
 
   def __init__(self, params):
@@ -470,7 +472,14 @@ class AsrModel(BaseTask):
       enc, enc_pad = self.encoder.FProp(
           self.theta.encoder, input_batch.src.src_inputs,
           input_batch.src.paddings)
-      hyps = self.decoder.GreedyDecode(self.theta.decoder, enc, enc_pad)
+      if self.p.decode_num_hyps > 1:
+        beam = self.decoder.BeamSearchDecode(
+            self.theta.decoder, enc, enc_pad,
+            num_hyps=self.p.decode_num_hyps)
+        hyps = beam.topk_ids[:, 0]
+      else:
+        hyps = self.decoder.GreedyDecode(self.theta.decoder, enc,
+                                         enc_pad)
     return NestedMap(topk_decoded=hyps,
                      transcripts=input_batch.tgt.ids)
 
