@@ -21,6 +21,7 @@ _PARAM_MODULES = [
     'lingvo_amd.models.params.lm.synthetic_packed_input',
     'lingvo_amd.models.params.asr.librispeech',
     'lingvo_amd.models.params.mt.wmt14_en_de',
+    'lingvo_amd.models.params.mt.wmtm16_en_de',
     'lingvo_amd.models.params.punctuator.codelab',
     'lingvo_amd.models.params.milan.cxc',
     'lingvo_amd.models.params.car.kitti',

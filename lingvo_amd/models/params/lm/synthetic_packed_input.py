@@ -110,3 +110,20 @@ class DenseLm128B8x8(DenseLm8B):
     # by parallel.tensor_parallel.LowerShardingAnnotations.
     p.lm.weight_split_dims_mapping = [-1, 0]
     return p
+
+
+@registry.RegisterSingleTaskModel
+class DenseLm1T16x16(DenseLm128B8x8):
+  """~1T-param dense LM config (reference synthetic_packed_input.py:330
+  DenseLm1T16x16). Registered for config parity: training it needs
+  multi-node TP x DP (16x16 in the reference's TPU terms); on MI355X
+  the layout is 8-way TP inside a node x DP across nodes with the same
+  sharding annotations."""
+
+  LAYERS = 128
+  DIM = 24576
+
+  def Task(self):
+    p = super().Task()
+    p.lm.name = 'dense_lm_1t'
+    return p
