@@ -11,10 +11,8 @@ north-star model.
 from __future__ import annotations
 
 import math
-from typing import Optional, Tuple
 
 import torch
-import torch.nn.functional as F
 
 from lingvo_amd.core import py_utils
 from lingvo_amd.core.base_input_generator import BaseSequenceInputGenerator

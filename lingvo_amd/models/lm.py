@@ -4,7 +4,6 @@ LanguageModel, layers.py:495 RnnLm / TransformerLm).
 
 from __future__ import annotations
 
-import math
 from typing import Optional
 
 import torch

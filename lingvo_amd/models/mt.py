@@ -5,7 +5,6 @@ TransformerModel, encoder.py:836, decoder.py:2361 batch-major variants).
 
 from __future__ import annotations
 
-from typing import Optional
 
 import torch
 

@@ -12,8 +12,6 @@ in fewer target forward passes.
 
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 
 from lingvo_amd.core.nested_map import NestedMap
