@@ -150,3 +150,33 @@ def GenericInput(processor: Callable, file_pattern: str,
                               file_buffer_size, file_parallelism, repeat)
   return RecordBatcher(yielder, processor, bucket_upper_bound,
                        bucket_batch_limit, num_threads=num_batcher_threads)
+
+
+class SequentialYielder:
+  """Strict-order record yielder (reference
+  sequential_record_yielder.cc): files in the given order, records in
+  file order, one epoch (eval/decode determinism — no shuffle buffer).
+  Matches the native RecordYielder's yield_record interface."""
+
+  def __init__(self, files, file_format: str = 'text'):
+    assert file_format == 'text', 'sequential yielder: text only'
+    self._files = list(files)
+    self._gen = self._Iter()
+
+  def _Iter(self):
+    for src_id, path in enumerate(self._files):
+      with open(path, 'rb') as f:
+        for line in f:
+          yield line.rstrip(b'\n'), src_id
+
+  def yield_record(self):
+    try:
+      return next(self._gen)
+    except StopIteration:
+      raise StopIteration from None
+
+  def current_epoch(self):
+    return 1
+
+  def stop(self):
+    self._gen = iter(())

@@ -211,3 +211,21 @@ def test_tfrecord_examples_through_native_yielder(tmp_path):
     assert ex['text'][0].startswith(b'line')
   assert seen == set(range(20))
   y.stop()
+
+
+def test_sequential_yielder_strict_order(tmp_path):
+  from lingvo_amd.core.generic_input import SequentialYielder
+  for i in range(2):
+    with open(tmp_path / f'f{i}.txt', 'w') as f:
+      for j in range(5):
+        f.write(f'{i}-{j}\n')
+  y = SequentialYielder([str(tmp_path / 'f0.txt'),
+                         str(tmp_path / 'f1.txt')])
+  recs = []
+  for _ in range(10):
+    rec, src = y.yield_record()
+    recs.append((rec.decode(), src))
+  assert recs[0] == ('0-0', 0) and recs[4] == ('0-4', 0)
+  assert recs[5] == ('1-0', 1) and recs[9] == ('1-4', 1)
+  with pytest.raises(StopIteration):
+    y.yield_record()
