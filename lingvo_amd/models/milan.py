@@ -123,3 +123,20 @@ class DualEncoder(BaseTask):
       preds = self.ComputePredictions(self.theta, input_batch)
       sims = preds.image_emb @ preds.text_emb.t()
     return NestedMap(ranks=sims.argsort(-1, descending=True))
+
+  def CreateDecoderMetrics(self) -> NestedMap:
+    from lingvo_amd.core import metrics as metrics_lib
+    return NestedMap(recall_at_1=metrics_lib.AverageMetric(),
+                     recall_at_5=metrics_lib.AverageMetric(),
+                     num_samples_in_batch=metrics_lib.AverageMetric())
+
+  def PostProcessDecodeOut(self, decode_out, decode_metrics) -> None:
+    """In-batch image->text retrieval recall@k (reference milan
+    score/eval utilities)."""
+    ranks = decode_out.ranks
+    b = ranks.shape[0]
+    labels = torch.arange(b, device=ranks.device).unsqueeze(1)
+    pos = (ranks == labels).float().argmax(dim=1)
+    decode_metrics.recall_at_1.Update(float((pos < 1).float().mean()), b)
+    decode_metrics.recall_at_5.Update(float((pos < 5).float().mean()), b)
+    decode_metrics.num_samples_in_batch.Update(float(b))

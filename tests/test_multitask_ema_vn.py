@@ -128,3 +128,18 @@ def test_multitask_program_schedule(tmp_path):
   assert seen <= {'a', 'b'} and seen
   total = sum(t.global_step for t in model.tasks)
   assert total == 6
+
+
+def test_milan_retrieval_metrics():
+  import torch
+  from lingvo_amd.core import registry
+  mp2 = registry.GetParams('milan.cxc.ImageTextDualEncoder', 'Train')
+  mp2.task.random_seed = 4
+  mp2.input.batch_size = 8
+  task = mp2.Instantiate().GetTask()
+  task.eval()
+  out = task.Decode(task.GetInputBatch())
+  dm = task.CreateDecoderMetrics()
+  task.PostProcessDecodeOut(out, dm)
+  assert 0.0 <= dm.recall_at_1.value <= dm.recall_at_5.value <= 1.0
+  assert dm.num_samples_in_batch.value == 8
