@@ -308,3 +308,78 @@ class LanguageModel(BaseTask):
                        avg_xent=xent.avg_xent)
 
     return NestedMap(default=default)
+
+
+class InsertionLm(BaseTask):
+  """Insertion-based LM (reference core/insertion.py consumers /
+  KERMIT): sample a canvas from each sequence, encode it
+  BIDIRECTIONALLY, and train per-slot content predictions for the
+  missing symbols (slot j = insert before canvas position j; an
+  appended learned END position covers slot C)."""
+
+  @classmethod
+  def Params(cls):
+    from lingvo_amd.core import insertion
+    p = super().Params()
+    p.Define('vocab_size', 32000, 'Vocab.')
+    p.Define('model_dim', 256, 'Model dim.')
+    p.Define('num_layers', 4, 'Bidirectional layers.')
+    p.Define('num_heads', 4, 'Heads.')
+    p.Define('insertion_tpl', insertion.SymbolInsertionLayer.Params(),
+             'Canvas sampler.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    self.CreateChild('insertion', p.insertion_tpl)
+    self.CreateChild('emb', lingvo_layers.EmbeddingLayer.Params().Set(
+        vocab_size=p.vocab_size, embedding_dim=p.model_dim,
+        scale_sqrt_depth=True))
+    self.CreateChild('pos_emb',
+                     lingvo_layers.PositionalEmbeddingLayer.Params().Set(
+                         embedding_dim=p.model_dim))
+    self.CreateVariable('end_emb', py_utils.WeightParams(
+        [p.model_dim], py_utils.WeightInit.Gaussian(0.02), p.dtype))
+    self.CreateChild(
+        'stack', transformer_lib.StackedTransformerLayers.Params().Set(
+            model_dim=p.model_dim, num_layers=p.num_layers,
+            num_heads=p.num_heads, mask_self_atten=False))
+    self.CreateChild('softmax',
+                     lingvo_layers.SimpleFullSoftmax.Params().Set(
+                         input_dim=p.model_dim, num_classes=p.vocab_size))
+
+  def ComputePredictions(self, theta, input_batch):
+    p = self.p
+    rollin = self.insertion.FProp(theta.insertion, input_batch.ids,
+                                  input_batch.paddings)
+    canvas, cpad = rollin.canvas, rollin.canvas_paddings
+    b, c = canvas.shape
+    x = self.emb.EmbLookup(theta.emb, canvas.long()).to(self.fprop_dtype)
+    # append the learned END slot position
+    end = theta.end_emb.reshape(1, 1, -1).expand(b, 1, -1).to(x.dtype)
+    x = torch.cat([x, end], dim=1)
+    pad = torch.cat([cpad, torch.zeros(b, 1, device=x.device)], dim=1)
+    pos = self.pos_emb.FProp(theta.pos_emb, c + 1, device=x.device)
+    x = x + pos.unsqueeze(0).to(x.dtype)
+    act = self.stack.FProp(theta.stack, x, pad)
+    return NestedMap(slot_acts=act, rollin=rollin)
+
+  def ComputeLoss(self, theta, predictions, input_batch):
+    tgt = predictions.rollin.target_indices   # [N, 3] (b, slot, symbol)
+    act = predictions.slot_acts
+    if tgt.shape[0] == 0:
+      zero = act.sum() * 0.0
+      w = torch.ones(())
+      return NestedMap(loss=(zero, w),
+                       num_samples_in_batch=(w, w)), NestedMap()
+    gathered = act[tgt[:, 0], tgt[:, 1]]      # [N, D]
+    logits = self.softmax.Logits(theta.softmax, gathered).float()
+    loss = torch.nn.functional.cross_entropy(logits, tgt[:, 2])
+    w = torch.tensor(float(tgt.shape[0]))
+    metrics = NestedMap(
+        loss=(loss, w),
+        num_samples_in_batch=(
+            torch.tensor(float(input_batch.ids.shape[0])),
+            torch.ones(())))
+    return metrics, NestedMap()

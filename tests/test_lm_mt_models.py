@@ -241,3 +241,21 @@ def test_mt_flat_beam_search_option():
   n = min(int(out_ref.topk_lens[0, 0]), int(out_flat.topk_lens[0, 0]))
   assert torch.equal(out_ref.topk_ids[0, 0, :n],
                      out_flat.topk_ids[0, 0, :n])
+
+
+def test_insertion_lm_trains():
+  import torch
+  from lingvo_amd.models import lm as lm_lib
+  from lingvo_amd.core.base_model import SingleTaskModel
+  task_p = lm_lib.InsertionLm.Params().Set(
+      name='ins', vocab_size=32, model_dim=32, num_layers=1,
+      num_heads=1, random_seed=9)
+  input_p = lm_lib.SyntheticLmInput.Params().Set(
+      name='in', batch_size=4, seq_len=10, vocab_size=32)
+  model = SingleTaskModel.Params().Set(name='m', task=task_p,
+                                       input=input_p).Instantiate()
+  task = model.GetTask()
+  losses = [float(task.TrainStep(task.GetInputBatch())['loss'][0])
+            for _ in range(3)]
+  assert all(l == l for l in losses)
+  assert all(l > 0 for l in losses)
