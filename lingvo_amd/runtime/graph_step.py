@@ -108,6 +108,13 @@ class GraphedTrainStep:
       group['lr'] = lr
     self.opt.step()
     task.global_step_var += 1
+    if task.p.train.pruner_hparams is not None:
+      if not hasattr(task, '_pruner'):
+        from lingvo_amd.core.pruning_utils import MagnitudePruner
+        task._pruner = MagnitudePruner(task,
+                                       **task.p.train.pruner_hparams)
+      task._pruner.Prune(int(task.global_step))
+    task.PostTrainingStepUpdate(task.global_step)
     if task.ema is not None:
       task.ema.Update(task.named_parameters())
     return self.metrics
