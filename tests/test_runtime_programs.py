@@ -305,3 +305,42 @@ def test_async_save_completes(tmp_path):
   payload = torch.load(tmp_path / 'ckpt-00000001.pt',
                        weights_only=False)
   assert torch.equal(payload['model']['w'], torch.arange(4.0))
+
+
+def test_evaler_decoder_poll_loop(tmp_path):
+  """Trainer saves -> Evaler/Decoder pick up the checkpoint once and
+  record it in the processed ledger (idempotent)."""
+  from lingvo_amd.core.checkpointer import Checkpointer
+  from lingvo_amd.runtime import runners
+
+  model_p = registry.GetParams('image.mnist.LeNet5', 'Train')
+  model_p.task.random_seed = 5
+  model = model_p.Instantiate()
+  task = model.GetTask()
+  task.TrainStep(task.GetInputBatch())
+  ck = Checkpointer(Checkpointer.Params(),
+                    str(tmp_path / 'train'), model,
+                    [l.EnsureOptimizer(task) for l in task.learners])
+  ck.Save()
+
+  ev = runners.Evaler(model_p, str(tmp_path), dataset='Dev',
+                      run_once=True, max_eval_batches=2, device='cpu')
+  ev.Start()
+  recs = [json.loads(l)
+          for l in open(tmp_path / 'eval_dev' / 'metrics.jsonl')]
+  assert len(recs) == 1 and recs[0]['step'] == 1
+
+  dec = runners.Decoder(model_p, str(tmp_path), dataset='Dev',
+                        run_once=True, max_eval_batches=1, device='cpu')
+  dec.Start()
+  drecs = [json.loads(l)
+           for l in open(tmp_path / 'decoder_dev' / 'metrics.jsonl')]
+  assert len(drecs) == 1
+
+  # re-running against the same checkpoint is a no-op (ledger)
+  ev2 = runners.Evaler(model_p, str(tmp_path), dataset='Dev',
+                       run_once=True, max_eval_batches=2, device='cpu')
+  ev2.Start()
+  recs2 = [json.loads(l)
+           for l in open(tmp_path / 'eval_dev' / 'metrics.jsonl')]
+  assert len(recs2) == 1
