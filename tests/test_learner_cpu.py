@@ -100,3 +100,13 @@ def test_mass_op_masks_and_targets():
   assert torch.equal(out.tgt_weights > 0, masked)
   # padding is never masked
   assert not masked[2, 8:].any()
+
+
+def test_bprop_variable_exclusion():
+  task = _mnist_task(bprop_variable_exclusion='fc')
+  before = {n: q.detach().clone() for n, q in task.named_parameters()}
+  task.TrainStep(task.GetInputBatch())
+  moved = {n: not torch.equal(before[n], q.detach())
+           for n, q in task.named_parameters()}
+  assert not any(v for n, v in moved.items() if 'fc' in n)
+  assert any(v for n, v in moved.items() if 'fc' not in n)
