@@ -229,3 +229,24 @@ def test_sequential_yielder_strict_order(tmp_path):
   assert recs[5] == ('1-0', 1) and recs[9] == ('1-4', 1)
   with pytest.raises(StopIteration):
     y.yield_record()
+
+
+def test_wpm_vocab_builder_roundtrip(tmp_path):
+  """A trained vocab encodes its own corpus with zero <unk>s and
+  decodes back exactly."""
+  import sys
+  sys.path.insert(0, 'tools')
+  from build_wpm_vocab import TrainWpmVocab
+  corpus = ['the cat sat on the mat',
+            'the dog sat on the log',
+            'a cat and a dog'] * 5
+  vocab = TrainWpmVocab(corpus, vocab_size=80)
+  assert vocab[:3] == ['<unk>', '<s>', '</s>']
+  tok = tokenizers.WpmTokenizer.Params().Set(
+      name='w', tokens=vocab).Instantiate()
+  for line in corpus[:3]:
+    ids = tok._TokensToIds(line)
+    assert 0 not in ids, (line, ids)       # no <unk>
+    assert tok._IdsToTokens(ids) == line
+  # frequent words merged to single pieces
+  assert '▁the' in vocab
