@@ -604,3 +604,24 @@ def test_conformer_moe_ffn_option():
   out.sum().backward()
   assert layer.fflayer_end.moe.wi.grad is not None
   assert float(layer.fflayer_end.AuxLoss()) >= 0
+
+
+def test_gru_and_sru_cells():
+  from lingvo_amd.layers import rnn_cell
+  from lingvo_amd.core.nested_map import NestedMap
+  for cls in (rnn_cell.GRUCell, rnn_cell.SRUCell):
+    cell = cls.Params().Set(name='c', num_input_nodes=8,
+                            num_output_nodes=8,
+                            random_seed=3).Instantiate()
+    st = cell.InitState(2, 'cpu', torch.float32)
+    x = torch.randn(2, 8, requires_grad=True)
+    for _ in range(3):
+      st = cell.FProp(cell.theta, st, NestedMap(act=x))
+    assert st.m.shape == (2, 8)
+    st.m.sum().backward()
+    assert x.grad is not None
+    # padded step leaves state unchanged
+    st2 = cell.FProp(cell.theta, st,
+                     NestedMap(act=x.detach(),
+                               padding=torch.ones(2, 1)))
+    assert torch.allclose(st2.m, st.m, atol=1e-6)
