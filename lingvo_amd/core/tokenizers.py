@@ -8,6 +8,8 @@ with EOS suffix, paddings) following the reference contract.
 
 from __future__ import annotations
 
+import re
+
 from typing import Dict, List, Optional, Sequence, Tuple
 
 import torch
@@ -170,12 +172,19 @@ class WpmTokenizer(VocabFileTokenizer):
       i = j
     return pieces
 
+  # Tokenization splits on ASCII whitespace ONLY (space/tab/CR/LF/FF/VT)
+  # — an explicit contract shared with the C++ encoder. (Python's
+  # str.split() also treats \x1c-\x1f and Unicode spaces as separators,
+  # which the byte-level native path must not.)
+  _WS = re.compile(r'[ \t\n\r\f\v]+')
+
   def _TokensToIds(self, text: str) -> List[int]:
     if self._native is not None:
       return list(self._native.encode(text))
     out = []
-    for w in text.split():
-      out.extend(self._EncodeWord(w))
+    for w in self._WS.split(text):
+      if w:
+        out.extend(self._EncodeWord(w))
     return out
 
   def _IdsToTokens(self, ids: Sequence[int]) -> str:

@@ -107,3 +107,48 @@ def test_pack_unpack_roundtrip(lens):
   seg = out.src_segment_ids
   starts = (seg != torch.roll(seg, 1, dims=1)) & (seg > 0)
   assert (pos[starts] == 0).all()
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.dictionaries(
+    _IDENT,
+    st.one_of(
+        st.lists(st.integers(-2**40, 2**40), min_size=1, max_size=5),
+        st.lists(st.floats(-1e4, 1e4, allow_nan=False, width=32),
+                 min_size=1, max_size=5),
+        st.lists(st.binary(min_size=0, max_size=12), min_size=1,
+                 max_size=3)),
+    min_size=1, max_size=5))
+def test_tf_example_codec_fuzz(features):
+  from lingvo_amd.core import tf_example
+  blob = tf_example.EncodeExample(features)
+  back = tf_example.ParseExample(blob)
+  assert set(back) == set(features)
+  for k, vals in features.items():
+    if isinstance(vals[0], float):
+      assert all(abs(a - b) < 1e-3 + abs(b) * 1e-5
+                 for a, b in zip(back[k], vals))
+    else:
+      assert back[k] == vals, k
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.lists(st.text(
+    alphabet=st.characters(blacklist_categories=('Cs',),
+                           max_codepoint=0x2FFF),
+    max_size=20), min_size=1, max_size=4))
+def test_wpm_native_matches_python_fuzz(texts):
+  from lingvo_amd.core import tokenizers
+  from lingvo_amd.ops import _loader
+  ext = _loader.get_ext()
+  if ext is None or not hasattr(ext, 'WpmEncoder'):
+    return
+  vocab = ['<unk>', '<s>', '</s>', '▁a', '▁b', 'a', 'b', 'c', '▁日',
+           '日', '本', '▁', 'ab']
+  tok = tokenizers.WpmTokenizer.Params().Set(
+      name='w', tokens=vocab).Instantiate()
+  py = tokenizers.WpmTokenizer.Params().Set(
+      name='w2', tokens=vocab).Instantiate()
+  py._native = None
+  for t in texts:
+    assert tok._TokensToIds(t) == py._TokensToIds(t), repr(t)
