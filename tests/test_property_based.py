@@ -152,3 +152,37 @@ def test_wpm_native_matches_python_fuzz(texts):
   py._native = None
   for t in texts:
     assert tok._TokensToIds(t) == py._TokensToIds(t), repr(t)
+
+
+@settings(max_examples=15, deadline=None)
+@given(st.integers(0, 10**6), st.integers(2, 5), st.integers(4, 8))
+def test_flat_vs_batched_beam_fuzz(seed, k, vocab):
+  """Flat and batched helpers agree on the top hypothesis for random
+  score tables (alpha=0)."""
+  from lingvo_amd.core import beam_search_helper as bsh
+  from lingvo_amd.core import flat_beam_search_helper as fbsh
+  g = torch.Generator().manual_seed(seed)
+  table = torch.log_softmax(torch.randn(vocab, vocab, generator=g) * 2,
+                            dim=-1)
+
+  def init_fn(b, kk):
+    return NestedMap(d=torch.zeros(b * kk))
+
+  def step_fn(state, prev):
+    return table[prev], state
+
+  def reorder_fn(state, gather):
+    state.d = state.d[gather]
+    return state
+
+  ref = bsh.BeamSearchHelper(bsh.BeamSearchHelper.Params().Set(
+      num_hyps_per_beam=k, max_steps=6))
+  flat = fbsh.FlatBeamSearchHelper(fbsh.FlatBeamSearchHelper.Params().Set(
+      num_hyps_per_beam=k, max_steps=6, length_norm_alpha=0.0))
+  o1 = ref.BeamSearchDecode(1, init_fn, step_fn, reorder_fn)
+  o2 = flat.BeamSearchDecode(1, init_fn, step_fn, reorder_fn)
+  n1, n2 = int(o1.topk_lens[0, 0]), int(o2.topk_lens[0, 0])
+  assert o1.topk_ids[0, 0, :n1].tolist() == \
+      o2.topk_ids[0, 0, :n2].tolist(), seed
+  assert abs(float(o1.topk_scores[0, 0]) -
+             float(o2.topk_scores[0, 0])) < 1e-4
