@@ -186,3 +186,38 @@ def test_flat_vs_batched_beam_fuzz(seed, k, vocab):
       o2.topk_ids[0, 0, :n2].tolist(), seed
   assert abs(float(o1.topk_scores[0, 0]) -
              float(o2.topk_scores[0, 0])) < 1e-4
+
+
+@settings(max_examples=10, deadline=None)
+@given(st.integers(0, 10**6), st.sampled_from([2, 3, 4, 6]),
+       st.integers(1, 3))
+def test_streaming_conformer_fuzz(seed, chunk, layers):
+  """Chunked StreamStep == full FProp across random chunkings/depths."""
+  from lingvo_amd.layers import conformer as conformer_lib
+  T = 12
+  p = conformer_lib.ConformerLayer.Params().Set(
+      name='c', input_dim=16, atten_num_heads=2, kernel_size=4,
+      is_causal=True, conv_norm='layer', atten_left_context=T,
+      random_seed=seed % 1000 + 1)
+  lays = [p.Copy().Set(name=f'c{i}').Instantiate()
+          for i in range(layers)]
+  for l in lays:
+    l.eval()
+  g = torch.Generator().manual_seed(seed)
+  x = torch.randn(2, T, 16, generator=g)
+  pad = torch.zeros(2, T)
+  full = x
+  for l in lays:
+    full = l.FProp(l.theta, full, pad)
+  states = [l.InitStreamState(l.theta, 2, T, 'cpu', torch.float32)
+            for l in lays]
+  outs = []
+  for c0 in range(0, T, chunk):
+    h = x[:, c0:c0 + chunk]
+    pc = pad[:, c0:c0 + chunk]
+    for i, l in enumerate(lays):
+      h, states[i] = l.StreamStep(l.theta, h, pc, states[i])
+    outs.append(h)
+  stream = torch.cat(outs, dim=1)
+  assert (full - stream).abs().max() < 2e-3, \
+      float((full - stream).abs().max())
