@@ -221,3 +221,38 @@ def test_streaming_conformer_fuzz(seed, chunk, layers):
   stream = torch.cat(outs, dim=1)
   assert (full - stream).abs().max() < 2e-3, \
       float((full - stream).abs().max())
+
+
+@settings(max_examples=8, deadline=None)
+@given(st.integers(0, 10**6),
+       st.lists(st.integers(2, 5), min_size=2, max_size=3))
+def test_packed_lm_fuzz(seed, seg_lens):
+  """Packed rows reproduce per-sequence outputs for random packings."""
+  from lingvo_amd.models import lm as lm_lib
+  lm = lm_lib.TransformerLm.Params().Set(
+      name='lm', vocab_size=32, model_dim=16, num_layers=1, num_heads=1,
+      hidden_dim=32, dropout_prob=0.0,
+      random_seed=seed % 997 + 1).Instantiate()
+  lm.eval()
+  g = torch.Generator().manual_seed(seed)
+  seqs = [torch.randint(3, 32, (1, L), generator=g) for L in seg_lens]
+  # unpacked reference: each sequence in its own padded row
+  tmax = max(seg_lens)
+  ids = torch.zeros(len(seqs), tmax, dtype=torch.long)
+  pad = torch.ones(len(seqs), tmax)
+  for i, s in enumerate(seqs):
+    ids[i, :s.shape[1]] = s
+    pad[i, :s.shape[1]] = 0.0
+  ref = lm.FProp(lm.theta, ids, pad)
+  # packed single row
+  packed = torch.cat(seqs, dim=1)
+  seg = torch.cat([torch.full((1, L), i + 1)
+                   for i, L in enumerate(seg_lens)], dim=1)
+  pos = torch.cat([torch.arange(L).unsqueeze(0) for L in seg_lens],
+                  dim=1)
+  out = lm.FProp(lm.theta, packed, torch.zeros_like(seg, dtype=torch.float),
+                 segment_ids=seg, segment_pos=pos)
+  off = 0
+  for i, L in enumerate(seg_lens):
+    assert (out[0, off:off + L] - ref[i, :L]).abs().max() < 1e-4, (i, L)
+    off += L
