@@ -256,3 +256,27 @@ def test_packed_lm_fuzz(seed, seg_lens):
   for i, L in enumerate(seg_lens):
     assert (out[0, off:off + L] - ref[i, :L]).abs().max() < 1e-4, (i, L)
     off += L
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.lists(st.integers(0, 4), max_size=8),
+       st.lists(st.integers(0, 4), max_size=8))
+def test_wer_matches_dp_oracle(ref_toks, hyp_toks):
+  from lingvo_amd.core import metrics
+  ref = ' '.join(map(str, ref_toks))
+  hyp = ' '.join(map(str, hyp_toks))
+  w = metrics.WerMetric()
+  w.Update(ref, hyp)
+  # DP edit distance oracle
+  r, h = ref.split(), hyp.split()
+  dp = [[0] * (len(h) + 1) for _ in range(len(r) + 1)]
+  for i in range(len(r) + 1):
+    dp[i][0] = i
+  for j in range(len(h) + 1):
+    dp[0][j] = j
+  for i in range(1, len(r) + 1):
+    for j in range(1, len(h) + 1):
+      dp[i][j] = min(dp[i - 1][j] + 1, dp[i][j - 1] + 1,
+                     dp[i - 1][j - 1] + (r[i - 1] != h[j - 1]))
+  want = dp[len(r)][len(h)] / len(r) if r else 0.0
+  assert abs(w.value - want) < 1e-6, (ref, hyp, w.value, want)
