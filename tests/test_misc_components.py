@@ -568,3 +568,26 @@ def test_random_permutation_and_cached_call():
   cc = pu.CachedCall(fn)
   a, b = cc(), cc()
   assert calls['n'] == 1 and torch.equal(a, b)
+
+
+def test_numeric_utils():
+  import torch
+  from lingvo_amd.core import py_utils as pu
+  from lingvo_amd.core import metrics as metrics_lib
+  from lingvo_amd.core.nested_map import NestedMap
+  # CheckNumerics raises on NaN/Inf, passes clean tensors through
+  x = torch.randn(4)
+  assert torch.equal(pu.CheckNumerics(x, 'ok'), x)
+  with pytest.raises(Exception):
+    pu.CheckNumerics(torch.tensor([1.0, float('nan')]), 'bad')
+  # GlobalGradNorm == norm of concatenated grads
+  grads = [torch.randn(3, 4), torch.randn(7)]
+  want = torch.cat([g.reshape(-1) for g in grads]).norm()
+  assert abs(float(pu.GlobalGradNorm(grads)) - float(want)) < 1e-5
+  # PackMetrics/UnpackMetrics round trip
+  m = NestedMap(loss=(torch.tensor(2.0), torch.tensor(3.0)),
+                acc=(torch.tensor(0.5), torch.tensor(4.0)))
+  packed = metrics_lib.PackMetrics(m)
+  back = metrics_lib.UnpackMetrics(sorted(m.keys()), packed)
+  assert abs(float(back['loss'][0]) - 2.0) < 1e-6
+  assert abs(float(back['acc'][1]) - 4.0) < 1e-6
