@@ -344,3 +344,22 @@ def test_evaler_decoder_poll_loop(tmp_path):
   recs2 = [json.loads(l)
            for l in open(tmp_path / 'eval_dev' / 'metrics.jsonl')]
   assert len(recs2) == 1
+
+
+def test_decode_program_in_schedule(tmp_path):
+  from lingvo_amd.runtime import program as program_lib
+  model_p = registry.GetParams('image.mnist.LeNet5', 'Train')
+  model_p.task.random_seed = 3
+  sched = program_lib.SimpleProgramSchedule.Params()
+  sched.train_program.steps_per_loop = 1
+  dp = program_lib.DecodeProgram.Params().Set(name='decode_dev',
+                                              steps_per_loop=2)
+  dp.Define('cls', program_lib.DecodeProgram, 'class')
+  sched.eval_programs = [dp]
+  ex = program_lib.Executor(model_p, str(tmp_path), sched, device='cpu',
+                            max_steps=1)
+  ex.Start()
+  recs = [json.loads(l)
+          for l in open(tmp_path / 'decode_dev' / 'metrics.jsonl')]
+  assert recs and 'accuracy' in recs[0] or recs[0].keys()
+  assert recs[0]['step'] == 1
