@@ -280,3 +280,48 @@ def test_wer_matches_dp_oracle(ref_toks, hyp_toks):
                      dp[i - 1][j - 1] + (r[i - 1] != h[j - 1]))
   want = dp[len(r)][len(h)] / len(r) if r else 0.0
   assert abs(w.value - want) < 1e-6, (ref, hyp, w.value, want)
+
+
+@settings(max_examples=10, deadline=None)
+@given(st.integers(0, 10**6), st.sampled_from([(2, 1), (4, 2), (4, 4)]))
+def test_extend_step_fuzz(seed, heads):
+  """KV-cache decode == full causal FProp across GQA configs."""
+  from lingvo_amd.layers import attention as attention_lib
+  n, nkv = heads
+  layer = attention_lib.MultiHeadedAttention.Params().Set(
+      name='m', input_dim=32, hidden_dim=32, num_heads=n,
+      num_kv_heads=nkv, causal=True,
+      random_seed=seed % 997 + 1).Instantiate()
+  layer.eval()
+  g = torch.Generator().manual_seed(seed)
+  x = torch.randn(2, 7, 32, generator=g)
+  full = layer.FProp(layer.theta, x)
+  st = layer.InitStates(layer.theta, 2, 7, 'cpu', torch.float32)
+  outs = []
+  for t in range(7):
+    o, st = layer.ExtendStep(layer.theta, x[:, t:t + 1], st)
+    outs.append(o)
+  assert (full - torch.cat(outs, 1)).abs().max() < 2e-3
+
+
+@settings(max_examples=20, deadline=None)
+@given(st.integers(0, 10**6), st.integers(8, 40), st.sampled_from([2, 4]))
+def test_top2_gating_invariants(seed, n, e):
+  from lingvo_amd.parallel import moe
+  g = torch.Generator().manual_seed(seed)
+  logits = torch.randn(n, e, generator=g)
+  cap = max(2, n // e)
+  out = moe.Top2Gating(logits, cap)
+  # gates non-negative; kept pairs renormalize to ~1
+  both = out.keep1 & out.keep2
+  pair = (out.g1 + out.g2)[both]
+  assert bool((pair - 1.0).abs().max() < 1e-4) if both.any() else True
+  assert (out.g1 >= 0).all() and (out.g2 >= 0).all()
+  # capacity respected: kept slots unique per expert and < cap
+  for ee in range(e):
+    slots = torch.cat([out.pos1[(out.top1 == ee) & out.keep1],
+                       out.pos2[(out.top2 == ee) & out.keep2]])
+    assert (slots < cap).all()
+    assert slots.unique().numel() == slots.numel(), (ee, slots)
+  # aux loss finite and >= lower bound 1.0 at perfect balance
+  assert torch.isfinite(out.aux_loss)
