@@ -28,6 +28,8 @@ def main():
   ap.add_argument('--warmup', type=int, default=5)
   ap.add_argument('--batch', type=int, default=128,
                   help='Per-GPU batch size.')
+  ap.add_argument('--memory', action='store_true',
+                  help='Report peak device memory in the JSON config.')
   ap.add_argument('--no-graph', action='store_true',
                   help='Disable hipGraph step capture (eager steps).')
   ap.add_argument('--model', default='asr.librispeech.'
@@ -111,6 +113,11 @@ def main():
     dist.all_reduce(e, op=dist.ReduceOp.MAX)
     elapsed = float(e.item())
 
+  peak_mem_gb = None
+  if args.memory and has_gpu:
+    peak_mem_gb = round(
+        torch.cuda.max_memory_allocated(device) / 2**30, 3)
+
   ms_per_step = elapsed / args.steps * 1000.0
   global_batch = args.batch * world
   examples_per_sec = global_batch * args.steps / elapsed
@@ -141,6 +148,8 @@ def main():
                         else model_p.input.Get('seq_len')),
             'parallelism': f'dp{world}',
             'final_loss': round(loss, 4),
+            **({'peak_mem_gb': peak_mem_gb}
+               if peak_mem_gb is not None else {}),
         },
     }
     print(json.dumps(out))
