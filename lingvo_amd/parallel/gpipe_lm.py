@@ -98,7 +98,9 @@ class TransformerLmStage(BaseLayer):
 
 
 def RunGPipeLmStep(stage: TransformerLmStage, runner: GPipeRunner,
-                   batches: List[NestedMap]) -> Optional[torch.Tensor]:
+                   batches: List[NestedMap],
+                   schedule: str = 'fill_drain'
+                   ) -> Optional[torch.Tensor]:
   """One pipelined train step over len(batches) microbatches.
 
   Stage 0 feeds microbatches; the last stage computes the xent loss.
@@ -119,4 +121,5 @@ def RunGPipeLmStep(stage: TransformerLmStage, runner: GPipeRunner,
     xent = stage.XentLoss(theta, nmap.act, b.labels, b.weights)
     return xent.avg_xent
 
-  return runner.RunStep(fprop, input_fn=input_fn, loss_fn=loss_fn)
+  return runner.RunStep(fprop, input_fn=input_fn, loss_fn=loss_fn,
+                        schedule=schedule)

@@ -401,3 +401,53 @@ def test_ppdp_grid_2x2_matches_single_process():
     for n in want:
       assert torch.allclose(got[n], want[n], atol=1e-4), \
           (s, n, (got[n] - want[n]).abs().max())
+
+
+def _run_lm_1f1b(rank, world, port, num_micro, schedule, results):
+  os.environ['MASTER_ADDR'] = '127.0.0.1'
+  os.environ['MASTER_PORT'] = str(port)
+  dist.init_process_group('gloo', rank=rank, world_size=world)
+  from lingvo_amd.parallel.gpipe_lm import (RunGPipeLmStep,
+                                            TransformerLmStage)
+  from lingvo_amd.parallel.pipeline import GPipeRunner
+  from lingvo_amd.core.nested_map import NestedMap
+  stage = TransformerLmStage.Params().Set(
+      name='stage', vocab_size=64, model_dim=32, num_layers_total=4,
+      num_heads=1, hidden_dim=64, stage_idx=rank, num_stages=world,
+      random_seed=21).Instantiate()
+  runner = GPipeRunner(rank, world, num_micro)
+  batches = []
+  for m in range(num_micro):
+    g = torch.Generator().manual_seed(900 + m)
+    ids = torch.randint(1, 64, (2, 8), generator=g)
+    batches.append(NestedMap(ids=ids, labels=ids.roll(-1, 1),
+                             paddings=torch.zeros(2, 8),
+                             weights=torch.ones(2, 8)))
+  loss = RunGPipeLmStep(stage, runner, batches, schedule=schedule)
+  results[f'{schedule}_loss{rank}'] = \
+      None if loss is None else float(loss)
+  results[f'{schedule}_g{rank}'] = torch.cat(
+      [p.grad.reshape(-1) for p in stage.parameters()
+       if p.grad is not None])
+  dist.destroy_process_group()
+
+
+def test_gpipe_lm_1f1b_matches_fill_drain():
+  ctx = mp.get_context('spawn')
+  with ctx.Manager() as mgr:
+    results = mgr.dict()
+    for schedule in ('fill_drain', '1f1b'):
+      procs = [ctx.Process(target=_run_lm_1f1b,
+                           args=(r, 2, dist_port(29602), 4, schedule,
+                                 results))
+               for r in range(2)]
+      for p in procs:
+        p.start()
+      for p in procs:
+        p.join(240)
+        assert p.exitcode == 0
+    results = dict(results)
+  assert abs(results['fill_drain_loss1'] - results['1f1b_loss1']) < 1e-6
+  for r in range(2):
+    assert torch.allclose(results[f'fill_drain_g{r}'],
+                          results[f'1f1b_g{r}'], atol=1e-6), r
