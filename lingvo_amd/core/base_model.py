@@ -9,6 +9,8 @@ runner (lingvo_amd/runtime) drives the loop and hipGraph capture.
 
 from __future__ import annotations
 
+import contextlib
+
 from typing import Dict, List, Optional, Tuple
 
 import torch
@@ -49,6 +51,22 @@ class ExponentialMovingAverage:
     for name, prm in module.named_parameters():
       if name in self._shadow:
         prm.copy_(self._shadow[name])
+
+  @contextlib.contextmanager
+  def AsWeights(self, module: torch.nn.Module):
+    """Temporarily swap the EMA shadows into the module (the
+    reference's eval/decode-under-EMA behavior), restoring the live
+    weights afterwards."""
+    saved = {n: p.detach().clone()
+             for n, p in module.named_parameters() if n in self._shadow}
+    self.CopyTo(module)
+    try:
+      yield
+    finally:
+      with torch.no_grad():
+        for n, p in module.named_parameters():
+          if n in saved:
+            p.copy_(saved[n])
 
 
 class BaseTask(BaseLayer):

@@ -250,10 +250,14 @@ class Evaler(_CheckpointPoller):
     task = model.GetTask()
     task.eval()
     agg: List[NestedMap] = []
-    for _ in range(self._max_eval_batches):
-      batch = task.GetInputBatch()
-      batch = task.input_generator.ToDevice(batch, self._device)
-      agg.append(task.EvalStep(batch))
+    import contextlib as _ctx
+    ema_scope = (task.ema.AsWeights(task) if task.ema is not None
+                 else _ctx.nullcontext())  # eval under EMA shadows
+    with ema_scope:
+      for _ in range(self._max_eval_batches):
+        batch = task.GetInputBatch()
+        batch = task.input_generator.ToDevice(batch, self._device)
+        agg.append(task.EvalStep(batch))
     avg = py_utils.WeightedAvgOfMetrics(agg)
     out_dir = os.path.join(self._logdir, f'eval_{self._dataset.lower()}')
     os.makedirs(out_dir, exist_ok=True)

@@ -143,3 +143,17 @@ def test_milan_retrieval_metrics():
   task.PostProcessDecodeOut(out, dm)
   assert 0.0 <= dm.recall_at_1.value <= dm.recall_at_5.value <= 1.0
   assert dm.num_samples_in_batch.value == 8
+
+
+def test_ema_as_weights_context():
+  import torch
+  from lingvo_amd.core.base_model import ExponentialMovingAverage
+  lin = torch.nn.Linear(4, 4, bias=False)
+  ema = ExponentialMovingAverage(0.5)
+  ema.Update(lin.named_parameters())
+  live = lin.weight.detach().clone()
+  with torch.no_grad():
+    lin.weight.add_(1.0)  # diverge live weights from shadows
+  with ema.AsWeights(lin):
+    assert torch.allclose(lin.weight.detach(), live)
+  assert torch.allclose(lin.weight.detach(), live + 1.0)  # restored
