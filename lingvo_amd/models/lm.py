@@ -98,6 +98,9 @@ class TransformerLm(BaseLayer):
     p.Define('dropout_prob', 0.1, 'Dropout.')
     p.Define('shared_emb', True, 'Tie softmax and embedding weights.')
     p.Define('remat', False, 'Checkpoint layers.')
+    p.Define('use_rope', False,
+             'Rotary position embedding in self-attention (replaces '
+             'the additive sinusoidal positions).')
     p.Define('moe_every_n', 0, 'Every n-th layer uses MoE FFN.')
     p.Define('num_experts', 0, 'MoE experts (when moe_every_n > 0).')
     p.Define('expert_capacity_factor', 2.0, 'MoE capacity factor.')
@@ -124,6 +127,8 @@ class TransformerLm(BaseLayer):
         model_dim=p.model_dim, num_layers=p.num_layers,
         num_heads=p.num_heads, hidden_dim=p.hidden_dim,
         mask_self_atten=True, remat=p.remat)
+    if p.use_rope:
+      stack_p.transformer_tpl.tr_atten_tpl.atten_tpl.use_rope = True
     # propagate GShard-style sharding annotations to the stack so
     # LowerShardingAnnotations can rewrite it to explicit TP layers
     if p.weight_split_dims_mapping is not None:
@@ -155,11 +160,13 @@ class TransformerLm(BaseLayer):
     segment_pos [B,T] restarts positions per segment (reference
     PackedBatchMajorLanguageModel, tasks/lm/model.py:408)."""
     x = self._Emb(theta, ids.long()).to(self.fprop_dtype)
-    pos = self.pos_emb.FProp(theta.pos_emb, ids.shape[1], device=ids.device)
-    if segment_pos is not None:
-      x = x + pos[segment_pos.long()].to(x.dtype)
-    else:
-      x = x + pos.unsqueeze(0).to(x.dtype)
+    if not self.p.use_rope:
+      pos = self.pos_emb.FProp(theta.pos_emb, ids.shape[1],
+                               device=ids.device)
+      if segment_pos is not None:
+        x = x + pos[segment_pos.long()].to(x.dtype)
+      else:
+        x = x + pos.unsqueeze(0).to(x.dtype)
     if self.p.dropout_prob and not self.do_eval:
       x = py_utils.DeterministicDropout(x, 1.0 - self.p.dropout_prob)
     return self.stack.FProp(theta.stack, x, paddings,
@@ -187,7 +194,9 @@ class TransformerLm(BaseLayer):
     for t in range(total - 1):
       cur = prefix[:, t] if t < t0 else tok
       x = self._Emb(theta, cur.long()).to(self.fprop_dtype)
-      x = (x + pos[t].to(x.dtype)).unsqueeze(1)          # [B, 1, D]
+      if not self.p.use_rope:
+        x = x + pos[t].to(x.dtype)
+      x = x.unsqueeze(1)                                 # [B, 1, D]
       act, states = self.stack.ExtendStep(theta.stack, x, states)
       if t < t0 - 1:
         continue  # prefill: just populate the cache

@@ -277,3 +277,20 @@ def test_label_smoothing_numerics():
   want_true = 1.0 - 0.1 + 0.1 / V if sm.p.uncertainty == 0.1 else None
   got_true = target[0, 0, 1]
   assert abs(float(got_true) - (1.0 - 0.1)) < 0.1 / V + 1e-6
+
+
+def test_rope_lm_generate_matches_full():
+  import torch
+  from lingvo_amd.models import lm as lm_lib
+  from lingvo_amd.runtime import speculative
+  lm = lm_lib.TransformerLm.Params().Set(
+      name='lm', vocab_size=32, model_dim=16, num_layers=2, num_heads=2,
+      hidden_dim=32, dropout_prob=0.0, use_rope=True,
+      random_seed=11).Instantiate()
+  lm.eval()
+  g = torch.Generator().manual_seed(2)
+  prefix = torch.randint(3, 32, (2, 4), generator=g)
+  fast = lm.Generate(lm.theta, prefix, max_new=8)
+  ref = speculative.GreedyReference(lm, lm.theta, prefix, 8)
+  n = min(fast.shape[1], ref.shape[1])
+  assert torch.equal(fast[:, :n], ref[:, :n])
