@@ -231,3 +231,50 @@ def test_zigzag_ring_attention_exact():
                           ref[:, pos].detach(), atol=1e-4), r
     assert torch.allclose(results[f'dq{r}'], q.grad[:, pos], atol=1e-4)
     assert torch.allclose(results[f'dk{r}'], k.grad[:, pos], atol=1e-4)
+
+
+def _run_ring_gqa(rank, world, port, results):
+  os.environ['MASTER_ADDR'] = '127.0.0.1'
+  os.environ['MASTER_PORT'] = str(port)
+  dist.init_process_group('gloo', rank=rank, world_size=world)
+  g = torch.Generator().manual_seed(321)
+  B, S, N, NKV, H = 2, 12, 4, 2, 8
+  q = torch.randn(B, S, N, H, generator=g)
+  k = torch.randn(B, S, NKV, H, generator=g)
+  v = torch.randn(B, S, NKV, H, generator=g)
+  ql = cp.ShardSequence(q, rank, world).detach().requires_grad_(True)
+  kl = cp.ShardSequence(k, rank, world).detach().requires_grad_(True)
+  vl = cp.ShardSequence(v, rank, world).detach().requires_grad_(True)
+  out = cp.RingAttention(ql, kl, vl, causal=True)
+  out.square().sum().backward()
+  results[f'out{rank}'] = out.detach()
+  results[f'dk{rank}'] = kl.grad.clone()
+  dist.destroy_process_group()
+
+
+def test_ring_attention_gqa():
+  """Ring attention with grouped KV heads (NKV < N) stays exact."""
+  ctx = mp.get_context('spawn')
+  with ctx.Manager() as mgr:
+    results = mgr.dict()
+    procs = [ctx.Process(target=_run_ring_gqa,
+                         args=(r, 2, dist_port(29604), results))
+             for r in range(2)]
+    for p in procs:
+      p.start()
+    for p in procs:
+      p.join(120)
+      assert p.exitcode == 0
+    results = dict(results)
+  g = torch.Generator().manual_seed(321)
+  B, S, N, NKV, H = 2, 12, 4, 2, 8
+  q = torch.randn(B, S, N, H, generator=g).requires_grad_(True)
+  k = torch.randn(B, S, NKV, H, generator=g).requires_grad_(True)
+  v = torch.randn(B, S, NKV, H, generator=g).requires_grad_(True)
+  ref = flash_attn.flash_attention(q, k, v, None, None, -1, 0)
+  ref.square().sum().backward()
+  for r in range(2):
+    sl = slice(r * 6, (r + 1) * 6)
+    assert torch.allclose(results[f'out{r}'], ref[:, sl].detach(),
+                          atol=1e-4)
+    assert torch.allclose(results[f'dk{r}'], k.grad[:, sl], atol=1e-4)
