@@ -30,6 +30,10 @@ def main():
                   help='Per-GPU batch size.')
   ap.add_argument('--memory', action='store_true',
                   help='Report peak device memory in the JSON config.')
+  ap.add_argument('--profile', default=None, metavar='PATH',
+                  help='Write a torch.profiler key_averages table for '
+                       'a few steps to PATH (after the timed region; '
+                       'complements rocprofv3 kernel traces).')
   ap.add_argument('--no-graph', action='store_true',
                   help='Disable hipGraph step capture (eager steps).')
   ap.add_argument('--model', default='asr.librispeech.'
@@ -117,6 +121,21 @@ def main():
   if args.memory and has_gpu:
     peak_mem_gb = round(
         torch.cuda.max_memory_allocated(device) / 2**30, 3)
+
+  if args.profile and rank == 0:
+    from torch.profiler import profile, ProfilerActivity
+    acts = [ProfilerActivity.CPU]
+    if has_gpu:
+      acts.append(ProfilerActivity.CUDA)
+    with profile(activities=acts) as prof:
+      for i in range(2):
+        step(args.steps + i)
+      if has_gpu:
+        torch.cuda.synchronize()
+    with open(args.profile, 'w') as f:
+      f.write(prof.key_averages().table(
+          sort_by='cuda_time_total' if has_gpu else 'cpu_time_total',
+          row_limit=60))
 
   ms_per_step = elapsed / args.steps * 1000.0
   global_batch = args.batch * world
