@@ -25,8 +25,12 @@ class InferenceGraphExporter:
   @staticmethod
   def Export(model_params, export_path: str,
              checkpoint_path: Optional[str] = None,
-             train_dir: Optional[str] = None) -> str:
-    """Builds the bundle; loads weights from checkpoint if given."""
+             train_dir: Optional[str] = None,
+             use_ema: bool = True) -> str:
+    """Builds the bundle; loads weights from checkpoint if given.
+    use_ema: when the checkpoint carries EMA shadows, export those
+    instead of the live weights (the reference's bfloat16_variables /
+    EMA export behavior for serving)."""
     model = model_params.Instantiate()
     if checkpoint_path is None and train_dir is not None:
       checkpoint_path = LatestCheckpoint(train_dir)
@@ -34,6 +38,15 @@ class InferenceGraphExporter:
       payload = torch.load(checkpoint_path, map_location='cpu',
                            weights_only=False)
       model.load_state_dict(payload['model'], strict=False)
+      if use_ema and payload.get('ema'):
+        with torch.no_grad():
+          sd = payload['ema']
+          for name, prm in model.named_parameters():
+            # task params are saved under the task prefix; match suffix
+            for k, v in sd.items():
+              if name.endswith(k) and prm.shape == v.shape:
+                prm.copy_(v)
+                break
     task = model.GetTask()
     subgraphs = sorted(task.Inference().keys()) if hasattr(
         task, 'Inference') else []

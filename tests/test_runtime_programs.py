@@ -363,3 +363,30 @@ def test_decode_program_in_schedule(tmp_path):
           for l in open(tmp_path / 'decode_dev' / 'metrics.jsonl')]
   assert recs and 'accuracy' in recs[0] or recs[0].keys()
   assert recs[0]['step'] == 1
+
+
+def test_export_uses_ema_shadows(tmp_path):
+  from lingvo_amd.core.checkpointer import Checkpointer
+  from lingvo_amd.runtime.inference import InferenceGraphExporter, Predictor
+  model_p = registry.GetParams('image.mnist.LeNet5', 'Train')
+  model_p.task.random_seed = 6
+  model_p.task.train.ema_decay = 0.5
+  model = model_p.Instantiate()
+  task = model.GetTask()
+  for _ in range(3):
+    task.TrainStep(task.GetInputBatch())
+  ck = Checkpointer(Checkpointer.Params(), str(tmp_path / 'train'),
+                    model, [l.EnsureOptimizer(task) for l in task.learners])
+  path = ck.Save()
+  bundle = str(tmp_path / 'inference.pt')
+  InferenceGraphExporter.Export(model_p, bundle, checkpoint_path=path,
+                                use_ema=True)
+  pred = Predictor(bundle, device='cpu')
+  # exported weights equal the EMA shadows, not the live weights
+  shadows = task.ema.StateDict()
+  name, shadow = next(iter(shadows.items()))
+  live = dict(task.named_parameters())[name].detach()
+  exported = dict(pred._model.named_parameters())
+  exp_t = next(v for k, v in exported.items() if k.endswith(name))
+  assert torch.allclose(exp_t.detach(), shadow)
+  assert not torch.allclose(exp_t.detach(), live)
