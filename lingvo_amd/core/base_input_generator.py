@@ -51,6 +51,13 @@ class BaseInputGenerator(BaseLayer):
       yield self.GetPreprocessedInputBatch()
 
   # ---- host->device (infeed replacement) --------------------------------
+  def SplitInputBatch(self, batch: NestedMap,
+                      num_splits: int) -> list:
+    """Tower batch splitting (reference base_input_generator.py:1006
+    SplitInputBatch / input_generator_helper.py ComputeSplits)."""
+    from lingvo_amd.utils import helpers
+    return helpers.SplitNestedMap(batch, num_splits)
+
   def ToDevice(self, batch: NestedMap, device) -> NestedMap:
     """Moves a batch to the device, overlapping the copy on a side stream."""
     if str(device).startswith('cuda') and torch.cuda.is_available():
