@@ -268,6 +268,27 @@ class Evaler(_CheckpointPoller):
     task.train()
 
 
+def _DecodeOutToRecords(out) -> list:
+  """Flattens a Decode() NestedMap into per-example JSON records."""
+  import torch
+  b = None
+  for _, v in out.FlattenItems():
+    if isinstance(v, torch.Tensor) and v.dim() >= 1:
+      b = v.shape[0]
+      break
+  if b is None:
+    return []
+  recs = []
+  for i in range(b):
+    rec = {}
+    for key, v in out.FlattenItems():
+      if isinstance(v, torch.Tensor) and v.dim() >= 1 and \
+          v.shape[0] == b:
+        rec[key] = v[i].tolist()
+    recs.append(rec)
+  return recs
+
+
 class Decoder(_CheckpointPoller):
   """Polls checkpoints, runs Decode + decoder metrics
   (reference runners.py:1105)."""
@@ -282,13 +303,19 @@ class Decoder(_CheckpointPoller):
     task = model.GetTask()
     task.eval()
     dec_metrics = task.CreateDecoderMetrics()
-    for _ in range(self._max_eval_batches):
-      batch = task.GetInputBatch()
-      batch = task.input_generator.ToDevice(batch, self._device)
-      out = task.Decode(batch)
-      task.PostProcessDecodeOut(out, dec_metrics)
     out_dir = os.path.join(self._logdir, f'decoder_{self._dataset.lower()}')
     os.makedirs(out_dir, exist_ok=True)
+    # decode outputs serialized per example (reference decoder_lib.py
+    # decode-output records; jsonl instead of tfrecord)
+    out_path = os.path.join(out_dir, f'decode_out-{step:08d}.jsonl')
+    with open(out_path, 'w') as outf:
+      for _ in range(self._max_eval_batches):
+        batch = task.GetInputBatch()
+        batch = task.input_generator.ToDevice(batch, self._device)
+        out = task.Decode(batch)
+        task.PostProcessDecodeOut(out, dec_metrics)
+        for rec in _DecodeOutToRecords(out):
+          outf.write(json.dumps(rec) + '\n')
     with open(os.path.join(out_dir, 'metrics.jsonl'), 'a') as f:
       f.write(json.dumps(
           {'step': step,

@@ -390,3 +390,24 @@ def test_export_uses_ema_shadows(tmp_path):
   exp_t = next(v for k, v in exported.items() if k.endswith(name))
   assert torch.allclose(exp_t.detach(), shadow)
   assert not torch.allclose(exp_t.detach(), live)
+
+
+def test_decoder_writes_decode_outputs(tmp_path):
+  from lingvo_amd.core.checkpointer import Checkpointer
+  from lingvo_amd.runtime import runners
+  model_p = registry.GetParams('image.mnist.LeNet5', 'Train')
+  model_p.task.random_seed = 5
+  model = model_p.Instantiate()
+  task = model.GetTask()
+  task.TrainStep(task.GetInputBatch())
+  ck = Checkpointer(Checkpointer.Params(), str(tmp_path / 'train'),
+                    model, [l.EnsureOptimizer(task) for l in task.learners])
+  ck.Save()
+  dec = runners.Decoder(model_p, str(tmp_path), dataset='Dev',
+                        run_once=True, max_eval_batches=1, device='cpu')
+  dec.Start()
+  import glob as globlib
+  outs = globlib.glob(str(tmp_path / 'decoder_dev' / 'decode_out-*.jsonl'))
+  assert len(outs) == 1
+  recs = [json.loads(l) for l in open(outs[0])]
+  assert recs and 'correct_top1' in recs[0]
