@@ -316,7 +316,13 @@ class LanguageModel(BaseTask):
       return NestedMap(per_example_xent=xent.per_example_xent,
                        avg_xent=xent.avg_xent)
 
-    return NestedMap(default=default)
+    def generate(prefix, max_new):
+      n = int(max_new.reshape(-1)[0]) if isinstance(
+          max_new, torch.Tensor) else int(max_new)
+      ids = self.lm.Generate(self.theta.lm, prefix.long(), n)
+      return NestedMap(ids=ids)
+
+    return NestedMap(default=default, generate=generate)
 
 
 class InsertionLm(BaseTask):

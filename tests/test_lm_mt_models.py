@@ -294,3 +294,29 @@ def test_rope_lm_generate_matches_full():
   ref = speculative.GreedyReference(lm, lm.theta, prefix, 8)
   n = min(fast.shape[1], ref.shape[1])
   assert torch.equal(fast[:, :n], ref[:, :n])
+
+
+def test_lm_generate_subgraph_through_server(tmp_path):
+  """Text generation end to end through export -> server."""
+  import json as json_lib
+  from lingvo_amd.core import registry
+  from lingvo_amd.runtime.inference import InferenceGraphExporter, Predictor
+  from lingvo_amd.runtime.server import MakeApp
+  from fastapi.testclient import TestClient
+  mp2 = registry.GetParams('lm.one_billion_wds.OneBWdsTransformerLm',
+                           'Train')
+  mp2.task.fprop_dtype = torch.float32
+  mp2.task.train.bf16_weights = False
+  mp2.task.random_seed = 3
+  mp2.task.lm.Set(vocab_size=32, model_dim=16, num_layers=1,
+                  num_heads=1, hidden_dim=32, dropout_prob=0.0)
+  mp2.input.Set(batch_size=2, seq_len=8, vocab_size=32)
+  bundle = str(tmp_path / 'lm.pt')
+  InferenceGraphExporter.Export(mp2, bundle)
+  app = MakeApp(Predictor(bundle, device='cpu'))
+  client = TestClient(app)
+  r = client.post('/predict/generate',
+                  json={'prefix': [[1, 5, 9]], 'max_new': [4]})
+  assert r.status_code == 200, r.text
+  ids = r.json()['ids']
+  assert len(ids[0]) >= 4 and ids[0][:3] == [1, 5, 9]
