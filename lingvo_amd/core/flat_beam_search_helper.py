@@ -8,6 +8,11 @@ python over hypotheses, so the decode loop stays on-device (and can be
 captured in a hipGraph). Contrast with beam_search_helper.py, whose
 n-best update loops over (batch, hyp) on the host.
 
+Policy note: unlike BeamSearchHelper there is no
+valid_eos_max_logit_delta gate — every step's EOS candidates enter the
+n-best pool and compete on normalized score (keeping the update fully
+tensorized).
+
 Same callback contract as BeamSearchHelper:
   state = init_fn(batch, K)
   log_probs [B*K, V], state = step_fn(state, prev_ids [B*K])

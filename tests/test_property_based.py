@@ -175,8 +175,13 @@ def test_flat_vs_batched_beam_fuzz(seed, k, vocab):
     state.d = state.d[gather]
     return state
 
+  # The batched helper additionally gates EOS admission by
+  # valid_eos_max_logit_delta (the reference x_ops rule); the flat
+  # helper admits EOS unconditionally. Disable the gate so the two
+  # policies coincide for the comparison.
   ref = bsh.BeamSearchHelper(bsh.BeamSearchHelper.Params().Set(
-      num_hyps_per_beam=k, max_steps=6))
+      num_hyps_per_beam=k, max_steps=6,
+      valid_eos_max_logit_delta=1e9))
   flat = fbsh.FlatBeamSearchHelper(fbsh.FlatBeamSearchHelper.Params().Set(
       num_hyps_per_beam=k, max_steps=6, length_norm_alpha=0.0))
   o1 = ref.BeamSearchDecode(1, init_fn, step_fn, reorder_fn)
