@@ -88,3 +88,30 @@ class Predictor:
              for k, v in feeds.items()}
     with torch.no_grad():
       return fn(**feeds)
+
+
+def main(argv=None):
+  """Export CLI (the reference's --mode=write_inference_graph):
+    python -m lingvo_amd.runtime.inference --model image.mnist.LeNet5 \
+        --logdir /tmp/run --output /tmp/inference.pt
+  """
+  import argparse
+  ap = argparse.ArgumentParser()
+  ap.add_argument('--model', required=True, help='Registry key.')
+  ap.add_argument('--output', required=True)
+  ap.add_argument('--logdir', default=None,
+                  help='Loads the latest checkpoint from <logdir>/train.')
+  ap.add_argument('--checkpoint', default=None)
+  ap.add_argument('--no-ema', action='store_true')
+  args = ap.parse_args(argv)
+  from lingvo_amd.core import registry
+  model_p = registry.GetParams(args.model, 'Train')
+  train_dir = os.path.join(args.logdir, 'train') if args.logdir else None
+  path = InferenceGraphExporter.Export(
+      model_p, args.output, checkpoint_path=args.checkpoint,
+      train_dir=train_dir, use_ema=not args.no_ema)
+  print(f'exported {args.model} -> {path}')
+
+
+if __name__ == '__main__':
+  main()
