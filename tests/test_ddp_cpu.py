@@ -272,3 +272,35 @@ def test_sync_batch_norm_matches_global_moments():
   ref = bn.FProp(bn.theta, x_all, torch.zeros(6, 6))
   assert torch.allclose(results['out0'], ref[:3], atol=1e-4)
   assert torch.allclose(results['out1'], ref[3:], atol=1e-4)
+
+
+def _run_two_shot_sizes(rank, world, port, results):
+  os.environ['MASTER_ADDR'] = '127.0.0.1'
+  os.environ['MASTER_PORT'] = str(port)
+  dist.init_process_group('gloo', rank=rank, world_size=world)
+  from lingvo_amd.parallel.ddp import TwoShotAllReduce
+  ok = True
+  for n in (1, 3, 64, 257):
+    g = torch.Generator().manual_seed(10 * n + rank)
+    t = torch.randn(n, generator=g)
+    ref = t.clone()
+    dist.all_reduce(ref)
+    out = TwoShotAllReduce(t)
+    ok = ok and torch.allclose(out, ref, atol=1e-6)
+  results[f'ok{rank}'] = ok
+  dist.destroy_process_group()
+
+
+def test_two_shot_all_reduce_size_sweep():
+  ctx = mp.get_context('spawn')
+  port = dist_port(29606)
+  with ctx.Manager() as mgr:
+    results = mgr.dict()
+    procs = [ctx.Process(target=_run_two_shot_sizes,
+                         args=(r, 2, port, results)) for r in range(2)]
+    for p in procs:
+      p.start()
+    for p in procs:
+      p.join(120)
+      assert p.exitcode == 0
+    assert results['ok0'] and results['ok1']
