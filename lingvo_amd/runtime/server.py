@@ -160,9 +160,15 @@ def main(argv: Optional[list] = None) -> None:
   ap.add_argument('--host', default='127.0.0.1')
   ap.add_argument('--port', type=int, default=8000)
   ap.add_argument('--device', default=None)
+  ap.add_argument('--micro-batch', action='store_true',
+                  help='Coalesce concurrent single-example requests.')
+  ap.add_argument('--max-batch', type=int, default=8)
+  ap.add_argument('--max-wait-ms', type=float, default=3.0)
   args = ap.parse_args(argv)
   import uvicorn
-  app = MakeApp(Predictor(args.bundle, device=args.device))
+  app = MakeApp(Predictor(args.bundle, device=args.device),
+                micro_batch=args.micro_batch,
+                max_batch=args.max_batch, max_wait_ms=args.max_wait_ms)
   uvicorn.run(app, host=args.host, port=args.port, log_level='info')
 
 
