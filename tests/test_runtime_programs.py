@@ -411,3 +411,19 @@ def test_decoder_writes_decode_outputs(tmp_path):
   assert len(outs) == 1
   recs = [json.loads(l) for l in open(outs[0])]
   assert recs and 'correct_top1' in recs[0]
+
+
+def test_summarize_metrics_tool(tmp_path):
+  import sys
+  sys.path.insert(0, 'tools')
+  from summarize_metrics import Summarize
+  from lingvo_amd.runtime import program as program_lib
+  mp2 = registry.GetParams('image.mnist.LeNet5', 'Train')
+  mp2.task.random_seed = 3
+  sched = program_lib.SimpleProgramSchedule.Params()
+  sched.train_program.steps_per_loop = 1
+  ex = program_lib.Executor(mp2, str(tmp_path), sched, device='cpu',
+                            max_steps=1)
+  ex.Start()
+  out = Summarize(str(tmp_path))
+  assert 'metrics.jsonl' in out and 'loss=' in out
