@@ -6,6 +6,12 @@ Measures whole-job examples/sec for the Conformer-L ASR train step on
 synthetic Librispeech-shaped data (random-init weights, bf16 compute),
 DP over RCCL/xGMI for N>1. Launched by the driver via
 torch.distributed.run for N>1 (one rank per GPU).
+
+Robustness contract: for a single-rank GPU run the measurement executes
+in a child process per mode (hipGraph first, eager fallback) so a GPU
+memory fault in the captured path degrades to a slower eager number
+instead of an rc=134 abort with no JSON. Phase markers go to stderr so a
+crash is attributable to model-build / warmup / capture / replay-N.
 """
 
 from __future__ import annotations
@@ -13,15 +19,18 @@ from __future__ import annotations
 import argparse
 import json
 import os
+import subprocess
 import sys
 import time
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
-import torch
+
+def _phase(msg):
+  print(f'# phase: {msg}', file=sys.stderr, flush=True)
 
 
-def main():
+def build_argparser():
   ap = argparse.ArgumentParser()
   ap.add_argument('--gpus', type=int, default=1)
   ap.add_argument('--steps', type=int, default=20)
@@ -38,13 +47,68 @@ def main():
                   help='Disable hipGraph step capture (eager steps).')
   ap.add_argument('--model', default='asr.librispeech.'
                   'Librispeech960WpmConformerL')
-  args = ap.parse_args()
+  ap.add_argument('--mode', choices=['auto', 'graph', 'eager'],
+                  default='auto',
+                  help='auto = child-process graph with eager fallback '
+                       '(single rank); graph/eager = run that mode '
+                       'inline in this process.')
+  return ap
 
+
+def main():
+  args = build_argparser().parse_args()
+  world = int(os.environ.get('WORLD_SIZE', '1'))
+  if args.no_graph:
+    args.mode = 'eager'
+
+  if args.mode == 'auto' and world == 1:
+    import torch
+    if not torch.cuda.is_available():
+      run_bench(args, world)
+      return
+    # Orchestrate: run each mode in a child so an asynchronous GPU fault
+    # (SIGABRT, uncatchable in-process) falls back instead of killing the
+    # bench. The child prints the one JSON line; we relay it.
+    base = [sys.executable, os.path.abspath(__file__),
+            '--gpus', str(args.gpus), '--steps', str(args.steps),
+            '--warmup', str(args.warmup), '--batch', str(args.batch),
+            '--model', args.model]
+    if args.memory:
+      base.append('--memory')
+    if args.profile:
+      base += ['--profile', args.profile]
+    for mode in ('graph', 'eager'):
+      _phase(f'launching child mode={mode}')
+      proc = subprocess.run(base + ['--mode', mode],
+                            stdout=subprocess.PIPE, text=True)
+      out = proc.stdout or ''
+      json_line = None
+      for line in out.splitlines():
+        line = line.strip()
+        if line.startswith('{') and '"metric"' in line:
+          json_line = line
+      if json_line is not None:
+        # Relay any non-JSON prefix lines for context, then the result.
+        for line in out.splitlines():
+          if line.strip() != json_line:
+            print(line)
+        print(json_line, flush=True)
+        return
+      _phase(f'child mode={mode} rc={proc.returncode} without a result'
+             f'{"; falling back to eager" if mode == "graph" else ""}')
+    print('# bench: all modes failed', file=sys.stderr, flush=True)
+    sys.exit(1)
+
+  run_bench(args, world)
+
+
+def run_bench(args, world):
+  _phase('importing torch')
+  import torch
   import torch.distributed as dist
   from lingvo_amd.core import registry
   from lingvo_amd.parallel import ddp
 
-  world = int(os.environ.get('WORLD_SIZE', '1'))
   # Keep the NCCL watchdog from probing streams during hipGraph capture
   # (capture falls back to eager if it still objects).
   os.environ.setdefault('TORCH_NCCL_ASYNC_ERROR_HANDLING', '0')
@@ -55,6 +119,9 @@ def main():
   if has_gpu:
     torch.cuda.set_device(device)
 
+  use_graph = args.mode != 'eager' and has_gpu
+
+  _phase('building model')
   model_p = registry.GetParams(args.model, 'Train')
   model_p.input.batch_size = args.batch
   if not has_gpu:  # CPU smoke of the bench harness only
@@ -67,8 +134,22 @@ def main():
   sync = ddp.GradSync(task) if world > 1 else None
   finalize = sync.Finalize if sync else None
 
+  if world > 1:
+    # Per-rank sanity before any timed work: proves the process group is
+    # live on this topology and catches rank bring-up bugs before the
+    # driver's 8-GPU budget is spent.
+    _phase(f'rank sanity: rank={rank}/{world} device={device}')
+    t = torch.ones(1, device=device if has_gpu else 'cpu')
+    dist.all_reduce(t)
+    ok = abs(float(t.item()) - world) < 1e-6
+    print(f'# rank {rank}/{world} device={device} allreduce='
+          f'{float(t.item())} ok={ok}', file=sys.stderr, flush=True)
+    if not ok:
+      raise RuntimeError(f'rank {rank}: allreduce sanity failed')
+
   # Pre-generate a handful of synthetic batches on device. Each DP rank
   # draws a distinct stream (weak scaling: different data per rank).
+  _phase('generating synthetic batches')
   gen = task.input_generator
   gen._batch_count = rank * 1009
   batches = []
@@ -78,8 +159,9 @@ def main():
         lambda t: t.to(device) if isinstance(t, torch.Tensor) else t))
 
   graphed = None
-  if has_gpu and not args.no_graph:
+  if use_graph:
     try:
+      _phase('hipGraph warmup + capture')
       from lingvo_amd.runtime.graph_step import GraphedTrainStep
       graphed = GraphedTrainStep(task, batches[0], grad_sync=sync)
       if rank == 0:
@@ -96,7 +178,11 @@ def main():
                           grad_sync_finalize=finalize)
 
   for i in range(args.warmup):
-    step(i)
+    _phase(f'warmup step {i}')
+    metrics = step(i)
+  if has_gpu:
+    torch.cuda.synchronize()
+  _phase('warmup done')
 
   if world > 1:
     dist.barrier()
@@ -105,11 +191,14 @@ def main():
   t0 = time.perf_counter()
   for i in range(args.steps):
     metrics = step(i)
+    if i % 5 == 0:
+      _phase(f'timed step {i}')
   if has_gpu:
     torch.cuda.synchronize()
   if world > 1:
     dist.barrier()
   elapsed = time.perf_counter() - t0
+  _phase('timed region done')
 
   # MAX step time over ranks == MIN throughput: reduce elapsed as MAX.
   if world > 1:
@@ -140,7 +229,7 @@ def main():
   ms_per_step = elapsed / args.steps * 1000.0
   global_batch = args.batch * world
   examples_per_sec = global_batch * args.steps / elapsed
-  loss = float(metrics['loss'][0])
+  loss = float(metrics['loss'][0].detach())
 
   if rank == 0:
     out = {
@@ -167,11 +256,12 @@ def main():
                         else model_p.input.Get('seq_len')),
             'parallelism': f'dp{world}',
             'final_loss': round(loss, 4),
+            'step_mode': 'hipgraph' if graphed is not None else 'eager',
             **({'peak_mem_gb': peak_mem_gb}
                if peak_mem_gb is not None else {}),
         },
     }
-    print(json.dumps(out))
+    print(json.dumps(out), flush=True)
   if world > 1:
     dist.destroy_process_group()
 
