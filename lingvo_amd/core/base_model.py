@@ -169,7 +169,8 @@ class BaseTask(BaseLayer):
       for i, lrn in enumerate(self.learners):
         ln = lrn.p.loss_name
         lmetrics = lrn.Apply(self, metrics[ln][0] if i else loss, step,
-                             grad_sync_finalize=grad_sync_finalize)
+                             grad_sync_finalize=grad_sync_finalize,
+                             retain_graph=i < len(self.learners) - 1)
         for k, v in lmetrics.items():
           metrics[k if i == 0 else f'{k}_{i}'] = v
     self.global_step_var += 1

@@ -111,6 +111,12 @@ class Saver:
     self._WriteStateFile()
     return path
 
+  def _DoSaveGuarded(self, payload: Dict, step: int) -> None:
+    try:
+      self._DoSave(payload, step)
+    except BaseException as e:  # surfaced by the next Sync()/Save()
+      self._bg_error = e
+
   def Save(self, payload: Dict, step: int) -> str:
     """Saves; if async, snapshots tensors to CPU and writes in background."""
     if self._async:
@@ -121,7 +127,7 @@ class Saver:
           for k, v in payload.items()
       }
       self.Sync()
-      self._thread = threading.Thread(target=self._DoSave,
+      self._thread = threading.Thread(target=self._DoSaveGuarded,
                                       args=(snap, step), daemon=True)
       self._thread.start()
       return CheckpointPath(self._dir, step)
@@ -136,6 +142,10 @@ class Saver:
     if self._thread is not None:
       self._thread.join()
       self._thread = None
+    err = getattr(self, '_bg_error', None)
+    if err is not None:
+      self._bg_error = None
+      raise RuntimeError('async checkpoint save failed') from err
 
 
 class Checkpointer:

@@ -78,8 +78,8 @@ class Learner(BaseLayer):
     return self.p.learning_rate * self.lr_schedule.Value(step)
 
   def Apply(self, task, loss: torch.Tensor, global_step: int,
-            grad_sync_finalize: Optional[Callable[[], None]] = None
-            ) -> NestedMap:
+            grad_sync_finalize: Optional[Callable[[], None]] = None,
+            retain_graph: bool = False) -> NestedMap:
     """Backward + clip + update. Returns eval metrics for this learner.
 
     grad_sync_finalize: callback run after backward, before clipping — the
@@ -99,7 +99,9 @@ class Learner(BaseLayer):
       loss = loss + p.l1_regularizer_weight * reg
 
     opt.zero_grad(set_to_none=True)
-    loss.backward()
+    # retain_graph: with multiple learners over one FProp, every backward
+    # but the last must keep the autograd graph alive.
+    loss.backward(retain_graph=retain_graph)
     if grad_sync_finalize is not None:
       grad_sync_finalize()
 

@@ -152,7 +152,10 @@ class MasterAdamW(torch.optim.Optimizer):
       params = [prm for prm in group['params'] if prm.grad is not None]
       if not params:
         continue
-      grads, masters, ms, vs = [], [], [], []
+      # State is created lazily on first grad, so params whose grads
+      # appear later can have a lower step count: bucket by per-param
+      # step so bias correction is exact (normally one bucket).
+      by_step = {}
       for prm in params:
         st = self.state[prm]
         if not st:
@@ -162,26 +165,30 @@ class MasterAdamW(torch.optim.Optimizer):
           st['v'] = torch.zeros_like(st['master'])
           st['g32'] = torch.zeros_like(st['master'])
         st['step'] += 1
-        grads.append(st['g32'])
-        masters.append(st['master'])
-        ms.append(st['m'])
-        vs.append(st['v'])
-      # One batched cast of all bf16 grads to the fp32 scratch buffers.
-      torch._foreach_copy_(grads, [prm.grad for prm in params])
-      t = self.state[params[0]]['step']
-      bc1 = 1.0 - beta1 ** t
-      bc2 = 1.0 - beta2 ** t
-      torch._foreach_mul_(ms, beta1)
-      torch._foreach_add_(ms, grads, alpha=1.0 - beta1)
-      torch._foreach_mul_(vs, beta2)
-      torch._foreach_addcmul_(vs, grads, grads, value=1.0 - beta2)
-      denom = torch._foreach_sqrt(vs)
-      torch._foreach_div_(denom, bc2 ** 0.5)
-      torch._foreach_add_(denom, eps)
-      if wd:
-        torch._foreach_mul_(masters, 1.0 - lr * wd)
-      torch._foreach_addcdiv_(masters, ms, denom, value=-(lr / bc1))
-      torch._foreach_copy_(params, masters)  # bf16 cast back
+        by_step.setdefault(st['step'], []).append(prm)
+      for t, bucket in by_step.items():
+        grads, masters, ms, vs = [], [], [], []
+        for prm in bucket:
+          st = self.state[prm]
+          grads.append(st['g32'])
+          masters.append(st['master'])
+          ms.append(st['m'])
+          vs.append(st['v'])
+        # One batched cast of all bf16 grads to the fp32 scratch buffers.
+        torch._foreach_copy_(grads, [prm.grad for prm in bucket])
+        bc1 = 1.0 - beta1 ** t
+        bc2 = 1.0 - beta2 ** t
+        torch._foreach_mul_(ms, beta1)
+        torch._foreach_add_(ms, grads, alpha=1.0 - beta1)
+        torch._foreach_mul_(vs, beta2)
+        torch._foreach_addcmul_(vs, grads, grads, value=1.0 - beta2)
+        denom = torch._foreach_sqrt(vs)
+        torch._foreach_div_(denom, bc2 ** 0.5)
+        torch._foreach_add_(denom, eps)
+        if wd:
+          torch._foreach_mul_(masters, 1.0 - lr * wd)
+        torch._foreach_addcdiv_(masters, ms, denom, value=-(lr / bc1))
+        torch._foreach_copy_(bucket, masters)  # bf16 cast back
     return None
 
 
@@ -274,7 +281,77 @@ class Accumulator(Base):
     self.CreateChild('inner', self.p.optimizer_tpl)
 
   def CreateTorchOptimizer(self, params, lr):
-    return self.inner.CreateTorchOptimizer(params, lr)
+    params = list(params)
+    inner = self.inner.CreateTorchOptimizer(params, lr)
+    if self.p.accum_steps <= 1:
+      return inner
+    return _AccumulatorImpl(inner, params, self.p.accum_steps)
+
+
+class _AccumulatorImpl:
+  """Duck-typed optimizer: sums grads into side buffers each step() call
+  and applies the inner optimizer with the averaged grads every
+  accum_steps calls (reference optimizer.py:507 Accumulator semantics:
+  intermediate steps are pure accumulation, no parameter update)."""
+
+  def __init__(self, inner, params, accum_steps):
+    self._inner = inner
+    self._params = params
+    self._n = int(accum_steps)
+    self._count = 0
+    self._acc = {}  # id(param) -> fp32 accumulation buffer
+
+  @property
+  def param_groups(self):
+    return self._inner.param_groups
+
+  @property
+  def state(self):
+    return self._inner.state
+
+  def zero_grad(self, set_to_none=True):
+    self._inner.zero_grad(set_to_none=set_to_none)
+
+  @torch.no_grad()
+  def step(self, closure=None):
+    for prm in self._params:
+      if prm.grad is None:
+        continue
+      buf = self._acc.get(id(prm))
+      if buf is None:
+        buf = torch.zeros_like(prm, dtype=torch.float32)
+        self._acc[id(prm)] = buf
+      buf.add_(prm.grad.float())
+    self._count += 1
+    if self._count < self._n:
+      return None
+    self._count = 0
+    inv = 1.0 / self._n
+    for prm in self._params:
+      buf = self._acc.get(id(prm))
+      if buf is None:
+        continue
+      if prm.grad is None:
+        prm.grad = torch.zeros_like(prm)
+      prm.grad.copy_((buf * inv).to(prm.grad.dtype))
+      buf.zero_()
+    return self._inner.step()
+
+  def state_dict(self):
+    sd = self._inner.state_dict()
+    return {'inner': sd, 'accum_count': self._count,
+            'accum_bufs': [self._acc.get(id(prm))
+                           for prm in self._params]}
+
+  def load_state_dict(self, state_dict):
+    if 'inner' not in state_dict:  # plain inner-optimizer checkpoint
+      self._inner.load_state_dict(state_dict)
+      return
+    self._inner.load_state_dict(state_dict['inner'])
+    self._count = state_dict.get('accum_count', 0)
+    for prm, buf in zip(self._params, state_dict.get('accum_bufs', [])):
+      if buf is not None:
+        self._acc[id(prm)] = buf.to(prm.device)
 
 
 class AdaDelta(Base):

@@ -50,8 +50,12 @@ class MicroBatcher:
   relies on TF-Serving for. Requests are grouped by (subgraph, feed
   keys, per-example shapes); results split back per caller."""
 
-  def __init__(self, run_fn, max_batch: int = 8, max_wait_ms: float = 3.0):
+  def __init__(self, run_fn, max_batch: int = 8, max_wait_ms: float = 3.0,
+               lock: Optional[threading.Lock] = None):
     self._run = run_fn
+    # Serializes model execution with non-batched requests that run the
+    # same predictor from request threads.
+    self._lock = lock or threading.Lock()
     self.max_batch = max_batch
     self.max_wait = max_wait_ms / 1000.0
     self._q: queue_lib.Queue = queue_lib.Queue()
@@ -100,7 +104,8 @@ class MicroBatcher:
       try:
         stacked = {k: torch.cat([g[0][k] for g in group], dim=0)
                    for k in group[0][0]}
-        out = self._run(subgraph, **stacked)
+        with self._lock:
+          out = self._run(subgraph, **stacked)
         self.batches_run += 1
         self.examples_run += len(group)
         for i, (_, ev_i, slot_i) in enumerate(group):
@@ -120,7 +125,8 @@ def MakeApp(predictor: Predictor, micro_batch: bool = False,
 
   app = FastAPI(title='lingvo_amd inference')
   lock = threading.Lock()
-  batcher = MicroBatcher(predictor.Run, max_batch, max_wait_ms) \
+  batcher = MicroBatcher(predictor.Run, max_batch, max_wait_ms,
+                         lock=lock) \
       if micro_batch else None
   app.state.batcher = batcher
 

@@ -74,3 +74,69 @@ def test_schedules():
   pw = schedule_lib.PiecewiseConstantSchedule.Params().Set(
       name='p', boundaries=[10, 20], values=[1.0, 0.5, 0.1]).Instantiate()
   assert pw.Value(5) == 1.0 and pw.Value(15) == 0.5 and pw.Value(25) == 0.1
+
+
+def test_accumulator_matches_large_batch():
+  """Accumulator(N) over N micro-batches == one step on the mean grad."""
+  torch.manual_seed(7)
+  data = [torch.randn(8, 16) for _ in range(4)]
+  tgt = [torch.randn(8, 4) for _ in range(4)]
+
+  def make(opt_params):
+    torch.manual_seed(3)
+    w = torch.nn.Parameter(torch.randn(16, 4))
+    opt = opt_params.Instantiate().CreateTorchOptimizer([w], lr=0.1)
+    return w, opt
+
+  # Accumulated: 4 micro steps.
+  wa, oa = make(optimizer_lib.Accumulator.Params().Set(
+      name='acc', accum_steps=4,
+      optimizer_tpl=optimizer_lib.SGD.Params()))
+  for x, y in zip(data, tgt):
+    oa.zero_grad()
+    ((x @ wa - y) ** 2).mean().backward()
+    oa.step()
+
+  # Reference: one step on the concatenated batch (same mean grad).
+  wb, ob = make(optimizer_lib.SGD.Params().Set(name='sgd'))
+  ob.zero_grad()
+  ((torch.cat(data) @ wb - torch.cat(tgt)) ** 2).mean().backward()
+  ob.step()
+
+  assert torch.allclose(wa.detach(), wb.detach(), atol=1e-6), \
+      (wa - wb).abs().max()
+
+
+def test_accumulator_no_update_between_applies():
+  w = torch.nn.Parameter(torch.ones(4))
+  p = optimizer_lib.Accumulator.Params().Set(
+      name='acc', accum_steps=3,
+      optimizer_tpl=optimizer_lib.SGD.Params())
+  opt = p.Instantiate().CreateTorchOptimizer([w], lr=1.0)
+  before = w.detach().clone()
+  for i in range(2):
+    opt.zero_grad()
+    (w.sum()).backward()
+    opt.step()
+    assert torch.equal(w.detach(), before)  # accumulating only
+  opt.zero_grad()
+  (w.sum()).backward()
+  opt.step()
+  assert not torch.equal(w.detach(), before)  # applied on 3rd
+
+
+def test_master_adamw_lazy_param_bias_correction():
+  """A param whose grad appears late gets its own bias correction."""
+  w1 = torch.nn.Parameter(torch.randn(8, dtype=torch.bfloat16))
+  w2 = torch.nn.Parameter(torch.randn(8, dtype=torch.bfloat16))
+  opt = optimizer_lib.MasterAdamW([w1, w2], lr=0.01)
+  for i in range(5):
+    opt.zero_grad()
+    w1.grad = torch.ones_like(w1)
+    if i >= 3:
+      w2.grad = torch.ones_like(w2)
+    opt.step()
+  # w2 saw 2 steps; its state step count must be 2 (not 5).
+  assert opt.state[w1]['step'] == 5
+  assert opt.state[w2]['step'] == 2
+  # Fresh optimizer stepping w2-like param twice gives same master.
