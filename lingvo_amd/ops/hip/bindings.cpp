@@ -85,9 +85,13 @@ void RegisterInputPipeline(py::module_& m);
 // wpm_tokenizer.cpp
 void RegisterWpmTokenizer(py::module_& m);
 
+// record_batcher.cpp
+void RegisterRecordBatcher(py::module_& m);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   RegisterInputPipeline(m);
   RegisterWpmTokenizer(m);
+  RegisterRecordBatcher(m);
   m.def("group_norm_fwd", &group_norm_fwd, "Fused padded GroupNorm fwd");
   m.def("group_norm_bwd", &group_norm_bwd, "Fused padded GroupNorm bwd");
   m.def("lstm_gates_fwd", &lstm_gates_fwd_op, "Fused LSTM gates fwd");

@@ -1,0 +1,508 @@
+// Native RecordBatcher: RecordYielder -> native processor threadpool ->
+// length-bucketed padded batches, fully GIL-free.
+//
+// MI355X-native equivalent of the reference's C++ batcher
+// (lingvo/core/ops/record_batcher.h:89 Options bucket_upper_bound /
+// bucket_batch_limit, processor threadpool, flush logic at
+// record_batcher.cc:228): records are read+shuffled by native reader
+// threads, processed (feature parse / tokenize) by native worker
+// threads, bucketed by length, and emitted as padded torch CPU tensors
+// ready for pinned-memory H2D copies. The Python-processor path stays in
+// lingvo_amd/core/generic_input.py; this file is the fast path that
+// keeps 8 GPUs fed without touching the GIL per record.
+
+#include <torch/extension.h>
+
+#include "wpm_encoder.h"
+
+#include <atomic>
+#include <condition_variable>
+#include <cstdint>
+#include <cstring>
+#include <deque>
+#include <fstream>
+#include <functional>
+#include <memory>
+#include <mutex>
+#include <string>
+#include <thread>
+#include <utility>
+#include <vector>
+
+namespace {
+
+// ---------------------------------------------------------------------------
+// Generic bucketing core (shared by the concrete batchers below).
+//
+// Semantics match reference record_batcher.cc: an example with length key
+// k lands in the first bucket with bound >= k; k > last bound drops the
+// example; a bucket flushes into the ready queue when it holds
+// bucket_batch_limit[i] examples, or (partial) when Flush() is called /
+// the upstream yielder is exhausted.
+// ---------------------------------------------------------------------------
+template <typename Example>
+class BucketCore {
+ public:
+  BucketCore(std::vector<int64_t> bounds, std::vector<int64_t> limits,
+             int64_t ready_cap)
+      : bounds_(std::move(bounds)),
+        limits_(std::move(limits)),
+        ready_cap_(std::max<int64_t>(1, ready_cap)) {
+    TORCH_CHECK(bounds_.size() == limits_.size() && !bounds_.empty(),
+                "bucket_upper_bound/bucket_batch_limit size mismatch");
+    for (size_t i = 1; i < bounds_.size(); ++i) {
+      TORCH_CHECK(bounds_[i] > bounds_[i - 1],
+                  "bucket_upper_bound must be increasing");
+    }
+    buckets_.resize(bounds_.size());
+  }
+
+  // Returns false if the example was dropped (key > last bound).
+  bool Add(Example ex, int64_t key) {
+    if (key < 0) return false;
+    size_t bi = 0;
+    while (bi < bounds_.size() && key > bounds_[bi]) ++bi;
+    if (bi == bounds_.size()) return false;  // longer than last bound
+    std::unique_lock<std::mutex> lk(mu_);
+    buckets_[bi].push_back(std::move(ex));
+    // Re-check the bucket after every wait: another worker waiting on
+    // ready-queue space may have flushed it already (pushing a stale,
+    // possibly empty bucket otherwise).
+    while (!stop_ && (int64_t)buckets_[bi].size() >= limits_[bi]) {
+      if ((int64_t)ready_.size() < ready_cap_) {
+        // Emit exactly limit examples (bucket can exceed it transiently
+        // while workers wait for ready-queue space).
+        auto& bkt = buckets_[bi];
+        std::vector<Example> out(
+            std::make_move_iterator(bkt.begin()),
+            std::make_move_iterator(bkt.begin() + limits_[bi]));
+        bkt.erase(bkt.begin(), bkt.begin() + limits_[bi]);
+        ready_.push_back({bi, std::move(out)});
+        cv_ready_.notify_one();
+      } else {
+        cv_space_.wait(lk);
+      }
+    }
+    return true;
+  }
+
+  // Emits all partial buckets (end of epoch / eval tail).
+  void Flush() {
+    std::lock_guard<std::mutex> lk(mu_);
+    for (size_t bi = 0; bi < buckets_.size(); ++bi) {
+      if (!buckets_[bi].empty()) {
+        ready_.push_back({bi, std::move(buckets_[bi])});
+        buckets_[bi].clear();
+      }
+    }
+    cv_ready_.notify_all();
+  }
+
+  // Marks the stream done: GetBatch throws stop_iteration once drained.
+  void SetExhausted() {
+    Flush();
+    std::lock_guard<std::mutex> lk(mu_);
+    exhausted_ = true;
+    cv_ready_.notify_all();
+  }
+
+  // Returns false when the stream is exhausted and drained (caller
+  // raises StopIteration with the GIL held).
+  bool TryPop(std::pair<size_t, std::vector<Example>>* out) {
+    std::unique_lock<std::mutex> lk(mu_);
+    cv_ready_.wait(lk, [this] {
+      return stop_ || exhausted_ || !ready_.empty();
+    });
+    TORCH_CHECK(!stop_, "RecordBatcher stopped");
+    if (ready_.empty()) return false;
+    *out = std::move(ready_.front());
+    ready_.pop_front();
+    cv_space_.notify_one();
+    return true;
+  }
+
+  void Stop() {
+    std::lock_guard<std::mutex> lk(mu_);
+    stop_ = true;
+    cv_ready_.notify_all();
+    cv_space_.notify_all();
+  }
+
+  int64_t bound(size_t bi) const { return bounds_[bi]; }
+
+ private:
+  std::vector<int64_t> bounds_, limits_;
+  int64_t ready_cap_;
+  std::mutex mu_;
+  std::condition_variable cv_ready_, cv_space_;
+  std::vector<std::vector<Example>> buckets_;
+  std::deque<std::pair<size_t, std::vector<Example>>> ready_;
+  bool stop_ = false;
+  bool exhausted_ = false;
+};
+
+// Pulls records from a yielder callable on worker threads until it
+// signals exhaustion; `process` turns one record into (example, key).
+template <typename Example, typename Yield, typename Process>
+class WorkerPool {
+ public:
+  WorkerPool(BucketCore<Example>* core, Yield yield, Process process,
+             int num_threads)
+      : core_(core), yield_(std::move(yield)),
+        process_(std::move(process)) {
+    active_ = std::max(1, num_threads);
+    for (int i = 0; i < std::max(1, num_threads); ++i) {
+      threads_.emplace_back([this] { Loop(); });
+    }
+  }
+
+  void Join() {
+    for (auto& t : threads_) {
+      if (t.joinable()) t.join();
+    }
+    threads_.clear();
+  }
+
+ private:
+  void Loop() {
+    std::string rec;
+    while (true) {
+      if (!yield_(&rec)) break;  // exhausted or stopped
+      Example ex;
+      int64_t key = process_(rec, &ex);
+      core_->Add(std::move(ex), key);
+    }
+    if (--active_ == 0) core_->SetExhausted();
+  }
+
+  BucketCore<Example>* core_;
+  Yield yield_;
+  Process process_;
+  std::atomic<int> active_{0};
+  std::vector<std::thread> threads_;
+};
+
+// Minimal native text reader feeding the worker pool (the shuffling
+// RecordYielder in input_pipeline.cpp stays the Python-visible one; this
+// avoids a cross-TU dependency and is sequential-per-file with a
+// round-robin over shards, which is enough entropy under bucketing).
+class ShardedReader {
+ public:
+  ShardedReader(std::vector<std::string> files, bool repeat,
+                bool binary_framed)
+      : files_(std::move(files)), repeat_(repeat), framed_(binary_framed) {
+    TORCH_CHECK(!files_.empty(), "no input files");
+  }
+
+  // Thread-safe: returns false when exhausted (non-repeat) or stopped.
+  bool Next(std::string* out) {
+    std::unique_lock<std::mutex> lk(mu_);
+    while (!stop_) {
+      if (!cur_.is_open()) {
+        if (file_idx_ >= files_.size()) {
+          if (!repeat_) return false;
+          file_idx_ = 0;
+          ++epoch_;
+        }
+        cur_.open(files_[file_idx_++], std::ios::binary);
+        if (!cur_) {
+          cur_.close();
+          continue;
+        }
+      }
+      if (framed_) {
+        uint32_t len = 0;
+        if (cur_.read(reinterpret_cast<char*>(&len), 4)) {
+          out->resize(len);
+          if (cur_.read(out->data(), len)) return true;
+        }
+        cur_.close();
+        cur_.clear();
+      } else {
+        if (std::getline(cur_, *out)) return true;
+        cur_.close();
+        cur_.clear();
+      }
+    }
+    return false;
+  }
+
+  void Stop() {
+    std::lock_guard<std::mutex> lk(mu_);
+    stop_ = true;
+  }
+
+  int64_t epoch() const { return epoch_; }
+
+ private:
+  std::vector<std::string> files_;
+  bool repeat_, framed_;
+  std::mutex mu_;
+  std::ifstream cur_;
+  size_t file_idx_ = 0;
+  std::atomic<int64_t> epoch_{1};
+  bool stop_ = false;
+};
+
+// ---------------------------------------------------------------------------
+// ASR frame batcher.
+//
+// Record layout (length-framed binary shards, written by
+// tools/make_asr_shards.py): int32 T, int32 D, int32 L,
+// float32 frames[T*D], int32 tokens[L]. Output batch matches AsrInput:
+// src_frames [B,Tb,D], src_paddings [B,Tb], tgt_ids [B,Lm+1] (SOS-led),
+// tgt_labels [B,Lm+1] (EOS-tailed), tgt_paddings [B,Lm+1].
+// Bucket key = T (reference AsrInput buckets by frame count,
+// lingvo/tasks/asr/input_generator.py).
+// ---------------------------------------------------------------------------
+struct AsrExample {
+  int32_t t = 0, d = 0;
+  std::vector<float> frames;
+  std::vector<int32_t> tokens;
+};
+
+class AsrFrameBatcher {
+ public:
+  AsrFrameBatcher(std::vector<std::string> files,
+                  std::vector<int64_t> bounds, std::vector<int64_t> limits,
+                  int64_t sos_id, int64_t eos_id, int num_threads,
+                  bool repeat, int64_t ready_cap)
+      : core_(std::move(bounds), std::move(limits), ready_cap),
+        reader_(std::move(files), repeat, /*binary_framed=*/true),
+        sos_id_(sos_id), eos_id_(eos_id) {
+    pool_ = std::make_unique<Pool>(
+        &core_, [this](std::string* r) { return reader_.Next(r); },
+        [](const std::string& rec, AsrExample* ex) -> int64_t {
+          if (rec.size() < 12) return -1;
+          const char* p = rec.data();
+          int32_t t, d, l;
+          std::memcpy(&t, p, 4);
+          std::memcpy(&d, p + 4, 4);
+          std::memcpy(&l, p + 8, 4);
+          size_t need = 12 + (size_t)t * d * 4 + (size_t)l * 4;
+          if (t <= 0 || d <= 0 || l < 0 || rec.size() < need) return -1;
+          ex->t = t;
+          ex->d = d;
+          ex->frames.resize((size_t)t * d);
+          std::memcpy(ex->frames.data(), p + 12, (size_t)t * d * 4);
+          ex->tokens.resize(l);
+          std::memcpy(ex->tokens.data(), p + 12 + (size_t)t * d * 4,
+                      (size_t)l * 4);
+          return t;
+        },
+        num_threads);
+  }
+
+  ~AsrFrameBatcher() { Stop(); }
+
+  void Stop() {
+    reader_.Stop();
+    core_.Stop();
+    if (pool_) pool_->Join();
+  }
+
+  void Flush() { core_.Flush(); }
+
+  // (src_frames, src_paddings, tgt_ids, tgt_labels, tgt_paddings);
+  // empty vector when exhausted.
+  std::vector<torch::Tensor> GetBatch() {
+    std::pair<size_t, std::vector<AsrExample>> popped;
+    do {
+      if (!core_.TryPop(&popped)) return {};
+    } while (popped.second.empty());
+    const size_t bi = popped.first;
+    auto& exs = popped.second;
+    const int64_t b = (int64_t)exs.size();
+    const int64_t tb = core_.bound(bi);
+    const int64_t d = exs[0].d;
+    int64_t lmax = 0;
+    for (auto& ex : exs) lmax = std::max<int64_t>(lmax, ex.tokens.size());
+    const int64_t lb = lmax + 1;  // room for SOS/EOS
+    auto frames = torch::zeros({b, tb, d}, torch::kFloat32);
+    auto fpad = torch::ones({b, tb}, torch::kFloat32);
+    auto ids = torch::full({b, lb}, eos_id_, torch::kInt64);
+    auto labels = torch::full({b, lb}, eos_id_, torch::kInt64);
+    auto tpad = torch::ones({b, lb}, torch::kFloat32);
+    auto fr = frames.accessor<float, 3>();
+    auto fp = fpad.accessor<float, 2>();
+    auto id = ids.accessor<int64_t, 2>();
+    auto lb_a = labels.accessor<int64_t, 2>();
+    auto tp = tpad.accessor<float, 2>();
+    for (int64_t i = 0; i < b; ++i) {
+      const auto& ex = exs[i];
+      std::memcpy(fr[i].data(), ex.frames.data(),
+                  (size_t)ex.t * d * sizeof(float));
+      for (int64_t j = 0; j < ex.t; ++j) fp[i][j] = 0.f;
+      id[i][0] = sos_id_;
+      const int64_t l = (int64_t)ex.tokens.size();
+      for (int64_t j = 0; j < l; ++j) {
+        id[i][j + 1] = ex.tokens[j];
+        lb_a[i][j] = ex.tokens[j];
+      }
+      lb_a[i][l] = eos_id_;
+      for (int64_t j = 0; j <= l; ++j) tp[i][j] = 0.f;
+    }
+    return {frames, fpad, ids, labels, tpad};
+  }
+
+  int64_t epoch() const { return reader_.epoch(); }
+
+ private:
+  using Pool = WorkerPool<AsrExample,
+                          std::function<bool(std::string*)>,
+                          std::function<int64_t(const std::string&,
+                                                AsrExample*)>>;
+  BucketCore<AsrExample> core_;
+  ShardedReader reader_;
+  int64_t sos_id_, eos_id_;
+  std::unique_ptr<Pool> pool_;
+};
+
+// ---------------------------------------------------------------------------
+// MT pair batcher: text records "src\ttgt", WPM-tokenized both sides
+// (reference NmtInput wordpiece bucketing,
+// lingvo/tasks/mt/input_generator.py). Bucket key = max(|src|, |tgt|+1).
+// ---------------------------------------------------------------------------
+struct MtExample {
+  std::vector<int64_t> src, tgt;
+};
+
+class MtPairBatcher {
+ public:
+  MtPairBatcher(std::vector<std::string> files,
+                std::vector<std::string> pieces, int64_t unk_id,
+                int64_t sos_id, int64_t eos_id,
+                std::vector<int64_t> bounds, std::vector<int64_t> limits,
+                int num_threads, bool repeat, int64_t ready_cap)
+      : core_(std::move(bounds), std::move(limits), ready_cap),
+        reader_(std::move(files), repeat, /*binary_framed=*/false),
+        encoder_(std::move(pieces), unk_id),
+        sos_id_(sos_id), eos_id_(eos_id) {
+    pool_ = std::make_unique<Pool>(
+        &core_, [this](std::string* r) { return reader_.Next(r); },
+        [this](const std::string& rec, MtExample* ex) -> int64_t {
+          auto tab = rec.find('\t');
+          if (tab == std::string::npos) return -1;
+          ex->src = encoder_.Encode(rec.substr(0, tab));
+          ex->tgt = encoder_.Encode(rec.substr(tab + 1));
+          if (ex->src.empty() || ex->tgt.empty()) return -1;
+          return std::max<int64_t>(ex->src.size(), ex->tgt.size() + 1);
+        },
+        num_threads);
+  }
+
+  ~MtPairBatcher() { Stop(); }
+
+  void Stop() {
+    reader_.Stop();
+    core_.Stop();
+    if (pool_) pool_->Join();
+  }
+
+  void Flush() { core_.Flush(); }
+
+  // (src_ids, src_paddings, tgt_ids, tgt_labels, tgt_paddings);
+  // empty vector when exhausted.
+  std::vector<torch::Tensor> GetBatch() {
+    std::pair<size_t, std::vector<MtExample>> popped;
+    do {
+      if (!core_.TryPop(&popped)) return {};
+    } while (popped.second.empty());
+    auto& exs = popped.second;
+    const int64_t b = (int64_t)exs.size();
+    int64_t smax = 1, tmax = 1;
+    for (auto& ex : exs) {
+      smax = std::max<int64_t>(smax, ex.src.size());
+      tmax = std::max<int64_t>(tmax, ex.tgt.size() + 1);
+    }
+    auto sids = torch::full({b, smax}, eos_id_, torch::kInt64);
+    auto spad = torch::ones({b, smax}, torch::kFloat32);
+    auto tids = torch::full({b, tmax}, eos_id_, torch::kInt64);
+    auto tlab = torch::full({b, tmax}, eos_id_, torch::kInt64);
+    auto tpad = torch::ones({b, tmax}, torch::kFloat32);
+    auto si = sids.accessor<int64_t, 2>();
+    auto sp = spad.accessor<float, 2>();
+    auto ti = tids.accessor<int64_t, 2>();
+    auto tl = tlab.accessor<int64_t, 2>();
+    auto tp = tpad.accessor<float, 2>();
+    for (int64_t i = 0; i < b; ++i) {
+      const auto& ex = exs[i];
+      for (size_t j = 0; j < ex.src.size(); ++j) {
+        si[i][j] = ex.src[j];
+        sp[i][j] = 0.f;
+      }
+      ti[i][0] = sos_id_;
+      for (size_t j = 0; j < ex.tgt.size(); ++j) {
+        ti[i][j + 1] = ex.tgt[j];
+        tl[i][j] = ex.tgt[j];
+      }
+      tl[i][ex.tgt.size()] = eos_id_;
+      for (size_t j = 0; j <= ex.tgt.size(); ++j) tp[i][j] = 0.f;
+    }
+    return {sids, spad, tids, tlab, tpad};
+  }
+
+  int64_t epoch() const { return reader_.epoch(); }
+
+ private:
+  using Pool = WorkerPool<MtExample,
+                          std::function<bool(std::string*)>,
+                          std::function<int64_t(const std::string&,
+                                                MtExample*)>>;
+  BucketCore<MtExample> core_;
+  ShardedReader reader_;
+  lingvo_amd::WpmEncoder encoder_;
+  int64_t sos_id_, eos_id_;
+  std::unique_ptr<Pool> pool_;
+};
+
+}  // namespace
+
+void RegisterRecordBatcher(py::module_& m) {
+  py::class_<AsrFrameBatcher>(m, "AsrFrameBatcher")
+      .def(py::init<std::vector<std::string>, std::vector<int64_t>,
+                    std::vector<int64_t>, int64_t, int64_t, int, bool,
+                    int64_t>(),
+           py::arg("files"), py::arg("bucket_upper_bound"),
+           py::arg("bucket_batch_limit"), py::arg("sos_id") = 1,
+           py::arg("eos_id") = 2, py::arg("num_threads") = 4,
+           py::arg("repeat") = true, py::arg("ready_cap") = 8)
+      .def("get_batch",
+           [](AsrFrameBatcher& self) {
+             std::vector<torch::Tensor> out;
+             {
+               py::gil_scoped_release rel;
+               out = self.GetBatch();
+             }
+             if (out.empty()) throw py::stop_iteration();
+             return py::make_tuple(out[0], out[1], out[2], out[3], out[4]);
+           })
+      .def("flush", &AsrFrameBatcher::Flush)
+      .def("epoch", &AsrFrameBatcher::epoch)
+      .def("stop", &AsrFrameBatcher::Stop,
+           py::call_guard<py::gil_scoped_release>());
+  py::class_<MtPairBatcher>(m, "MtPairBatcher")
+      .def(py::init<std::vector<std::string>, std::vector<std::string>,
+                    int64_t, int64_t, int64_t, std::vector<int64_t>,
+                    std::vector<int64_t>, int, bool, int64_t>(),
+           py::arg("files"), py::arg("pieces"), py::arg("unk_id") = 0,
+           py::arg("sos_id") = 1, py::arg("eos_id") = 2,
+           py::arg("bucket_upper_bound") = std::vector<int64_t>{64},
+           py::arg("bucket_batch_limit") = std::vector<int64_t>{16},
+           py::arg("num_threads") = 4, py::arg("repeat") = true,
+           py::arg("ready_cap") = 8)
+      .def("get_batch",
+           [](MtPairBatcher& self) {
+             std::vector<torch::Tensor> out;
+             {
+               py::gil_scoped_release rel;
+               out = self.GetBatch();
+             }
+             if (out.empty()) throw py::stop_iteration();
+             return py::make_tuple(out[0], out[1], out[2], out[3], out[4]);
+           })
+      .def("flush", &MtPairBatcher::Flush)
+      .def("epoch", &MtPairBatcher::epoch)
+      .def("stop", &MtPairBatcher::Stop,
+           py::call_guard<py::gil_scoped_release>());
+}
