@@ -59,6 +59,13 @@ class RecordBatcher:
       try:
         record, source_id = self._yielder.yield_record()
       except StopIteration:
+        # Flush partial buckets (eval tail; reference record_batcher
+        # flush semantics) before signaling end-of-stream.
+        with self._lock:
+          tails = [b for b in self._buckets if b]
+          self._buckets = [[] for _ in self._bounds]
+        for tail in tails:
+          self._out.put(self._Collate(tail))
         self._out.put(None)
         return
       out = self._processor(record)
