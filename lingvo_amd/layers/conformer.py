@@ -233,19 +233,63 @@ class ConformerLayer(BaseLayer):
     return self.final_ln.FProp(theta.final_ln, x), state
 
 
+class _Conv3x3S2Nhwc(torch.autograd.Function):
+  """3x3 stride-2 same-pad conv on NHWC input as 9 offset GEMMs over
+  strided slices of the padded input.
+
+  Replaces the im2col+col2im formulation: the K=9*C contraction is
+  decomposed into 9 plain [B*Ho*Wo, C] x [C, Co] hipBLASLt GEMMs that
+  accumulate in-place (addmm_ beta=1), so NO cols buffer exists in
+  either direction. The old path materialized ~2 GB cols buffers whose
+  col2im backward overflowed 32-bit byte offsets (GPU write faults at
+  large batch, heap-layout dependent) and cost ~90 ms/step at B=128."""
+
+  @staticmethod
+  def forward(ctx, x, w, bias):
+    """x [B,H,W,C] NHWC; w [3,3,C,Co] (kh,kw,cin,cout); bias [Co]."""
+    B, H, W, C = x.shape
+    Co = w.shape[3]
+    Ho, Wo = (H - 1) // 2 + 1, (W - 1) // 2 + 1
+    xp = F.pad(x, (0, 0, 1, 1, 1, 1))  # pad W then H
+    R = B * Ho * Wo
+    out = bias.to(x.dtype).expand(R, Co).contiguous()
+    for dt in range(3):
+      for df in range(3):
+        sl = xp[:, dt:dt + 2 * Ho - 1:2,
+                df:df + 2 * Wo - 1:2, :].contiguous().reshape(R, C)
+        out.addmm_(sl, w[dt, df])
+    ctx.save_for_backward(x, w)
+    ctx.dims = (B, H, W, C, Ho, Wo, Co)
+    ctx.bias_dtype = bias.dtype
+    return out.reshape(B, Ho, Wo, Co)
+
+  @staticmethod
+  def backward(ctx, dout):
+    x, w = ctx.saved_tensors
+    B, H, W, C, Ho, Wo, Co = ctx.dims
+    R = B * Ho * Wo
+    dout = dout.reshape(R, Co)
+    xp = F.pad(x, (0, 0, 1, 1, 1, 1))
+    dxp = torch.zeros_like(xp)
+    dw = torch.empty_like(w)
+    for dt in range(3):
+      for df in range(3):
+        sl = xp[:, dt:dt + 2 * Ho - 1:2,
+                df:df + 2 * Wo - 1:2, :].contiguous().reshape(R, C)
+        dw[dt, df] = (sl.t() @ dout).to(w.dtype)
+        partial = (dout @ w[dt, df].t()).reshape(B, Ho, Wo, C)
+        # Offsets overlap in the padded buffer: accumulate sequentially.
+        dxp[:, dt:dt + 2 * Ho - 1:2, df:df + 2 * Wo - 1:2, :] += partial
+    dx = dxp[:, 1:1 + H, 1:1 + W, :].contiguous()
+    dbias = dout.float().sum(0).to(ctx.bias_dtype)
+    return dx, dw, dbias
+
+
 class ConvSubsampling(BaseLayer):
   """2x Conv2D stride-2 frontend: [B, T, F] mel -> [B, T/4, D]
-  (reference tasks/asr/encoder conv subsampling)."""
-
-  # im2col buffers near 2 GB trip 32-bit byte-offset overflows in the
-  # upstream col2im/baddbmm backward kernels (observed as GPU write
-  # faults whose occurrence depends on heap layout — box-to-box
-  # nondeterminism at B=128, hard fault at B>=160). Keep chunks far
-  # below the edge: backward materializes fp32/extra copies of the cols
-  # buffer, so the safe bound is ~2^29 bytes, not 2^31.
-  # Env-overridable for fault-localization tests.
-  MAX_COLS_BYTES = int(os.environ.get('LINGVO_AMD_MAX_COLS_BYTES',
-                                      2 ** 28))
+  (reference tasks/asr/encoder conv subsampling). conv1 (Cin=1, K=9) is
+  a small unfold+GEMM; conv2 (Cin=Cout=C) runs the offset-GEMM
+  _Conv3x3S2Nhwc path."""
 
   @classmethod
   def Params(cls):
@@ -274,45 +318,24 @@ class ConvSubsampling(BaseLayer):
     self.CreateVariable('proj_b', py_utils.WeightParams(
         [p.output_dim], py_utils.WeightInit.Constant(0.0), p.dtype))
 
-  @staticmethod
-  def _ConvGemm(x: torch.Tensor, w_oihw: torch.Tensor, bias: torch.Tensor,
-                stride: int = 2, pad: int = 1) -> torch.Tensor:
-    """3x3 strided conv as im2col + hipBLASLt GEMM. MIOpen's algorithm
-    search can fall back to a naive NCHW kernel for these shapes on
-    gfx950 (observed ~1000x regression under rocprof); the unfold+GEMM
-    path always lands on Tensile.
-
-    The batch is chunked so the im2col buffer stays under 2^31 BYTES:
-    beyond that, 32-bit byte offsets overflow in the col2im backward
-    (observed as a GPU write fault at per-buffer >= 2.2 GB)."""
-    o = w_oihw.shape[0]
-    bsz, cin = x.shape[0], x.shape[1]
-    hout = (x.shape[2] + 2 * pad - 3) // stride + 1
-    wout = (x.shape[3] + 2 * pad - 3) // stride + 1
-    cols_bytes_per_ex = cin * 9 * hout * wout * x.element_size()
-    max_chunk = max(1, int(ConvSubsampling.MAX_COLS_BYTES //
-                           cols_bytes_per_ex))
-    outs = []
-    for s in range(0, bsz, max_chunk):
-      xc = x[s:s + max_chunk]
-      bc = xc.shape[0]
-      cols = F.unfold(xc, kernel_size=3, stride=stride, padding=pad)
-      out = torch.baddbmm(
-          bias.reshape(1, o, 1),
-          w_oihw.reshape(1, o, -1).expand(bc, -1, -1), cols)
-      outs.append(out.reshape(bc, o, hout, wout))
-    return outs[0] if len(outs) == 1 else torch.cat(outs, dim=0)
-
   def FProp(self, theta: NestedMap, inputs: torch.Tensor,
             paddings: torch.Tensor):
     """inputs [B, T, F] -> (out [B, ceil(T/4), D], out_paddings)."""
-    x = inputs.unsqueeze(1)  # [B,1,T,F]
-    w1 = theta.conv1_w.permute(3, 2, 0, 1).contiguous()
-    x = F.relu(self._ConvGemm(x, w1, theta.conv1_b))
-    w2 = theta.conv2_w.permute(3, 2, 0, 1).contiguous()
-    x = F.relu(self._ConvGemm(x, w2, theta.conv2_b))
-    b, ch, t4, f4 = x.shape
-    x = x.permute(0, 2, 3, 1).reshape(b, t4, f4 * ch)
+    b, t, f = inputs.shape
+    ch = self._ch
+    # conv1 (Cin=1): unfold + GEMM, output NHWC directly. cols are tiny
+    # (9 x T*F/4 per example).
+    cols = F.unfold(inputs.unsqueeze(1), kernel_size=3, stride=2,
+                    padding=1)  # [B, 9, HW]
+    h1 = (f + 1) // 2
+    t1 = (t + 1) // 2
+    x = torch.matmul(cols.transpose(1, 2),
+                     theta.conv1_w.reshape(9, ch)) + theta.conv1_b
+    x = F.relu(x).reshape(b, t1, h1, ch)  # NHWC
+    # conv2: offset-GEMM path (no cols buffer).
+    x = F.relu(_Conv3x3S2Nhwc.apply(x, theta.conv2_w, theta.conv2_b))
+    _, t4, f4, _ = x.shape
+    x = x.reshape(b, t4, f4 * ch)
     out = torch.addmm(theta.proj_b, x.reshape(-1, f4 * ch),
                       theta.proj_w).reshape(b, t4, -1)
     out_paddings = paddings[:, ::2][:, ::2]

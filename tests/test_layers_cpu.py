@@ -327,21 +327,41 @@ def test_mha_gqa_kv_heads_layer():
   assert out.shape == (2, 6, 128)
 
 
-def test_conv_subsampling_chunking_equivalence():
+def test_conv_subsampling_matches_conv2d():
+  """Offset-GEMM frontend == F.conv2d reference (fwd + grads)."""
+  import torch.nn.functional as F
   p = conformer_lib.ConvSubsampling.Params().Set(
       name='sub', input_freq_dim=16, output_dim=32, channels=8,
       random_seed=1)
   sub = p.Instantiate()
-  x = torch.randn(6, 24, 16)
-  pad = torch.zeros(6, 24)
-  full, _ = sub.FProp(sub.theta, x, pad)
-  old = conformer_lib.ConvSubsampling.MAX_COLS_BYTES
-  try:
-    conformer_lib.ConvSubsampling.MAX_COLS_BYTES = 16 * 1024  # force chunks
-    chunked, _ = sub.FProp(sub.theta, x, pad)
-  finally:
-    conformer_lib.ConvSubsampling.MAX_COLS_BYTES = old
-  assert torch.allclose(full, chunked, atol=1e-5)
+  x = torch.randn(6, 25, 16)  # odd T exercises the edge rows
+  pad = torch.zeros(6, 25)
+  out, out_pad = sub.FProp(sub.theta, x, pad)
+  loss = out.float().square().sum()
+  loss.backward()
+  got_grads = {n: prm.grad.clone() for n, prm in sub.named_parameters()}
+  sub.zero_grad(set_to_none=True)
+
+  # Reference: plain conv2d stack on NCHW.
+  w1 = sub.theta.conv1_w.permute(3, 2, 0, 1)  # [ch,1,3,3]
+  w2 = sub.theta.conv2_w.permute(3, 2, 0, 1)
+  r = F.relu(F.conv2d(x.unsqueeze(1), w1, sub.theta.conv1_b,
+                      stride=2, padding=1))
+  r = F.relu(F.conv2d(r, w2, sub.theta.conv2_b, stride=2, padding=1))
+  b, ch, t4, f4 = r.shape
+  r = r.permute(0, 2, 3, 1).reshape(b, t4, f4 * ch)
+  ref = torch.addmm(sub.theta.proj_b, r.reshape(-1, f4 * ch),
+                    sub.theta.proj_w).reshape(b, t4, -1)
+  ref_pad = pad[:, ::2][:, ::2][:, :t4]
+  from lingvo_amd.core import py_utils as pu
+  ref = pu.ApplyPadding(ref_pad, ref)
+  assert torch.allclose(out, ref, atol=1e-4), (out - ref).abs().max()
+  ref.float().square().sum().backward()
+  ref_grads = {n: prm.grad.clone() for n, prm in sub.named_parameters()}
+  for n in got_grads:
+    assert torch.allclose(got_grads[n], ref_grads[n], atol=1e-3,
+                          rtol=1e-3), (n, (got_grads[n] -
+                                           ref_grads[n]).abs().max())
 
 
 def test_rope_relative_property():
