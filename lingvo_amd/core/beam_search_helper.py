@@ -29,126 +29,99 @@ class BeamSearchHelper:
     p = Params()
     p.Define('num_hyps_per_beam', 8, 'Beam width K.')
     p.Define('beam_size', 3.0,
-             'Stop when best active score is worse than best terminated '
-             'minus this margin (reference x_ops.cc:116 beam_size).')
+             'A beam is done when every live hyp scores below the best '
+             'terminated score minus this margin (x_ops.cc:116).')
     p.Define('length_normalization', 0.0, 'Length-norm alpha.')
     p.Define('valid_eos_max_logit_delta', 5.0,
-             'EOS may terminate only if its score is within this delta '
-             'of the best non-EOS extension.')
+             'EOS may terminate a hyp only if its global score is '
+             'within this delta of that hyp\'s best extension.')
+    p.Define('local_eos_threshold', -100.0,
+             'EOS must additionally beat this local log-prob.')
+    p.Define('merge_paths', False,
+             'Epsilon-emitting models: merge hyps whose epsilon-'
+             'stripped histories match (log-sum-exp scores).')
+    p.Define('ensure_full_beam', False,
+             'Beam not done until it holds K terminated hyps.')
+    p.Define('force_eos_in_last_step', False,
+             'Accept EOS terminations unconditionally at max_steps-1.')
+    p.Define('batch_major_state', True, 'Unused; API parity.')
     p.Define('target_sos_id', 1, 'SOS token.')
     p.Define('target_eos_id', 2, 'EOS token.')
+    p.Define('target_eoc_id', -1, 'Epsilon/end-of-chunk token (RNN-T).')
     p.Define('max_steps', 128, 'Maximum decode length.')
     return p
 
   def __init__(self, params: Params):
     self.p = params
 
-  def _Norm(self, length: torch.Tensor) -> torch.Tensor:
-    alpha = self.p.length_normalization
-    if alpha == 0.0:
-      return torch.ones_like(length, dtype=torch.float32)
-    return ((5.0 + length.float()) / 6.0) ** alpha
-
   def BeamSearchDecode(self, batch: int, init_fn, step_fn,
                        reorder_fn) -> NestedMap:
-    """Runs the search. reorder_fn(state, gather_idx [B*K]) -> state."""
+    """Runs the search with the faithful step op
+    (beam_search_step.BeamSearchStep).
+
+    Callback contract (reference beam_search_helper.py:203-260):
+      init_fn(batch, k) -> decoder state (tensors indexed beam-major,
+        m = beam*k + j)
+      step_fn(state, prev_ids [B*K]) -> (log_probs [B*K, V], state)
+      reorder_fn(state, gather [B*K]) -> state reshuffled
+    """
+    from lingvo_amd.core import beam_search_step as bss
     p = self.p
     k = p.num_hyps_per_beam
     state = init_fn(batch, k)
-    device = state.device if hasattr(state, 'device') else \
-        next(iter(s for s in state.Flatten()
-                  if isinstance(s, torch.Tensor))).device
-
+    device = next(iter(s for s in state.Flatten()
+                       if isinstance(s, torch.Tensor))).device
     bk = batch * k
+    # beam-major (tasks) <-> hyp-major (step op) index maps.
+    j_idx = torch.arange(bk, device=device) % k
+    b_idx = torch.arange(bk, device=device) // k
+    to_hyp_major = (j_idx * batch + b_idx)           # m -> i
+    jj = torch.arange(bk, device=device) // batch
+    bb = torch.arange(bk, device=device) % batch
+    to_beam_major = (bb * k + jj)                    # i -> m
+
+    ss = bss.BeamSearchState.Init(batch, k, p.max_steps)
     prev_ids = torch.full((bk,), p.target_sos_id, dtype=torch.long,
                           device=device)
-    # Only hyp 0 of each beam starts alive (others -inf) so the first
-    # step doesn't produce k duplicates.
-    cum_scores = torch.full((batch, k), -1e30, device=device)
-    cum_scores[:, 0] = 0.0
-    histories = torch.zeros(batch, k, p.max_steps, dtype=torch.long,
-                            device=device)
-    done_scores = torch.full((batch, k), -1e30, device=device)
-    done_norm_scores = torch.full((batch, k), -1e30, device=device)
-    done_ids = torch.zeros(batch, k, p.max_steps, dtype=torch.long,
-                           device=device)
-    done_lens = torch.zeros(batch, k, dtype=torch.long, device=device)
-
     for t in range(p.max_steps):
-      log_probs, state = step_fn(state, prev_ids)  # [B*K, V]
-      v = log_probs.shape[-1]
-      total = cum_scores.reshape(bk, 1) + log_probs.float()  # [B*K, V]
-
-      # EOS handling: candidate terminations this step.
-      eos_scores = total[:, p.target_eos_id].reshape(batch, k)
-      no_eos = total.clone()
-      no_eos[:, p.target_eos_id] = -1e30
-      best_no_eos = no_eos.max(dim=-1).values.reshape(batch, k)
-
-      # Top-k over the flattened (hyp, vocab) extension space.
-      flat = no_eos.reshape(batch, k * v)
-      top_scores, top_idx = flat.topk(k, dim=-1)  # [B, K]
-      prev_hyp = top_idx // v  # [B, K] index into previous hyps
-      new_tok = top_idx % v
-
-      # Terminations: EOS within delta of the best extension of that hyp.
-      eos_valid = eos_scores >= (best_no_eos -
-                                 p.valid_eos_max_logit_delta)
-      eos_norm = eos_scores * 0 + eos_scores  # placeholder for clarity
-      lens = torch.full((batch, k), t + 1, device=device)
-      eos_norm = eos_scores / self._Norm(lens)
-      improve = eos_valid & (eos_norm > done_norm_scores.min(
-          dim=-1, keepdim=True).values)
-      if bool(improve.any()):
-        for b in range(batch):
-          for h in range(k):
-            if bool(improve[b, h]):
-              slot = int(done_norm_scores[b].argmin())
-              if float(eos_norm[b, h]) > float(done_norm_scores[b, slot]):
-                done_norm_scores[b, slot] = eos_norm[b, h]
-                done_scores[b, slot] = eos_scores[b, h]
-                done_ids[b, slot, :t] = histories[b, h, :t]
-                done_ids[b, slot, t] = p.target_eos_id
-                done_lens[b, slot] = t + 1
-
-      # Reshuffle live hyps.
-      gather = (torch.arange(batch, device=device).unsqueeze(1) * k +
-                prev_hyp).reshape(bk)
-      histories = histories.reshape(bk, -1)[gather].reshape(
-          batch, k, -1)
-      histories[:, :, t] = new_tok
-      cum_scores = top_scores
-      prev_ids = new_tok.reshape(bk)
-      state = reorder_fn(state, gather)
-
-      # Early stop: best possible live score worse than worst kept done.
-      best_live = (cum_scores.max(dim=-1).values /
-                   self._Norm(torch.full((batch,), t + 1, device=device)))
-      worst_done = done_norm_scores.min(dim=-1).values
-      have_all = (done_norm_scores > -1e29).all(dim=-1)
-      if bool((have_all &
-               (best_live + p.beam_size < worst_done)).all()):
+      log_probs, state = step_fn(state, prev_ids)    # [B*K (beam-major), V]
+      scores_hyp_major = log_probs.float()[to_hyp_major]
+      gather_h = bss.BeamSearchStep(
+          scores_hyp_major, ss, t,
+          eos_id=p.target_eos_id, eoc_id=p.target_eoc_id,
+          beam_size=p.beam_size,
+          valid_eos_max_logit_delta=p.valid_eos_max_logit_delta,
+          local_eos_threshold=p.local_eos_threshold,
+          merge_paths=p.merge_paths,
+          ensure_full_beam=p.ensure_full_beam,
+          force_eos_in_last_step=p.force_eos_in_last_step)
+      # Convert the hyp-major gather (new i <- old hyp gather_h[i]) into
+      # the tasks' beam-major layout.
+      gather_h = gather_h.to(device)
+      gather_m = to_beam_major[gather_h[to_hyp_major]]
+      state = reorder_fn(state, gather_m)
+      prev_ids = ss.hyps[t].to(device)[to_hyp_major]
+      if ss.all_done:
         break
 
-    # Fill any empty done slots with live hyps.
-    live_norm = cum_scores / self._Norm(
-        torch.full((batch, k), p.max_steps, device=device))
-    for b in range(batch):
-      for slot in range(k):
-        if float(done_norm_scores[b, slot]) < -1e29:
-          h = int(live_norm[b].argmax())
-          done_norm_scores[b, slot] = live_norm[b, h]
-          done_scores[b, slot] = cum_scores[b, h]
-          done_ids[b, slot] = histories[b, h]
-          done_lens[b, slot] = p.max_steps
-          live_norm[b, h] = -1e30
-
-    order = done_norm_scores.argsort(dim=-1, descending=True)
-    gather3 = order.unsqueeze(-1).expand_as(done_ids)
+    out = bss.TopKTerminatedHyps(
+        ss, k, length_normalization=p.length_normalization)
+    # Robustness: beams that never terminated get their best live hyp
+    # (the reference leaves them empty; empty hyps break downstream
+    # metrics, so surface the live content instead).
+    for beam in range(batch):
+      if float(out.topk_scores[beam, 0]) < -1e29:
+        ids, _ = bss._TraceIds(ss, beam, min(p.max_steps, ss.max_steps))
+        ln = min(len(ids), out.topk_ids.shape[-1])
+        if ln:
+          out.topk_ids[beam, 0, :ln] = torch.tensor(ids[:ln])
+        out.topk_lens[beam, 0] = ln
+        out.topk_scores[beam, 0] = float(
+            ss.cumulative_scores[beam])
     return NestedMap(
-        topk_ids=done_ids.gather(1, gather3),
-        topk_lens=done_lens.gather(1, order),
-        topk_scores=done_norm_scores.gather(1, order))
+        topk_ids=out.topk_ids.to(device),
+        topk_lens=out.topk_lens.to(device),
+        topk_scores=out.topk_scores.to(device))
 
 
 class GreedySearchHelper:

@@ -1,0 +1,184 @@
+"""Semantics tests for the faithful beam-search step op
+(lingvo_amd/core/beam_search_step.py), checked against hand-computed
+oracles of the reference C++ behavior
+(lingvo/core/ops/beam_search_step_op_kernels.cc:111,681,845)."""
+
+import math
+import os
+import sys
+
+import torch
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), '..'))
+
+from lingvo_amd.core import beam_search_step as bss  # noqa: E402
+from lingvo_amd.core.beam_search_helper import BeamSearchHelper  # noqa
+from lingvo_amd.core.nested_map import NestedMap  # noqa: E402
+
+EOS = 2
+
+
+def _scores(n, v, entries):
+  """entries: {(hyp, tok): logprob}; everything else very low."""
+  s = torch.full((n, v), -20.0)
+  for (h, t), val in entries.items():
+    s[h, t] = val
+  return s
+
+
+def test_first_step_only_first_hyp():
+  """At t=0 only hyp 0 of each beam proposes (reference :171)."""
+  st = bss.BeamSearchState.Init(num_beams=1, k=2, max_steps=4)
+  # hyp 0 proposes tok 5 (-1) and tok 6 (-2); hyp 1 would propose tok 7
+  # at score -0.1 but must be ignored on the first step.
+  sc = _scores(2, 10, {(0, 5): -1.0, (0, 6): -2.0, (1, 7): -0.1})
+  bss.BeamSearchStep(sc, st, t=0)
+  assert int(st.hyps[0, 0]) == 5
+  assert int(st.hyps[0, 1]) == 6
+  assert abs(float(st.cumulative_scores[0]) - (-1.0)) < 1e-5
+
+
+def test_valid_eos_max_logit_delta_gates_termination():
+  st = bss.BeamSearchState.Init(num_beams=1, k=2, max_steps=4)
+  # Best extension -1.0; EOS at -7.0 is outside delta=5 -> no done hyp.
+  sc = _scores(2, 10, {(0, 5): -1.0, (0, EOS): -7.0})
+  bss.BeamSearchStep(sc, st, t=0, valid_eos_max_logit_delta=5.0)
+  assert not st.done_hyps
+  # EOS within delta -> terminates.
+  st2 = bss.BeamSearchState.Init(num_beams=1, k=2, max_steps=4)
+  sc2 = _scores(2, 10, {(0, 5): -1.0, (0, EOS): -3.0})
+  bss.BeamSearchStep(sc2, st2, t=0, valid_eos_max_logit_delta=5.0)
+  assert len(st2.done_hyps) == 1
+  assert st2.done_hyps[0].ids == [EOS]
+  assert abs(st2.done_hyps[0].global_score - (-3.0)) < 1e-5
+  assert abs(st2.best_scores[0] - (-3.0)) < 1e-5
+
+
+def test_local_eos_threshold():
+  st = bss.BeamSearchState.Init(num_beams=1, k=2, max_steps=4)
+  sc = _scores(2, 10, {(0, 5): -1.0, (0, EOS): -3.0})
+  bss.BeamSearchStep(sc, st, t=0, local_eos_threshold=-2.0)
+  assert not st.done_hyps  # local -3.0 below threshold -2.0
+
+
+def test_force_eos_in_last_step():
+  st = bss.BeamSearchState.Init(num_beams=1, k=1, max_steps=2)
+  sc0 = _scores(1, 10, {(0, 5): -1.0})
+  bss.BeamSearchStep(sc0, st, t=0, force_eos_in_last_step=True)
+  assert not st.done_hyps
+  # Last step: EOS terminates even far outside the delta.
+  sc1 = _scores(1, 10, {(0, 6): -1.0, (0, EOS): -15.0})
+  bss.BeamSearchStep(sc1, st, t=1, force_eos_in_last_step=True,
+                     valid_eos_max_logit_delta=5.0)
+  assert len(st.done_hyps) == 1
+  assert st.done_hyps[0].ids == [5, EOS]
+
+
+def test_beam_done_uses_beam_size_margin():
+  st = bss.BeamSearchState.Init(num_beams=1, k=1, max_steps=8)
+  # Step 0: a good EOS (-1.0) terminates; live continuation is -1.5.
+  sc = _scores(1, 10, {(0, 5): -1.5, (0, EOS): -1.0})
+  bss.BeamSearchStep(sc, st, t=0, beam_size=3.0)
+  assert not st.all_done  # -1.5 > best(-1.0) - 3.0
+  # Step 1: live falls to -6.0 < -1.0 - 3.0 -> beam done.
+  sc1 = _scores(1, 10, {(0, 5): -4.5})
+  bss.BeamSearchStep(sc1, st, t=1, beam_size=3.0)
+  assert st.all_done
+
+
+def test_ensure_full_beam_blocks_done():
+  st = bss.BeamSearchState.Init(num_beams=1, k=2, max_steps=8)
+  sc = _scores(2, 10, {(0, 5): -9.0, (0, EOS): -1.0})
+  bss.BeamSearchStep(sc, st, t=0, beam_size=3.0, ensure_full_beam=True)
+  # 1 done hyp < K=2 -> not done even though live hyps are terrible.
+  assert len(st.done_hyps) == 1
+  assert not st.all_done
+  # Without ensure_full_beam the same state IS done.
+  st2 = bss.BeamSearchState.Init(num_beams=1, k=2, max_steps=8)
+  bss.BeamSearchStep(sc.clone(), st2, t=0, beam_size=3.0,
+                     ensure_full_beam=False)
+  assert st2.all_done
+
+
+def test_merge_paths_logsumexp():
+  """Two paths that differ only by epsilon placement merge with
+  log-sum-exp scores (InsertHypWithEpsilonDedupe, kernels.h:138)."""
+  EOC = 3
+  st = bss.BeamSearchState.Init(num_beams=1, k=2, max_steps=4)
+  # t=0: two hyps: hyp0 emits token 5, hyp1 emits epsilon... first step
+  # considers only hyp 0; set up two live paths at t=1 instead:
+  sc0 = _scores(2, 10, {(0, 5): -1.0, (0, EOC): -1.2})
+  bss.BeamSearchStep(sc0, st, t=0, eoc_id=EOC, merge_paths=True)
+  # Now hyp A = [5] (score -1.0), hyp B = [eps] (score -1.2).
+  # t=1: A emits eps (-0.7); B emits 5 (-0.5). Both paths strip to [5]
+  # -> must merge into ONE hyp with score logsumexp(-1.7, -1.7).
+  scA = {(0, EOC): -0.7, (1, 5): -0.5}
+  sc1 = _scores(2, 10, scA)
+  bss.BeamSearchStep(sc1, st, t=1, eoc_id=EOC, merge_paths=True)
+  merged = _log_sum_exp(-1.0 + -0.7, -1.2 + -0.5)
+  # The merged hyp is the best live hyp.
+  assert abs(float(st.cumulative_scores[0]) - merged) < 1e-5
+  # And the second slot must NOT be the duplicate [5] path: its score
+  # differs from both raw continuations.
+  assert abs(float(st.cumulative_scores[1]) - (-1.7)) > 1e-6
+
+
+def _log_sum_exp(a, b):
+  m = max(a, b)
+  return m + math.log(math.exp(a - m) + math.exp(b - m))
+
+
+def test_topk_terminated_hyps_length_norm():
+  st = bss.BeamSearchState.Init(num_beams=1, k=2, max_steps=8)
+  st.done_hyps = [
+      bss.DoneHyp(0, [5, EOS], [-1.0, -1.0], -2.0, 1),
+      bss.DoneHyp(0, [5, 6, 7, EOS], [-0.6] * 4, -2.4, 3),
+  ]
+  # alpha=0: raw scores, short hyp wins.
+  out = bss.TopKTerminatedHyps(st, 2, length_normalization=0.0)
+  assert int(out.topk_lens[0, 0]) == 2
+  # Strong length norm: longer hyp's normalized score wins.
+  out = bss.TopKTerminatedHyps(st, 2, length_normalization=2.0)
+  n_short = -2.0 / (((2 + 5) / 6.0) ** 2)
+  n_long = -2.4 / (((4 + 5) / 6.0) ** 2)
+  assert n_long > n_short
+  assert int(out.topk_lens[0, 0]) == 4
+
+
+def test_helper_end_to_end_prefers_high_prob_sequence():
+  """Driver-level: a toy model whose argmax path is 7 7 7 EOS."""
+  p = BeamSearchHelper.Params().Set(
+      num_hyps_per_beam=3, max_steps=6, force_eos_in_last_step=True)
+  helper = BeamSearchHelper(p)
+  V = 10
+
+  def init_fn(batch, k):
+    return NestedMap(step=torch.zeros(batch * k, dtype=torch.long))
+
+  def step_fn(state, prev_ids):
+    n = prev_ids.shape[0]
+    logp = torch.full((n, V), -8.0)
+    # favor token 7 for 3 steps, then EOS.
+    for i in range(n):
+      t = int(state.step[i])
+      if t < 3:
+        logp[i, 7] = -0.1
+        logp[i, 6] = -1.0
+      else:
+        logp[i, EOS] = -0.05
+        logp[i, 7] = -3.0
+    state.step = state.step + 1
+    return logp, state
+
+  def reorder_fn(state, gather):
+    state.step = state.step[gather]
+    return state
+
+  out = helper.BeamSearchDecode(2, init_fn, step_fn, reorder_fn)
+  best = out.topk_ids[0, 0, :int(out.topk_lens[0, 0])].tolist()
+  assert best == [7, 7, 7, EOS], best
+  best1 = out.topk_ids[1, 0, :int(out.topk_lens[1, 0])].tolist()
+  assert best1 == [7, 7, 7, EOS], best1
+  # Scores sorted descending.
+  sc = out.topk_scores[0]
+  assert sc[0] >= sc[1] >= sc[2]
