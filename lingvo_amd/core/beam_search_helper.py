@@ -85,7 +85,9 @@ class BeamSearchHelper:
                           device=device)
     for t in range(p.max_steps):
       log_probs, state = step_fn(state, prev_ids)    # [B*K (beam-major), V]
-      scores_hyp_major = log_probs.float()[to_hyp_major]
+      # Row i (hyp-major) holds the scores of slot m(i) (beam-major):
+      # index with the i->m map.
+      scores_hyp_major = log_probs.float()[to_beam_major]
       gather_h = bss.BeamSearchStep(
           scores_hyp_major, ss, t,
           eos_id=p.target_eos_id, eoc_id=p.target_eoc_id,
@@ -109,15 +111,19 @@ class BeamSearchHelper:
     # Robustness: beams that never terminated get their best live hyp
     # (the reference leaves them empty; empty hyps break downstream
     # metrics, so surface the live content instead).
-    for beam in range(batch):
-      if float(out.topk_scores[beam, 0]) < -1e29:
-        ids, _ = bss._TraceIds(ss, beam, min(p.max_steps, ss.max_steps))
-        ln = min(len(ids), out.topk_ids.shape[-1])
-        if ln:
-          out.topk_ids[beam, 0, :ln] = torch.tensor(ids[:ln])
-        out.topk_lens[beam, 0] = ln
-        out.topk_scores[beam, 0] = float(
-            ss.cumulative_scores[beam])
+    steps_run = min(t + 1, p.max_steps)
+    if any(float(out.topk_scores[beam, 0]) < -1e29
+           for beam in range(batch)):
+      if out.topk_ids.shape[-1] < steps_run:
+        pad = steps_run - out.topk_ids.shape[-1]
+        out.topk_ids = torch.nn.functional.pad(out.topk_ids, (0, pad))
+      for beam in range(batch):
+        if float(out.topk_scores[beam, 0]) < -1e29:
+          ids, _ = bss._TraceIds(ss, beam, steps_run)
+          if ids:
+            out.topk_ids[beam, 0, :len(ids)] = torch.tensor(ids)
+          out.topk_lens[beam, 0] = len(ids)
+          out.topk_scores[beam, 0] = float(ss.cumulative_scores[beam])
     return NestedMap(
         topk_ids=out.topk_ids.to(device),
         topk_lens=out.topk_lens.to(device),
