@@ -44,6 +44,9 @@ class BeamSearchHelper:
              'Beam not done until it holds K terminated hyps.')
     p.Define('force_eos_in_last_step', False,
              'Accept EOS terminations unconditionally at max_steps-1.')
+    p.Define('force_eos_in_top_k', False,
+             'Always evaluate the EOS extension of every hyp, even if '
+             'it falls outside the per-hyp top-k (x_ops.cc:177).')
     p.Define('batch_major_state', True, 'Unused; API parity.')
     p.Define('target_sos_id', 1, 'SOS token.')
     p.Define('target_eos_id', 2, 'EOS token.')
@@ -96,7 +99,8 @@ class BeamSearchHelper:
           local_eos_threshold=p.local_eos_threshold,
           merge_paths=p.merge_paths,
           ensure_full_beam=p.ensure_full_beam,
-          force_eos_in_last_step=p.force_eos_in_last_step)
+          force_eos_in_last_step=p.force_eos_in_last_step,
+          force_eos_in_top_k=p.force_eos_in_top_k)
       # Convert the hyp-major gather (new i <- old hyp gather_h[i]) into
       # the tasks' beam-major layout.
       gather_h = gather_h.to(device)

@@ -181,7 +181,7 @@ def test_flat_vs_batched_beam_fuzz(seed, k, vocab):
   # policies coincide for the comparison.
   ref = bsh.BeamSearchHelper(bsh.BeamSearchHelper.Params().Set(
       num_hyps_per_beam=k, max_steps=6,
-      valid_eos_max_logit_delta=1e9))
+      valid_eos_max_logit_delta=1e9, force_eos_in_top_k=True))
   flat = fbsh.FlatBeamSearchHelper(fbsh.FlatBeamSearchHelper.Params().Set(
       num_hyps_per_beam=k, max_steps=6, length_norm_alpha=0.0))
   o1 = ref.BeamSearchDecode(1, init_fn, step_fn, reorder_fn)
