@@ -181,3 +181,44 @@ def test_lstm_gates_matches_ref():
   assert (m1.float() - m_ref.detach()).abs().max() < 0.03
   assert (gates.grad.float() - gr.grad).abs().max() < 0.05
   assert (c0.grad.float() - cr.grad).abs().max() < 0.05
+
+
+@gpu
+def test_conv_subsampling_gpu_matches_fp32_conv2d():
+  """Offset-GEMM frontend (bf16, hipBLASLt) vs fp32 conv2d reference at
+  a bench-shaped slice."""
+  import torch.nn.functional as F
+  from lingvo_amd.layers import conformer as conformer_lib
+  torch.manual_seed(5)
+  p = conformer_lib.ConvSubsampling.Params().Set(
+      name='sub', input_freq_dim=80, output_dim=64, channels=16,
+      random_seed=3)
+  p.dtype = torch.bfloat16
+  sub = p.Instantiate().to('cuda')
+  x = torch.randn(4, 300, 80, device='cuda', dtype=torch.bfloat16)
+  pad = torch.zeros(4, 300, device='cuda')
+  out, _ = sub.FProp(sub.theta, x, pad)
+  out.float().square().sum().backward()
+  got_w2 = sub.conv2_w.grad.clone().float()
+  sub.zero_grad(set_to_none=True)
+
+  w1 = sub.theta.conv1_w.detach().float().permute(3, 2, 0, 1)
+  w2 = sub.theta.conv2_w.detach().float().requires_grad_(True)
+  r = F.relu(F.conv2d(x.float().unsqueeze(1), w1,
+                      sub.theta.conv1_b.detach().float(), stride=2,
+                      padding=1))
+  r = F.relu(F.conv2d(r, w2.permute(3, 2, 0, 1),
+                      sub.theta.conv2_b.detach().float(), stride=2,
+                      padding=1))
+  b, ch, t4, f4 = r.shape
+  r = r.permute(0, 2, 3, 1).reshape(b, t4, f4 * ch)
+  ref = torch.addmm(sub.theta.proj_b.detach().float(),
+                    r.reshape(-1, f4 * ch),
+                    sub.theta.proj_w.detach().float()
+                    ).reshape(b, t4, -1)
+  rel = (out.float() - ref).abs().max() / ref.abs().max().clamp_min(1)
+  assert rel < 0.05, rel
+  ref.square().sum().backward()
+  wrel = (got_w2 - w2.grad).abs().max() / \
+      w2.grad.abs().max().clamp_min(1e-6)
+  assert wrel < 0.08, wrel
