@@ -125,6 +125,13 @@ class MoEFeedForwardLayer(BaseLayer):
     p.Define('moe_group', None,
              'torch.distributed group for EP (None = default group when '
              'initialized).')
+    p.Define('shard_experts', True,
+             'Shard expert weights over the EP group (E-dim sharding, '
+             'reference gshard_builder.py:2269): each rank stores and '
+             'updates only E/W experts. Per-rank expert memory = '
+             'total/W; expert grads need no DP all-reduce (each rank '
+             'already sees every token routed to its experts via the '
+             'all-to-all backward).')
     return p
 
   def __init__(self, params):
@@ -132,12 +139,29 @@ class MoEFeedForwardLayer(BaseLayer):
     p = self.p
     self.CreateVariable('gate_w', py_utils.WeightParams(
         [p.input_dim, p.num_experts], p.params_init, p.dtype))
-    # All experts' weights live on every rank's checkpoint view, but each
-    # rank only computes its local shard (E/W experts).
+    world, rank, _ = self._EpWorld()
+    self._ep_shard = (p.shard_experts and world > 1 and
+                      p.num_experts % world == 0)
+    self._ep_world_at_init = world
+    e_param = p.num_experts // world if self._ep_shard else p.num_experts
     self.CreateVariable('wi', py_utils.WeightParams(
-        [p.num_experts, p.input_dim, p.hidden_dim], p.params_init, p.dtype))
+        [e_param, p.input_dim, p.hidden_dim], p.params_init, p.dtype))
     self.CreateVariable('wo', py_utils.WeightParams(
-        [p.num_experts, p.hidden_dim, p.input_dim], p.params_init, p.dtype))
+        [e_param, p.hidden_dim, p.input_dim], p.params_init, p.dtype))
+    if self._ep_shard:
+      # Initialize the local shard as the rank's slice of the FULL
+      # deterministic init so sharded EP == unsharded reference
+      # bit-for-bit (same seed stream).
+      for name, dim1 in (('wi', p.input_dim), ('wo', p.hidden_dim)):
+        g = self._InitGenerator(name)
+        dim2 = p.hidden_dim if name == 'wi' else p.input_dim
+        full = py_utils.InitWeight([p.num_experts, dim1, dim2],
+                                   p.params_init, g, p.dtype)
+        with torch.no_grad():
+          getattr(self, name).copy_(
+              full[rank * e_param:(rank + 1) * e_param])
+        # Rank-local parameter: DP gradient sync must skip it.
+        getattr(self, name)._ep_sharded = True
     self._last_aux_loss = None
 
   def _EpWorld(self):
@@ -179,6 +203,10 @@ class MoEFeedForwardLayer(BaseLayer):
     world, rank, group = self._EpWorld()
     act_fn = activations.GetFn(p.activation)
     if world > 1 and e % world == 0:
+      if self._ep_shard and world != self._ep_world_at_init:
+        raise RuntimeError(
+            f'MoE expert shards were built for EP world '
+            f'{self._ep_world_at_init}, but FProp runs at {world}')
       e_local = e // world
       # all-to-all: send expert-shard slices to their owner ranks.
       buf = dispatch.reshape(world, e_local * capacity, d).contiguous()
@@ -186,8 +214,11 @@ class MoEFeedForwardLayer(BaseLayer):
       # recv: [W, e_local*C, D] = every rank's tokens for MY experts.
       h = recv.reshape(world, e_local, capacity, d).permute(1, 0, 2, 3) \
           .reshape(e_local, world * capacity, d)
-      wi = theta.wi[rank * e_local:(rank + 1) * e_local]
-      wo = theta.wo[rank * e_local:(rank + 1) * e_local]
+      if self._ep_shard:
+        wi, wo = theta.wi, theta.wo  # already the local shard
+      else:
+        wi = theta.wi[rank * e_local:(rank + 1) * e_local]
+        wo = theta.wo[rank * e_local:(rank + 1) * e_local]
       h = act_fn(torch.bmm(h, wi))
       h = torch.bmm(h, wo)
       h = h.reshape(e_local, world, capacity, d).permute(1, 0, 2, 3) \

@@ -32,14 +32,11 @@ class GraphedTrainStep:
     task.MaybeConvertBf16Weights()
     self.task = task
     self.grad_sync = grad_sync
-    if grad_sync is not None:
-      # Replay doesn't fire python hooks; use Finalize's pull-from-grad
-      # path instead, and keep hook all-reduces out of the capture.
-      grad_sync.Close()
     self.learner = task.learners[0]
     self.opt = self.learner.EnsureOptimizer(task)
     self.loss_name = self.learner.p.loss_name
     self._seed = task.p.random_seed or 1234
+    self._sync_in_graph = False
 
     self.static_batch = example_batch.Transform(
         lambda t: t.clone() if isinstance(t, torch.Tensor) else t)
@@ -49,25 +46,50 @@ class GraphedTrainStep:
         prm.grad = torch.zeros_like(prm)
     self.grads = [prm.grad for prm in self.params]
 
-    def fwd_bwd():
+    def fwd_bwd(with_sync: bool):
       torch._foreach_zero_(self.grads)
       with py_utils.StepSeedScope(self._seed, 0):
         metrics, _ = task.FProp(task.theta, self.static_batch)
       loss = metrics[self.loss_name][0]
       loss.backward()
+      if with_sync and grad_sync is not None:
+        # Captured finalize: the bucket copies + RCCL all-reduces fired
+        # from the post-accumulate-grad hooks during this (captured)
+        # backward, overlapping comm with the rest of backward on
+        # replay exactly as in eager mode (reference deferred-CRS
+        # precedent py_utils.py:3056). Finalize drains the stragglers
+        # and writes averaged grads — all stream-ordered, so the whole
+        # sequence replays correctly.
+        grad_sync.Finalize()
       return metrics
 
-    # Warmup on a side stream (materializes workspaces/allocations).
-    side = torch.cuda.Stream()
-    side.wait_stream(torch.cuda.current_stream())
-    with torch.cuda.stream(side):
-      for _ in range(warmup_iters):
-        self.metrics = fwd_bwd()
-    torch.cuda.current_stream().wait_stream(side)
+    def warmup_and_capture(with_sync: bool):
+      side = torch.cuda.Stream()
+      side.wait_stream(torch.cuda.current_stream())
+      with torch.cuda.stream(side):
+        for _ in range(warmup_iters):
+          self.metrics = fwd_bwd(with_sync)
+      torch.cuda.current_stream().wait_stream(side)
+      graph = torch.cuda.CUDAGraph()
+      with torch.cuda.graph(graph):
+        self.metrics = fwd_bwd(with_sync)
+      return graph
 
-    self.graph = torch.cuda.CUDAGraph()
-    with torch.cuda.graph(self.graph):
-      self.metrics = fwd_bwd()
+    if grad_sync is not None:
+      try:
+        # Preferred: hooks + collectives captured INSIDE the graph
+        # (grad buffers and bucket buffers are static).
+        self.graph = warmup_and_capture(with_sync=True)
+        self._sync_in_graph = True
+      except Exception:
+        # RCCL capture unsupported on this stack: fall back to the
+        # pull-from-grad path (hooks off; Finalize launches the bucket
+        # all-reduces eagerly after each replay — correct but without
+        # backward overlap).
+        grad_sync.Close()
+        self.graph = warmup_and_capture(with_sync=False)
+    else:
+      self.graph = warmup_and_capture(with_sync=False)
 
   def Step(self, batch: NestedMap) -> NestedMap:
     task = self.task
@@ -84,7 +106,7 @@ class GraphedTrainStep:
 
     self.graph.replay()
 
-    if self.grad_sync is not None:
+    if self.grad_sync is not None and not self._sync_in_graph:
       self.grad_sync.Finalize()  # pulls from (static) param.grad in place
 
     grad_norm = py_utils.GlobalGradNorm(self.grads)
