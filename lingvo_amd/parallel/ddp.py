@@ -73,7 +73,12 @@ class GradSync:
     self._names = {id(p): n for n, p in module.named_parameters()}
     if self._world <= 1:
       return
-    params = [p for p in module.parameters() if p.requires_grad]
+    # EP-sharded expert weights are rank-local parameters: each rank
+    # already accumulates gradients from every token routed to its
+    # experts (through the all-to-all backward), so DP all-reduce would
+    # corrupt them (parallel/moe.py shard_experts).
+    params = [p for p in module.parameters()
+              if p.requires_grad and not getattr(p, '_ep_sharded', False)]
     if not params:
       return
     # bf16 buckets on RCCL (reference py_utils.py:3042 precedent); fp32 on

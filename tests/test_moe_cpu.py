@@ -56,13 +56,18 @@ def _run_ep(rank, world, port, results):
   out = layer.FProp(layer.theta, x)
   (out.sum() + layer.AuxLoss()).backward()
   results[f'out{rank}'] = out.detach()
+  results[f'wi_shape{rank}'] = tuple(layer.wi.shape)
+  results[f'wi_sharded{rank}'] = getattr(layer.wi, '_ep_sharded', False)
   results[f'wi_grad{rank}'] = layer.wi.grad.clone()
   dist.destroy_process_group()
 
 
 def test_moe_ep2_matches_local():
-  """EP=2 output on each rank == local all-expert computation on that
-  rank's tokens; expert grads accumulate across ranks' losses."""
+  """EP=2 with E-dim sharded expert weights: each rank stores E/W
+  experts (memory = total/W), outputs equal the local all-expert
+  computation on that rank's tokens, and each rank's expert grads
+  accumulate contributions from BOTH ranks' losses (via the
+  all-to-all backward) with no DP all-reduce needed."""
   ctx = mp.get_context('spawn')
   with ctx.Manager() as mgr:
     results = mgr.dict()
@@ -75,6 +80,12 @@ def test_moe_ep2_matches_local():
       assert p.exitcode == 0
     outs = {r: results[f'out{r}'] for r in range(2)}
     wigrads = {r: results[f'wi_grad{r}'] for r in range(2)}
+    shapes = {r: results[f'wi_shape{r}'] for r in range(2)}
+    sharded = {r: results[f'wi_sharded{r}'] for r in range(2)}
+
+  # Per-rank expert memory = total / EP world (2 of 4 experts).
+  assert shapes[0] == (2, 16, 32) and shapes[1] == (2, 16, 32)
+  assert sharded[0] and sharded[1]
 
   ref_grads = []
   for rank in range(2):
@@ -86,13 +97,11 @@ def test_moe_ep2_matches_local():
     assert torch.allclose(outs[rank], out.detach(), atol=1e-5), rank
     ref_grads.append(layer.wi.grad.clone())
 
-  # EP wi grad: rank r holds sum over BOTH ranks' losses for its local
-  # experts (2 experts per rank), zeros elsewhere.
+  # Sharded wi grad on rank r == the summed full-model grad rows of
+  # rank r's experts (experts 0-1 on rank 0, 2-3 on rank 1).
   total = ref_grads[0] + ref_grads[1]
-  assert torch.allclose(wigrads[0][:2], total[:2], atol=1e-5)
-  assert torch.allclose(wigrads[1][2:], total[2:], atol=1e-5)
-  assert wigrads[0][2:].abs().max() == 0
-  assert wigrads[1][:2].abs().max() == 0
+  assert torch.allclose(wigrads[0], total[:2], atol=1e-5)
+  assert torch.allclose(wigrads[1], total[2:], atol=1e-5)
 
 
 def test_moe_lm_train_step():
