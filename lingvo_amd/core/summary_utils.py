@@ -78,3 +78,110 @@ def ModelAnalysis(model: torch.nn.Module) -> str:
     lines.append(f'{name} {tuple(prm.shape)} {n}')
   lines.append(f'total #params: {total}')
   return '\n'.join(lines) + '\n'
+
+
+# ---------------------------------------------------------------------------
+# TensorBoard-compatible event files (reference summary_utils.py:42-95
+# writes TF summaries; here the events-file format itself is produced
+# natively: TFRecord framing with masked crc32c + hand-encoded Event
+# protos, so `tensorboard --logdir` works without TensorFlow installed).
+# ---------------------------------------------------------------------------
+
+_CRC_TABLE = []
+
+
+def _Crc32c(data: bytes) -> int:
+  global _CRC_TABLE
+  if not _CRC_TABLE:
+    poly = 0x82F63B78
+    for n in range(256):
+      c = n
+      for _ in range(8):
+        c = (c >> 1) ^ poly if c & 1 else c >> 1
+      _CRC_TABLE.append(c)
+  crc = 0xFFFFFFFF
+  for b in data:
+    crc = _CRC_TABLE[(crc ^ b) & 0xFF] ^ (crc >> 8)
+  return crc ^ 0xFFFFFFFF
+
+
+def _MaskedCrc(data: bytes) -> int:
+  crc = _Crc32c(data)
+  return (((crc >> 15) | (crc << 17)) + 0xA282EAD8) & 0xFFFFFFFF
+
+
+def _Varint(n: int) -> bytes:
+  out = b''
+  while True:
+    b = n & 0x7F
+    n >>= 7
+    if n:
+      out += bytes([b | 0x80])
+    else:
+      return out + bytes([b])
+
+
+def _Field(num: int, wire: int) -> bytes:
+  return _Varint((num << 3) | wire)
+
+
+def _EncodeEvent(wall_time: float, step: int = 0,
+                 file_version: Optional[str] = None,
+                 scalars: Optional[Dict[str, float]] = None) -> bytes:
+  """Minimal tensorflow.Event proto encoder (event.proto wire format)."""
+  import struct as _struct
+  out = _Field(1, 1) + _struct.pack('<d', wall_time)     # wall_time
+  if step:
+    out += _Field(2, 0) + _Varint(step & 0xFFFFFFFFFFFFFFFF)
+  if file_version is not None:
+    fv = file_version.encode()
+    out += _Field(3, 2) + _Varint(len(fv)) + fv
+  if scalars:
+    summ = b''
+    for tag, value in scalars.items():
+      tb = tag.encode()
+      val = (_Field(1, 2) + _Varint(len(tb)) + tb +       # Value.tag
+             _Field(2, 5) + _struct.pack('<f', value))    # simple_value
+      summ += _Field(1, 2) + _Varint(len(val)) + val      # Summary.value
+    out += _Field(5, 2) + _Varint(len(summ)) + summ       # Event.summary
+  return out
+
+
+class TbEventWriter:
+  """Writes TensorBoard events files: `events.out.tfevents.<ts>.<host>`
+  in the given directory, TFRecord-framed Event protos with valid
+  masked crc32c. Scalars only (the breadth TensorBoard actually needs
+  for training curves)."""
+
+  def __init__(self, logdir: str):
+    os.makedirs(logdir, exist_ok=True)
+    import socket
+    ts = int(time.time())
+    host = socket.gethostname()
+    self._path = os.path.join(logdir, f'events.out.tfevents.{ts}.{host}')
+    self._f = open(self._path, 'ab')
+    self._WriteRecord(_EncodeEvent(time.time(),
+                                   file_version='brain.Event:2'))
+
+  def _WriteRecord(self, data: bytes) -> None:
+    import struct as _struct
+    header = _struct.pack('<Q', len(data))
+    self._f.write(header)
+    self._f.write(_struct.pack('<I', _MaskedCrc(header)))
+    self._f.write(data)
+    self._f.write(_struct.pack('<I', _MaskedCrc(data)))
+    self._f.flush()
+
+  def scalar(self, tag: str, value, step: int) -> None:
+    if isinstance(value, torch.Tensor):
+      value = float(value.detach().cpu())
+    self._WriteRecord(_EncodeEvent(time.time(), step,
+                                   scalars={tag: float(value)}))
+
+  def scalars(self, values: Dict[str, float], step: int) -> None:
+    vals = {k: (float(v.detach().cpu()) if isinstance(v, torch.Tensor)
+                else float(v)) for k, v in values.items()}
+    self._WriteRecord(_EncodeEvent(time.time(), step, scalars=vals))
+
+  def close(self) -> None:
+    self._f.close()

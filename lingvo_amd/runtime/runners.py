@@ -115,6 +115,7 @@ class Trainer(BaseRunner):
     self._grad_sync = grad_sync
     self._tracker = StepRateTracker()
     self._metrics_log = os.path.join(logdir, 'train', 'metrics.jsonl')
+    self._tb = None  # lazily created TensorBoard events writer
 
   def Start(self) -> None:
     self._RunLoop(self._Loop)
@@ -153,6 +154,14 @@ class Trainer(BaseRunner):
           f.write(json.dumps({
               'step': step, 'loss': loss,
               'steps_per_sec': round(self._tracker.steps_per_sec, 4)}) + '\n')
+        if self._tb is None:
+          from lingvo_amd.core.summary_utils import TbEventWriter
+          self._tb = TbEventWriter(os.path.join(self._logdir, 'train'))
+        self._tb.scalars(
+            {'loss': loss,
+             'steps_per_sec': self._tracker.steps_per_sec,
+             'examples_per_sec': self._tracker.examples_per_sec},
+            step)
       ckpt.MaybeSave()
     ckpt.Save()
     ckpt.Sync()
@@ -261,10 +270,13 @@ class Evaler(_CheckpointPoller):
     avg = py_utils.WeightedAvgOfMetrics(agg)
     out_dir = os.path.join(self._logdir, f'eval_{self._dataset.lower()}')
     os.makedirs(out_dir, exist_ok=True)
+    vals = {k: py_utils.ToScalar(v[0]) for k, v in avg.items()}
     with open(os.path.join(out_dir, 'metrics.jsonl'), 'a') as f:
-      f.write(json.dumps(
-          {'step': step,
-           **{k: py_utils.ToScalar(v[0]) for k, v in avg.items()}}) + '\n')
+      f.write(json.dumps({'step': step, **vals}) + '\n')
+    if not hasattr(self, '_tb') or self._tb is None:
+      from lingvo_amd.core.summary_utils import TbEventWriter
+      self._tb = TbEventWriter(out_dir)
+    self._tb.scalars(vals, step)
     task.train()
 
 

@@ -605,3 +605,31 @@ def test_split_input_batch():
   assert total == batch.data.shape[0]
   recon = torch.cat([p2.data for p2 in parts])
   assert torch.equal(recon, batch.data)
+
+
+def test_tb_event_writer_format(tmp_path):
+  """TensorBoard events file: TFRecord framing with valid masked
+  crc32c + parseable Event protos (file_version, step, simple_value)."""
+  import struct
+  from lingvo_amd.core import summary_utils as su
+  w = su.TbEventWriter(str(tmp_path))
+  w.scalar('loss', 2.5, 7)
+  w.close()
+  path = [p for p in tmp_path.iterdir()
+          if p.name.startswith('events.out.tfevents')][0]
+  data = path.read_bytes()
+  off, recs = 0, []
+  while off < len(data):
+    (ln,) = struct.unpack('<Q', data[off:off + 8])
+    assert struct.unpack('<I', data[off + 8:off + 12])[0] == \
+        su._MaskedCrc(data[off:off + 8])
+    rec = data[off + 12:off + 12 + ln]
+    assert struct.unpack('<I', data[off + 12 + ln:off + 16 + ln])[0] == \
+        su._MaskedCrc(rec)
+    recs.append(rec)
+    off += 16 + ln
+  assert len(recs) == 2
+  assert b'brain.Event:2' in recs[0]
+  assert b'loss' in recs[1]
+  # simple_value float 2.5 little-endian appears in the scalar record
+  assert struct.pack('<f', 2.5) in recs[1]
