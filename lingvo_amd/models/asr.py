@@ -271,6 +271,17 @@ class AsrDecoder(BaseLayer):
     fgb = self.rnns[0].p.forget_gate_bias
     use_fused = enc.is_cuda and dt == torch.bfloat16
 
+    if use_fused:
+      # Whole-recurrence fused path (ops/las_decoder.py): per-step
+      # small-M MFMA GEMMs + fused attention kernel, wgrads batched
+      # over the L steps. Identical math to the loop below.
+      from lingvo_amd.ops import las_decoder
+      out = las_decoder.decoder_recurrence(
+          emb_gates, w_cm0, b1, wm1, wq, enc,
+          enc_paddings, fgb, cap if cap is not None else 0.0,
+          p.source_dim)
+      return NestedMap(atten_vecs=out)
+
     def lstm_pointwise(gates, c_prev):
       if use_fused:
         from lingvo_amd.ops import lstm_gates as lstm_ops

@@ -88,6 +88,18 @@ void RegisterWpmTokenizer(py::module_& m);
 // record_batcher.cpp
 void RegisterRecordBatcher(py::module_& m);
 
+// las_decoder.hip
+void smallm_gemm(torch::Tensor a, torch::Tensor wt,
+                 c10::optional<torch::Tensor> pre, torch::Tensor out,
+                 int64_t k, int64_t wt_col0, double alpha,
+                 int64_t pre_mode);
+std::vector<torch::Tensor> attend_fwd(torch::Tensor q, torch::Tensor enc,
+                                      torch::Tensor pad, double scale);
+std::vector<torch::Tensor> attend_bwd(torch::Tensor dctx,
+                                      torch::Tensor probs, torch::Tensor q,
+                                      torch::Tensor enc, torch::Tensor denc,
+                                      double scale);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   RegisterInputPipeline(m);
   RegisterWpmTokenizer(m);
@@ -107,4 +119,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe", &mfma_probe, "MFMA 16x16x32 layout probe");
   m.def("fa_fwd", &fa_fwd, "Flash attention forward");
   m.def("fa_bwd", &fa_bwd, "Flash attention backward");
+  m.def("smallm_gemm", &smallm_gemm,
+        "Small-M MFMA GEMM (decode-step projections)");
+  m.def("attend_fwd", &attend_fwd, "Fused dot-attention step fwd");
+  m.def("attend_bwd", &attend_bwd, "Fused dot-attention step bwd");
 }
