@@ -20,8 +20,9 @@ from lingvo_amd.core.base_layer import BaseLayer
 from lingvo_amd.core.nested_map import NestedMap
 from lingvo_amd.layers import layers as lingvo_layers
 from lingvo_amd.layers import transformer as transformer_lib
-from lingvo_amd.parallel.pipeline import (GPipeRunner,
-                                          PartitionSequentialLayers)
+from lingvo_amd.parallel.pipeline import (GPipeRunner, PartitionByCost,
+                                          SoftmaxFlops,
+                                          TransformerLayerFlops)
 
 
 class TransformerLmStage(BaseLayer):
@@ -43,8 +44,15 @@ class TransformerLmStage(BaseLayer):
   def __init__(self, params):
     super().__init__(params)
     p = self.p
-    parts = PartitionSequentialLayers(list(range(p.num_layers_total)),
-                                      p.num_stages)
+    # Cost-balanced stage partition: the last stage also carries the
+    # softmax projection, so it takes fewer transformer layers
+    # (reference FPropMeta-driven partitioning, gpipe.py:339-375).
+    hidden = p.hidden_dim or 4 * p.model_dim
+    layer_cost = TransformerLayerFlops(p.model_dim, hidden)
+    parts = PartitionByCost([layer_cost] * p.num_layers_total,
+                            p.num_stages,
+                            extra_last=SoftmaxFlops(p.model_dim,
+                                                    p.vocab_size))
     self._my_layers = parts[p.stage_idx]
     self.is_first = p.stage_idx == 0
     self.is_last = p.stage_idx == p.num_stages - 1

@@ -289,3 +289,74 @@ def PartitionSequentialLayers(layer_params: List, num_stages: int
     out.append(layer_params[idx:idx + take])
     idx += take
   return out
+
+
+def PartitionByCost(costs: List[float], num_stages: int,
+                    extra_first: float = 0.0,
+                    extra_last: float = 0.0) -> List[List[int]]:
+  """Contiguous partition of len(costs) units over num_stages stages
+  minimizing the max per-stage cost (the reference partitions stages by
+  FPropMeta flop estimates, gpipe.py:339-375; this is the explicit
+  min-max DP over measured/analytic costs).
+
+  extra_first/extra_last: fixed cost carried by stage 0 / the last
+  stage (embedding table, final softmax) so those stages receive
+  correspondingly fewer layers. Returns index lists per stage."""
+  n = len(costs)
+  k = min(num_stages, n) if n else num_stages
+  if n == 0:
+    return [[] for _ in range(num_stages)]
+  prefix = [0.0]
+  for c in costs:
+    prefix.append(prefix[-1] + c)
+
+  def seg(i, j):  # cost of units [i, j)
+    return prefix[j] - prefix[i]
+
+  INF = float('inf')
+  # dp[s][j] = min over partitions of first j units into s stages of the
+  # max stage cost; track split points.
+  dp = [[INF] * (n + 1) for _ in range(k + 1)]
+  cut = [[0] * (n + 1) for _ in range(k + 1)]
+  dp[0][0] = 0.0
+  for s in range(1, k + 1):
+    for j in range(s, n + 1):
+      # stage s-1 covers units [i, j)
+      for i in range(s - 1, j):
+        load = seg(i, j)
+        if s == 1:
+          load += extra_first
+        if s == k:
+          load += extra_last
+        # (a single stage carrying both extras handled by both adds)
+        cand = max(dp[s - 1][i], load)
+        if cand < dp[s][j]:
+          dp[s][j] = cand
+          cut[s][j] = i
+  # Recover ranges.
+  bounds = [n]
+  j = n
+  for s in range(k, 0, -1):
+    j = cut[s][j]
+    bounds.append(j)
+  bounds.reverse()
+  out = [list(range(bounds[s], bounds[s + 1])) for s in range(k)]
+  while len(out) < num_stages:
+    out.append([])
+  return out
+
+
+def TransformerLayerFlops(model_dim: int, hidden_dim: int,
+                          seq_len: int = 1) -> float:
+  """Per-token matmul flops of one transformer layer (QKV/out
+  projections + FFN + O(T) attention term) — the analytic FPropMeta
+  stand-in feeding PartitionByCost."""
+  proj = 8 * model_dim * model_dim
+  ffn = 4 * model_dim * hidden_dim
+  attn = 4 * model_dim * seq_len
+  return float(proj + ffn + attn)
+
+
+def SoftmaxFlops(model_dim: int, vocab_size: int) -> float:
+  """Per-token flops of the full-softmax projection."""
+  return float(2 * model_dim * vocab_size)
