@@ -71,8 +71,7 @@ class LConvLayer(BaseLayer):
     p = self.p
     x = self.ln.FProp(theta.ln, inputs)
     x = py_utils.MatmulBias(x, theta.pw1_w, theta.pw1_b)
-    a, b = x.chunk(2, dim=-1)
-    x = a * torch.sigmoid(b)  # GLU
+    x = F.glu(x, dim=-1)  # fused GLU (single kernel fwd + bwd)
     if paddings is not None:
       x = py_utils.ApplyPadding(paddings, x)
     x = conv1d_ops.depthwise_conv1d(x, theta.dw_w, theta.dw_b,
@@ -106,8 +105,7 @@ class LConvLayer(BaseLayer):
     assert p.is_causal, 'StreamStep requires a causal LConv'
     x = self.ln.FProp(theta.ln, inputs)
     x = py_utils.MatmulBias(x, theta.pw1_w, theta.pw1_b)
-    a, b = x.chunk(2, dim=-1)
-    x = a * torch.sigmoid(b)
+    x = F.glu(x, dim=-1)
     x = py_utils.ApplyPadding(paddings, x)
     # Depthwise causal conv over [state | chunk].
     full = torch.cat([state.conv.to(x.dtype), x], dim=1)
