@@ -361,7 +361,7 @@ __global__ __launch_bounds__(NWB * WAVE_SIZE) void fa_bwd_kernel(
     float* __restrict__ dq_acc,
     unsigned short* __restrict__ dk, unsigned short* __restrict__ dv,
     float* __restrict__ dbias, int B, int T, int S, int N, int NKV,
-    int win_l, int win_r, int bias_clip, float scale) {
+    int win_l, int win_r, int bias_clip, float scale, int skip) {
   constexpr int ROWB = H * 2;
   constexpr int KH = H / 32;
   constexpr int HF = H / 16;
@@ -439,15 +439,17 @@ __global__ __launch_bounds__(NWB * WAVE_SIZE) void fa_bwd_kernel(
       stage_regular<H, QT, NWB * WAVE_SIZE>(
           q + (((long)b * T + qb) * N + n) * H, (long)N * H, T - qb,
           q_lds);
-      stage_transposed<H, QT, NWB * WAVE_SIZE>(
-          q + (((long)b * T + qb) * N + n) * H, (long)N * H, T - qb,
-          qt_lds);
       stage_regular<H, QT, NWB * WAVE_SIZE>(
           dout + (((long)b * T + qb) * N + n) * H, (long)N * H, T - qb,
           do_lds);
-      stage_transposed<H, QT, NWB * WAVE_SIZE>(
-          dout + (((long)b * T + qb) * N + n) * H, (long)N * H, T - qb,
-          dot_lds);
+      if (!(skip & 2)) {
+        stage_transposed<H, QT, NWB * WAVE_SIZE>(
+            q + (((long)b * T + qb) * N + n) * H, (long)N * H, T - qb,
+            qt_lds);
+        stage_transposed<H, QT, NWB * WAVE_SIZE>(
+            dout + (((long)b * T + qb) * N + n) * H, (long)N * H, T - qb,
+            dot_lds);
+      }
       for (int i = threadIdx.x; i < QT; i += NWB * WAVE_SIZE) {
         int qrow = qb + i;
         lse_s[i] = qrow < T ? lse[((long)b * N + n) * T + qrow] : NEG_INF;
@@ -562,6 +564,7 @@ __global__ __launch_bounds__(NWB * WAVE_SIZE) void fa_bwd_kernel(
       }
 
       // Full dS tile to LDS (transposed store: ds_lds[q][key]) for dQ.
+      if (!(skip & 4))
 #pragma unroll
       for (int nf = 0; nf < 4; ++nf) {
 #pragma unroll
@@ -578,7 +581,7 @@ __global__ __launch_bounds__(NWB * WAVE_SIZE) void fa_bwd_kernel(
       // dQ strips: rows q = qb + dw*16 + .. (A from ds_lds), B = K from
       // kt_lds; atomic-accumulate fp32. Only QT/16 strips exist, so with
       // NWB > QT/16 the extra waves skip this phase.
-      if (wid < QT / 16) {
+      if (wid < QT / 16 && !(skip & 1)) {
         f32x4 acc_dq[HF];
 #pragma unroll
         for (int hf = 0; hf < HF; ++hf) acc_dq[hf] = {0.f, 0.f, 0.f, 0.f};
@@ -803,6 +806,13 @@ std::vector<torch::Tensor> fa_bwd(torch::Tensor dout, torch::Tensor q,
   const int* ksp = kseg.has_value() ? kseg->data_ptr<int>() : nullptr;
   bool bg = bias.has_value() && bias_grad;
 
+  // Phase-skip bitmask for perf attribution only (results invalid when
+  // nonzero): 1 = skip dQ math+atomics, 2 = skip transposed q/do
+  // staging, 4 = skip dS-full store.
+  static int skip_phases = []() {
+    const char* e = getenv("LINGVO_FA_BWD_SKIP");
+    return e ? atoi(e) : 0;
+  }();
   // 128-key tiles with 8 waves (q-tile staging amortized 2x). H=128
   // uses 144KB LDS -> 1 block/CU; staging savings outweigh occupancy.
   const int ktb = 128;
@@ -826,7 +836,7 @@ std::vector<torch::Tensor> fa_bwd(torch::Tensor dout, torch::Tensor q,
       dq_acc.data_ptr<float>(),                                              \
       (unsigned short*)dk_t.data_ptr(), (unsigned short*)dv_t.data_ptr(),    \
       dbias_t.numel() ? dbias_t.data_ptr<float>() : nullptr, B, T, S, N,     \
-      NKV, (int)win_l, (int)win_r, (int)bias_clip, (float)scale)
+      NKV, (int)win_l, (int)win_r, (int)bias_clip, (float)scale, skip_phases)
   if (H == 64) {
     if (bg) FA_BWD(64, true); else FA_BWD(64, false);
   } else {

@@ -43,19 +43,36 @@ __global__ __launch_bounds__(256) void smallm_gemm_kernel(
   const int g = lane >> 4;            // k-group 0..3
   const int cl = lane & 15;           // row-in-frag / col-in-frag
 
-  f32x4 acc[2] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
-  const int col = col0 + cl;
-  const bool col_ok = col < n;
+  // Double-buffered LDS staging of the Wt tile [16 cols][32 k]:
+  // coalesced global reads (16 threads x 4B per column row) instead of
+  // 16 lanes each issuing a strided 16B read.
+  constexpr int ROWP = 40;  // padded row stride in shorts (80 B)
+  __shared__ unsigned short wtile[2][16 * ROWP];
+  const int tid = threadIdx.x;
+  const int ld_row = tid / 16;        // 0..15 (wt column)
+  const int ld_off = (tid % 16) * 2;  // shorts within the 32-k row
 
-  for (int k0 = 0; k0 < k; k0 += 32) {
-    // B fragment: Wt[col][k0 + g*8 .. +8] (8 contiguous K elems).
-    bf16x8 bf;
-    if (col_ok) {
-      bf = load_bf16x8_bits(wt + (long)col * ldw + k0 + g * 8);
-    } else {
-      float z[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-      bf = pack_bf16x8(z);
+  auto stage = [&](int buf, int k0) {
+    const int col = col0 + ld_row;
+    unsigned int v = 0;
+    if (col < n && k0 + ld_off < k) {
+      v = *reinterpret_cast<const unsigned int*>(
+          wt + (long)col * ldw + k0 + ld_off);
     }
+    *reinterpret_cast<unsigned int*>(
+        &wtile[buf][ld_row * ROWP + ld_off]) = v;
+  };
+
+  f32x4 acc[2] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+
+  stage(0, 0);
+  __syncthreads();
+  int cur = 0;
+  for (int k0 = 0; k0 < k; k0 += 32) {
+    if (k0 + 32 < k) {
+      stage(1 - cur, k0 + 32);
+    }
+    bf16x8 bf = load_bf16x8_bits(&wtile[cur][cl * ROWP + g * 8]);
 #pragma unroll
     for (int mt = 0; mt < 2; ++mt) {
       const int row = wid * 32 + mt * 16 + cl;
@@ -68,7 +85,11 @@ __global__ __launch_bounds__(256) void smallm_gemm_kernel(
       }
       acc[mt] = mfma16x16x32_bf16(af, bf, acc[mt]);
     }
+    __syncthreads();
+    cur = 1 - cur;
   }
+  const bool col_ok = col0 + cl < n;
+  const int col = col0 + cl;
 
   // Epilogue: C layout row = g*4+r, col = cl.
 #pragma unroll
@@ -116,15 +137,30 @@ __global__ __launch_bounds__(256) void attend_fwd_kernel(
   for (int i = threadIdx.x; i < d; i += 256) q_s[i] = q[(long)b * d + i];
   __syncthreads();
 
-  // Pass 1: logits, one s per wave-iteration.
+  // Cache this lane's q slice in registers (vectorized over the s loop).
+  constexpr int MAXDV = 4;  // supports d up to 2048
+  const int ndv = d / (WAVE_SIZE * 8) + (d % (WAVE_SIZE * 8) ? 1 : 0);
+  float qreg[MAXDV][8];
+  for (int v = 0; v < ndv; ++v) {
+    const int i = lane * 8 + v * WAVE_SIZE * 8;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      qreg[v][j] = i < d ? bf16_bits_to_float(q_s[i + j]) : 0.f;
+    }
+  }
+
+  // Pass 1: logits, one s per wave-iteration (vector loads of enc).
   const unsigned short* eb = enc + (long)b * s_len * d;
   for (int s = wid; s < s_len; s += 4) {
     float part = 0.f;
-    for (int i = lane * 8; i < d; i += WAVE_SIZE * 8) {
+    for (int v = 0; v < ndv; ++v) {
+      const int i = lane * 8 + v * WAVE_SIZE * 8;
+      if (i < d) {
+        bf16x8 ev = load_bf16x8_bits(eb + (long)s * d + i);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        part += bf16_bits_to_float(q_s[i + j]) *
-                bf16_bits_to_float(eb[(long)s * d + i + j]);
+        for (int j = 0; j < 8; ++j) {
+          part += qreg[v][j] * (float)ev[j];
+        }
       }
     }
     part = wave_reduce_sum(part);
@@ -164,10 +200,14 @@ __global__ __launch_bounds__(256) void attend_fwd_kernel(
   for (int s = wid; s < s_len; s += 4) {
     const float p = logit_s[s] * inv;
     if (lane == 0) probs[(long)b * s_len + s] = p;
-    for (int i = lane * 8; i < d; i += WAVE_SIZE * 8) {
+    for (int v = 0; v < ndv; ++v) {
+      const int i = lane * 8 + v * WAVE_SIZE * 8;
+      if (i < d) {
+        bf16x8 ev = load_bf16x8_bits(eb + (long)s * d + i);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        my_ctx[i + j] += p * bf16_bits_to_float(eb[(long)s * d + i + j]);
+        for (int j = 0; j < 8; ++j) {
+          my_ctx[i + j] += p * (float)ev[j];
+        }
       }
     }
   }
@@ -207,16 +247,32 @@ __global__ __launch_bounds__(256) void attend_bwd_kernel(
   }
   __syncthreads();
 
+  // Per-lane register caches of dctx and q slices.
+  constexpr int MAXDV = 4;
+  const int ndv = d / (WAVE_SIZE * 8) + (d % (WAVE_SIZE * 8) ? 1 : 0);
+  float dcreg[MAXDV][8], qreg[MAXDV][8];
+  for (int v = 0; v < ndv; ++v) {
+    const int i = lane * 8 + v * WAVE_SIZE * 8;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      dcreg[v][j] = i < d ? bf16_bits_to_float(dctx_s[i + j]) : 0.f;
+      qreg[v][j] = i < d ? bf16_bits_to_float(q_s[i + j]) : 0.f;
+    }
+  }
+
   // Pass 1: dprobs (stored to dl_s) + t reduction.
   const unsigned short* eb = enc + (long)b * s_len * d;
   const float* pb = probs + (long)b * s_len;
   for (int s = wid; s < s_len; s += 4) {
     float part = 0.f;
-    for (int i = lane * 8; i < d; i += WAVE_SIZE * 8) {
+    for (int v = 0; v < ndv; ++v) {
+      const int i = lane * 8 + v * WAVE_SIZE * 8;
+      if (i < d) {
+        bf16x8 ev = load_bf16x8_bits(eb + (long)s * d + i);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        part += bf16_bits_to_float(dctx_s[i + j]) *
-                bf16_bits_to_float(eb[(long)s * d + i + j]);
+        for (int j = 0; j < 8; ++j) {
+          part += dcreg[v][j] * (float)ev[j];
+        }
       }
     }
     part = wave_reduce_sum(part);
@@ -247,14 +303,24 @@ __global__ __launch_bounds__(256) void attend_bwd_kernel(
   for (int s = wid; s < s_len; s += 4) {
     const float dl = dl_s[s];
     const float p = pb[s];
-    for (int i = lane * 8; i < d; i += WAVE_SIZE * 8) {
+    for (int v = 0; v < ndv; ++v) {
+      const int i = lane * 8 + v * WAVE_SIZE * 8;
+      if (i < d) {
+        bf16x8 ev = load_bf16x8_bits(eb + (long)s * d + i);
+        floatx4* dbv = reinterpret_cast<floatx4*>(db + (long)s * d + i);
+        floatx4 lo = dbv[0], hi = dbv[1];
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const float ev = bf16_bits_to_float(eb[(long)s * d + i + j]);
-        my_dq[i + j] += dl * ev;
-        db[(long)s * d + i + j] +=
-            dl * bf16_bits_to_float(q_s[i + j]) +
-            p * bf16_bits_to_float(dctx_s[i + j]);
+        for (int j = 0; j < 4; ++j) {
+          my_dq[i + j] += dl * (float)ev[j];
+          lo[j] += dl * qreg[v][j] + p * dcreg[v][j];
+        }
+#pragma unroll
+        for (int j = 4; j < 8; ++j) {
+          my_dq[i + j] += dl * (float)ev[j];
+          hi[j - 4] += dl * qreg[v][j] + p * dcreg[v][j];
+        }
+        dbv[0] = lo;
+        dbv[1] = hi;
       }
     }
   }

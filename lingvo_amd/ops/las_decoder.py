@@ -48,14 +48,15 @@ class _DecoderRecurrenceFn(torch.autograd.Function):
     gates0 = torch.empty(L, B, G4, dtype=dt, device=dev)
     gates1 = torch.empty(L, B, G4, dtype=dt, device=dev)
     qs = torch.empty(L, B, D, dtype=dt, device=dev)
+    # One [L,B,4H] copy up front instead of a per-step .contiguous().
+    eg_t = emb_gates_b.permute(1, 0, 2).contiguous()
     c0s, m0s, c1s, m1s, ctxs, probs_l = [], [], [], [], [], []
     zeros_h = torch.zeros(B, H, dtype=dt, device=dev)
     c0, m0, c1, m1 = zeros_h, zeros_h, zeros_h, zeros_h
     ctxv = torch.zeros(B, D, dtype=dt, device=dev)
     for t in range(L):
       g0 = gates0[t]
-      ext.smallm_gemm(ctxv, w1t, emb_gates_b[:, t].contiguous(), g0,
-                      src_dim, 0, 1.0, 2)
+      ext.smallm_gemm(ctxv, w1t, eg_t[t], g0, src_dim, 0, 1.0, 2)
       ext.smallm_gemm(m0, w1t, None, g0, H, src_dim, 1.0, 3)
       c0, m0 = ext.lstm_gates_fwd(g0, c0, fgb, cap)
       g1 = gates1[t]
@@ -114,12 +115,15 @@ class _DecoderRecurrenceFn(torch.autograd.Function):
     dmm = torch.empty(B, 2 * H, dtype=dt, device=dev)
     dcm = torch.empty(B, src_dim + H, dtype=dt, device=dev)
     dm1_total = torch.empty(B, H, dtype=dt, device=dev)
+    dout_t = dout.permute(1, 0, 2)
+    dm1_parts = dout_t[:, :, :H].contiguous()   # [L, B, H]
+    dctx_parts = dout_t[:, :, H:].contiguous()  # [L, B, D]
     for t in range(L - 1, -1, -1):
-      dctx_total = (dout[:, t, H:] + dctx_carry).contiguous()
+      dctx_total = dctx_parts[t] + dctx_carry
       dq = ext.attend_bwd(dctx_total, probs[t], qs[t], enc_b, denc,
                           scale)[0]
       dqs[t] = dq
-      pre_m1 = (dout[:, t, :H] + dm1_carry).contiguous()
+      pre_m1 = dm1_parts[t] + dm1_carry
       # dm1_total = dq @ wq^T + (out grad + carry)
       ext.smallm_gemm(dq, wq_b, pre_m1, dm1_total, D, 0, 1.0, 2)
       c1_prev = c1s[t - 1] if t > 0 else zeros_h
