@@ -210,10 +210,10 @@ class EmbeddingLayer(BaseLayer):
         p.dtype))
 
   def EmbLookup(self, theta: NestedMap, ids: torch.Tensor) -> torch.Tensor:
-    out = F.embedding(ids, theta.wm)
-    if self.p.scale_sqrt_depth:
-      out = out * (self.p.embedding_dim ** 0.5)
-    return out
+    scale = (self.p.embedding_dim ** 0.5 if self.p.scale_sqrt_depth
+             else 1.0)
+    from lingvo_amd.ops import embedding as emb_ops
+    return emb_ops.embedding_lookup(theta.wm, ids, scale)
 
   def FProp(self, theta: NestedMap, ids: torch.Tensor) -> torch.Tensor:
     return self.EmbLookup(theta, ids)
@@ -362,6 +362,60 @@ class DropoutLayer(BaseLayer):
     return py_utils.DeterministicDropout(inputs, self.p.keep_prob)
 
 
+class _ChunkedLogitsXent(torch.autograd.Function):
+  """Streaming vocab-chunked logits+xent: peak extra memory is
+  [R, chunk] instead of [R, V] in both directions."""
+
+  @staticmethod
+  def forward(ctx, x, w, b, labels, chunk):
+    xf = x.float()
+    r = xf.shape[0]
+    v = w.shape[1]
+    run_max = torch.full((r,), -1e30, device=x.device)
+    run_sum = torch.zeros(r, device=x.device)
+    tgt = torch.zeros(r, device=x.device)
+    for c0 in range(0, v, chunk):
+      c1 = min(v, c0 + chunk)
+      logits = torch.addmm(b[c0:c1].float(), xf, w[:, c0:c1].float())
+      m = logits.max(dim=1).values
+      new_max = torch.maximum(run_max, m)
+      run_sum = run_sum * torch.exp(run_max - new_max) + \
+          torch.exp(logits - new_max.unsqueeze(1)).sum(dim=1)
+      run_max = new_max
+      in_chunk = (labels >= c0) & (labels < c1)
+      if bool(in_chunk.any()):
+        idx = in_chunk.nonzero(as_tuple=True)[0]
+        tgt[idx] = logits[idx, labels[idx] - c0]
+    lse = run_max + torch.log(run_sum)
+    ctx.save_for_backward(x, w, b, labels, lse)
+    ctx.chunk = chunk
+    return lse - tgt
+
+  @staticmethod
+  def backward(ctx, dy):
+    x, w, b, labels, lse = ctx.saved_tensors
+    chunk = ctx.chunk
+    xf = x.float()
+    v = w.shape[1]
+    dyf = dy.float().unsqueeze(1)
+    dx = torch.zeros_like(xf)
+    dw = torch.zeros_like(w, dtype=torch.float32)
+    db = torch.zeros(v, device=x.device)
+    for c0 in range(0, v, chunk):
+      c1 = min(v, c0 + chunk)
+      logits = torch.addmm(b[c0:c1].float(), xf, w[:, c0:c1].float())
+      p = torch.exp(logits - lse.unsqueeze(1))
+      in_chunk = (labels >= c0) & (labels < c1)
+      if bool(in_chunk.any()):
+        idx = in_chunk.nonzero(as_tuple=True)[0]
+        p[idx, labels[idx] - c0] -= 1.0
+      p = p * dyf
+      dx += p @ w[:, c0:c1].float().t()
+      dw[:, c0:c1] = xf.t() @ p
+      db[c0:c1] = p.sum(dim=0)
+    return (dx.to(x.dtype), dw.to(w.dtype), db.to(b.dtype), None, None)
+
+
 class SimpleFullSoftmax(BaseLayer):
   """Softmax + cross-entropy over a full vocab (reference layers.py:3697).
 
@@ -442,6 +496,14 @@ class SimpleFullSoftmax(BaseLayer):
       from lingvo_amd.ops import softmax_xent
       per_example = softmax_xent.logits_xent(
           inputs2d, theta.linear_w, theta.bias, class_ids.reshape(-1))
+    elif p.chunk_size > 0 and class_probabilities is None:
+      # Vocab-chunked xent (reference SimpleFullSoftmax
+      # softmax_max_alloc chunking, layers.py:3697): never materializes
+      # [R, V]; fwd streams a running logsumexp over vocab chunks, bwd
+      # recomputes each chunk's logits.
+      per_example = _ChunkedLogitsXent.apply(
+          inputs2d, theta.linear_w, theta.bias,
+          class_ids.reshape(-1).long(), p.chunk_size)
     else:
       logits = self.Logits(theta, inputs2d)
       per_example = self.XentLossFromLogits(

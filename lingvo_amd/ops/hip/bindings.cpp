@@ -88,6 +88,12 @@ void RegisterWpmTokenizer(py::module_& m);
 // record_batcher.cpp
 void RegisterRecordBatcher(py::module_& m);
 
+// embedding.hip
+torch::Tensor emb_gather(torch::Tensor table, torch::Tensor ids,
+                         double scale);
+torch::Tensor emb_scatter_add(torch::Tensor dy, torch::Tensor ids,
+                              int64_t vocab, double scale);
+
 // las_decoder.hip
 void smallm_gemm(torch::Tensor a, torch::Tensor wt,
                  c10::optional<torch::Tensor> pre, torch::Tensor out,
@@ -119,6 +125,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe", &mfma_probe, "MFMA 16x16x32 layout probe");
   m.def("fa_fwd", &fa_fwd, "Flash attention forward");
   m.def("fa_bwd", &fa_bwd, "Flash attention backward");
+  m.def("emb_gather", &emb_gather, "Embedding gather fwd");
+  m.def("emb_scatter_add", &emb_scatter_add,
+        "Deterministic embedding scatter-add bwd");
   m.def("smallm_gemm", &smallm_gemm,
         "Small-M MFMA GEMM (decode-step projections)");
   m.def("attend_fwd", &attend_fwd, "Fused dot-attention step fwd");

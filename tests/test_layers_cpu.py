@@ -645,3 +645,28 @@ def test_gru_and_sru_cells():
                      NestedMap(act=x.detach(),
                                padding=torch.ones(2, 1)))
     assert torch.allclose(st2.m, st.m, atol=1e-6)
+
+
+def test_chunked_softmax_matches_full():
+  """chunk_size>0 streaming xent == full-logits path (fwd + grads)."""
+  from lingvo_amd.layers import layers as lingvo_layers
+  torch.manual_seed(4)
+
+  def run(chunk):
+    torch.manual_seed(4)
+    p = lingvo_layers.SimpleFullSoftmax.Params().Set(
+        name='sm', input_dim=24, num_classes=50, chunk_size=chunk,
+        random_seed=7)
+    sm = p.Instantiate()
+    x = torch.randn(6, 3, 24, requires_grad=True)
+    ids = torch.randint(0, 50, (6, 3))
+    wts = torch.ones(6, 3)
+    out = sm.XentLoss(sm.theta, x, class_weights=wts, class_ids=ids)
+    out.total_xent.backward()
+    return (out.per_example_xent.detach(), x.grad.clone(),
+            sm.linear_w.grad.clone(), sm.bias.grad.clone())
+
+  full = run(0)
+  chunked = run(16)  # 50 classes in chunks of 16 (ragged tail)
+  for a, b in zip(full, chunked):
+    assert torch.allclose(a, b, atol=1e-4), (a - b).abs().max()
