@@ -190,3 +190,77 @@ class WpmTokenizer(VocabFileTokenizer):
   def _IdsToTokens(self, ids: Sequence[int]) -> str:
     s = ''.join(self._inv.get(t, '?') for t in ids)
     return s.replace(self.WORD_MARK, ' ').strip()
+
+
+class BpeTokenizer(BaseTokenizer):
+  """Byte-pair-encoding tokenizer (reference core/tokenizers.py:305
+  BpeTokenizer + BpeWordsToIds / BpeIdsToWords ops, x_ops.cc:613-849).
+
+  `codes`: merge rules in priority order, each 'a b' (the pair merged
+  into 'ab'); words end with the '</w>' marker. `tokens`: subword ->
+  id vocabulary list (index = id). Common words can be pre-cached via
+  `words_to_ids` {word: [ids]} (the reference's words_to_ids file)."""
+
+  WORD_END = '</w>'
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('codes', [], "Merge rules ['a b', ...] in priority order.")
+    p.Define('codes_filepath', None, 'File of merge rules (one per line).')
+    p.Define('tokens', [], 'Subword vocab (index = id).')
+    p.Define('vocab_filepath', None, 'File of subwords (one per line).')
+    p.Define('words_to_ids', {}, 'Optional pre-tokenized word cache.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    codes = list(p.codes)
+    if p.codes_filepath:
+      with open(p.codes_filepath) as f:
+        codes = [ln.rstrip('\n') for ln in f
+                 if ln.strip() and not ln.startswith('#')]
+    self._ranks = {}
+    for i, line in enumerate(codes):
+      a, b = line.split()
+      self._ranks[(a, b)] = i
+    tokens = list(p.tokens)
+    if p.vocab_filepath:
+      with open(p.vocab_filepath) as f:
+        tokens = [ln.rstrip('\n') for ln in f if ln.rstrip('\n')]
+    self._vocab = {t: i for i, t in enumerate(tokens)}
+    self._inv = {i: t for t, i in self._vocab.items()}
+    self._cache = {w: list(ids) for w, ids in p.words_to_ids.items()}
+
+  def _BpeWord(self, word: str) -> List[str]:
+    pieces = list(word[:-1]) + [word[-1] + self.WORD_END]
+    while len(pieces) > 1:
+      best, best_rank = None, None
+      for i in range(len(pieces) - 1):
+        r = self._ranks.get((pieces[i], pieces[i + 1]))
+        if r is not None and (best_rank is None or r < best_rank):
+          best, best_rank = i, r
+      if best is None:
+        break
+      pieces = (pieces[:best] + [pieces[best] + pieces[best + 1]] +
+                pieces[best + 2:])
+    return pieces
+
+  def _TokensToIds(self, text: str) -> List[int]:
+    unk = self.p.target_unk_id
+    out: List[int] = []
+    for w in text.split():
+      if not w:
+        continue
+      cached = self._cache.get(w)
+      if cached is None:
+        cached = [self._vocab.get(piece, unk)
+                  for piece in self._BpeWord(w)]
+        self._cache[w] = cached
+      out.extend(cached)
+    return out
+
+  def _IdsToTokens(self, ids: Sequence[int]) -> str:
+    s = ''.join(self._inv.get(t, '?') for t in ids)
+    return s.replace(self.WORD_END, ' ').strip()
