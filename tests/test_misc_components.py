@@ -662,5 +662,5 @@ def test_bpe_tokenizer_roundtrip():
   tok2 = p2.Instantiate()
   assert tok2._TokensToIds('low') == [3]
   # StringsToIds padding contract
-  out = tok.StringsToIds(['low', 'lower low'], max_length=6)
-  assert out.ids.shape == (2, 6)
+  ids_t = tok.StringsToIds(['low', 'lower low'], max_length=6)[0]
+  assert ids_t.shape == (2, 6)
