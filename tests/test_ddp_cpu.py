@@ -344,3 +344,56 @@ def test_gradsync_compressor_hook_path():
     results = dict(results)
   assert torch.allclose(results['g0'], results['g1'], atol=1e-6)
   assert results['res0'].abs().sum() > 0  # something was dropped
+
+
+def _run_syncbn(rank, world, port, results):
+  os.environ['MASTER_ADDR'] = '127.0.0.1'
+  os.environ['MASTER_PORT'] = str(port)
+  dist.init_process_group('gloo', rank=rank, world_size=world)
+  from lingvo_amd.layers import bn_layers
+  p = bn_layers.BatchNormLayer.Params().Set(
+      name='bn', dim=8, enable_cross_replica_sum_on_tpu=True,
+      random_seed=1)
+  bn = p.Instantiate()
+  torch.manual_seed(100 + rank)  # different data per rank
+  x = torch.randn(3, 5, 8) * (rank + 1) + rank  # different mean/var
+  pad = torch.zeros(3, 5)
+  pad[0, 3:] = 1.0
+  out = bn.FProp(bn.theta, x, pad)
+  results[f'out{rank}'] = out.detach()
+  results[f'x{rank}'] = x
+  results[f'pad{rank}'] = pad
+  dist.destroy_process_group()
+
+
+def test_sync_batch_norm_uses_global_moments():
+  """Cross-replica BN: each rank normalizes with GLOBAL moments
+  (sufficient-statistics all-reduce, reference bn_layers.py:139
+  cross-replica option)."""
+  port = dist_port(29561)
+  ctx = mp.get_context('spawn')
+  with ctx.Manager() as mgr:
+    results = mgr.dict()
+    procs = [ctx.Process(target=_run_syncbn, args=(r, 2, port, results))
+             for r in range(2)]
+    for p in procs:
+      p.start()
+    for p in procs:
+      p.join(120)
+      assert p.exitcode == 0
+    outs = {r: results[f'out{r}'] for r in range(2)}
+    xs = {r: results[f'x{r}'] for r in range(2)}
+    pads = {r: results[f'pad{r}'] for r in range(2)}
+
+  # Global moments over BOTH ranks' valid frames.
+  mask = torch.cat([(1 - pads[0]).reshape(-1), (1 - pads[1]).reshape(-1)])
+  allx = torch.cat([xs[0].reshape(-1, 8), xs[1].reshape(-1, 8)])
+  m = mask.unsqueeze(1)
+  count = mask.sum()
+  mean = (allx * m).sum(0) / count
+  var = ((allx - mean) ** 2 * m).sum(0) / count
+  want0 = (xs[0] - mean) * torch.rsqrt(var + 1e-3)
+  want0 = want0 * (1 - pads[0]).unsqueeze(-1)
+  got = outs[0]  # gamma=0 init => scale (1+0)=1, beta 0
+  assert torch.allclose(got, want0.to(got.dtype), atol=1e-3), \
+      (got - want0).abs().max()
