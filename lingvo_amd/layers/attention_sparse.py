@@ -46,75 +46,27 @@ class ChunkwiseSelfAttention(attention_lib.MultiHeadedAttention):
   def FProp(self, theta: NestedMap, query_vec: torch.Tensor,
             paddings: Optional[torch.Tensor] = None,
             segment_ids: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Runs the chunk mask NATIVELY inside the flash kernel (chunk_size/
+    left_chunks args): no widened K/V copies, and key tiles outside the
+    chunk window are skipped in-kernel — O(T*W) work on GPU and CPU."""
     p = self.p
     assert segment_ids is None, 'packed inputs: use the flash seg path'
     b, t, _ = query_vec.shape
-    w = p.chunk_size
-    pad_t = (-t) % w
-    x = F.pad(query_vec, (0, 0, 0, pad_t))
-    pads = F.pad(paddings, (0, pad_t), value=1.0) if paddings is not None \
-        else F.pad(torch.zeros(b, t, device=x.device), (0, pad_t),
-                   value=1.0)
-    q, k, v = self._Project(theta, x)
+    q, k, v = self._Project(theta, query_vec)
     if p.use_rope:
       q = self.rope.FProp(theta.rope, q)
       k = self.rope.FProp(theta.rope, k)
-    nc = x.shape[1] // w
-    n, nkv, h = self._n, self._nkv, self._h
-    # fold chunks into batch: [B*nc, W(+left), ...]
-    qc = q.reshape(b, nc, w, n, h).reshape(b * nc, w, n, h)
-    lc = p.left_chunks
-
-    def widen(tensor, nch):
-      tc = tensor.reshape(b, nc, w, nch, h)
-      parts = []
-      for d in range(lc, 0, -1):
-        shifted = F.pad(tc[:, :-d], (0, 0, 0, 0, 0, 0, d, 0))
-        parts.append(shifted)
-      parts.append(tc)
-      return torch.cat(parts, dim=2).reshape(b * nc, (lc + 1) * w, nch, h)
-
-    kc = widen(k, nkv)
-    vc = widen(v, nkv)
-    padc = pads.reshape(b, nc, w)
-    wide_pads = []
-    for d in range(lc, 0, -1):
-      wide_pads.append(F.pad(padc[:, :-d], (0, 0, d, 0), value=1.0))
-    wide_pads.append(padc)
-    padw = torch.cat(wide_pads, dim=2).reshape(b * nc, (lc + 1) * w)
-    # mask via segment ids: 1 for valid positions, query always seg 1
-    q_seg = (1.0 - padc.reshape(b * nc, w)).long()
-    k_seg = (1.0 - padw).long()
-    # avoid all-masked query rows attending nothing: padded queries get
-    # seg 2 (matches nothing, output 0 — flash zeroes those rows).
-    q_seg = torch.where(q_seg > 0, q_seg, torch.full_like(q_seg, 2))
-    if p.causal:
-      out = self._CausalChunk(qc, kc, vc, q_seg, k_seg, lc, w)
-    else:
-      out = flash_attn.flash_attention(
-          qc, kc, vc, None, None, -1, -1,
-          q_segment_ids=q_seg, k_segment_ids=k_seg)
-    out = out.reshape(b, nc * w, n * h)[:, :t]
+    klen = (py_utils.LengthsFromPaddings(paddings).to(torch.int32)
+            if paddings is not None else None)
+    out = flash_attn.flash_attention(
+        q, k, v, klen=klen, win_l=-1, win_r=0 if p.causal else -1,
+        chunk_size=p.chunk_size, left_chunks=p.left_chunks)
+    out = out.reshape(b, t, self._n * self._h)
     post = py_utils.MatmulBias(out, theta.post_w,
                                theta.post_b if p.use_bias else None)
     if paddings is not None:
       post = py_utils.ApplyPadding(paddings, post)
     return post
-
-  def _CausalChunk(self, qc, kc, vc, q_seg, k_seg, lc, w):
-    # query local row i sits at widened position lc*w + i: win_r=0 in
-    # that coordinate == win_l=-1, win_r = lc*w applied with the query
-    # offset. flash's windows are relative (k <= q + win_r) with q,k in
-    # the SAME coordinates, so shift queries by prepending nothing and
-    # using win_r = lc*w - 0... simpler: pad queries on the left to
-    # align coordinates, then slice.
-    pad_q = lc * w
-    qp = F.pad(qc, (0, 0, 0, 0, pad_q, 0))
-    qs = F.pad(q_seg, (pad_q, 0), value=3)  # pad rows match nothing
-    out = flash_attn.flash_attention(
-        qp, kc, vc, None, None, -1, 0,
-        q_segment_ids=qs, k_segment_ids=k_seg)
-    return out[:, pad_q:]
 
 
 class RoutingAttention(attention_lib.MultiHeadedAttention):

@@ -23,7 +23,8 @@ from lingvo_amd.ops import _loader
 
 
 def _ref_attention(q, k, v, klen, bias, win_l, win_r, bias_clip, scale,
-                   q_segment_ids=None, k_segment_ids=None):
+                   q_segment_ids=None, k_segment_ids=None,
+                   chunk_size=0, left_chunks=0):
   """fp32 reference with identical semantics (any backend)."""
   B, T, N, H = q.shape
   S, NKV = k.shape[1], k.shape[2]
@@ -45,6 +46,9 @@ def _ref_attention(q, k, v, klen, bias, win_l, win_r, bias_clip, scale,
     mask &= kpos >= qpos - win_l
   if win_r >= 0:
     mask &= kpos <= qpos + win_r
+  if chunk_size > 0:
+    qc, kc = qpos // chunk_size, kpos // chunk_size
+    mask &= (kc <= qc) & (kc >= qc - left_chunks)
   mask = mask[None, None]
   if klen is not None:
     mask = mask & (kpos[None, None] < klen[:, None, None, None])
@@ -64,11 +68,11 @@ class _FlashAttnFn(torch.autograd.Function):
 
   @staticmethod
   def forward(ctx, q, k, v, klen, bias, qseg, kseg, win_l, win_r,
-              bias_clip, scale):
+              bias_clip, scale, chunk_size, left_chunks):
     ext = _loader.get_ext(required=True)
     bias_b = None if bias is None else bias.to(torch.bfloat16).contiguous()
     o, lse = ext.fa_fwd(q, k, v, klen, bias_b, qseg, kseg, win_l, win_r,
-                        bias_clip, scale)
+                        bias_clip, scale, chunk_size, left_chunks)
     empty = torch.empty(0)
     ctx.save_for_backward(q, k, v, o, lse,
                           klen if klen is not None else empty,
@@ -76,24 +80,27 @@ class _FlashAttnFn(torch.autograd.Function):
                           qseg if qseg is not None else empty,
                           kseg if kseg is not None else empty)
     ctx.cfg = (win_l, win_r, bias_clip, scale, bias is not None and
-               bias.requires_grad, None if bias is None else bias.dtype)
+               bias.requires_grad, None if bias is None else bias.dtype,
+               chunk_size, left_chunks)
     return o
 
   @staticmethod
   def backward(ctx, dout):
     ext = _loader.get_ext(required=True)
     q, k, v, o, lse, klen, bias_b, qseg, kseg = ctx.saved_tensors
-    win_l, win_r, bias_clip, scale, bias_grad, bias_dtype = ctx.cfg
+    (win_l, win_r, bias_clip, scale, bias_grad, bias_dtype, chunk_size,
+     left_chunks) = ctx.cfg
     klen = klen if klen.numel() else None
     bias_b = bias_b if bias_b.numel() else None
     qseg = qseg if qseg.numel() else None
     kseg = kseg if kseg.numel() else None
     dq, dk, dv, dbias = ext.fa_bwd(
         dout.contiguous(), q, k, v, o, lse, klen, bias_b, qseg, kseg,
-        bias_grad, win_l, win_r, bias_clip, scale)
+        bias_grad, win_l, win_r, bias_clip, scale, chunk_size,
+        left_chunks)
     dbias_out = dbias.to(bias_dtype) if bias_grad else None
     return (dq, dk, dv, None, dbias_out, None, None, None, None, None,
-            None)
+            None, None, None)
 
 
 def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
@@ -103,7 +110,8 @@ def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                     bias_clip: int = 127,
                     scale: Optional[float] = None,
                     q_segment_ids: Optional[torch.Tensor] = None,
-                    k_segment_ids: Optional[torch.Tensor] = None
+                    k_segment_ids: Optional[torch.Tensor] = None,
+                    chunk_size: int = 0, left_chunks: int = 0
                     ) -> torch.Tensor:
   """q [B,T,N,H], k/v [B,S,NKV,H] -> [B,T,N,H]. Optional packed-input
   segment ids [B,T]/[B,S]: attention is blocked across segments
@@ -120,8 +128,10 @@ def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     out = _FlashAttnFn.apply(
         q.to(torch.bfloat16).contiguous(), k.to(torch.bfloat16).contiguous(),
         v.to(torch.bfloat16).contiguous(), klen, bias, q_segment_ids,
-        k_segment_ids, win_l, win_r, bias_clip, scale)
+        k_segment_ids, win_l, win_r, bias_clip, scale, chunk_size,
+        left_chunks)
     return out.to(orig) if orig != torch.bfloat16 else out
   out = _ref_attention(q, k, v, klen, bias, win_l, win_r, bias_clip, scale,
-                       q_segment_ids, k_segment_ids)
+                       q_segment_ids, k_segment_ids, chunk_size,
+                       left_chunks)
   return out.to(q.dtype)

@@ -108,10 +108,17 @@ __device__ __forceinline__ bf16x8 lds_frag(const char* tile, int row,
 }
 
 __device__ __forceinline__ bool visible(int q, int k, int klen, int win_l,
-                                        int win_r) {
+                                        int win_r, int chunk, int lc) {
   if (k >= klen) return false;
   if (win_l >= 0 && k < q - win_l) return false;
   if (win_r >= 0 && k > q + win_r) return false;
+  if (chunk > 0) {
+    // Chunkwise mask (reference ChunkwiseSelfAttention,
+    // batch_major_attention.py:4008): keys visible iff their chunk is
+    // within [q_chunk - lc, q_chunk].
+    const int qc = q / chunk, kc = k / chunk;
+    if (kc > qc || kc < qc - lc) return false;
+  }
   return true;
 }
 
@@ -131,7 +138,8 @@ __global__ __launch_bounds__(FWD_BLOCK) void fa_fwd_kernel(
     const int* __restrict__ qseg,              // [B][T] or null (packed)
     const int* __restrict__ kseg,              // [B][S] or null
     unsigned short* __restrict__ o, float* __restrict__ lse, int B, int T,
-    int S, int N, int NKV, int win_l, int win_r, int bias_clip, float scale) {
+    int S, int N, int NKV, int win_l, int win_r, int bias_clip, float scale,
+    int chunk, int lc) {
   constexpr int ROWB = H * 2;
   constexpr int KH = H / 32;   // mfma K-steps over head dim
   constexpr int HF = H / 16;   // output col frags
@@ -182,6 +190,14 @@ __global__ __launch_bounds__(FWD_BLOCK) void fa_fwd_kernel(
   int kmax_excl =
       min(klen, win_r < 0 ? S : min(S, qt * FQT + FQT - 1 + win_r + 1));
   int kmin = win_l < 0 ? 0 : max(0, qt * FQT - win_l);
+  if (chunk > 0) {
+    // Skip key tiles entirely outside the q-tile's chunk window.
+    const int qc_lo = (qt * FQT) / chunk;
+    const int qc_hi = (qt * FQT + FQT - 1) / chunk;
+    kmin = max(kmin, max(0, (qc_lo - lc)) * chunk);
+    kmax_excl = min(kmax_excl, (qc_hi + 1) * chunk);
+    if (kmax_excl <= kmin) return;
+  }
   const int kt_lo = kmin / KTF;
   const int kt_hi = (max(kmax_excl, 1) - 1) / KTF;
 
@@ -226,7 +242,7 @@ __global__ __launch_bounds__(FWD_BLOCK) void fa_fwd_kernel(
           val += bf16_bits_to_float(
               bias[(long)n * (2 * bias_clip + 1) + d + bias_clip]);
         }
-        if (!visible(qrow, kcol, klen, win_l, win_r) || qrow >= T)
+        if (!visible(qrow, kcol, klen, win_l, win_r, chunk, lc) || qrow >= T)
           val = NEG_INF;
         if (SEG) {
           if (kcol < S && qrow < T &&
@@ -361,7 +377,8 @@ __global__ __launch_bounds__(NWB * WAVE_SIZE) void fa_bwd_kernel(
     float* __restrict__ dq_acc,
     unsigned short* __restrict__ dk, unsigned short* __restrict__ dv,
     float* __restrict__ dbias, int B, int T, int S, int N, int NKV,
-    int win_l, int win_r, int bias_clip, float scale, int skip) {
+    int win_l, int win_r, int bias_clip, float scale, int skip,
+    int chunk, int lc) {
   constexpr int ROWB = H * 2;
   constexpr int KH = H / 32;
   constexpr int HF = H / 16;
@@ -417,6 +434,13 @@ __global__ __launch_bounds__(NWB * WAVE_SIZE) void fa_bwd_kernel(
   // q-tile range for this key tile.
   int qlo = win_r < 0 ? 0 : max(0, kbase - win_r);
   int qhi = win_l < 0 ? T - 1 : min(T - 1, kbase + KTB - 1 + win_l);
+  if (chunk > 0) {
+    const int kc_lo = kbase / chunk;
+    const int kc_hi = (min(kbase + KTB, S) - 1) / chunk;
+    qlo = max(qlo, kc_lo * chunk);
+    qhi = min(qhi, (kc_hi + lc + 1) * chunk - 1);
+    if (qhi < qlo) return;
+  }
   const int qt_lo = qlo / QT, qt_hi = max(qhi, 0) / QT;
 
   for (int head = 0; head < group; ++head) {
@@ -479,7 +503,7 @@ __global__ __launch_bounds__(NWB * WAVE_SIZE) void fa_bwd_kernel(
             d = d < -bias_clip ? -bias_clip : (d > bias_clip ? bias_clip : d);
             val += bf16_bits_to_float(bias[(long)n * nbias + d + bias_clip]);
           }
-          bool vis = visible(qcol, key, klen, win_l, win_r) && qcol < T;
+          bool vis = visible(qcol, key, klen, win_l, win_r, chunk, lc) && qcol < T;
           if (vis && qseg &&
               qseg[(long)b * T + qcol] != kseg[(long)b * S + key])
             vis = false;
@@ -673,7 +697,8 @@ std::vector<torch::Tensor> fa_fwd(torch::Tensor q, torch::Tensor k,
                                   c10::optional<torch::Tensor> qseg,
                                   c10::optional<torch::Tensor> kseg,
                                   int64_t win_l, int64_t win_r,
-                                  int64_t bias_clip, double scale) {
+                                  int64_t bias_clip, double scale,
+                                  int64_t chunk_size, int64_t left_chunks) {
   check_btnh(q, "q");
   check_btnh(k, "k");
   check_btnh(v, "v");
@@ -705,7 +730,7 @@ std::vector<torch::Tensor> fa_fwd(torch::Tensor q, torch::Tensor k,
                      (const unsigned short*)v.data_ptr(), klp, bp, qsp,     \
                      ksp, (unsigned short*)o.data_ptr(),                    \
                      lse.data_ptr<float>(), B, T, S, N, NKV, (int)win_l,    \
-                     (int)win_r, (int)bias_clip, (float)scale)
+                     (int)win_r, (int)bias_clip, (float)scale, (int)chunk_size, (int)left_chunks)
 #define FA_FWD64(HH)                                                        \
   hipLaunchKernelGGL((fa_fwd_kernel<HH, 64, false>), grid,                  \
                      dim3(FWD_BLOCK),                                       \
@@ -714,7 +739,7 @@ std::vector<torch::Tensor> fa_fwd(torch::Tensor q, torch::Tensor k,
                      (const unsigned short*)v.data_ptr(), klp, bp, qsp,     \
                      ksp, (unsigned short*)o.data_ptr(),                    \
                      lse.data_ptr<float>(), B, T, S, N, NKV, (int)win_l,    \
-                     (int)win_r, (int)bias_clip, (float)scale)
+                     (int)win_r, (int)bias_clip, (float)scale, (int)chunk_size, (int)left_chunks)
 #define FA_FWD_SEG(HH)                                                      \
   if (ktf == 128)                                                           \
     hipLaunchKernelGGL((fa_fwd_kernel<HH, 128, true>), grid,                \
@@ -724,7 +749,7 @@ std::vector<torch::Tensor> fa_fwd(torch::Tensor q, torch::Tensor k,
                        (const unsigned short*)v.data_ptr(), klp, bp, qsp,   \
                        ksp, (unsigned short*)o.data_ptr(),                  \
                        lse.data_ptr<float>(), B, T, S, N, NKV, (int)win_l,  \
-                       (int)win_r, (int)bias_clip, (float)scale);           \
+                       (int)win_r, (int)bias_clip, (float)scale, (int)chunk_size, (int)left_chunks);           \
   else                                                                      \
     hipLaunchKernelGGL((fa_fwd_kernel<HH, 64, true>), grid,                 \
                        dim3(FWD_BLOCK), shmem, stream,                      \
@@ -733,7 +758,7 @@ std::vector<torch::Tensor> fa_fwd(torch::Tensor q, torch::Tensor k,
                        (const unsigned short*)v.data_ptr(), klp, bp, qsp,   \
                        ksp, (unsigned short*)o.data_ptr(),                  \
                        lse.data_ptr<float>(), B, T, S, N, NKV, (int)win_l,  \
-                       (int)win_r, (int)bias_clip, (float)scale)
+                       (int)win_r, (int)bias_clip, (float)scale, (int)chunk_size, (int)left_chunks)
   if (qsp != nullptr) {
     if (H == 64) {
       FA_FWD_SEG(64);
@@ -762,7 +787,8 @@ std::vector<torch::Tensor> fa_bwd(torch::Tensor dout, torch::Tensor q,
                                   c10::optional<torch::Tensor> kseg,
                                   bool bias_grad, int64_t win_l,
                                   int64_t win_r, int64_t bias_clip,
-                                  double scale) {
+                                  double scale, int64_t chunk_size,
+                                  int64_t left_chunks) {
   check_btnh(q, "q");
   const int B = q.size(0), T = q.size(1), N = q.size(2), H = q.size(3);
   const int S = k.size(1), NKV = k.size(2);
@@ -836,7 +862,7 @@ std::vector<torch::Tensor> fa_bwd(torch::Tensor dout, torch::Tensor q,
       dq_acc.data_ptr<float>(),                                              \
       (unsigned short*)dk_t.data_ptr(), (unsigned short*)dv_t.data_ptr(),    \
       dbias_t.numel() ? dbias_t.data_ptr<float>() : nullptr, B, T, S, N,     \
-      NKV, (int)win_l, (int)win_r, (int)bias_clip, (float)scale, skip_phases)
+      NKV, (int)win_l, (int)win_r, (int)bias_clip, (float)scale, skip_phases, (int)chunk_size, (int)left_chunks)
   if (H == 64) {
     if (bg) FA_BWD(64, true); else FA_BWD(64, false);
   } else {

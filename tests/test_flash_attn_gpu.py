@@ -174,3 +174,41 @@ def test_flash_segment_mask_fwd_bwd():
   out2 = fa.flash_attention(q.detach(), k2, v.detach(), win_r=0,
                             q_segment_ids=seg, k_segment_ids=seg)
   assert torch.equal(out[:, :40].detach(), out2[:, :40])
+
+
+@gpu
+def test_flash_chunk_mask_matches_ref():
+  """Native chunk-mask mode vs the fp32 reference (fwd + grads)."""
+  from lingvo_amd.ops import flash_attn as fa
+  torch.manual_seed(6)
+  B, T, N, H = 3, 200, 2, 64
+  for chunk, lc, causal in [(64, 0, False), (64, 1, True), (48, 2, False)]:
+    q = torch.randn(B, T, N, H, device='cuda',
+                    dtype=torch.bfloat16).requires_grad_()
+    k = torch.randn_like(q).requires_grad_()
+    v = torch.randn_like(q).requires_grad_()
+    klen = torch.tensor([200, 150, 90], device='cuda',
+                        dtype=torch.int32)
+    win_r = 0 if causal else -1
+    out = fa.flash_attention(q, k, v, klen=klen, win_r=win_r,
+                             chunk_size=chunk, left_chunks=lc)
+    g = torch.randn_like(out)
+    out.backward(g)
+    grads = [q.grad.float().clone(), k.grad.float().clone(),
+             v.grad.float().clone()]
+    q.grad = k.grad = v.grad = None
+
+    ref = fa._ref_attention(q.detach().float().requires_grad_(),
+                            k.detach().float().requires_grad_(),
+                            v.detach().float().requires_grad_(),
+                            klen, None, -1, win_r, 127, H ** -0.5,
+                            chunk_size=chunk, left_chunks=lc)
+    assert (out.float() - ref.detach()).abs().max() < 0.05, (chunk, lc)
+    qr = q.detach().float().requires_grad_()
+    kr = k.detach().float().requires_grad_()
+    vr = v.detach().float().requires_grad_()
+    fa._ref_attention(qr, kr, vr, klen, None, -1, win_r, 127,
+                      H ** -0.5, chunk_size=chunk,
+                      left_chunks=lc).backward(g.float())
+    for got, want in zip(grads, [qr.grad, kr.grad, vr.grad]):
+      assert (got - want).abs().max() < 0.1, (chunk, lc)

@@ -64,14 +64,14 @@ def main():
   flops_dir = 4.0 * B * N * T * T * H  # QK^T + PV (2 GEMMs, 2 flops/MAC)
 
   o, lse = ext.fa_fwd(q, k, v, klen, bias, None, None, -1, -1, clip,
-                      scale)
+                      scale, 0, 0)
   dout = torch.randn_like(o)
 
   t_fwd = bench(lambda: ext.fa_fwd(q, k, v, klen, bias, None, None, -1,
-                                   -1, clip, scale), args.iters)
+                                   -1, clip, scale, 0, 0), args.iters)
   t_bwd = bench(lambda: ext.fa_bwd(dout, q, k, v, o, lse, klen, bias,
                                    None, None, bias is not None, -1, -1,
-                                   clip, scale), args.iters)
+                                   clip, scale, 0, 0), args.iters)
   print(f'shape={args.shape} B={B} T={T} N={N} H={H} '
         f'bias={"clip" + str(clip) if bias is not None else "none"}')
   print(f'fwd: {t_fwd * 1e3:8.3f} ms  {flops_dir / t_fwd / 1e12:7.1f} TF/s')
