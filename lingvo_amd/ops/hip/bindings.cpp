@@ -24,7 +24,8 @@ std::vector<torch::Tensor> fa_fwd(torch::Tensor q, torch::Tensor k,
                                   c10::optional<torch::Tensor> qseg,
                                   c10::optional<torch::Tensor> kseg,
                                   int64_t win_l, int64_t win_r,
-                                  int64_t bias_clip, double scale);
+                                  int64_t bias_clip, double scale,
+                                  int64_t chunk_size, int64_t left_chunks);
 std::vector<torch::Tensor> fa_bwd(torch::Tensor dout, torch::Tensor q,
                                   torch::Tensor k, torch::Tensor v,
                                   torch::Tensor o, torch::Tensor lse,
@@ -34,7 +35,8 @@ std::vector<torch::Tensor> fa_bwd(torch::Tensor dout, torch::Tensor q,
                                   c10::optional<torch::Tensor> kseg,
                                   bool bias_grad, int64_t win_l,
                                   int64_t win_r, int64_t bias_clip,
-                                  double scale);
+                                  double scale, int64_t chunk_size,
+                                  int64_t left_chunks);
 
 // conv1d.hip
 torch::Tensor dwconv1d_fwd(torch::Tensor x, torch::Tensor w,
