@@ -88,3 +88,76 @@ class BaseSequenceInputGenerator(BaseInputGenerator):
 
   def scaled_bucket_batch_limit(self) -> List[int]:
     return list(self.p.bucket_batch_limit)
+
+
+class BaseInputGeneratorFromFiles(BaseSequenceInputGenerator):
+  """File-pattern inputs with within-batch mixing (reference
+  base_input_generator.py:1216 BaseInputGeneratorFromFiles).
+
+  `file_pattern` is either one 'format:glob' string or a weighted list
+  [(pattern, weight), ...]: records are then mixed EXAMPLE-level (every
+  batch contains a weighted mix, unlike CrossBatchMixingDataSource which
+  alternates whole batches). Subclasses implement ProcessRecord()
+  (record bytes -> (NestedMap example, bucket key))."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('file_pattern', '', 'Pattern or [(pattern, weight), ...].')
+    p.Define('file_random_seed', 301, 'Shuffle seed.')
+    p.Define('file_buffer_size', 10000, 'Shuffle buffer records.')
+    p.Define('file_parallelism', 4, 'Reader threads per pattern.')
+    p.Define('num_batcher_threads', 2, 'Processor threads.')
+    p.Define('repeat', True, 'Loop forever (False = one epoch).')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    self._pipeline = None
+
+  def ProcessRecord(self, record: bytes):
+    """record -> (NestedMap example, int bucket_key) or None to drop."""
+    raise NotImplementedError
+
+  def _BuildPipeline(self):
+    from lingvo_amd.core import generic_input
+    from lingvo_amd.ops import _loader
+    p = self.p
+    ext = _loader.get_ext(required=True)
+    patterns = p.file_pattern
+    if isinstance(patterns, str):
+      patterns = [(patterns, 1.0)]
+    yielders = []
+    weights = []
+    for pattern, w in patterns:
+      fmt, files = generic_input.ExpandFilePattern(pattern)
+      yielders.append(ext.RecordYielder(
+          files, fmt, p.file_random_seed, p.file_buffer_size,
+          p.file_parallelism, p.repeat))
+      weights.append(w)
+    if len(yielders) == 1:
+      yielder = yielders[0]
+    else:
+      # Within-batch mixing: record-level weighted sampling feeds ONE
+      # batcher, so every batch mixes sources by weight.
+      yielder = generic_input.WeightedMixYielder(
+          yielders, weights, seed=p.file_random_seed)
+    bounds = p.bucket_upper_bound or [2 ** 30]
+    limits = p.bucket_batch_limit or [p.batch_size]
+    return generic_input.RecordBatcher(
+        yielder, lambda rec: self.ProcessRecord(rec), bounds, limits,
+        num_threads=p.num_batcher_threads)
+
+  def _InputBatch(self) -> NestedMap:
+    if self._pipeline is None:
+      self._pipeline = self._BuildPipeline()
+    batch = self._pipeline.GetNext()
+    if batch is None:
+      raise StopIteration
+    return batch
+
+  def Reset(self) -> None:
+    super().Reset()
+    if self._pipeline is not None:
+      self._pipeline.Stop()
+      self._pipeline = None

@@ -89,3 +89,100 @@ class CurriculumDataSource(DataSource):
       if self._step >= b:
         idx += 1
     return self.sources[idx].GetNext()
+
+
+class SequentialDataSource(DataSource):
+  """Runs sub-sources to exhaustion in order (reference
+  sequential_record_yielder + datasource chaining): source i+1 starts
+  when source i raises StopIteration; raises StopIteration after the
+  last (one pass)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('sub', [], 'Sub-datasource params, consumed in order.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    self.CreateChildren('sources', [sp.Copy() for sp in self.p.sub])
+    self._idx = 0
+
+  def GetNext(self) -> NestedMap:
+    while self._idx < len(self.sources):
+      try:
+        batch = self.sources[self._idx].GetNext()
+        batch.source_id = self._idx
+        return batch
+      except StopIteration:
+        self._idx += 1
+    raise StopIteration
+
+  def Reset(self) -> None:
+    self._idx = 0
+    for s in self.sources:
+      s.Reset()
+
+
+class WithinBatchMixingDataSource(DataSource):
+  """Example-level mixing: each output batch contains rows drawn from
+  the sub-sources in (expected) weight proportion (reference
+  base_input_generator.py:1216 within-batch mixing semantics, lifted to
+  the datasource layer). Sub-batches must be structurally compatible."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('sub', [], 'Sub-datasource params.')
+    p.Define('weights', [], 'Mixing weights.')
+    p.Define('batch_size', 0, 'Rows per mixed batch (0 = first sub '
+             'batch size).')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    assert len(self.p.sub) == len(self.p.weights)
+    self.CreateChildren('sources', [sp.Copy() for sp in self.p.sub])
+    self._rng = random.Random(self.p.random_seed or 301)
+    self._pools: List[Optional[NestedMap]] = [None] * len(self.p.sub)
+    self._cursor = [0] * len(self.p.sub)
+
+  def _Row(self, i: int) -> NestedMap:
+    import torch
+    if self._pools[i] is None or self._cursor[i] >= next(
+        v.shape[0] for v in self._pools[i].Flatten()
+        if isinstance(v, torch.Tensor)):
+      self._pools[i] = self.sources[i].GetNext()
+      self._cursor[i] = 0
+    r = self._cursor[i]
+    self._cursor[i] += 1
+    return self._pools[i].Transform(
+        lambda t: t[r:r + 1] if isinstance(t, torch.Tensor) else t)
+
+  def GetNext(self) -> NestedMap:
+    import torch
+    n = self.p.batch_size
+    if not n:
+      if self._pools[0] is None:
+        self._pools[0] = self.sources[0].GetNext()
+        self._cursor[0] = 0
+      n = next(v.shape[0] for v in self._pools[0].Flatten()
+               if isinstance(v, torch.Tensor))
+    rows = []
+    srcs = []
+    for _ in range(n):
+      i = self._rng.choices(range(len(self.sources)),
+                            weights=self.p.weights, k=1)[0]
+      rows.append(self._Row(i))
+      srcs.append(i)
+    flat = [r.Flatten() for r in rows]
+    out_vals = []
+    for j in range(len(flat[0])):
+      vals = [f[j] for f in flat]
+      if isinstance(vals[0], torch.Tensor):
+        out_vals.append(torch.cat(vals, dim=0))
+      else:
+        out_vals.append(vals[0])
+    out = rows[0].Pack(out_vals)
+    out.source_id = torch.tensor(srcs)
+    return out

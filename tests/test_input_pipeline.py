@@ -250,3 +250,107 @@ def test_wpm_vocab_builder_roundtrip(tmp_path):
     assert tok._IdsToTokens(ids) == line
   # frequent words merged to single pieces
   assert '▁the' in vocab
+
+
+def test_within_batch_mixing_datasource():
+  import torch
+  from lingvo_amd.core import datasource as ds
+  from lingvo_amd.core.base_input_generator import BaseInputGenerator
+  from lingvo_amd.core.nested_map import NestedMap
+
+  class Const(BaseInputGenerator):
+    @classmethod
+    def Params(cls):
+      p = super().Params()
+      p.Define('value', 0, 'Row fill value.')
+      return p
+
+    def _InputBatch(self):
+      return NestedMap(x=torch.full((4, 3), float(self.p.value)))
+
+  def src(v):
+    return ds.SimpleDataSource.Params().Set(
+        input_generator=Const.Params().Set(value=v, name=f'g{v}'))
+
+  p = ds.WithinBatchMixingDataSource.Params().Set(
+      name='mix', sub=[src(1), src(2)], weights=[0.8, 0.2],
+      batch_size=16, random_seed=7)
+  mix = p.Instantiate()
+  counts = {1: 0, 2: 0}
+  for _ in range(20):
+    b = mix.GetNext()
+    assert b.x.shape == (16, 3)
+    for r in range(16):
+      counts[int(b.x[r, 0])] += 1
+    # source_id bookkeeping matches row contents
+    assert all(int(b.x[r, 0]) == int(b.source_id[r]) + 1
+               for r in range(16))
+  total = counts[1] + counts[2]
+  assert 0.7 < counts[1] / total < 0.9  # ~0.8 mixing ratio
+
+
+def test_sequential_datasource_epochs():
+  import torch
+  from lingvo_amd.core import datasource as ds
+  from lingvo_amd.core.base_input_generator import BaseInputGenerator
+  from lingvo_amd.core.nested_map import NestedMap
+
+  class Finite(BaseInputGenerator):
+    @classmethod
+    def Params(cls):
+      p = super().Params()
+      p.Define('value', 0, '')
+      p.Define('batches', 2, '')
+      return p
+
+    def _InputBatch(self):
+      if self._batch_count >= self.p.batches:
+        raise StopIteration
+      return NestedMap(x=torch.full((2,), float(self.p.value)))
+
+  def src(v):
+    return ds.SimpleDataSource.Params().Set(
+        input_generator=Finite.Params().Set(value=v, name=f'f{v}'))
+
+  seq = ds.SequentialDataSource.Params().Set(
+      name='seq', sub=[src(1), src(2)]).Instantiate()
+  seen = []
+  try:
+    while True:
+      seen.append(int(seq.GetNext().x[0]))
+  except StopIteration:
+    pass
+  assert seen == [1, 1, 2, 2]
+  seq.Reset()
+  assert int(seq.GetNext().x[0]) == 1
+
+
+def test_file_input_generator_within_batch_mixing(tmp_path):
+  import torch
+  from lingvo_amd.core.base_input_generator import \
+      BaseInputGeneratorFromFiles
+  from lingvo_amd.core.nested_map import NestedMap
+
+  for name, tok in [('a.txt', 'aaa'), ('b.txt', 'bbb')]:
+    with open(tmp_path / name, 'w') as f:
+      for i in range(200):
+        f.write(f'{tok}\n')
+
+  class Gen(BaseInputGeneratorFromFiles):
+    def ProcessRecord(self, record):
+      val = 1.0 if record == b'aaa' else 2.0
+      return NestedMap(x=torch.tensor([val])), 1
+
+  p = Gen.Params().Set(
+      name='g', batch_size=8,
+      file_pattern=[(f'text:{tmp_path}/a.txt', 0.75),
+                    (f'text:{tmp_path}/b.txt', 0.25)])
+  gen = p.Instantiate()
+  vals = []
+  for _ in range(10):
+    b = gen.GetPreprocessedInputBatch()
+    assert b.x.shape[0] == 8
+    vals.extend(b.x.reshape(-1).tolist())
+  frac_a = sum(1 for v in vals if v == 1.0) / len(vals)
+  assert 0.6 < frac_a < 0.9  # ~0.75 example-level mix
+  gen.Reset()
