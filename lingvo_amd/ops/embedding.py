@@ -26,6 +26,17 @@ class _EmbeddingFn(torch.autograd.Function):
   def backward(ctx, dy):
     ext = _loader.get_ext(required=True)
     (ids,) = ctx.saved_tensors
+    if torch.cuda.is_current_stream_capturing():
+      # The deterministic sorted-segment path synchronizes (dynamic
+      # unique counts) and cannot be captured in a hipGraph; fall back
+      # to capture-safe atomic index_add (nondeterministic summation
+      # order, standard embedding-backward semantics).
+      d = dy.shape[-1]
+      dtable = torch.zeros(ctx.vocab, d, dtype=torch.float32,
+                           device=dy.device)
+      dtable.index_add_(0, ids.reshape(-1),
+                        dy.float().reshape(-1, d) * ctx.scale)
+      return dtable.to(dy.dtype), None, None
     dtable = ext.emb_scatter_add(dy.to(torch.bfloat16).contiguous(), ids,
                                  ctx.vocab, ctx.scale)
     return dtable, None, None

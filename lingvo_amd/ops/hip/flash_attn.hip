@@ -196,7 +196,22 @@ __global__ __launch_bounds__(FWD_BLOCK) void fa_fwd_kernel(
     const int qc_hi = (qt * FQT + FQT - 1) / chunk;
     kmin = max(kmin, max(0, (qc_lo - lc)) * chunk);
     kmax_excl = min(kmax_excl, (qc_hi + 1) * chunk);
-    if (kmax_excl <= kmin) return;
+    if (kmax_excl <= kmin) {
+      // No visible keys for this whole q-tile: emit zeros (fully-masked
+      // rows output 0, matching the reference) and a NEG_INF lse.
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int qrow = q0 + g * 4 + r;
+        if (qrow >= T) continue;
+        unsigned short* orow = o + (((long)b * T + qrow) * N + n) * H;
+#pragma unroll
+        for (int hf = 0; hf < HF; ++hf) {
+          orow[hf * 16 + cl] = float_to_bf16_bits(0.f);
+        }
+        if (cl == 0) lse[((long)b * N + n) * T + qrow] = NEG_INF;
+      }
+      return;
+    }
   }
   const int kt_lo = kmin / KTF;
   const int kt_hi = (max(kmax_excl, 1) - 1) / KTF;
@@ -439,7 +454,24 @@ __global__ __launch_bounds__(NWB * WAVE_SIZE) void fa_bwd_kernel(
     const int kc_hi = (min(kbase + KTB, S) - 1) / chunk;
     qlo = max(qlo, kc_lo * chunk);
     qhi = min(qhi, (kc_hi + lc + 1) * chunk - 1);
-    if (qhi < qlo) return;
+    if (qhi < qlo) {
+      // No visible queries: dK/dV of this key tile are zero. Write them
+      // (group==1 outputs are uninitialized buffers).
+      if (group == 1) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int key = k0 + g * 4 + r;
+          if (key >= S) continue;
+          long off = (((long)b * S + key) * NKV + nkv) * H;
+#pragma unroll
+          for (int hf = 0; hf < HF; ++hf) {
+            dk[off + hf * 16 + cl] = float_to_bf16_bits(0.f);
+            dv[off + hf * 16 + cl] = float_to_bf16_bits(0.f);
+          }
+        }
+      }
+      return;
+    }
   }
   const int qt_lo = qlo / QT, qt_hi = max(qhi, 0) / QT;
 
@@ -551,16 +583,6 @@ __global__ __launch_bounds__(NWB * WAVE_SIZE) void fa_bwd_kernel(
         for (int r = 0; r < 4; ++r) {
           float dlogits = pt[nf][r] * (acc[r] - delta_s[nf * 16 + cl]);
           dlg[nf][r] = dlogits;
-          if (BIAS_GRAD) {
-            const int qcol = qb + nf * 16 + cl;
-            const int key = k0 + g * 4 + r;
-            if (dlogits != 0.f) {
-              int d = qcol - key;
-              d = d < -bias_clip ? -bias_clip
-                                 : (d > bias_clip ? bias_clip : d);
-              atomicAdd(&dbias_s[d + bias_clip], dlogits);
-            }
-          }
         }
       }
 
@@ -632,6 +654,34 @@ __global__ __launch_bounds__(NWB * WAVE_SIZE) void fa_bwd_kernel(
                       cl,
                   acc_dq[hf][r]);
             }
+          }
+        }
+      } else if (BIAS_GRAD && wid >= QT / 16) {
+        // Bias grad via per-diagonal sums over the dS tile in ds_lds
+        // (entries hold dlogits*scale; invisible positions are exact
+        // zeros). One lane per diagonal d = qcol - key: elements
+        // (qrow, key) with qb + qrow - kbase - key == d. Runs on the
+        // waves otherwise idle during the dQ phase.
+        const int ndiag = QT + KTB - 1;
+        const int lane_global = (wid - QT / 16) * WAVE_SIZE + lane;
+        const int nworkers = (NWB - QT / 16) * WAVE_SIZE;
+        const float inv_scale = 1.f / scale;
+        for (int di = lane_global; di < ndiag; di += nworkers) {
+          // diagonal offset within the tile: qrow - key = di - (KTB-1)
+          const int doff = di - (KTB - 1);
+          float sum = 0.f;
+          const int q_lo = max(0, doff);
+          const int q_hi = min(QT - 1, KTB - 1 + doff);
+          for (int qrow = q_lo; qrow <= q_hi; ++qrow) {
+            const int key = qrow - doff;
+            sum += bf16_bits_to_float(*reinterpret_cast<unsigned short*>(
+                ds_lds + qrow * (KTB * 2) + swz(qrow, key * 2)));
+          }
+          if (sum != 0.f) {
+            int d = (qb + doff) - kbase;  // global q - k distance
+            d = d < -bias_clip ? -bias_clip
+                               : (d > bias_clip ? bias_clip : d);
+            atomicAdd(&dbias_s[d + bias_clip], sum * inv_scale);
           }
         }
       }
