@@ -211,3 +211,50 @@ class PassiveAsymQDomain(QDomain):
   def state_dict_ranges(self):
     return {k: (float(v[0]), float(v[1]))
             for k, v in self._ranges.items()}
+
+
+class SymmetricScheduledClipQDomain(QDomain):
+  """Fake quant against a SCHEDULED clip cap instead of a running max
+  (reference quant_utils.py SymmetricScheduledClipQDomain: the clipping
+  ramp of FakeQuantizationSchedule drives the quantization range, so
+  early training sees soft clipping and late training a fixed cap)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('cc_schedule', FakeQuantizationSchedule.Params(),
+             'Clip-cap schedule.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    self.CreateChild('schedule', self.p.cc_schedule)
+
+  def QuantizeTensor(self, x: torch.Tensor,
+                     calibrate: bool = True) -> torch.Tensor:
+    p = self.p
+    cap = self.schedule.CurrentCap(self._step)
+    x = torch.clamp(x, -cap, cap)
+    if not self.schedule.ShouldQuantize(self._step):
+      return x
+    qmax = 2.0 ** (p.bits - 1) - 1
+    scale = torch.tensor(cap / qmax, dtype=x.dtype, device=x.device)
+    return _FakeQuantFn.apply(x, scale, p.bits)
+
+
+def MaterializeInt8Weights(module: torch.nn.Module):
+  """Post-training int8 weight export: returns
+  {name: (int8 tensor, fp32 scale)} with symmetric per-tensor scales
+  (the serving-side counterpart of QuantizeWeight)."""
+  out = {}
+  for name, prm in module.named_parameters():
+    w = prm.detach().float()
+    scale = w.abs().max().clamp_min(1e-6) / 127.0
+    q = torch.clamp(torch.round(w / scale), -127, 127).to(torch.int8)
+    out[name] = (q, float(scale))
+  return out
+
+
+def DequantizeInt8(q: torch.Tensor, scale: float,
+                   dtype=torch.float32) -> torch.Tensor:
+  return q.to(dtype) * scale
