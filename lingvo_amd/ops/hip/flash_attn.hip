@@ -108,17 +108,16 @@ __device__ __forceinline__ bf16x8 lds_frag(const char* tile, int row,
 }
 
 __device__ __forceinline__ bool visible(int q, int k, int klen, int win_l,
-                                        int win_r, int chunk, int lc) {
+                                        int win_r, int chunk_lo,
+                                        int chunk_hi) {
   if (k >= klen) return false;
   if (win_l >= 0 && k < q - win_l) return false;
   if (win_r >= 0 && k > q + win_r) return false;
-  if (chunk > 0) {
-    // Chunkwise mask (reference ChunkwiseSelfAttention,
-    // batch_major_attention.py:4008): keys visible iff their chunk is
-    // within [q_chunk - lc, q_chunk].
-    const int qc = q / chunk, kc = k / chunk;
-    if (kc > qc || kc < qc - lc) return false;
-  }
+  // Chunkwise mask (reference ChunkwiseSelfAttention,
+  // batch_major_attention.py:4008): caller precomputes the visible key
+  // range [chunk_lo, chunk_hi) PER QUERY (one division per q, not per
+  // element — per-element integer division cost 1.7x fwd time).
+  if (k < chunk_lo || k >= chunk_hi) return false;
   return true;
 }
 
@@ -173,6 +172,20 @@ __global__ __launch_bounds__(FWD_BLOCK) void fa_fwd_kernel(
     } else {
       float z[8] = {0, 0, 0, 0, 0, 0, 0, 0};
       qfrag[kk] = pack_bf16x8(z);
+    }
+  }
+
+  // Per-lane chunk windows for its 4 q rows (one division each).
+  int cw_lo[4], cw_hi[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    if (chunk > 0) {
+      const int qc = (q0 + g * 4 + r) / chunk;
+      cw_lo[r] = (qc - lc) * chunk;
+      cw_hi[r] = (qc + 1) * chunk;
+    } else {
+      cw_lo[r] = 0;
+      cw_hi[r] = 0x7fffffff;
     }
   }
 
@@ -257,7 +270,8 @@ __global__ __launch_bounds__(FWD_BLOCK) void fa_fwd_kernel(
           val += bf16_bits_to_float(
               bias[(long)n * (2 * bias_clip + 1) + d + bias_clip]);
         }
-        if (!visible(qrow, kcol, klen, win_l, win_r, chunk, lc) || qrow >= T)
+        if (!visible(qrow, kcol, klen, win_l, win_r, cw_lo[r], cw_hi[r]) ||
+            qrow >= T)
           val = NEG_INF;
         if (SEG) {
           if (kcol < S && qrow < T &&
@@ -513,6 +527,20 @@ __global__ __launch_bounds__(NWB * WAVE_SIZE) void fa_bwd_kernel(
       }
       __syncthreads();
 
+      // Per-lane chunk windows for this q-tile's 4 q columns.
+      int bw_lo[4], bw_hi[4];
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        if (chunk > 0) {
+          const int qc = (qb + nf * 16 + cl) / chunk;
+          bw_lo[nf] = (qc - lc) * chunk;
+          bw_hi[nf] = (qc + 1) * chunk;
+        } else {
+          bw_lo[nf] = 0;
+          bw_hi[nf] = 0x7fffffff;
+        }
+      }
+
       // S^T strip: rows = 16 keys (this wave), cols = QT queries.
       float pt[4][4];   // P^T
       float dlg[4][4];  // dLogits
@@ -535,7 +563,8 @@ __global__ __launch_bounds__(NWB * WAVE_SIZE) void fa_bwd_kernel(
             d = d < -bias_clip ? -bias_clip : (d > bias_clip ? bias_clip : d);
             val += bf16_bits_to_float(bias[(long)n * nbias + d + bias_clip]);
           }
-          bool vis = visible(qcol, key, klen, win_l, win_r, chunk, lc) && qcol < T;
+          bool vis = visible(qcol, key, klen, win_l, win_r, bw_lo[nf],
+                             bw_hi[nf]) && qcol < T;
           if (vis && qseg &&
               qseg[(long)b * T + qcol] != kseg[(long)b * S + key])
             vis = false;
