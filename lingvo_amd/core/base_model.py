@@ -92,6 +92,11 @@ class BaseTask(BaseLayer):
               'dict of MagnitudePruner kwargs (reference '
               'base_model.py:1105 _GetMaskUpdateOp model_pruning hook); '
               'None disables magnitude pruning.')
+    tp.Define('early_stop', None,
+              'EarlyStop params (core/early_stop.py) with '
+              'p.metric_name set: the trainer stops when the eval '
+              'metric history shows no improvement for p.window steps '
+              '(reference early_stop.py:126 wired through runners).')
     p.Define('train', tp, 'Training hyperparameters subtree.')
     ep = Params()
     ep.Define('samples_per_summary', 1000, 'Eval samples per summary.')

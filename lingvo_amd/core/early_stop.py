@@ -55,6 +55,10 @@ class EarlyStop:
   def Params(cls) -> Params:
     p = Params()
     p.Define('metric_history', None, 'MetricHistory instance or None.')
+    p.Define('metric_name', 'loss',
+             'Eval metric the runners track into MetricHistory when '
+             'wiring EarlyStop from task params.')
+    p.Define('minimize', True, 'Whether lower metric is better.')
     p.Define('window', 10000, 'Steps without improvement to stop.')
     p.Define('min_steps', 1000, 'Never stop before this step.')
     p.Define('tolerance', 0.0, 'Improvement tolerance.')

@@ -49,11 +49,13 @@ class BaseRunner:
   """Owns model/checkpointer; _RunLoop retries transient failures."""
 
   def __init__(self, model_params, logdir: str, job_name: str,
-               device: Optional[str] = None, max_retries: int = 10):
+               device: Optional[str] = None, max_retries: int = 10,
+               trial=None):
     self._params = model_params
     self._logdir = logdir
     self._job = job_name
     self._max_retries = max_retries
+    self._trial = trial  # lingvo_amd.runtime.trial.Trial or None
     self._device = device or (
         'cuda:0' if torch.cuda.is_available() else 'cpu')
     self._train_dir = os.path.join(logdir, 'train')
@@ -116,13 +118,32 @@ class Trainer(BaseRunner):
     self._tracker = StepRateTracker()
     self._metrics_log = os.path.join(logdir, 'train', 'metrics.jsonl')
     self._tb = None  # lazily created TensorBoard events writer
+    self._early_stop = None  # lazily built from p.train.early_stop
 
   def Start(self) -> None:
     self._RunLoop(self._Loop)
 
   def _ShouldStop(self, task) -> bool:
     limit = self._max_steps or task.p.train.max_steps
-    return limit is not None and task.global_step >= limit
+    if limit is not None and task.global_step >= limit:
+      return True
+    if self._trial is not None and self._trial.ShouldStop():
+      self._SetStatusMessage('trial requested stop')
+      return True
+    es_p = task.p.train.early_stop
+    if es_p is not None:
+      if self._early_stop is None:
+        from lingvo_amd.core.early_stop import EarlyStop, MetricHistory
+        es = es_p.Copy()
+        es.metric_history = MetricHistory(
+            self._logdir, 'eval', es_p.metric_name,
+            minimize=es_p.minimize)
+        self._early_stop = EarlyStop(es)
+      if self._early_stop.Stop(int(task.global_step)):
+        self._SetStatusMessage(
+            f'early stop at step {int(task.global_step)}')
+        return True
+    return False
 
   def _Loop(self) -> None:
     model = self.model
@@ -277,6 +298,16 @@ class Evaler(_CheckpointPoller):
       from lingvo_amd.core.summary_utils import TbEventWriter
       self._tb = TbEventWriter(out_dir)
     self._tb.scalars(vals, step)
+    # Early-stop history + trial report (reference early_stop.py:126
+    # BestStep wiring + base_trial hooks in runners).
+    es_p = task.p.train.early_stop
+    if es_p is not None and es_p.metric_name in vals:
+      from lingvo_amd.core.early_stop import MetricHistory
+      MetricHistory(self._logdir, 'eval', es_p.metric_name,
+                    minimize=es_p.minimize).ConditionalAppend(
+                        step, vals[es_p.metric_name])
+    if self._trial is not None:
+      self._trial.ShouldStopAndMaybeReport(step, vals)
     task.train()
 
 

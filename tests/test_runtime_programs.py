@@ -427,3 +427,40 @@ def test_summarize_metrics_tool(tmp_path):
   ex.Start()
   out = Summarize(str(tmp_path))
   assert 'metrics.jsonl' in out and 'loss=' in out
+
+
+def test_early_stop_and_trial_wiring(tmp_path):
+  """Trainer stops when (a) the eval metric history shows no
+  improvement for `window` steps, (b) a FileTrial requests a stop."""
+  import torch
+  from lingvo_amd.core.early_stop import EarlyStop, MetricHistory
+  from lingvo_amd.runtime.trial import FileTrial, NoOpTrial
+  from lingvo_amd.runtime import runners
+  from lingvo_amd.core import registry
+
+  logdir = str(tmp_path / 'log')
+  # Simulate an evaler-written history: best at step 1, flat after.
+  mh = MetricHistory(logdir, 'eval', 'loss', minimize=True)
+  for step, v in [(1, 1.0), (5, 1.5), (9, 2.0)]:
+    mh.ConditionalAppend(step, v)
+  assert mh.BestStep() == 1
+
+  p = registry.GetParams('image.mnist.LeNet5', 'Train')
+  p.task.train.early_stop = EarlyStop.Params().Set(
+      metric_name='loss', window=3, min_steps=2)
+  tr = runners.Trainer(p, logdir, max_steps=100)
+  task = tr.model.GetTask()
+  task.global_step_var += 6  # step 6: best=1, 6-1 > window=3 -> stop
+  assert tr._ShouldStop(task)
+
+  # Trial stop file
+  trial = FileTrial(str(tmp_path / 'trial'))
+  p2 = registry.GetParams('image.mnist.LeNet5', 'Train')
+  tr2 = runners.Trainer(p2, str(tmp_path / 'log2'), max_steps=100,
+                        trial=trial)
+  task2 = tr2.model.GetTask()
+  assert not tr2._ShouldStop(task2)
+  trial.ReportEvalMeasure(3, {'loss': 0.5})
+  open(trial._stopfile, 'w').close()
+  assert tr2._ShouldStop(task2)
+  assert not NoOpTrial().ShouldStop()
