@@ -23,6 +23,90 @@ from lingvo_amd.core.nested_map import NestedMap
 
 TRANSIENT_ERRORS = (ConnectionError, TimeoutError, BrokenPipeError, OSError)
 
+# RuntimeError substrings that indicate a transient comm/runtime fault
+# (the reference's Aborted/FailedPrecondition/DataLoss/Cancelled taxonomy,
+# base_runner.py:399-527, mapped to the RCCL/HIP world).
+TRANSIENT_RUNTIME_MARKERS = (
+    'NCCL', 'RCCL', 'Connection reset', 'Socket Timeout', 'EOF',
+    'Watchdog caught', 'Broken pipe', 'transport error',
+)
+# Fatal marker strings: retrying cannot help (the reference's
+# compile-error class -> os._exit(1) in daemon mode).
+FATAL_RUNTIME_MARKERS = (
+    'out of memory', 'HIP error: invalid device function',
+    'no kernel image',
+)
+
+
+def ClassifyError(e: BaseException) -> str:
+  """'transient' | 'fatal' | 'oor' (out-of-range: clean end of data)."""
+  if isinstance(e, StopIteration):
+    return 'oor'
+  if isinstance(e, FloatingPointError):
+    return 'fatal'
+  msg = str(e)
+  if isinstance(e, RuntimeError):
+    if any(m.lower() in msg.lower() for m in FATAL_RUNTIME_MARKERS):
+      return 'fatal'
+    if any(m.lower() in msg.lower() for m in TRANSIENT_RUNTIME_MARKERS):
+      return 'transient'
+    return 'fatal'
+  if isinstance(e, TRANSIENT_ERRORS):
+    return 'transient'
+  return 'fatal'
+
+
+class Watchdog:
+  """Per-rank hang detector (SURVEY SS5 failure detection): if Pet() is
+  not called within `timeout_s`, dumps all python stacks, tears down the
+  torch.distributed process group (so RCCL peers unblock instead of
+  hanging the collective), and hard-exits with code 42 for an external
+  scheduler to restart (the reference daemon-mode exit-code protocol,
+  base_runner.py:413-420)."""
+
+  def __init__(self, timeout_s: float, tag: str = 'trainer',
+               exit_fn=None):
+    import threading
+    self._timeout = timeout_s
+    self._tag = tag
+    self._last = time.monotonic()
+    self._stop = False
+    self._exit_fn = exit_fn or (lambda code: os._exit(code))
+    self._fired = False
+    self._thread = threading.Thread(target=self._Loop, daemon=True)
+    self._thread.start()
+
+  def Pet(self) -> None:
+    self._last = time.monotonic()
+
+  def Stop(self) -> None:
+    self._stop = True
+
+  def _Loop(self) -> None:
+    import faulthandler
+    import sys
+    while not self._stop:
+      time.sleep(min(5.0, self._timeout / 4))
+      if self._stop:
+        return
+      if time.monotonic() - self._last > self._timeout:
+        self._fired = True
+        sys.stderr.write(
+            f'[watchdog:{self._tag}] no progress for '
+            f'{self._timeout:.0f}s; dumping stacks and aborting\n')
+        try:
+          faulthandler.dump_traceback(file=sys.stderr)
+        except Exception:
+          pass
+        try:
+          import torch.distributed as dist
+          if dist.is_available() and dist.is_initialized():
+            dist.destroy_process_group()
+        except Exception:
+          pass
+        self._exit_fn(42)
+        return
+
 
 class StepRateTracker:
   """steps/sec + examples/sec EMA (reference summary_utils.py:393)."""
@@ -85,7 +169,15 @@ class BaseRunner:
       try:
         loop_fn()
         return
-      except TRANSIENT_ERRORS as e:
+      except Exception as e:  # noqa: BLE001 - classified below
+        kind = ClassifyError(e)
+        if kind == 'oor':
+          # End of data in eval/decode: clean finish (reference treats
+          # OutOfRange in eval jobs as loop completion).
+          self._SetStatusMessage('end of data')
+          return
+        if kind == 'fatal':
+          raise
         retries += 1
         if retries > self._max_retries:
           raise
@@ -93,8 +185,6 @@ class BaseRunner:
         self._SetStatusMessage(
             f'transient error, retry {retries} in {wait:.0f}s: {e}')
         time.sleep(wait)
-      except FloatingPointError:
-        raise  # NaN loss: fatal, like reference compile errors.
 
   def Start(self) -> None:
     raise NotImplementedError
@@ -104,9 +194,12 @@ class Trainer(BaseRunner):
   """Synchronous training loop (reference runners.py:192)."""
 
   def __init__(self, model_params, logdir: str, max_steps: Optional[int]
-               = None, grad_sync=None, **kwargs):
+               = None, grad_sync=None,
+               watchdog_timeout: Optional[float] = None, **kwargs):
     super().__init__(model_params, logdir, 'trainer', **kwargs)
     self._max_steps = max_steps
+    self._watchdog = (Watchdog(watchdog_timeout, 'trainer')
+                      if watchdog_timeout else None)
     if grad_sync is None:
       # Auto-enable DP grad sync when launched under torchrun.
       import torch.distributed as dist
@@ -165,6 +258,8 @@ class Trainer(BaseRunner):
       examples = py_utils.ToScalar(
           metrics.get('num_samples_in_batch', (batch_size_of(batch), 1))[0])
       self._tracker.Update(step, examples)
+      if self._watchdog is not None:
+        self._watchdog.Pet()
       if step % 10 == 0 or step <= 1:
         loss = py_utils.ToScalar(metrics[task.learners[0].p.loss_name][0])
         self._SetStatusMessage(

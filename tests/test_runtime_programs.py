@@ -464,3 +464,44 @@ def test_early_stop_and_trial_wiring(tmp_path):
   open(trial._stopfile, 'w').close()
   assert tr2._ShouldStop(task2)
   assert not NoOpTrial().ShouldStop()
+
+
+def test_error_classification_taxonomy():
+  from lingvo_amd.runtime.runners import ClassifyError
+  assert ClassifyError(StopIteration()) == 'oor'
+  assert ClassifyError(ConnectionError('x')) == 'transient'
+  assert ClassifyError(RuntimeError('NCCL communicator was aborted')) \
+      == 'transient'
+  assert ClassifyError(RuntimeError('HIP out of memory')) == 'fatal'
+  assert ClassifyError(RuntimeError('shape mismatch')) == 'fatal'
+  assert ClassifyError(ValueError('bad')) == 'fatal'
+  assert ClassifyError(FloatingPointError('nan')) == 'fatal'
+
+
+def test_watchdog_fires_and_pets():
+  import time as _t
+  from lingvo_amd.runtime.runners import Watchdog
+  fired = []
+  wd = Watchdog(0.4, 'test', exit_fn=lambda code: fired.append(code))
+  for _ in range(4):      # petted: must not fire
+    _t.sleep(0.15)
+    wd.Pet()
+  assert not fired
+  _t.sleep(1.2)           # starved: fires with exit code 42
+  assert fired == [42]
+  wd.Stop()
+
+
+def test_runloop_oor_is_clean_finish(tmp_path):
+  from lingvo_amd.runtime import runners
+  from lingvo_amd.core import registry
+  p = registry.GetParams('image.mnist.LeNet5', 'Train')
+  tr = runners.Trainer(p, str(tmp_path), max_steps=5)
+  calls = []
+
+  def loop():
+    calls.append(1)
+    raise StopIteration  # end of data
+
+  tr._RunLoop(loop)      # returns cleanly, no retry storm
+  assert len(calls) == 1
