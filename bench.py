@@ -52,12 +52,21 @@ def build_argparser():
                   help='auto = child-process graph with eager fallback '
                        '(single rank); graph/eager = run that mode '
                        'inline in this process.')
+  ap.add_argument('--decode', action='store_true',
+                  help='Measure inference Decode() examples/sec instead '
+                       'of the train step (beam width = model '
+                       'decode_num_hyps; greedy when 1).')
+  ap.add_argument('--decode-hyps', type=int, default=1,
+                  help='Beam width for --decode.')
   return ap
 
 
 def main():
   args = build_argparser().parse_args()
   world = int(os.environ.get('WORLD_SIZE', '1'))
+  if args.decode:
+    run_decode_bench(args)
+    return
   if args.no_graph:
     args.mode = 'eager'
 
@@ -100,6 +109,60 @@ def main():
     sys.exit(1)
 
   run_bench(args, world)
+
+
+def run_decode_bench(args):
+  """Inference decode throughput (decode examples/sec).
+
+  Times AsrModel.Decode() — encoder + autoregressive LSTM-attention
+  decode (greedy or beam) — on synthetic batches."""
+  _phase('importing torch')
+  import torch
+  from lingvo_amd.core import registry
+  has_gpu = torch.cuda.is_available()
+  device = 'cuda:0' if has_gpu else 'cpu'
+  _phase('building model')
+  model_p = registry.GetParams(args.model, 'Train')
+  model_p.input.batch_size = args.batch
+  if not has_gpu:
+    model_p.task.fprop_dtype = torch.float32
+    model_p.input.frame_len = 80
+    model_p.task.encoder.num_layers = 1
+  model_p.task.random_seed = 1234
+  if 'decode_num_hyps' in model_p.task:
+    model_p.task.decode_num_hyps = args.decode_hyps
+  model = model_p.Instantiate().to(device)
+  task = model.GetTask()
+  task.eval()
+  gen = task.input_generator
+  batch = gen.GetPreprocessedInputBatch().Transform(
+      lambda t: t.to(device) if isinstance(t, torch.Tensor) else t)
+  _phase('decode warmup')
+  for _ in range(max(1, args.warmup)):
+    out = task.Decode(batch)
+  if has_gpu:
+    torch.cuda.synchronize()
+  t0 = time.perf_counter()
+  for i in range(args.steps):
+    out = task.Decode(batch)
+    if i % 5 == 0:
+      _phase(f'decode step {i}')
+  if has_gpu:
+    torch.cuda.synchronize()
+  elapsed = time.perf_counter() - t0
+  examples_per_sec = args.batch * args.steps / elapsed
+  print(json.dumps({
+      'metric': 'decode examples/sec, Librispeech Conformer-L',
+      'value': round(examples_per_sec, 3),
+      'unit': 'examples/sec',
+      'n_gpus': 1, 'steps': args.steps, 'warmup': args.warmup,
+      'ms_per_step': round(elapsed / args.steps * 1000.0, 3),
+      'higher_is_better': True, 'scaling': 'weak', 'vs_baseline': None,
+      'dtype': 'bf16' if has_gpu else 'fp32', 'data': 'synthetic',
+      'config': {'model': args.model, 'global_batch': args.batch,
+                 'beam': args.decode_hyps,
+                 'parallelism': 'dp1'},
+  }), flush=True)
 
 
 def run_bench(args, world):
