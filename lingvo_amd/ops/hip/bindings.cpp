@@ -96,6 +96,11 @@ torch::Tensor emb_gather(torch::Tensor table, torch::Tensor ids,
 torch::Tensor emb_scatter_add(torch::Tensor dy, torch::Tensor ids,
                               int64_t vocab, double scale);
 
+// moe_gating.hip
+std::vector<torch::Tensor> moe_positions(torch::Tensor top1,
+                                         torch::Tensor top2,
+                                         int64_t num_experts);
+
 // las_decoder.hip
 void smallm_gemm(torch::Tensor a, torch::Tensor wt,
                  c10::optional<torch::Tensor> pre, torch::Tensor out,
@@ -130,6 +135,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("emb_gather", &emb_gather, "Embedding gather fwd");
   m.def("emb_scatter_add", &emb_scatter_add,
         "Deterministic embedding scatter-add bwd");
+  m.def("moe_positions", &moe_positions,
+        "Deterministic MoE top-2 position scan");
   m.def("smallm_gemm", &smallm_gemm,
         "Small-M MFMA GEMM (decode-step projections)");
   m.def("attend_fwd", &attend_fwd, "Fused dot-attention step fwd");

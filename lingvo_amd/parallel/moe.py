@@ -76,13 +76,22 @@ def Top2Gating(logits: torch.Tensor, capacity: int,
   aux_loss = (density * mean_probs).sum() * e
 
   # Position in expert via cumsum over token order (top1 first, then
-  # top2, matching the reference's ordering).
-  one1 = F.one_hot(top1, e).to(torch.int32)
-  pos1 = (one1.cumsum(dim=0) - 1).gather(1, top1.unsqueeze(1)).squeeze(1)
-  count1 = one1.sum(dim=0)  # [E]
-  one2 = F.one_hot(top2, e).to(torch.int32)
-  pos2 = (one2.cumsum(dim=0) - 1).gather(
-      1, top2.unsqueeze(1)).squeeze(1) + count1.gather(0, top2)
+  # top2, matching the reference's ordering). GPU path: deterministic
+  # per-expert block scan in-kernel (ops/hip/moe_gating.hip, K9).
+  if logits.is_cuda:
+    from lingvo_amd.ops import _loader
+    ext = _loader.get_ext(required=True)
+    p1, p2, _ = ext.moe_positions(top1.to(torch.int32).contiguous(),
+                                  top2.to(torch.int32).contiguous(), e)
+    pos1, pos2 = p1.long(), p2.long()
+  else:
+    one1 = F.one_hot(top1, e).to(torch.int32)
+    pos1 = (one1.cumsum(dim=0) - 1).gather(
+        1, top1.unsqueeze(1)).squeeze(1)
+    count1 = one1.sum(dim=0)  # [E]
+    one2 = F.one_hot(top2, e).to(torch.int32)
+    pos2 = (one2.cumsum(dim=0) - 1).gather(
+        1, top2.unsqueeze(1)).squeeze(1) + count1.gather(0, top2)
 
   keep1 = pos1 < capacity
   keep2 = pos2 < capacity

@@ -301,3 +301,25 @@ def test_embedding_kernel_matches_torch_and_deterministic():
   rr.square().sum().backward()
   rel = (g1 - tref.grad).abs().max() / tref.grad.abs().max().clamp_min(1)
   assert rel < 0.05, rel
+
+
+@gpu
+def test_moe_positions_kernel_matches_cumsum():
+  import torch.nn.functional as F
+  import lingvo_amd.ops._lingvo_ops as ext
+  torch.manual_seed(8)
+  n, e = 5000, 16
+  top1 = torch.randint(0, e, (n,), device='cuda', dtype=torch.int32)
+  top2 = torch.randint(0, e, (n,), device='cuda', dtype=torch.int32)
+  p1, p2, c1 = ext.moe_positions(top1, top2, e)
+  one1 = F.one_hot(top1.long(), e).to(torch.int32)
+  ref1 = (one1.cumsum(dim=0) - 1).gather(
+      1, top1.long().unsqueeze(1)).squeeze(1)
+  count1 = one1.sum(dim=0)
+  one2 = F.one_hot(top2.long(), e).to(torch.int32)
+  ref2 = (one2.cumsum(dim=0) - 1).gather(
+      1, top2.long().unsqueeze(1)).squeeze(1) + \
+      count1.gather(0, top2.long())
+  assert torch.equal(p1.long(), ref1)
+  assert torch.equal(p2.long(), ref2)
+  assert torch.equal(c1.long(), count1.long())
