@@ -715,3 +715,74 @@ class MultitaskAdapterLayer(BaseLayer):
     act = activations.GetFn(p.activation)
     h = act(torch.baddbmm(db.unsqueeze(1), x, dw))
     return inputs + torch.baddbmm(ub.unsqueeze(1), h, uw)
+
+
+class ShardedEmbeddingLayer(BaseLayer):
+  """Vocab-sharded embedding table over the data/model-parallel group
+  (the MI355X counterpart of the reference's TPU-embedding subsystem,
+  core/tpu_embedding_layers*.py, 2329 LoC: giant tables sharded across
+  accelerators with gradients routed back to the owning shard).
+
+  Rank r stores rows [r*V/W, (r+1)*V/W). Lookup: each rank gathers the
+  ids it owns (others contribute zeros) and one all-reduce sums the
+  partial embeddings — 288 GB HBM per GPU means even 100B-row tables
+  shard across a single node without a separate parameter-server
+  engine. Gradients flow only to the owning rank's shard (marked
+  _ep_sharded so DP GradSync skips them)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('vocab_size', 0, 'GLOBAL vocabulary size.')
+    p.Define('embedding_dim', 0, 'Embedding dimension.')
+    p.Define('scale_sqrt_depth', False, 'Scale outputs by sqrt(dim).')
+    p.Define('shard_group', None, 'torch.distributed group (None = '
+             'default).')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    import torch.distributed as dist
+    if dist.is_available() and dist.is_initialized():
+      self._world = dist.get_world_size(p.shard_group)
+      self._rank = dist.get_rank(p.shard_group)
+    else:
+      self._world, self._rank = 1, 0
+    assert p.vocab_size % self._world == 0, (
+        'vocab must divide the shard world')
+    self._vshard = p.vocab_size // self._world
+    self.CreateVariable('wm', py_utils.WeightParams(
+        [self._vshard, p.embedding_dim],
+        py_utils.WeightInit.Gaussian(1.0 / math.sqrt(p.embedding_dim)),
+        p.dtype))
+    if self._world > 1:
+      # Deterministic slice of the full-table init (parity with the
+      # unsharded layer under the same seed).
+      g = self._InitGenerator('wm')
+      full = py_utils.InitWeight(
+          [p.vocab_size, p.embedding_dim],
+          py_utils.WeightInit.Gaussian(1.0 / math.sqrt(p.embedding_dim)),
+          g, p.dtype)
+      with torch.no_grad():
+        self.wm.copy_(full[self._rank * self._vshard:
+                           (self._rank + 1) * self._vshard])
+      self.wm._ep_sharded = True
+
+  def EmbLookup(self, theta: NestedMap, ids: torch.Tensor) -> torch.Tensor:
+    p = self.p
+    scale = p.embedding_dim ** 0.5 if p.scale_sqrt_depth else 1.0
+    ids = ids.long()
+    if self._world == 1:
+      from lingvo_amd.ops import embedding as emb_ops
+      return emb_ops.embedding_lookup(theta.wm, ids, scale)
+    lo = self._rank * self._vshard
+    mine = (ids >= lo) & (ids < lo + self._vshard)
+    local_ids = torch.where(mine, ids - lo, torch.zeros_like(ids))
+    out = F.embedding(local_ids, theta.wm) * scale
+    out = out * mine.unsqueeze(-1).to(out.dtype)
+    from lingvo_amd.parallel.tensor_parallel import _ReduceFromTp
+    return _ReduceFromTp.apply(out, p.shard_group)
+
+  def FProp(self, theta: NestedMap, ids: torch.Tensor) -> torch.Tensor:
+    return self.EmbLookup(theta, ids)
