@@ -36,46 +36,42 @@ __global__ __launch_bounds__(256) void smallm_gemm_kernel(
     const unsigned short* __restrict__ a, const unsigned short* __restrict__ wt,
     const unsigned short* __restrict__ pre, unsigned short* __restrict__ out,
     int m, int n, int k, long ldw, float alpha, int pre_mode) {
+  // K-SPLIT layout: every wave computes ALL 8 m-tiles over its quarter
+  // of K (no per-iteration block syncs; the serial K chain is 4x
+  // shorter than an M-split), then waves 1-3 spill partials to LDS and
+  // wave 0 reduces + writes. The recurrence calls this back-to-back
+  // with dependent inputs, so per-call LATENCY is the metric.
   const int nt = blockIdx.x;          // n-tile (16 cols)
   const int col0 = nt * 16;
   const int wid = threadIdx.x / WAVE_SIZE;
   const int lane = threadIdx.x & 63;
   const int g = lane >> 4;            // k-group 0..3
   const int cl = lane & 15;           // row-in-frag / col-in-frag
+  const int col = col0 + cl;
+  const bool col_ok = col < n;
 
-  // Double-buffered LDS staging of the Wt tile [16 cols][32 k]:
-  // coalesced global reads (16 threads x 4B per column row) instead of
-  // 16 lanes each issuing a strided 16B read.
-  constexpr int ROWP = 40;  // padded row stride in shorts (80 B)
-  __shared__ unsigned short wtile[2][16 * ROWP];
-  const int tid = threadIdx.x;
-  const int ld_row = tid / 16;        // 0..15 (wt column)
-  const int ld_off = (tid % 16) * 2;  // shorts within the 32-k row
+  constexpr int MT = 8;               // m-tiles (M <= 128)
+  __shared__ float red[3][128][16];   // waves 1-3 partials
 
-  auto stage = [&](int buf, int k0) {
-    const int col = col0 + ld_row;
-    unsigned int v = 0;
-    if (col < n && k0 + ld_off < k) {
-      v = *reinterpret_cast<const unsigned int*>(
-          wt + (long)col * ldw + k0 + ld_off);
-    }
-    *reinterpret_cast<unsigned int*>(
-        &wtile[buf][ld_row * ROWP + ld_off]) = v;
-  };
+  const int kq = (k / 32 + 3) / 4 * 32;  // per-wave K quota (mult of 32)
+  const int k_lo = wid * kq;
+  const int k_hi = min(k, k_lo + kq);
 
-  f32x4 acc[2] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
-
-  stage(0, 0);
-  __syncthreads();
-  int cur = 0;
-  for (int k0 = 0; k0 < k; k0 += 32) {
-    if (k0 + 32 < k) {
-      stage(1 - cur, k0 + 32);
-    }
-    bf16x8 bf = load_bf16x8_bits(&wtile[cur][cl * ROWP + g * 8]);
+  f32x4 acc[MT];
 #pragma unroll
-    for (int mt = 0; mt < 2; ++mt) {
-      const int row = wid * 32 + mt * 16 + cl;
+  for (int mt = 0; mt < MT; ++mt) acc[mt] = {0.f, 0.f, 0.f, 0.f};
+
+  for (int k0 = k_lo; k0 < k_hi; k0 += 32) {
+    bf16x8 bf;
+    if (col_ok) {
+      bf = load_bf16x8_bits(wt + (long)col * ldw + k0 + g * 8);
+    } else {
+      float z[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+      bf = pack_bf16x8(z);
+    }
+#pragma unroll
+    for (int mt = 0; mt < MT; ++mt) {
+      const int row = mt * 16 + cl;
       bf16x8 af;
       if (row < m) {
         af = load_bf16x8_bits(a + (long)row * k + k0 + g * 8);
@@ -85,20 +81,28 @@ __global__ __launch_bounds__(256) void smallm_gemm_kernel(
       }
       acc[mt] = mfma16x16x32_bf16(af, bf, acc[mt]);
     }
-    __syncthreads();
-    cur = 1 - cur;
   }
-  const bool col_ok = col0 + cl < n;
-  const int col = col0 + cl;
 
-  // Epilogue: C layout row = g*4+r, col = cl.
+  // Cross-wave reduction: C layout row = mt*16 + g*4 + r, col = cl.
+  if (wid > 0) {
 #pragma unroll
-  for (int mt = 0; mt < 2; ++mt) {
+    for (int mt = 0; mt < MT; ++mt) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        red[wid - 1][mt * 16 + g * 4 + r][cl] = acc[mt][r];
+      }
+    }
+  }
+  __syncthreads();
+  if (wid != 0) return;
+#pragma unroll
+  for (int mt = 0; mt < MT; ++mt) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      const int row = wid * 32 + mt * 16 + g * 4 + r;
+      const int row = mt * 16 + g * 4 + r;
       if (row >= m || !col_ok) continue;
-      float v = acc[mt][r] * alpha;
+      float v = (acc[mt][r] + red[0][row][cl] + red[1][row][cl] +
+                 red[2][row][cl]) * alpha;
       long off = (long)row * n + col;
       if (pre_mode == 1) {
         v += bf16_bits_to_float(pre[col]);
@@ -149,24 +153,42 @@ __global__ __launch_bounds__(256) void attend_fwd_kernel(
     }
   }
 
-  // Pass 1: logits, one s per wave-iteration (vector loads of enc).
+  // Pass 1: logits. Each wave processes 4 consecutive s per iteration
+  // (independent loads pipeline the HBM/L2 latency that a one-s-at-a-
+  // time loop serializes).
   const unsigned short* eb = enc + (long)b * s_len * d;
-  for (int s = wid; s < s_len; s += 4) {
-    float part = 0.f;
+  for (int s0 = wid * 4; s0 < s_len; s0 += 16) {
+    float part[4] = {0.f, 0.f, 0.f, 0.f};
     for (int v = 0; v < ndv; ++v) {
       const int i = lane * 8 + v * WAVE_SIZE * 8;
       if (i < d) {
-        bf16x8 ev = load_bf16x8_bits(eb + (long)s * d + i);
+        bf16x8 ev[4];
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          part += qreg[v][j] * (float)ev[j];
+        for (int u = 0; u < 4; ++u) {
+          if (s0 + u < s_len) {
+            ev[u] = load_bf16x8_bits(eb + (long)(s0 + u) * d + i);
+          }
+        }
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          if (s0 + u < s_len) {
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+              part[u] += qreg[v][j] * (float)ev[u][j];
+            }
+          }
         }
       }
     }
-    part = wave_reduce_sum(part);
-    if (lane == 0) {
-      logit_s[s] = part * scale + (pad[(long)b * s_len + s] > 0.5f
-                                       ? -1e30f : 0.f);
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      if (s0 + u < s_len) {
+        float t = wave_reduce_sum(part[u]);
+        if (lane == 0) {
+          logit_s[s0 + u] = t * scale +
+              (pad[(long)b * s_len + s0 + u] > 0.5f ? -1e30f : 0.f);
+        }
+      }
     }
   }
   __syncthreads();
@@ -197,16 +219,38 @@ __global__ __launch_bounds__(256) void attend_fwd_kernel(
   for (int i = threadIdx.x; i < 4 * d; i += 256) ctx_s[i] = 0.f;
   __syncthreads();
   float* my_ctx = ctx_s + wid * d;
-  for (int s = wid; s < s_len; s += 4) {
-    const float p = logit_s[s] * inv;
-    if (lane == 0) probs[(long)b * s_len + s] = p;
+  for (int s0 = wid * 4; s0 < s_len; s0 += 16) {
+    float pv[4];
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      pv[u] = s0 + u < s_len ? logit_s[s0 + u] * inv : 0.f;
+      if (lane == 0 && s0 + u < s_len) {
+        probs[(long)b * s_len + s0 + u] = pv[u];
+      }
+    }
     for (int v = 0; v < ndv; ++v) {
       const int i = lane * 8 + v * WAVE_SIZE * 8;
       if (i < d) {
-        bf16x8 ev = load_bf16x8_bits(eb + (long)s * d + i);
+        bf16x8 ev[4];
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          if (s0 + u < s_len) {
+            ev[u] = load_bf16x8_bits(eb + (long)(s0 + u) * d + i);
+          }
+        }
+        float accv[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          if (s0 + u < s_len) {
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+              accv[j] += pv[u] * (float)ev[u][j];
+            }
+          }
+        }
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
-          my_ctx[i + j] += p * (float)ev[j];
+          my_ctx[i + j] += accv[j];
         }
       }
     }
@@ -263,20 +307,36 @@ __global__ __launch_bounds__(256) void attend_bwd_kernel(
   // Pass 1: dprobs (stored to dl_s) + t reduction.
   const unsigned short* eb = enc + (long)b * s_len * d;
   const float* pb = probs + (long)b * s_len;
-  for (int s = wid; s < s_len; s += 4) {
-    float part = 0.f;
+  for (int s0 = wid * 4; s0 < s_len; s0 += 16) {
+    float part[4] = {0.f, 0.f, 0.f, 0.f};
     for (int v = 0; v < ndv; ++v) {
       const int i = lane * 8 + v * WAVE_SIZE * 8;
       if (i < d) {
-        bf16x8 ev = load_bf16x8_bits(eb + (long)s * d + i);
+        bf16x8 ev[4];
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          part += dcreg[v][j] * (float)ev[j];
+        for (int u = 0; u < 4; ++u) {
+          if (s0 + u < s_len) {
+            ev[u] = load_bf16x8_bits(eb + (long)(s0 + u) * d + i);
+          }
+        }
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          if (s0 + u < s_len) {
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+              part[u] += dcreg[v][j] * (float)ev[u][j];
+            }
+          }
         }
       }
     }
-    part = wave_reduce_sum(part);
-    if (lane == 0) dl_s[s] = part;
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      if (s0 + u < s_len) {
+        float t = wave_reduce_sum(part[u]);
+        if (lane == 0) dl_s[s0 + u] = t;
+      }
+    }
   }
   __syncthreads();
   float t_part = 0.f;
@@ -300,27 +360,57 @@ __global__ __launch_bounds__(256) void attend_bwd_kernel(
   __syncthreads();
   float* my_dq = dq_s + wid * d;
   float* db = denc + (long)b * s_len * d;
-  for (int s = wid; s < s_len; s += 4) {
-    const float dl = dl_s[s];
-    const float p = pb[s];
+  for (int s0 = wid * 4; s0 < s_len; s0 += 16) {
+    float dlv[4], ppv[4];
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      dlv[u] = s0 + u < s_len ? dl_s[s0 + u] : 0.f;
+      ppv[u] = s0 + u < s_len ? pb[s0 + u] : 0.f;
+    }
     for (int v = 0; v < ndv; ++v) {
       const int i = lane * 8 + v * WAVE_SIZE * 8;
       if (i < d) {
-        bf16x8 ev = load_bf16x8_bits(eb + (long)s * d + i);
-        floatx4* dbv = reinterpret_cast<floatx4*>(db + (long)s * d + i);
-        floatx4 lo = dbv[0], hi = dbv[1];
+        bf16x8 ev[4];
+        floatx4 lo[4], hi[4];
 #pragma unroll
-        for (int j = 0; j < 4; ++j) {
-          my_dq[i + j] += dl * (float)ev[j];
-          lo[j] += dl * qreg[v][j] + p * dcreg[v][j];
+        for (int u = 0; u < 4; ++u) {
+          if (s0 + u < s_len) {
+            ev[u] = load_bf16x8_bits(eb + (long)(s0 + u) * d + i);
+            floatx4* dbv =
+                reinterpret_cast<floatx4*>(db + (long)(s0 + u) * d + i);
+            lo[u] = dbv[0];
+            hi[u] = dbv[1];
+          }
+        }
+        float dqacc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          if (s0 + u < s_len) {
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+              dqacc[j] += dlv[u] * (float)ev[u][j];
+              lo[u][j] += dlv[u] * qreg[v][j] + ppv[u] * dcreg[v][j];
+            }
+#pragma unroll
+            for (int j = 4; j < 8; ++j) {
+              dqacc[j] += dlv[u] * (float)ev[u][j];
+              hi[u][j - 4] += dlv[u] * qreg[v][j] + ppv[u] * dcreg[v][j];
+            }
+          }
         }
 #pragma unroll
-        for (int j = 4; j < 8; ++j) {
-          my_dq[i + j] += dl * (float)ev[j];
-          hi[j - 4] += dl * qreg[v][j] + p * dcreg[v][j];
+        for (int u = 0; u < 4; ++u) {
+          if (s0 + u < s_len) {
+            floatx4* dbv =
+                reinterpret_cast<floatx4*>(db + (long)(s0 + u) * d + i);
+            dbv[0] = lo[u];
+            dbv[1] = hi[u];
+          }
         }
-        dbv[0] = lo;
-        dbv[1] = hi;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          my_dq[i + j] += dqacc[j];
+        }
       }
     }
   }
