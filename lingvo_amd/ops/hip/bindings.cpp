@@ -15,6 +15,7 @@ std::vector<torch::Tensor> layer_norm_bwd(torch::Tensor dy, torch::Tensor x,
 
 // mfma_probe.hip
 torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor bT);
+torch::Tensor tr16_probe(torch::Tensor a, torch::Tensor b);
 
 // flash_attn.hip
 std::vector<torch::Tensor> fa_fwd(torch::Tensor q, torch::Tensor k,
@@ -130,6 +131,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layer_norm_fwd", &layer_norm_fwd, "Fused LayerNorm/RMSNorm fwd");
   m.def("layer_norm_bwd", &layer_norm_bwd, "Fused LayerNorm/RMSNorm bwd");
   m.def("mfma_probe", &mfma_probe, "MFMA 16x16x32 layout probe");
+  m.def("tr16_probe", &tr16_probe, "ds_read_tr16_b64 layout probe");
   m.def("fa_fwd", &fa_fwd, "Flash attention forward");
   m.def("fa_bwd", &fa_bwd, "Flash attention backward");
   m.def("emb_gather", &emb_gather, "Embedding gather fwd");

@@ -43,6 +43,7 @@ __global__ __launch_bounds__(256) void smallm_gemm_kernel(
   // with dependent inputs, so per-call LATENCY is the metric.
   const int nt = blockIdx.x;          // n-tile (16 cols)
   const int col0 = nt * 16;
+  const int m0 = blockIdx.y * 128;    // m-block (rows handled here)
   const int wid = threadIdx.x / WAVE_SIZE;
   const int lane = threadIdx.x & 63;
   const int g = lane >> 4;            // k-group 0..3
@@ -50,7 +51,7 @@ __global__ __launch_bounds__(256) void smallm_gemm_kernel(
   const int col = col0 + cl;
   const bool col_ok = col < n;
 
-  constexpr int MT = 8;               // m-tiles (M <= 128)
+  constexpr int MT = 8;               // m-tiles per block (128 rows)
   __shared__ float red[3][128][16];   // waves 1-3 partials
 
   const int kq = (k / 32 + 3) / 4 * 32;  // per-wave K quota (mult of 32)
@@ -71,7 +72,7 @@ __global__ __launch_bounds__(256) void smallm_gemm_kernel(
     }
 #pragma unroll
     for (int mt = 0; mt < MT; ++mt) {
-      const int row = mt * 16 + cl;
+      const int row = m0 + mt * 16 + cl;
       bf16x8 af;
       if (row < m) {
         af = load_bf16x8_bits(a + (long)row * k + k0 + g * 8);
@@ -99,10 +100,11 @@ __global__ __launch_bounds__(256) void smallm_gemm_kernel(
   for (int mt = 0; mt < MT; ++mt) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      const int row = mt * 16 + g * 4 + r;
+      const int lrow = mt * 16 + g * 4 + r;  // LDS-local row
+      const int row = m0 + lrow;
       if (row >= m || !col_ok) continue;
-      float v = (acc[mt][r] + red[0][row][cl] + red[1][row][cl] +
-                 red[2][row][cl]) * alpha;
+      float v = (acc[mt][r] + red[0][lrow][cl] + red[1][lrow][cl] +
+                 red[2][lrow][cl]) * alpha;
       long off = (long)row * n + col;
       if (pre_mode == 1) {
         v += bf16_bits_to_float(pre[col]);
@@ -440,7 +442,7 @@ void smallm_gemm(torch::Tensor a, torch::Tensor wt,
   const int n = out.size(1);
   TORCH_CHECK(a.size(1) >= k && k % 32 == 0, "K%32");
   TORCH_CHECK(k == a.size(1), "a must be [M,K] exactly");
-  TORCH_CHECK(m <= 128, "smallm_gemm: M<=128");
+
   TORCH_CHECK(wt.size(0) >= n, "wt rows >= N");
   const long ldw = wt.size(1);
   TORCH_CHECK(wt_col0 + k <= ldw, "wt K slice OOB");
@@ -451,7 +453,8 @@ void smallm_gemm(torch::Tensor a, torch::Tensor wt,
     prep = (const unsigned short*)pre->data_ptr();
   }
   auto stream = at::cuda::getCurrentCUDAStream();
-  hipLaunchKernelGGL(smallm_gemm_kernel, dim3(cdiv(n, 16)), dim3(256), 0,
+  hipLaunchKernelGGL(smallm_gemm_kernel,
+                     dim3(cdiv(n, 16), cdiv(m, 128)), dim3(256), 0,
                      stream, (const unsigned short*)a.data_ptr(),
                      (const unsigned short*)wt.data_ptr() + wt_col0,
                      prep, (unsigned short*)out.data_ptr(), m, n, (int)k,

@@ -44,3 +44,80 @@ torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor bT) {
                      c.data_ptr<float>());
   return c;
 }
+
+// ---------------------------------------------------------------------------
+// ds_read_tr16_b64 probe: verifies the paneled-LDS B-fragment read used
+// by flash_attn bwd. LDS holds a [K=32][16] row-major bf16 panel (row
+// stride exactly 16 elements); per lane addr = panel + k_lane*32B +
+// col*2B with k_lane = (lane>>4)*8, col = lane&15; two tr-reads deliver
+// elems j=0..3 and 4..7 as B[k_lane + j][col] (guide T10 / learn_hip
+// m156 pattern). The probe runs a full MFMA against a regular A-frag so
+// the test checks end-to-end math, not just the gather.
+// ---------------------------------------------------------------------------
+
+typedef __attribute__((__vector_size__(4 * sizeof(__bf16)))) __bf16 bf16x4_v;
+#define LV_LDS __attribute__((address_space(3)))
+
+__device__ __forceinline__ bf16x8 lds_b_frag_tr16(const char* panel,
+                                                  int k0) {
+  const int lane = threadIdx.x & 63;
+  const int g = lane >> 4;
+  const int cl = lane & 15;
+  const char* addr = panel + ((k0 + g * 8) * 16 + cl) * 2;
+  bf16x4_v lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (LV_LDS bf16x4_v*)(addr));
+  bf16x4_v hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (LV_LDS bf16x4_v*)(addr + 4 * 16 * 2));
+  bf16x8 out;
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    out[j] = lo[j];
+    out[j + 4] = hi[j];
+  }
+  return out;
+}
+
+namespace {
+
+__global__ void tr16_probe_kernel(const unsigned short* __restrict__ a,
+                                  const unsigned short* __restrict__ b,
+                                  float* __restrict__ c) {
+  // b is the [32][16] row-major panel staged to LDS with plain
+  // vectorized copies.
+  __shared__ unsigned short panel[32 * 16];
+  for (int i = threadIdx.x; i < 32 * 16 / 8; i += 64) {
+    *reinterpret_cast<ushortx8*>(&panel[i * 8]) =
+        *reinterpret_cast<const ushortx8*>(b + i * 8);
+  }
+  __syncthreads();
+  const int lane = threadIdx.x & 63;
+  float af[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    af[j] = bf16_bits_to_float(a[(lane & 15) * 32 + (lane >> 4) * 8 + j]);
+  }
+  bf16x8 bfrag = lds_b_frag_tr16(reinterpret_cast<const char*>(panel), 0);
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  acc = mfma16x16x32_bf16(pack_bf16x8(af), bfrag, acc);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    c[((lane >> 4) * 4 + r) * 16 + (lane & 15)] = acc[r];
+  }
+}
+
+}  // namespace
+
+torch::Tensor tr16_probe(torch::Tensor a, torch::Tensor b) {
+  // a: [16, 32] bf16 row-major (A); b: [32, 16] bf16 row-major (B).
+  // Returns A @ B [16, 16] with B read through ds_read_tr16_b64.
+  TORCH_CHECK(a.is_cuda() && a.is_contiguous() && b.is_contiguous());
+  TORCH_CHECK(a.size(0) == 16 && a.size(1) == 32);
+  TORCH_CHECK(b.size(0) == 32 && b.size(1) == 16);
+  auto c = torch::empty({16, 16}, a.options().dtype(torch::kFloat32));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(tr16_probe_kernel, dim3(1), dim3(64), 0, stream,
+                     (const unsigned short*)a.data_ptr(),
+                     (const unsigned short*)b.data_ptr(),
+                     c.data_ptr<float>());
+  return c;
+}
