@@ -167,3 +167,79 @@ def SamplePoints(points: torch.Tensor, num_samples: int,
     dists = torch.minimum(dists,
                           (points - points[nxt]).pow(2).sum(-1))
   return torch.tensor(chosen, dtype=torch.long)
+
+
+def PointToGrid(points: torch.Tensor, num_points_per_cell: int,
+                x_intervals: int, y_intervals: int, z_intervals: int,
+                x_range, y_range, z_range, seed: int = 0):
+  """Bins points into an equally spaced 3-D grid (reference
+  tasks/car/ops/car_ops.cc:38 PointToGrid / point_grid_op.cc).
+
+  points [n, d] (first 3 dims = xyz). Returns (output_points
+  [gx, gy, gz, P, d], grid_centers [gx, gy, gz, 3], num_points
+  [gx, gy, gz]). Cells beyond capacity drop random points (shuffled);
+  short cells are padded with the cell center on xyz and zeros on the
+  remaining features.
+  """
+  n, d = points.shape
+  gx, gy, gz = x_intervals, y_intervals, z_intervals
+  device = points.device
+  sizes = torch.tensor(
+      [(x_range[1] - x_range[0]) / gx, (y_range[1] - y_range[0]) / gy,
+       (z_range[1] - z_range[0]) / gz], device=device)
+  lo = torch.tensor([x_range[0], y_range[0], z_range[0]], device=device)
+  cell = ((points[:, :3] - lo) / sizes).floor().long()
+  in_range = ((cell >= 0) & (cell < torch.tensor([gx, gy, gz],
+                                                 device=device))).all(-1)
+  flat = (cell[:, 0] * gy + cell[:, 1]) * gz + cell[:, 2]
+  flat = torch.where(in_range, flat, torch.full_like(flat, -1))
+
+  # Grid centers.
+  ix = torch.arange(gx, device=device)
+  iy = torch.arange(gy, device=device)
+  iz = torch.arange(gz, device=device)
+  cx, cy, cz = torch.meshgrid(ix, iy, iz, indexing='ij')
+  centers = torch.stack(
+      [(cx + 0.5) * sizes[0] + lo[0], (cy + 0.5) * sizes[1] + lo[1],
+       (cz + 0.5) * sizes[2] + lo[2]], dim=-1).float()
+
+  p = num_points_per_cell
+  out = torch.zeros(gx * gy * gz, p, d, device=device)
+  out[:, :, :3] = centers.reshape(-1, 1, 3)  # padding = cell center
+  counts = torch.zeros(gx * gy * gz, dtype=torch.int32, device=device)
+  g = torch.Generator(device='cpu').manual_seed(seed)
+  order = torch.randperm(n, generator=g).to(device)  # shuffle per doc
+  slots = {}
+  for i in order.tolist():
+    c = int(flat[i])
+    if c < 0:
+      continue
+    k = slots.get(c, 0)
+    if k < p:
+      out[c, k] = points[i]
+      slots[c] = k + 1
+  for c, k in slots.items():
+    counts[c] = k
+  return (out.reshape(gx, gy, gz, p, d), centers,
+          counts.reshape(gx, gy, gz))
+
+
+def BallQuery(points: torch.Tensor, centers: torch.Tensor, radius: float,
+              num_neighbors: int) -> torch.Tensor:
+  """Fixed-radius neighborhood gather (reference ps_utils.cc
+  neighborhood sampling): for each center, up to `num_neighbors` point
+  indices within `radius` (first index repeated as padding).
+  points [n, 3], centers [m, 3] -> [m, num_neighbors] long."""
+  d2 = (centers[:, None, :] - points[None, :, :]).pow(2).sum(-1)
+  within = d2 <= radius * radius
+  idx = torch.zeros(centers.shape[0], num_neighbors, dtype=torch.long,
+                    device=points.device)
+  for i in range(centers.shape[0]):
+    cand = within[i].nonzero(as_tuple=True)[0]
+    if cand.numel() == 0:
+      continue
+    take = cand[:num_neighbors]
+    idx[i, :take.numel()] = take
+    if take.numel() < num_neighbors:
+      idx[i, take.numel():] = take[0]
+  return idx
