@@ -55,6 +55,15 @@ class DualEncoder(BaseTask):
     p.Define('text_layers', 2, 'Text transformer layers.')
     p.Define('vocab_size', 1000, 'Caption vocab.')
     p.Define('temperature', 0.07, 'Softmax temperature.')
+    p.Define('score_function', 'dot',
+             "'dot' | 'bilinear' (learned W between towers; reference "
+             "milan/score_functions.py).")
+    p.Define('label_smoothing', 0.0,
+             'Label smoothing on the in-batch softmax loss.')
+    p.Define('id_feature', '',
+             'Optional batch field of example ids: pairs sharing an id '
+             'with the diagonal are masked out of the loss (reference '
+             'dual_encoder.py:52 id-based duplicate masking).')
     return p
 
   def __init__(self, params):
@@ -80,6 +89,9 @@ class DualEncoder(BaseTask):
     self.CreateChild('text_proj', lingvo_layers.ProjectionLayer.Params()
                      .Set(input_dim=p.text_dim, output_dim=p.joint_dim,
                           has_bias=True))
+    if p.score_function == 'bilinear':
+      self.CreateVariable('score_w', py_utils.WeightParams(
+          [p.joint_dim, p.joint_dim], p.params_init, p.dtype))
 
   def EncodeImage(self, theta, images):
     x = images.to(self.fprop_dtype)
@@ -104,12 +116,28 @@ class DualEncoder(BaseTask):
                           input_batch.text_paddings)
     return NestedMap(image_emb=img, text_emb=txt)
 
+  def Score(self, theta, image_emb, text_emb):
+    """Tower-pair similarity (reference score_functions.py: dot product
+    or learned bilinear form)."""
+    if self.p.score_function == 'bilinear':
+      return image_emb @ theta.score_w.float() @ text_emb.t()
+    return image_emb @ text_emb.t()
+
   def ComputeLoss(self, theta, predictions, input_batch):
     p = self.p
-    sims = predictions.image_emb @ predictions.text_emb.t() / p.temperature
+    sims = self.Score(theta, predictions.image_emb,
+                      predictions.text_emb) / p.temperature
     labels = torch.arange(sims.shape[0], device=sims.device)
-    loss_i2t = F.cross_entropy(sims, labels)
-    loss_t2i = F.cross_entropy(sims.t(), labels)
+    if p.id_feature and p.id_feature in input_batch:
+      # Mask off-diagonal pairs that are actually positives (same id):
+      # they must not be treated as negatives (reference id masking).
+      ids = input_batch[p.id_feature].reshape(-1)
+      dup = (ids[:, None] == ids[None, :]) &           ~torch.eye(len(ids), dtype=torch.bool, device=sims.device)
+      sims = sims.masked_fill(dup, -1e30)
+    loss_i2t = F.cross_entropy(sims, labels,
+                               label_smoothing=p.label_smoothing)
+    loss_t2i = F.cross_entropy(sims.t(), labels,
+                               label_smoothing=p.label_smoothing)
     loss = 0.5 * (loss_i2t + loss_t2i)
     acc = (sims.argmax(-1) == labels).float().mean()
     w = torch.tensor(float(sims.shape[0]))

@@ -235,3 +235,23 @@ def test_ball_query():
   idx = car_ops.BallQuery(pts, centers, radius=0.5, num_neighbors=3)
   assert set(idx[0].tolist()) <= {0, 1, 3}
   assert idx[1, 0] == 2 and idx[1, 1] == 2  # padding repeats first
+
+
+def test_milan_score_functions_and_id_masking():
+  from lingvo_amd.models import milan
+  p = milan.DualEncoder.Params().Set(
+      name='de', joint_dim=32, image_channels=[8], text_dim=32,
+      text_layers=1, vocab_size=100, score_function='bilinear',
+      label_smoothing=0.1, id_feature='ids', random_seed=3)
+  p.input = milan.SyntheticImageTextInput.Params().Set(
+      batch_size=4, image_size=16, text_len=6, vocab_size=100)
+  task = p.Instantiate()
+  batch = task.GetInputBatch()
+  batch.ids = torch.tensor([0, 0, 1, 2])  # examples 0,1 share an id
+  m = task.TrainStep(batch)
+  assert torch.isfinite(m['loss'][0])
+  assert hasattr(task, 'score_w')
+  # id masking: duplicate pair contributes -inf logits, loss finite
+  preds = task.ComputePredictions(task.theta, batch)
+  mm, _ = task.ComputeLoss(task.theta, preds, batch)
+  assert torch.isfinite(mm.loss[0])
