@@ -35,8 +35,10 @@ def build_argparser():
   ap.add_argument('--gpus', type=int, default=1)
   ap.add_argument('--steps', type=int, default=20)
   ap.add_argument('--warmup', type=int, default=5)
-  ap.add_argument('--batch', type=int, default=128,
-                  help='Per-GPU batch size.')
+  ap.add_argument('--batch', type=int, default=512,
+                  help='Per-GPU batch size (512 saturates one MI355X at '
+                       'Conformer-L: 921 ex/s vs 674 at 128, using '
+                       '~101 of 288 GB HBM; see BASELINE.md sweep).')
   ap.add_argument('--memory', action='store_true',
                   help='Report peak device memory in the JSON config.')
   ap.add_argument('--profile', default=None, metavar='PATH',
