@@ -46,24 +46,34 @@ torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor bT) {
 }
 
 // ---------------------------------------------------------------------------
-// ds_read_tr16_b64 probe: verifies the paneled-LDS B-fragment read used
-// by flash_attn bwd. LDS holds a [K=32][16] row-major bf16 panel (row
-// stride exactly 16 elements); per lane addr = panel + k_lane*32B +
-// col*2B with k_lane = (lane>>4)*8, col = lane&15; two tr-reads deliver
-// elems j=0..3 and 4..7 as B[k_lane + j][col] (guide T10 / learn_hip
-// m156 pattern). The probe runs a full MFMA against a regular A-frag so
-// the test checks end-to-end math, not just the gather.
+// ds_read_tr16_b64 probe. HW semantics VERIFIED on gfx950
+// (tools/tr16_diag.hip raw-pattern dump): given a per-lane element
+// address a, the instruction reads the 16-element-aligned block
+// containing a as a row-major 4x4 bf16 tile and returns COLUMN (a & 3)
+// — out_j = (a & ~0xF) + 4*j + (a & 3). Address bits 2-3 are ignored.
+// (The guide's m156 [4][16] formula describes a different per-lane
+// address layout; the 4x4-tile model here is the ground truth we
+// measured.)
+//
+// Consequence: a transpose-readable B panel must be stored TILED as
+// [K/4][NCOLS/4][4][4]; a lane building the MFMA B-frag
+// B[k0 + g*8 + j][col] issues two tr-reads at the blocks
+// (k0/4 + 2g, col/4) and (k0/4 + 2g + 1, col/4) with low bits col&3.
 // ---------------------------------------------------------------------------
 
 typedef __attribute__((__vector_size__(4 * sizeof(__bf16)))) __bf16 bf16x4_v;
 #define LV_LDS __attribute__((address_space(3)))
 
+// panel: tiled [K/4][4 col-blocks][4][4] layout (16-col logical panel).
 __device__ __forceinline__ bf16x8 lds_b_frag_tr16(const char* panel,
                                                   int k0) {
   const int lane = threadIdx.x & 63;
   const int g = lane >> 4;
   const int cl = lane & 15;
-  const char* addr = panel + ((k0 + g * 8) * 16 + cl) * 2;
+  const int kb = (k0 + g * 8) / 4;       // first 4-row block
+  const int cb = cl / 4;
+  const char* addr =
+      panel + (((kb * 4 + cb) * 16) + (cl & 3)) * 2;
   bf16x4_v lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
       (LV_LDS bf16x4_v*)(addr));
   bf16x4_v hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
@@ -82,12 +92,13 @@ namespace {
 __global__ void tr16_probe_kernel(const unsigned short* __restrict__ a,
                                   const unsigned short* __restrict__ b,
                                   float* __restrict__ c) {
-  // b is the [32][16] row-major panel staged to LDS with plain
-  // vectorized copies.
+  // Stage b [32][16] row-major into the TILED [8][4][4][4] layout.
   __shared__ unsigned short panel[32 * 16];
-  for (int i = threadIdx.x; i < 32 * 16 / 8; i += 64) {
-    *reinterpret_cast<ushortx8*>(&panel[i * 8]) =
-        *reinterpret_cast<const ushortx8*>(b + i * 8);
+  for (int i = threadIdx.x; i < 32 * 16; i += 64) {
+    const int row = i / 16;
+    const int col = i % 16;
+    panel[((row / 4) * 4 + col / 4) * 16 + (row % 4) * 4 + (col % 4)] =
+        b[i];
   }
   __syncthreads();
   const int lane = threadIdx.x & 63;
