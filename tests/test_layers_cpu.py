@@ -670,3 +670,41 @@ def test_chunked_softmax_matches_full():
   chunked = run(16)  # 50 classes in chunks of 16 (ragged tail)
   for a, b in zip(full, chunked):
     assert torch.allclose(a, b, atol=1e-4), (a - b).abs().max()
+
+
+def test_depthwise_conv2d_with_padding():
+  from lingvo_amd.layers import conv_layers_with_time_padding as ctp
+  p = ctp.DepthwiseConv2DLayer.Params().Set(
+      name='dw2', filter_shape=(3, 3, 8, 2), filter_stride=(2, 1),
+      random_seed=1)
+  layer = p.Instantiate()
+  x = torch.randn(2, 12, 6, 8)
+  pad = py_utils.PaddingsFromLengths(torch.tensor([12, 7]), 12)
+  out, opad = layer.FProp(layer.theta, x, pad)
+  assert out.shape == (2, 6, 6, 16)  # stride-2 time, mult=2 channels
+  assert opad.shape == (2, 6)
+  # causal variant: future mutation does not leak
+  pc = ctp.CausalDepthwiseConv2DLayer.Params().Set(
+      name='cdw2', filter_shape=(3, 3, 8, 1), random_seed=1)
+  cl = pc.Instantiate()
+  o1, _ = cl.FProp(cl.theta, x, pad)
+  x2 = x.clone()
+  x2[:, 6:] = 9.0
+  o2, _ = cl.FProp(cl.theta, x2, pad)
+  assert torch.allclose(o1[:, :6], o2[:, :6], atol=1e-5)
+
+
+def test_depthwise_conv1d_dilation():
+  from lingvo_amd.layers import conv_layers_with_time_padding as ctp
+  p = ctp.DepthwiseConv1DLayer.Params().Set(
+      name='dil', kernel_size=3, dim=4, dilation=2, is_causal=True,
+      random_seed=1)
+  layer = p.Instantiate()
+  x = torch.randn(1, 10, 4)
+  out, _ = layer.FProp(layer.theta, x, torch.zeros(1, 10))
+  # causal dilated: out[t] uses x[t], x[t-2], x[t-4]
+  want0 = x[:, 0] * layer.theta.w[2] + layer.theta.b
+  assert torch.allclose(out[:, 0], want0, atol=1e-5)
+  want4 = (x[:, 4] * layer.theta.w[2] + x[:, 2] * layer.theta.w[1] +
+           x[:, 0] * layer.theta.w[0] + layer.theta.b)
+  assert torch.allclose(out[:, 4], want4, atol=1e-5)
