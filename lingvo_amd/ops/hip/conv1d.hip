@@ -105,24 +105,31 @@ __global__ __launch_bounds__(256) void dwconv_bwd_dw(
     const unsigned short* __restrict__ x, float* __restrict__ dw_acc,
     float* __restrict__ db_acc, int B, int T, int D, int K, int pad,
     int ngroups) {
+  // Thread decomposition (fully vectorized LDS reads): 16 d-groups of
+  // 8 channels x 4 tap-quarters of 8 taps x 4 row-quarters of 16 rows.
+  // Every x/dy access is a ushortx8 load; accumulators live in
+  // registers across all chunks and flush once.
   __shared__ unsigned short x_s[DW_CHUNK + MAXK - 1][DW_DBLK];
   __shared__ unsigned short dy_s[DW_CHUNK][DW_DBLK];
   const int tid = threadIdx.x;
-  const int d_local = tid % DW_DBLK;
-  const int jg = tid / DW_DBLK;  // tap parity: owns taps j with j%2==jg
-  const int d = blockIdx.x * DW_DBLK + d_local;
+  const int dg = tid & 15;          // d-group (8 channels)
+  const int tq = (tid >> 4) & 3;    // tap quarter (taps 8tq..8tq+7)
+  const int rq = tid >> 6;          // row quarter (rows 16rq..16rq+15)
+  const int d0 = blockIdx.x * DW_DBLK + dg * 8;
   const long rows = (long)B * T;
   const int xrows = DW_CHUNK + K - 1;
 
-  float dw[MAXK / 2 + 1];
-  for (int j = 0; j < MAXK / 2 + 1; ++j) dw[j] = 0.f;
-  float db = 0.f;
-  const bool d_ok = d < D;
+  float dw8[8][8];
+#pragma unroll
+  for (int jj = 0; jj < 8; ++jj) {
+#pragma unroll
+    for (int e = 0; e < 8; ++e) dw8[jj][e] = 0.f;
+  }
+  float db8[8] = {0, 0, 0, 0, 0, 0, 0, 0};
 
   const long nchunks = (rows + DW_CHUNK - 1) / DW_CHUNK;
   for (long c = blockIdx.y; c < nchunks; c += ngroups) {
     const long r0 = c * DW_CHUNK;
-    // Stage x window [r0 - pad, r0 - pad + xrows) and dy [r0, r0+CHUNK).
     for (int i = tid; i < xrows * (DW_DBLK / 8); i += 256) {
       int row = i / (DW_DBLK / 8);
       int d8 = (i % (DW_DBLK / 8)) * 8;
@@ -151,28 +158,54 @@ __global__ __launch_bounds__(256) void dwconv_bwd_dw(
     }
     __syncthreads();
 
-    if (d_ok) {
-      const int nrows = (int)min((long)DW_CHUNK, rows - r0);
-      for (int lr = 0; lr < nrows; ++lr) {
-        float g = bf16_bits_to_float(dy_s[lr][d_local]);
-        const int t = (int)((r0 + lr) % T);
-        if (jg == 0) db += g;
-        for (int j = jg, ji = 0; j < K; j += 2, ++ji) {
-          int ts = t + j - pad;
-          if (ts >= 0 && ts < T) {
-            dw[ji] += g * bf16_bits_to_float(x_s[lr + j][d_local]);
-          }
+    const int nrows = (int)min((long)DW_CHUNK, rows - r0);
+    const int lr_hi = min((rq + 1) * 16, nrows);
+    for (int lr = rq * 16; lr < lr_hi; ++lr) {
+      ushortx8 gv = *reinterpret_cast<const ushortx8*>(&dy_s[lr][dg * 8]);
+      float g[8];
+#pragma unroll
+      for (int e = 0; e < 8; ++e) g[e] = bf16_bits_to_float(gv[e]);
+      const int t = (int)((r0 + lr) % T);
+      if (tq == 0) {
+#pragma unroll
+        for (int e = 0; e < 8; ++e) db8[e] += g[e];
+      }
+#pragma unroll
+      for (int jj = 0; jj < 8; ++jj) {
+        const int j = tq * 8 + jj;
+        if (j >= K) break;
+        const int ts = t + j - pad;
+        if (ts < 0 || ts >= T) continue;
+        ushortx8 xv =
+            *reinterpret_cast<const ushortx8*>(&x_s[lr + j][dg * 8]);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          dw8[jj][e] += g[e] * bf16_bits_to_float(xv[e]);
         }
       }
     }
     __syncthreads();
   }
 
-  if (d_ok) {
-    for (int j = jg, ji = 0; j < K; j += 2, ++ji) {
-      if (dw[ji] != 0.f) atomicAdd(dw_acc + (long)j * D + d, dw[ji]);
+  if (d0 + 7 < D || d0 < D) {
+#pragma unroll
+    for (int jj = 0; jj < 8; ++jj) {
+      const int j = tq * 8 + jj;
+      if (j >= K) break;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        if (d0 + e < D && dw8[jj][e] != 0.f) {
+          atomicAdd(dw_acc + (long)j * D + d0 + e, dw8[jj][e]);
+        }
+      }
     }
-    if (jg == 0 && db != 0.f) atomicAdd(db_acc + d, db);
+    if (tq == 0) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        if (d0 + e < D && db8[e] != 0.f) atomicAdd(db_acc + d0 + e,
+                                                   db8[e]);
+      }
+    }
   }
 }
 
