@@ -46,19 +46,20 @@ torch::Tensor mfma_probe(torch::Tensor a, torch::Tensor bT) {
 }
 
 // ---------------------------------------------------------------------------
-// ds_read_tr16_b64 probe. HW semantics VERIFIED on gfx950
-// (tools/tr16_diag.hip raw-pattern dump): given a per-lane element
-// address a, the instruction reads the 16-element-aligned block
-// containing a as a row-major 4x4 bf16 tile and returns COLUMN (a & 3)
-// — out_j = (a & ~0xF) + 4*j + (a & 3). Address bits 2-3 are ignored.
-// (The guide's m156 [4][16] formula describes a different per-lane
-// address layout; the 4x4-tile model here is the ground truth we
-// measured.)
-//
-// Consequence: a transpose-readable B panel must be stored TILED as
-// [K/4][NCOLS/4][4][4]; a lane building the MFMA B-frag
-// B[k0 + g*8 + j][col] issues two tr-reads at the blocks
-// (k0/4 + 2g, col/4) and (k0/4 + 2g + 1, col/4) with low bits col&3.
+// ds_read_tr16_b64 exploration probe. Measured gfx950 semantics
+// (tools/tr16_diag.hip + tr16_diag2.hip raw-pattern dumps):
+//   out(lane l, elem j) = LDS[ addr supplied by lane (l&~15)|((l&3)+4j) ]
+// within each 16-lane group — a cooperative cross-lane gather where
+// every source element is REPLICATED into 4 output lanes (lanes 0,4,8,
+// 12 of a group produce identical results). A single read therefore
+// covers only 4 distinct output columns, NOT the 16-distinct-column
+// MFMA B fragment this probe originally assumed; building B-frags needs
+// a different (multi-read / different-layout) composition, as in
+// HipKittens' 192-read kernels. Since phase-skip attribution measured
+// the transposed-staging cost this would remove at only ~4% of
+// flash-attention backward, the kernels keep their scalar-write
+// transposed staging; this probe + the diag tools document the measured
+// instruction behavior for future schedule work.
 // ---------------------------------------------------------------------------
 
 typedef __attribute__((__vector_size__(4 * sizeof(__bf16)))) __bf16 bf16x4_v;

@@ -325,14 +325,3 @@ def test_moe_positions_kernel_matches_cumsum():
   assert torch.equal(c1.long(), count1.long())
 
 
-@gpu
-def test_tr16_transpose_read_probe():
-  """ds_read_tr16_b64 paneled B-frag == torch.matmul (asymmetric
-  random inputs; transpose-detecting)."""
-  import lingvo_amd.ops._lingvo_ops as ext
-  torch.manual_seed(13)
-  a = torch.randn(16, 32, device='cuda', dtype=torch.bfloat16)
-  b = torch.randn(32, 16, device='cuda', dtype=torch.bfloat16)
-  c = ext.tr16_probe(a, b)
-  ref = a.float() @ b.float()
-  assert (c - ref).abs().max() < 0.1, (c - ref).abs().max()
