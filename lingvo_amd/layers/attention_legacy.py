@@ -300,3 +300,151 @@ class MergerLayer(BaseLayer):
       gates = torch.softmax(torch.matmul(flat, theta.gate_w), dim=-1)
       return torch.einsum('...nd,...n->...d', x, gates.to(x.dtype))
     raise ValueError(p.merger_op)
+
+
+class MultiHeadedAttention(BaseAttentionLayer):
+  """Per-step multi-headed wrapper around an inner attention
+  (reference attention.py:1425): projects sources/queries/contexts per
+  head, folds heads into the batch, runs `inner_atten` per head, and
+  combines with an output projection."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('num_attention_heads', 4, 'Heads.')
+    p.Define('inner_atten_tpl', DotProductAttention.Params(),
+             'Per-head attention template.')
+    p.Define('context_dim', 0, 'Value dim (defaults to source_dim).')
+    p.Define('use_source_vec_as_attention_value', True,
+             'Use keys as values (reference default).')
+    p.Define('enable_query_proj', True, 'Project queries.')
+    p.Define('enable_ctx_post_proj', True, 'Output projection.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    n = p.num_attention_heads
+    assert p.hidden_dim % n == 0, 'hidden_dim % heads'
+    ctx_dim = p.context_dim or p.source_dim
+    self.CreateVariable('source_proj', py_utils.WeightParams(
+        [p.source_dim, p.hidden_dim], p.params_init, p.dtype))
+    if p.enable_query_proj:
+      self.CreateVariable('query_proj', py_utils.WeightParams(
+          [p.query_dim, p.hidden_dim], p.params_init, p.dtype))
+    if not p.use_source_vec_as_attention_value:
+      self.CreateVariable('ctx_proj', py_utils.WeightParams(
+          [ctx_dim, p.hidden_dim], p.params_init, p.dtype))
+    if p.enable_ctx_post_proj:
+      self.CreateVariable('ctx_post_proj', py_utils.WeightParams(
+          [p.hidden_dim, p.hidden_dim], p.params_init, p.dtype))
+    inner = p.inner_atten_tpl.Copy().Set(
+        name='inner', source_dim=p.hidden_dim // n,
+        query_dim=p.hidden_dim // n, hidden_dim=p.hidden_dim // n,
+        atten_dropout_prob=p.atten_dropout_prob)
+    self.CreateChild('inner', inner)
+
+  def InitForSourcePacked(self, theta, source_vecs, source_contexts,
+                          source_padding):
+    p = self.p
+    n = p.num_attention_heads
+    b, s, _ = source_vecs.shape
+    h = p.hidden_dim // n
+    keys = torch.matmul(source_vecs, theta.source_proj)
+    if p.use_source_vec_as_attention_value or source_contexts is None:
+      vals = keys
+    else:
+      vals = torch.matmul(source_contexts, theta.ctx_proj)
+    # fold heads into batch: [B*N, S, H]
+    keys = keys.reshape(b, s, n, h).permute(0, 2, 1, 3).reshape(
+        b * n, s, h)
+    vals = vals.reshape(b, s, n, h).permute(0, 2, 1, 3).reshape(
+        b * n, s, h)
+    pad = source_padding.repeat_interleave(n, dim=0)
+    packed = self.inner.InitForSourcePacked(theta.inner, keys, vals, pad)
+    packed.batch = b
+    return packed
+
+  def ZeroAttentionState(self, source_len, batch, device=None,
+                         dtype=torch.float32):
+    return self.inner.ZeroAttentionState(
+        source_len, batch * self.p.num_attention_heads, device, dtype)
+
+  def ComputeContextVector(self, theta, packed, query_vec, state):
+    p = self.p
+    n = p.num_attention_heads
+    b = packed.batch
+    h = p.hidden_dim // n
+    q = query_vec
+    if p.enable_query_proj:
+      q = torch.matmul(q, theta.query_proj)
+    q = q.reshape(b, n, h).reshape(b * n, h)
+    ctx, probs, state = self.inner.ComputeContextVector(
+        theta.inner, packed, q, state)
+    ctx = ctx.reshape(b, n * h)
+    if p.enable_ctx_post_proj:
+      ctx = torch.matmul(ctx, theta.ctx_post_proj)
+    # head-averaged probs (reference returns per-head; average for the
+    # [B, S] contract used by decoder callbacks)
+    probs = probs.reshape(b, n, -1).mean(dim=1)
+    return ctx, probs, state
+
+
+class MultiSourceAttention(BaseAttentionLayer):
+  """Attention over multiple named source sets, merged
+  (reference attention.py:3856): one child attention per source,
+  combined by a MergerLayer."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('source_atten_tpls', [],
+             'List of (name, attention Params).')
+    p.Define('primary_source_key', '',
+             'Source whose probs are returned (default: first).')
+    p.Define('atten_merger_tpl', None,
+             'MergerLayer params (default: mean).')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    self._names = [name for name, _ in p.source_atten_tpls]
+    for name, tpl in p.source_atten_tpls:
+      self.CreateChild(f'atten_{name}', tpl.Copy())
+    if p.atten_merger_tpl is not None:
+      self.CreateChild('merger', p.atten_merger_tpl.Copy())
+
+  def InitForSourcePacked(self, theta, source_vecs, source_contexts,
+                          source_padding):
+    """source_vecs/contexts/padding: NestedMaps keyed by source name."""
+    packed = NestedMap()
+    for name in self._names:
+      child = getattr(self, f'atten_{name}')
+      packed[name] = child.InitForSourcePacked(
+          theta[f'atten_{name}'], source_vecs[name],
+          None if source_contexts is None else source_contexts[name],
+          source_padding[name])
+    return packed
+
+  def ZeroAttentionState(self, source_len, batch, device=None,
+                         dtype=torch.float32):
+    return NestedMap()
+
+  def ComputeContextVector(self, theta, packed, query_vec, state):
+    p = self.p
+    ctxs = []
+    primary_probs = None
+    primary = p.primary_source_key or self._names[0]
+    for name in self._names:
+      child = getattr(self, f'atten_{name}')
+      ctx, probs, _ = child.ComputeContextVector(
+          theta[f'atten_{name}'], packed[name], query_vec, NestedMap())
+      ctxs.append(ctx)
+      if name == primary:
+        primary_probs = probs
+    if p.atten_merger_tpl is not None:
+      merged = self.merger.FProp(theta.merger, ctxs)
+    else:
+      merged = torch.stack(ctxs).mean(dim=0)
+    return merged, primary_probs, state

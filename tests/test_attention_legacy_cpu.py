@@ -118,3 +118,44 @@ def test_merger_layer_modes():
   m = al.MergerLayer.Params().Set(name='m', merger_op='mean').Instantiate()
   assert torch.allclose(m.FProp(m.theta, xs),
                         (xs[0] + xs[1] + xs[2]) / 3, atol=1e-6)
+
+
+def test_legacy_multiheaded_wrapper():
+  from lingvo_amd.layers import attention_legacy as al
+  p = al.MultiHeadedAttention.Params().Set(
+      name='mha', source_dim=16, query_dim=12, hidden_dim=32,
+      num_attention_heads=4, random_seed=1)
+  layer = p.Instantiate()
+  src = torch.randn(2, 7, 16)
+  pad = torch.zeros(2, 7)
+  pad[1, 5:] = 1.0
+  packed = layer.InitForSourcePacked(layer.theta, src, None, pad)
+  q = torch.randn(2, 12)
+  ctx, probs, _ = layer.ComputeContextVector(
+      layer.theta, packed, q, layer.ZeroAttentionState(7, 2))
+  assert ctx.shape == (2, 32)
+  assert probs.shape == (2, 7)
+  # padded keys get ~zero prob; probs normalized
+  assert float(probs[1, 5:].sum()) < 1e-6
+  assert torch.allclose(probs.sum(-1), torch.ones(2), atol=1e-5)
+  ctx.sum().backward()
+  assert layer.source_proj.grad is not None
+
+
+def test_multi_source_attention():
+  from lingvo_amd.core.nested_map import NestedMap
+  from lingvo_amd.layers import attention_legacy as al
+  mk = lambda name: (name, al.DotProductAttention.Params().Set(
+      source_dim=8, query_dim=8, hidden_dim=8, random_seed=1))
+  p = al.MultiSourceAttention.Params().Set(
+      name='ms', source_atten_tpls=[mk('a'), mk('b')],
+      primary_source_key='b')
+  layer = p.Instantiate()
+  srcs = NestedMap(a=torch.randn(2, 5, 8), b=torch.randn(2, 3, 8))
+  pads = NestedMap(a=torch.zeros(2, 5), b=torch.zeros(2, 3))
+  packed = layer.InitForSourcePacked(layer.theta, srcs, None, pads)
+  q = torch.randn(2, 8)
+  ctx, probs, _ = layer.ComputeContextVector(layer.theta, packed, q,
+                                             NestedMap())
+  assert ctx.shape == (2, 8)
+  assert probs.shape == (2, 3)  # primary source 'b' probs
