@@ -664,3 +664,22 @@ def test_bpe_tokenizer_roundtrip():
   # StringsToIds padding contract
   ids_t = tok.StringsToIds(['low', 'lower low'], max_length=6)[0]
   assert ids_t.shape == (2, 6)
+
+
+def test_symbolic_shapes():
+  from lingvo_amd.core import symbolic, tshape
+  b = symbolic.NewSymbol('batch')
+  t = symbolic.NewSymbol('time')
+  s = tshape.Shape([b, t * 2, 128])
+  flops = s.num_elements() * 4
+  with symbolic.SymbolToValueMap({b: 8, t: 10}):
+    assert s.ToTensorShape() == [8, 20, 128]
+    assert symbolic.EvalExpr(flops) == 8 * 20 * 128 * 4
+    # nesting overrides then restores
+    with symbolic.SymbolToValueMap({t: 3}):
+      assert s.ToTensorShape() == [8, 6, 128]
+    assert s.ToTensorShape() == [8, 20, 128]
+  cat = s[:2] + tshape.Shape([4])
+  with symbolic.SymbolToValueMap({'batch': 2, 'time': 5}):
+    assert cat.ToTensorShape() == [2, 10, 4]
+  assert symbolic.IsExpr(b * 2) and not symbolic.IsExpr(7)
