@@ -232,54 +232,121 @@ class ConformerLayer(BaseLayer):
 
 
 class _Conv3x3S2Nhwc(torch.autograd.Function):
-  """3x3 stride-2 same-pad conv on NHWC input as 9 offset GEMMs over
-  strided slices of the padded input.
+  """3x3 stride-2 same-pad conv on NHWC input via SPACE-TO-DEPTH +
+  4 flat GEMMs.
 
-  Replaces the im2col+col2im formulation: the K=9*C contraction is
-  decomposed into 9 plain [B*Ho*Wo, C] x [C, Co] hipBLASLt GEMMs that
-  accumulate in-place (addmm_ beta=1), so NO cols buffer exists in
-  either direction. The old path materialized ~2 GB cols buffers whose
-  col2im backward overflowed 32-bit byte offsets (GPU write faults at
-  large batch, heap-layout dependent) and cost ~90 ms/step at B=128."""
+  The input is space-to-depth'd once (2x2 cells -> 4C channels, block
+  order [(a0,b1),(a1,b1),(a1,b0),(a0,b0)]) into a zero-padded flat
+  buffer X [B*(Ho+1)*(Wo+1), 4C]. In s2d coordinates the 9 conv taps
+  collapse into FOUR cell offsets {(-1,-1),(0,-1),(-1,0),(0,0)}; each
+  is a UNIFORM ROW OFFSET of the flat buffer and touches a CONTIGUOUS
+  channel range, so forward is 4 addmm_ calls on lda-strided views
+  with in-GEMM accumulation — no im2col, no slice copies, and the
+  backward's dX lands via 4 sequential in-place GEMMs (no strided
+  scatter-adds). Two earlier formulations lost here: im2col's ~2 GB
+  cols buffers overflowed 32-bit offsets in the upstream col2im (the
+  round-1 GPU faults), and the 9-offset strided-slice form spent
+  ~70 ms/step at B=512 on backward copies + skinny GEMMs.
+
+  Tap map (cell = (dtau, dphi); within a cell, taps ordered by s2d
+  block; bijection (cell, block) <-> (dt, df)):
+    (-1,-1): [w[0,0]]                     blocks [s1]
+    ( 0,-1): [w[1,0], w[2,0]]             blocks [s0, s1]
+    (-1, 0): [w[0,2], w[0,1]]             blocks [s1, s2]
+    ( 0, 0): [w[1,2], w[2,2], w[2,1], w[1,1]]  blocks [s0..s3]
+  """
+
+  CELLS = [
+      ((-1, -1), [(0, 0, 1)]),
+      ((0, -1), [(1, 0, 0), (2, 0, 1)]),
+      ((-1, 0), [(0, 2, 1), (0, 1, 2)]),
+      ((0, 0), [(1, 2, 0), (2, 2, 1), (2, 1, 2), (1, 1, 3)]),
+  ]
+
+  @staticmethod
+  def _s2d(x):
+    """[B,H,W,C] (H,W even) -> padded flat [B*(Ho+1)*(Wo+1), 4C]."""
+    B, H, W, C = x.shape
+    Ho, Wo = H // 2, W // 2
+    xr = x.reshape(B, Ho, 2, Wo, 2, C).permute(0, 1, 3, 2, 4, 5)
+    xr = xr.reshape(B, Ho, Wo, 4, C)
+    # (a,b) flat index a*2+b; block order [(0,1),(1,1),(1,0),(0,0)].
+    order = torch.tensor([1, 3, 2, 0], device=x.device)
+    xr = xr.index_select(3, order).reshape(B, Ho, Wo, 4 * C)
+    X = x.new_zeros(B, Ho + 1, Wo + 1, 4 * C)
+    X[:, 1:, 1:, :] = xr
+    return X.reshape(B * (Ho + 1) * (Wo + 1), 4 * C), Ho, Wo
+
+  @staticmethod
+  def _cell_weight(w, taps, transpose=False):
+    """Per-cell weight [nblk*C, Co] from taps sorted by block."""
+    mats = [w[dt, df] for (dt, df, _) in
+            sorted(taps, key=lambda t: t[2])]
+    cat = torch.cat(mats, dim=0)
+    return cat.t().contiguous() if transpose else cat.contiguous()
 
   @staticmethod
   def forward(ctx, x, w, bias):
     """x [B,H,W,C] NHWC; w [3,3,C,Co] (kh,kw,cin,cout); bias [Co]."""
     B, H, W, C = x.shape
     Co = w.shape[3]
-    Ho, Wo = (H - 1) // 2 + 1, (W - 1) // 2 + 1
-    xp = F.pad(x, (0, 0, 1, 1, 1, 1))  # pad W then H
-    R = B * Ho * Wo
-    out = bias.to(x.dtype).expand(R, Co).contiguous()
-    for dt in range(3):
-      for df in range(3):
-        sl = xp[:, dt:dt + 2 * Ho - 1:2,
-                df:df + 2 * Wo - 1:2, :].contiguous().reshape(R, C)
-        out.addmm_(sl, w[dt, df])
+    pad_h, pad_w = H % 2, W % 2
+    if pad_h or pad_w:
+      x = F.pad(x, (0, 0, 0, pad_w, 0, pad_h))
+    Hp2, Wp2 = x.shape[1], x.shape[2]
+    X, Ho, Wo = _Conv3x3S2Nhwc._s2d(x)
+    Wp = Wo + 1
+    Rp = B * (Ho + 1) * Wp
+    base = Wp + 1  # first real out row of batch 0
+    O = bias.to(x.dtype).expand(Rp, Co).contiguous()
+    for (dtau, dphi), taps in _Conv3x3S2Nhwc.CELLS:
+      off = dtau * Wp + dphi
+      blks = sorted(t[2] for t in taps)
+      k0, k1 = blks[0] * C, (blks[-1] + 1) * C
+      wc = _Conv3x3S2Nhwc._cell_weight(w.to(x.dtype), taps)
+      O[base:].addmm_(X[base + off:Rp + off, k0:k1], wc)
     ctx.save_for_backward(x, w)
-    ctx.dims = (B, H, W, C, Ho, Wo, Co)
+    ctx.dims = (B, H, W, C, Ho, Wo, Co, pad_h, pad_w)
     ctx.bias_dtype = bias.dtype
-    return out.reshape(B, Ho, Wo, Co)
+    out = O.reshape(B, Ho + 1, Wp, Co)[:, 1:, 1:, :]
+    return out.contiguous()
 
   @staticmethod
   def backward(ctx, dout):
     x, w = ctx.saved_tensors
-    B, H, W, C, Ho, Wo, Co = ctx.dims
-    R = B * Ho * Wo
-    dout = dout.reshape(R, Co)
-    xp = F.pad(x, (0, 0, 1, 1, 1, 1))
-    dxp = torch.zeros_like(xp)
+    B, H, W, C, Ho, Wo, Co, pad_h, pad_w = ctx.dims
+    Wp = Wo + 1
+    Rp = B * (Ho + 1) * Wp
+    base = Wp + 1
+    dO = dout.new_zeros(B, Ho + 1, Wp, Co)
+    dO[:, 1:, 1:, :] = dout.reshape(B, Ho, Wo, Co)
+    dO = dO.reshape(Rp, Co)
+    X, _, _ = _Conv3x3S2Nhwc._s2d(x)
+    dX = torch.zeros_like(X)
     dw = torch.empty_like(w)
-    for dt in range(3):
-      for df in range(3):
-        sl = xp[:, dt:dt + 2 * Ho - 1:2,
-                df:df + 2 * Wo - 1:2, :].contiguous().reshape(R, C)
-        dw[dt, df] = (sl.t() @ dout).to(w.dtype)
-        partial = (dout @ w[dt, df].t()).reshape(B, Ho, Wo, C)
-        # Offsets overlap in the padded buffer: accumulate sequentially.
-        dxp[:, dt:dt + 2 * Ho - 1:2, df:df + 2 * Wo - 1:2, :] += partial
-    dx = dxp[:, 1:1 + H, 1:1 + W, :].contiguous()
-    dbias = dout.float().sum(0).to(ctx.bias_dtype)
+    for (dtau, dphi), taps in _Conv3x3S2Nhwc.CELLS:
+      off = dtau * Wp + dphi
+      blks = sorted(t[2] for t in taps)
+      k0, k1 = blks[0] * C, (blks[-1] + 1) * C
+      # dW_cell = X_slice^T @ dO: one GEMM per cell, rows split back to
+      # taps (each block within a cell is exactly one tap).
+      g = X[base + off:Rp + off, k0:k1].t() @ dO[base:]
+      for i, (dt, df, _) in enumerate(sorted(taps,
+                                             key=lambda t: t[2])):
+        dw[dt, df] = g[i * C:(i + 1) * C].to(w.dtype)
+      # dX rows/channels overlap across cells: sequential in-place.
+      wc_t = _Conv3x3S2Nhwc._cell_weight(w.to(dO.dtype), taps,
+                                         transpose=True)
+      dX[base + off:Rp + off, k0:k1].addmm_(dO[base:], wc_t)
+    # Inverse s2d.
+    dXr = dX.reshape(B, Ho + 1, Wp, 4 * C)[:, 1:, 1:, :]
+    dXr = dXr.reshape(B, Ho, Wo, 4, C)
+    inv_order = torch.tensor([3, 0, 2, 1], device=x.device)
+    dXr = dXr.index_select(3, inv_order).reshape(B, Ho, Wo, 2, 2, C)
+    dx = dXr.permute(0, 1, 3, 2, 4, 5).reshape(B, 2 * Ho, 2 * Wo, C)
+    if pad_h or pad_w:
+      dx = dx[:, :2 * Ho - pad_h, :2 * Wo - pad_w, :].contiguous()
+    dbias = dO[base:].float().sum(0).to(ctx.bias_dtype)
     return dx, dw, dbias
 
 
