@@ -231,6 +231,20 @@ class ConformerLayer(BaseLayer):
     return self.final_ln.FProp(theta.final_ln, x), state
 
 
+_S2D_ORDER_CACHE = {}
+
+
+def _s2d_order(key, vals, device):
+  """Capture-safe cached index tensor (H2D copies are illegal inside
+  hipGraph capture; warmup populates the cache before capture)."""
+  ck = (key, device)
+  t = _S2D_ORDER_CACHE.get(ck)
+  if t is None:
+    t = torch.tensor(vals, device=device)
+    _S2D_ORDER_CACHE[ck] = t
+  return t
+
+
 class _Conv3x3S2Nhwc(torch.autograd.Function):
   """3x3 stride-2 same-pad conv on NHWC input via SPACE-TO-DEPTH +
   4 flat GEMMs.
@@ -271,7 +285,7 @@ class _Conv3x3S2Nhwc(torch.autograd.Function):
     xr = x.reshape(B, Ho, 2, Wo, 2, C).permute(0, 1, 3, 2, 4, 5)
     xr = xr.reshape(B, Ho, Wo, 4, C)
     # (a,b) flat index a*2+b; block order [(0,1),(1,1),(1,0),(0,0)].
-    order = torch.tensor([1, 3, 2, 0], device=x.device)
+    order = _s2d_order('fwd', [1, 3, 2, 0], x.device)
     xr = xr.index_select(3, order).reshape(B, Ho, Wo, 4 * C)
     X = x.new_zeros(B, Ho + 1, Wo + 1, 4 * C)
     X[:, 1:, 1:, :] = xr
@@ -341,7 +355,7 @@ class _Conv3x3S2Nhwc(torch.autograd.Function):
     # Inverse s2d.
     dXr = dX.reshape(B, Ho + 1, Wp, 4 * C)[:, 1:, 1:, :]
     dXr = dXr.reshape(B, Ho, Wo, 4, C)
-    inv_order = torch.tensor([3, 0, 2, 1], device=x.device)
+    inv_order = _s2d_order('inv', [3, 0, 2, 1], x.device)
     dXr = dXr.index_select(3, inv_order).reshape(B, Ho, Wo, 2, 2, C)
     dx = dXr.permute(0, 1, 3, 2, 4, 5).reshape(B, 2 * Ho, 2 * Wo, C)
     if pad_h or pad_w:
