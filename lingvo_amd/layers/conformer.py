@@ -231,20 +231,6 @@ class ConformerLayer(BaseLayer):
     return self.final_ln.FProp(theta.final_ln, x), state
 
 
-_S2D_ORDER_CACHE = {}
-
-
-def _s2d_order(key, vals, device):
-  """Capture-safe cached index tensor (H2D copies are illegal inside
-  hipGraph capture; warmup populates the cache before capture)."""
-  ck = (key, device)
-  t = _S2D_ORDER_CACHE.get(ck)
-  if t is None:
-    t = torch.tensor(vals, device=device)
-    _S2D_ORDER_CACHE[ck] = t
-  return t
-
-
 class _Conv3x3S2Nhwc(torch.autograd.Function):
   """3x3 stride-2 same-pad conv on NHWC input via SPACE-TO-DEPTH +
   4 flat GEMMs.
@@ -277,18 +263,21 @@ class _Conv3x3S2Nhwc(torch.autograd.Function):
       ((0, 0), [(1, 2, 0), (2, 2, 1), (2, 1, 2), (1, 1, 3)]),
   ]
 
+  # s2d block order: s0=(0,1), s1=(1,1), s2=(1,0), s3=(0,0) in (a,b)
+  # cell coordinates — the order that makes every cell's channel range
+  # contiguous.
+  BLOCKS = [(0, 1), (1, 1), (1, 0), (0, 0)]
+
   @staticmethod
   def _s2d(x):
-    """[B,H,W,C] (H,W even) -> padded flat [B*(Ho+1)*(Wo+1), 4C]."""
+    """[B,H,W,C] (H,W even) -> padded flat [B*(Ho+1)*(Wo+1), 4C].
+    One strided copy per block straight into the padded buffer (no
+    intermediate permute/index_select materialization)."""
     B, H, W, C = x.shape
     Ho, Wo = H // 2, W // 2
-    xr = x.reshape(B, Ho, 2, Wo, 2, C).permute(0, 1, 3, 2, 4, 5)
-    xr = xr.reshape(B, Ho, Wo, 4, C)
-    # (a,b) flat index a*2+b; block order [(0,1),(1,1),(1,0),(0,0)].
-    order = _s2d_order('fwd', [1, 3, 2, 0], x.device)
-    xr = xr.index_select(3, order).reshape(B, Ho, Wo, 4 * C)
     X = x.new_zeros(B, Ho + 1, Wo + 1, 4 * C)
-    X[:, 1:, 1:, :] = xr
+    for i, (a, b) in enumerate(_Conv3x3S2Nhwc.BLOCKS):
+      X[:, 1:, 1:, i * C:(i + 1) * C] = x[:, a::2, b::2, :]
     return X.reshape(B * (Ho + 1) * (Wo + 1), 4 * C), Ho, Wo
 
   @staticmethod
@@ -352,12 +341,12 @@ class _Conv3x3S2Nhwc(torch.autograd.Function):
       wc_t = _Conv3x3S2Nhwc._cell_weight(w.to(dO.dtype), taps,
                                          transpose=True)
       dX[base + off:Rp + off, k0:k1].addmm_(dO[base:], wc_t)
-    # Inverse s2d.
+    # Inverse s2d: strided scatter of each block's channel slice.
     dXr = dX.reshape(B, Ho + 1, Wp, 4 * C)[:, 1:, 1:, :]
-    dXr = dXr.reshape(B, Ho, Wo, 4, C)
-    inv_order = _s2d_order('inv', [3, 0, 2, 1], x.device)
-    dXr = dXr.index_select(3, inv_order).reshape(B, Ho, Wo, 2, 2, C)
-    dx = dXr.permute(0, 1, 3, 2, 4, 5).reshape(B, 2 * Ho, 2 * Wo, C)
+    dx_full = dout.new_empty(B, 2 * Ho, 2 * Wo, C)
+    for i, (a, b) in enumerate(_Conv3x3S2Nhwc.BLOCKS):
+      dx_full[:, a::2, b::2, :] = dXr[..., i * C:(i + 1) * C]
+    dx = dx_full
     if pad_h or pad_w:
       dx = dx[:, :2 * Ho - pad_h, :2 * Wo - pad_w, :].contiguous()
     dbias = dO[base:].float().sum(0).to(ctx.bias_dtype)
