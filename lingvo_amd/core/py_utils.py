@@ -202,19 +202,30 @@ def MakeStepGenerator(device: Union[str, torch.device],
 
 
 def DeterministicDropout(x: torch.Tensor, keep_prob: float,
-                         op_seed: Optional[int] = None) -> torch.Tensor:
+                         op_seed: Optional[int] = None,
+                         act: str = 'NONE') -> torch.Tensor:
   """Dropout reproducible under StepSeedScope (reference py_utils.py:3978).
 
   GPU: one fused HIP kernel (mask recomputed from the seed in backward);
-  CPU: torch.Generator reference path.
+  CPU: torch.Generator reference path. act in {'NONE','SWISH','RELU'}
+  applies the activation INSIDE the same kernel (dropout(act(x))) —
+  the FFN hidden tensor is read/written once instead of twice.
   """
   if keep_prob >= 1.0:
+    if act == 'SWISH':
+      return torch.nn.functional.silu(x)
+    if act == 'RELU':
+      return torch.relu(x)
     return x
   if x.is_cuda and x.numel() % 8 == 0:
     from lingvo_amd.ops import dropout as dropout_ops
     s1, _ = GenerateStepSeedPair(op_seed)
     # step-dependence comes from the device step-seed buffer (graph-safe)
-    return dropout_ops.dropout(x, keep_prob, s1)
+    return dropout_ops.dropout(x, keep_prob, s1, act=act)
+  if act == 'SWISH':
+    x = torch.nn.functional.silu(x)
+  elif act == 'RELU':
+    x = torch.relu(x)
   g = MakeStepGenerator(x.device, op_seed)
   mask = (torch.rand(x.shape, generator=g, device=x.device,
                      dtype=torch.float32) < keep_prob)

@@ -115,9 +115,16 @@ class TransformerFeedForwardLayer(BaseLayer):
     p = self.p
     x = self.layer_norm.FProp(theta.layer_norm, inputs)
     h = py_utils.MatmulBias(x, theta.w1, theta.b1)
-    h = activations.GetFn(p.activation)(h)
-    if p.relu_dropout_prob and not self.do_eval:
-      h = py_utils.DeterministicDropout(h, 1.0 - p.relu_dropout_prob)
+    if (p.relu_dropout_prob and not self.do_eval and
+        p.activation in ('SWISH', 'RELU')):
+      # Activation fused into the dropout kernel: one pass over the
+      # [*, hidden] tensor instead of two, each direction.
+      h = py_utils.DeterministicDropout(h, 1.0 - p.relu_dropout_prob,
+                                        act=p.activation)
+    else:
+      h = activations.GetFn(p.activation)(h)
+      if p.relu_dropout_prob and not self.do_eval:
+        h = py_utils.DeterministicDropout(h, 1.0 - p.relu_dropout_prob)
     out = py_utils.MatmulBias(h, theta.w2, theta.b2)
     if p.residual_weight != 1.0:
       out = out * p.residual_weight

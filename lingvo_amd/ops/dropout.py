@@ -38,38 +38,56 @@ def SetStepSeed(step: int, global_seed: int = 0) -> None:
     _StepSeedBuf(dev).fill_(val)
 
 
+ACT_CODES = {'NONE': 0, 'SWISH': 1, 'SILU': 1, 'RELU': 2}
+
+
 class _DropoutFn(torch.autograd.Function):
 
   @staticmethod
-  def forward(ctx, x, residual, seed, keep):
+  def forward(ctx, x, residual, seed, keep, act):
     ext = _loader.get_ext(required=True)
     buf = _StepSeedBuf(x.device)
-    y = ext.dropout_fwd(x, residual, seed, buf, keep)
+    y = ext.dropout_fwd(x, residual, seed, buf, keep, act)
     ctx.seed = seed
     ctx.keep = keep
+    ctx.act = act
     ctx.has_res = residual is not None
     ctx.dev = x.device
+    if act:
+      ctx.save_for_backward(x)
     return y
 
   @staticmethod
   def backward(ctx, dy):
     ext = _loader.get_ext(required=True)
     dy = dy.contiguous()
-    dx = ext.dropout_bwd(dy, ctx.seed, _StepSeedBuf(ctx.dev), ctx.keep)
+    x = ctx.saved_tensors[0] if ctx.act else None
+    dx = ext.dropout_bwd(dy, x, ctx.seed, _StepSeedBuf(ctx.dev), ctx.keep,
+                         ctx.act)
     dres = dy if ctx.has_res else None
-    return dx, dres, None, None
+    return dx, dres, None, None, None
 
 
 def dropout(x: torch.Tensor, keep_prob: float, seed: int,
-            residual: Optional[torch.Tensor] = None) -> torch.Tensor:
-  """y = dropout(x) (+ residual). GPU bf16 fast path; generic otherwise."""
+            residual: Optional[torch.Tensor] = None,
+            act: str = 'NONE') -> torch.Tensor:
+  """y = dropout(act(x)) (+ residual). One HIP kernel each way on GPU
+  bf16 (the FFN activation pass fuses in); generic composed path
+  otherwise. act in {'NONE','SWISH','RELU'}; residual excludes act."""
+  act_code = ACT_CODES[act.upper()]
+  assert not (act_code and residual is not None)
   if x.is_cuda and x.numel() % 8 == 0:
     orig = x.dtype
     y = _DropoutFn.apply(
         x.to(torch.bfloat16).contiguous(),
         None if residual is None else
-        residual.to(torch.bfloat16).contiguous(), seed, keep_prob)
+        residual.to(torch.bfloat16).contiguous(), seed, keep_prob,
+        act_code)
     return y.to(orig) if orig != torch.bfloat16 else y
+  if act_code == 1:
+    x = torch.nn.functional.silu(x)
+  elif act_code == 2:
+    x = torch.relu(x)
   g = torch.Generator(device=x.device)
   g.manual_seed(seed & 0x7FFFFFFFFFFFFFFF)
   mask = (torch.rand(x.shape, generator=g, device=x.device,

@@ -23,7 +23,26 @@ __device__ __forceinline__ unsigned int hash_u32(unsigned long long seed,
   return (unsigned int)h;
 }
 
-template <bool RESIDUAL>
+// ACT: 0 = identity, 1 = SiLU/Swish, 2 = ReLU — applied BEFORE the
+// dropout mask, fusing the FFN activation pass into this kernel.
+template <int ACT>
+__device__ __forceinline__ float apply_act(float v) {
+  if (ACT == 1) return v / (1.f + __expf(-v)) ;
+  if (ACT == 2) return v > 0.f ? v : 0.f;
+  return v;
+}
+
+template <int ACT>
+__device__ __forceinline__ float act_grad(float v) {
+  if (ACT == 1) {
+    const float s = 1.f / (1.f + __expf(-v));
+    return s * (1.f + v * (1.f - s));
+  }
+  if (ACT == 2) return v > 0.f ? 1.f : 0.f;
+  return 1.f;
+}
+
+template <bool RESIDUAL, int ACT>
 __global__ void dropout_fwd_kernel(const unsigned short* __restrict__ x,
                                    const unsigned short* __restrict__ res,
                                    unsigned short* __restrict__ y,
@@ -39,7 +58,8 @@ __global__ void dropout_fwd_kernel(const unsigned short* __restrict__ x,
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
       bool kept = hash_u32(seed, i * 8 + e) < thresh;
-      float f = kept ? bf16_bits_to_float(v[e]) * inv_keep : 0.f;
+      float f = kept ? apply_act<ACT>(bf16_bits_to_float(v[e])) * inv_keep
+                     : 0.f;
       o[e] = float_to_bf16_bits(f);
     }
     if (RESIDUAL) {
@@ -53,7 +73,9 @@ __global__ void dropout_fwd_kernel(const unsigned short* __restrict__ x,
   }
 }
 
+template <int ACT>
 __global__ void dropout_bwd_kernel(const unsigned short* __restrict__ dy,
+                                   const unsigned short* __restrict__ x,
                                    unsigned short* __restrict__ dx,
                                    long nvec, unsigned long long seed,
                                    const long* __restrict__ step_seed,
@@ -63,11 +85,14 @@ __global__ void dropout_bwd_kernel(const unsigned short* __restrict__ dy,
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
        i += (long)gridDim.x * blockDim.x) {
     ushortx8 v = *reinterpret_cast<const ushortx8*>(dy + i * 8);
+    ushortx8 xv;
+    if (ACT != 0) xv = *reinterpret_cast<const ushortx8*>(x + i * 8);
     ushortx8 o;
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
       bool kept = hash_u32(seed, i * 8 + e) < thresh;
       float f = kept ? bf16_bits_to_float(v[e]) * inv_keep : 0.f;
+      if (ACT != 0) f *= act_grad<ACT>(bf16_bits_to_float(xv[e]));
       o[e] = float_to_bf16_bits(f);
     }
     *reinterpret_cast<ushortx8*>(dx + i * 8) = o;
@@ -78,46 +103,62 @@ __global__ void dropout_bwd_kernel(const unsigned short* __restrict__ dy,
 
 torch::Tensor dropout_fwd(torch::Tensor x, c10::optional<torch::Tensor> res,
                           int64_t seed, c10::optional<torch::Tensor> step_seed,
-                          double keep) {
+                          double keep, int64_t act) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() &&
               x.scalar_type() == torch::kBFloat16 && x.numel() % 8 == 0);
+  TORCH_CHECK(!(res.has_value() && act != 0),
+              "residual+activation fusion unsupported");
   auto y = torch::empty_like(x);
   long nvec = x.numel() / 8;
   auto stream = at::cuda::getCurrentCUDAStream();
   const long* ssp = step_seed.has_value() ?
       step_seed->data_ptr<long>() : nullptr;
+#define LAUNCH_FWD(RES, ACT, RESPTR)                                         \
+  hipLaunchKernelGGL((dropout_fwd_kernel<RES, ACT>),                         \
+                     dim3(memory_bound_grid(nvec, 256)), dim3(256), 0,       \
+                     stream, (const unsigned short*)x.data_ptr(), RESPTR,    \
+                     (unsigned short*)y.data_ptr(), nvec,                    \
+                     (unsigned long long)seed, ssp, (float)keep,             \
+                     (float)(1.0 / keep))
   if (res.has_value()) {
-    hipLaunchKernelGGL((dropout_fwd_kernel<true>),
-                       dim3(memory_bound_grid(nvec, 256)), dim3(256), 0,
-                       stream, (const unsigned short*)x.data_ptr(),
-                       (const unsigned short*)res->data_ptr(),
-                       (unsigned short*)y.data_ptr(), nvec,
-                       (unsigned long long)seed, ssp, (float)keep,
-                       (float)(1.0 / keep));
+    LAUNCH_FWD(true, 0, (const unsigned short*)res->data_ptr());
+  } else if (act == 1) {
+    LAUNCH_FWD(false, 1, nullptr);
+  } else if (act == 2) {
+    LAUNCH_FWD(false, 2, nullptr);
   } else {
-    hipLaunchKernelGGL((dropout_fwd_kernel<false>),
-                       dim3(memory_bound_grid(nvec, 256)), dim3(256), 0,
-                       stream, (const unsigned short*)x.data_ptr(), nullptr,
-                       (unsigned short*)y.data_ptr(), nvec,
-                       (unsigned long long)seed, ssp, (float)keep,
-                       (float)(1.0 / keep));
+    LAUNCH_FWD(false, 0, nullptr);
   }
+#undef LAUNCH_FWD
   return y;
 }
 
-torch::Tensor dropout_bwd(torch::Tensor dy, int64_t seed,
-                          c10::optional<torch::Tensor> step_seed,
-                          double keep) {
+torch::Tensor dropout_bwd(torch::Tensor dy, c10::optional<torch::Tensor> x,
+                          int64_t seed, c10::optional<torch::Tensor> step_seed,
+                          double keep, int64_t act) {
+  TORCH_CHECK(act == 0 || x.has_value(),
+              "activation-fused dropout bwd needs the pre-activation input");
   auto dx = torch::empty_like(dy);
   long nvec = dy.numel() / 8;
   auto stream = at::cuda::getCurrentCUDAStream();
   const long* ssp = step_seed.has_value() ?
       step_seed->data_ptr<long>() : nullptr;
-  hipLaunchKernelGGL(dropout_bwd_kernel,
-                     dim3(memory_bound_grid(nvec, 256)), dim3(256), 0,
-                     stream, (const unsigned short*)dy.data_ptr(),
-                     (unsigned short*)dx.data_ptr(), nvec,
-                     (unsigned long long)seed, ssp, (float)keep,
-                     (float)(1.0 / keep));
+  const unsigned short* xp = x.has_value() ?
+      (const unsigned short*)x->data_ptr() : nullptr;
+#define LAUNCH_BWD(ACT)                                                      \
+  hipLaunchKernelGGL((dropout_bwd_kernel<ACT>),                              \
+                     dim3(memory_bound_grid(nvec, 256)), dim3(256), 0,       \
+                     stream, (const unsigned short*)dy.data_ptr(), xp,       \
+                     (unsigned short*)dx.data_ptr(), nvec,                   \
+                     (unsigned long long)seed, ssp, (float)keep,             \
+                     (float)(1.0 / keep))
+  if (act == 1) {
+    LAUNCH_BWD(1);
+  } else if (act == 2) {
+    LAUNCH_BWD(2);
+  } else {
+    LAUNCH_BWD(0);
+  }
+#undef LAUNCH_BWD
   return dx;
 }

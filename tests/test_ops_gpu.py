@@ -119,6 +119,30 @@ def test_fused_dropout_fwd_bwd():
 
 
 @gpu
+def test_act_fused_dropout_matches_composed():
+  """dropout(act(x)) fused == silu/relu then plain dropout, same seed."""
+  from lingvo_amd.ops import dropout as dropout_ops
+  torch.manual_seed(1)
+  keep = 0.85
+  for act, fn, grad_fn in [
+      ('SWISH', torch.nn.functional.silu, None),
+      ('RELU', torch.relu, None)]:
+    x = torch.randn(8, 64, 256, device='cuda',
+                    dtype=torch.bfloat16).requires_grad_(True)
+    y = dropout_ops.dropout(x, keep, seed=7, act=act)
+    a = fn(x.detach())
+    y_ref = dropout_ops.dropout(a, keep, seed=7)
+    assert (y.detach().float() - y_ref.float()).abs().max() < 0.02, act
+    # backward vs autograd through the composed fp32 reference.
+    g = torch.randn_like(y)
+    y.backward(g)
+    x32 = x.detach().float().requires_grad_(True)
+    mask = (y_ref != 0) | (a == 0)
+    (fn(x32) * mask / keep).backward(g.float())
+    assert (x.grad.float() - x32.grad).abs().max() < 0.05, act
+
+
+@gpu
 def test_group_norm_matches_ref():
   from lingvo_amd.ops import group_norm as gn_ops
   from lingvo_amd.core import py_utils as pu
