@@ -109,8 +109,12 @@ __global__ __launch_bounds__(256) void dwconv_bwd_dw(
   // 8 channels x 4 tap-quarters of 8 taps x 4 row-quarters of 16 rows.
   // Every x/dy access is a ushortx8 load; accumulators live in
   // registers across all chunks and flush once.
-  __shared__ unsigned short x_s[DW_CHUNK + MAXK - 1][DW_DBLK];
-  __shared__ unsigned short dy_s[DW_CHUNK][DW_DBLK];
+  // Row stride padded by 8 ushorts (4 banks): with an unpadded 128-
+  // ushort (64-bank) stride, the 4 tap-quarter groups of a wave read
+  // x_s rows 8 apart that land on IDENTICAL banks -> 4-way conflict on
+  // every inner-loop LDS read.
+  __shared__ unsigned short x_s[DW_CHUNK + MAXK - 1][DW_DBLK + 8];
+  __shared__ unsigned short dy_s[DW_CHUNK][DW_DBLK + 8];
   const int tid = threadIdx.x;
   const int dg = tid & 15;          // d-group (8 channels)
   const int tq = (tid >> 4) & 3;    // tap quarter (taps 8tq..8tq+7)
