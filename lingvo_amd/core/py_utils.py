@@ -234,14 +234,29 @@ def DeterministicDropout(x: torch.Tensor, keep_prob: float,
 
 def DeterministicDropoutAdd(x: torch.Tensor, keep_prob: float,
                             residual: torch.Tensor,
-                            op_seed: Optional[int] = None) -> torch.Tensor:
-  """residual + dropout(x): fused into one HIP kernel on GPU."""
+                            op_seed: Optional[int] = None,
+                            scale: float = 1.0,
+                            paddings: Optional[torch.Tensor] = None
+                            ) -> torch.Tensor:
+  """residual + dropout(x * scale * (1 - paddings)): one HIP kernel on
+  GPU — the residual weight and ApplyPadding mask fuse into the same
+  pass instead of separate elementwise kernels."""
   if keep_prob >= 1.0:
+    if scale != 1.0:
+      x = x * scale
+    if paddings is not None:
+      x = ApplyPadding(paddings, x)
     return residual + x
-  if x.is_cuda and x.numel() % 8 == 0:
+  if x.is_cuda and x.numel() % 8 == 0 and \
+      (paddings is None or x.shape[-1] % 8 == 0):
     from lingvo_amd.ops import dropout as dropout_ops
     s1, _ = GenerateStepSeedPair(op_seed)
-    return dropout_ops.dropout(x, keep_prob, s1, residual=residual)
+    return dropout_ops.dropout(x, keep_prob, s1, residual=residual,
+                               scale=scale, paddings=paddings)
+  if scale != 1.0:
+    x = x * scale
+  if paddings is not None:
+    x = ApplyPadding(paddings, x)
   return residual + DeterministicDropout(x, keep_prob, op_seed)
 
 

@@ -84,11 +84,12 @@ class LConvLayer(BaseLayer):
       x = self.norm.FProp(theta.norm, x, paddings)
     x = F.silu(x)
     x = py_utils.MatmulBias(x, theta.pw2_w, theta.pw2_b)
+    if p.dropout_prob and not self.do_eval:
+      # padding mask folds into the dropout kernel's elementwise pass.
+      return py_utils.DeterministicDropoutAdd(x, 1.0 - p.dropout_prob,
+                                              inputs, paddings=paddings)
     if paddings is not None:
       x = py_utils.ApplyPadding(paddings, x)
-    if p.dropout_prob and not self.do_eval:
-      return py_utils.DeterministicDropoutAdd(x, 1.0 - p.dropout_prob,
-                                              inputs)
     return inputs + x
 
   def InitStreamState(self, batch: int, device, dtype) -> NestedMap:

@@ -126,13 +126,15 @@ class TransformerFeedForwardLayer(BaseLayer):
       if p.relu_dropout_prob and not self.do_eval:
         h = py_utils.DeterministicDropout(h, 1.0 - p.relu_dropout_prob)
     out = py_utils.MatmulBias(h, theta.w2, theta.b2)
+    if p.residual_dropout_prob and not self.do_eval:
+      # residual weight + padding mask fold into the dropout kernel.
+      return py_utils.DeterministicDropoutAdd(
+          out, 1.0 - p.residual_dropout_prob, inputs,
+          scale=p.residual_weight, paddings=paddings)
     if p.residual_weight != 1.0:
       out = out * p.residual_weight
     if paddings is not None:
       out = py_utils.ApplyPadding(paddings, out)
-    if p.residual_dropout_prob and not self.do_eval:
-      return py_utils.DeterministicDropoutAdd(
-          out, 1.0 - p.residual_dropout_prob, inputs)
     return inputs + out
 
 

@@ -44,50 +44,74 @@ ACT_CODES = {'NONE': 0, 'SWISH': 1, 'SILU': 1, 'RELU': 2}
 class _DropoutFn(torch.autograd.Function):
 
   @staticmethod
-  def forward(ctx, x, residual, seed, keep, act):
+  def forward(ctx, x, residual, seed, keep, act, scale, pad, d):
     ext = _loader.get_ext(required=True)
     buf = _StepSeedBuf(x.device)
-    y = ext.dropout_fwd(x, residual, seed, buf, keep, act)
+    y = ext.dropout_fwd(x, residual, seed, buf, keep, act, scale, pad, d)
     ctx.seed = seed
     ctx.keep = keep
     ctx.act = act
+    ctx.scale = scale
+    ctx.d = d
     ctx.has_res = residual is not None
     ctx.dev = x.device
+    saved = []
     if act:
-      ctx.save_for_backward(x)
+      saved.append(x)
+    ctx.has_pad = pad is not None
+    if pad is not None:
+      saved.append(pad)
+    ctx.save_for_backward(*saved)
     return y
 
   @staticmethod
   def backward(ctx, dy):
     ext = _loader.get_ext(required=True)
     dy = dy.contiguous()
-    x = ctx.saved_tensors[0] if ctx.act else None
+    saved = list(ctx.saved_tensors)
+    x = saved.pop(0) if ctx.act else None
+    pad = saved.pop(0) if ctx.has_pad else None
     dx = ext.dropout_bwd(dy, x, ctx.seed, _StepSeedBuf(ctx.dev), ctx.keep,
-                         ctx.act)
+                         ctx.act, ctx.scale, pad, ctx.d)
     dres = dy if ctx.has_res else None
-    return dx, dres, None, None, None
+    return dx, dres, None, None, None, None, None, None
 
 
 def dropout(x: torch.Tensor, keep_prob: float, seed: int,
             residual: Optional[torch.Tensor] = None,
-            act: str = 'NONE') -> torch.Tensor:
-  """y = dropout(act(x)) (+ residual). One HIP kernel each way on GPU
-  bf16 (the FFN activation pass fuses in); generic composed path
-  otherwise. act in {'NONE','SWISH','RELU'}; residual excludes act."""
+            act: str = 'NONE', scale: float = 1.0,
+            paddings: Optional[torch.Tensor] = None) -> torch.Tensor:
+  """y = dropout(act(x) * scale * (1 - pad)) (+ residual). One HIP
+  kernel each way on GPU bf16 — the FFN activation, residual weight
+  and ApplyPadding mask all fuse into the same pass. act in
+  {'NONE','SWISH','RELU'}; residual excludes act. paddings is [B, T]
+  (or [rows]) with 1.0 == padded, row = flat_index // D where
+  D = x.shape[-1] (D % 8 == 0)."""
   act_code = ACT_CODES[act.upper()]
   assert not (act_code and residual is not None)
-  if x.is_cuda and x.numel() % 8 == 0:
+  d = x.shape[-1]
+  if (x.is_cuda and x.numel() % 8 == 0 and
+      (paddings is None or d % 8 == 0)):
     orig = x.dtype
+    pad = None
+    if paddings is not None:
+      pad = paddings.to(torch.bfloat16).contiguous()
+      assert pad.numel() * d == x.numel()
     y = _DropoutFn.apply(
         x.to(torch.bfloat16).contiguous(),
         None if residual is None else
         residual.to(torch.bfloat16).contiguous(), seed, keep_prob,
-        act_code)
+        act_code, scale, pad, d)
     return y.to(orig) if orig != torch.bfloat16 else y
   if act_code == 1:
     x = torch.nn.functional.silu(x)
   elif act_code == 2:
     x = torch.relu(x)
+  if scale != 1.0:
+    x = x * scale
+  if paddings is not None:
+    x = x * (1.0 - paddings.to(x.dtype)).reshape(
+        *paddings.shape, *([1] * (x.dim() - paddings.dim())))
   g = torch.Generator(device=x.device)
   g.manual_seed(seed & 0x7FFFFFFFFFFFFFFF)
   mask = (torch.rand(x.shape, generator=g, device=x.device,

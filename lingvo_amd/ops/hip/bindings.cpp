@@ -55,11 +55,13 @@ torch::Tensor xent_bwd(torch::Tensor logits, torch::Tensor labels,
 torch::Tensor dropout_fwd(torch::Tensor x, c10::optional<torch::Tensor> res,
                           int64_t seed,
                           c10::optional<torch::Tensor> step_seed,
-                          double keep, int64_t act);
+                          double keep, int64_t act, double scale,
+                          c10::optional<torch::Tensor> pad, int64_t d);
 torch::Tensor dropout_bwd(torch::Tensor dy, c10::optional<torch::Tensor> x,
                           int64_t seed,
                           c10::optional<torch::Tensor> step_seed,
-                          double keep, int64_t act);
+                          double keep, int64_t act, double scale,
+                          c10::optional<torch::Tensor> pad, int64_t d);
 
 // group_norm.hip
 std::vector<torch::Tensor> group_norm_fwd(torch::Tensor x,
