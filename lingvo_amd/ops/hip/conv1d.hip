@@ -102,8 +102,10 @@ constexpr int DW_DBLK = 128;
 
 __global__ __launch_bounds__(256) void dwconv_bwd_dw(
     const unsigned short* __restrict__ dy,
-    const unsigned short* __restrict__ x, float* __restrict__ dw_acc,
-    float* __restrict__ db_acc, int B, int T, int D, int K, int pad,
+    const unsigned short* __restrict__ x,
+    float* __restrict__ dw_part,  // [ngroups, K, D] plain stores
+    float* __restrict__ db_part,  // [ngroups, D]
+    int B, int T, int D, int K, int pad,
     int ngroups) {
   // Thread decomposition (fully vectorized LDS reads): 16 d-groups of
   // 8 channels x 4 tap-quarters of 8 taps x 4 row-quarters of 16 rows.
@@ -191,24 +193,61 @@ __global__ __launch_bounds__(256) void dwconv_bwd_dw(
     __syncthreads();
   }
 
-  if (d0 + 7 < D || d0 < D) {
+  // Cross-wave (rq) reduction in LDS, then ONE plain store per element
+  // into this block's [K, DW_DBLK] partial slice. The previous global
+  // atomicAdd flush serialized ~4k adds per dw address across the grid
+  // and dominated the kernel's runtime.
+  __shared__ float dw_lds[MAXK - 1][DW_DBLK];
+  __shared__ float db_lds[DW_DBLK];
+  for (int i = tid; i < K * DW_DBLK; i += 256) {
+    dw_lds[i / DW_DBLK][i % DW_DBLK] = 0.f;
+  }
+  for (int i = tid; i < DW_DBLK; i += 256) db_lds[i] = 0.f;
+  __syncthreads();
 #pragma unroll
-    for (int jj = 0; jj < 8; ++jj) {
-      const int j = tq * 8 + jj;
-      if (j >= K) break;
+  for (int jj = 0; jj < 8; ++jj) {
+    const int j = tq * 8 + jj;
+    if (j >= K) break;
 #pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        if (d0 + e < D && dw8[jj][e] != 0.f) {
-          atomicAdd(dw_acc + (long)j * D + d0 + e, dw8[jj][e]);
-        }
-      }
+    for (int e = 0; e < 8; ++e) {
+      if (dw8[jj][e] != 0.f) atomicAdd(&dw_lds[j][dg * 8 + e], dw8[jj][e]);
     }
-    if (tq == 0) {
+  }
+  if (tq == 0) {
 #pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        if (d0 + e < D && db8[e] != 0.f) atomicAdd(db_acc + d0 + e,
-                                                   db8[e]);
-      }
+    for (int e = 0; e < 8; ++e) {
+      if (db8[e] != 0.f) atomicAdd(&db_lds[dg * 8 + e], db8[e]);
+    }
+  }
+  __syncthreads();
+  const long gbase = (long)blockIdx.y * K * D;
+  for (int i = tid; i < K * DW_DBLK; i += 256) {
+    const int j = i / DW_DBLK;
+    const int d = blockIdx.x * DW_DBLK + i % DW_DBLK;
+    if (d < D) dw_part[gbase + (long)j * D + d] = dw_lds[j][i % DW_DBLK];
+  }
+  for (int i = tid; i < DW_DBLK; i += 256) {
+    const int d = blockIdx.x * DW_DBLK + i;
+    if (d < D) db_part[(long)blockIdx.y * D + d] = db_lds[i];
+  }
+}
+
+// Sums the per-group partials: dw[j][d] = sum_g dw_part[g][j][d].
+__global__ void dwconv_bwd_dw_reduce(const float* __restrict__ dw_part,
+                                     const float* __restrict__ db_part,
+                                     float* __restrict__ dw,
+                                     float* __restrict__ db, long kd, int D,
+                                     int ngroups) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < kd + D;
+       i += (long)gridDim.x * blockDim.x) {
+    float sum = 0.f;
+    if (i < kd) {
+      for (int g = 0; g < ngroups; ++g) sum += dw_part[(long)g * kd + i];
+      dw[i] = sum;
+    } else {
+      const long d = i - kd;
+      for (int g = 0; g < ngroups; ++g) sum += db_part[(long)g * D + d];
+      db[d] = sum;
     }
   }
 }
@@ -254,11 +293,19 @@ std::vector<torch::Tensor> dwconv1d_bwd(torch::Tensor dy, torch::Tensor x,
   int dblks = (D + DW_DBLK - 1) / DW_DBLK;
   int ngroups = (int)std::min<long>(std::max<long>(1, 1024 / dblks),
                                     nchunks);
+  auto dw_part = torch::empty({(long)ngroups * K * D}, opts);
+  auto db_part = torch::empty({(long)ngroups * D}, opts);
   dim3 grid_w(dblks, ngroups);
   hipLaunchKernelGGL(dwconv_bwd_dw, grid_w, dim3(256), 0, stream,
                      (const unsigned short*)dy.data_ptr(),
                      (const unsigned short*)x.data_ptr(),
-                     dw.data_ptr<float>(), db.data_ptr<float>(), B, T, D, K,
-                     (int)pad, ngroups);
+                     dw_part.data_ptr<float>(), db_part.data_ptr<float>(),
+                     B, T, D, K, (int)pad, ngroups);
+  const long kd = (long)K * D;
+  hipLaunchKernelGGL(dwconv_bwd_dw_reduce,
+                     dim3((int)std::min<long>((kd + D + 255) / 256, 1024)),
+                     dim3(256), 0, stream, dw_part.data_ptr<float>(),
+                     db_part.data_ptr<float>(), dw.data_ptr<float>(),
+                     db.data_ptr<float>(), kd, D, ngroups);
   return {dx, dw, db};
 }
