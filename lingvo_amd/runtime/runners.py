@@ -250,9 +250,13 @@ class Trainer(BaseRunner):
     model.train()
     finalize = self._grad_sync.Finalize if self._grad_sync else None
     while not self._ShouldStop(task):
+      ig = task.input_generator
+      if ig is not None and hasattr(ig, 'SetGlobalStep'):
+        # Curriculum-style inputs switch stages on the global step.
+        ig.SetGlobalStep(int(task.global_step))
       batch = task.GetInputBatch()
-      if task.input_generator is not None:
-        batch = task.input_generator.ToDevice(batch, self._device)
+      if ig is not None:
+        batch = ig.ToDevice(batch, self._device)
       metrics = task.TrainStep(batch, grad_sync_finalize=finalize)
       step = task.global_step
       examples = py_utils.ToScalar(

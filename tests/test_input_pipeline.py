@@ -354,3 +354,42 @@ def test_file_input_generator_within_batch_mixing(tmp_path):
   frac_a = sum(1 for v in vals if v == 1.0) / len(vals)
   assert 0.6 < frac_a < 0.9  # ~0.75 example-level mix
   gen.Reset()
+
+
+def test_curriculum_datasource_stage_switching():
+  import torch
+  from lingvo_amd.core import datasource as ds
+  from lingvo_amd.core.base_input_generator import BaseInputGenerator
+  from lingvo_amd.core.nested_map import NestedMap
+
+  class Const(BaseInputGenerator):
+
+    @classmethod
+    def Params(cls):
+      p = super().Params()
+      p.Define('value', 0, '')
+      return p
+
+    def _InputBatch(self):
+      return NestedMap(x=torch.full((2,), float(self.p.value)))
+
+  def src(v):
+    return ds.SimpleDataSource.Params().Set(
+        input_generator=Const.Params().Set(value=v, name=f'c{v}'))
+
+  import pytest
+  with pytest.raises(ValueError):
+    ds.CurriculumDataSource.Params().Set(
+        name='bad', sub=[src(1)], boundaries=[5]).Instantiate()
+
+  cur = ds.CurriculumDataSource.Params().Set(
+      name='cur', sub=[src(1), src(2), src(3)],
+      boundaries=[10, 20]).Instantiate()
+  assert cur.current_stage == 0
+  assert int(cur.GetNext().x[0]) == 1
+  cur.SetGlobalStep(10)
+  assert cur.current_stage == 1
+  assert int(cur.GetNext().x[0]) == 2
+  cur.SetGlobalStep(25)
+  assert cur.current_stage == 2
+  assert int(cur.GetNext().x[0]) == 3
