@@ -240,6 +240,203 @@ class FnLayer(BaseLayer):
     return self.p.fn(*args)
 
 
+class FirstNLayer(BaseLayer):
+  """Returns the first n positional args (reference builder_layers.py:29)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('n', 1, 'Number of args to return.')
+    return p
+
+  def FProp(self, theta: NestedMap, *args):
+    n = self.p.n
+    assert len(args) >= n
+    return args[0] if n == 1 else tuple(args[:n])
+
+
+class ArgIndexLayer(BaseLayer):
+  """Selects args by index (reference builder_layers.py:60)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('idx', [], 'Indices of args to return.')
+    return p
+
+  def FProp(self, theta: NestedMap, *args):
+    out = tuple(args[i] for i in self.p.idx)
+    return out[0] if len(out) == 1 else out
+
+
+class CreateNestedMapLayer(BaseLayer):
+  """Packs positional args into a NestedMap by key
+  (reference builder_layers.py:100)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('keys', [], 'Dotted keys, one per positional arg.')
+    return p
+
+  def FProp(self, theta: NestedMap, *args):
+    out = NestedMap()
+    for key, value in zip(self.p.keys, args):
+      out.Set(key, value)
+    return out
+
+
+class UnarySequentialLayer(BaseLayer):
+  """Sequential over exactly one tensor (reference builder_layers.py:554)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('sub', [], 'Sub-layer params.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    self.CreateChildren(
+        'seq', [sp.Copy().Set(name=f'sub_{i}')
+                for i, sp in enumerate(self.p.sub)])
+
+  def FProp(self, theta: NestedMap, x):
+    for i, layer in enumerate(self.seq):
+      x = layer.FProp(theta.seq[i], x)
+    return x
+
+
+class BranchLayer(BaseLayer):
+  """Runs a GraphLayer body and appends named intermediate tensors to
+  its outputs (reference builder_layers.py:1256 BranchLayer; fetches
+  here are graph tensor names rather than TF activation fetch points)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('body', None, 'GraphLayer params.')
+    p.Define('fetches', [], 'Graph tensor names to append to outputs.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    body = self.p.body.Copy()
+    body.output_endpoints = list(body.output_endpoints) +         [f for f in self.p.fetches if f not in body.output_endpoints]
+    self.CreateChild('body', body)
+
+  def FProp(self, theta: NestedMap, *args):
+    out = self.body.FProp(theta.body, *args)
+    return out
+
+
+class PrintShapeLayer(BaseLayer):
+  """Identity that logs arg shapes (reference builder_layers.py:1402)."""
+
+  def FProp(self, theta: NestedMap, *args):
+    for i, a in enumerate(args):
+      if isinstance(a, torch.Tensor):
+        print(f'{self.p.name} arg{i}: shape={tuple(a.shape)} '
+              f'dtype={a.dtype}')
+      else:
+        print(f'{self.p.name} arg{i}: {a!r}')
+    return args[0] if len(args) == 1 else args
+
+
+class ReshapeLayer(BaseLayer):
+  """Reshapes input (reference builder_layers.py:1428); -1 allowed."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('shape', [], 'Target shape.')
+    return p
+
+  def FProp(self, theta: NestedMap, x: torch.Tensor) -> torch.Tensor:
+    return x.reshape(*self.p.shape)
+
+
+class ConcatLayer(BaseLayer):
+  """Concatenates args along an axis (reference builder_layers.py:1451)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('axis', -1, 'Concat axis.')
+    return p
+
+  def FProp(self, theta: NestedMap, *args):
+    return torch.cat(list(args), dim=self.p.axis)
+
+
+class SliceLayer(BaseLayer):
+  """Slices the last dim (reference builder_layers.py:1474)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('begin', 0, 'Start index (last dim).')
+    p.Define('size', -1, 'Length (-1 = to end).')
+    return p
+
+  def FProp(self, theta: NestedMap, x: torch.Tensor) -> torch.Tensor:
+    b = self.p.begin
+    return x[..., b:] if self.p.size < 0 else         x[..., b:b + self.p.size]
+
+
+class SoftCondLayer(BaseLayer):
+  """Soft conditional computation (reference builder_layers.py:274;
+  arXiv:1904.04971): body runs with a per-batch sigmoid-weighted
+  average over num_experts copies of its theta. Weights come from the
+  mean input embedding (the reference also collapses the batch)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('body', None, 'Params of the wrapped layer.')
+    p.Define('num_experts', 0, 'Expert (theta copy) count.')
+    p.Define('cond_dim', 0, 'Input dim for the gating projection.')
+    p.Define('nonzeros_mean', False,
+             'Mean over nonzero rows only (packed inputs).')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    assert p.num_experts and p.cond_dim
+    self.CreateChildren(
+        'experts',
+        [p.body.Copy().Set(name=f'expert_{i}')
+         for i in range(p.num_experts)])
+    self.CreateVariable('w', py_utils.WeightParams(
+        [p.cond_dim, p.num_experts], p.params_init, p.dtype))
+
+  def _GetExpertDist(self, theta: NestedMap,
+                     inputs: torch.Tensor) -> torch.Tensor:
+    flat = inputs.reshape(-1, self.p.cond_dim).float()
+    if self.p.nonzeros_mean:
+      nonzero = (flat.abs().sum(-1, keepdim=True) > 0).float()
+      emb = (flat * nonzero).sum(0) / nonzero.sum().clamp_min(1e-10)
+    else:
+      emb = flat.mean(0)
+    return torch.sigmoid(emb @ theta.w.float())
+
+  def FProp(self, theta: NestedMap, inputs: torch.Tensor, *args):
+    dist = self._GetExpertDist(theta, inputs)
+    flat_thetas = [theta.experts[i].Flatten()
+                   for i in range(self.p.num_experts)]
+    mixed = []
+    for vals in zip(*flat_thetas):
+      if isinstance(vals[0], torch.Tensor):
+        stacked = torch.stack([v.float() for v in vals])
+        mix = torch.einsum('e,e...->...', dist, stacked)
+        mixed.append(mix.to(vals[0].dtype))
+      else:
+        mixed.append(vals[0])
+    weighted = theta.experts[0].Pack(mixed)
+    return self.experts[0].FProp(weighted, inputs, *args)
+
+
 class Builder:
   """Pattern-based model-stack builder (reference builder.py:38): each
   method returns a Params TREE; composition happens on params, and a
