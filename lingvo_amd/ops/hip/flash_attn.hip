@@ -395,7 +395,7 @@ __global__ void fa_bwd_delta(const unsigned short* __restrict__ dout,
 // ---------------------------------------------------------------------------
 // Backward main
 // ---------------------------------------------------------------------------
-template <int H, bool BIAS_GRAD, int KTB, int NWB>
+template <int H, bool BIAS_GRAD, int KTB, int NWB, int QTB>
 __global__ __launch_bounds__(NWB * WAVE_SIZE) void fa_bwd_kernel(
     const unsigned short* __restrict__ dout,
     const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
@@ -413,16 +413,16 @@ __global__ __launch_bounds__(NWB * WAVE_SIZE) void fa_bwd_kernel(
   constexpr int HF = H / 16;
   const int nbias = 2 * bias_clip + 1;
   extern __shared__ char smem[];
-  char* q_lds = smem;                        // [QT][H] swz
-  char* qt_lds = q_lds + QT * ROWB;          // [H][QT] swz
-  char* do_lds = qt_lds + H * QT * 2;        // [QT][H] swz
-  char* dot_lds = do_lds + QT * ROWB;        // [H][QT] swz
-  char* kt_lds = dot_lds + H * QT * 2;       // [H][KTB] swz
-  char* ds_lds = kt_lds + H * KTB * 2;       // [QT][KTB] swz
-  char* a_lds = ds_lds + QT * KTB * 2;       // [NWB][16][QT] swz
-  float* lse_s = (float*)(a_lds + NWB * 16 * QT * 2);  // [QT]
-  float* delta_s = lse_s + QT;                         // [QT]
-  float* dbias_s = delta_s + QT;                       // [nbias] if BIAS_GRAD
+  char* q_lds = smem;                        // [QTB][H] swz
+  char* qt_lds = q_lds + QTB * ROWB;          // [H][QTB] swz
+  char* do_lds = qt_lds + H * QTB * 2;        // [QTB][H] swz
+  char* dot_lds = do_lds + QTB * ROWB;        // [H][QTB] swz
+  char* kt_lds = dot_lds + H * QTB * 2;       // [H][KTB] swz
+  char* ds_lds = kt_lds + H * KTB * 2;       // [QTB][KTB] swz
+  char* a_lds = ds_lds + QTB * KTB * 2;       // [NWB][16][QTB] swz
+  float* lse_s = (float*)(a_lds + NWB * 16 * QTB * 2);  // [QTB]
+  float* delta_s = lse_s + QTB;                         // [QTB]
+  float* dbias_s = delta_s + QTB;                       // [nbias] if BIAS_GRAD
 
   const int kt = blockIdx.x;
   const int nkv = blockIdx.y;
@@ -487,7 +487,7 @@ __global__ __launch_bounds__(NWB * WAVE_SIZE) void fa_bwd_kernel(
       return;
     }
   }
-  const int qt_lo = qlo / QT, qt_hi = max(qhi, 0) / QT;
+  const int qt_lo = qlo / QTB, qt_hi = max(qhi, 0) / QTB;
 
   for (int head = 0; head < group; ++head) {
     const int n = nkv * group + head;
@@ -504,23 +504,23 @@ __global__ __launch_bounds__(NWB * WAVE_SIZE) void fa_bwd_kernel(
     }
 
     for (int qt2 = qt_lo; qt2 <= qt_hi; ++qt2) {
-      const int qb = qt2 * QT;
+      const int qb = qt2 * QTB;
       __syncthreads();
-      stage_regular<H, QT, NWB * WAVE_SIZE>(
+      stage_regular<H, QTB, NWB * WAVE_SIZE>(
           q + (((long)b * T + qb) * N + n) * H, (long)N * H, T - qb,
           q_lds);
-      stage_regular<H, QT, NWB * WAVE_SIZE>(
+      stage_regular<H, QTB, NWB * WAVE_SIZE>(
           dout + (((long)b * T + qb) * N + n) * H, (long)N * H, T - qb,
           do_lds);
       if (!(skip & 2)) {
-        stage_transposed<H, QT, NWB * WAVE_SIZE>(
+        stage_transposed<H, QTB, NWB * WAVE_SIZE>(
             q + (((long)b * T + qb) * N + n) * H, (long)N * H, T - qb,
             qt_lds);
-        stage_transposed<H, QT, NWB * WAVE_SIZE>(
+        stage_transposed<H, QTB, NWB * WAVE_SIZE>(
             dout + (((long)b * T + qb) * N + n) * H, (long)N * H, T - qb,
             dot_lds);
       }
-      for (int i = threadIdx.x; i < QT; i += NWB * WAVE_SIZE) {
+      for (int i = threadIdx.x; i < QTB; i += NWB * WAVE_SIZE) {
         int qrow = qb + i;
         lse_s[i] = qrow < T ? lse[((long)b * N + n) * T + qrow] : NEG_INF;
         delta_s[i] = qrow < T ? delta[((long)b * N + n) * T + qrow] : 0.f;
@@ -528,9 +528,10 @@ __global__ __launch_bounds__(NWB * WAVE_SIZE) void fa_bwd_kernel(
       __syncthreads();
 
       // Per-lane chunk windows for this q-tile's 4 q columns.
-      int bw_lo[4], bw_hi[4];
+      constexpr int NQ = QTB / 16;
+      int bw_lo[NQ], bw_hi[NQ];
 #pragma unroll
-      for (int nf = 0; nf < 4; ++nf) {
+      for (int nf = 0; nf < NQ; ++nf) {
         if (chunk > 0) {
           const int qc = (qb + nf * 16 + cl) / chunk;
           bw_lo[nf] = (qc - lc) * chunk;
@@ -541,11 +542,11 @@ __global__ __launch_bounds__(NWB * WAVE_SIZE) void fa_bwd_kernel(
         }
       }
 
-      // S^T strip: rows = 16 keys (this wave), cols = QT queries.
-      float pt[4][4];   // P^T
-      float dlg[4][4];  // dLogits
+      // S^T strip: rows = 16 keys (this wave), cols = QTB queries.
+      float pt[NQ][4];  // P^T
+      float dlg[NQ][4];  // dLogits
 #pragma unroll
-      for (int nf = 0; nf < 4; ++nf) {
+      for (int nf = 0; nf < NQ; ++nf) {
         f32x4 acc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
         for (int kk = 0; kk < KH; ++kk) {
@@ -575,32 +576,32 @@ __global__ __launch_bounds__(NWB * WAVE_SIZE) void fa_bwd_kernel(
       }
 
       // P^T -> a_lds; dV += P^T @ dO (B from dOt).
-      char* aw = a_lds + wid * 16 * (QT * 2);
+      char* aw = a_lds + wid * 16 * (QTB * 2);
 #pragma unroll
-      for (int nf = 0; nf < 4; ++nf) {
+      for (int nf = 0; nf < NQ; ++nf) {
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           int row = g * 4 + r;
           int col = nf * 16 + cl;
           *reinterpret_cast<unsigned short*>(
-              aw + row * (QT * 2) + swz(row, col * 2)) =
+              aw + row * (QTB * 2) + swz(row, col * 2)) =
               float_to_bf16_bits(pt[nf][r]);
         }
       }
 #pragma unroll
-      for (int kk2 = 0; kk2 < QT / 32; ++kk2) {
-        bf16x8 pa = lds_frag(aw, cl, QT * 2, (kk2 * 32 + g * 8) * 2);
+      for (int kk2 = 0; kk2 < QTB / 32; ++kk2) {
+        bf16x8 pa = lds_frag(aw, cl, QTB * 2, (kk2 * 32 + g * 8) * 2);
 #pragma unroll
         for (int hf = 0; hf < HF; ++hf) {
           bf16x8 bd =
-              lds_frag(dot_lds, hf * 16 + cl, QT * 2, (kk2 * 32 + g * 8) * 2);
+              lds_frag(dot_lds, hf * 16 + cl, QTB * 2, (kk2 * 32 + g * 8) * 2);
           acc_dv[hf] = mfma16x16x32_bf16(pa, bd, acc_dv[hf]);
         }
       }
 
       // dP^T = V @ dO^T (B from do_lds rows).
 #pragma unroll
-      for (int nf = 0; nf < 4; ++nf) {
+      for (int nf = 0; nf < NQ; ++nf) {
         f32x4 acc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
         for (int kk = 0; kk < KH; ++kk) {
@@ -617,23 +618,23 @@ __global__ __launch_bounds__(NWB * WAVE_SIZE) void fa_bwd_kernel(
 
       // dS^T (scaled, bf16) -> a_lds (overwrite); dK += dS^T @ Q (B=Qt).
 #pragma unroll
-      for (int nf = 0; nf < 4; ++nf) {
+      for (int nf = 0; nf < NQ; ++nf) {
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           int row = g * 4 + r;
           int col = nf * 16 + cl;
           *reinterpret_cast<unsigned short*>(
-              aw + row * (QT * 2) + swz(row, col * 2)) =
+              aw + row * (QTB * 2) + swz(row, col * 2)) =
               float_to_bf16_bits(dlg[nf][r] * scale);
         }
       }
 #pragma unroll
-      for (int kk2 = 0; kk2 < QT / 32; ++kk2) {
-        bf16x8 da = lds_frag(aw, cl, QT * 2, (kk2 * 32 + g * 8) * 2);
+      for (int kk2 = 0; kk2 < QTB / 32; ++kk2) {
+        bf16x8 da = lds_frag(aw, cl, QTB * 2, (kk2 * 32 + g * 8) * 2);
 #pragma unroll
         for (int hf = 0; hf < HF; ++hf) {
           bf16x8 bq =
-              lds_frag(qt_lds, hf * 16 + cl, QT * 2, (kk2 * 32 + g * 8) * 2);
+              lds_frag(qt_lds, hf * 16 + cl, QTB * 2, (kk2 * 32 + g * 8) * 2);
           acc_dk[hf] = mfma16x16x32_bf16(da, bq, acc_dk[hf]);
         }
       }
@@ -641,7 +642,7 @@ __global__ __launch_bounds__(NWB * WAVE_SIZE) void fa_bwd_kernel(
       // Full dS tile to LDS (transposed store: ds_lds[q][key]) for dQ.
       if (!(skip & 4))
 #pragma unroll
-      for (int nf = 0; nf < 4; ++nf) {
+      for (int nf = 0; nf < NQ; ++nf) {
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           int qrow = nf * 16 + cl;
@@ -654,9 +655,9 @@ __global__ __launch_bounds__(NWB * WAVE_SIZE) void fa_bwd_kernel(
       __syncthreads();
 
       // dQ strips: rows q = qb + dw*16 + .. (A from ds_lds), B = K from
-      // kt_lds; atomic-accumulate fp32. Only QT/16 strips exist, so with
-      // NWB > QT/16 the extra waves skip this phase.
-      if (wid < QT / 16 && !(skip & 1)) {
+      // kt_lds; atomic-accumulate fp32. Only QTB/16 strips exist, so with
+      // NWB > QTB/16 the extra waves skip this phase.
+      if (wid < QTB / 16 && !(skip & 1)) {
         f32x4 acc_dq[HF];
 #pragma unroll
         for (int hf = 0; hf < HF; ++hf) acc_dq[hf] = {0.f, 0.f, 0.f, 0.f};
@@ -685,22 +686,22 @@ __global__ __launch_bounds__(NWB * WAVE_SIZE) void fa_bwd_kernel(
             }
           }
         }
-      } else if (BIAS_GRAD && wid >= QT / 16) {
+      } else if (BIAS_GRAD && wid >= QTB / 16) {
         // Bias grad via per-diagonal sums over the dS tile in ds_lds
         // (entries hold dlogits*scale; invisible positions are exact
         // zeros). One lane per diagonal d = qcol - key: elements
         // (qrow, key) with qb + qrow - kbase - key == d. Runs on the
         // waves otherwise idle during the dQ phase.
-        const int ndiag = QT + KTB - 1;
-        const int lane_global = (wid - QT / 16) * WAVE_SIZE + lane;
-        const int nworkers = (NWB - QT / 16) * WAVE_SIZE;
+        const int ndiag = QTB + KTB - 1;
+        const int lane_global = (wid - QTB / 16) * WAVE_SIZE + lane;
+        const int nworkers = (NWB - QTB / 16) * WAVE_SIZE;
         const float inv_scale = 1.f / scale;
         for (int di = lane_global; di < ndiag; di += nworkers) {
           // diagonal offset within the tile: qrow - key = di - (KTB-1)
           const int doff = di - (KTB - 1);
           float sum = 0.f;
           const int q_lo = max(0, doff);
-          const int q_hi = min(QT - 1, KTB - 1 + doff);
+          const int q_hi = min(QTB - 1, KTB - 1 + doff);
           for (int qrow = q_lo; qrow <= q_hi; ++qrow) {
             const int key = qrow - doff;
             sum += bf16_bits_to_float(*reinterpret_cast<unsigned short*>(
@@ -922,16 +923,21 @@ std::vector<torch::Tensor> fa_bwd(torch::Tensor dout, torch::Tensor q,
   // uses 144KB LDS -> 1 block/CU; staging savings outweigh occupancy.
   const int ktb = 128;
   const int nwb = 8;
-  size_t shmem = (size_t)QT * H * 2 * 2     // q_lds + do_lds
-                 + (size_t)H * QT * 2 * 2   // qt_lds + dot_lds
-                 + (size_t)H * ktb * 2      // kt_lds
-                 + (size_t)QT * ktb * 2     // ds_lds
-                 + (size_t)nwb * 16 * QT * 2  // a_lds
-                 + 2 * QT * sizeof(float) + (bg ? nbias * sizeof(float) : 0);
+  // q-tile 32 (not the fwd's 64): halves q/ds/a LDS to ~64 KB so TWO
+  // blocks co-reside per CU (4 waves/SIMD) — the MFMA chain is latency-
+  // bound at 1 block/CU. Staging traffic is unchanged (same bytes over
+  // 2x the tiles).
+  const int qtb = 32;
+  size_t shmem = (size_t)qtb * H * 2 * 2     // q_lds + do_lds
+                 + (size_t)H * qtb * 2 * 2   // qt_lds + dot_lds
+                 + (size_t)H * ktb * 2       // kt_lds
+                 + (size_t)qtb * ktb * 2     // ds_lds
+                 + (size_t)nwb * 16 * qtb * 2  // a_lds
+                 + 2 * qtb * sizeof(float) + (bg ? nbias * sizeof(float) : 0);
   dim3 grid((S + ktb - 1) / ktb, NKV, B);
 #define FA_BWD(HH, BG)                                                       \
   hipLaunchKernelGGL(                                                        \
-      (fa_bwd_kernel<HH, BG, 128, 8>), grid, dim3(8 * WAVE_SIZE), shmem,     \
+      (fa_bwd_kernel<HH, BG, 128, 8, 32>), grid, dim3(8 * WAVE_SIZE), shmem, \
       stream,                                                                \
       (const unsigned short*)dout.data_ptr(),                                \
       (const unsigned short*)q.data_ptr(),                                   \
