@@ -76,19 +76,36 @@ class CurriculumDataSource(DataSource):
 
   def __init__(self, params):
     super().__init__(params)
-    assert len(self.p.sub) == len(self.p.boundaries) + 1
-    self.CreateChildren('sources', [sp.Copy() for sp in self.p.sub])
+    p = self.p
+    if len(p.sub) != len(p.boundaries) + 1:
+      raise ValueError(
+          'Expected one more sub than boundaries; got %d sub, %d '
+          'boundaries' % (len(p.sub), len(p.boundaries)))
+    if list(p.boundaries) != sorted(p.boundaries):
+      raise ValueError('boundaries must be monotonically increasing')
+    self.CreateChildren('sources', [sp.Copy() for sp in p.sub])
     self._step = 0
 
   def SetStep(self, step: int) -> None:
-    self._step = step
+    self._step = int(step)
 
-  def GetNext(self) -> NestedMap:
+  # Runner-facing alias (Trainer calls SetGlobalStep each step).
+  SetGlobalStep = SetStep
+
+  @property
+  def current_stage(self) -> int:
     idx = 0
     for b in self.p.boundaries:
       if self._step >= b:
         idx += 1
-    return self.sources[idx].GetNext()
+    return idx
+
+  def GetNext(self) -> NestedMap:
+    return self.sources[self.current_stage].GetNext()
+
+  def Reset(self) -> None:
+    for src in self.sources:
+      src.Reset()
 
 
 class SequentialDataSource(DataSource):
@@ -187,45 +204,3 @@ class WithinBatchMixingDataSource(DataSource):
     out.source_id = torch.tensor(srcs)
     return out
 
-
-class CurriculumDataSource(DataSource):
-  """Reads different sub-sources in stages gated by the training global
-  step (reference datasource.py:253). p.sub has one more entry than
-  p.boundaries; stage i serves while step < boundaries[i]. The runner
-  advances the step via SetGlobalStep each train step."""
-
-  @classmethod
-  def Params(cls):
-    p = super().Params()
-    p.Define('sub', [], 'DataSource params, one per stage.')
-    p.Define('boundaries', [], 'Global-step thresholds between stages.')
-    return p
-
-  def __init__(self, params):
-    super().__init__(params)
-    p = self.p
-    if len(p.sub) != len(p.boundaries) + 1:
-      raise ValueError(
-          'Expected one more sub than boundaries; got %d sub, %d '
-          'boundaries' % (len(p.sub), len(p.boundaries)))
-    if list(p.boundaries) != sorted(p.boundaries):
-      raise ValueError('boundaries must be monotonically increasing')
-    self.CreateChildren('sources', [sp.Copy() for sp in p.sub])
-    self._step = 0
-
-  def SetGlobalStep(self, step: int) -> None:
-    self._step = int(step)
-
-  @property
-  def current_stage(self) -> int:
-    for i, b in enumerate(self.p.boundaries):
-      if self._step < b:
-        return i
-    return len(self.p.boundaries)
-
-  def GetNext(self) -> NestedMap:
-    return self.sources[self.current_stage].GetNext()
-
-  def Reset(self) -> None:
-    for src in self.sources:
-      src.Reset()

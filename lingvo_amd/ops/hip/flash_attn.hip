@@ -67,9 +67,11 @@ __device__ void stage_regular(const unsigned short* src, long src_stride,
 
 // Stage ROWS x H bf16 from global into LDS TRANSPOSED as [H][ROWS]
 // (ROWS=64, 128B rows), swizzled per h-row. Scalar 2B writes.
-template <int H, int ROWS, int NT = BLOCK>
+template <int H, int ROWS, int NT = BLOCK, int PITCHB = ROWS * 2>
 __device__ void stage_transposed(const unsigned short* src, long src_stride,
                                  int valid_rows, char* dst) {
+  // PITCHB must be >= 128 when the swizzle is in play: swz() XORs up to
+  // 112 bytes into the row, so a 64-byte pitch would corrupt neighbors.
   constexpr int ROWB = H * 2;
   constexpr int TPR = ROWB / 16;
   constexpr int RPP = NT / TPR;
@@ -90,7 +92,7 @@ __device__ void stage_transposed(const unsigned short* src, long src_stride,
     for (int j = 0; j < 8; ++j) {
       int h = h0 + j;
       *reinterpret_cast<unsigned short*>(
-          dst + h * (ROWS * 2) + swz(h, row * 2)) = v[j];
+          dst + h * PITCHB + swz(h, row * 2)) = v[j];
     }
   }
 }
@@ -413,14 +415,17 @@ __global__ __launch_bounds__(NWB * WAVE_SIZE) void fa_bwd_kernel(
   constexpr int HF = H / 16;
   const int nbias = 2 * bias_clip + 1;
   extern __shared__ char smem[];
+  // qt/dot/a tiles keep a 128-byte row pitch regardless of QTB: the
+  // swizzle XORs up to 112 bytes within a row.
+  constexpr int TP = 128;
   char* q_lds = smem;                        // [QTB][H] swz
-  char* qt_lds = q_lds + QTB * ROWB;          // [H][QTB] swz
-  char* do_lds = qt_lds + H * QTB * 2;        // [QTB][H] swz
-  char* dot_lds = do_lds + QTB * ROWB;        // [H][QTB] swz
-  char* kt_lds = dot_lds + H * QTB * 2;       // [H][KTB] swz
+  char* qt_lds = q_lds + QTB * ROWB;         // [H][QTB] swz, pitch TP
+  char* do_lds = qt_lds + H * TP;            // [QTB][H] swz
+  char* dot_lds = do_lds + QTB * ROWB;       // [H][QTB] swz, pitch TP
+  char* kt_lds = dot_lds + H * TP;           // [H][KTB] swz
   char* ds_lds = kt_lds + H * KTB * 2;       // [QTB][KTB] swz
-  char* a_lds = ds_lds + QTB * KTB * 2;       // [NWB][16][QTB] swz
-  float* lse_s = (float*)(a_lds + NWB * 16 * QTB * 2);  // [QTB]
+  char* a_lds = ds_lds + QTB * KTB * 2;      // [NWB][16][QTB] swz, pitch TP
+  float* lse_s = (float*)(a_lds + NWB * 16 * TP);       // [QTB]
   float* delta_s = lse_s + QTB;                         // [QTB]
   float* dbias_s = delta_s + QTB;                       // [nbias] if BIAS_GRAD
 
@@ -513,10 +518,10 @@ __global__ __launch_bounds__(NWB * WAVE_SIZE) void fa_bwd_kernel(
           dout + (((long)b * T + qb) * N + n) * H, (long)N * H, T - qb,
           do_lds);
       if (!(skip & 2)) {
-        stage_transposed<H, QTB, NWB * WAVE_SIZE>(
+        stage_transposed<H, QTB, NWB * WAVE_SIZE, TP>(
             q + (((long)b * T + qb) * N + n) * H, (long)N * H, T - qb,
             qt_lds);
-        stage_transposed<H, QTB, NWB * WAVE_SIZE>(
+        stage_transposed<H, QTB, NWB * WAVE_SIZE, TP>(
             dout + (((long)b * T + qb) * N + n) * H, (long)N * H, T - qb,
             dot_lds);
       }
@@ -576,7 +581,7 @@ __global__ __launch_bounds__(NWB * WAVE_SIZE) void fa_bwd_kernel(
       }
 
       // P^T -> a_lds; dV += P^T @ dO (B from dOt).
-      char* aw = a_lds + wid * 16 * (QTB * 2);
+      char* aw = a_lds + wid * 16 * TP;
 #pragma unroll
       for (int nf = 0; nf < NQ; ++nf) {
 #pragma unroll
@@ -584,17 +589,17 @@ __global__ __launch_bounds__(NWB * WAVE_SIZE) void fa_bwd_kernel(
           int row = g * 4 + r;
           int col = nf * 16 + cl;
           *reinterpret_cast<unsigned short*>(
-              aw + row * (QTB * 2) + swz(row, col * 2)) =
+              aw + row * TP + swz(row, col * 2)) =
               float_to_bf16_bits(pt[nf][r]);
         }
       }
 #pragma unroll
       for (int kk2 = 0; kk2 < QTB / 32; ++kk2) {
-        bf16x8 pa = lds_frag(aw, cl, QTB * 2, (kk2 * 32 + g * 8) * 2);
+        bf16x8 pa = lds_frag(aw, cl, TP, (kk2 * 32 + g * 8) * 2);
 #pragma unroll
         for (int hf = 0; hf < HF; ++hf) {
           bf16x8 bd =
-              lds_frag(dot_lds, hf * 16 + cl, QTB * 2, (kk2 * 32 + g * 8) * 2);
+              lds_frag(dot_lds, hf * 16 + cl, TP, (kk2 * 32 + g * 8) * 2);
           acc_dv[hf] = mfma16x16x32_bf16(pa, bd, acc_dv[hf]);
         }
       }
@@ -624,17 +629,17 @@ __global__ __launch_bounds__(NWB * WAVE_SIZE) void fa_bwd_kernel(
           int row = g * 4 + r;
           int col = nf * 16 + cl;
           *reinterpret_cast<unsigned short*>(
-              aw + row * (QTB * 2) + swz(row, col * 2)) =
+              aw + row * TP + swz(row, col * 2)) =
               float_to_bf16_bits(dlg[nf][r] * scale);
         }
       }
 #pragma unroll
       for (int kk2 = 0; kk2 < QTB / 32; ++kk2) {
-        bf16x8 da = lds_frag(aw, cl, QTB * 2, (kk2 * 32 + g * 8) * 2);
+        bf16x8 da = lds_frag(aw, cl, TP, (kk2 * 32 + g * 8) * 2);
 #pragma unroll
         for (int hf = 0; hf < HF; ++hf) {
           bf16x8 bq =
-              lds_frag(qt_lds, hf * 16 + cl, QTB * 2, (kk2 * 32 + g * 8) * 2);
+              lds_frag(qt_lds, hf * 16 + cl, TP, (kk2 * 32 + g * 8) * 2);
           acc_dk[hf] = mfma16x16x32_bf16(da, bq, acc_dk[hf]);
         }
       }
@@ -929,10 +934,10 @@ std::vector<torch::Tensor> fa_bwd(torch::Tensor dout, torch::Tensor q,
   // 2x the tiles).
   const int qtb = 32;
   size_t shmem = (size_t)qtb * H * 2 * 2     // q_lds + do_lds
-                 + (size_t)H * qtb * 2 * 2   // qt_lds + dot_lds
+                 + (size_t)H * 128 * 2       // qt_lds + dot_lds (128B pitch)
                  + (size_t)H * ktb * 2       // kt_lds
                  + (size_t)qtb * ktb * 2     // ds_lds
-                 + (size_t)nwb * 16 * qtb * 2  // a_lds
+                 + (size_t)nwb * 16 * 128    // a_lds (128B pitch)
                  + 2 * qtb * sizeof(float) + (bg ? nbias * sizeof(float) : 0);
   dim3 grid((S + ktb - 1) / ktb, NKV, B);
 #define FA_BWD(HH, BG)                                                       \
