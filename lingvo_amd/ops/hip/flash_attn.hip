@@ -52,8 +52,9 @@ __device__ void stage_regular(const unsigned short* src, long src_stride,
   const int r0 = tid / TPR;
   const int byte0 = (tid % TPR) * 16;
 #pragma unroll
-  for (int rp = 0; rp < ROWS / RPP; ++rp) {
+  for (int rp = 0; rp < (ROWS + RPP - 1) / RPP; ++rp) {
     int row = r0 + rp * RPP;
+    if (row >= ROWS) break;  // ROWS < RPP: excess threads idle
     ushortx8 v;
     if (row < valid_rows) {
       v = *reinterpret_cast<const ushortx8*>(src + (long)row * src_stride +
@@ -79,8 +80,9 @@ __device__ void stage_transposed(const unsigned short* src, long src_stride,
   const int r0 = tid / TPR;
   const int h0 = (tid % TPR) * 8;
 #pragma unroll
-  for (int rp = 0; rp < ROWS / RPP; ++rp) {
+  for (int rp = 0; rp < (ROWS + RPP - 1) / RPP; ++rp) {
     int row = r0 + rp * RPP;
+    if (row >= ROWS) break;  // ROWS < RPP: excess threads idle
     ushortx8 v;
     if (row < valid_rows) {
       v = *reinterpret_cast<const ushortx8*>(src + (long)row * src_stride +
