@@ -54,7 +54,8 @@ __device__ void stage_regular(const unsigned short* src, long src_stride,
 #pragma unroll
   for (int rp = 0; rp < (ROWS + RPP - 1) / RPP; ++rp) {
     int row = r0 + rp * RPP;
-    if (row >= ROWS) break;  // ROWS < RPP: excess threads idle
+    // constexpr-folded for ROWS % RPP == 0 (the fwd tile shapes).
+    if ((ROWS % RPP) != 0 && row >= ROWS) break;
     ushortx8 v;
     if (row < valid_rows) {
       v = *reinterpret_cast<const ushortx8*>(src + (long)row * src_stride +
@@ -82,7 +83,7 @@ __device__ void stage_transposed(const unsigned short* src, long src_stride,
 #pragma unroll
   for (int rp = 0; rp < (ROWS + RPP - 1) / RPP; ++rp) {
     int row = r0 + rp * RPP;
-    if (row >= ROWS) break;  // ROWS < RPP: excess threads idle
+    if ((ROWS % RPP) != 0 && row >= ROWS) break;
     ushortx8 v;
     if (row < valid_rows) {
       v = *reinterpret_cast<const ushortx8*>(src + (long)row * src_stride +
@@ -930,11 +931,10 @@ std::vector<torch::Tensor> fa_bwd(torch::Tensor dout, torch::Tensor q,
   // uses 144KB LDS -> 1 block/CU; staging savings outweigh occupancy.
   const int ktb = 128;
   const int nwb = 8;
-  // q-tile 32 (not the fwd's 64): halves q/ds/a LDS to ~64 KB so TWO
-  // blocks co-reside per CU (4 waves/SIMD) — the MFMA chain is latency-
-  // bound at 1 block/CU. Staging traffic is unchanged (same bytes over
-  // 2x the tiles).
-  const int qtb = 32;
+  // q-tile 64. A 32-row variant (2 blocks/CU via ~64 KB LDS) measured
+  // SLOWER (bwd 0.73 -> 0.81 ms at the bench shape): doubled staging
+  // sync passes outweigh the occupancy gain. QTB stays a template knob.
+  const int qtb = 64;
   size_t shmem = (size_t)qtb * H * 2 * 2     // q_lds + do_lds
                  + (size_t)H * 128 * 2       // qt_lds + dot_lds (128B pitch)
                  + (size_t)H * ktb * 2       // kt_lds
@@ -944,7 +944,7 @@ std::vector<torch::Tensor> fa_bwd(torch::Tensor dout, torch::Tensor q,
   dim3 grid((S + ktb - 1) / ktb, NKV, B);
 #define FA_BWD(HH, BG)                                                       \
   hipLaunchKernelGGL(                                                        \
-      (fa_bwd_kernel<HH, BG, 128, 8, 32>), grid, dim3(8 * WAVE_SIZE), shmem, \
+      (fa_bwd_kernel<HH, BG, 128, 8, 64>), grid, dim3(8 * WAVE_SIZE), shmem, \
       stream,                                                                \
       (const unsigned short*)dout.data_ptr(),                                \
       (const unsigned short*)q.data_ptr(),                                   \
