@@ -258,3 +258,37 @@ def MaterializeInt8Weights(module: torch.nn.Module):
 def DequantizeInt8(q: torch.Tensor, scale: float,
                    dtype=torch.float32) -> torch.Tensor:
   return q.to(dtype) * scale
+
+
+class LinearClippingCapSchedule(BaseLayer):
+  """Linearly narrows a symmetric clipping cap from start_cap to
+  end_cap over [start_step, end_step] (reference quant_utils.py:1246)."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('start_step', 0, 'Step to begin narrowing.')
+    p.Define('end_step', 15000, 'Step end_cap is reached.')
+    p.Define('start_cap', 8.0, 'Initial clip cap.')
+    p.Define('end_cap', 1.0, 'Final clip cap.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    self._step = 0
+
+  def SetStep(self, step: int) -> None:
+    self._step = int(step)
+
+  def Value(self) -> float:
+    p = self.p
+    if self._step <= p.start_step:
+      return p.start_cap
+    if self._step >= p.end_step:
+      return p.end_cap
+    frac = (self._step - p.start_step) / float(p.end_step - p.start_step)
+    return p.start_cap + frac * (p.end_cap - p.start_cap)
+
+  def ApplyClipping(self, theta, x: torch.Tensor) -> torch.Tensor:
+    cap = self.Value()
+    return torch.clamp(x, -cap, cap)

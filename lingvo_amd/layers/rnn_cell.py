@@ -479,3 +479,48 @@ class ConvLSTMCell(RNNCell):
       c1 = c1 * (1 - pad_t) + state0.c * pad_t
       m1 = m1 * (1 - pad_t) + state0.m * pad_t
     return NestedMap(c=c1, m=m1)
+
+
+class QuantizedLSTMCell(RNNCell):
+  """Quantization-friendly LSTM (reference rnn_cell.py:900): no bias,
+  no forget-gate bias, no output nonlinearity; the cell state is
+  clipped by a LinearClippingCapSchedule that narrows over training."""
+
+  @classmethod
+  def Params(cls):
+    from lingvo_amd.core import quant_utils
+    p = super().Params()
+    p.Define('cc_schedule',
+             quant_utils.LinearClippingCapSchedule.Params(),
+             'Clipping cap schedule.')
+    return p
+
+  def __init__(self, params):
+    super().__init__(params)
+    p = self.p
+    self.CreateChild('cc_schedule', p.cc_schedule)
+    self.CreateVariable('wm', py_utils.WeightParams(
+        [p.num_input_nodes + p.num_output_nodes,
+         4 * p.num_output_nodes], p.params_init, p.dtype))
+
+  def PostTrainingStepUpdate(self, global_step: int) -> None:
+    self.cc_schedule.SetStep(global_step)
+
+  def InitState(self, batch, device, dtype) -> NestedMap:
+    h = self.p.num_output_nodes
+    return NestedMap(c=torch.zeros(batch, h, device=device, dtype=dtype),
+                     m=torch.zeros(batch, h, device=device, dtype=dtype))
+
+  def FProp(self, theta, state0, inputs):
+    xm = torch.cat([inputs.act, state0.m], dim=-1)
+    gates = torch.matmul(xm, theta.wm)
+    i_i, i_g, f_g, o_g = gates.chunk(4, dim=-1)
+    c1 = torch.sigmoid(f_g) * state0.c + \
+        torch.sigmoid(i_g) * torch.tanh(i_i)
+    c1 = self.cc_schedule.ApplyClipping(theta.cc_schedule, c1)
+    m1 = torch.sigmoid(o_g) * c1
+    pad = inputs.Get('padding')
+    if pad is not None:
+      c1 = c1 * (1 - pad) + state0.c * pad
+      m1 = m1 * (1 - pad) + state0.m * pad
+    return NestedMap(c=c1, m=m1)

@@ -141,3 +141,23 @@ def test_conv_lstm_cell():
   assert s1.m.shape == (2, 5, 6, 4)
   assert torch.equal(s1.c[1], s0.c[1]) and torch.equal(s1.m[1], s0.m[1])
   assert not torch.equal(s1.c[0], s0.c[0])
+
+
+def test_quantized_lstm_cell_clipping_schedule():
+  torch.manual_seed(0)
+  p = rnn_cell.QuantizedLSTMCell.Params().Set(
+      name='q', num_input_nodes=6, num_output_nodes=4)
+  p.cc_schedule.Set(start_step=0, end_step=100, start_cap=8.0,
+                    end_cap=1.0)
+  cell = p.Instantiate()
+  assert cell.cc_schedule.Value() == 8.0
+  cell.PostTrainingStepUpdate(50)
+  assert abs(cell.cc_schedule.Value() - 4.5) < 1e-6
+  cell.PostTrainingStepUpdate(1000)
+  assert cell.cc_schedule.Value() == 1.0
+  s0 = cell.InitState(3, 'cpu', torch.float32)
+  s0.c = torch.full((3, 4), 100.0)
+  ins = NestedMap(act=torch.randn(3, 6), padding=torch.zeros(3, 1))
+  s1 = cell.FProp(cell.theta, s0, ins)
+  assert s1.c.abs().max() <= 1.0 + 1e-6
+  assert s1.m.abs().max() <= 1.0 + 1e-6
