@@ -243,3 +243,52 @@ def BallQuery(points: torch.Tensor, centers: torch.Tensor, radius: float,
     if take.numel() < num_neighbors:
       idx[i, take.numel():] = take[0]
   return idx
+
+
+def FarthestPointSampler(points: torch.Tensor, padding: torch.Tensor,
+                         num_sampled_points: int,
+                         precomputed_squared_distance=None,
+                         num_seeded_points: int = 0, random_seed=None):
+  """Batched farthest-point sampling (reference car_lib.py:244).
+
+  points [N, P, dims]; padding [N, P] (1 = padded). Returns
+  (sampled_idx [N, S] long, closest_idx [N, P] long) where closest_idx
+  maps every input point to its nearest sampled point (PCNN pooling).
+  The first num_seeded_points are taken as-is (assumed unpadded).
+  """
+  n, p1, _ = points.shape
+  s_count = min(num_sampled_points, p1)
+  g = torch.Generator().manual_seed(
+      random_seed if random_seed is not None else 0)
+  big = torch.finfo(torch.float32).max
+
+  def pair_dist(idx):
+    # squared distance from every point to the sampled point idx [N].
+    if precomputed_squared_distance is not None:
+      return precomputed_squared_distance[
+          torch.arange(n), idx]  # [N, P]
+    sel = points[torch.arange(n), idx].unsqueeze(1)  # [N, 1, dims]
+    return (points.float() - sel.float()).pow(2).sum(-1)
+
+  pad_mask = padding > 0.5
+  sampled = torch.zeros(n, s_count, dtype=torch.long)
+  closest = torch.zeros(n, p1, dtype=torch.long)
+  min_dist = torch.full((n, p1), big)
+
+  for step in range(s_count):
+    if step < num_seeded_points:
+      idx = torch.full((n,), step, dtype=torch.long)
+    elif step == 0:
+      # Random unpadded start per batch row.
+      r = torch.rand(n, p1, generator=g).masked_fill(pad_mask, -1.0)
+      idx = r.argmax(dim=1)
+    else:
+      scores = min_dist.masked_fill(pad_mask, -big)
+      idx = scores.argmax(dim=1)
+    sampled[:, step] = idx
+    d = pair_dist(idx)
+    improved = d < min_dist
+    closest = torch.where(improved,
+                          torch.full_like(closest, step), closest)
+    min_dist = torch.minimum(min_dist, d)
+  return sampled, closest

@@ -255,3 +255,30 @@ def test_milan_score_functions_and_id_masking():
   preds = task.ComputePredictions(task.theta, batch)
   mm, _ = task.ComputeLoss(task.theta, preds, batch)
   assert torch.isfinite(mm.loss[0])
+
+
+def test_farthest_point_sampler():
+  import torch
+  from lingvo_amd.models import car_ops
+  torch.manual_seed(0)
+  pts = torch.randn(2, 32, 3)
+  pad = torch.zeros(2, 32)
+  pad[1, 16:] = 1.0  # second row: only first 16 points are real
+  sampled, closest = car_ops.FarthestPointSampler(pts, pad, 8,
+                                                  random_seed=3)
+  assert sampled.shape == (2, 8) and closest.shape == (2, 32)
+  # No duplicates among sampled points per row.
+  for r in range(2):
+    assert len(set(sampled[r].tolist())) == 8
+  # Padded points are never sampled.
+  assert (sampled[1] < 16).all()
+  # closest_idx maps each point to the nearest sampled point.
+  for r in range(2):
+    d = (pts[r, :, None, :] - pts[r, sampled[r]][None]).pow(2).sum(-1)
+    want = d.argmin(dim=1)
+    real = pad[r] < 0.5
+    assert (closest[r][real] == want[real]).all()
+  # Seeded points come first, in order.
+  sampled2, _ = car_ops.FarthestPointSampler(pts, pad, 8,
+                                             num_seeded_points=3)
+  assert sampled2[:, :3].tolist() == [[0, 1, 2], [0, 1, 2]]
