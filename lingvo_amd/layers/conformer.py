@@ -272,10 +272,14 @@ class _Conv3x3S2Nhwc(torch.autograd.Function):
   @staticmethod
   def _s2d(x):
     """[B,H,W,C] (H,W even) -> padded flat [B*(Ho+1)*(Wo+1), 4C].
-    One strided copy per block straight into the padded buffer (no
-    intermediate permute/index_select materialization)."""
+    GPU bf16 with C%8==0 runs the one-pass HIP kernel; otherwise 4
+    strided copies straight into the padded buffer."""
     B, H, W, C = x.shape
     Ho, Wo = H // 2, W // 2
+    if x.is_cuda and x.dtype == torch.bfloat16 and C % 8 == 0:
+      from lingvo_amd.ops import _loader
+      X = _loader.get_ext(required=True).s2d_fwd(x)
+      return X.reshape(B * (Ho + 1) * (Wo + 1), 4 * C), Ho, Wo
     X = x.new_zeros(B, Ho + 1, Wo + 1, 4 * C)
     for i, (a, b) in enumerate(_Conv3x3S2Nhwc.BLOCKS):
       X[:, 1:, 1:, i * C:(i + 1) * C] = x[:, a::2, b::2, :]
@@ -309,7 +313,9 @@ class _Conv3x3S2Nhwc(torch.autograd.Function):
       k0, k1 = blks[0] * C, (blks[-1] + 1) * C
       wc = _Conv3x3S2Nhwc._cell_weight(w.to(x.dtype), taps)
       O[base:].addmm_(X[base + off:Rp + off, k0:k1], wc)
-    ctx.save_for_backward(x, w)
+    # Save the s2d buffer (not x): backward reuses it for the dW and
+    # dX GEMMs without re-running the transform.
+    ctx.save_for_backward(X, w)
     ctx.dims = (B, H, W, C, Ho, Wo, Co, pad_h, pad_w)
     ctx.bias_dtype = bias.dtype
     out = O.reshape(B, Ho + 1, Wp, Co)[:, 1:, 1:, :]
@@ -317,15 +323,20 @@ class _Conv3x3S2Nhwc(torch.autograd.Function):
 
   @staticmethod
   def backward(ctx, dout):
-    x, w = ctx.saved_tensors
+    X, w = ctx.saved_tensors
     B, H, W, C, Ho, Wo, Co, pad_h, pad_w = ctx.dims
     Wp = Wo + 1
     Rp = B * (Ho + 1) * Wp
     base = Wp + 1
-    dO = dout.new_zeros(B, Ho + 1, Wp, Co)
-    dO[:, 1:, 1:, :] = dout.reshape(B, Ho, Wo, Co)
+    dout4 = dout.reshape(B, Ho, Wo, Co)
+    if dout.is_cuda and dout.dtype == torch.bfloat16 and Co % 8 == 0:
+      from lingvo_amd.ops import _loader
+      dO = _loader.get_ext(required=True).pad_scatter(
+          dout4.contiguous())
+    else:
+      dO = dout.new_zeros(B, Ho + 1, Wp, Co)
+      dO[:, 1:, 1:, :] = dout4
     dO = dO.reshape(Rp, Co)
-    X, _, _ = _Conv3x3S2Nhwc._s2d(x)
     dX = torch.zeros_like(X)
     dw = torch.empty_like(w)
     for (dtau, dphi), taps in _Conv3x3S2Nhwc.CELLS:
@@ -342,12 +353,18 @@ class _Conv3x3S2Nhwc(torch.autograd.Function):
       wc_t = _Conv3x3S2Nhwc._cell_weight(w.to(dO.dtype), taps,
                                          transpose=True)
       dX[base + off:Rp + off, k0:k1].addmm_(dO[base:], wc_t)
-    # Inverse s2d: strided scatter of each block's channel slice.
-    dXr = dX.reshape(B, Ho + 1, Wp, 4 * C)[:, 1:, 1:, :]
-    dx_full = dout.new_empty(B, 2 * Ho, 2 * Wo, C)
-    for i, (a, b) in enumerate(_Conv3x3S2Nhwc.BLOCKS):
-      dx_full[:, a::2, b::2, :] = dXr[..., i * C:(i + 1) * C]
-    dx = dx_full
+    # Inverse s2d: one-pass HIP kernel on GPU bf16, else strided
+    # scatter of each block's channel slice.
+    if dX.is_cuda and dX.dtype == torch.bfloat16 and C % 8 == 0:
+      from lingvo_amd.ops import _loader
+      dx = _loader.get_ext(required=True).s2d_inv(
+          dX.reshape(B, Ho + 1, Wp, 4 * C), C)
+    else:
+      dXr = dX.reshape(B, Ho + 1, Wp, 4 * C)[:, 1:, 1:, :]
+      dx_full = dout.new_empty(B, 2 * Ho, 2 * Wo, C)
+      for i, (a, b) in enumerate(_Conv3x3S2Nhwc.BLOCKS):
+        dx_full[:, a::2, b::2, :] = dXr[..., i * C:(i + 1) * C]
+      dx = dx_full
     if pad_h or pad_w:
       dx = dx[:, :2 * Ho - pad_h, :2 * Wo - pad_w, :].contiguous()
     dbias = dO[base:].float().sum(0).to(ctx.bias_dtype)

@@ -105,6 +105,11 @@ std::vector<torch::Tensor> moe_positions(torch::Tensor top1,
                                          torch::Tensor top2,
                                          int64_t num_experts);
 
+// s2d.hip
+torch::Tensor s2d_fwd(torch::Tensor x);
+torch::Tensor s2d_inv(torch::Tensor dX, int64_t C);
+torch::Tensor pad_scatter(torch::Tensor dout);
+
 // las_decoder.hip
 void smallm_gemm(torch::Tensor a, torch::Tensor wt,
                  c10::optional<torch::Tensor> pre, torch::Tensor out,
@@ -146,4 +151,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Small-M MFMA GEMM (decode-step projections)");
   m.def("attend_fwd", &attend_fwd, "Fused dot-attention step fwd");
   m.def("attend_bwd", &attend_bwd, "Fused dot-attention step bwd");
+  m.def("s2d_fwd", &s2d_fwd, "Space-to-depth + pad (conv frontend)");
+  m.def("s2d_inv", &s2d_inv, "Inverse space-to-depth");
+  m.def("pad_scatter", &pad_scatter, "Zero-border pad scatter");
 }
