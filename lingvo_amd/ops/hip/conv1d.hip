@@ -21,11 +21,15 @@ namespace {
 
 constexpr int MAXK = 33;
 
+// KT > 0: compile-time tap count (full unroll + software pipelining);
+// KT == 0: runtime K fallback.
+template <int KT>
 __global__ void dwconv_fwd(const unsigned short* __restrict__ x,
                            const unsigned short* __restrict__ w,
                            const unsigned short* __restrict__ bias,
                            unsigned short* __restrict__ y, int B, int T,
-                           int D, int K, int pad) {
+                           int D, int Krt, int pad) {
+  const int K = KT > 0 ? KT : Krt;
   const long nvec = (long)B * T * (D / 8);
   const int dvec = D / 8;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
@@ -43,7 +47,9 @@ __global__ void dwconv_fwd(const unsigned short* __restrict__ x,
 #pragma unroll
       for (int e = 0; e < 8; ++e) acc[e] = 0.f;
     }
-    for (int j = 0; j < K; ++j) {
+#pragma unroll
+    for (int j = 0; j < (KT > 0 ? KT : 1); ++j) {
+      if (KT == 0) break;
       int ts = t + j - pad;
       if (ts < 0 || ts >= T) continue;
       ushortx8 xv = *reinterpret_cast<const ushortx8*>(
@@ -53,6 +59,19 @@ __global__ void dwconv_fwd(const unsigned short* __restrict__ x,
       for (int e = 0; e < 8; ++e)
         acc[e] += bf16_bits_to_float(xv[e]) * bf16_bits_to_float(wv[e]);
     }
+    if (KT == 0) {
+      for (int j = 0; j < K; ++j) {
+        int ts = t + j - pad;
+        if (ts < 0 || ts >= T) continue;
+        ushortx8 xv = *reinterpret_cast<const ushortx8*>(
+            x + (b * T + ts) * D + dv * 8);
+        ushortx8 wv =
+            *reinterpret_cast<const ushortx8*>(w + j * D + dv * 8);
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          acc[e] += bf16_bits_to_float(xv[e]) * bf16_bits_to_float(wv[e]);
+      }
+    }
     ushortx8 ov;
 #pragma unroll
     for (int e = 0; e < 8; ++e) ov[e] = float_to_bf16_bits(acc[e]);
@@ -60,10 +79,12 @@ __global__ void dwconv_fwd(const unsigned short* __restrict__ x,
   }
 }
 
+template <int KT>
 __global__ void dwconv_bwd_dx(const unsigned short* __restrict__ dy,
                               const unsigned short* __restrict__ w,
                               unsigned short* __restrict__ dx, int B, int T,
-                              int D, int K, int pad) {
+                              int D, int Krt, int pad) {
+  const int K = KT > 0 ? KT : Krt;
   const long nvec = (long)B * T * (D / 8);
   const int dvec = D / 8;
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
@@ -75,7 +96,9 @@ __global__ void dwconv_bwd_dx(const unsigned short* __restrict__ dy,
     float acc[8];
 #pragma unroll
     for (int e = 0; e < 8; ++e) acc[e] = 0.f;
-    for (int j = 0; j < K; ++j) {
+#pragma unroll
+    for (int j = 0; j < (KT > 0 ? KT : 1); ++j) {
+      if (KT == 0) break;
       int ty = t - j + pad;  // y position whose tap j touched x[t]
       if (ty < 0 || ty >= T) continue;
       ushortx8 gv = *reinterpret_cast<const ushortx8*>(
@@ -84,6 +107,19 @@ __global__ void dwconv_bwd_dx(const unsigned short* __restrict__ dy,
 #pragma unroll
       for (int e = 0; e < 8; ++e)
         acc[e] += bf16_bits_to_float(gv[e]) * bf16_bits_to_float(wv[e]);
+    }
+    if (KT == 0) {
+      for (int j = 0; j < K; ++j) {
+        int ty = t - j + pad;
+        if (ty < 0 || ty >= T) continue;
+        ushortx8 gv = *reinterpret_cast<const ushortx8*>(
+            dy + (b * T + ty) * D + dv * 8);
+        ushortx8 wv =
+            *reinterpret_cast<const ushortx8*>(w + j * D + dv * 8);
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          acc[e] += bf16_bits_to_float(gv[e]) * bf16_bits_to_float(wv[e]);
+      }
     }
     ushortx8 ov;
 #pragma unroll
@@ -265,7 +301,10 @@ torch::Tensor dwconv1d_fwd(torch::Tensor x, torch::Tensor w,
   auto stream = at::cuda::getCurrentCUDAStream();
   long nvec = (long)B * T * (D / 8);
   int grid = memory_bound_grid(nvec, 256);
-  hipLaunchKernelGGL(dwconv_fwd, dim3(grid), dim3(256), 0, stream,
+  auto fwd_kern = K == 32 ? dwconv_fwd<32>
+                : K == 16 ? dwconv_fwd<16>
+                : K == 8 ? dwconv_fwd<8> : dwconv_fwd<0>;
+  hipLaunchKernelGGL(fwd_kern, dim3(grid), dim3(256), 0, stream,
                      (const unsigned short*)x.data_ptr(),
                      (const unsigned short*)w.data_ptr(),
                      bias.has_value() ? (const unsigned short*)
@@ -284,7 +323,10 @@ std::vector<torch::Tensor> dwconv1d_bwd(torch::Tensor dy, torch::Tensor x,
   auto db = torch::zeros({D}, opts);
   auto stream = at::cuda::getCurrentCUDAStream();
   long nvec = (long)B * T * (D / 8);
-  hipLaunchKernelGGL(dwconv_bwd_dx, dim3(memory_bound_grid(nvec, 256)),
+  auto dx_kern = K == 32 ? dwconv_bwd_dx<32>
+               : K == 16 ? dwconv_bwd_dx<16>
+               : K == 8 ? dwconv_bwd_dx<8> : dwconv_bwd_dx<0>;
+  hipLaunchKernelGGL(dx_kern, dim3(memory_bound_grid(nvec, 256)),
                      dim3(256), 0, stream,
                      (const unsigned short*)dy.data_ptr(),
                      (const unsigned short*)w.data_ptr(),
