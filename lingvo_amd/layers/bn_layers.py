@@ -109,14 +109,16 @@ class GroupNormLayer(BaseLayer):
         [p.dim], py_utils.WeightInit.Constant(0.0), p.dtype))
 
   def FProp(self, theta: NestedMap, inputs: torch.Tensor,
-            paddings: Optional[torch.Tensor] = None) -> torch.Tensor:
+            paddings: Optional[torch.Tensor] = None,
+            act: str = 'NONE') -> torch.Tensor:
     p = self.p
     b, t, d = inputs.shape
     g = p.num_groups
-    if inputs.is_cuda and d % g == 0 and d // g <= 64:
+    if (inputs.is_cuda and d % g == 0 and d // g <= 64 and
+        (act == 'NONE' or (d // g) % 8 == 0)):
       from lingvo_amd.ops import group_norm as gn_ops
       return gn_ops.group_norm(inputs, theta.gamma, theta.beta, paddings,
-                               g, p.epsilon)
+                               g, p.epsilon, act=act)
     xf = inputs.float().reshape(b, t, g, d // g)
     # Per (b, group) moments over (t, d/g), excluding padded frames.
     if paddings is not None:
@@ -130,6 +132,8 @@ class GroupNormLayer(BaseLayer):
     out = (xf - mean) * torch.rsqrt(var + p.epsilon)
     out = out.reshape(b, t, d)
     out = out * (1.0 + theta.gamma.float()) + theta.beta.float()
+    if act in ('SILU', 'SWISH'):
+      out = torch.nn.functional.silu(out)
     out = out.to(inputs.dtype)
     if paddings is not None:
       out = py_utils.ApplyPadding(paddings, out)

@@ -80,9 +80,13 @@ class LConvLayer(BaseLayer):
       x = self.norm.FProp(theta.norm, x)
       if paddings is not None:
         x = py_utils.ApplyPadding(paddings, x)
+      x = F.silu(x)
+    elif p.conv_norm == 'group':
+      # swish folds into the GroupNorm kernel (fwd epilogue + bwd chain)
+      x = self.norm.FProp(theta.norm, x, paddings, act='SILU')
     else:
       x = self.norm.FProp(theta.norm, x, paddings)
-    x = F.silu(x)
+      x = F.silu(x)
     x = py_utils.MatmulBias(x, theta.pw2_w, theta.pw2_b)
     if p.dropout_prob and not self.do_eval:
       # padding mask folds into the dropout kernel's elementwise pass.

@@ -68,13 +68,15 @@ std::vector<torch::Tensor> group_norm_fwd(torch::Tensor x,
                                           torch::Tensor gamma,
                                           torch::Tensor beta,
                                           c10::optional<torch::Tensor> pad,
-                                          int64_t groups, double eps);
+                                          int64_t groups, double eps,
+                                          int64_t act);
 std::vector<torch::Tensor> group_norm_bwd(torch::Tensor dy, torch::Tensor x,
                                           torch::Tensor gamma,
+                                          torch::Tensor beta,
                                           c10::optional<torch::Tensor> pad,
                                           torch::Tensor mean,
                                           torch::Tensor rstd,
-                                          int64_t groups);
+                                          int64_t groups, int64_t act);
 
 // lstm_gates.hip
 std::vector<torch::Tensor> lstm_gates_fwd_op(torch::Tensor gates,

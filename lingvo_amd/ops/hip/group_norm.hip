@@ -25,7 +25,8 @@ __global__ __launch_bounds__(GN_BLOCK) void gn_fwd_kernel(
     const unsigned short* __restrict__ beta,
     const unsigned short* __restrict__ paddings,  // [B,T] bf16 or null
     unsigned short* __restrict__ y, float* __restrict__ mean_out,
-    float* __restrict__ rstd_out, int B, int T, int D, int G, float eps) {
+    float* __restrict__ rstd_out, int B, int T, int D, int G, float eps,
+    int act) {
   __shared__ float scratch[GN_WAVES];
   const int b = blockIdx.x / G;
   const int g = blockIdx.x % G;
@@ -65,7 +66,9 @@ __global__ __launch_bounds__(GN_BLOCK) void gn_fwd_kernel(
     float v = bf16_bits_to_float(x[((long)b * T + t) * D + c]);
     float w = 1.f + bf16_bits_to_float(gamma[c]);
     float bb = bf16_bits_to_float(beta[c]);
-    float out = ((v - mu) * rstd * w + bb) * (1.f - pad);
+    float out = (v - mu) * rstd * w + bb;
+    if (act == 1) out = out / (1.f + __expf(-out));  // fused SiLU
+    out *= (1.f - pad);
     y[((long)b * T + t) * D + c] = float_to_bf16_bits(out);
   }
 }
@@ -74,10 +77,11 @@ __global__ __launch_bounds__(GN_BLOCK) void gn_bwd_kernel(
     const unsigned short* __restrict__ dy,
     const unsigned short* __restrict__ x,
     const unsigned short* __restrict__ gamma,
+    const unsigned short* __restrict__ beta,
     const unsigned short* __restrict__ paddings,
     const float* __restrict__ mean, const float* __restrict__ rstd,
     unsigned short* __restrict__ dx, float* __restrict__ dgamma,
-    float* __restrict__ dbeta, int B, int T, int D, int G) {
+    float* __restrict__ dbeta, int B, int T, int D, int G, int act) {
   __shared__ float scratch[GN_WAVES];
   const int b = blockIdx.x / G;
   const int g = blockIdx.x % G;
@@ -147,14 +151,18 @@ __global__ __launch_bounds__(GN_BLOCK) void gn_bwd_kernel(
 // all x/dy traffic is ushortx8, and dgamma/dbeta partials accumulate
 // in REGISTERS (the scalar kernel above pays two LDS atomics per
 // element in its hot loop, which dominates its runtime).
+// act == 1: the forward emitted silu(y_pre); incoming dy is w.r.t.
+// silu's output, so fold silu'(y_pre) in (y_pre recomputed from the
+// saved moments + beta — nothing extra stored).
 __global__ __launch_bounds__(GN_BLOCK) void gn_bwd_vec_kernel(
     const unsigned short* __restrict__ dy,
     const unsigned short* __restrict__ x,
     const unsigned short* __restrict__ gamma,
+    const unsigned short* __restrict__ beta,
     const unsigned short* __restrict__ paddings,
     const float* __restrict__ mean, const float* __restrict__ rstd,
     unsigned short* __restrict__ dx, float* __restrict__ dgamma,
-    float* __restrict__ dbeta, int B, int T, int D, int G) {
+    float* __restrict__ dbeta, int B, int T, int D, int G, int act) {
   __shared__ float scratch[GN_WAVES];
   __shared__ float dg_s[64];
   __shared__ float db_s[64];
@@ -170,11 +178,15 @@ __global__ __launch_bounds__(GN_BLOCK) void gn_bwd_vec_kernel(
   const float rs = rstd[(long)b * G + g];
   const long col0 = (long)g * cg + oct * 8;
 
-  float w[8];
+  float w[8], bb[8];
   {
     ushortx8 gv = *reinterpret_cast<const ushortx8*>(gamma + col0);
+    ushortx8 bv = *reinterpret_cast<const ushortx8*>(beta + col0);
 #pragma unroll
-    for (int e = 0; e < 8; ++e) w[e] = 1.f + bf16_bits_to_float(gv[e]);
+    for (int e = 0; e < 8; ++e) {
+      w[e] = 1.f + bf16_bits_to_float(gv[e]);
+      bb[e] = bf16_bits_to_float(bv[e]);
+    }
   }
   for (int i = tid; i < cg; i += GN_BLOCK) {
     dg_s[i] = 0.f;
@@ -200,6 +212,11 @@ __global__ __launch_bounds__(GN_BLOCK) void gn_bwd_vec_kernel(
       float xf = bf16_bits_to_float(xv[e]);
       float df = bf16_bits_to_float(dyv[e]);
       float xhat = (xf - mu) * rs;
+      if (act == 1) {
+        const float ypre = xhat * w[e] + bb[e];
+        const float sg = 1.f / (1.f + __expf(-ypre));
+        df *= sg * (1.f + ypre * (1.f - sg));
+      }
       float dxhat = df * w[e];
       s1 += dxhat;
       s2 += dxhat * xhat;
@@ -238,6 +255,11 @@ __global__ __launch_bounds__(GN_BLOCK) void gn_bwd_vec_kernel(
       float xf = bf16_bits_to_float(xv[e]);
       float df = bf16_bits_to_float(dyv[e]);
       float xhat = (xf - mu) * rs;
+      if (act == 1) {
+        const float ypre = xhat * w[e] + bb[e];
+        const float sg = 1.f / (1.f + __expf(-ypre));
+        df *= sg * (1.f + ypre * (1.f - sg));
+      }
       float dxhat = df * w[e];
       float val = rs * (dxhat - inv_n * (s1 + xhat * s2)) * (1.f - pad);
       o[e] = float_to_bf16_bits(val);
@@ -252,7 +274,8 @@ std::vector<torch::Tensor> group_norm_fwd(torch::Tensor x,
                                           torch::Tensor gamma,
                                           torch::Tensor beta,
                                           c10::optional<torch::Tensor> pad,
-                                          int64_t groups, double eps) {
+                                          int64_t groups, double eps,
+                                          int64_t act) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 3 &&
               x.scalar_type() == torch::kBFloat16);
   const int B = x.size(0), T = x.size(1), D = x.size(2);
@@ -263,6 +286,8 @@ std::vector<torch::Tensor> group_norm_fwd(torch::Tensor x,
   auto mean = torch::empty({(long)B * G}, opts);
   auto rstd = torch::empty({(long)B * G}, opts);
   auto stream = at::cuda::getCurrentCUDAStream();
+  TORCH_CHECK(act == 0 || (D / G) % 8 == 0,
+              "fused activation needs cg % 8 == 0");
   hipLaunchKernelGGL(gn_fwd_kernel, dim3(B * G), dim3(GN_BLOCK), 0, stream,
                      (const unsigned short*)x.data_ptr(),
                      (const unsigned short*)gamma.data_ptr(),
@@ -271,16 +296,18 @@ std::vector<torch::Tensor> group_norm_fwd(torch::Tensor x,
                          ? (const unsigned short*)pad->data_ptr()
                          : nullptr,
                      (unsigned short*)y.data_ptr(), mean.data_ptr<float>(),
-                     rstd.data_ptr<float>(), B, T, D, G, (float)eps);
+                     rstd.data_ptr<float>(), B, T, D, G, (float)eps,
+                     (int)act);
   return {y, mean, rstd};
 }
 
 std::vector<torch::Tensor> group_norm_bwd(torch::Tensor dy, torch::Tensor x,
                                           torch::Tensor gamma,
+                                          torch::Tensor beta,
                                           c10::optional<torch::Tensor> pad,
                                           torch::Tensor mean,
                                           torch::Tensor rstd,
-                                          int64_t groups) {
+                                          int64_t groups, int64_t act) {
   const int B = x.size(0), T = x.size(1), D = x.size(2);
   const int G = (int)groups;
   auto dx = torch::empty_like(x);
@@ -289,18 +316,20 @@ std::vector<torch::Tensor> group_norm_bwd(torch::Tensor dy, torch::Tensor x,
   auto dbeta = torch::zeros({D}, opts);
   auto stream = at::cuda::getCurrentCUDAStream();
   const int cg = D / G;
-  auto kern = (cg % 8 == 0 && GN_BLOCK % (cg / 8) == 0)
-                  ? gn_bwd_vec_kernel : gn_bwd_kernel;
+  const bool vec = cg % 8 == 0 && GN_BLOCK % (cg / 8) == 0;
+  TORCH_CHECK(act == 0 || vec, "fused activation needs cg % 8 == 0");
+  auto kern = vec ? gn_bwd_vec_kernel : gn_bwd_kernel;
   hipLaunchKernelGGL(kern, dim3(B * G), dim3(GN_BLOCK), 0, stream,
                      (const unsigned short*)dy.data_ptr(),
                      (const unsigned short*)x.data_ptr(),
                      (const unsigned short*)gamma.data_ptr(),
+                     (const unsigned short*)beta.data_ptr(),
                      pad.has_value()
                          ? (const unsigned short*)pad->data_ptr()
                          : nullptr,
                      mean.data_ptr<float>(), rstd.data_ptr<float>(),
                      (unsigned short*)dx.data_ptr(),
                      dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), B,
-                     T, D, G);
+                     T, D, G, (int)act);
   return {dx, dgamma, dbeta};
 }

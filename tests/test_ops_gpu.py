@@ -147,6 +147,44 @@ def test_act_fused_dropout_matches_composed():
 
 
 @gpu
+def test_group_norm_fused_silu_matches_composed():
+  from lingvo_amd.ops import group_norm as gn_ops
+  torch.manual_seed(5)
+  B, T, D, G = 3, 40, 128, 8
+  x = torch.randn(B, T, D, device='cuda',
+                  dtype=torch.bfloat16).requires_grad_(True)
+  gamma = torch.randn(D, device='cuda', dtype=torch.float32) * 0.1
+  beta = torch.randn(D, device='cuda', dtype=torch.float32) * 0.1
+  gamma.requires_grad_(True)
+  beta.requires_grad_(True)
+  pad = torch.zeros(B, T, device='cuda')
+  pad[:, -7:] = 1.0
+  y = gn_ops.group_norm(x, gamma, beta, pad, G, act='SILU')
+  # fp32 composed reference.
+  x32 = x.detach().float().requires_grad_(True)
+  g32 = gamma.detach().clone().requires_grad_(True)
+  b32 = beta.detach().clone().requires_grad_(True)
+  mask = (1.0 - pad)[:, :, None]
+  xm = x32.reshape(B, T, G, D // G)
+  m32 = mask[..., None]
+  cnt = (m32.sum(dim=(1, 3), keepdim=True) * (D // G)).clamp_min(1.0)
+  mu = (xm * m32).sum(dim=(1, 3), keepdim=True) / cnt
+  var = ((xm - mu) ** 2 * m32).sum(dim=(1, 3), keepdim=True) / cnt
+  ref = ((xm - mu) * torch.rsqrt(var + 1e-3)).reshape(B, T, D)
+  ref = ref * (1.0 + g32) + b32
+  ref = torch.nn.functional.silu(ref) * mask
+  assert (y.float() - ref).abs().max() < 0.03
+  g = torch.randn_like(y)
+  y.backward(g)
+  ref.backward(g.float())
+  assert (x.grad.float() - x32.grad).abs().max() < 0.05
+  assert (gamma.grad - g32.grad).abs().max() / \
+      g32.grad.abs().max().clamp_min(1.0) < 0.05
+  assert (beta.grad - b32.grad).abs().max() / \
+      b32.grad.abs().max().clamp_min(1.0) < 0.05
+
+
+@gpu
 def test_group_norm_matches_ref():
   from lingvo_amd.ops import group_norm as gn_ops
   from lingvo_amd.core import py_utils as pu
