@@ -321,6 +321,72 @@ def ConcatenatePaddedSequences(x: torch.Tensor, y: torch.Tensor,
 # --------------------------------------------------------------------------
 # Numeric checks & misc
 # --------------------------------------------------------------------------
+def TrimTrailingPaddings(x: torch.Tensor, paddings: torch.Tensor):
+  """Trims time steps that are padded in EVERY batch row
+  (reference py_utils.py TrimTrailingPaddings). x [B, T, ...],
+  paddings [B, T]; returns (trimmed_x, trimmed_paddings)."""
+  lengths = LengthsFromPaddings(paddings)
+  max_len = int(lengths.max().item()) if lengths.numel() else 0
+  max_len = max(max_len, 1)
+  return x[:, :max_len], paddings[:, :max_len]
+
+
+def ReversePaddedSequence(x: torch.Tensor,
+                          paddings: torch.Tensor) -> torch.Tensor:
+  """Reverses the VALID prefix of each row, keeping padding in place
+  (reference py_utils.py ReversePaddedSequence). x [B, T, ...]."""
+  b, t = paddings.shape
+  lengths = LengthsFromPaddings(paddings).to(torch.long)  # [B]
+  pos = torch.arange(t, device=x.device).unsqueeze(0).expand(b, t)
+  rev = (lengths.unsqueeze(1) - 1 - pos).clamp_min(0)
+  idx = torch.where(pos < lengths.unsqueeze(1), rev, pos)
+  shaped = idx.reshape(b, t, *([1] * (x.dim() - 2))).expand_as(x)
+  return torch.gather(x, 1, shaped)
+
+
+def ShiftLeft(x: torch.Tensor, shift: int,
+              pad_val: float = 0.0) -> torch.Tensor:
+  """Shifts [B, T, ...] left along time, padding the tail."""
+  if shift <= 0:
+    return x
+  pad = x.new_full((x.shape[0], shift, *x.shape[2:]), pad_val)
+  return torch.cat([x[:, shift:], pad], dim=1)
+
+
+def MixByWeight(fns, weights, seed: Optional[int] = None):
+  """Calls one of fns sampled by normalized weights; returns
+  (result, index) (reference py_utils.py MixByWeight)."""
+  import random as _random
+  rng = _random.Random(seed)
+  total = float(sum(weights))
+  r = rng.uniform(0.0, total)
+  acc = 0.0
+  for i, (fn, w) in enumerate(zip(fns, weights)):
+    acc += float(w)
+    if r <= acc:
+      return fn(), i
+  return fns[-1](), len(fns) - 1
+
+
+def SplitRecursively(x, num_splits: int, axis: int = -1):
+  """Splits tensors (or NestedMaps/lists of them) into num_splits
+  equal parts along axis (reference py_utils.py SplitRecursively).
+  Returns a list of num_splits structures mirroring x."""
+  if isinstance(x, torch.Tensor):
+    assert x.shape[axis] % num_splits == 0
+    return list(x.chunk(num_splits, dim=axis))
+  if isinstance(x, (list, tuple)):
+    split_elems = [SplitRecursively(e, num_splits, axis) for e in x]
+    return [type(x)(parts[i] for parts in split_elems)
+            for i in range(num_splits)]
+  if isinstance(x, NestedMap):
+    flat = x.Flatten()
+    split_flat = [SplitRecursively(e, num_splits, axis) for e in flat]
+    return [x.Pack([parts[i] for parts in split_flat])
+            for i in range(num_splits)]
+  raise TypeError('Unsupported type for SplitRecursively: %r' % type(x))
+
+
 def CheckNumerics(x: torch.Tensor, message: str = '') -> torch.Tensor:
   """Raises if x contains NaN/Inf (reference py_utils.py:208)."""
   if not torch.isfinite(x).all():

@@ -759,3 +759,24 @@ def test_branch_layer_fetches_graph_tensors():
   x = torch.randn(2, 2)
   z, h = br.FProp(br.theta, x)
   assert torch.allclose(h, x + 1) and torch.allclose(z, (x + 1) * 3)
+
+
+def test_py_utils_sequence_helpers():
+  import torch
+  from lingvo_amd.core import py_utils as pu
+  from lingvo_amd.core.nested_map import NestedMap
+  x = torch.arange(24.).reshape(2, 4, 3)
+  pad = torch.tensor([[0., 0., 1., 1.], [0., 0., 0., 1.]])
+  tx, tp = pu.TrimTrailingPaddings(x, pad)
+  assert tx.shape == (2, 3, 3) and tp.shape == (2, 3)
+  r = pu.ReversePaddedSequence(x, pad)
+  assert torch.equal(r[0, 0], x[0, 1]) and torch.equal(r[0, 1], x[0, 0])
+  assert torch.equal(r[0, 2], x[0, 2])  # padding untouched
+  assert torch.equal(r[1, 0], x[1, 2])
+  s = pu.ShiftLeft(x, 2, -1.0)
+  assert torch.equal(s[:, :2], x[:, 2:]) and (s[:, 2:] == -1).all()
+  out, i = pu.MixByWeight([lambda: 'a', lambda: 'b'], [0.0, 5.0], seed=0)
+  assert out == 'b' and i == 1
+  parts = pu.SplitRecursively(NestedMap(a=x, b=[x]), 3)
+  assert len(parts) == 3 and parts[2].a.shape == (2, 4, 1)
+  assert torch.equal(torch.cat([q.a for q in parts], dim=-1), x)
