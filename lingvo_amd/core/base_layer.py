@@ -148,6 +148,22 @@ class BaseLayer(nn.Module):
     self.add_module(name, child)
     self._children_names.append(name)
 
+  def AddChild(self, name: str, child: 'BaseLayer',
+               replace: bool = False) -> None:
+    """Registers an ALREADY-INSTANTIATED layer as a child (module
+    sharing across tasks — reference multitask_model.py:42 AddChild).
+    Gradients accumulate across all users; torch's parameter iterators
+    dedupe the shared weights. replace=True swaps out an existing
+    child of the same name (its own weights are discarded)."""
+    if name in self._children_names:
+      if not replace:
+        raise ValueError(f'Child {name!r} already exists on '
+                         f'{self.layer_name}')
+      del self._modules[name]
+      self._children_names.remove(name)
+    self.add_module(name, child)
+    self._children_names.append(name)
+
   def CreateChildren(self, name: str,
                      children_params: Sequence[InstantiableParams]) -> None:
     """Instantiates a list of sub-layers as an nn.ModuleList."""

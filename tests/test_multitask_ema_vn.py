@@ -157,3 +157,37 @@ def test_ema_as_weights_context():
   with ema.AsWeights(lin):
     assert torch.allclose(lin.weight.detach(), live)
   assert torch.allclose(lin.weight.detach(), live + 1.0)  # restored
+
+
+def test_shared_encoder_model():
+  from lingvo_amd.core.multitask_model import SharedEncoderModel
+  from lingvo_amd.models import mt as mt_model
+
+  def _mt_task(seed):
+    p = mt_model.TransformerModel.Params().Set(name=f'mt{seed}',
+                                               random_seed=seed)
+    p.encoder.Set(model_dim=16, num_layers=1, num_heads=2, vocab_size=32,
+                  hidden_dim=32)
+    p.decoder.Set(model_dim=16, num_layers=1, num_heads=2, vocab_size=32,
+                  hidden_dim=32)
+    return p
+
+  mp_ = SharedEncoderModel.Params().Set(name='m',
+                                        encoder_to_share='a')
+  mp_.task_params = Params()
+  mp_.task_params.Define('a', _mt_task(1), '')
+  mp_.task_params.Define('b', _mt_task(2), '')
+  mp_.task_probs = Params()
+  mp_.task_probs.Define('a', 0.5, '')
+  mp_.task_probs.Define('b', 0.5, '')
+  model = mp_.Instantiate()
+  ta, tb = model.GetTask('a'), model.GetTask('b')
+  assert ta.encoder is tb.encoder
+  # One parameter set for the shared encoder across the whole model.
+  enc_params = {id(q) for q in ta.encoder.parameters()}
+  shared = [n for n, q in model.named_parameters() if id(q) in enc_params]
+  assert shared  # present exactly once in the deduped iterator
+  seen = set()
+  for n, q in model.named_parameters():
+    assert id(q) not in seen or id(q) not in enc_params
+    seen.add(id(q))
