@@ -780,3 +780,36 @@ def test_py_utils_sequence_helpers():
   parts = pu.SplitRecursively(NestedMap(a=x, b=[x]), 3)
   assert len(parts) == 3 and parts[2].a.shape == (2, 4, 1)
   assert torch.equal(torch.cat([q.a for q in parts], dim=-1), x)
+
+
+def test_batch_utils_scaling():
+  from lingvo_amd.core import batch_utils
+  assert batch_utils.scale_infeed_to_global(16) == 16  # world size 1
+  assert batch_utils.scale_global_to_infeed(16) == 16
+  assert batch_utils.scale_split_to_infeed(8) == 8
+  assert batch_utils.scale_global_to_worker(32) == 32
+
+
+def test_gradient_combiners():
+  import torch
+  from lingvo_amd.core import gradient_combiner as gc
+  from lingvo_amd.core.nested_map import NestedMap
+  vmap = NestedMap(w=torch.zeros(3))
+  g1 = NestedMap(loss_metric=(None, 1.0),
+                 grads=NestedMap(w=torch.tensor([1.0, 0.0, 0.0])))
+  g2 = NestedMap(loss_metric=(None, 2.0),
+                 grads=NestedMap(w=torch.tensor([0.0, 1.0, 0.0])))
+  comb = gc.SumGradientCombiner.Params().Set(name='s').Instantiate()
+  out = comb.Combine(vmap, {'a': g1, 'b': g2})
+  assert torch.allclose(out.w, torch.tensor([1.0, 2.0, 0.0]))
+  # PCGrad: orthogonal grads pass through unchanged.
+  pc = gc.PCGradCombiner.Params().Set(name='p').Instantiate()
+  out2 = pc.Combine(vmap, {'a': g1, 'b': g2})
+  assert torch.allclose(out2.w, torch.tensor([1.0, 2.0, 0.0]))
+  # Conflicting grads: the conflicting component is projected out, so
+  # the combined grad has no negative dot with either input.
+  g3 = NestedMap(loss_metric=(None, 1.0),
+                 grads=NestedMap(w=torch.tensor([-1.0, 0.5, 0.0])))
+  out3 = pc.Combine(vmap, {'a': g1, 'b': g3})
+  assert float(out3.w @ g1.grads.w) >= -1e-6
+  assert float(out3.w @ g3.grads.w) >= -1e-6
