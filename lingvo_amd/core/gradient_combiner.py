@@ -47,23 +47,27 @@ class PCGradCombiner(GradientCombiner):
 
   def Combine(self, vmap: NestedMap, losses_and_gradients) -> NestedMap:
     tasks = list(losses_and_gradients.values())
-    flats = []
+    orig = []
     for entry in tasks:
       _, weight = entry.loss_metric
-      flats.append([None if g is None else (g * weight).clone()
-                    for g in entry.grads.Flatten()])
+      orig.append([None if g is None else g * weight
+                   for g in entry.grads.Flatten()])
+    # Projections use the ORIGINAL other-task gradients (PCGrad alg. 1):
+    # after the pass, each projected g_i has non-negative dot with every
+    # original g_j it conflicted with.
+    flats = [[None if g is None else g.clone() for g in f] for f in orig]
     n = len(flats)
     for i in range(n):
       for j in range(n):
         if i == j:
           continue
-        dot = sum((a * b).sum() for a, b in zip(flats[i], flats[j])
+        dot = sum((a * b).sum() for a, b in zip(flats[i], orig[j])
                   if a is not None and b is not None)
         if float(dot) >= 0.0:
           continue
-        sq = sum((b * b).sum() for b in flats[j] if b is not None)
+        sq = sum((b * b).sum() for b in orig[j] if b is not None)
         coef = dot / sq.clamp_min(1e-12)
-        for k, b in enumerate(flats[j]):
+        for k, b in enumerate(orig[j]):
           if flats[i][k] is not None and b is not None:
             flats[i][k] = flats[i][k] - coef * b
     out = flats[0]

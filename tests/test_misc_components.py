@@ -806,10 +806,14 @@ def test_gradient_combiners():
   pc = gc.PCGradCombiner.Params().Set(name='p').Instantiate()
   out2 = pc.Combine(vmap, {'a': g1, 'b': g2})
   assert torch.allclose(out2.w, torch.tensor([1.0, 2.0, 0.0]))
-  # Conflicting grads: the conflicting component is projected out, so
-  # the combined grad has no negative dot with either input.
+  # Conflicting grads: g1 projected onto g3's normal plane + g3
+  # projected onto g1's normal plane (PCGrad): sum is the sum of the
+  # two projections, computed against the ORIGINAL gradients.
   g3 = NestedMap(loss_metric=(None, 1.0),
                  grads=NestedMap(w=torch.tensor([-1.0, 0.5, 0.0])))
   out3 = pc.Combine(vmap, {'a': g1, 'b': g3})
-  assert float(out3.w @ g1.grads.w) >= -1e-6
-  assert float(out3.w @ g3.grads.w) >= -1e-6
+  g1v, g3v = g1.grads.w, g3.grads.w
+  p1 = g1v - (g1v @ g3v) / (g3v @ g3v) * g3v
+  p3 = g3v - (g3v @ g1v) / (g1v @ g1v) * g1v
+  assert torch.allclose(out3.w, p1 + p3, atol=1e-6)
+  assert float(p1 @ g3v) > -1e-5 and float(p3 @ g1v) > -1e-5
