@@ -708,3 +708,39 @@ def test_depthwise_conv1d_dilation():
   want4 = (x[:, 4] * layer.theta.w[2] + x[:, 2] * layer.theta.w[1] +
            x[:, 0] * layer.theta.w[0] + layer.theta.b)
   assert torch.allclose(out[:, 4], want4, atol=1e-5)
+
+
+def test_conv_builder_stacks():
+  import torch
+  from lingvo_amd.layers import conv_layers_builder as cb
+  torch.manual_seed(0)
+  b = cb.Builder(norm='batch', activation='RELU')
+  stack = b._Seq('s',
+                 b.Conv2D('c1', (3, 3, 2, 4), (2, 2)),
+                 b.SeparableConv2D('c2', (3, 3, 4, 8), depth_multiplier=2),
+                 b.GlobalPooling('gp'))
+  layer = stack.Instantiate()
+  x = torch.randn(2, 8, 6, 2)
+  pad = torch.zeros(2, 8)
+  pad[1, 6:] = 1.0
+  out, _ = layer.FProp(layer.theta, x, pad)
+  assert out.shape == (2, 3, 8)  # freq 6->3 after stride 2, 8 channels
+  out.sum().backward()
+
+
+def test_causal_pooling_layer():
+  import torch
+  from lingvo_amd.layers import conv_layers_builder as cb
+  x = torch.arange(8.).reshape(1, 4, 1, 2)
+  x = x.expand(1, 4, 1, 2).clone()
+  pad = torch.zeros(1, 4)
+  p = cb.CausalPoolingLayer.Params().Set(name='cp', pooling_type='AVG',
+                                         left_context=2).Instantiate()
+  out, _ = p.FProp(p.theta, x, pad)
+  # t=0 averages itself; t>=1 averages the last two frames.
+  assert torch.allclose(out[0, 0], x[0, 0])
+  assert torch.allclose(out[0, 2], (x[0, 1] + x[0, 2]) / 2)
+  m = cb.CausalPoolingLayer.Params().Set(name='cm', pooling_type='MAX',
+                                         left_context=-1).Instantiate()
+  om, _ = m.FProp(m.theta, x, pad)
+  assert torch.allclose(om[0, 3], x[0].max(dim=0).values)
