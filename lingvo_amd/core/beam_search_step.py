@@ -157,14 +157,35 @@ def BeamSearchStep(scores: torch.Tensor, state: BeamSearchState, t: int,
 
   # Device-side per-hyp top-(k+2): covers k survivors + eos (+eoc).
   topk_size = min(k + 2, scores.shape[1])
-  scores_f = scores.float().cpu()
-  total = cum.unsqueeze(1) + scores_f                     # [N, V]
-  top_vals, top_idx = total.topk(topk_size, dim=-1)
-  eos_local = scores_f[:, eos_id]
-  eos_global = total[:, eos_id]
-  if eoc_id >= 0:
-    eoc_local = scores_f[:, eoc_id]
-    eoc_global = total[:, eoc_id]
+  if scores.is_cuda and topk_size <= 32:
+    # HIP top-k kernel (K12): prune [N, V] on the GPU, ship only
+    # [N, k+2] to the host hyp bookkeeping below.
+    from lingvo_amd.ops import _loader
+    ext = _loader.get_ext(required=True)
+    total_dev = cum.to(scores.device).float().unsqueeze(1) + \
+        scores.float()
+    tv, ti = ext.topk_rows(total_dev.contiguous(), topk_size)
+    eos_cols = total_dev[:, eos_id]
+    scr_eos = scores[:, eos_id].float()
+    if eoc_id >= 0:
+      eoc_cols = total_dev[:, eoc_id]
+      scr_eoc = scores[:, eoc_id].float()
+    top_vals, top_idx = tv.cpu(), ti.cpu().long()
+    eos_global = eos_cols.cpu()
+    eos_local = scr_eos.cpu()
+    if eoc_id >= 0:
+      eoc_global = eoc_cols.cpu()
+      eoc_local = scr_eoc.cpu()
+    scores_f = None
+  else:
+    scores_f = scores.float().cpu()
+    total = cum.unsqueeze(1) + scores_f                   # [N, V]
+    top_vals, top_idx = total.topk(topk_size, dim=-1)
+    eos_local = scores_f[:, eos_id]
+    eos_global = total[:, eos_id]
+    if eoc_id >= 0:
+      eoc_local = scores_f[:, eoc_id]
+      eoc_global = total[:, eoc_id]
 
   # Host-side merge per beam (reference merged_topk_vec).
   merged: List[List[Hyp]] = [[] for _ in range(b)]

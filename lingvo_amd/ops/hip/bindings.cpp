@@ -112,6 +112,9 @@ torch::Tensor s2d_fwd(torch::Tensor x);
 torch::Tensor s2d_inv(torch::Tensor dX, int64_t C);
 torch::Tensor pad_scatter(torch::Tensor dout);
 
+// topk.hip
+std::vector<torch::Tensor> topk_rows(torch::Tensor scores, int64_t k);
+
 // las_decoder.hip
 void smallm_gemm(torch::Tensor a, torch::Tensor wt,
                  c10::optional<torch::Tensor> pre, torch::Tensor out,
@@ -154,6 +157,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attend_fwd", &attend_fwd, "Fused dot-attention step fwd");
   m.def("attend_bwd", &attend_bwd, "Fused dot-attention step bwd");
   m.def("s2d_fwd", &s2d_fwd, "Space-to-depth + pad (conv frontend)");
+  m.def("topk_rows", &topk_rows, "Row-wise top-k (beam pruning, K12)");
   m.def("s2d_inv", &s2d_inv, "Inverse space-to-depth");
   m.def("pad_scatter", &pad_scatter, "Zero-border pad scatter");
 }

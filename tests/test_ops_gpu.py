@@ -147,6 +147,50 @@ def test_act_fused_dropout_matches_composed():
 
 
 @gpu
+def test_topk_rows_matches_torch():
+  from lingvo_amd.ops import _loader
+  ext = _loader.get_ext(required=True)
+  torch.manual_seed(2)
+  for dtype in (torch.float32, torch.bfloat16):
+    for r, v, k in [(8, 1000, 10), (64, 32000, 18), (3, 40, 32)]:
+      x = torch.randn(r, v, device='cuda').to(dtype).contiguous()
+      tv, ti = ext.topk_rows(x, k)
+      want_v, want_i = x.float().topk(k, dim=-1)
+      assert torch.equal(tv, want_v), (dtype, r, v, k)
+      # Indices must point at the returned values (ties may reorder).
+      got = x.float().gather(1, ti.long())
+      assert torch.equal(got, tv)
+      # No duplicate indices within a row.
+      for rr in range(r):
+        assert len(set(ti[rr].tolist())) == k
+
+
+@gpu
+def test_beam_search_step_gpu_topk_matches_cpu():
+  """The beam step's GPU top-k path must agree with the CPU path."""
+  from lingvo_amd.core import beam_search_step as bss
+  torch.manual_seed(3)
+  b, k, vocab = 2, 4, 64
+  n = b * k
+
+  def run(device):
+    state = bss.BeamSearchState.Init(num_beams=b, k=k, max_steps=8)
+    torch.manual_seed(7)
+    outs = []
+    for t in range(3):
+      scores = torch.log_softmax(
+          torch.randn(n, vocab, dtype=torch.float32), dim=-1)
+      bss.BeamSearchStep(scores.to(device), state, t, eos_id=2)
+      outs.append(state.cumulative_scores.clone())
+    return outs
+
+  cpu_out = run('cpu')
+  gpu_out = run('cuda')
+  for a, c in zip(cpu_out, gpu_out):
+    assert torch.allclose(a, c, atol=1e-5)
+
+
+@gpu
 def test_group_norm_fused_silu_matches_composed():
   from lingvo_amd.ops import group_norm as gn_ops
   torch.manual_seed(5)
