@@ -146,6 +146,16 @@ class BaseRunner:
     os.makedirs(self._train_dir, exist_ok=True)
     self._model = None
     self._status_path = os.path.join(logdir, f'{job_name}_status.txt')
+    # Pluggable metric export (reference base_runner.py:174
+    # _ExportMetrics): callable(step=int, **metrics) or None.
+    self._export_metrics_fn = None
+
+  def SetExportMetricsFn(self, fn) -> None:
+    self._export_metrics_fn = fn
+
+  def _ExportMetrics(self, step: int, **metrics) -> None:
+    if self._export_metrics_fn is not None:
+      self._export_metrics_fn(step=step, **metrics)
 
   @property
   def model(self):
@@ -195,9 +205,14 @@ class Trainer(BaseRunner):
 
   def __init__(self, model_params, logdir: str, max_steps: Optional[int]
                = None, grad_sync=None,
-               watchdog_timeout: Optional[float] = None, **kwargs):
+               watchdog_timeout: Optional[float] = None,
+               detect_anomaly: bool = False, **kwargs):
     super().__init__(model_params, logdir, 'trainer', **kwargs)
     self._max_steps = max_steps
+    if detect_anomaly:
+      # autograd anomaly mode (SURVEY §5 sanitizer mapping): NaN
+      # sources raise with the producing op's forward stack.
+      torch.autograd.set_detect_anomaly(True)
     self._watchdog = (Watchdog(watchdog_timeout, 'trainer')
                       if watchdog_timeout else None)
     if grad_sync is None:
@@ -282,6 +297,10 @@ class Trainer(BaseRunner):
              'steps_per_sec': self._tracker.steps_per_sec,
              'examples_per_sec': self._tracker.examples_per_sec},
             step)
+        self._ExportMetrics(
+            int(step), loss=loss,
+            steps_per_sec=self._tracker.steps_per_sec,
+            examples_per_sec=self._tracker.examples_per_sec)
       ckpt.MaybeSave()
     ckpt.Save()
     ckpt.Sync()

@@ -47,8 +47,10 @@ class RunnerManager:
     args = self.args
     device = args.device
     if job == 'trainer':
-      return runners.Trainer(self.GetParamsForDataset('Train'), args.logdir,
-                             max_steps=args.max_steps, device=device)
+      return runners.Trainer(
+          self.GetParamsForDataset('Train'), args.logdir,
+          max_steps=args.max_steps, device=device,
+          detect_anomaly=getattr(args, 'detect_anomaly', False))
     if job == 'controller':
       return runners.Controller(self.GetParamsForDataset('Train'),
                                 args.logdir, device=device)
@@ -103,6 +105,8 @@ def MakeParser() -> argparse.ArgumentParser:
   ap.add_argument('--model_params_override', action='append', default=[],
                   help='dotted.path=value (repeatable)')
   ap.add_argument('--list_models', action='store_true')
+  ap.add_argument('--detect_anomaly', action='store_true',
+                  help='torch.autograd anomaly mode (NaN provenance).')
   return ap
 
 

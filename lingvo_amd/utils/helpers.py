@@ -15,29 +15,14 @@ import torch
 from lingvo_amd.core.nested_map import NestedMap
 
 
-def ComputeSplits(batch_size: int, num_splits: int) -> List[int]:
-  """Evenly splits batch_size into num_splits parts
-  (reference input_generator_helper.py:21)."""
-  assert num_splits >= 1
-  base = batch_size // num_splits
-  rem = batch_size % num_splits
-  return [base + (1 if i < rem else 0) for i in range(num_splits)]
-
-
-def SplitTensors(tensors: Sequence[torch.Tensor],
-                 num_splits: int) -> List[List[torch.Tensor]]:
-  """Splits each tensor along dim 0 into num_splits pieces; returns one
-  list per split (reference input_generator_helper.py:45)."""
-  b = tensors[0].shape[0]
-  sizes = ComputeSplits(b, num_splits)
-  per_tensor = [list(torch.split(t, sizes)) for t in tensors]
-  return [[pt[i] for pt in per_tensor] for i in range(num_splits)]
-
-
-def SplitNestedMap(batch: NestedMap, num_splits: int) -> List[NestedMap]:
-  flat = batch.Flatten()
-  splits = SplitTensors([t for t in flat], num_splits)
-  return [batch.Pack(s) for s in splits]
+# Canonical implementations live in core (richer APIs there):
+# core/input_generator_helper.py (ComputeSplits/SplitTensors/
+# SplitDictOfTensors/SplitNestedMap), core/batch_utils.py
+# (world-size-aware scaling), core/gradient_combiner.py (layer-style
+# Sum + PCGrad combiners on grads). These wrappers keep the original
+# helper names working.
+from lingvo_amd.core.input_generator_helper import (  # noqa: F401
+    ComputeSplits, SplitNestedMap, SplitTensors)
 
 
 def ScaleInfeedToGlobal(infeed_batch_size: int,
@@ -52,8 +37,8 @@ def ScaleGlobalToInfeed(global_batch_size: int, num_replicas: int) -> int:
 
 
 class GradientCombiner:
-  """Multi-loss gradient combination interface
-  (reference gradient_combiner.py:44). Default: weighted sum."""
+  """Loss-level weighted-sum combiner. For gradient-level combination
+  (sum / PCGrad on per-loss grads) use core/gradient_combiner.py."""
 
   def __init__(self, weights: Optional[Sequence[float]] = None):
     self._weights = weights

@@ -505,3 +505,19 @@ def test_runloop_oor_is_clean_finish(tmp_path):
 
   tr._RunLoop(loop)      # returns cleanly, no retry storm
   assert len(calls) == 1
+
+
+def test_export_metrics_hook(tmp_path):
+  from lingvo_amd.runtime import runners
+  from lingvo_amd.models import mnist as mnist_model
+  p = mnist_model.ModelV1.Params().Set(
+      name='m', hidden_dim=8, filter_shapes=[(3, 3, 1, 2)])
+  p.softmax.num_classes = 10
+  p.input = mnist_model.FakeMnistData.Params().Set(batch_size=4)
+  from lingvo_amd.core.base_model import SingleTaskModel
+  mp_ = SingleTaskModel.Params(p)
+  tr = runners.Trainer(mp_, str(tmp_path), max_steps=2)
+  exported = []
+  tr.SetExportMetricsFn(lambda **kw: exported.append(kw))
+  tr.Start()
+  assert exported and 'loss' in exported[0] and 'step' in exported[0]
