@@ -79,6 +79,89 @@ __global__ void dwconv_fwd(const unsigned short* __restrict__ x,
   }
 }
 
+// Register-window fwd variant: thread computes R consecutive t rows of
+// one 8-channel vec. Taps live in REGISTERS (KT compile-time), the x
+// window shifts by one vec per unrolled tap step: (KT + R - 1) x-loads
+// per R outputs instead of KT per output (the plain kernel is
+// L1-bandwidth-bound re-reading x KT times). High VGPR use is the
+// deliberate trade (W: KT*4 regs).
+template <int KT, int R>
+__global__ __launch_bounds__(256) void dwconv_fwd_reg(
+    const unsigned short* __restrict__ x,
+    const unsigned short* __restrict__ w,
+    const unsigned short* __restrict__ bias,
+    unsigned short* __restrict__ y, int B, int T, int D, int pad) {
+  const int dvec = D / 8;
+  const int tblks = (T + R - 1) / R;
+  const long nwork = (long)B * tblks * dvec;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nwork;
+       i += (long)gridDim.x * blockDim.x) {
+    const int dv = (int)(i % dvec);
+    const long bt = i / dvec;
+    const int t0 = (int)(bt % tblks) * R;
+    const long b = bt / tblks;
+    // Tap registers for this channel octet.
+    ushortx8 wv[KT];
+#pragma unroll
+    for (int j = 0; j < KT; ++j) {
+      wv[j] = *reinterpret_cast<const ushortx8*>(w + j * D + dv * 8);
+    }
+    float acc[R][8];
+#pragma unroll
+    for (int r = 0; r < R; ++r) {
+      if (bias) {
+        ushortx8 bv = *reinterpret_cast<const ushortx8*>(bias + dv * 8);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) acc[r][e] = bf16_bits_to_float(bv[e]);
+      } else {
+#pragma unroll
+        for (int e = 0; e < 8; ++e) acc[r][e] = 0.f;
+      }
+    }
+    const long rowbase = b * T;
+    auto loadx = [&](int t) -> ushortx8 {
+      ushortx8 v;
+      if (t >= 0 && t < T) {
+        v = *reinterpret_cast<const ushortx8*>(
+            x + (rowbase + t) * D + dv * 8);
+      } else {
+#pragma unroll
+        for (int e = 0; e < 8; ++e) v[e] = 0;
+      }
+      return v;
+    };
+    // Window W[r] = x[t0 + r + j - pad] for the current tap j.
+    ushortx8 W[R];
+#pragma unroll
+    for (int r = 0; r < R; ++r) W[r] = loadx(t0 + r - pad);
+#pragma unroll
+    for (int j = 0; j < KT; ++j) {
+#pragma unroll
+      for (int r = 0; r < R; ++r) {
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          acc[r][e] += bf16_bits_to_float(W[r][e]) *
+                       bf16_bits_to_float(wv[j][e]);
+        }
+      }
+      if (j + 1 < KT) {
+#pragma unroll
+        for (int r = 0; r + 1 < R; ++r) W[r] = W[r + 1];
+        W[R - 1] = loadx(t0 + (R - 1) + (j + 1) - pad);
+      }
+    }
+#pragma unroll
+    for (int r = 0; r < R; ++r) {
+      const int t = t0 + r;
+      if (t >= T) continue;
+      ushortx8 ov;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) ov[e] = float_to_bf16_bits(acc[r][e]);
+      *reinterpret_cast<ushortx8*>(y + (rowbase + t) * D + dv * 8) = ov;
+    }
+  }
+}
+
 template <int KT>
 __global__ void dwconv_bwd_dx(const unsigned short* __restrict__ dy,
                               const unsigned short* __restrict__ w,
@@ -302,7 +385,25 @@ torch::Tensor dwconv1d_fwd(torch::Tensor x, torch::Tensor w,
   long nvec = (long)B * T * (D / 8);
   int grid = memory_bound_grid(nvec, 256);
   // Measured: full K unroll (KT=32) raises register pressure and costs
-  // ~50% (8.2 -> 12.5 ms/step at the bench shape). Runtime-K path wins.
+  // ~50% (8.2 -> 12.5 ms/step at the bench shape). Runtime-K path wins
+  // for the naive form; LINGVO_DWCONV_REG=1 selects the register-window
+  // variant (A/B knob).
+  static const bool use_reg = []() {
+    const char* e = getenv("LINGVO_DWCONV_REG");
+    return e && e[0] == '1';
+  }();
+  if (use_reg && K == 32) {
+    long nwork = (long)B * ((T + 3) / 4) * (D / 8);
+    hipLaunchKernelGGL((dwconv_fwd_reg<32, 4>),
+                       dim3(memory_bound_grid(nwork, 256)), dim3(256), 0,
+                       stream, (const unsigned short*)x.data_ptr(),
+                       (const unsigned short*)w.data_ptr(),
+                       bias.has_value() ? (const unsigned short*)
+                                              bias->data_ptr()
+                                        : nullptr,
+                       (unsigned short*)y.data_ptr(), B, T, D, (int)pad);
+    return y;
+  }
   auto fwd_kern = dwconv_fwd<0>;
   hipLaunchKernelGGL(fwd_kern, dim3(grid), dim3(256), 0, stream,
                      (const unsigned short*)x.data_ptr(),
