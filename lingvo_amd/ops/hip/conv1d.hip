@@ -79,12 +79,13 @@ __global__ void dwconv_fwd(const unsigned short* __restrict__ x,
   }
 }
 
-// Register-window fwd variant: thread computes R consecutive t rows of
-// one 8-channel vec. Taps live in REGISTERS (KT compile-time), the x
-// window shifts by one vec per unrolled tap step: (KT + R - 1) x-loads
-// per R outputs instead of KT per output (the plain kernel is
-// L1-bandwidth-bound re-reading x KT times). High VGPR use is the
-// deliberate trade (W: KT*4 regs).
+// Register-window fwd variant (A/B, default OFF — measured SLOWER:
+// 0.77 vs 0.44 ms at the bench shape): thread computes R consecutive
+// t rows of one 8-channel vec with taps in registers and a shifting x
+// window, cutting x re-reads from KT to (KT+R-1)/R per output. The
+// loss shows the plain kernel is NOT L1-bound; the ~128-VGPR tap
+// array costs more occupancy than the traffic saving returns. Kept
+// behind LINGVO_DWCONV_REG=1 as measured methodology.
 template <int KT, int R>
 __global__ __launch_bounds__(256) void dwconv_fwd_reg(
     const unsigned short* __restrict__ x,
