@@ -182,3 +182,82 @@ class GradDropCompressor:
     kept = torch.where(mask, flat, torch.zeros_like(flat))
     res.copy_((flat - kept).reshape_as(res))
     return kept.reshape_as(grad)
+
+
+class EGDD(optimizer_lib.Base):
+  """Exponentiated Gradient Delta-Delta (reference egdd.py:49; Kivinen
+  & Warmuth 1997): momentum SGD with per-weight multiplicative gains
+  and a per-parameter learning-rate scale, both updated by
+  unnormalized exponentiated gradient."""
+
+  @classmethod
+  def Params(cls):
+    p = super().Params()
+    p.Define('momentum', 0.9, 'Heavy-ball momentum.')
+    p.Define('beta', 0.9, 'Gradient-EMA decay for the gain update.')
+    p.Define('gain_learning_rate', 0.01, 'EG rate for per-weight gain.')
+    p.Define('scale_learning_rate', 0.001, 'EG rate for lr scale.')
+    p.Define('initial_gain', 1.0, 'Initial per-weight gain.')
+    p.Define('min_gain', 1e-2, 'Gain floor.')
+    p.Define('max_gain', 1e2, 'Gain cap.')
+    p.Define('initial_scale', 1.0, 'Initial lr scale.')
+    p.Define('min_scale', 1e-1, 'Scale floor.')
+    p.Define('max_scale', 1e1, 'Scale cap.')
+    p.Define('use_directions', True,
+             'Scale update uses normalized grad/momentum directions.')
+    p.Define('use_signs', True,
+             'Gain update uses sign(grad)*sign(gbar).')
+    return p
+
+  def CreateTorchOptimizer(self, params, lr):
+    return _EgddImpl(params, self.p, lr)
+
+
+class _EgddImpl(torch.optim.Optimizer):
+
+  def __init__(self, params, p, lr):
+    super().__init__(params, dict(lr=lr))
+    self._p = p
+
+  @torch.no_grad()
+  def step(self, closure=None):
+    loss = closure() if closure is not None else None
+    p_ = self._p
+    for group in self.param_groups:
+      lr = group['lr']
+      for w in group['params']:
+        if w.grad is None:
+          continue
+        g = w.grad.float()
+        st = self.state[w]
+        if not st:
+          st['momentum'] = torch.zeros_like(g)
+          st['gbar'] = torch.zeros_like(g)
+          st['gain'] = torch.full_like(g, p_.initial_gain)
+          st['lr_scale'] = torch.tensor(float(p_.initial_scale),
+                                        device=g.device)
+          st['counter'] = 0
+        m, gbar, gain = st['momentum'], st['gbar'], st['gain']
+        st['counter'] += 1
+        # lr-scale EG update from grad/momentum alignment.
+        if p_.use_directions:
+          ng = g / (g.norm() + 1e-10)
+          nm = m / (m.norm() + 1e-10)
+          align = (ng * nm).sum()
+        else:
+          align = (g * m).sum()
+        st['lr_scale'] = (st['lr_scale'] *
+                          torch.exp(p_.scale_learning_rate * align)
+                          ).clamp(p_.min_scale, p_.max_scale)
+        # per-weight gain EG update.
+        if p_.use_signs:
+          upd = torch.exp(p_.gain_learning_rate * torch.sign(g) *
+                          torch.sign(gbar))
+        else:
+          corr = gbar / (1.0 - p_.beta ** max(st['counter'] - 1, 1))
+          upd = torch.exp(p_.gain_learning_rate * g * corr)
+        gain.mul_(upd).clamp_(p_.min_gain, p_.max_gain)
+        m.mul_(p_.momentum).add_(lr * gain * g)
+        gbar.mul_(p_.beta).add_(g, alpha=1.0 - p_.beta)
+        w.add_((-st['lr_scale'] * m).to(w.dtype))
+    return loss

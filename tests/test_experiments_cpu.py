@@ -143,3 +143,21 @@ def test_magnitude_pruner_schedule_and_masks():
   pruner.ApplyMasks()
   mask = pruner.masks['weight']
   assert (model.weight.detach()[~mask] == 0).all()
+
+
+def test_egdd_optimizer_converges_and_clips():
+  import torch
+  from lingvo_amd.core import optimizer_experiments as oe
+  torch.manual_seed(0)
+  w = torch.nn.Parameter(torch.tensor([4.0, -3.0]))
+  opt_p = oe.EGDD.Params().Set(name='egdd', momentum=0.9)
+  opt = opt_p.Instantiate().CreateTorchOptimizer([w], lr=0.05)
+  for _ in range(200):
+    opt.zero_grad()
+    loss = (w ** 2).sum()
+    loss.backward()
+    opt.step()
+  assert float((w ** 2).sum()) < 1e-2
+  st = opt.state[w]
+  assert float(st['gain'].max()) <= 100.0
+  assert 0.1 <= float(st['lr_scale']) <= 10.0
