@@ -147,6 +147,30 @@ def test_act_fused_dropout_matches_composed():
 
 
 @gpu
+def test_s2d_kernels_match_torch_fallback():
+  from lingvo_amd.ops import _loader
+  ext = _loader.get_ext(required=True)
+  torch.manual_seed(4)
+  B, H, W, C = 3, 10, 6, 16
+  x = torch.randn(B, H, W, C, device='cuda', dtype=torch.bfloat16)
+  X = ext.s2d_fwd(x)
+  Ho, Wo = H // 2, W // 2
+  blocks = [(0, 1), (1, 1), (1, 0), (0, 0)]
+  ref = x.new_zeros(B, Ho + 1, Wo + 1, 4 * C)
+  for i, (a, b) in enumerate(blocks):
+    ref[:, 1:, 1:, i * C:(i + 1) * C] = x[:, a::2, b::2, :]
+  assert torch.equal(X, ref)
+  # inverse round-trips back to x.
+  assert torch.equal(ext.s2d_inv(X, C), x)
+  # pad_scatter zero border + copy.
+  d = torch.randn(B, Ho, Wo, 8, device='cuda', dtype=torch.bfloat16)
+  P = ext.pad_scatter(d)
+  assert torch.equal(P[:, 1:, 1:, :], d)
+  assert float(P[:, 0].abs().sum()) == 0.0
+  assert float(P[:, :, 0].abs().sum()) == 0.0
+
+
+@gpu
 def test_topk_rows_matches_torch():
   from lingvo_amd.ops import _loader
   ext = _loader.get_ext(required=True)
