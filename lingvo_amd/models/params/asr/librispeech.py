@@ -105,3 +105,20 @@ class Librispeech960BaseGrapheme(Librispeech960Base):
   """Grapheme-target LAS (reference librispeech.py:156)."""
 
   VOCAB = 76
+
+
+@registry.RegisterSingleTaskModel
+class Librispeech960Wpm(Librispeech960Base):
+  """WPM-target LAS (reference librispeech.py:239
+  Librispeech960Wpm: 16k word pieces, 96-dim embeddings, target len
+  140). Synthetic inputs mirror the WPM shapes; plug a vocab file into
+  tokenizers.WpmTokenizer for real data."""
+
+  VOCAB = 16328
+  TARGET_LEN = 140
+
+  def Task(self):
+    p = super().Task()
+    p.name = 'librispeech_las_wpm'
+    p.decoder.Set(vocab_size=self.VOCAB, emb_dim=96)
+    return p
