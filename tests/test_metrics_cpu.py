@@ -109,3 +109,23 @@ def test_record_batcher_negative_key_drops(tmp_path):
   assert all(v % 2 == 0 for v in vals)
   rb.Stop()
   y.stop()
+
+
+def test_bleu_scorer_and_unsegmenter():
+  from lingvo_amd.core import scorers
+  s = scorers.BleuScorer()
+  s.AddSentence('the cat sat on the mat', 'the cat sat on the mat')
+  assert abs(s.ComputeOverallScore() - 1.0) < 1e-9
+  s2 = scorers.BleuScorer()
+  s2.AddSentence('the cat sat on the mat', 'a dog ran in a park yes')
+  assert s2.ComputeOverallScore() == 0.0
+  # Partial overlap is between 0 and 1, and shorter hyps get a brevity
+  # penalty.
+  s3 = scorers.BleuScorer()
+  s3.AddSentence('the cat sat on the mat', 'the cat sat on the')
+  v = s3.ComputeOverallScore()
+  assert 0.0 < v < 1.0
+  u = scorers.Unsegmenter('bpe')
+  assert u('th@@ e ca@@ t') == 'the cat'
+  w = scorers.Unsegmenter('wpm')
+  assert w('▁the ▁ca t') == 'the cat'
