@@ -35,7 +35,10 @@ class FRNN(BaseLayer):
     p = self.p
     b, t, _ = inputs.shape
     x = inputs.transpose(0, 1)  # time-major for the scan
-    pad = (paddings.transpose(0, 1).unsqueeze(-1)
+    # Cast paddings to the activation dtype: the cells' padded-state
+    # carry (c1*(1-pad) + c0*pad) would otherwise promote the whole
+    # recurrent state to fp32 after the first step.
+    pad = (paddings.transpose(0, 1).unsqueeze(-1).to(inputs.dtype)
            if paddings is not None else torch.zeros(
                t, b, 1, dtype=inputs.dtype, device=inputs.device))
     if p.reverse:

@@ -328,3 +328,19 @@ def test_asr_beam_search_decode():
                               max_steps=6)
   s = out4.topk_scores
   assert bool((s[:, :-1] >= s[:, 1:]).all())  # ranked
+
+
+def test_las_models_train_in_bf16():
+  """Regression: fp32 paddings used to promote the biLSTM recurrent
+  state to fp32 on step 2 (dtype-mismatch crash in bf16 training)."""
+  import torch
+  from lingvo_amd.core import registry
+  registry.ImportAllParams()
+  p = registry.GetParams('asr.librispeech.Librispeech960Base', 'Train')
+  p.input.batch_size = 2
+  p.input.frame_len = 40
+  task = p.Instantiate().GetTask()
+  task.to(torch.bfloat16)
+  m = task.TrainStep(task.input_generator.GetPreprocessedInputBatch())
+  loss = float(m[task.learners[0].p.loss_name][0].detach())
+  assert loss == loss  # not NaN
