@@ -325,7 +325,8 @@ def run_bench(args, world):
         'config': {
             'model': ('Conformer-L (17 blocks, d=512, h=8, kernel 32) + '
                       'LSTM attention decoder'
-                      if 'Conformer' in args.model else args.model),
+                      if args.model.endswith('WpmConformerL')
+                      else args.model),
             'global_batch': global_batch,
             'seq_len': (model_p.input.frame_len
                         if 'frame_len' in model_p.input
