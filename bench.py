@@ -328,9 +328,10 @@ def run_bench(args, world):
                       if args.model.endswith('WpmConformerL')
                       else args.model),
             'global_batch': global_batch,
-            'seq_len': (model_p.input.frame_len
-                        if 'frame_len' in model_p.input
-                        else model_p.input.Get('seq_len')),
+            'seq_len': next(
+                (model_p.input.Get(k) for k in
+                 ('frame_len', 'seq_len', 'src_len')
+                 if k in model_p.input), None),
             'parallelism': f'dp{world}',
             'final_loss': round(loss, 4),
             'step_mode': 'hipgraph' if graphed is not None else 'eager',
