@@ -6,6 +6,9 @@ import os
 import subprocess
 import sys
 
+sys.path.insert(0, os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))))
+
 CASES = [
     {'batch': 8, 'frame_len': 1200, 'layers': 4, 'dropout': 0.2},
     {'batch': 128, 'frame_len': 240, 'layers': 4, 'dropout': 0.2},
