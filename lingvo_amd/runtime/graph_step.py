@@ -18,6 +18,8 @@ from __future__ import annotations
 
 from typing import Optional
 
+import os
+
 import torch
 
 from lingvo_amd.core import py_utils
@@ -63,13 +65,33 @@ class GraphedTrainStep:
         grad_sync.Finalize()
       return metrics
 
+    # HIP segfaults instantiating/replaying graphs beyond roughly
+    # 30-50k nodes (measured: the LAS 4x-biLSTM scan at T=300 steps x 8
+    # directions crashes; half the steps or a quarter of the layers
+    # captures fine — tools/las_graph_diag.py). Count kernel launches
+    # in one profiled warmup step and decline capture cleanly above
+    # the threshold so callers' eager fallbacks engage instead of a
+    # SIGSEGV.
+    MAX_CAPTURE_KERNELS = int(os.environ.get(
+        'LINGVO_GRAPH_MAX_KERNELS', '30000'))
+
     def warmup_and_capture(with_sync: bool):
       side = torch.cuda.Stream()
       side.wait_stream(torch.cuda.current_stream())
       with torch.cuda.stream(side):
-        for _ in range(warmup_iters):
+        for _ in range(max(warmup_iters - 1, 0)):
+          self.metrics = fwd_bwd(with_sync)
+        from torch.profiler import ProfilerActivity, profile
+        with profile(activities=[ProfilerActivity.CUDA]) as prof:
           self.metrics = fwd_bwd(with_sync)
       torch.cuda.current_stream().wait_stream(side)
+      n_kernels = sum(1 for e in prof.events()
+                      if e.device_type != torch.autograd.DeviceType.CPU)
+      if n_kernels > MAX_CAPTURE_KERNELS:
+        raise RuntimeError(
+            f'step launches ~{n_kernels} kernels; hipGraph capture '
+            f'above {MAX_CAPTURE_KERNELS} is unstable '
+            '(LINGVO_GRAPH_MAX_KERNELS to override)')
       graph = torch.cuda.CUDAGraph()
       with torch.cuda.graph(graph):
         self.metrics = fwd_bwd(with_sync)
