@@ -744,3 +744,32 @@ def test_causal_pooling_layer():
                                          left_context=-1).Instantiate()
   om, _ = m.FProp(m.theta, x, pad)
   assert torch.allclose(om[0, 3], x[0].max(dim=0).values)
+
+
+def test_lstm_frnn_matches_plain_frnn():
+  """Hoisted-projection LstmFRNN == plain FRNN scan, fwd + grads
+  (reference lstm_frnn_layer.py)."""
+  import torch
+  from lingvo_amd.layers import lstm_frnn_layer as lf
+  from lingvo_amd.layers import rnn_cell, rnn_layers
+  torch.manual_seed(0)
+  cp = rnn_cell.LSTMCellSimple.Params().Set(
+      num_input_nodes=6, num_output_nodes=4, forget_gate_bias=1.0)
+  plain = rnn_layers.BidirectionalFRNN.Params().Set(
+      name='p', fwd=cp.Copy(), bak=cp.Copy()).Instantiate()
+  fast = lf.BidirectionalLstmFRNN.Params().Set(
+      name='f', fwd=cp.Copy(), bak=cp.Copy()).Instantiate()
+  fast.load_state_dict(plain.state_dict())
+  x = torch.randn(3, 7, 6)
+  pad = torch.zeros(3, 7)
+  pad[1, 5:] = 1.0
+  a = plain.FProp(plain.theta, x, pad)
+  b = fast.FProp(fast.theta, x, pad)
+  assert torch.allclose(a, b, atol=1e-5)
+  x1 = x.clone().requires_grad_(True)
+  x2 = x.clone().requires_grad_(True)
+  plain.FProp(plain.theta, x1, pad).square().sum().backward()
+  fast.FProp(fast.theta, x2, pad).square().sum().backward()
+  assert torch.allclose(x1.grad, x2.grad, atol=1e-4)
+  assert torch.allclose(plain.fwd_rnn.cell.vars.wm.grad,
+                        fast.fwd_rnn.cell.vars.wm.grad, atol=1e-4)

@@ -147,6 +147,27 @@ def test_act_fused_dropout_matches_composed():
 
 
 @gpu
+def test_lstm_frnn_fused_gates_matches_fp32():
+  """GPU bf16 LstmFRNN (fused K11 gate path) vs CPU fp32 reference."""
+  import torch
+  from lingvo_amd.layers import lstm_frnn_layer as lf
+  torch.manual_seed(1)
+  cp = lf.LSTMCellSimpleExt.Params().Set(
+      num_input_nodes=16, num_output_nodes=8, forget_gate_bias=1.0)
+  layer = lf.LstmFRNN.Params().Set(name='f', cell=cp).Instantiate()
+  x = torch.randn(4, 12, 16)
+  pad = torch.zeros(4, 12)
+  pad[2, 9:] = 1.0
+  ref, _ = layer.FProp(layer.theta, x, pad)
+  gpu_layer = layer.to('cuda').to(torch.bfloat16)
+  out, _ = gpu_layer.FProp(gpu_layer.theta, x.cuda().bfloat16(),
+                           pad.cuda())
+  assert (out.float().cpu() - ref).abs().max() < 0.05
+  out.square().sum().backward()
+  assert gpu_layer.cell.vars.wm.grad is not None
+
+
+@gpu
 def test_s2d_kernels_match_torch_fallback():
   from lingvo_amd.ops import _loader
   ext = _loader.get_ext(required=True)
