@@ -159,7 +159,11 @@ def test_lstm_frnn_fused_gates_matches_fp32():
   pad = torch.zeros(4, 12)
   pad[2, 9:] = 1.0
   ref, _ = layer.FProp(layer.theta, x, pad)
-  gpu_layer = layer.to('cuda').to(torch.bfloat16)
+  gpu_p = lf.LstmFRNN.Params().Set(name='g', cell=cp.Copy())
+  gpu_p.fprop_dtype = torch.bfloat16
+  gpu_layer = gpu_p.Instantiate()
+  gpu_layer.load_state_dict(layer.state_dict())
+  gpu_layer = gpu_layer.to('cuda').to(torch.bfloat16)
   out, _ = gpu_layer.FProp(gpu_layer.theta, x.cuda().bfloat16(),
                            pad.cuda())
   assert (out.float().cpu() - ref).abs().max() < 0.05
