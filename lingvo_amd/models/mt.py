@@ -293,18 +293,19 @@ class RnmtEncoder(BaseLayer):
   def __init__(self, params):
     super().__init__(params)
     p = self.p
-    from lingvo_amd.layers import rnn_cell
-    from lingvo_amd.layers import rnn_layers
+    from lingvo_amd.layers import lstm_frnn_layer
     self.CreateChild('emb', lingvo_layers.EmbeddingLayer.Params().Set(
         vocab_size=p.vocab_size, embedding_dim=p.model_dim,
         scale_sqrt_depth=True))
     half = p.model_dim // 2
     layer_ps = []
     for i in range(p.num_lstm_layers):
-      cell = rnn_cell.LSTMCellSimple.Params().Set(
+      # Hoisted-projection biLSTM (see layers/lstm_frnn_layer.py).
+      cell = lstm_frnn_layer.LSTMCellSimpleExt.Params().Set(
           num_input_nodes=p.model_dim, num_output_nodes=half)
-      layer_ps.append(rnn_layers.BidirectionalFRNN.Params().Set(
-          name=f'blstm_{i}', fwd=cell.Copy(), bak=cell.Copy()))
+      layer_ps.append(
+          lstm_frnn_layer.BidirectionalLstmFRNN.Params().Set(
+              name=f'blstm_{i}', fwd=cell.Copy(), bak=cell.Copy()))
     self.CreateChildren('rnn', layer_ps)
 
   def FProp(self, theta: NestedMap, ids: torch.Tensor,
