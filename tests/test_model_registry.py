@@ -91,3 +91,29 @@ def test_waymo_pillars_small_train_step():
   task = model.GetTask()
   m = task.TrainStep(task.GetInputBatch())
   assert torch.isfinite(m['loss'][0])
+
+
+def test_base_model_params_contract():
+  from lingvo_amd.core import base_model_params as bmp
+  from lingvo_amd.models import mnist as mnist_model
+
+  class MyParams(bmp.SingleTaskModelParams):
+
+    def Train(self):
+      return mnist_model.FakeMnistData.Params().Set(batch_size=4,
+                                                    name='train')
+
+    def Task(self):
+      p = mnist_model.ModelV1.Params().Set(
+          name='m', hidden_dim=8, filter_shapes=[(3, 3, 1, 2)])
+      p.softmax.num_classes = 10
+      return p
+
+  mp = MyParams()
+  assert mp.GetDatasetParams('Train').batch_size == 4
+  import pytest
+  with pytest.raises(bmp.DatasetError):
+    mp.GetDatasetParams('Nope')
+  model_p = mp.Model()
+  model = model_p.Instantiate()
+  assert model.GetTask().p.name == 'm'
