@@ -1,5 +1,6 @@
 """Executor/programs, early stop, inference export/predictor tests."""
 
+import re
 import json
 import os
 
@@ -521,3 +522,40 @@ def test_export_metrics_hook(tmp_path):
   tr.SetExportMetricsFn(lambda **kw: exported.append(kw))
   tr.Start()
   assert exported and 'loss' in exported[0] and 'step' in exported[0]
+
+
+def test_warm_start_rules_remap_and_restore(tmp_path):
+  """init_from_checkpoint_rules: regex-remapped partial restore
+  (reference checkpointer init rules)."""
+  import torch
+  from lingvo_amd.core.checkpointer import Checkpointer
+  from lingvo_amd.models import mnist as mnist_model
+  from lingvo_amd.core.base_model import SingleTaskModel
+
+  def build(name, seed):
+    p = mnist_model.ModelV1.Params().Set(
+        name=name, hidden_dim=8, filter_shapes=[(3, 3, 1, 2)],
+        random_seed=seed)
+    p.softmax.num_classes = 10
+    p.input = mnist_model.FakeMnistData.Params().Set(batch_size=4)
+    return SingleTaskModel.Params(p).Instantiate()
+
+  src = build('a', 1)
+  ckpt_path = tmp_path / 'src.pt'
+  torch.save({'model': src.state_dict(), 'step': 7}, ckpt_path)
+
+  dst = build('b', 2)
+  # Pick one parameter name; remap identity.
+  names = [n for n, _ in dst.named_parameters()]
+  target = names[0]
+  before = dict(dst.named_parameters())[target].detach().clone()
+  cp = Checkpointer.Params()
+  cp.init_from_checkpoint_rules = {
+      str(ckpt_path): [(re.escape(target), target)]}
+  ck = Checkpointer(cp, str(tmp_path / 'train'), dst)
+  # Warm-start applies on Restore() when no checkpoint exists yet.
+  assert ck.Restore() is None
+  after = dict(dst.named_parameters())[target].detach()
+  want = dict(src.named_parameters())[target].detach()
+  assert torch.allclose(after, want)
+  assert not torch.allclose(after, before)
