@@ -261,8 +261,10 @@ class TransformerXLAttention(MultiHeadedAttention):
   with learned per-head biases u, v and sinusoidal relative embeddings
   r projected per head. `memory` (the previous segment's input hidden
   states, no grad) is prepended on the key/value side, giving XL-style
-  segment recurrence. The O(T^2) gather below is the CPU oracle; the
-  rel-shift fused form inside the flash kernel is the round-2 GPU path.
+  segment recurrence. This composed-einsum form runs on CPU and GPU but
+  materializes the [B, N, T, L] rel-logits; the in-kernel rel-shift
+  variant (extra MFMA against a staged R tile) is a round-3 item
+  (docs/ROADMAP.md #6).
   """
 
   @classmethod
