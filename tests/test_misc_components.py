@@ -862,3 +862,24 @@ def test_numeric_gradient_matches_autograd():
   test_utils.AssertAllClose(num, xg.grad, rtol=2e-2, atol=2e-3)
   # Golden scalar with fixed seed.
   test_utils.CompareToGoldenSingleFloat(float(f(x)), float(f(x)))
+
+
+def test_dropout_cpu_fallback_act_scale_padding_composition():
+  """CPU fallback of the fused dropout must equal the manual
+  composition act -> scale -> padding-mask -> mask/keep."""
+  import torch
+  from lingvo_amd.ops import dropout as dropout_ops
+  torch.manual_seed(0)
+  x = torch.randn(3, 4, 8)
+  pad = torch.zeros(3, 4)
+  pad[1, 2:] = 1.0
+  keep = 0.8
+  y = dropout_ops.dropout(x, keep, seed=11, act='SWISH', scale=0.5,
+                          paddings=pad)
+  a = torch.nn.functional.silu(x) * 0.5 * \
+      (1.0 - pad)[:, :, None]
+  g = torch.Generator(device='cpu')
+  g.manual_seed(11)
+  mask = (torch.rand(a.shape, generator=g) < keep)
+  want = a * mask.to(a.dtype) / keep
+  assert torch.allclose(y, want, atol=1e-6)
