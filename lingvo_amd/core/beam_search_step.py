@@ -62,6 +62,11 @@ class DoneHyp:
   step: int
 
 
+# Test knob: force the generic merge-capable loop even when the
+# vectorized fast path applies (equivalence fuzzing).
+_DISABLE_FAST_PATH = False
+
+
 @dataclass
 class BeamSearchState:
   """Mutable search state across steps (the reference op's in/out
@@ -191,6 +196,90 @@ def BeamSearchStep(scores: torch.Tensor, state: BeamSearchState, t: int,
   merged: List[List[Hyp]] = [[] for _ in range(b)]
   eos_done: List[Optional[Tuple[Hyp, int]]] = [None] * n
 
+  # Vectorized fast path (no path merging, no epsilon): candidate
+  # selection and table writes as tensor ops; only TERMINATED hyps
+  # materialize python Hyp objects. Identical semantics to the generic
+  # loop below (sort key (-global, word, hyp); EOS entries never join
+  # the live pool; t==0 admits only hyp 0 per beam).
+  if not merge_paths and eoc_id < 0 and not _DISABLE_FAST_PATH:
+    k2 = top_vals.shape[1]
+    vals = top_vals.clone()                        # [N, K2] global
+    words = top_idx.clone()                        # [N, K2]
+    if force_eos_in_top_k:
+      no_eos = (words != eos_id).all(dim=1)
+      if bool(no_eos.any()):
+        rows = no_eos.nonzero(as_tuple=True)[0]
+        words[rows, -1] = eos_id
+        vals[rows, -1] = eos_global[rows]
+    locals_ = vals - cum.unsqueeze(1)
+    active = torch.tensor(
+        [not skip_beam[i % b] and (t > 0 or i < b) for i in range(n)])
+    # --- EOS terminations: first qualifying eos entry per row.
+    best_global = vals[:, 0]
+    thr = best_global - valid_eos_max_logit_delta
+    is_eos = words == eos_id
+    ok = torch.ones_like(vals, dtype=torch.bool) if last_step_force \
+        else ((vals > thr.unsqueeze(1)) &
+              (locals_ > local_eos_threshold))
+    eos_ok = is_eos & ok & active.unsqueeze(1)
+    has_eos = eos_ok.any(dim=1)
+    first_eos = eos_ok.float().argmax(dim=1)
+    for i in has_eos.nonzero(as_tuple=True)[0].tolist():
+      c = int(first_eos[i])
+      eos_done[i] = (Hyp(i % b, i, eos_id, float(locals_[i, c]),
+                         float(vals[i, c]), ()), eos_id)
+    # --- Live pool per beam: layout (j asc, col asc) == python order.
+    NEG = -1e30
+    pool_ok = (~is_eos) & active.unsqueeze(1)      # [N, K2]
+    # [N, K2] -> [K, b, K2] -> [b, K*K2]
+    def to_pool(x):
+      return x.reshape(k, b, k2).permute(1, 0, 2).reshape(b, k * k2)
+    pg = to_pool(vals.clone())
+    pw = to_pool(words.clone())
+    pl = to_pool(locals_.clone())
+    pm = to_pool(pool_ok)
+    hyp_ids = torch.arange(n).unsqueeze(1).expand(n, k2)
+    ph = to_pool(hyp_ids.clone())
+    pg = torch.where(pm, pg, torch.full_like(pg, NEG))
+    big = torch.iinfo(torch.long).max
+    pw = torch.where(pm, pw, torch.full_like(pw, big))
+    ph = torch.where(pm, ph, torch.full_like(ph, big))
+    #
+
+    # Lexicographic stable sort: minor keys first.
+    o1 = torch.argsort(ph, dim=1, stable=True)
+    pg, pw, pl, ph, pm = (x.gather(1, o1) for x in (pg, pw, pl, ph, pm))
+    o2 = torch.argsort(pw, dim=1, stable=True)
+    pg, pw, pl, ph, pm = (x.gather(1, o2) for x in (pg, pw, pl, ph, pm))
+    o3 = torch.argsort(-pg, dim=1, stable=True)
+    pg, pw, pl, ph, pm = (x.gather(1, o3) for x in (pg, pw, pl, ph, pm))
+
+    gather = torch.arange(n, dtype=torch.long)
+    new_cum = cum.clone()
+    for beam_id in range(b):
+      if skip_beam[beam_id]:
+        continue
+      valid = int(pm[beam_id].sum())
+      for j in range(k):
+        i = j * b + beam_id
+        if j < valid:
+          state.hyps[t, i] = int(pw[beam_id, j])
+          state.prev_hyps[t, i] = int(ph[beam_id, j])
+          state.step_scores[t, i] = float(pl[beam_id, j])
+          new_cum[i] = float(pg[beam_id, j])
+          gather[i] = int(ph[beam_id, j])
+        else:
+          state.hyps[t, i] = eos_id
+          state.prev_hyps[t, i] = beam_id
+          state.step_scores[t, i] = -1e30
+          new_cum[i] = -1e30
+          gather[i] = beam_id
+    state.cumulative_scores = new_cum
+    _RecordTerminations(state, eos_done, t, b, k, n, merge_paths,
+                        beam_size, ensure_full_beam, new_cum)
+    return gather
+
+
   def _insert(beam: List[Hyp], h: Hyp):
     if merge_paths:
       for i, other in enumerate(beam):
@@ -266,8 +355,16 @@ def BeamSearchStep(scores: torch.Tensor, state: BeamSearchState, t: int,
         new_cum[i] = -1e30
         gather[i] = beam_id
   state.cumulative_scores = new_cum
+  _RecordTerminations(state, eos_done, t, b, k, n, merge_paths,
+                      beam_size, ensure_full_beam, new_cum)
+  return gather
 
-  # Record terminations + update best scores.
+
+def _RecordTerminations(state, eos_done, t, b, k, n, merge_paths,
+                        beam_size, ensure_full_beam, new_cum):
+  """Appends DoneHyps, updates best scores and beam_done/all_done
+  (reference UpdateAllDone :845). Shared by the vectorized fast path
+  and the generic merge-capable loop."""
   for i in range(n):
     if eos_done[i] is None:
       continue
@@ -285,7 +382,6 @@ def BeamSearchStep(scores: torch.Tensor, state: BeamSearchState, t: int,
     state.done_hyps.append(DoneHyp(beam_id, ids, step_scores,
                                    h.global_score, t))
 
-  # all_done / beam_done (reference UpdateAllDone :845).
   for beam_id in range(b):
     if state.beam_done[beam_id]:
       continue
@@ -298,7 +394,6 @@ def BeamSearchStep(scores: torch.Tensor, state: BeamSearchState, t: int,
     state.beam_done[beam_id] = all(
         s < state.best_scores[beam_id] - beam_size for s in live)
   state.all_done = all(state.beam_done)
-  return gather
 
 
 def _Higher(x: Hyp, y: Hyp) -> bool:

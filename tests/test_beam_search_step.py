@@ -182,3 +182,50 @@ def test_helper_end_to_end_prefers_high_prob_sequence():
   # Scores sorted descending.
   sc = out.topk_scores[0]
   assert sc[0] >= sc[1] >= sc[2]
+
+
+def test_fast_path_matches_generic_loop_fuzz():
+  """The vectorized fast path must evolve the EXACT same state as the
+  generic merge-capable loop over random multi-step searches."""
+  import torch
+  torch.manual_seed(0)
+  for trial in range(8):
+    b, k, v = 2 + trial % 2, 3 + trial % 3, 20
+    n = b * k
+    force = trial % 2 == 0
+    torch.manual_seed(100 + trial)
+    score_seq = [torch.log_softmax(torch.randn(n, v), dim=-1)
+                 for _ in range(4)]
+
+    def run(disable_fast):
+      bss._DISABLE_FAST_PATH = disable_fast
+      try:
+        st = bss.BeamSearchState.Init(num_beams=b, k=k, max_steps=4)
+        gathers = []
+        for t, sc in enumerate(score_seq):
+          g = bss.BeamSearchStep(
+              sc.clone(), st, t, eos_id=2,
+              valid_eos_max_logit_delta=4.0,
+              force_eos_in_top_k=force,
+              ensure_full_beam=(trial % 3 == 0))
+          gathers.append(g.clone())
+        return st, gathers
+      finally:
+        bss._DISABLE_FAST_PATH = False
+
+    st_f, g_f = run(False)
+    st_g, g_g = run(True)
+    assert torch.equal(st_f.hyps, st_g.hyps), trial
+    assert torch.equal(st_f.prev_hyps, st_g.prev_hyps), trial
+    assert torch.allclose(st_f.cumulative_scores,
+                          st_g.cumulative_scores), trial
+    assert torch.allclose(st_f.step_scores, st_g.step_scores,
+                          atol=1e-5), trial
+    assert st_f.best_scores == st_g.best_scores, trial
+    assert st_f.beam_done == st_g.beam_done, trial
+    assert len(st_f.done_hyps) == len(st_g.done_hyps), trial
+    for a, c in zip(st_f.done_hyps, st_g.done_hyps):
+      assert a.ids == c.ids and a.beam_id == c.beam_id, trial
+      assert abs(a.global_score - c.global_score) < 1e-5, trial
+    for a, c in zip(g_f, g_g):
+      assert torch.equal(a, c), trial
