@@ -206,6 +206,10 @@ def test_fast_path_matches_generic_loop_fuzz():
           g = bss.BeamSearchStep(
               sc.clone(), st, t, eos_id=2,
               valid_eos_max_logit_delta=4.0,
+              local_eos_threshold=(-100.0 if trial % 2 else -3.0),
+              beam_size=(3.0 if trial % 2 else 1.0),
+              force_eos_in_last_step=(trial % 4 == 1),
+              beam_independence=(trial % 4 == 2),
               force_eos_in_top_k=force,
               ensure_full_beam=(trial % 3 == 0))
           gathers.append(g.clone())

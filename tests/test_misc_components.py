@@ -883,3 +883,18 @@ def test_dropout_cpu_fallback_act_scale_padding_composition():
   mask = (torch.rand(a.shape, generator=g) < keep)
   want = a * mask.to(a.dtype) / keep
   assert torch.allclose(y, want, atol=1e-6)
+
+
+def test_prof_summary_tool(tmp_path):
+  import subprocess
+  import sys
+  csv = tmp_path / 'k.csv'
+  csv.write_text(
+      '"Name","Calls","TotalDurationNs","AverageNs","Percentage"\n'
+      '"kern_a",10,5000000,500000,62.5\n'
+      '"kern_b",5,3000000,600000,37.5\n')
+  r = subprocess.run(
+      [sys.executable, 'tools/prof_summary.py', str(csv)],
+      capture_output=True, text=True)
+  assert r.returncode == 0, r.stderr
+  assert 'kern_a' in r.stdout and 'kern_b' in r.stdout
