@@ -226,3 +226,17 @@ def test_lm_loss_decreases():
       first = float(m['loss'][0])
   last = float(m['loss'][0])
   assert last < first * 0.5, (first, last)
+
+
+def test_trainer_cli_list_models_and_inspect(tmp_path, capsys):
+  from lingvo_amd.runtime import trainer as trainer_cli
+  trainer_cli.main(['--list_models'])
+  out = capsys.readouterr().out
+  assert 'image.mnist.LeNet5' in out
+  assert 'asr.librispeech.Librispeech960WpmConformerL' in out
+  assert 'lm.one_billion_wds.OneBWdsTransformerLm' in out
+  # inspect_model prints a parameter summary without training.
+  trainer_cli.main(['--model', 'image.mnist.LeNet5',
+                    '--logdir', str(tmp_path), '--mode', 'inspect_model'])
+  out = capsys.readouterr().out
+  assert 'params' in out.lower() or 'total' in out.lower()
