@@ -199,11 +199,12 @@ def run_bench(args, world):
 
   _phase('building model')
   model_p = registry.GetParams(args.model, 'Train')
-  model_p.input.batch_size = args.batch
   if not has_gpu:  # CPU smoke of the bench harness only
+    args.batch = min(args.batch, 8)
     model_p.task.fprop_dtype = torch.float32
     model_p.input.frame_len = 80
     model_p.task.encoder.num_layers = 1
+  model_p.input.batch_size = args.batch
   model_p.task.random_seed = 1234
   model = model_p.Instantiate().to(device)
   task = model.GetTask()
